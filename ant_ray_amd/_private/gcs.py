@@ -1,0 +1,660 @@
+"""GCS — the cluster-global control plane.
+
+Role parity: reference gcs_server (src/ray/gcs/gcs_server.cc:270 DoStart wires
+NodeManager/ResourceManager/HealthCheck/JobManager/PlacementGroupManager/
+ActorManager/WorkerManager + KV + pubsub). Same responsibilities, one asyncio
+process, msgpack-RPC transport (see protocol.py):
+
+  * node table + health (heartbeats; dead after miss window — parity with
+    gcs_health_check_manager.h:45, 3s period / 5 misses),
+  * actor table with scheduling (lease a worker from a raylet, push the
+    creation task, restart per max_restarts — parity with
+    gcs_actor_manager.h:94 + gcs_actor_scheduler.h:108),
+  * named actors, job ids, cluster KV (function table lives here —
+    gcs_kv_manager.cc / gcs_function_manager.h),
+  * placement groups (reserve bundles on raylets, 2PC-lite — parity with
+    gcs_placement_group_manager.h:55),
+  * pubsub channels pushed over registered connections (src/ray/pubsub/).
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import logging
+import os
+import time
+from typing import Any, Dict, List, Optional, Tuple
+
+from ant_ray_amd._private import protocol
+from ant_ray_amd._private.protocol import Connection
+
+logger = logging.getLogger("antray.gcs")
+
+# Actor states (parity with reference rpc::ActorTableData states)
+DEPENDENCIES_UNREADY = "DEPENDENCIES_UNREADY"
+PENDING_CREATION = "PENDING_CREATION"
+ALIVE = "ALIVE"
+RESTARTING = "RESTARTING"
+DEAD = "DEAD"
+
+HEARTBEAT_PERIOD_S = 1.0
+HEARTBEAT_MISS_LIMIT = 5
+
+
+class NodeInfo:
+    def __init__(self, node_id, addr, resources, store_path, object_store_bytes=0):
+        self.node_id: bytes = node_id
+        self.addr: Tuple[str, int] = tuple(addr)
+        self.resources_total: Dict[str, float] = dict(resources)
+        self.resources_available: Dict[str, float] = dict(resources)
+        self.store_path: str = store_path
+        self.object_store_bytes = object_store_bytes
+        self.alive = True
+        self.last_heartbeat = time.monotonic()
+        self.conn: Optional[Connection] = None
+
+    def view(self):
+        return {
+            "node_id": self.node_id,
+            "addr": list(self.addr),
+            "resources_total": self.resources_total,
+            "resources_available": self.resources_available,
+            "store_path": self.store_path,
+            "alive": self.alive,
+        }
+
+
+class ActorInfo:
+    def __init__(self, actor_id, owner, name, namespace, opts, create_payload):
+        self.actor_id: bytes = actor_id
+        self.owner: Optional[bytes] = owner
+        self.name: str = name or ""
+        self.namespace: str = namespace or ""
+        self.opts: Dict[str, Any] = opts or {}
+        self.create_payload = create_payload  # msgpack-able dict to push_task
+        self.state = PENDING_CREATION
+        self.addr: Optional[Tuple[str, int]] = None
+        self.worker_id: Optional[bytes] = None
+        self.node_id: Optional[bytes] = None
+        self.num_restarts = 0
+        self.death_cause = ""
+        self.pending_waiters: List[asyncio.Future] = []
+
+    def view(self):
+        return {
+            "actor_id": self.actor_id,
+            "state": self.state,
+            "addr": list(self.addr) if self.addr else None,
+            "worker_id": self.worker_id,
+            "node_id": self.node_id,
+            "name": self.name,
+            "namespace": self.namespace,
+            "num_restarts": self.num_restarts,
+            "death_cause": self.death_cause,
+            "max_restarts": int(self.opts.get("max_restarts", 0)),
+        }
+
+
+class PlacementGroupInfo:
+    def __init__(self, pg_id, bundles, strategy, name=""):
+        self.pg_id: bytes = pg_id
+        self.bundles: List[Dict[str, float]] = bundles
+        self.strategy = strategy
+        self.name = name
+        self.state = "PENDING"
+        self.bundle_nodes: List[Optional[bytes]] = [None] * len(bundles)
+
+    def view(self):
+        return {
+            "pg_id": self.pg_id,
+            "state": self.state,
+            "strategy": self.strategy,
+            "bundles": self.bundles,
+            "bundle_nodes": self.bundle_nodes,
+            "name": self.name,
+        }
+
+
+class GcsServer:
+    def __init__(self):
+        self.nodes: Dict[bytes, NodeInfo] = {}
+        self.actors: Dict[bytes, ActorInfo] = {}
+        self.named_actors: Dict[Tuple[str, str], bytes] = {}
+        self.pgs: Dict[bytes, PlacementGroupInfo] = {}
+        self.kv: Dict[str, Dict[bytes, bytes]] = {}
+        self.workers: Dict[bytes, Dict[str, Any]] = {}  # worker_id -> info
+        self.jobs: Dict[int, Dict[str, Any]] = {}
+        self._job_counter = 0
+        self._subscribers: Dict[str, List[Connection]] = {}
+        self._worker_conns: Dict[Tuple[str, int], Connection] = {}
+        self._server = None
+        self.port = None
+        self._shutdown = asyncio.Event()
+
+    # ------------------------------------------------------------------ serve
+    async def start(self, host="127.0.0.1", port=0):
+        self._server, self.port = await protocol.serve(self._handle, host, port)
+        asyncio.get_running_loop().create_task(self._health_loop())
+        logger.info("GCS listening on %s:%s", host, self.port)
+        return self.port
+
+    async def _health_loop(self):
+        while not self._shutdown.is_set():
+            await asyncio.sleep(HEARTBEAT_PERIOD_S)
+            now = time.monotonic()
+            for node in list(self.nodes.values()):
+                if node.alive and now - node.last_heartbeat > HEARTBEAT_PERIOD_S * HEARTBEAT_MISS_LIMIT:
+                    logger.warning("node %s missed heartbeats; marking dead", node.node_id.hex()[:8])
+                    await self._on_node_dead(node)
+
+    # --------------------------------------------------------------- dispatch
+    async def _handle(self, conn: Connection, method: str, p: Any):
+        fn = getattr(self, "rpc_" + method, None)
+        if fn is None:
+            raise ValueError(f"unknown GCS method {method}")
+        return await fn(conn, p or {})
+
+    # ------------------------------------------------------------------ nodes
+    async def rpc_register_node(self, conn, p):
+        node = NodeInfo(
+            p["node_id"], tuple(p["addr"]), p.get("resources", {}),
+            p.get("store_path", ""), p.get("object_store_bytes", 0),
+        )
+        node.conn = conn
+        self.nodes[node.node_id] = node
+        conn.session["node_id"] = node.node_id
+        conn.on_close = self._make_node_close_cb(node)
+        await self._publish("NODE", node.node_id, node.view())
+        return {"ok": True}
+
+    def _make_node_close_cb(self, node: NodeInfo):
+        def cb(conn):
+            if node.alive:
+                asyncio.get_running_loop().create_task(self._on_node_dead(node))
+        return cb
+
+    async def _on_node_dead(self, node: NodeInfo):
+        node.alive = False
+        await self._publish("NODE", node.node_id, node.view())
+        # fail actors on that node
+        for actor in list(self.actors.values()):
+            if actor.node_id == node.node_id and actor.state in (ALIVE, PENDING_CREATION, RESTARTING):
+                await self._on_actor_worker_died(actor, f"node {node.node_id.hex()[:8]} died")
+
+    async def rpc_heartbeat(self, conn, p):
+        node = self.nodes.get(p["node_id"])
+        if node:
+            node.last_heartbeat = time.monotonic()
+            if "resources_available" in p:
+                node.resources_available = p["resources_available"]
+        return {"ok": True}
+
+    async def rpc_node_table(self, conn, p):
+        return [n.view() for n in self.nodes.values()]
+
+    async def rpc_get_local_node(self, conn, p):
+        """Pick the node whose store_path the caller should attach to."""
+        ip = p.get("ip")
+        for n in self.nodes.values():
+            if n.alive and (ip is None or n.addr[0] == ip):
+                return n.view()
+        return None
+
+    # ---------------------------------------------------------------- workers
+    async def rpc_register_worker(self, conn, p):
+        wid = p["worker_id"]
+        is_driver = p.get("is_driver", False)
+        job_id = p.get("job_id")
+        if is_driver and job_id is None:
+            self._job_counter += 1
+            job_id = self._job_counter
+            self.jobs[job_id] = {
+                "job_id": job_id,
+                "start_time": time.time(),
+                "driver": wid,
+                "state": "RUNNING",
+            }
+        self.workers[wid] = {
+            "worker_id": wid,
+            "addr": p.get("addr"),
+            "node_id": p.get("node_id"),
+            "is_driver": is_driver,
+            "job_id": job_id,
+            "pid": p.get("pid"),
+        }
+        conn.session["worker_id"] = wid
+        return {"job_id": job_id}
+
+    async def rpc_report_worker_failure(self, conn, p):
+        wid = p["worker_id"]
+        self.workers.pop(wid, None)
+        for actor in list(self.actors.values()):
+            if actor.worker_id == wid and actor.state in (ALIVE, PENDING_CREATION, RESTARTING):
+                await self._on_actor_worker_died(actor, p.get("reason", "worker died"))
+        return {"ok": True}
+
+    # --------------------------------------------------------------------- kv
+    async def rpc_kv_put(self, conn, p):
+        ns = self.kv.setdefault(p.get("ns", ""), {})
+        key = p["key"]
+        if not p.get("overwrite", True) and key in ns:
+            return {"added": False}
+        ns[key] = p["value"]
+        return {"added": True}
+
+    async def rpc_kv_get(self, conn, p):
+        ns = self.kv.get(p.get("ns", ""), {})
+        return {"value": ns.get(p["key"])}
+
+    async def rpc_kv_del(self, conn, p):
+        ns = self.kv.get(p.get("ns", ""), {})
+        existed = ns.pop(p["key"], None) is not None
+        return {"deleted": existed}
+
+    async def rpc_kv_keys(self, conn, p):
+        ns = self.kv.get(p.get("ns", ""), {})
+        prefix = p.get("prefix", b"")
+        return {"keys": [k for k in ns.keys() if k.startswith(prefix)]}
+
+    async def rpc_next_job_id(self, conn, p):
+        self._job_counter += 1
+        return {"job_id": self._job_counter}
+
+    # ----------------------------------------------------------------- pubsub
+    async def rpc_subscribe(self, conn, p):
+        for ch in p["channels"]:
+            subs = self._subscribers.setdefault(ch, [])
+            if conn not in subs:
+                subs.append(conn)
+        return {"ok": True}
+
+    async def _publish(self, channel: str, key: bytes, data: Any):
+        subs = self._subscribers.get(channel, [])
+        dead = []
+        for c in subs:
+            if c.closed:
+                dead.append(c)
+                continue
+            try:
+                await c.notify("pub", {"channel": channel, "key": key, "data": data})
+            except Exception:
+                dead.append(c)
+        for c in dead:
+            try:
+                subs.remove(c)
+            except ValueError:
+                pass
+
+    # ----------------------------------------------------------------- actors
+    async def rpc_create_actor(self, conn, p):
+        actor_id = p["actor_id"]
+        name = p.get("name") or ""
+        namespace = p.get("namespace") or ""
+        if name:
+            key = (namespace, name)
+            existing_id = self.named_actors.get(key)
+            if existing_id is not None:
+                existing = self.actors.get(existing_id)
+                if existing is not None and existing.state != DEAD:
+                    if p.get("get_if_exists"):
+                        return {"existing": True, "actor_id": existing_id}
+                    raise ValueError(f"actor name '{name}' already taken")
+            self.named_actors[key] = actor_id
+        info = ActorInfo(
+            actor_id, p.get("owner"), name, namespace, p.get("opts", {}), p["create_payload"],
+        )
+        self.actors[actor_id] = info
+        asyncio.get_running_loop().create_task(self._schedule_actor(info))
+        return {"existing": False, "actor_id": actor_id}
+
+    def _pick_node(self, resources: Dict[str, float], pg: Optional[dict] = None,
+                   node_affinity: Optional[bytes] = None) -> Optional[NodeInfo]:
+        if pg:
+            pg_info = self.pgs.get(pg["pg_id"])
+            if pg_info and pg_info.state == "CREATED":
+                idx = pg.get("bundle_index", 0)
+                if idx < 0:
+                    idx = 0
+                nid = pg_info.bundle_nodes[idx]
+                node = self.nodes.get(nid)
+                if node and node.alive:
+                    return node
+            return None
+        if node_affinity is not None:
+            node = self.nodes.get(node_affinity)
+            return node if node and node.alive else None
+        # feasibility + best-fit by available CPU fraction
+        best, best_score = None, -1.0
+        for node in self.nodes.values():
+            if not node.alive:
+                continue
+            feasible = all(
+                node.resources_total.get(k, 0) >= v for k, v in resources.items()
+            )
+            if not feasible:
+                continue
+            available = all(
+                node.resources_available.get(k, 0) >= v for k, v in resources.items()
+            )
+            score = 1.0 if available else 0.0
+            score += node.resources_available.get("CPU", 0) / max(
+                node.resources_total.get("CPU", 1), 1
+            )
+            if score > best_score:
+                best, best_score = node, score
+        return best
+
+    async def _schedule_actor(self, info: ActorInfo, delay: float = 0.0):
+        if delay:
+            await asyncio.sleep(delay)
+        opts = info.opts
+        resources = dict(opts.get("resources") or {})
+        resources.setdefault("CPU", float(opts.get("num_cpus", 1)))
+        if opts.get("num_gpus"):
+            resources["GPU"] = float(opts["num_gpus"])
+        pg = opts.get("placement_group")
+        deadline = time.monotonic() + float(opts.get("_scheduling_timeout", 3600.0))
+        while True:
+            node = self._pick_node(resources, pg, opts.get("_node_affinity"))
+            if node is not None and node.conn is not None and not node.conn.closed:
+                try:
+                    lease = await node.conn.call(
+                        "lease_worker",
+                        {
+                            "resources": resources,
+                            "actor_id": info.actor_id,
+                            "runtime_env": opts.get("runtime_env"),
+                            "pg": pg,
+                            "detached": bool(opts.get("lifetime") == "detached"),
+                        },
+                        timeout=120,
+                    )
+                except Exception as e:
+                    logger.warning("actor lease on node failed: %s", e)
+                    lease = None
+                if lease and lease.get("granted"):
+                    await self._push_actor_creation(info, node, lease)
+                    return
+            if time.monotonic() > deadline or self._shutdown.is_set():
+                info.state = DEAD
+                info.death_cause = "scheduling timed out (insufficient resources)"
+                await self._publish("ACTOR", info.actor_id, info.view())
+                self._wake_waiters(info)
+                return
+            await asyncio.sleep(0.2)
+
+    async def _push_actor_creation(self, info: ActorInfo, node: NodeInfo, lease):
+        addr = tuple(lease["addr"])
+        info.worker_id = lease["worker_id"]
+        info.node_id = node.node_id
+        try:
+            wconn = await self._get_worker_conn(addr)
+            payload = dict(info.create_payload)
+            payload["actor_id"] = info.actor_id
+            payload["num_restarts"] = info.num_restarts
+            reply = await wconn.call("push_task", payload, timeout=None)
+        except Exception as e:
+            logger.warning("actor creation push failed: %s", e)
+            await self._on_actor_worker_died(info, f"creation push failed: {e}")
+            return
+        if reply.get("status") == "ok":
+            info.state = ALIVE
+            info.addr = addr
+            await self._publish("ACTOR", info.actor_id, info.view())
+            self._wake_waiters(info)
+        else:
+            info.state = DEAD
+            info.death_cause = reply.get("error", "actor __init__ failed")
+            info.creation_error = reply.get("error_payload")
+            await self._publish("ACTOR", info.actor_id, info.view())
+            self._wake_waiters(info)
+
+    def _wake_waiters(self, info: ActorInfo):
+        for fut in info.pending_waiters:
+            if not fut.done():
+                fut.set_result(info.view())
+        info.pending_waiters.clear()
+
+    async def _get_worker_conn(self, addr: Tuple[str, int]) -> Connection:
+        conn = self._worker_conns.get(addr)
+        if conn is None or conn.closed:
+            conn = await protocol.connect(addr, self._handle, name=f"gcs->worker{addr}")
+            self._worker_conns[addr] = conn
+        return conn
+
+    async def _on_actor_worker_died(self, info: ActorInfo, reason: str):
+        if info.state == DEAD:
+            return
+        max_restarts = int(info.opts.get("max_restarts", 0))
+        if max_restarts == -1 or info.num_restarts < max_restarts:
+            info.num_restarts += 1
+            info.state = RESTARTING
+            info.addr = None
+            info.worker_id = None
+            await self._publish("ACTOR", info.actor_id, info.view())
+            asyncio.get_running_loop().create_task(self._schedule_actor(info, delay=0.1))
+        else:
+            info.state = DEAD
+            info.death_cause = reason
+            info.addr = None
+            await self._publish("ACTOR", info.actor_id, info.view())
+            self._wake_waiters(info)
+
+    async def rpc_get_actor(self, conn, p):
+        info = self.actors.get(p["actor_id"])
+        return info.view() if info else None
+
+    async def rpc_get_actor_by_name(self, conn, p):
+        aid = self.named_actors.get((p.get("namespace") or "", p["name"]))
+        if aid is None:
+            return None
+        info = self.actors.get(aid)
+        return info.view() if info else None
+
+    async def rpc_wait_actor_ready(self, conn, p):
+        info = self.actors.get(p["actor_id"])
+        if info is None:
+            return None
+        if info.state in (ALIVE, DEAD):
+            view = info.view()
+            if info.state == DEAD:
+                view["creation_error"] = getattr(info, "creation_error", None)
+            return view
+        fut = asyncio.get_running_loop().create_future()
+        info.pending_waiters.append(fut)
+        view = await fut
+        if view.get("state") == DEAD:
+            view["creation_error"] = getattr(info, "creation_error", None)
+        return view
+
+    async def rpc_kill_actor(self, conn, p):
+        info = self.actors.get(p["actor_id"])
+        if info is None:
+            return {"ok": False}
+        no_restart = p.get("no_restart", True)
+        if no_restart:
+            info.opts["max_restarts"] = 0
+        if info.addr:
+            try:
+                wconn = await self._get_worker_conn(tuple(info.addr))
+                await wconn.notify("exit_worker", {"reason": "ray.kill"})
+            except Exception:
+                pass
+        # actual state transition happens when the raylet reports worker death
+        return {"ok": True}
+
+    async def rpc_actor_out_of_scope(self, conn, p):
+        """All handles to a non-detached actor dropped -> tear it down."""
+        return await self.rpc_kill_actor(conn, p)
+
+    async def rpc_list_actors(self, conn, p):
+        return [a.view() for a in self.actors.values()]
+
+    async def rpc_list_workers(self, conn, p):
+        return list(self.workers.values())
+
+    async def rpc_list_jobs(self, conn, p):
+        return list(self.jobs.values())
+
+    # ------------------------------------------------------------ placement
+    async def rpc_create_placement_group(self, conn, p):
+        pg = PlacementGroupInfo(p["pg_id"], p["bundles"], p.get("strategy", "PACK"), p.get("name", ""))
+        self.pgs[pg.pg_id] = pg
+        asyncio.get_running_loop().create_task(self._schedule_pg(pg))
+        return {"ok": True}
+
+    async def _schedule_pg(self, pg: PlacementGroupInfo, timeout_s: float = 3600.0):
+        deadline = time.monotonic() + timeout_s
+        while time.monotonic() < deadline and not self._shutdown.is_set():
+            placement = self._plan_pg(pg)
+            if placement is not None:
+                # 2PC-lite: reserve each bundle on its raylet; rollback on fail
+                reserved = []
+                ok = True
+                for idx, node in placement:
+                    try:
+                        r = await node.conn.call(
+                            "reserve_bundle",
+                            {"pg_id": pg.pg_id, "bundle_index": idx, "resources": pg.bundles[idx]},
+                            timeout=30,
+                        )
+                    except Exception:
+                        r = None
+                    if not r or not r.get("ok"):
+                        ok = False
+                        break
+                    reserved.append((idx, node))
+                if ok:
+                    for idx, node in placement:
+                        pg.bundle_nodes[idx] = node.node_id
+                    pg.state = "CREATED"
+                    await self._publish("PG", pg.pg_id, pg.view())
+                    return
+                for idx, node in reserved:
+                    try:
+                        await node.conn.call("return_bundle", {"pg_id": pg.pg_id, "bundle_index": idx})
+                    except Exception:
+                        pass
+            await asyncio.sleep(0.2)
+        pg.state = "FAILED"
+        await self._publish("PG", pg.pg_id, pg.view())
+
+    def _plan_pg(self, pg: PlacementGroupInfo):
+        """Return [(bundle_index, NodeInfo)] or None if not placeable now."""
+        alive = [n for n in self.nodes.values() if n.alive]
+        if not alive:
+            return None
+        avail = {n.node_id: dict(n.resources_available) for n in alive}
+
+        def fits(nid, res):
+            return all(avail[nid].get(k, 0) >= v for k, v in res.items())
+
+        def take(nid, res):
+            for k, v in res.items():
+                avail[nid][k] = avail[nid].get(k, 0) - v
+
+        placement = []
+        strategy = pg.strategy
+        if strategy in ("STRICT_PACK", "PACK"):
+            # try to fit all on one node first
+            for n in alive:
+                trial = dict(avail[n.node_id])
+                ok = True
+                for b in pg.bundles:
+                    if all(trial.get(k, 0) >= v for k, v in b.items()):
+                        for k, v in b.items():
+                            trial[k] = trial.get(k, 0) - v
+                    else:
+                        ok = False
+                        break
+                if ok:
+                    return [(i, n) for i in range(len(pg.bundles))]
+            if strategy == "STRICT_PACK":
+                return None
+        if strategy == "STRICT_SPREAD" and len(pg.bundles) > len(alive):
+            return None
+        # spread/fallback: round-robin over nodes with capacity
+        used_nodes = set()
+        for i, b in enumerate(pg.bundles):
+            candidates = sorted(
+                alive,
+                key=lambda n: (n.node_id in used_nodes, -avail[n.node_id].get("CPU", 0)),
+            )
+            placed = False
+            for n in candidates:
+                if strategy == "STRICT_SPREAD" and n.node_id in used_nodes:
+                    continue
+                if fits(n.node_id, b):
+                    take(n.node_id, b)
+                    placement.append((i, n))
+                    used_nodes.add(n.node_id)
+                    placed = True
+                    break
+            if not placed:
+                return None
+        return placement
+
+    async def rpc_get_placement_group(self, conn, p):
+        pg = self.pgs.get(p["pg_id"])
+        return pg.view() if pg else None
+
+    async def rpc_remove_placement_group(self, conn, p):
+        pg = self.pgs.pop(p["pg_id"], None)
+        if pg and pg.state == "CREATED":
+            for idx, nid in enumerate(pg.bundle_nodes):
+                node = self.nodes.get(nid)
+                if node and node.conn and not node.conn.closed:
+                    try:
+                        await node.conn.call("return_bundle", {"pg_id": pg.pg_id, "bundle_index": idx})
+                    except Exception:
+                        pass
+        return {"ok": True}
+
+    async def rpc_list_placement_groups(self, conn, p):
+        return [pg.view() for pg in self.pgs.values()]
+
+    # ----------------------------------------------------------------- state
+    async def rpc_cluster_resources(self, conn, p):
+        total: Dict[str, float] = {}
+        avail: Dict[str, float] = {}
+        for n in self.nodes.values():
+            if not n.alive:
+                continue
+            for k, v in n.resources_total.items():
+                total[k] = total.get(k, 0) + v
+            for k, v in n.resources_available.items():
+                avail[k] = avail.get(k, 0) + v
+        return {"total": total, "available": avail}
+
+    async def rpc_shutdown(self, conn, p):
+        self._shutdown.set()
+        for node in self.nodes.values():
+            if node.conn and not node.conn.closed:
+                try:
+                    await node.conn.notify("shutdown", {})
+                except Exception:
+                    pass
+        asyncio.get_running_loop().call_later(0.2, lambda: os._exit(0))
+        return {"ok": True}
+
+
+async def run_gcs(host="127.0.0.1", port=0, announce_fd: int = None):
+    gcs = GcsServer()
+    bound = await gcs.start(host, port)
+    if announce_fd is not None:
+        os.write(announce_fd, (str(bound) + "\n").encode())
+        os.close(announce_fd)
+    await gcs._shutdown.wait()
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("--port", type=int, default=0)
+    args = ap.parse_args()
+    logging.basicConfig(level=logging.INFO)
+    asyncio.run(run_gcs(args.host, args.port))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,526 @@
+"""Raylet — the per-node daemon: worker pool + lease-based local scheduler +
+shm object store host.
+
+Role parity: reference raylet (src/ray/raylet/node_manager.h:134 NodeManager;
+lease protocol HandleRequestWorkerLease node_manager.cc:1794 ->
+ClusterLeaseManager -> LocalLeaseManager; WorkerPool worker_pool.h:285 forks
+language workers; placement-group bundle reservation
+placement_group_resource_manager.cc). Differences by design: the shm store is
+direct-mapped (no store server thread), and scheduling is lease-queue +
+local-first with spillback hints from the GCS node table rather than a full
+cluster resource view (the GCS owns the global view).
+
+GPU accounting: the raylet owns the node's HIP device id pool; a lease that
+asks for GPUs is granted specific ids which the worker exports as
+HIP_VISIBLE_DEVICES before user code initializes the runtime (parity with
+python/ray/_private/accelerators/amd_gpu.py:10).
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import logging
+import os
+import signal
+import subprocess
+import sys
+import time
+from typing import Any, Dict, List, Optional, Tuple
+
+from ant_ray_amd._private import protocol
+from ant_ray_amd._private.ids import NodeID, WorkerID
+from ant_ray_amd._private.protocol import Connection
+
+logger = logging.getLogger("antray.raylet")
+
+IDLE_WORKER_KILL_S = 60.0
+
+
+def detect_num_gpus() -> int:
+    """Count AMD GPUs without importing torch (KFD topology)."""
+    if "HIP_VISIBLE_DEVICES" in os.environ:
+        v = os.environ["HIP_VISIBLE_DEVICES"]
+        return 0 if v == "" else len(v.split(","))
+    base = "/sys/class/kfd/kfd/topology/nodes"
+    count = 0
+    try:
+        for d in os.listdir(base):
+            p = os.path.join(base, d, "gpu_id")
+            try:
+                with open(p) as f:
+                    if int(f.read().strip()) != 0:
+                        count += 1
+            except Exception:
+                pass
+    except FileNotFoundError:
+        pass
+    return count
+
+
+class WorkerProc:
+    def __init__(self, worker_id: bytes, proc: subprocess.Popen):
+        self.worker_id = worker_id
+        self.proc = proc
+        self.conn: Optional[Connection] = None
+        self.addr: Optional[Tuple[str, int]] = None
+        self.pid = proc.pid
+        self.registered = asyncio.Event()
+        self.leased = False
+        self.lease_owner: Optional[Connection] = None
+        self.lease_resources: Dict[str, float] = {}
+        self.gpu_ids: List[int] = []
+        self.is_actor = False
+        self.detached_actor = False
+        self.idle_since = time.monotonic()
+
+
+class Raylet:
+    def __init__(self, node_ip: str, gcs_addr: Tuple[str, int], resources: Dict[str, float],
+                 store_path: str, store_capacity: int, session_dir: str):
+        self.node_id = NodeID.from_random().binary()
+        self.node_ip = node_ip
+        self.gcs_addr = gcs_addr
+        self.resources_total = dict(resources)
+        self.resources_available = dict(resources)
+        self.store_path = store_path
+        self.store_capacity = store_capacity
+        self.session_dir = session_dir
+        self.workers: Dict[bytes, WorkerProc] = {}
+        self.idle_workers: List[WorkerProc] = []
+        self.pending_leases: List[Tuple[dict, asyncio.Future]] = []
+        self.gpu_pool: Dict[int, float] = {}  # gpu id -> available fraction
+        ngpus = int(self.resources_total.get("GPU", 0))
+        for i in range(ngpus):
+            self.gpu_pool[i] = 1.0
+        # placement group bundle reservations: (pg_id, idx) -> available dict
+        self.bundles: Dict[Tuple[bytes, int], Dict[str, float]] = {}
+        self.gcs_conn: Optional[Connection] = None
+        self.port = None
+        self._server = None
+        self._store = None
+        self._shutdown = asyncio.Event()
+
+    # ----------------------------------------------------------------- start
+    async def start(self, port=0):
+        # host the node's shm store
+        from ant_ray_amd._shm_store import ShmStore
+
+        if not os.path.exists(self.store_path):
+            self._store = ShmStore.create(self.store_path, self.store_capacity)
+        else:
+            self._store = ShmStore.open(self.store_path)
+        self._server, self.port = await protocol.serve(self._handle, self.node_ip, port)
+        self.gcs_conn = await protocol.connect(self.gcs_addr, self._handle, name="raylet->gcs")
+        self.gcs_conn.on_close = lambda c: os._exit(1)  # fate-share with GCS
+        await self.gcs_conn.call(
+            "register_node",
+            {
+                "node_id": self.node_id,
+                "addr": [self.node_ip, self.port],
+                "resources": self.resources_total,
+                "store_path": self.store_path,
+                "object_store_bytes": self.store_capacity,
+            },
+        )
+        asyncio.get_running_loop().create_task(self._heartbeat_loop())
+        asyncio.get_running_loop().create_task(self._reap_loop())
+        logger.info(
+            "raylet %s on %s:%s (%s)", self.node_id.hex()[:8], self.node_ip, self.port,
+            {k: v for k, v in self.resources_total.items()},
+        )
+        return self.port
+
+    async def _heartbeat_loop(self):
+        while not self._shutdown.is_set():
+            try:
+                await self.gcs_conn.call(
+                    "heartbeat",
+                    {"node_id": self.node_id, "resources_available": self.resources_available},
+                    timeout=5,
+                )
+            except Exception:
+                pass
+            await asyncio.sleep(1.0)
+
+    async def _reap_loop(self):
+        """Detect dead worker processes + kill over-provisioned idle workers."""
+        while not self._shutdown.is_set():
+            await asyncio.sleep(0.5)
+            for w in list(self.workers.values()):
+                if w.proc.poll() is not None:
+                    await self._on_worker_dead(w, f"exit code {w.proc.returncode}")
+            now = time.monotonic()
+            max_idle = int(self.resources_total.get("CPU", 1))
+            if len(self.idle_workers) > max_idle:
+                for w in [w for w in self.idle_workers if now - w.idle_since > IDLE_WORKER_KILL_S][: len(self.idle_workers) - max_idle]:
+                    self._kill_worker(w)
+
+    async def _on_worker_dead(self, w: WorkerProc, reason: str):
+        self.workers.pop(w.worker_id, None)
+        if w in self.idle_workers:
+            self.idle_workers.remove(w)
+        if w.leased:
+            self._release_resources(w)
+        try:
+            await self.gcs_conn.call(
+                "report_worker_failure", {"worker_id": w.worker_id, "reason": reason}, timeout=5
+            )
+        except Exception:
+            pass
+        self._pump_leases()
+
+    # -------------------------------------------------------------- dispatch
+    async def _handle(self, conn: Connection, method: str, p: Any):
+        fn = getattr(self, "rpc_" + method, None)
+        if fn is None:
+            raise ValueError(f"unknown raylet method {method}")
+        return await fn(conn, p or {})
+
+    # --------------------------------------------------------------- workers
+    def _spawn_worker(self) -> WorkerProc:
+        worker_id = WorkerID.from_random().binary()
+        log_dir = os.path.join(self.session_dir, "logs")
+        os.makedirs(log_dir, exist_ok=True)
+        out = open(os.path.join(log_dir, f"worker-{worker_id.hex()[:8]}.log"), "wb")
+        env = dict(os.environ)
+        env["ANTRAY_WORKER_ID"] = worker_id.hex()
+        env["ANTRAY_GCS"] = f"{self.gcs_addr[0]}:{self.gcs_addr[1]}"
+        env["ANTRAY_RAYLET"] = f"{self.node_ip}:{self.port}"
+        env["ANTRAY_NODE_ID"] = self.node_id.hex()
+        env["ANTRAY_STORE"] = self.store_path
+        env["ANTRAY_SESSION_DIR"] = self.session_dir
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "ant_ray_amd._private.workers.default_worker"],
+            env=env,
+            stdout=out,
+            stderr=subprocess.STDOUT,
+            start_new_session=True,
+        )
+        out.close()
+        w = WorkerProc(worker_id, proc)
+        self.workers[worker_id] = w
+        return w
+
+    async def rpc_register_worker(self, conn, p):
+        w = self.workers.get(p["worker_id"])
+        if w is None:
+            raise ValueError("unknown worker")
+        w.conn = conn
+        w.addr = tuple(p["addr"])
+        conn.session["worker"] = w
+        conn.on_close = lambda c: asyncio.get_running_loop().create_task(
+            self._on_worker_conn_closed(w)
+        )
+        w.registered.set()
+        return {"ok": True}
+
+    async def _on_worker_conn_closed(self, w: WorkerProc):
+        if w.proc.poll() is None:
+            # connection lost but process alive; give it a moment then kill
+            await asyncio.sleep(1.0)
+            if w.proc.poll() is None:
+                self._kill_worker(w)
+        await self._on_worker_dead(w, "connection closed")
+
+    def _kill_worker(self, w: WorkerProc):
+        if w in self.idle_workers:
+            self.idle_workers.remove(w)
+        try:
+            os.killpg(w.proc.pid, signal.SIGKILL)
+        except Exception:
+            try:
+                w.proc.kill()
+            except Exception:
+                pass
+
+    # ---------------------------------------------------------------- leases
+    def _fits(self, avail: Dict[str, float], res: Dict[str, float]) -> bool:
+        return all(avail.get(k, 0.0) + 1e-9 >= v for k, v in res.items())
+
+    def _feasible(self, res: Dict[str, float]) -> bool:
+        return all(self.resources_total.get(k, 0.0) + 1e-9 >= v for k, v in res.items())
+
+    def _take(self, res: Dict[str, float], pg: Optional[dict]):
+        pool = (
+            self.bundles.get((pg["pg_id"], pg.get("bundle_index", 0)))
+            if pg
+            else self.resources_available
+        )
+        for k, v in res.items():
+            pool[k] = pool.get(k, 0.0) - v
+        gpu_ids: List[int] = []
+        need = res.get("GPU", 0.0)
+        if need > 0 and not pg:
+            remaining = need
+            for gid in sorted(self.gpu_pool, key=lambda g: -self.gpu_pool[g]):
+                if remaining <= 1e-9:
+                    break
+                grab = min(1.0, remaining, self.gpu_pool[gid])
+                if grab > 1e-9 and self.gpu_pool[gid] >= min(1.0, remaining) - 1e-9:
+                    self.gpu_pool[gid] -= grab
+                    remaining -= grab
+                    gpu_ids.append(gid)
+        elif need > 0 and pg:
+            # GPU ids for PG bundles come from the general pool too
+            remaining = need
+            for gid in sorted(self.gpu_pool, key=lambda g: -self.gpu_pool[g]):
+                if remaining <= 1e-9:
+                    break
+                grab = min(1.0, remaining, self.gpu_pool[gid])
+                if grab > 1e-9:
+                    self.gpu_pool[gid] -= grab
+                    remaining -= grab
+                    gpu_ids.append(gid)
+        return gpu_ids
+
+    def _give_back(self, res: Dict[str, float], pg: Optional[dict], gpu_ids: List[int]):
+        pool = (
+            self.bundles.get((pg["pg_id"], pg.get("bundle_index", 0)))
+            if pg
+            else self.resources_available
+        )
+        if pool is not None:
+            for k, v in res.items():
+                pool[k] = pool.get(k, 0.0) + v
+        need = res.get("GPU", 0.0)
+        if need > 0:
+            remaining = need
+            for gid in gpu_ids:
+                back = min(1.0, remaining)
+                self.gpu_pool[gid] = min(1.0, self.gpu_pool[gid] + back)
+                remaining -= back
+
+    def _release_resources(self, w: WorkerProc):
+        if not w.leased:
+            return
+        self._give_back(w.lease_resources, getattr(w, "lease_pg", None), w.gpu_ids)
+        w.leased = False
+        w.lease_resources = {}
+        w.gpu_ids = []
+        w.lease_pg = None
+
+    async def rpc_lease_worker(self, conn, p):
+        """Grant a worker lease; queue if resources busy; spillback hint if
+        infeasible on this node."""
+        res = dict(p.get("resources") or {})
+        pg = p.get("pg")
+        if not pg and not self._feasible(res):
+            return {"granted": False, "infeasible": True}
+        fut = asyncio.get_running_loop().create_future()
+        self.pending_leases.append(((p, conn), fut))
+        self._pump_leases()
+        return await fut
+
+    def _pump_leases(self):
+        if not self.pending_leases:
+            return
+        granted_any = True
+        while granted_any and self.pending_leases:
+            granted_any = False
+            for item in list(self.pending_leases):
+                (p, requester), fut = item
+                if fut.done():
+                    self.pending_leases.remove(item)
+                    continue
+                res = dict(p.get("resources") or {})
+                pg = p.get("pg")
+                pool = (
+                    self.bundles.get((pg["pg_id"], pg.get("bundle_index", 0)))
+                    if pg
+                    else self.resources_available
+                )
+                if pool is None:
+                    continue  # bundle not reserved yet
+                if not self._fits(pool, {k: v for k, v in res.items() if k != "GPU"} if pg else res):
+                    continue
+                if res.get("GPU", 0) > 0:
+                    free_gpu = sum(self.gpu_pool.values())
+                    if free_gpu + 1e-9 < res["GPU"]:
+                        continue
+                asyncio.get_running_loop().create_task(self._grant_lease(item))
+                self.pending_leases.remove(item)
+                granted_any = True
+
+    async def _grant_lease(self, item):
+        (p, requester), fut = item
+        res = dict(p.get("resources") or {})
+        pg = p.get("pg")
+        gpu_ids = self._take(res, pg)
+        w = None
+        while w is None:
+            if self.idle_workers:
+                w = self.idle_workers.pop()
+                if w.proc.poll() is not None:
+                    self.workers.pop(w.worker_id, None)
+                    w = None
+                    continue
+            else:
+                w = self._spawn_worker()
+            try:
+                await asyncio.wait_for(w.registered.wait(), timeout=60)
+            except asyncio.TimeoutError:
+                self._kill_worker(w)
+                self.workers.pop(w.worker_id, None)
+                w = None
+        w.leased = True
+        w.lease_owner = requester
+        w.lease_resources = res
+        w.lease_pg = pg
+        w.gpu_ids = gpu_ids
+        w.is_actor = p.get("actor_id") is not None
+        w.detached_actor = bool(p.get("detached"))
+        # tell the worker its lease context (GPU visibility, job env)
+        try:
+            await w.conn.call(
+                "set_lease",
+                {
+                    "gpu_ids": gpu_ids,
+                    "resources": res,
+                    "actor_id": p.get("actor_id"),
+                    "runtime_env": p.get("runtime_env"),
+                },
+                timeout=30,
+            )
+        except Exception as e:
+            logger.warning("set_lease failed: %s", e)
+            self._release_resources(w)
+            self._kill_worker(w)
+            if not fut.done():
+                fut.set_result({"granted": False, "error": str(e)})
+            return
+        if not fut.done():
+            fut.set_result(
+                {
+                    "granted": True,
+                    "worker_id": w.worker_id,
+                    "addr": list(w.addr),
+                    "gpu_ids": gpu_ids,
+                    "node_id": self.node_id,
+                }
+            )
+
+    async def rpc_return_worker(self, conn, p):
+        w = self.workers.get(p["worker_id"])
+        if w is None:
+            return {"ok": False}
+        self._release_resources(w)
+        kill = p.get("kill", False) or w.is_actor or w.gpu_ids
+        if kill or w.proc.poll() is not None:
+            self._kill_worker(w)
+        else:
+            w.idle_since = time.monotonic()
+            w.leased = False
+            self.idle_workers.append(w)
+        self._pump_leases()
+        return {"ok": True}
+
+    async def rpc_prestart_workers(self, conn, p):
+        n = int(p.get("n", 1))
+        for _ in range(n):
+            if len(self.workers) < int(self.resources_total.get("CPU", 1)) + 2:
+                w = self._spawn_worker()
+
+                async def _pool(w=w):
+                    try:
+                        await asyncio.wait_for(w.registered.wait(), timeout=60)
+                        if not w.leased:
+                            self.idle_workers.append(w)
+                            self._pump_leases()
+                    except asyncio.TimeoutError:
+                        pass
+
+                asyncio.get_running_loop().create_task(_pool())
+        return {"ok": True}
+
+    # ---------------------------------------------------------------- bundles
+    async def rpc_reserve_bundle(self, conn, p):
+        res = dict(p["resources"])
+        # CPU/mem come out of the node pool now; GPU ids are taken at lease time
+        non_gpu = {k: v for k, v in res.items() if k != "GPU"}
+        if not self._fits(self.resources_available, non_gpu):
+            return {"ok": False}
+        gpu_need = res.get("GPU", 0.0)
+        if gpu_need > 0 and sum(self.gpu_pool.values()) + 1e-9 < gpu_need:
+            return {"ok": False}
+        for k, v in non_gpu.items():
+            self.resources_available[k] = self.resources_available.get(k, 0.0) - v
+        self.bundles[(p["pg_id"], p["bundle_index"])] = dict(res)
+        self._pump_leases()
+        return {"ok": True}
+
+    async def rpc_return_bundle(self, conn, p):
+        b = self.bundles.pop((p["pg_id"], p["bundle_index"]), None)
+        if b is not None:
+            for k, v in b.items():
+                if k != "GPU":
+                    self.resources_available[k] = self.resources_available.get(k, 0.0) + v
+        return {"ok": True}
+
+    # ------------------------------------------------------------------ misc
+    async def rpc_node_info(self, conn, p):
+        return {
+            "node_id": self.node_id,
+            "addr": [self.node_ip, self.port],
+            "resources_total": self.resources_total,
+            "resources_available": self.resources_available,
+            "store_path": self.store_path,
+            "num_workers": len(self.workers),
+        }
+
+    async def rpc_shutdown(self, conn, p):
+        await self._do_shutdown()
+        return {"ok": True}
+
+    async def _do_shutdown(self):
+        self._shutdown.set()
+        for w in list(self.workers.values()):
+            self._kill_worker(w)
+        try:
+            if os.path.exists(self.store_path):
+                os.unlink(self.store_path)
+        except Exception:
+            pass
+        asyncio.get_running_loop().call_later(0.1, lambda: os._exit(0))
+
+
+async def run_raylet(args):
+    resources = json.loads(args.resources) if args.resources else {}
+    ncpu = args.num_cpus if args.num_cpus is not None else os.cpu_count()
+    ngpu = args.num_gpus if args.num_gpus is not None else detect_num_gpus()
+    resources.setdefault("CPU", float(ncpu))
+    resources.setdefault("GPU", float(ngpu))
+    resources.setdefault("memory", float(os.sysconf("SC_PHYS_PAGES") * os.sysconf("SC_PAGE_SIZE")))
+    resources.setdefault("object_store_memory", float(args.store_capacity))
+    resources.setdefault(f"node:{args.node_ip}", 1.0)
+    host, port = args.gcs.split(":")
+    raylet = Raylet(
+        args.node_ip, (host, int(port)), resources, args.store_path,
+        args.store_capacity, args.session_dir,
+    )
+    await raylet.start(args.port)
+    if args.announce_fd:
+        os.write(args.announce_fd, (str(raylet.port) + "\n").encode())
+        os.close(args.announce_fd)
+    await raylet._shutdown.wait()
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gcs", required=True)
+    ap.add_argument("--node-ip", default="127.0.0.1")
+    ap.add_argument("--port", type=int, default=0)
+    ap.add_argument("--num-cpus", type=int, default=None)
+    ap.add_argument("--num-gpus", type=int, default=None)
+    ap.add_argument("--resources", default="")
+    ap.add_argument("--store-path", required=True)
+    ap.add_argument("--store-capacity", type=int, default=2 * 1024**3)
+    ap.add_argument("--session-dir", required=True)
+    ap.add_argument("--announce-fd", type=int, default=0)
+    args = ap.parse_args()
+    logging.basicConfig(level=logging.INFO)
+    asyncio.run(run_raylet(args))
+
+
+if __name__ == "__main__":
+    main()

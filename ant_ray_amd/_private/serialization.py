@@ -1,0 +1,194 @@
+"""Object serialization: cloudpickle + protocol-5 out-of-band buffers.
+
+Role parity: reference python/ray/_private/serialization.py (cloudpickle +
+Arrow zero-copy buffers + out-of-band tensor hooks). Design here:
+
+  * Values are pickled with `cloudpickle` at protocol 5; large contiguous
+    buffers (numpy arrays, torch CPU tensor storage, bytes>threshold) are
+    collected out-of-band via `buffer_callback`.
+  * The stored wire format keeps buffers 64-byte aligned so a reader can map
+    them zero-copy straight out of the shm segment (numpy arrays deserialized
+    from the object store alias shm memory, read-only — same behavior as the
+    reference's plasma-backed numpy reads).
+  * ObjectRefs nested inside values are serialized by binary id + owner
+    address and re-hydrated on read; the contained-ref list travels in the
+    header so the runtime can track borrows.
+
+Layout of a serialized object:
+    u32 magic | u32 n_buffers | u64 pickle_size
+    (u64 offset, u64 size) * n_buffers      # relative to start of data area
+    pickle bytes | padding | buffer0 | padding | buffer1 ...
+"""
+from __future__ import annotations
+
+import pickle
+import struct
+from typing import Any, List, Optional, Tuple
+
+import cloudpickle
+
+MAGIC = 0x41525A31  # "ARZ1"
+_HDR = struct.Struct("<IIQ")
+_BUF = struct.Struct("<QQ")
+ALIGN = 64
+
+# Metadata tags (stored in the shm object's metadata field).
+META_PICKLE = b"py"
+META_ERROR = b"err"
+META_RAW = b"raw"  # plain bytes payload, no pickle
+META_ACTOR_DIED = b"actor_died"
+META_GPU = b"gpu"  # GPU object: payload is a descriptor, tensors live on-device
+
+
+def _align(n: int) -> int:
+    return (n + ALIGN - 1) & ~(ALIGN - 1)
+
+
+class SerializedObject:
+    """A pickled value plus its out-of-band buffers, ready to write."""
+
+    __slots__ = ("pickle_bytes", "buffers", "contained_refs", "total_size", "metadata")
+
+    def __init__(self, pickle_bytes: bytes, buffers: List[memoryview], contained_refs, metadata=META_PICKLE):
+        self.pickle_bytes = pickle_bytes
+        self.buffers = buffers
+        self.contained_refs = contained_refs
+        self.metadata = metadata
+        hdr = _HDR.size + _BUF.size * len(buffers)
+        off = _align(hdr + len(pickle_bytes))
+        for b in buffers:
+            off = _align(off + b.nbytes)
+        self.total_size = off
+
+    def write_into(self, dest: memoryview):
+        """Write the full serialized form into `dest` (length >= total_size)."""
+        n = len(self.buffers)
+        hdr_size = _HDR.size + _BUF.size * n
+        _HDR.pack_into(dest, 0, MAGIC, n, len(self.pickle_bytes))
+        off = _align(hdr_size + len(self.pickle_bytes))
+        pos = _HDR.size
+        offsets = []
+        for b in self.buffers:
+            offsets.append((off, b.nbytes))
+            _BUF.pack_into(dest, pos, off, b.nbytes)
+            pos += _BUF.size
+            off = _align(off + b.nbytes)
+        dest[hdr_size : hdr_size + len(self.pickle_bytes)] = self.pickle_bytes
+        for (boff, bsize), b in zip(offsets, self.buffers):
+            dest[boff : boff + bsize] = b  # PickleBuffer.raw() views are 1-D bytes
+
+    def to_bytes(self) -> bytes:
+        out = bytearray(self.total_size)
+        self.write_into(memoryview(out))
+        return bytes(out)
+
+
+def _torch_cpu_tensor_reducer(tensor):
+    """Out-of-band reducer for CPU torch tensors (dense, contiguous path)."""
+    import torch
+
+    if tensor.device.type != "cpu" or not tensor.is_contiguous() or tensor.is_sparse:
+        # fall back to torch's own reducer
+        return NotImplemented
+    # byte view works for every dtype incl. bf16 (numpy has no bfloat16)
+    np_view = tensor.detach().reshape(-1).view(torch.uint8).numpy()
+    return (
+        _rebuild_torch_tensor,
+        (pickle.PickleBuffer(np_view), str(tensor.dtype), tuple(tensor.shape)),
+    )
+
+
+def _rebuild_torch_tensor(buf, dtype_str, shape):
+    import warnings
+
+    import numpy as np
+    import torch
+
+    dtype = getattr(torch, dtype_str.replace("torch.", ""))
+    mv = buf.raw()
+    np_arr = np.frombuffer(mv, dtype=np.uint8)
+    with warnings.catch_warnings():
+        # zero-copy view over (possibly read-only) shm memory; writes would
+        # corrupt the store, so the returned tensor must be treated immutable
+        warnings.simplefilter("ignore")
+        t = torch.from_numpy(np_arr)
+    return t.view(dtype).reshape(shape)
+
+
+class _Pickler(cloudpickle.CloudPickler):
+    def __init__(self, file, buffer_callback=None):
+        super().__init__(file, protocol=5, buffer_callback=buffer_callback)
+        self.contained_refs = []
+
+    def reducer_override(self, obj):
+        from ant_ray_amd._private.object_ref import ObjectRef
+
+        if isinstance(obj, ObjectRef):
+            self.contained_refs.append(obj)
+            return (ObjectRef._rehydrate, (obj.binary(), obj.owner_addr))
+        try:
+            import torch
+
+            if isinstance(obj, torch.Tensor):
+                r = _torch_cpu_tensor_reducer(obj)
+                if r is not NotImplemented:
+                    return r
+        except ImportError:
+            pass
+        return super().reducer_override(obj)
+
+
+_BUFFER_THRESHOLD = 512  # buffers below this get pickled in-band
+
+
+def serialize(value: Any, metadata: bytes = META_PICKLE) -> SerializedObject:
+    import io
+
+    buffers: List[memoryview] = []
+
+    def cb(pb: pickle.PickleBuffer):
+        mv = pb.raw()
+        if mv.nbytes < _BUFFER_THRESHOLD:
+            return True  # keep in-band
+        buffers.append(mv)
+        return False
+
+    f = io.BytesIO()
+    p = _Pickler(f, buffer_callback=cb)
+    p.dump(value)
+    return SerializedObject(f.getvalue(), buffers, p.contained_refs, metadata)
+
+
+def deserialize(data: memoryview, metadata: bytes = META_PICKLE) -> Any:
+    if metadata == META_RAW:
+        return bytes(data)
+    magic, n, psize = _HDR.unpack_from(data, 0)
+    if magic != MAGIC:
+        raise ValueError("corrupt serialized object")
+    hdr_size = _HDR.size + _BUF.size * n
+    buffers = []
+    pos = _HDR.size
+    for _ in range(n):
+        off, size = _BUF.unpack_from(data, pos)
+        pos += _BUF.size
+        buffers.append(data[off : off + size])
+    pbytes = data[hdr_size : hdr_size + psize]
+    value = pickle.loads(pbytes, buffers=buffers)
+    if metadata in (META_ERROR, META_ACTOR_DIED):
+        # value is the exception instance to raise at the get() site
+        return value
+    return value
+
+
+def serialize_error(exc: BaseException) -> SerializedObject:
+    try:
+        return serialize(exc, metadata=META_ERROR)
+    except Exception:
+        from ant_ray_amd.exceptions import RayTaskError
+
+        fallback = RayTaskError(
+            function_name=getattr(exc, "function_name", "unknown"),
+            traceback_str=str(exc),
+            cause=None,
+        )
+        return serialize(fallback, metadata=META_ERROR)

@@ -1,0 +1,281 @@
+"""Task execution engine for worker processes.
+
+Role parity: reference TaskReceiver + scheduling queues
+(src/ray/core_worker/task_execution/task_receiver.h:44,
+actor_scheduling_queue.cc, concurrency_group_manager.cc, fiber.h for async
+actors) and the Cython execution path (_raylet.pyx:2141
+task_execution_handler). Execution models:
+
+  * normal tasks + plain actors: one serial executor thread (strict order),
+  * threaded actors (max_concurrency>1): ThreadPoolExecutor,
+  * async actors (coroutine methods): dedicated asyncio loop thread with a
+    max_concurrency semaphore,
+  * actor task ordering: per-caller sequence numbers; out-of-order arrivals
+    are buffered until their turn (parity with SequentialActorSubmitQueue).
+"""
+from __future__ import annotations
+
+import asyncio
+import concurrent.futures
+import inspect
+import logging
+import os
+import queue
+import threading
+from typing import Any, Dict, Optional
+
+from ant_ray_amd._private import serialization
+from ant_ray_amd._private.ids import ObjectID, TaskID
+from ant_ray_amd._private.object_ref import ObjectRef
+from ant_ray_amd._private.object_store import INLINE_OBJECT_MAX
+from ant_ray_amd.exceptions import RayTaskError
+
+logger = logging.getLogger("antray.executor")
+
+
+class TaskExecutor:
+    def __init__(self, core_worker):
+        self.cw = core_worker
+        self._fn_cache: Dict[bytes, Any] = {}
+        self.actor_instance = None
+        self.actor_id: Optional[bytes] = None
+        self.max_concurrency = 1
+        self._serial_q: "queue.Queue" = queue.Queue()
+        self._serial_thread = threading.Thread(target=self._serial_loop, daemon=True)
+        self._serial_thread.start()
+        self._pool: Optional[concurrent.futures.ThreadPoolExecutor] = None
+        self._actor_loop: Optional[Any] = None  # EventLoopThread for async actors
+        self._async_sem: Optional[asyncio.Semaphore] = None
+        self._exit_requested = False
+        self._inflight = 0
+        self._inflight_cv = threading.Condition()
+        # actor-task ordering: caller -> expected next seq / buffered payloads
+        self._expected_seq: Dict[bytes, int] = {}
+        self._buffered: Dict[bytes, Dict[int, tuple]] = {}
+
+    # ------------------------------------------------------------- entrypoint
+    async def submit(self, payload: dict) -> dict:
+        """Called on the io loop; returns the reply dict when the task is done."""
+        loop = asyncio.get_running_loop()
+        fut: asyncio.Future = loop.create_future()
+        ttype = payload.get("type")
+        if ttype == "actor_task":
+            self._submit_actor_ordered(payload, fut, loop)
+        else:
+            self._dispatch(payload, fut, loop)
+        return await fut
+
+    def _submit_actor_ordered(self, payload, fut, loop):
+        caller = payload["caller"]
+        seq = payload["seq"]
+        expected = self._expected_seq.get(caller, 1)
+        if seq == expected:
+            self._dispatch(payload, fut, loop)
+            self._expected_seq[caller] = expected + 1
+            # drain any buffered successors
+            buf = self._buffered.get(caller)
+            while buf:
+                nxt = self._expected_seq[caller]
+                item = buf.pop(nxt, None)
+                if item is None:
+                    break
+                self._dispatch(item[0], item[1], item[2])
+                self._expected_seq[caller] = nxt + 1
+        else:
+            self._buffered.setdefault(caller, {})[seq] = (payload, fut, loop)
+
+    def _dispatch(self, payload, fut, loop):
+        with self._inflight_cv:
+            self._inflight += 1
+
+        def done(reply):
+            with self._inflight_cv:
+                self._inflight -= 1
+                self._inflight_cv.notify_all()
+            loop.call_soon_threadsafe(
+                lambda: fut.set_result(reply) if not fut.done() else None
+            )
+            if self._exit_requested:
+                self._maybe_exit()
+
+        method = payload.get("method", "")
+        if (
+            payload.get("type") == "actor_task"
+            and self.actor_instance is not None
+            and self._is_async_method(method)
+        ):
+            self._run_async_actor_task(payload, done)
+        elif payload.get("type") == "actor_task" and self.max_concurrency > 1:
+            self._pool.submit(self._run_and_reply, payload, done)
+        else:
+            self._serial_q.put((payload, done))
+
+    def _serial_loop(self):
+        while True:
+            payload, done = self._serial_q.get()
+            if payload is None:
+                return
+            self._run_and_reply(payload, done)
+
+    def request_exit(self):
+        self._exit_requested = True
+        self._maybe_exit()
+
+    def _maybe_exit(self):
+        with self._inflight_cv:
+            busy = self._inflight > 0
+        if not busy:
+            try:
+                if self.actor_instance is not None:
+                    ray_terminate = getattr(self.actor_instance, "__ray_terminate__", None)
+                    if ray_terminate:
+                        ray_terminate()
+            except Exception:
+                pass
+            os._exit(0)
+
+    # -------------------------------------------------------------- execution
+    def _is_async_method(self, name: str) -> bool:
+        if self.actor_instance is None:
+            return False
+        m = getattr(type(self.actor_instance), name, None)
+        return m is not None and (
+            inspect.iscoroutinefunction(m)
+            or inspect.isasyncgenfunction(m)
+        )
+
+    def _resolve_fn(self, payload):
+        fn_id = payload.get("fn_id")
+        if payload.get("fn") is not None:
+            import cloudpickle
+
+            fn = cloudpickle.loads(payload["fn"])
+            if fn_id:
+                self._fn_cache[fn_id] = fn
+            return fn
+        if fn_id in self._fn_cache:
+            return self._fn_cache[fn_id]
+        return None
+
+    def _deserialize_args(self, payload):
+        args, kwargs = serialization.deserialize(
+            memoryview(payload["args"]), serialization.META_PICKLE
+        )
+        # top-level ObjectRef args are fetched before the call (reference
+        # semantics: dependencies resolved by the worker before execution)
+        def resolve(v):
+            if isinstance(v, ObjectRef):
+                return self.cw.get([v], None)[0]
+            return v
+
+        args = tuple(resolve(a) for a in args)
+        kwargs = {k: resolve(v) for k, v in kwargs.items()}
+        return args, kwargs
+
+    def _run_and_reply(self, payload, done):
+        try:
+            reply = self._execute(payload)
+        except BaseException as e:  # noqa: BLE001
+            logger.exception("task execution crashed")
+            reply = self._error_reply(payload, e)
+        done(reply)
+
+    def _execute(self, payload) -> dict:
+        ttype = payload["type"]
+        task_id = payload["task_id"]
+        self.cw.current_task_id = task_id
+        try:
+            if ttype == "normal":
+                fn = self._resolve_fn(payload)
+                if fn is None:
+                    return {"status": "need_fn"}
+                args, kwargs = self._deserialize_args(payload)
+                result = fn(*args, **kwargs)
+                return self._reply_results(payload, result)
+            if ttype == "actor_create":
+                import cloudpickle
+
+                cls = cloudpickle.loads(payload["cls"])
+                args, kwargs = self._deserialize_args(payload)
+                self.max_concurrency = int(payload.get("max_concurrency", 1))
+                if self.max_concurrency > 1:
+                    self._pool = concurrent.futures.ThreadPoolExecutor(
+                        max_workers=self.max_concurrency
+                    )
+                self.actor_id = payload.get("actor_id")
+                self.cw.actor_id = self.actor_id
+                self.actor_instance = cls(*args, **kwargs)
+                has_async = any(
+                    inspect.iscoroutinefunction(m)
+                    for _, m in inspect.getmembers(cls, inspect.isfunction)
+                )
+                if has_async:
+                    from ant_ray_amd._private.protocol import EventLoopThread
+
+                    self._actor_loop = EventLoopThread(name="actor-async")
+                    sem_size = self.max_concurrency if self.max_concurrency > 1 else 1000
+                    self._async_sem = None
+                    self._async_sem_size = sem_size
+                return {"status": "ok", "results": []}
+            if ttype == "actor_task":
+                if self.actor_instance is None:
+                    raise RuntimeError("no actor instance in this worker")
+                method = getattr(self.actor_instance, payload["method"])
+                args, kwargs = self._deserialize_args(payload)
+                result = method(*args, **kwargs)
+                return self._reply_results(payload, result)
+            raise ValueError(f"unknown task type {ttype}")
+        except BaseException as e:  # noqa: BLE001
+            return self._error_reply(payload, e)
+        finally:
+            self.cw.current_task_id = None
+
+    def _run_async_actor_task(self, payload, done):
+        async def runner():
+            if self._async_sem is None:
+                self._async_sem = asyncio.Semaphore(self._async_sem_size)
+            async with self._async_sem:
+                try:
+                    method = getattr(self.actor_instance, payload["method"])
+                    args, kwargs = self._deserialize_args(payload)
+                    result = await method(*args, **kwargs)
+                    reply = self._reply_results(payload, result)
+                except BaseException as e:  # noqa: BLE001
+                    reply = self._error_reply(payload, e)
+                done(reply)
+
+        self._actor_loop.submit(runner())
+
+    # ---------------------------------------------------------------- replies
+    def _reply_results(self, payload, result) -> dict:
+        n = payload.get("n_returns", 1)
+        if n == 0:
+            return {"status": "ok", "results": []}
+        values = (result,) if n == 1 else tuple(result)
+        if n > 1 and len(values) != n:
+            raise ValueError(f"task returned {len(values)} values, expected {n}")
+        out = []
+        for i, v in enumerate(values):
+            oid = ObjectID.for_return(TaskID(payload["task_id"]), i).binary()
+            sobj = serialization.serialize(v)
+            self.cw._register_escapes(sobj)
+            if sobj.total_size <= INLINE_OBJECT_MAX:
+                out.append({"oid": oid, "inline": sobj.to_bytes(), "meta": sobj.metadata})
+            else:
+                self.cw.store.put_serialized_to_shm(oid, sobj)
+                out.append({"oid": oid, "inline": None, "holder": list(self.cw.addr)})
+        return {"status": "ok", "results": out}
+
+    def _error_reply(self, payload, exc: BaseException) -> dict:
+        if isinstance(exc, RayTaskError):
+            err = exc
+        else:
+            name = payload.get("method") or payload.get("name") or "task"
+            err = RayTaskError.from_exception(exc, name)
+        sobj = serialization.serialize_error(err)
+        return {
+            "status": "error",
+            "error": str(exc),
+            "error_payload": sobj.to_bytes(),
+            "error_meta": sobj.metadata,
+        }

@@ -1,0 +1,922 @@
+"""CoreWorker + global worker state.
+
+Role parity: the reference's per-process runtime
+(src/ray/core_worker/core_worker.h:167 / core_worker.cc — SubmitTask :1969,
+CreateActor :2053, SubmitActorTask :2349, Put :960, Get :1294) plus the Python
+driver state in python/ray/_private/worker.py:461. One class serves both the
+driver and executing workers:
+
+  * task submission with the lease protocol (resolve deps -> lease a worker
+    from the local raylet per scheduling key -> push tasks directly to the
+    leased worker, pipelined; parity with
+    task_submission/normal_task_submitter.cc:34),
+  * actor creation via GCS + direct ordered pushes to the actor's worker
+    (parity with actor_task_submitter.h:68 + sequential_actor_submit_queue),
+  * two-tier object store reads/writes (memory store + direct-mapped shm),
+    owner-mediated pulls for objects living in another node's store
+    (parity with ownership_object_directory + object_manager Push/Pull),
+  * local reference counting with free-on-zero for owned objects
+    (subset of reference_counter.h:44; cross-process borrow counts are
+    tracked conservatively: escaped refs pin their object).
+
+Threading: public API is synchronous (caller threads); all networking runs on
+a dedicated asyncio thread (protocol.EventLoopThread) — parity with the
+reference's io_service threads.
+"""
+from __future__ import annotations
+
+import atexit
+import logging
+import os
+import threading
+import time
+from collections import defaultdict
+from typing import Any, Dict, List, Optional, Sequence, Tuple
+
+from ant_ray_amd._private import protocol, serialization
+from ant_ray_amd._private.ids import ActorID, JobID, ObjectID, TaskID, WorkerID
+from ant_ray_amd._private.object_ref import ObjectRef
+from ant_ray_amd._private.object_store import IN_PLASMA, INLINE_OBJECT_MAX, ObjectStore
+from ant_ray_amd.exceptions import (
+    ActorDiedError,
+    GetTimeoutError,
+    ObjectLostError,
+    RayActorError,
+    RayTaskError,
+    RaySystemError,
+)
+
+logger = logging.getLogger("antray.worker")
+
+LOCAL_MODE = "local"
+DRIVER_MODE = "driver"
+WORKER_MODE = "worker"
+
+PIPELINE_DEPTH = 4          # tasks pushed per leased worker before waiting
+LEASE_IDLE_RELEASE_S = 2.0  # return leased workers after this idle time
+
+
+class _RawResult:
+    """Memory-store entry holding a serialized value (deserialize at get)."""
+
+    __slots__ = ("data", "meta")
+
+    def __init__(self, data: bytes, meta: bytes):
+        self.data = data
+        self.meta = meta
+
+
+class _ErrorResult:
+    __slots__ = ("exc",)
+
+    def __init__(self, exc: BaseException):
+        self.exc = exc
+
+
+class LeasedWorker:
+    def __init__(self, worker_id, addr, node_id, conn):
+        self.worker_id = worker_id
+        self.addr = tuple(addr)
+        self.node_id = node_id
+        self.conn = conn
+        self.in_flight = 0
+        self.idle_since = time.monotonic()
+        self.sent_fns: set = set()
+
+
+class ActorHandleState:
+    def __init__(self, actor_id: bytes):
+        self.actor_id = actor_id
+        self.addr: Optional[Tuple[str, int]] = None
+        self.conn = None
+        self.seq = 0
+        self.state = "PENDING"
+        self.death_cause = ""
+        self.lock = threading.Lock()
+        self.pending: List[dict] = []  # queued before addr known
+        self.inflight: Dict[int, dict] = {}  # seq -> task payload (for errors)
+        self.handle_count = 0
+        self.is_owner = False
+        self.detached = False
+
+
+class CoreWorker:
+    def __init__(self, mode: str, *, gcs_addr=None, store_path=None, node_ip="127.0.0.1",
+                 worker_id: Optional[bytes] = None, session_dir: str = ""):
+        self.mode = mode
+        self.worker_id = worker_id or WorkerID.from_random().binary()
+        self.node_ip = node_ip
+        self.session_dir = session_dir
+        self.job_id: Optional[int] = None
+        self.node_id: Optional[bytes] = None
+        self.raylet_addr: Optional[Tuple[str, int]] = None
+        self.store = ObjectStore(None)
+        self.addr: Optional[Tuple[str, int]] = None
+        self._put_index = 0
+        self._lock = threading.RLock()
+        self._local_refs: Dict[bytes, int] = defaultdict(int)
+        self._owned: Dict[bytes, dict] = {}  # oid -> {escaped, size}
+        self._object_locations: Dict[bytes, Tuple[str, int]] = {}
+        self._actors: Dict[bytes, ActorHandleState] = {}
+        self._actor_results: Dict[bytes, dict] = {}
+        # lease state per scheduling key
+        self._leases: Dict[str, List[LeasedWorker]] = defaultdict(list)
+        self._lease_queue: Dict[str, List[dict]] = defaultdict(list)
+        self._lease_requests: Dict[str, int] = defaultdict(int)
+        self._fn_cache: Dict[bytes, Any] = {}
+        self._pushed_fns: set = set()
+        self.io: Optional[protocol.EventLoopThread] = None
+        self.gcs = None
+        self.raylet = None
+        self._server = None
+        self._worker_conns: Dict[Tuple[str, int], Any] = {}
+        self.connected = False
+        self.executor = None  # TaskExecutor in worker mode
+        self.current_task_id: Optional[bytes] = None
+        self.actor_id: Optional[bytes] = None
+        self.gpu_ids: List[int] = []
+        self._shutdown_handlers = []
+
+    # ================================================================ connect
+    def connect(self, gcs_addr: Tuple[str, int], *, is_driver: bool, raylet_addr=None,
+                store_path=None, node_id=None):
+        self.io = protocol.EventLoopThread()
+        self.gcs_addr = gcs_addr
+        self.io.run(self._async_connect(gcs_addr, is_driver, raylet_addr, store_path, node_id), timeout=60)
+        self.connected = True
+
+    async def _async_connect(self, gcs_addr, is_driver, raylet_addr, store_path, node_id):
+        self.gcs = await protocol.connect(tuple(gcs_addr), self._handle_rpc, name="->gcs")
+        if is_driver:
+            node = await self.gcs.call("get_local_node", {"ip": None})
+            if node is None:
+                raise RaySystemError("no alive node registered in GCS")
+            raylet_addr = tuple(node["addr"])
+            store_path = node["store_path"]
+            node_id = node["node_id"]
+        self.node_id = node_id
+        self.raylet_addr = tuple(raylet_addr)
+        # object store attach
+        from ant_ray_amd._shm_store import ShmStore
+
+        self.store = ObjectStore(ShmStore.open(store_path))
+        # our RPC server (direct task pushes / object pulls)
+        self._server, port = await protocol.serve(self._handle_rpc, self.node_ip, 0)
+        self.addr = (self.node_ip, port)
+        self.raylet = await protocol.connect(self.raylet_addr, self._handle_rpc, name="->raylet")
+        if self.mode == WORKER_MODE:
+            self.raylet.on_close = lambda c: os._exit(0)  # fate-share with raylet
+            await self.raylet.call(
+                "register_worker", {"worker_id": self.worker_id, "addr": list(self.addr), "pid": os.getpid()}
+            )
+        reply = await self.gcs.call(
+            "register_worker",
+            {
+                "worker_id": self.worker_id,
+                "addr": list(self.addr),
+                "node_id": self.node_id,
+                "is_driver": is_driver,
+                "pid": os.getpid(),
+            },
+        )
+        self.job_id = reply.get("job_id") or 0
+        await self.gcs.call("subscribe", {"channels": ["ACTOR", "NODE"]})
+
+    def connect_local_mode(self):
+        self.job_id = 1
+        self.connected = True
+        from ant_ray_amd._private.local_mode import LocalModeExecutor
+
+        self.executor = LocalModeExecutor(self)
+
+    # =============================================================== RPC serve
+    async def _handle_rpc(self, conn, method, p):
+        fn = getattr(self, "rpc_" + method, None)
+        if fn is None:
+            raise ValueError(f"unknown worker method {method}")
+        return await fn(conn, p or {})
+
+    async def rpc_pub(self, conn, p):
+        """GCS pubsub push."""
+        if p["channel"] == "ACTOR":
+            self._on_actor_update(p["data"])
+        return None
+
+    def _on_actor_update(self, view: dict):
+        st = self._actors.get(view["actor_id"])
+        if st is None:
+            return
+        with st.lock:
+            st.state = view["state"]
+            st.death_cause = view.get("death_cause", "")
+            new_addr = tuple(view["addr"]) if view.get("addr") else None
+            if new_addr != st.addr:
+                st.addr = new_addr
+                st.conn = None  # reconnect lazily
+        if view["state"] == "DEAD":
+            self._fail_actor_tasks(st)
+        elif view["state"] == "ALIVE":
+            self.io.loop.create_task(self._drain_actor_queue(st))
+
+    async def rpc_set_lease(self, conn, p):
+        gpu_ids = p.get("gpu_ids") or []
+        self.gpu_ids = gpu_ids
+        if gpu_ids:
+            os.environ["HIP_VISIBLE_DEVICES"] = ",".join(str(g) for g in gpu_ids)
+            os.environ["CUDA_VISIBLE_DEVICES"] = os.environ["HIP_VISIBLE_DEVICES"]
+        env = (p.get("runtime_env") or {}).get("env_vars") or {}
+        os.environ.update({str(k): str(v) for k, v in env.items()})
+        return {"ok": True}
+
+    async def rpc_exit_worker(self, conn, p):
+        logger.info("worker exiting: %s", p.get("reason"))
+        if self.executor is not None:
+            self.executor.request_exit()
+        else:
+            os._exit(0)
+        return {"ok": True}
+
+    async def rpc_push_task(self, conn, p):
+        """Execute a pushed task (worker mode). Returns the completion reply."""
+        if self.executor is None:
+            raise RaySystemError("this process does not execute tasks")
+        return await self.executor.submit(p)
+
+    async def rpc_ping(self, conn, p):
+        return {"ok": True, "worker_id": self.worker_id}
+
+    async def rpc_pull_object(self, conn, p):
+        """Owner-mediated object fetch (cross-node data plane)."""
+        oid = p["oid"]
+        found, v = self.store.memory.get_now(oid)
+        if found and isinstance(v, _RawResult):
+            return {"data": v.data, "meta": v.meta}
+        if found and v is not IN_PLASMA and not isinstance(v, _ErrorResult):
+            sobj = serialization.serialize(v)
+            return {"data": sobj.to_bytes(), "meta": sobj.metadata}
+        if self.store.shm is not None:
+            buf, meta = self.store.shm.get_buffer(oid, 0.0)
+            if buf is not None:
+                return {"data": bytes(memoryview(buf)), "meta": bytes(meta)}
+        holder = self._object_locations.get(oid)
+        if holder is not None and tuple(holder) != self.addr:
+            return {"redirect": list(holder)}
+        return {"missing": True}
+
+    async def rpc_free_objects(self, conn, p):
+        self.store.free(p["oids"])
+        return {"ok": True}
+
+    # ================================================================== puts
+    def put(self, value, *, _owner_inline=False) -> ObjectRef:
+        with self._lock:
+            self._put_index += 1
+            oid = ObjectID.for_put(WorkerID(self.worker_id), self._put_index).binary()
+        if self.mode == LOCAL_MODE:
+            self.store.memory.put(oid, value)
+            return ObjectRef(oid, None, worker=self)
+        sobj = serialization.serialize(value)
+        self._register_escapes(sobj)
+        if _owner_inline and sobj.total_size <= INLINE_OBJECT_MAX:
+            self.store.memory.put(oid, _RawResult(sobj.to_bytes(), sobj.metadata))
+        else:
+            self.store.put_serialized_to_shm(oid, sobj)
+            self.store.memory.put(oid, IN_PLASMA)
+        with self._lock:
+            self._owned[oid] = {"escaped": False, "size": sobj.total_size}
+        return ObjectRef(oid, self.addr, worker=self)
+
+    def _register_escapes(self, sobj: serialization.SerializedObject):
+        # refs serialized into a stored value may be read anywhere -> pin them
+        for ref in sobj.contained_refs:
+            with self._lock:
+                info = self._owned.get(ref.binary())
+                if info is not None:
+                    info["escaped"] = True
+
+    # =================================================================== gets
+    def get(self, refs: Sequence[ObjectRef], timeout: Optional[float] = None):
+        deadline = None if timeout is None else time.monotonic() + timeout
+        out = []
+        for ref in refs:
+            remaining = None if deadline is None else max(0.0, deadline - time.monotonic())
+            out.append(self._get_one(ref, remaining))
+        return out
+
+    def _resolve_memory_entry(self, v):
+        if isinstance(v, _RawResult):
+            value = serialization.deserialize(memoryview(v.data), v.meta)
+            if v.meta in (serialization.META_ERROR, serialization.META_ACTOR_DIED):
+                if isinstance(value, RayTaskError):
+                    raise value.as_instanceof_cause()
+                raise value if isinstance(value, BaseException) else RayTaskError(cause=value)
+            return value
+        if isinstance(v, _ErrorResult):
+            if isinstance(v.exc, RayTaskError):
+                raise v.exc.as_instanceof_cause()
+            raise v.exc
+        return v
+
+    def _get_one(self, ref: ObjectRef, timeout: Optional[float]):
+        oid = ref.binary()
+        # 1) memory store (inlined results / local-mode values)
+        found, v = self.store.memory.get_now(oid)
+        if found and v is not IN_PLASMA:
+            value = self._resolve_memory_entry(v)
+            if isinstance(value, RayTaskError):
+                raise value.as_instanceof_cause()
+            return value
+        if self.store.memory.is_pending(oid):
+            ok = self.store.memory.wait(oid, timeout)
+            if not ok:
+                raise GetTimeoutError(f"Get timed out on {oid.hex()[:16]}")
+            found, v = self.store.memory.get_now(oid)
+            if found and v is not IN_PLASMA:
+                value = self._resolve_memory_entry(v)
+                if isinstance(value, RayTaskError):
+                    raise value.as_instanceof_cause()
+                return value
+        # 2) shm store (+ owner-mediated pull fallback)
+        deadline = None if timeout is None else time.monotonic() + timeout
+        pull_addr = self._object_locations.get(oid) or (
+            tuple(ref.owner_addr) if ref.owner_addr else None
+        )
+        attempt = 0
+        while True:
+            slice_t = 0.2
+            if deadline is not None:
+                slice_t = min(slice_t, max(0.0, deadline - time.monotonic()))
+            if self.store.shm is not None:
+                buf, meta = self.store.shm.get_buffer(oid, slice_t)
+                if buf is not None:
+                    return self._deserialize_buffer(buf, bytes(meta))
+            if pull_addr is not None and tuple(pull_addr) != self.addr:
+                value, ok = self._try_pull(oid, pull_addr)
+                if ok:
+                    return value
+            attempt += 1
+            if deadline is not None and time.monotonic() >= deadline:
+                raise GetTimeoutError(f"Get timed out on object {oid.hex()[:16]}")
+            if self.store.shm is None:
+                time.sleep(0.05)
+
+    def _deserialize_buffer(self, buf, meta: bytes):
+        value = serialization.deserialize(memoryview(buf), meta)
+        if meta in (serialization.META_ERROR, serialization.META_ACTOR_DIED):
+            if isinstance(value, RayTaskError):
+                raise value.as_instanceof_cause()
+            raise value if isinstance(value, BaseException) else RaySystemError(str(value))
+        return value
+
+    def _try_pull(self, oid: bytes, addr):
+        """Fetch object bytes from a holder (owner or redirect target)."""
+        for _ in range(3):
+            try:
+                conn = self._get_worker_conn(tuple(addr))
+                reply = self.io.run(conn.call("pull_object", {"oid": oid}, timeout=30), timeout=35)
+            except Exception:
+                return None, False
+            if reply.get("data") is not None:
+                data = reply["data"]
+                meta = reply.get("meta", serialization.META_PICKLE)
+                value = serialization.deserialize(memoryview(data), meta)
+                if meta in (serialization.META_ERROR, serialization.META_ACTOR_DIED):
+                    if isinstance(value, RayTaskError):
+                        raise value.as_instanceof_cause()
+                    raise value
+                return value, True
+            if reply.get("redirect"):
+                addr = tuple(reply["redirect"])
+                continue
+            return None, False
+        return None, False
+
+    def get_async(self, ref: ObjectRef):
+        """concurrent.futures.Future for a ref (used by ObjectRef.future())."""
+        import concurrent.futures
+
+        fut = concurrent.futures.Future()
+
+        def run():
+            try:
+                fut.set_result(self._get_one(ref, None))
+            except BaseException as e:  # noqa: BLE001
+                fut.set_exception(e)
+
+        threading.Thread(target=run, daemon=True).start()
+        return fut
+
+    # =================================================================== wait
+    def wait(self, refs: Sequence[ObjectRef], num_returns=1, timeout=None, fetch_local=True):
+        deadline = None if timeout is None else time.monotonic() + timeout
+        pending = list(refs)
+        ready: List[ObjectRef] = []
+        while True:
+            still = []
+            for ref in pending:
+                oid = ref.binary()
+                found, v = self.store.memory.get_now(oid)
+                ok = found and v is not IN_PLASMA
+                if not ok and self.store.shm is not None and self.store.shm.contains(oid):
+                    ok = True
+                (ready if ok else still).append(ref)
+            pending = still
+            if len(ready) >= num_returns or not pending:
+                break
+            if deadline is not None and time.monotonic() >= deadline:
+                break
+            time.sleep(0.005)
+        if len(ready) > num_returns:
+            pending = ready[num_returns:] + pending
+            ready = ready[:num_returns]
+        return ready, pending
+
+    # ============================================================ task submit
+    def _scheduling_key(self, fn_id: bytes, opts: dict) -> str:
+        res = opts.get("resources") or {}
+        return fn_id.hex() + "|" + ",".join(f"{k}={v}" for k, v in sorted(res.items()))
+
+    def submit_task(self, fn, fn_id: bytes, args, kwargs, opts: dict) -> List[ObjectRef]:
+        n_returns = opts.get("num_returns", 1)
+        task_id = TaskID.for_task(JobID.from_int(self.job_id or 0)).binary()
+        refs = [
+            ObjectRef(ObjectID.for_return(TaskID(task_id), i).binary(), self.addr, worker=self)
+            for i in range(n_returns)
+        ]
+        for r in refs:
+            self.store.memory.mark_pending(r.binary())
+        sobj = serialization.serialize((args, kwargs))
+        self._register_escapes(sobj)
+        if fn_id not in self._pushed_fns:
+            import cloudpickle
+
+            self._fn_cache[fn_id] = cloudpickle.dumps(fn)
+            self._pushed_fns.add(fn_id)
+        payload = {
+            "type": "normal",
+            "task_id": task_id,
+            "fn_id": fn_id,
+            "fn": None,
+            "args": sobj.to_bytes(),
+            "n_returns": n_returns,
+            "caller": self.worker_id,
+            "caller_addr": list(self.addr),
+            "name": opts.get("name", ""),
+            "max_retries": opts.get("max_retries", 3),
+        }
+        resources = dict(opts.get("resources") or {})
+        resources.setdefault("CPU", float(opts.get("num_cpus", 1)))
+        if opts.get("num_gpus"):
+            resources["GPU"] = float(opts["num_gpus"])
+        key = self._scheduling_key(fn_id, {"resources": resources})
+        self.io.submit(self._enqueue_task(key, payload, resources, opts)).result()
+        return refs
+
+    async def _enqueue_task(self, key, payload, resources, opts):
+        self._lease_queue[key].append({"payload": payload, "resources": resources, "opts": opts})
+        await self._pump_tasks(key)
+
+    async def _pump_tasks(self, key):
+        queue = self._lease_queue[key]
+        leases = self._leases[key]
+        # push queued tasks onto least-loaded leased workers
+        while queue:
+            target = None
+            for lw in leases:
+                if lw.in_flight < PIPELINE_DEPTH and (target is None or lw.in_flight < target.in_flight):
+                    target = lw
+            if target is None:
+                break
+            item = queue.pop(0)
+            target.in_flight += 1
+            self.io.loop.create_task(self._push_and_complete(key, target, item))
+        # request more leases if there's still queued work
+        want = len(queue)
+        if want > 0 and self._lease_requests[key] < min(want, 16):
+            self._lease_requests[key] += 1
+            self.io.loop.create_task(self._request_lease(key, queue[0]))
+
+    async def _request_lease(self, key, sample_item):
+        try:
+            reply = await self.raylet.call(
+                "lease_worker",
+                {
+                    "resources": sample_item["resources"],
+                    "pg": sample_item["opts"].get("placement_group"),
+                    "runtime_env": sample_item["opts"].get("runtime_env"),
+                },
+                timeout=None,
+            )
+        except Exception as e:
+            logger.warning("lease request failed: %s", e)
+            self._lease_requests[key] -= 1
+            return
+        self._lease_requests[key] -= 1
+        if not reply.get("granted"):
+            # infeasible locally: fail queued tasks for now (spillback in gcs
+            # handles actors; cross-node normal tasks arrive in round 2)
+            err = RaySystemError(
+                f"no feasible node for resources {sample_item['resources']}"
+            )
+            for item in self._lease_queue[key]:
+                self._fail_task(item["payload"], err)
+            self._lease_queue[key].clear()
+            return
+        conn = self._get_worker_conn_async_cached(tuple(reply["addr"]))
+        conn = await conn
+        lw = LeasedWorker(reply["worker_id"], reply["addr"], reply.get("node_id"), conn)
+        self._leases[key].append(lw)
+        await self._pump_tasks(key)
+        self.io.loop.create_task(self._lease_idle_watch(key, lw))
+
+    async def _lease_idle_watch(self, key, lw: LeasedWorker):
+        while True:
+            await protocol.asyncio.sleep(0.5)
+            if lw.in_flight == 0 and not self._lease_queue[key]:
+                if time.monotonic() - lw.idle_since > LEASE_IDLE_RELEASE_S:
+                    try:
+                        self._leases[key].remove(lw)
+                    except ValueError:
+                        pass
+                    try:
+                        await self.raylet.call("return_worker", {"worker_id": lw.worker_id}, timeout=10)
+                    except Exception:
+                        pass
+                    return
+
+    async def _push_and_complete(self, key, lw: LeasedWorker, item):
+        payload = item["payload"]
+        try:
+            fn_id = payload.get("fn_id")
+            if fn_id is not None and fn_id not in lw.sent_fns:
+                payload = dict(payload, fn=self._fn_cache.get(fn_id))
+                lw.sent_fns.add(fn_id)
+            reply = await lw.conn.call("push_task", payload, timeout=None)
+            if reply.get("status") == "need_fn":
+                payload = dict(payload, fn=self._fn_cache.get(fn_id))
+                reply = await lw.conn.call("push_task", payload, timeout=None)
+            self._handle_task_reply(payload, reply)
+        except Exception as e:  # worker died mid-task
+            retries = payload.get("max_retries", 3)
+            if retries > 0 and not isinstance(e, protocol.RpcError):
+                payload["max_retries"] = retries - 1
+                payload["fn"] = payload.get("fn")  # keep blob for resubmit
+                logger.warning("task %s failed (%s); retrying", payload["task_id"].hex()[:8], e)
+                try:
+                    self._leases[key].remove(lw)
+                except ValueError:
+                    pass
+                await self._enqueue_task(key, payload, item["resources"], item["opts"])
+            else:
+                self._fail_task(payload, RaySystemError(f"task push failed: {e}"))
+            return
+        finally:
+            lw.in_flight -= 1
+            lw.idle_since = time.monotonic()
+        await self._pump_tasks(key)
+
+    def _fail_task(self, payload, exc: BaseException):
+        task_id = payload["task_id"]
+        for i in range(payload.get("n_returns", 1)):
+            oid = ObjectID.for_return(TaskID(task_id), i).binary()
+            self.store.memory.put(oid, _ErrorResult(exc))
+
+    def _handle_task_reply(self, payload, reply):
+        if reply.get("status") == "ok":
+            for r in reply.get("results", []):
+                oid = r["oid"]
+                if r.get("inline") is not None:
+                    self.store.memory.put(oid, _RawResult(r["inline"], r.get("meta", b"py")))
+                else:
+                    self.store.memory.put(oid, IN_PLASMA)
+                    if r.get("holder"):
+                        self._object_locations[oid] = tuple(r["holder"])
+        else:
+            data = reply.get("error_payload")
+            meta = reply.get("error_meta", serialization.META_ERROR)
+            for i in range(payload.get("n_returns", 1)):
+                oid = ObjectID.for_return(TaskID(payload["task_id"]), i).binary()
+                if data is not None:
+                    self.store.memory.put(oid, _RawResult(data, meta))
+                else:
+                    self.store.memory.put(
+                        oid, _ErrorResult(RaySystemError(reply.get("error", "task failed")))
+                    )
+
+    # ============================================================= actor path
+    def create_actor(self, cls, actor_id: bytes, args, kwargs, opts: dict):
+        import cloudpickle
+
+        sobj = serialization.serialize((args, kwargs))
+        self._register_escapes(sobj)
+        payload = {
+            "type": "actor_create",
+            "task_id": TaskID.for_task(JobID.from_int(self.job_id or 0)).binary(),
+            "cls": cloudpickle.dumps(cls),
+            "args": sobj.to_bytes(),
+            "caller": self.worker_id,
+            "caller_addr": list(self.addr),
+            "max_concurrency": opts.get("max_concurrency", 1),
+            "n_returns": 0,
+        }
+        st = ActorHandleState(actor_id)
+        st.is_owner = True
+        st.detached = opts.get("lifetime") == "detached"
+        with self._lock:
+            self._actors[actor_id] = st
+        reply = self.io.run(
+            self.gcs.call(
+                "create_actor",
+                {
+                    "actor_id": actor_id,
+                    "owner": self.worker_id,
+                    "name": opts.get("name"),
+                    "namespace": opts.get("namespace"),
+                    "opts": {
+                        k: v
+                        for k, v in opts.items()
+                        if k
+                        in (
+                            "num_cpus", "num_gpus", "resources", "max_restarts",
+                            "max_task_retries", "max_concurrency", "lifetime",
+                            "placement_group", "runtime_env", "_node_affinity",
+                            "_scheduling_timeout",
+                        )
+                    },
+                    "create_payload": payload,
+                    "get_if_exists": opts.get("get_if_exists", False),
+                },
+                timeout=60,
+            ),
+            timeout=65,
+        )
+        return reply
+
+    def _get_actor_state(self, actor_id: bytes) -> ActorHandleState:
+        with self._lock:
+            st = self._actors.get(actor_id)
+            if st is None:
+                st = ActorHandleState(actor_id)
+                self._actors[actor_id] = st
+            return st
+
+    def submit_actor_task(self, actor_id: bytes, method_name: str, args, kwargs, opts: dict):
+        n_returns = opts.get("num_returns", 1)
+        st = self._get_actor_state(actor_id)
+        with st.lock:
+            st.seq += 1
+            seq = st.seq
+        task_id = TaskID.for_actor_task(ActorID(actor_id), seq).binary()
+        refs = [
+            ObjectRef(ObjectID.for_return(TaskID(task_id), i).binary(), self.addr, worker=self)
+            for i in range(n_returns)
+        ]
+        for r in refs:
+            self.store.memory.mark_pending(r.binary())
+        sobj = serialization.serialize((args, kwargs))
+        self._register_escapes(sobj)
+        payload = {
+            "type": "actor_task",
+            "task_id": task_id,
+            "actor_id": actor_id,
+            "method": method_name,
+            "args": sobj.to_bytes(),
+            "n_returns": n_returns,
+            "seq": seq,
+            "caller": self.worker_id,
+            "caller_addr": list(self.addr),
+            "concurrency_group": opts.get("concurrency_group"),
+        }
+        self.io.submit(self._submit_actor_async(st, payload)).result()
+        return refs
+
+    async def _submit_actor_async(self, st: ActorHandleState, payload):
+        with st.lock:
+            if st.state == "DEAD":
+                self._fail_task(
+                    payload,
+                    ActorDiedError(f"actor {st.actor_id.hex()[:8]} is dead: {st.death_cause}"),
+                )
+                return
+            st.pending.append(payload)
+        await self._drain_actor_queue(st)
+
+    async def _drain_actor_queue(self, st: ActorHandleState):
+        if st.addr is None:
+            # resolve address from GCS (once; pubsub keeps it fresh after)
+            view = await self.gcs.call("get_actor", {"actor_id": st.actor_id})
+            if view is None:
+                view = {"state": "DEAD", "death_cause": "unknown actor"}
+            with st.lock:
+                st.state = view["state"]
+                st.death_cause = view.get("death_cause", "")
+                st.addr = tuple(view["addr"]) if view.get("addr") else None
+            if st.state == "DEAD":
+                self._fail_actor_tasks(st)
+                return
+            if st.addr is None:
+                # still pending creation; wait for pubsub or poll
+                self.io.loop.create_task(self._wait_actor_alive(st))
+                return
+        with st.lock:
+            batch = st.pending[:]
+            st.pending.clear()
+            addr = st.addr
+        if not batch:
+            return
+        try:
+            conn = await self._get_worker_conn_async_cached(addr)
+        except Exception:
+            with st.lock:
+                st.pending = batch + st.pending
+                st.addr = None
+                st.conn = None
+            self.io.loop.create_task(self._wait_actor_alive(st))
+            return
+        for payload in batch:
+            self.io.loop.create_task(self._push_actor_task(st, conn, payload))
+
+    async def _wait_actor_alive(self, st: ActorHandleState):
+        view = await self.gcs.call("wait_actor_ready", {"actor_id": st.actor_id}, timeout=None)
+        if view is None:
+            view = {"state": "DEAD", "death_cause": "unknown actor"}
+        with st.lock:
+            st.state = view["state"]
+            st.death_cause = view.get("death_cause", "")
+            st.addr = tuple(view["addr"]) if view.get("addr") else None
+            if view.get("creation_error"):
+                st.creation_error = view["creation_error"]
+        if st.state == "DEAD":
+            self._fail_actor_tasks(st)
+        else:
+            await self._drain_actor_queue(st)
+
+    async def _push_actor_task(self, st: ActorHandleState, conn, payload):
+        with st.lock:
+            st.inflight[payload["seq"]] = payload
+        try:
+            reply = await conn.call("push_task", payload, timeout=None)
+            self._handle_task_reply(payload, reply)
+            with st.lock:
+                st.inflight.pop(payload["seq"], None)
+        except Exception:
+            # connection to actor lost: wait for GCS verdict (restart/dead)
+            with st.lock:
+                st.conn = None
+                st.addr = None
+            self.io.loop.create_task(self._wait_actor_failure_verdict(st))
+
+    async def _wait_actor_failure_verdict(self, st: ActorHandleState):
+        # Ask GCS until state changes away from ALIVE-with-old-addr
+        for _ in range(600):
+            view = await self.gcs.call("get_actor", {"actor_id": st.actor_id})
+            if view is None:
+                break
+            state = view["state"]
+            if state == "DEAD":
+                with st.lock:
+                    st.state = "DEAD"
+                    st.death_cause = view.get("death_cause", "")
+                self._fail_actor_tasks(st, include_inflight=True)
+                return
+            if state == "ALIVE" and view.get("addr"):
+                addr = tuple(view["addr"])
+                with st.lock:
+                    st.addr = addr
+                    st.state = "ALIVE"
+                    inflight = list(st.inflight.values())
+                    st.inflight.clear()
+                # default semantics: in-flight tasks on a restarted actor fail
+                for payload in inflight:
+                    self._fail_task(
+                        payload,
+                        RayActorError(
+                            f"actor {st.actor_id.hex()[:8]} restarted; task lost"
+                        ),
+                    )
+                await self._drain_actor_queue(st)
+                return
+            await protocol.asyncio.sleep(0.2)
+
+    def _fail_actor_tasks(self, st: ActorHandleState, include_inflight=True):
+        err = ActorDiedError(
+            f"The actor {st.actor_id.hex()[:8]} died: {st.death_cause}"
+        )
+        if getattr(st, "creation_error", None):
+            try:
+                payload = st.creation_error
+                exc = serialization.deserialize(memoryview(payload), serialization.META_ERROR)
+                if isinstance(exc, BaseException):
+                    err = ActorDiedError(
+                        f"The actor died because of an error raised in its creation task:\n"
+                        f"{getattr(exc, 'traceback_str', exc)}"
+                    )
+            except Exception:
+                pass
+        with st.lock:
+            pending = st.pending[:]
+            st.pending.clear()
+            inflight = list(st.inflight.values()) if include_inflight else []
+            st.inflight.clear()
+        for payload in pending + inflight:
+            self._fail_task(payload, err)
+
+    def kill_actor(self, actor_id: bytes, no_restart=True):
+        self.io.run(
+            self.gcs.call("kill_actor", {"actor_id": actor_id, "no_restart": no_restart}, timeout=30),
+            timeout=35,
+        )
+
+    def actor_handle_added(self, actor_id: bytes):
+        st = self._get_actor_state(actor_id)
+        with st.lock:
+            st.handle_count += 1
+
+    def actor_handle_removed(self, actor_id: bytes):
+        st = self._actors.get(actor_id)
+        if st is None:
+            return
+        with st.lock:
+            st.handle_count -= 1
+            should_kill = (
+                st.handle_count <= 0 and st.is_owner and not st.detached
+            )
+        if should_kill and self.connected and self.io is not None:
+            try:
+                self.io.submit(
+                    self.gcs.call("actor_out_of_scope", {"actor_id": actor_id})
+                )
+            except Exception:
+                pass
+
+    # ========================================================== conn caching
+    async def _get_worker_conn_async_cached(self, addr):
+        conn = self._worker_conns.get(addr)
+        if conn is None or conn.closed:
+            conn = await protocol.connect(addr, self._handle_rpc, name=f"->worker{addr}")
+            self._worker_conns[addr] = conn
+        return conn
+
+    def _get_worker_conn(self, addr):
+        return self.io.run(self._get_worker_conn_async_cached(addr), timeout=30)
+
+    # ============================================================== refcounts
+    def add_local_ref(self, oid: bytes):
+        with self._lock:
+            self._local_refs[oid] += 1
+
+    def remove_local_ref(self, oid: bytes):
+        try:
+            with self._lock:
+                self._local_refs[oid] -= 1
+                if self._local_refs[oid] > 0:
+                    return
+                del self._local_refs[oid]
+                info = self._owned.pop(oid, None)
+                if info is None or info.get("escaped"):
+                    return
+            # owned, unreferenced, never escaped -> free storage
+            self.store.free([oid])
+        except Exception:
+            pass
+
+    # ================================================================= close
+    def shutdown(self):
+        self.connected = False
+        for h in self._shutdown_handlers:
+            try:
+                h()
+            except Exception:
+                pass
+        if self.io is not None:
+            self.io.stop()
+
+
+class Worker:
+    """Global per-process worker state (parity: python/ray/_private/worker.py:461)."""
+
+    def __init__(self):
+        self.core_worker: Optional[CoreWorker] = None
+        self.mode: Optional[str] = None
+        self.namespace: str = ""
+        self._head_proc = None
+        self.session_dir = ""
+
+    @property
+    def connected(self):
+        return self.core_worker is not None and self.core_worker.connected
+
+    def add_local_ref(self, oid):
+        if self.core_worker:
+            self.core_worker.add_local_ref(oid)
+
+    def remove_local_ref(self, oid):
+        if self.core_worker:
+            self.core_worker.remove_local_ref(oid)
+
+
+global_worker = Worker()
+
+
+def get_global_worker() -> Worker:
+    return global_worker
