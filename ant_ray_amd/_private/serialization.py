@@ -105,7 +105,7 @@ def _rebuild_torch_tensor(buf, dtype_str, shape):
     import torch
 
     dtype = getattr(torch, dtype_str.replace("torch.", ""))
-    mv = buf.raw()
+    mv = buf.raw() if isinstance(buf, pickle.PickleBuffer) else memoryview(buf)
     np_arr = np.frombuffer(mv, dtype=np.uint8)
     with warnings.catch_warnings():
         # zero-copy view over (possibly read-only) shm memory; writes would
@@ -141,6 +141,38 @@ class _Pickler(cloudpickle.CloudPickler):
 _BUFFER_THRESHOLD = 512  # buffers below this get pickled in-band
 
 
+def _make_dispatch(contained_refs):
+    """copyreg-style dispatch table for the C pickler fast path."""
+    from ant_ray_amd._private.object_ref import ObjectRef
+
+    table = {}
+
+    def reduce_ref(obj):
+        contained_refs.append(obj)
+        return (ObjectRef._rehydrate, (obj.binary(), obj.owner_addr))
+
+    table[ObjectRef] = reduce_ref
+    try:
+        import torch
+
+        def reduce_tensor(t):
+            r = _torch_cpu_tensor_reducer(t)
+            if r is NotImplemented:
+                return t.__reduce_ex__(5)
+            return r
+
+        table[torch.Tensor] = reduce_tensor
+        try:
+            from torch.nn import Parameter
+
+            table[Parameter] = lambda t: t.__reduce_ex__(5)
+        except ImportError:
+            pass
+    except ImportError:
+        pass
+    return table
+
+
 def serialize(value: Any, metadata: bytes = META_PICKLE) -> SerializedObject:
     import io
 
@@ -153,6 +185,18 @@ def serialize(value: Any, metadata: bytes = META_PICKLE) -> SerializedObject:
         buffers.append(mv)
         return False
 
+    # Fast path: C pickler (ObjectRef / plain torch tensors via dispatch
+    # table). Functions/lambdas/local classes raise -> cloudpickle fallback.
+    contained_refs: list = []
+    f = io.BytesIO()
+    try:
+        p = pickle.Pickler(f, protocol=5, buffer_callback=cb)
+        p.dispatch_table = _make_dispatch(contained_refs)
+        p.dump(value)
+        return SerializedObject(f.getvalue(), buffers, contained_refs, metadata)
+    except (pickle.PicklingError, TypeError, AttributeError):
+        pass
+    buffers.clear()
     f = io.BytesIO()
     p = _Pickler(f, buffer_callback=cb)
     p.dump(value)
