@@ -1,0 +1,258 @@
+"""Core API semantics, local mode + distributed single node.
+
+Modeled on the reference's python/ray/tests/test_basic*.py coverage.
+"""
+import time
+
+import numpy as np
+import pytest
+
+
+class TestLocalMode:
+    def test_task_roundtrip(self, ray_local):
+        ray = ray_local
+
+        @ray.remote
+        def f(x):
+            return x + 1
+
+        assert ray.get(f.remote(1)) == 2
+
+    def test_actor_roundtrip(self, ray_local):
+        ray = ray_local
+
+        @ray.remote
+        class C:
+            def __init__(self, v):
+                self.v = v
+
+            def add(self, k):
+                self.v += k
+                return self.v
+
+        c = C.remote(5)
+        assert ray.get(c.add.remote(3)) == 8
+
+    def test_error(self, ray_local):
+        ray = ray_local
+
+        @ray.remote
+        def boom():
+            raise KeyError("nope")
+
+        with pytest.raises(KeyError):
+            ray.get(boom.remote())
+
+    def test_multi_returns(self, ray_local):
+        ray = ray_local
+
+        @ray.remote(num_returns=2)
+        def two():
+            return 1, 2
+
+        a, b = two.remote()
+        assert ray.get([a, b]) == [1, 2]
+
+    def test_nested_refs(self, ray_local):
+        ray = ray_local
+        r = ray.put(41)
+
+        @ray.remote
+        def f(x):
+            return x + 1
+
+        assert ray.get(f.remote(r)) == 42
+
+
+@pytest.mark.usefixtures("ray_start_regular_module")
+class TestDistributed:
+    def test_tasks(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+
+        @ray.remote
+        def sq(x):
+            return x * x
+
+        assert ray.get([sq.remote(i) for i in range(10)]) == [i * i for i in range(10)]
+
+    def test_put_get_large(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+        arr = np.random.rand(500_000)
+        ref = ray.put(arr)
+        out = ray.get(ref)
+        assert np.array_equal(out, arr)
+
+    def test_task_large_arg_and_return(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+
+        @ray.remote
+        def double(a):
+            return a * 2
+
+        arr = np.ones(400_000)
+        out = ray.get(double.remote(ray.put(arr)))
+        assert out.sum() == 800_000
+
+    def test_actor_state_and_order(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+
+        @ray.remote
+        class Counter:
+            def __init__(self):
+                self.n = 0
+
+            def incr(self):
+                self.n += 1
+                return self.n
+
+        c = Counter.remote()
+        results = ray.get([c.incr.remote() for _ in range(20)])
+        assert results == list(range(1, 21))
+
+    def test_actor_error_propagation(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+
+        @ray.remote
+        class Bad:
+            def fail(self):
+                raise ValueError("actor boom")
+
+        b = Bad.remote()
+        with pytest.raises(ValueError):
+            ray.get(b.fail.remote())
+
+    def test_actor_init_failure(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+        from ant_ray_amd.exceptions import RayActorError
+
+        @ray.remote
+        class BadInit:
+            def __init__(self):
+                raise RuntimeError("init boom")
+
+            def m(self):
+                return 1
+
+        b = BadInit.remote()
+        with pytest.raises(RayActorError):
+            ray.get(b.m.remote(), timeout=30)
+
+    def test_wait(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+
+        @ray.remote
+        def fast():
+            return 1
+
+        @ray.remote
+        def slow():
+            time.sleep(5)
+            return 2
+
+        a, b = fast.remote(), slow.remote()
+        ready, not_ready = ray.wait([a, b], num_returns=1, timeout=3)
+        assert ready == [a] and not_ready == [b]
+
+    def test_get_timeout(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+        from ant_ray_amd.exceptions import GetTimeoutError
+
+        @ray.remote
+        def slow():
+            time.sleep(10)
+
+        with pytest.raises(GetTimeoutError):
+            ray.get(slow.remote(), timeout=0.5)
+
+    def test_named_actor(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+
+        @ray.remote
+        class Reg:
+            def who(self):
+                return "reg"
+
+        Reg.options(name="registry").remote()
+        h = ray.get_actor("registry")
+        assert ray.get(h.who.remote()) == "reg"
+
+    def test_async_actor(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+
+        @ray.remote
+        class Async:
+            async def work(self, x):
+                import asyncio
+
+                await asyncio.sleep(0.01)
+                return x * 10
+
+        a = Async.remote()
+        assert ray.get([a.work.remote(i) for i in range(5)]) == [0, 10, 20, 30, 40]
+
+    def test_threaded_actor(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+
+        @ray.remote(max_concurrency=4)
+        class T:
+            def slow(self):
+                time.sleep(0.3)
+                return 1
+
+        t = T.remote()
+        ray.get(t.slow.remote())  # warm: actor worker spawn happens here
+        t0 = time.time()
+        assert sum(ray.get([t.slow.remote() for _ in range(4)])) == 4
+        assert time.time() - t0 < 1.1  # concurrent, not 1.2s serial
+
+    def test_nested_tasks(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+
+        @ray.remote
+        def inner(x):
+            return x + 1
+
+        @ray.remote
+        def outer(x):
+            import ant_ray_amd as ray2
+
+            return ray2.get(inner.remote(x)) + 100
+
+        assert ray.get(outer.remote(1)) == 102
+
+    def test_actor_handle_passing(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+
+        @ray.remote
+        class Holder:
+            def __init__(self):
+                self.v = 7
+
+            def get(self):
+                return self.v
+
+        @ray.remote
+        def reader(h):
+            import ant_ray_amd as ray2
+
+            return ray2.get(h.get.remote())
+
+        h = Holder.remote()
+        assert ray.get(reader.remote(h)) == 7
+
+    def test_cluster_resources(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+        assert ray.cluster_resources().get("CPU", 0) >= 4
+
+    def test_runtime_context(self, ray_start_regular_module):
+        ray = ray_start_regular_module
+
+        @ray.remote
+        def ctx():
+            import ant_ray_amd as ray2
+
+            c = ray2.get_runtime_context()
+            return (c.get_worker_id(), c.get_task_id())
+
+        wid, tid = ray.get(ctx.remote())
+        assert wid and tid
