@@ -1,0 +1,194 @@
+// Torch bindings for the CDNA4 kernels (host-only C++; device code lives in
+// kernels/*.hip, linked as hipcc-compiled objects — see csrc/build.py).
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void launch_rmsnorm_fwd(const void*, const void*, void*, float*, void*, int64_t,
+                        int, float, hipStream_t);
+void launch_rmsnorm_bwd(const void*, const void*, const void*, const float*,
+                        void*, float*, int64_t, int, hipStream_t);
+void launch_rope(void*, void*, const float*, const float*, int64_t, int64_t,
+                 int, int, int, int64_t, int64_t, int, hipStream_t);
+void launch_swiglu_fwd(const void*, void*, int64_t, int64_t, hipStream_t);
+void launch_swiglu_bwd(const void*, const void*, void*, int64_t, int64_t,
+                       hipStream_t);
+void launch_cross_entropy(void*, const int32_t*, float*, int64_t, int64_t,
+                          float, int32_t, int, hipStream_t);
+void launch_adamw(float*, void*, const void*, float*, float*, int64_t, float,
+                  float, float, float, float, float, float, float, hipStream_t);
+void launch_bf16_scale(void*, float, int64_t, hipStream_t);
+void launch_bf16_to_f32(const void*, float*, int64_t, hipStream_t);
+void launch_f32_to_bf16(const float*, void*, int64_t, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream().stream(); }
+
+void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+std::tuple<torch::Tensor, torch::Tensor> rmsnorm_fwd(
+    torch::Tensor x, torch::Tensor w,
+    c10::optional<torch::Tensor> residual, double eps) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  int64_t H = x.size(-1);
+  int64_t N = x.numel() / H;
+  TORCH_CHECK(H % 8 == 0, "H must be a multiple of 8");
+  auto y = torch::empty_like(x);
+  auto rstd = torch::empty({N}, x.options().dtype(torch::kFloat32));
+  void* res_ptr = nullptr;
+  if (residual.has_value()) {
+    check_bf16(*residual, "residual");
+    res_ptr = residual->data_ptr();
+  }
+  launch_rmsnorm_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                     rstd.data_ptr<float>(), res_ptr, N, (int)H, (float)eps,
+                     cur_stream());
+  return {y, rstd};
+}
+
+std::tuple<torch::Tensor, torch::Tensor> rmsnorm_bwd(torch::Tensor dy,
+                                                     torch::Tensor r,
+                                                     torch::Tensor w,
+                                                     torch::Tensor rstd) {
+  check_bf16(dy, "dy");
+  check_bf16(r, "r");
+  check_bf16(w, "w");
+  int64_t H = dy.size(-1);
+  int64_t N = dy.numel() / H;
+  TORCH_CHECK(H <= 8192, "rmsnorm_bwd supports H<=8192");
+  auto dx = torch::empty_like(dy);
+  auto dw = torch::zeros({H}, dy.options().dtype(torch::kFloat32));
+  launch_rmsnorm_bwd(dy.data_ptr(), r.data_ptr(), w.data_ptr(),
+                     rstd.data_ptr<float>(), dx.data_ptr(),
+                     dw.data_ptr<float>(), N, (int)H, cur_stream());
+  return {dx, dw};
+}
+
+// q/k are [B,S,H,D] views (possibly strided slices of a fused qkv buffer);
+// requires the innermost [H,D] block per token to be contiguous.
+void rope_(torch::Tensor q, torch::Tensor k, torch::Tensor cos_t,
+           torch::Tensor sin_t, bool backward) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(k.is_cuda() && k.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(q.dim() == 4 && k.dim() == 4, "q/k must be [B,S,H,D]");
+  TORCH_CHECK(q.stride(3) == 1 && q.stride(2) == q.size(3), "q [H,D] must be contiguous");
+  TORCH_CHECK(k.stride(3) == 1 && k.stride(2) == k.size(3), "k [H,D] must be contiguous");
+  TORCH_CHECK(q.stride(0) == q.size(1) * q.stride(1), "q batch stride must fold");
+  TORCH_CHECK(k.stride(0) == k.size(1) * k.stride(1), "k batch stride must fold");
+  TORCH_CHECK(cos_t.scalar_type() == torch::kFloat32 && cos_t.is_contiguous());
+  int64_t B = q.size(0), S = q.size(1);
+  int Hq = (int)q.size(2), Hk = (int)k.size(2), D = (int)q.size(3);
+  TORCH_CHECK(cos_t.size(0) >= S && cos_t.size(1) == D / 2, "cos table too small");
+  launch_rope(q.data_ptr(), k.data_ptr(), cos_t.data_ptr<float>(),
+              sin_t.data_ptr<float>(), B, S, Hq, Hk, D, q.stride(1), k.stride(1),
+              backward ? 1 : 0, cur_stream());
+}
+
+torch::Tensor swiglu_fwd(torch::Tensor gu) {
+  check_bf16(gu, "gate_up");
+  int64_t I2 = gu.size(-1);
+  int64_t N = gu.numel() / I2;
+  int64_t I = I2 / 2;
+  TORCH_CHECK(I % 8 == 0, "intermediate size must be a multiple of 8");
+  auto sizes = gu.sizes().vec();
+  sizes.back() = I;
+  auto out = torch::empty(sizes, gu.options());
+  launch_swiglu_fwd(gu.data_ptr(), out.data_ptr(), N, I, cur_stream());
+  return out;
+}
+
+torch::Tensor swiglu_bwd(torch::Tensor dout, torch::Tensor gu) {
+  check_bf16(dout, "dout");
+  check_bf16(gu, "gate_up");
+  int64_t I2 = gu.size(-1);
+  int64_t N = gu.numel() / I2;
+  auto dgu = torch::empty_like(gu);
+  launch_swiglu_bwd(dout.data_ptr(), gu.data_ptr(), dgu.data_ptr(), N, I2 / 2,
+                    cur_stream());
+  return dgu;
+}
+
+torch::Tensor cross_entropy_fwd_bwd(torch::Tensor logits, torch::Tensor targets,
+                                    double grad_scale, int64_t ignore_index,
+                                    bool write_dlogits) {
+  check_bf16(logits, "logits");
+  TORCH_CHECK(targets.scalar_type() == torch::kInt32 && targets.is_contiguous());
+  int64_t V = logits.size(-1);
+  int64_t N = logits.numel() / V;
+  TORCH_CHECK(targets.numel() == N, "targets size mismatch");
+  TORCH_CHECK(V % 8 == 0, "V must be a multiple of 8");
+  auto loss = torch::empty({N}, logits.options().dtype(torch::kFloat32));
+  launch_cross_entropy(logits.data_ptr(), targets.data_ptr<int32_t>(),
+                       loss.data_ptr<float>(), N, V, (float)grad_scale,
+                       (int32_t)ignore_index, write_dlogits ? 1 : 0,
+                       cur_stream());
+  return loss;
+}
+
+void adamw_(torch::Tensor p, torch::Tensor p_bf16, torch::Tensor g,
+            torch::Tensor m, torch::Tensor v, double lr, double b1, double b2,
+            double eps, double wd, int64_t step, double grad_scale) {
+  TORCH_CHECK(p.scalar_type() == torch::kFloat32 && p.is_contiguous());
+  check_bf16(p_bf16, "p_bf16");
+  check_bf16(g, "grad");
+  int64_t n = p.numel();
+  TORCH_CHECK(n % 4 == 0, "flat param count must be a multiple of 4");
+  TORCH_CHECK(p_bf16.numel() == n && g.numel() == n && m.numel() == n &&
+              v.numel() == n);
+  float bc1 = 1.f / (1.f - powf((float)b1, (float)step));
+  float bc2 = 1.f / (1.f - powf((float)b2, (float)step));
+  launch_adamw(p.data_ptr<float>(), p_bf16.data_ptr(), g.data_ptr(),
+               m.data_ptr<float>(), v.data_ptr<float>(), n, (float)lr,
+               (float)b1, (float)b2, (float)eps, (float)wd, bc1, bc2,
+               (float)grad_scale, cur_stream());
+}
+
+void bf16_scale_(torch::Tensor x, double scale) {
+  check_bf16(x, "x");
+  TORCH_CHECK(x.numel() % 8 == 0);
+  launch_bf16_scale(x.data_ptr(), (float)scale, x.numel(), cur_stream());
+}
+
+torch::Tensor bf16_to_f32(torch::Tensor x) {
+  check_bf16(x, "x");
+  TORCH_CHECK(x.numel() % 8 == 0);
+  auto y = torch::empty(x.sizes(), x.options().dtype(torch::kFloat32));
+  launch_bf16_to_f32(x.data_ptr(), y.data_ptr<float>(), x.numel(), cur_stream());
+  return y;
+}
+
+void f32_to_bf16_(torch::Tensor x, torch::Tensor y) {
+  TORCH_CHECK(x.scalar_type() == torch::kFloat32 && x.is_contiguous());
+  check_bf16(y, "y");
+  TORCH_CHECK(x.numel() == y.numel() && x.numel() % 8 == 0);
+  launch_f32_to_bf16(x.data_ptr<float>(), y.data_ptr(), x.numel(), cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused (add+)rmsnorm forward",
+        py::arg("x"), py::arg("w"), py::arg("residual") = py::none(),
+        py::arg("eps") = 1e-5);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rope_", &rope_, py::arg("q"), py::arg("k"), py::arg("cos"),
+        py::arg("sin"), py::arg("backward") = false);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("cross_entropy_fwd_bwd", &cross_entropy_fwd_bwd, py::arg("logits"),
+        py::arg("targets"), py::arg("grad_scale") = 1.0,
+        py::arg("ignore_index") = -100, py::arg("write_dlogits") = true);
+  m.def("adamw_", &adamw_);
+  m.def("bf16_scale_", &bf16_scale_);
+  m.def("bf16_to_f32", &bf16_to_f32);
+  m.def("f32_to_bf16_", &f32_to_bf16_);
+}

@@ -1,0 +1,156 @@
+"""Llama-3 family, MI355X-first implementation.
+
+Design (not a port of any reference model code):
+  * bf16 weights/activations; GEMMs via F.linear (hipBLASLt); everything
+    memory-bound is a hand-written CDNA4 kernel (ant_ray_amd.ops):
+      - fused residual-add + RMSNorm (one HBM pass)
+      - RoPE applied IN PLACE on the fused QKV projection buffer (zero copies)
+      - SwiGLU on the packed gate_up projection
+      - chunked fused vocab-projection + cross-entropy (logits never
+        materialize: 4.2 GB saved at B4xS4096xV128256)
+  * attention: torch SDPA (GQA-aware) — fused flash-style HIP attention is the
+    next kernel on the roadmap
+  * parameters live as views into flat buffers when wrapped by
+    parallel.FlatParamManager (optimizer = one fused kernel pass).
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ant_ray_amd import ops
+
+
+@dataclass
+class LlamaConfig:
+    hidden: int = 4096
+    n_layers: int = 32
+    n_heads: int = 32
+    n_kv_heads: int = 8
+    head_dim: int = 128
+    intermediate: int = 14336
+    vocab: int = 128256
+    rope_theta: float = 500000.0
+    max_seq: int = 4096
+    eps: float = 1e-5
+    tie_embeddings: bool = False
+
+    @classmethod
+    def llama3_8b(cls, max_seq=4096):
+        return cls(max_seq=max_seq)
+
+    @classmethod
+    def tiny(cls, max_seq=512):
+        return cls(hidden=256, n_layers=4, n_heads=4, n_kv_heads=2, head_dim=64,
+                   intermediate=512, vocab=1024, max_seq=max_seq)
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        Hq, Hk, D = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim
+        self.wqkv = nn.Linear(cfg.hidden, (Hq + 2 * Hk) * D, bias=False)
+        self.wo = nn.Linear(Hq * D, cfg.hidden, bias=False)
+
+    def forward(self, y, cos, sin):
+        cfg = self.cfg
+        Hq, Hk, D = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim
+        B, S, _ = y.shape
+        qkv = self.wqkv(y)
+        qkv = ops.rope_qkv(qkv, cos, sin, Hq, Hk, D)
+        q = qkv[..., : Hq * D].view(B, S, Hq, D).transpose(1, 2)
+        k = qkv[..., Hq * D : (Hq + Hk) * D].view(B, S, Hk, D).transpose(1, 2)
+        v = qkv[..., (Hq + Hk) * D :].view(B, S, Hk, D).transpose(1, 2)
+        o = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
+        o = o.transpose(1, 2).reshape(B, S, Hq * D)
+        return self.wo(o)
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.w_gate_up = nn.Linear(cfg.hidden, 2 * cfg.intermediate, bias=False)
+        self.w_down = nn.Linear(cfg.intermediate, cfg.hidden, bias=False)
+
+    def forward(self, y):
+        return self.w_down(ops.swiglu(self.w_gate_up(y)))
+
+
+class LlamaBlock(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.attn_norm = nn.Parameter(torch.ones(cfg.hidden))
+        self.mlp_norm = nn.Parameter(torch.ones(cfg.hidden))
+        self.attn = LlamaAttention(cfg)
+        self.mlp = LlamaMLP(cfg)
+
+    def forward(self, h, res, cos, sin):
+        # h = contribution from the previous sublayer; res = residual stream
+        if res is None:
+            res = h
+            y = ops.rmsnorm(h, self.attn_norm, self.cfg.eps)
+        else:
+            y, res = ops.fused_add_rmsnorm(h, res, self.attn_norm, self.cfg.eps)
+        a = self.attn(y, cos, sin)
+        y2, res = ops.fused_add_rmsnorm(a, res, self.mlp_norm, self.cfg.eps)
+        return self.mlp(y2), res
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, cfg: LlamaConfig, device=None):
+        super().__init__()
+        self.cfg = cfg
+        factory = {"device": device, "dtype": torch.bfloat16}
+        with torch.device(device or "cpu"):
+            self.embed = nn.Embedding(cfg.vocab, cfg.hidden)
+            self.blocks = nn.ModuleList(LlamaBlock(cfg) for _ in range(cfg.n_layers))
+            self.final_norm = nn.Parameter(torch.ones(cfg.hidden))
+            if cfg.tie_embeddings:
+                self.lm_head = None
+            else:
+                self.lm_head = nn.Linear(cfg.hidden, cfg.vocab, bias=False)
+        self.to(dtype=torch.bfloat16)
+        if device is not None:
+            self.to(device)
+        self._init_weights()
+        cos, sin = ops.rope_tables(cfg.head_dim, cfg.max_seq, cfg.rope_theta,
+                                   device=device or "cpu")
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+
+    def _init_weights(self):
+        std = 0.02
+        out_std = std / math.sqrt(2 * self.cfg.n_layers)
+        for name, p in self.named_parameters():
+            if p.dim() >= 2:
+                s = out_std if ("wo" in name or "w_down" in name) else std
+                nn.init.normal_(p, mean=0.0, std=s)
+
+    @property
+    def head_weight(self):
+        return self.embed.weight if self.lm_head is None else self.lm_head.weight
+
+    def forward(self, tokens, targets=None):
+        B, S = tokens.shape
+        h = self.embed(tokens)
+        res = None
+        cos, sin = self.rope_cos, self.rope_sin
+        for blk in self.blocks:
+            h, res = blk(h, res, cos, sin)
+        if res is None:
+            y = ops.rmsnorm(h, self.final_norm, self.cfg.eps)
+        else:
+            y, _ = ops.fused_add_rmsnorm(h, res, self.final_norm, self.cfg.eps)
+        flat = y.reshape(B * S, self.cfg.hidden)
+        if targets is not None:
+            return ops.linear_cross_entropy(flat, self.head_weight, targets.reshape(-1))
+        return (flat @ self.head_weight.t()).view(B, S, -1)
+
+    def num_params(self):
+        return sum(p.numel() for p in self.parameters())

@@ -1,0 +1,158 @@
+"""Autograd wrappers over the CDNA4 kernels (GPU) / torch references (CPU).
+
+These replace the library-call sites the reference delegates to PyTorch
+(the reference ships zero CUDA kernels) with hand-written
+HIP kernels on the MI355X path.
+"""
+from __future__ import annotations
+
+import torch
+
+from ant_ray_amd.ops import reference as ref
+
+
+def _hip():
+    from ant_ray_amd.ops import hip_ops
+
+    return hip_ops()
+
+
+def _use_hip(t: torch.Tensor) -> bool:
+    return t.is_cuda
+
+
+# ---------------------------------------------------------------- rmsnorm
+
+
+class _RMSNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, eps):
+        y, rstd = _hip().rmsnorm_fwd(x, w, None, eps)
+        ctx.save_for_backward(x, w, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, rstd = ctx.saved_tensors
+        dx, dw = _hip().rmsnorm_bwd(dy.contiguous(), x, w, rstd)
+        return dx, dw.to(w.dtype), None
+
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if _use_hip(x):
+        return _RMSNormFn.apply(x.contiguous(), w, eps)
+    # CPU reference with autograd
+    xf = x.float()
+    rstd = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (xf * rstd * w.float()).to(x.dtype)
+
+
+class _FusedAddRMSNormFn(torch.autograd.Function):
+    """(x, residual) -> (y, h) where h = x+residual (bf16-rounded),
+    y = rmsnorm(h)*w. The kernel updates `residual` in place to h."""
+
+    @staticmethod
+    def forward(ctx, x, residual, w, eps):
+        y, rstd = _hip().rmsnorm_fwd(x, w, residual, eps)
+        ctx.save_for_backward(residual, w, rstd)
+        ctx.mark_dirty(residual)
+        return y, residual
+
+    @staticmethod
+    def backward(ctx, dy, dh_out):
+        h, w, rstd = ctx.saved_tensors
+        dx, dw = _hip().rmsnorm_bwd(dy.contiguous(), h, w, rstd)
+        if dh_out is not None:
+            dx = dx + dh_out
+        return dx, dx, dw.to(w.dtype), None
+
+
+def fused_add_rmsnorm(x, residual, w, eps: float = 1e-5):
+    """Returns (normalized, new_residual). new_residual aliases `residual`'s
+    storage on GPU (in-place update) — callers must treat the old value as
+    consumed."""
+    if _use_hip(x):
+        return _FusedAddRMSNormFn.apply(x.contiguous(), residual, w, eps)
+    h = (x.float() + residual.float()).to(x.dtype)
+    return rmsnorm(h, w, eps), h
+
+
+# ------------------------------------------------------------------- rope
+
+
+class _RopeQKVFn(torch.autograd.Function):
+    """Rotates the q,k regions of the fused qkv buffer in place."""
+
+    @staticmethod
+    def forward(ctx, qkv, cos, sin, Hq, Hk, D):
+        B, S, _ = qkv.shape
+        q = qkv[:, :, : Hq * D].view(B, S, Hq, D)
+        k = qkv[:, :, Hq * D : (Hq + Hk) * D].view(B, S, Hk, D)
+        _hip().rope_(q, k, cos, sin, False)
+        ctx.meta = (Hq, Hk, D)
+        ctx.tables = (cos, sin)
+        ctx.mark_dirty(qkv)
+        return qkv
+
+    @staticmethod
+    def backward(ctx, dqkv):
+        Hq, Hk, D = ctx.meta
+        cos, sin = ctx.tables
+        B, S, _ = dqkv.shape
+        dqkv = dqkv.contiguous()
+        dq = dqkv[:, :, : Hq * D].view(B, S, Hq, D)
+        dk = dqkv[:, :, Hq * D : (Hq + Hk) * D].view(B, S, Hk, D)
+        _hip().rope_(dq, dk, cos, sin, True)
+        return dqkv, None, None, None, None, None
+
+
+def rope_qkv(qkv: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+             Hq: int, Hk: int, D: int) -> torch.Tensor:
+    """qkv: [B, S, (Hq+2*Hk)*D] fused projection output; rotates q,k in place
+    (GPU) and returns the buffer."""
+    if _use_hip(qkv):
+        return _RopeQKVFn.apply(qkv, cos, sin, Hq, Hk, D)
+    B, S, _ = qkv.shape
+    q = qkv[:, :, : Hq * D].view(B, S, Hq, D)
+    k = qkv[:, :, Hq * D : (Hq + Hk) * D].view(B, S, Hk, D)
+    v = qkv[:, :, (Hq + Hk) * D :]
+    qr = ref.rope_apply(q, cos, sin).reshape(B, S, Hq * D)
+    kr = ref.rope_apply(k, cos, sin).reshape(B, S, Hk * D)
+    return torch.cat([qr, kr, v], dim=-1)
+
+
+# ----------------------------------------------------------------- swiglu
+
+
+class _SwiGLUFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gu):
+        out = _hip().swiglu_fwd(gu)
+        ctx.save_for_backward(gu)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        (gu,) = ctx.saved_tensors
+        return _hip().swiglu_bwd(dout.contiguous(), gu)
+
+
+def swiglu(gate_up: torch.Tensor) -> torch.Tensor:
+    if _use_hip(gate_up):
+        return _SwiGLUFn.apply(gate_up.contiguous())
+    I = gate_up.shape[-1] // 2
+    g = gate_up[..., :I].float()
+    u = gate_up[..., I:].float()
+    return (torch.nn.functional.silu(g) * u).to(gate_up.dtype)
+
+
+# ------------------------------------------------------------------ adamw
+
+
+def adamw_step(p32, p_bf16, g, m, v, *, lr, b1=0.9, b2=0.95, eps=1e-8,
+               wd=0.0, step, grad_scale=1.0):
+    """Fused AdamW over flat buffers (fp32 master + bf16 working copy)."""
+    if p32.is_cuda:
+        _hip().adamw_(p32, p_bf16, g, m, v, lr, b1, b2, eps, wd, step, grad_scale)
+    else:
+        ref.adamw_step(p32, p_bf16, g, m, v, lr, b1, b2, eps, wd, step, grad_scale)

@@ -1,0 +1,143 @@
+"""Flagship benchmark: tokens/sec, Llama-3-8B DDP training (BASELINE.json).
+
+Single process per GPU. The driver launches N>1 as:
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N --steps K --warmup W
+
+Each rank: Llama-3-8B (random init, bf16) + FlatDDP (bucketed RCCL all-reduce
+overlapped with backward) + fused flat AdamW — the same training engine Ray
+Train's TorchTrainer workers use (ant_ray_amd/train). Synthetic token data.
+
+Rank 0 prints ONE JSON line with the whole-job aggregate tokens/sec.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--batch", type=int, default=0, help="per-GPU batch size")
+    ap.add_argument("--seq", type=int, default=4096)
+    ap.add_argument("--bucket-mb", type=int, default=64)
+    ap.add_argument("--lr", type=float, default=3e-4)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    if world > 1:
+        import torch.distributed as dist
+
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend="nccl", rank=rank, world_size=world)
+    else:
+        torch.cuda.set_device(local_rank)
+
+    device = f"cuda:{local_rank}"
+    if args.model.startswith("gpt2"):
+        args.seq = min(args.seq, 1024)
+        batch = args.batch or 16
+    else:
+        batch = args.batch or 4
+
+    from ant_ray_amd.models import build_model
+    from ant_ray_amd.parallel import FlatAdamW, FlatDDP, FlatParamManager
+
+    torch.manual_seed(1234 + rank)
+    t_build = time.time()
+    model = build_model(args.model, device=device, seq_len=args.seq)
+    mgr = FlatParamManager(model, device=device)
+    ddp = FlatDDP(model, manager=mgr, bucket_mb=args.bucket_mb)
+    opt = FlatAdamW(mgr, lr=args.lr, weight_decay=0.1)
+    vocab = model.cfg.vocab
+    if rank == 0:
+        print(
+            f"# model={args.model} params={model.num_params()/1e9:.2f}B "
+            f"world={world} batch/gpu={batch} seq={args.seq} "
+            f"build={time.time()-t_build:.1f}s",
+            flush=True,
+        )
+
+    def make_batch():
+        tokens = torch.randint(0, vocab, (batch, args.seq), device=device)
+        targets = torch.randint(0, vocab, (batch, args.seq), device=device)
+        return tokens, targets
+
+    def step():
+        tokens, targets = make_batch()
+        loss = ddp(tokens, targets)
+        loss.backward()
+        ddp.finish_grad_sync()
+        opt.step()
+        opt.zero_grad()
+        return loss
+
+    # warmup
+    for _ in range(args.warmup):
+        loss = step()
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(args.steps):
+        loss = step()
+    torch.cuda.synchronize()
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.barrier()
+    elapsed = time.time() - t0
+    if world > 1:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+
+    tokens_total = batch * args.seq * args.steps * world
+    value = tokens_total / elapsed
+    if rank == 0:
+        out = {
+            "metric": "tokens/sec Llama-3-8B Ray Train DDP"
+            if not args.model.startswith("gpt2")
+            else "tokens/sec GPT-2-small DDP",
+            "value": round(value, 1),
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": batch * world,
+                "seq_len": args.seq,
+                "parallelism": f"dp{world}",
+                "final_loss": round(float(loss.item()), 4),
+            },
+        }
+        print(json.dumps(out), flush=True)
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
