@@ -62,24 +62,24 @@ rmsnorm_fwd_kernel(const ushort_t* __restrict__ x, const ushort_t* __restrict__ 
 }
 
 // Backward. Grid-stride over rows so each thread owns a fixed column slice
-// across all its rows; dw accumulates in registers (H<=4096: 16 cols/thread)
-// and hits global memory once per block via atomics.
-// For H > MAX_COLS_PER_THREAD*256*8 this kernel is launched with
-// dw_atomic_per_iter (slower path handled by caller splitting H).
-#define DW_MAX_ITERS 4  // supports H <= 256*8*4 = 8192
-extern "C" __global__ void __launch_bounds__(256)
+// across all its rows; dw accumulates in registers (compile-time ITERS so the
+// accumulator array never spills to scratch — playbook rule: runtime-indexed
+// register arrays go to local memory) and hits global memory once per block
+// via atomics. ITERS = ceil(H / 2048); H <= 8192 supported.
+template <int ITERS>
+__global__ void __launch_bounds__(256)
 rmsnorm_bwd_kernel(const ushort_t* __restrict__ dy, const ushort_t* __restrict__ r,
                    const ushort_t* __restrict__ w, const float* __restrict__ rstd,
                    ushort_t* __restrict__ dx, float* __restrict__ dw,
                    int64_t N, int H) {
   __shared__ float scratch[16];
-  float dw_acc[DW_MAX_ITERS][8];
+  float dw_acc[ITERS][8];
 #pragma unroll
-  for (int i = 0; i < DW_MAX_ITERS; ++i)
+  for (int i = 0; i < ITERS; ++i)
 #pragma unroll
     for (int j = 0; j < 8; ++j) dw_acc[i][j] = 0.f;
 
-  int iters = H / (blockDim.x * 8);
+  constexpr int iters = ITERS;
 
   for (int64_t row = blockIdx.x; row < N; row += gridDim.x) {
     const ushort_t* dyr = dy + row * H;
@@ -88,8 +88,10 @@ rmsnorm_bwd_kernel(const ushort_t* __restrict__ dy, const ushort_t* __restrict__
     float rs = rstd[row];
     float c1 = 0.f;
     // pass 1: c1 = sum(dy*w*r)
+#pragma unroll
     for (int it = 0; it < iters; ++it) {
       int h = it * blockDim.x * 8 + threadIdx.x * 8;
+      if (h >= H) break;
       floatx8 dyf = bf8_to_f32x8(*reinterpret_cast<const ushortx8*>(dyr + h));
       floatx8 rf = bf8_to_f32x8(*reinterpret_cast<const ushortx8*>(rr + h));
       floatx8 wf = bf8_to_f32x8(*reinterpret_cast<const ushortx8*>(w + h));
@@ -100,8 +102,10 @@ rmsnorm_bwd_kernel(const ushort_t* __restrict__ dy, const ushort_t* __restrict__
     float k = total / (float)H * rs * rs;
     __syncthreads();  // scratch reuse safety across rows
     // pass 2: dx + dw accumulation
+#pragma unroll
     for (int it = 0; it < iters; ++it) {
       int h = it * blockDim.x * 8 + threadIdx.x * 8;
+      if (h >= H) break;
       floatx8 dyf = bf8_to_f32x8(*reinterpret_cast<const ushortx8*>(dyr + h));
       floatx8 rf = bf8_to_f32x8(*reinterpret_cast<const ushortx8*>(rr + h));
       floatx8 wf = bf8_to_f32x8(*reinterpret_cast<const ushortx8*>(w + h));
@@ -115,8 +119,10 @@ rmsnorm_bwd_kernel(const ushort_t* __restrict__ dy, const ushort_t* __restrict__
     }
   }
   // flush dw accumulators (one atomic per owned column)
-  for (int it = 0; it < iters && it < DW_MAX_ITERS; ++it) {
+#pragma unroll
+  for (int it = 0; it < iters; ++it) {
     int h = it * blockDim.x * 8 + threadIdx.x * 8;
+    if (h >= H) break;
 #pragma unroll
     for (int i = 0; i < 8; ++i) atomicAdd(&dw[h + i], dw_acc[it][i]);
   }
@@ -135,7 +141,16 @@ extern "C" void launch_rmsnorm_bwd(const void* dy, const void* r, const void* w,
                                    const float* rstd, void* dx, float* dw,
                                    int64_t N, int H, hipStream_t s) {
   int grid = (int)(N < 2048 ? N : 2048);
-  hipLaunchKernelGGL(rmsnorm_bwd_kernel, dim3(grid), dim3(256), 0, s,
-                     (const ushort_t*)dy, (const ushort_t*)r, (const ushort_t*)w,
-                     rstd, (ushort_t*)dx, dw, N, H);
+  int iters = (H + 2047) / 2048;
+#define LAUNCH_BWD(IT)                                                        \
+  hipLaunchKernelGGL(rmsnorm_bwd_kernel<IT>, dim3(grid), dim3(256), 0, s,     \
+                     (const ushort_t*)dy, (const ushort_t*)r,                 \
+                     (const ushort_t*)w, rstd, (ushort_t*)dx, dw, N, H)
+  switch (iters) {
+    case 1: LAUNCH_BWD(1); break;
+    case 2: LAUNCH_BWD(2); break;
+    case 3: LAUNCH_BWD(3); break;
+    default: LAUNCH_BWD(4); break;
+  }
+#undef LAUNCH_BWD
 }
