@@ -6,8 +6,8 @@
 #include <hip/hip_runtime.h>
 
 extern "C" {
-void launch_rmsnorm_fwd(const void*, const void*, void*, float*, void*, int64_t,
-                        int, float, hipStream_t);
+void launch_rmsnorm_fwd(const void*, const void*, void*, float*, const void*,
+                        void*, int64_t, int, float, hipStream_t);
 void launch_rmsnorm_bwd(const void*, const void*, const void*, const float*,
                         void*, float*, int64_t, int, hipStream_t);
 void launch_rope(void*, void*, const float*, const float*, int64_t, int64_t,
@@ -34,7 +34,9 @@ void check_bf16(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
 }
 
-std::tuple<torch::Tensor, torch::Tensor> rmsnorm_fwd(
+// Returns (y, rstd, h) — h is the fresh residual-sum buffer when `residual`
+// is given, else an alias of x.
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> rmsnorm_fwd(
     torch::Tensor x, torch::Tensor w,
     c10::optional<torch::Tensor> residual, double eps) {
   check_bf16(x, "x");
@@ -44,15 +46,19 @@ std::tuple<torch::Tensor, torch::Tensor> rmsnorm_fwd(
   TORCH_CHECK(H % 8 == 0, "H must be a multiple of 8");
   auto y = torch::empty_like(x);
   auto rstd = torch::empty({N}, x.options().dtype(torch::kFloat32));
-  void* res_ptr = nullptr;
+  const void* res_ptr = nullptr;
+  void* h_ptr = nullptr;
+  torch::Tensor h = x;
   if (residual.has_value()) {
     check_bf16(*residual, "residual");
     res_ptr = residual->data_ptr();
+    h = torch::empty_like(x);
+    h_ptr = h.data_ptr();
   }
   launch_rmsnorm_fwd(x.data_ptr(), w.data_ptr(), y.data_ptr(),
-                     rstd.data_ptr<float>(), res_ptr, N, (int)H, (float)eps,
-                     cur_stream());
-  return {y, rstd};
+                     rstd.data_ptr<float>(), res_ptr, h_ptr, N, (int)H,
+                     (float)eps, cur_stream());
+  return {y, rstd, h};
 }
 
 std::tuple<torch::Tensor, torch::Tensor> rmsnorm_bwd(torch::Tensor dy,

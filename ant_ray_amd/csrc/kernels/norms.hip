@@ -13,16 +13,21 @@
 #include "common.hip.h"
 
 // One block per row; row length H must be a multiple of 8.
+// With residual: h_out = bf16(x + residual) is written to a SEPARATE buffer
+// (training keeps every layer's h for backward; h_out == residual is legal
+// for inference-style in-place reuse).
 extern "C" __global__ void __launch_bounds__(256)
 rmsnorm_fwd_kernel(const ushort_t* __restrict__ x, const ushort_t* __restrict__ w,
                    ushort_t* __restrict__ y, float* __restrict__ rstd_out,
-                   ushort_t* __restrict__ residual,  // in/out; may be null
+                   const ushort_t* __restrict__ residual,  // may be null
+                   ushort_t* __restrict__ h_out,           // required if residual
                    int H, float eps) {
   __shared__ float scratch[16];
   int64_t row = blockIdx.x;
   const ushort_t* xr = x + row * H;
   ushort_t* yr = y + row * H;
-  ushort_t* rr = residual ? residual + row * H : nullptr;
+  const ushort_t* rr = residual ? residual + row * H : nullptr;
+  ushort_t* hr = residual ? h_out + row * H : nullptr;
 
   float sumsq = 0.f;
   // pass 1: (optional residual add) + sum of squares
@@ -37,7 +42,7 @@ rmsnorm_fwd_kernel(const ushort_t* __restrict__ x, const ushort_t* __restrict__ 
       // store the bf16-rounded sum; recompute f from it so the normalized
       // output matches what later layers will re-read
       ushortx8 sv = f32x8_to_bf8(f);
-      *reinterpret_cast<ushortx8*>(rr + h) = sv;
+      *reinterpret_cast<ushortx8*>(hr + h) = sv;
       f = bf8_to_f32x8(sv);
     }
 #pragma unroll
@@ -48,7 +53,7 @@ rmsnorm_fwd_kernel(const ushort_t* __restrict__ x, const ushort_t* __restrict__ 
   if (threadIdx.x == 0 && rstd_out) rstd_out[row] = rstd;
 
   // pass 2: normalize (rows are L1/L2 resident after pass 1)
-  const ushort_t* src = rr ? rr : xr;
+  const ushort_t* src = rr ? hr : xr;
   for (int h = threadIdx.x * 8; h < H; h += blockDim.x * 8) {
     ushortx8 xv = *reinterpret_cast<const ushortx8*>(src + h);
     ushortx8 wv = *reinterpret_cast<const ushortx8*>(w + h);
@@ -129,12 +134,13 @@ rmsnorm_bwd_kernel(const ushort_t* __restrict__ dy, const ushort_t* __restrict__
 }
 
 extern "C" void launch_rmsnorm_fwd(const void* x, const void* w, void* y,
-                                   float* rstd, void* residual, int64_t N,
-                                   int H, float eps, hipStream_t s) {
+                                   float* rstd, const void* residual,
+                                   void* h_out, int64_t N, int H, float eps,
+                                   hipStream_t s) {
   dim3 grid((uint32_t)N), block(256);
   hipLaunchKernelGGL(rmsnorm_fwd_kernel, grid, block, 0, s,
                      (const ushort_t*)x, (const ushort_t*)w, (ushort_t*)y,
-                     rstd, (ushort_t*)residual, H, eps);
+                     rstd, (const ushort_t*)residual, (ushort_t*)h_out, H, eps);
 }
 
 extern "C" void launch_rmsnorm_bwd(const void* dy, const void* r, const void* w,

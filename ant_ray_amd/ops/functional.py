@@ -27,7 +27,7 @@ def _use_hip(t: torch.Tensor) -> bool:
 class _RMSNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, eps):
-        y, rstd = _hip().rmsnorm_fwd(x, w, None, eps)
+        y, rstd, _ = _hip().rmsnorm_fwd(x, w, None, eps)
         ctx.save_for_backward(x, w, rstd)
         return y
 
@@ -48,15 +48,15 @@ def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor
 
 
 class _FusedAddRMSNormFn(torch.autograd.Function):
-    """(x, residual) -> (y, h) where h = x+residual (bf16-rounded),
-    y = rmsnorm(h)*w. The kernel updates `residual` in place to h."""
+    """(x, residual) -> (y, h) where h = bf16(x+residual) is a FRESH buffer
+    (training keeps each layer's h for backward), y = rmsnorm(h)*w — one
+    fused kernel pass."""
 
     @staticmethod
     def forward(ctx, x, residual, w, eps):
-        y, rstd = _hip().rmsnorm_fwd(x, w, residual, eps)
-        ctx.save_for_backward(residual, w, rstd)
-        ctx.mark_dirty(residual)
-        return y, residual
+        y, rstd, h = _hip().rmsnorm_fwd(x, w, residual, eps)
+        ctx.save_for_backward(h, w, rstd)
+        return y, h
 
     @staticmethod
     def backward(ctx, dy, dh_out):
@@ -68,11 +68,9 @@ class _FusedAddRMSNormFn(torch.autograd.Function):
 
 
 def fused_add_rmsnorm(x, residual, w, eps: float = 1e-5):
-    """Returns (normalized, new_residual). new_residual aliases `residual`'s
-    storage on GPU (in-place update) — callers must treat the old value as
-    consumed."""
+    """Returns (normalized, h=x+residual). h is a fresh tensor."""
     if _use_hip(x):
-        return _FusedAddRMSNormFn.apply(x.contiguous(), residual, w, eps)
+        return _FusedAddRMSNormFn.apply(x.contiguous(), residual.contiguous(), w, eps)
     h = (x.float() + residual.float()).to(x.dtype)
     return rmsnorm(h, w, eps), h
 
@@ -99,7 +97,9 @@ class _RopeQKVFn(torch.autograd.Function):
         Hq, Hk, D = ctx.meta
         cos, sin = ctx.tables
         B, S, _ = dqkv.shape
-        dqkv = dqkv.contiguous()
+        # clone: the incoming gradient may be used elsewhere by autograd and
+        # the inverse rotation below is in-place
+        dqkv = dqkv.contiguous().clone()
         dq = dqkv[:, :, : Hq * D].view(B, S, Hq, D)
         dk = dqkv[:, :, Hq * D : (Hq + Hk) * D].view(B, S, Hk, D)
         _hip().rope_(dq, dk, cos, sin, True)
