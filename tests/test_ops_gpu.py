@@ -122,7 +122,7 @@ class TestCrossEntropy:
     def test_linear_ce_end_to_end(self):
         N, H, V = 128, 256, 512
         x = torch.randn(N, H, dtype=torch.bfloat16, device=DEV, requires_grad=True)
-        w = torch.randn(V, H, dtype=torch.bfloat16, device=DEV, requires_grad=True) * 0.05
+        w = (torch.randn(V, H, dtype=torch.bfloat16, device=DEV) * 0.05).requires_grad_(True)
         targets = torch.randint(0, V, (N,), device=DEV)
         loss = ops.linear_cross_entropy(x, w, targets, chunk_rows=32)
         loss.backward()
