@@ -9,7 +9,8 @@ extern "C" {
 void launch_rmsnorm_fwd(const void*, const void*, void*, float*, const void*,
                         void*, int64_t, int, float, hipStream_t);
 void launch_rmsnorm_bwd(const void*, const void*, const void*, const float*,
-                        void*, float*, int64_t, int, hipStream_t);
+                        void*, float*, float*, int64_t, int, hipStream_t);
+int rmsnorm_bwd_grid(int64_t);
 void launch_rope(void*, void*, const float*, const float*, int64_t, int64_t,
                  int, int, int, int64_t, int64_t, int, hipStream_t);
 void launch_swiglu_fwd(const void*, void*, int64_t, int64_t, hipStream_t);
@@ -72,10 +73,13 @@ std::tuple<torch::Tensor, torch::Tensor> rmsnorm_bwd(torch::Tensor dy,
   int64_t N = dy.numel() / H;
   TORCH_CHECK(H <= 8192, "rmsnorm_bwd supports H<=8192");
   auto dx = torch::empty_like(dy);
-  auto dw = torch::zeros({H}, dy.options().dtype(torch::kFloat32));
+  auto dw = torch::empty({H}, dy.options().dtype(torch::kFloat32));
+  int grid = rmsnorm_bwd_grid(N);
+  auto partials = torch::empty({grid, H}, dy.options().dtype(torch::kFloat32));
   launch_rmsnorm_bwd(dy.data_ptr(), r.data_ptr(), w.data_ptr(),
                      rstd.data_ptr<float>(), dx.data_ptr(),
-                     dw.data_ptr<float>(), N, (int)H, cur_stream());
+                     partials.data_ptr<float>(), dw.data_ptr<float>(), N,
+                     (int)H, cur_stream());
   return {dx, dw};
 }
 

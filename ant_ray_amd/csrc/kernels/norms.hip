@@ -123,14 +123,32 @@ rmsnorm_bwd_kernel(const ushort_t* __restrict__ dy, const ushort_t* __restrict__
       *reinterpret_cast<ushortx8*>(dxr + h) = f32x8_to_bf8(o);
     }
   }
-  // flush dw accumulators (one atomic per owned column)
+  // flush dw accumulators to this block's private partial row (atomics into
+  // one shared dw[H] serialize 2048 blocks on 4096 addresses — measured 8x
+  // over roofline; partials + a tiny second-stage reduce fix that)
 #pragma unroll
   for (int it = 0; it < iters; ++it) {
     int h = it * blockDim.x * 8 + threadIdx.x * 8;
     if (h >= H) break;
+    float* dst = dw + (int64_t)blockIdx.x * H + h;
 #pragma unroll
-    for (int i = 0; i < 8; ++i) atomicAdd(&dw[h + i], dw_acc[it][i]);
+    for (int i = 0; i < 8; ++i) dst[i] = dw_acc[it][i];
   }
+}
+
+// Second stage: dw[H] = sum over G partial rows (column-parallel).
+extern "C" __global__ void __launch_bounds__(256)
+dw_reduce_kernel(const float* __restrict__ partials, float* __restrict__ dw,
+                 int G, int H) {
+  int h = (blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (h >= H) return;
+  floatx4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int g = 0; g < G; ++g) {
+    floatx4 v = *reinterpret_cast<const floatx4*>(partials + (int64_t)g * H + h);
+#pragma unroll
+    for (int i = 0; i < 4; ++i) acc[i] += v[i];
+  }
+  *reinterpret_cast<floatx4*>(dw + h) = acc;
 }
 
 extern "C" void launch_rmsnorm_fwd(const void* x, const void* w, void* y,
@@ -143,15 +161,20 @@ extern "C" void launch_rmsnorm_fwd(const void* x, const void* w, void* y,
                      rstd, (const ushort_t*)residual, (ushort_t*)h_out, H, eps);
 }
 
+// dw_partials must hold grid*H floats; returns grid used via launch config
+// chosen here (<=1024 blocks).
+extern "C" int rmsnorm_bwd_grid(int64_t N) { return (int)(N < 1024 ? N : 1024); }
+
 extern "C" void launch_rmsnorm_bwd(const void* dy, const void* r, const void* w,
-                                   const float* rstd, void* dx, float* dw,
-                                   int64_t N, int H, hipStream_t s) {
-  int grid = (int)(N < 2048 ? N : 2048);
+                                   const float* rstd, void* dx,
+                                   float* dw_partials, float* dw, int64_t N,
+                                   int H, hipStream_t s) {
+  int grid = rmsnorm_bwd_grid(N);
   int iters = (H + 2047) / 2048;
 #define LAUNCH_BWD(IT)                                                        \
   hipLaunchKernelGGL(rmsnorm_bwd_kernel<IT>, dim3(grid), dim3(256), 0, s,     \
                      (const ushort_t*)dy, (const ushort_t*)r,                 \
-                     (const ushort_t*)w, rstd, (ushort_t*)dx, dw, N, H)
+                     (const ushort_t*)w, rstd, (ushort_t*)dx, dw_partials, N, H)
   switch (iters) {
     case 1: LAUNCH_BWD(1); break;
     case 2: LAUNCH_BWD(2); break;
@@ -159,4 +182,7 @@ extern "C" void launch_rmsnorm_bwd(const void* dy, const void* r, const void* w,
     default: LAUNCH_BWD(4); break;
   }
 #undef LAUNCH_BWD
+  int rgrid = (H / 4 + 255) / 256;
+  hipLaunchKernelGGL(dw_reduce_kernel, dim3(rgrid), dim3(256), 0, s,
+                     dw_partials, dw, grid, H);
 }
