@@ -48,7 +48,9 @@ def main():
         args.seq = min(args.seq, 1024)
         batch = args.batch or 16
     else:
-        batch = args.batch or 4
+        # batch 6 x seq 4096 ~ 250 GB of 288 GB HBM (measured): biggest safe
+        # per-GPU batch with headroom for RCCL workspaces at 8 GPUs
+        batch = args.batch or 6
 
     from ant_ray_amd.models import build_model
     from ant_ray_amd.parallel import FlatAdamW, FlatDDP, FlatParamManager
