@@ -171,11 +171,11 @@ def remote(*args, **kwargs):
     return make
 
 
-def put(value: Any, *, _owner=None) -> ObjectRef:
+def put(value: Any, *, _owner=None, _tensor_transport: Optional[str] = None) -> ObjectRef:
     _check_connected()
     if isinstance(value, ObjectRef):
         raise TypeError("Calling 'put' on an ObjectRef is not allowed")
-    return global_worker.core_worker.put(value)
+    return global_worker.core_worker.put(value, tensor_transport=_tensor_transport)
 
 
 def get(

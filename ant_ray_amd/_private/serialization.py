@@ -130,6 +130,18 @@ class _Pickler(cloudpickle.CloudPickler):
             import torch
 
             if isinstance(obj, torch.Tensor):
+                if obj.is_cuda:
+                    ctx = getattr(_transport_ctx, "ctx", None)
+                    if ctx is not None and ctx.mode == "hip_ipc":
+                        from ant_ray_amd.experimental.gpu_object_manager.gpu_object_store import (
+                            _rebuild_gpu_tensor,
+                            export_tensor,
+                        )
+
+                        meta, pinned = export_tensor(obj)
+                        ctx.pinned.append(pinned)
+                        return (_rebuild_gpu_tensor, (meta,))
+                    obj = obj.detach().cpu()
                 r = _torch_cpu_tensor_reducer(obj)
                 if r is not NotImplemented:
                     return r
@@ -139,6 +151,29 @@ class _Pickler(cloudpickle.CloudPickler):
 
 
 _BUFFER_THRESHOLD = 512  # buffers below this get pickled in-band
+
+
+import threading as _threading
+
+_transport_ctx = _threading.local()
+
+
+class gpu_transport_context:
+    """While active, CUDA tensors serialized in this thread are exported via
+    the given transport ("hip_ipc") instead of copied to CPU. Exported
+    tensors are collected into `pinned` — the caller must register them in
+    the GPUObjectStore under the object id."""
+
+    def __init__(self, mode: str):
+        self.mode = mode
+        self.pinned = []
+
+    def __enter__(self):
+        _transport_ctx.ctx = self
+        return self
+
+    def __exit__(self, *a):
+        _transport_ctx.ctx = None
 
 
 def _make_dispatch(contained_refs):
@@ -156,6 +191,20 @@ def _make_dispatch(contained_refs):
         import torch
 
         def reduce_tensor(t):
+            if t.is_cuda:
+                ctx = getattr(_transport_ctx, "ctx", None)
+                if ctx is not None and ctx.mode == "hip_ipc":
+                    from ant_ray_amd.experimental.gpu_object_manager.gpu_object_store import (
+                        _rebuild_gpu_tensor,
+                        export_tensor,
+                    )
+
+                    meta, pinned = export_tensor(t)
+                    ctx.pinned.append(pinned)
+                    return (_rebuild_gpu_tensor, (meta,))
+                # default object-store transport: device -> host copy; the
+                # reader gets a CPU tensor (reference default behavior)
+                t = t.detach().cpu()
             r = _torch_cpu_tensor_reducer(t)
             if r is NotImplemented:
                 return t.__reduce_ex__(5)

@@ -254,13 +254,29 @@ class TaskExecutor:
         values = (result,) if n == 1 else tuple(result)
         if n > 1 and len(values) != n:
             raise ValueError(f"task returned {len(values)} values, expected {n}")
+        transport = payload.get("tensor_transport")
         out = []
         for i, v in enumerate(values):
             oid = ObjectID.for_return(TaskID(payload["task_id"]), i).binary()
-            sobj = serialization.serialize(v)
+            if transport == "hip_ipc":
+                with serialization.gpu_transport_context("hip_ipc") as gctx:
+                    sobj = serialization.serialize(v)
+                if gctx.pinned:
+                    from ant_ray_amd.experimental.gpu_object_manager import (
+                        gpu_object_store,
+                    )
+
+                    gpu_object_store.add(oid, gctx.pinned)
+            else:
+                sobj = serialization.serialize(v)
             self.cw._register_escapes(sobj)
             if sobj.total_size <= INLINE_OBJECT_MAX:
-                out.append({"oid": oid, "inline": sobj.to_bytes(), "meta": sobj.metadata})
+                entry = {"oid": oid, "inline": sobj.to_bytes(), "meta": sobj.metadata}
+                if transport == "hip_ipc":
+                    # the device tensors stay pinned here; record us as holder
+                    # so frees reach this process
+                    entry["holder"] = list(self.cw.addr)
+                out.append(entry)
             else:
                 self.cw.store.put_serialized_to_shm(oid, sobj)
                 out.append({"oid": oid, "inline": None, "holder": list(self.cw.addr)})

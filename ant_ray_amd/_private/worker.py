@@ -265,17 +265,31 @@ class CoreWorker:
 
     async def rpc_free_objects(self, conn, p):
         self.store.free(p["oids"])
+        try:
+            from ant_ray_amd.experimental.gpu_object_manager import gpu_object_store
+
+            gpu_object_store.free(p["oids"])
+        except Exception:
+            pass
         return {"ok": True}
 
     # ================================================================== puts
-    def put(self, value, *, _owner_inline=False) -> ObjectRef:
+    def put(self, value, *, _owner_inline=False, tensor_transport=None) -> ObjectRef:
         with self._lock:
             self._put_index += 1
             oid = ObjectID.for_put(WorkerID(self.worker_id), self._put_index).binary()
         if self.mode == LOCAL_MODE:
             self.store.memory.put(oid, value)
             return ObjectRef(oid, None, worker=self)
-        sobj = serialization.serialize(value)
+        if tensor_transport == "hip_ipc":
+            with serialization.gpu_transport_context("hip_ipc") as gctx:
+                sobj = serialization.serialize(value)
+            if gctx.pinned:
+                from ant_ray_amd.experimental.gpu_object_manager import gpu_object_store
+
+                gpu_object_store.add(oid, gctx.pinned)
+        else:
+            sobj = serialization.serialize(value)
         self._register_escapes(sobj)
         if _owner_inline and sobj.total_size <= INLINE_OBJECT_MAX:
             self.store.memory.put(oid, _RawResult(sobj.to_bytes(), sobj.metadata))
@@ -585,12 +599,12 @@ class CoreWorker:
         if reply.get("status") == "ok":
             for r in reply.get("results", []):
                 oid = r["oid"]
+                if r.get("holder"):
+                    self._object_locations[oid] = tuple(r["holder"])
                 if r.get("inline") is not None:
                     self.store.memory.put(oid, _RawResult(r["inline"], r.get("meta", b"py")))
                 else:
                     self.store.memory.put(oid, IN_PLASMA)
-                    if r.get("holder"):
-                        self._object_locations[oid] = tuple(r["holder"])
         else:
             data = reply.get("error_payload")
             meta = reply.get("error_meta", serialization.META_ERROR)
@@ -686,6 +700,7 @@ class CoreWorker:
             "caller": self.worker_id,
             "caller_addr": list(self.addr),
             "concurrency_group": opts.get("concurrency_group"),
+            "tensor_transport": opts.get("tensor_transport"),
         }
         self.io.submit(self._submit_actor_async(st, payload)).result()
         return refs

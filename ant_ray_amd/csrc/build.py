@@ -77,6 +77,28 @@ def build(verbose=True):
               f"-L{tdir}/lib", "-ltorch", "-ltorch_cpu", "-ltorch_python",
               "-lc10", "-ltorch_hip", "-lc10_hip", "-lamdhip64",
               f"-Wl,-rpath,{tdir}/lib"])
+
+    # gpu_ipc extension (host-only hip runtime calls + torch)
+    ipc_src = os.path.join(CSRC, "gpu_ipc.cpp")
+    ipc_obj = os.path.join(BUILD, "gpu_ipc.o")
+    ipc_so = os.path.join(PKG, "_gpu_ipc" + ext_suffix)
+    if _newer(ipc_src, ipc_obj):
+        if verbose:
+            print("[hipcc] gpu_ipc.cpp")
+        py_inc = sysconfig.get_paths()["include"]
+        _run([HIPCC, "-O2", "-std=c++17", "-fPIC", "-c", ipc_src, "-o", ipc_obj,
+              f"-I{tdir}/include", f"-I{tdir}/include/torch/csrc/api/include",
+              f"-I{py_inc}",
+              "-D__HIP_PLATFORM_AMD__", "-DUSE_ROCM",
+              "-DTORCH_EXTENSION_NAME=_gpu_ipc",
+              "-D_GLIBCXX_USE_CXX11_ABI=1"])
+    if _newer(ipc_obj, ipc_so):
+        if verbose:
+            print("[link] _gpu_ipc")
+        _run([HIPCC, "-shared", ipc_obj, "-o", ipc_so,
+              f"-L{tdir}/lib", "-ltorch", "-ltorch_cpu", "-ltorch_python",
+              "-lc10", "-ltorch_hip", "-lc10_hip", "-lamdhip64",
+              f"-Wl,-rpath,{tdir}/lib"])
     return out_so
 
 
