@@ -40,9 +40,10 @@ def _aligned(n: int) -> int:
 
 
 class FlatParamManager:
-    """Re-homes a module's parameters and gradients into flat bf16 buffers."""
+    """Re-homes a module's parameters and gradients into flat buffers
+    (bf16 on GPU; pass dtype=torch.float32 for CPU/gloo paths)."""
 
-    def __init__(self, module: nn.Module, device=None):
+    def __init__(self, module: nn.Module, device=None, dtype=torch.bfloat16):
         self.module = module
         params = [p for p in module.parameters() if p.requires_grad]
         # reverse registration order approximates backward completion order,
@@ -56,8 +57,9 @@ class FlatParamManager:
             total += _aligned(p.numel())
         self.numel = _aligned(total)
         dev = device or (params[0].device if params else "cpu")
-        self.flat_param = torch.zeros(self.numel, dtype=torch.bfloat16, device=dev)
-        self.flat_grad = torch.zeros(self.numel, dtype=torch.bfloat16, device=dev)
+        self.dtype = dtype
+        self.flat_param = torch.zeros(self.numel, dtype=dtype, device=dev)
+        self.flat_grad = torch.zeros(self.numel, dtype=dtype, device=dev)
         for p in params:
             off = self.offsets[id(p)]
             n = p.numel()
@@ -77,11 +79,14 @@ class FlatDDP(nn.Module):
     """Data-parallel wrapper with bucketed, overlapped RCCL all-reduce."""
 
     def __init__(self, module: nn.Module, manager: Optional[FlatParamManager] = None,
-                 bucket_mb: int = 64, process_group=None, device=None):
+                 bucket_mb: int = 64, process_group=None, device=None,
+                 auto_sync: bool = False):
         super().__init__()
         self.module = module
         self.manager = manager or FlatParamManager(module, device=device)
         self.pg = process_group
+        self.auto_sync = auto_sync
+        self._final_cb_queued = False
         self.world = dist.get_world_size(process_group) if dist.is_initialized() else 1
         self._works: List = []
         self._bucket_of: Dict[int, int] = {}
@@ -132,7 +137,33 @@ class FlatDDP(nn.Module):
             self._works.append(work)
 
     def forward(self, *args, **kwargs):
-        return self.module(*args, **kwargs)
+        out = self.module(*args, **kwargs)
+        if self.auto_sync and self.world > 1 and torch.is_grad_enabled():
+            self._attach_final_sync(out)
+        return out
+
+    def _attach_final_sync(self, out):
+        """Queue finish_grad_sync to run at the END of the next backward, so
+        user loops written for torch DDP (no explicit sync call) are correct.
+        The hook fires at backward START; queue_callback defers to its end."""
+        tensors = [t for t in torch.utils._pytree.tree_leaves(out)
+                   if isinstance(t, torch.Tensor) and t.requires_grad]
+        if not tensors:
+            return
+
+        def _on_backward_start(_grad):
+            if not self._final_cb_queued:
+                self._final_cb_queued = True
+                torch.autograd.Variable._execution_engine.queue_callback(
+                    self._final_sync_cb
+                )
+            return _grad
+
+        tensors[0].register_hook(_on_backward_start)
+
+    def _final_sync_cb(self):
+        self._final_cb_queued = False
+        self.finish_grad_sync()
 
     def finish_grad_sync(self):
         """Wait for outstanding bucket all-reduces (call before optimizer)."""
