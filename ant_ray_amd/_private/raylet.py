@@ -280,9 +280,13 @@ class Raylet:
             if pg
             else self.resources_available
         )
-        if pool is not None:
-            for k, v in res.items():
-                pool[k] = pool.get(k, 0.0) + v
+        if pool is None:
+            # bundle already returned (PG removed while the lease was live):
+            # return_bundle only gave back what the bundle had free at that
+            # moment, so the lease's share goes straight to the node pool
+            pool = self.resources_available
+        for k, v in res.items():
+            pool[k] = pool.get(k, 0.0) + v
         need = res.get("GPU", 0.0)
         if need > 0:
             remaining = need

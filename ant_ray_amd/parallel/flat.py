@@ -80,12 +80,16 @@ class FlatDDP(nn.Module):
 
     def __init__(self, module: nn.Module, manager: Optional[FlatParamManager] = None,
                  bucket_mb: int = 64, process_group=None, device=None,
-                 auto_sync: bool = False):
+                 auto_sync: bool = False, average_grads: bool = False):
         super().__init__()
         self.module = module
         self.manager = manager or FlatParamManager(module, device=device)
         self.pg = process_group
         self.auto_sync = auto_sync
+        # True → divide summed grads by world in finish_grad_sync (torch-DDP
+        # semantics for user-supplied optimizers); False → the caller folds
+        # 1/world into the optimizer (FlatAdamW grad_scale)
+        self.average_grads = average_grads
         self._final_cb_queued = False
         self.world = dist.get_world_size(process_group) if dist.is_initialized() else 1
         self._works: List = []
@@ -169,6 +173,8 @@ class FlatDDP(nn.Module):
         """Wait for outstanding bucket all-reduces (call before optimizer)."""
         for w in self._works:
             w.wait()
+        if self._works and self.average_grads and self.world > 1:
+            self.manager.flat_grad.div_(self.world)
         self._works.clear()
         self._bucket_pending = list(self._bucket_total)
 

@@ -1,0 +1,111 @@
+"""Train/AIR config dataclasses.
+
+Role parity: reference python/ray/air/config.py (ScalingConfig :99,
+RunConfig, FailureConfig, CheckpointConfig) and
+python/ray/train/v2/api/config.py. Semantics kept: ScalingConfig drives the
+worker group size/resources, FailureConfig.max_failures drives group
+restarts, CheckpointConfig(num_to_keep, checkpoint_score_attribute) drives
+retention in the checkpoint manager.
+"""
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass, field
+from typing import Any, Dict, Optional
+
+
+@dataclass
+class ScalingConfig:
+    num_workers: int = 1
+    use_gpu: bool = False
+    resources_per_worker: Optional[Dict[str, float]] = None
+    trainer_resources: Optional[Dict[str, float]] = None
+    placement_strategy: str = "PACK"
+    accelerator_type: Optional[str] = None
+
+    @property
+    def _resources_per_worker_not_none(self) -> Dict[str, float]:
+        if self.resources_per_worker is not None:
+            res = dict(self.resources_per_worker)
+            if self.use_gpu and "GPU" not in res:
+                res["GPU"] = 1
+            return res
+        return {"CPU": 1, "GPU": 1} if self.use_gpu else {"CPU": 1}
+
+    def num_gpus_per_worker(self) -> float:
+        return float(self._resources_per_worker_not_none.get("GPU", 0))
+
+
+@dataclass
+class FailureConfig:
+    max_failures: int = 0
+
+
+@dataclass
+class CheckpointConfig:
+    num_to_keep: Optional[int] = None
+    checkpoint_score_attribute: Optional[str] = None
+    checkpoint_score_order: str = "max"
+    checkpoint_frequency: int = 0
+    checkpoint_at_end: Optional[bool] = None
+
+
+@dataclass
+class RunConfig:
+    name: Optional[str] = None
+    storage_path: Optional[str] = None
+    failure_config: FailureConfig = field(default_factory=FailureConfig)
+    checkpoint_config: CheckpointConfig = field(default_factory=CheckpointConfig)
+    verbose: int = 1
+    log_to_file: bool = False
+    callbacks: Optional[list] = None
+
+    def __post_init__(self):
+        if self.storage_path is None:
+            self.storage_path = os.path.expanduser(
+                os.environ.get("ANTRAY_STORAGE_PATH", "~/ray_results")
+            )
+
+
+@dataclass
+class TorchConfig:
+    """Backend config for torch process-group setup.
+
+    Parity: python/ray/train/torch/config.py:33 (TorchConfig: backend,
+    init_method, timeout_s). backend=None → nccl (RCCL) when the worker has
+    a GPU else gloo.
+    """
+
+    backend: Optional[str] = None
+    init_method: str = "tcp"
+    timeout_s: int = 1800
+
+    def resolved_backend(self, use_gpu: bool) -> str:
+        if self.backend:
+            return self.backend
+        return "nccl" if use_gpu else "gloo"
+
+
+@dataclass
+class Result:
+    """Outcome of Trainer.fit(). Parity: python/ray/air/result.py:38."""
+
+    metrics: Optional[Dict[str, Any]] = None
+    checkpoint: Optional[Any] = None
+    path: Optional[str] = None
+    error: Optional[BaseException] = None
+    metrics_dataframe: Optional[Any] = None
+    best_checkpoints: Optional[list] = None
+
+    def get_best_checkpoint(self, metric: str, mode: str = "max"):
+        if not self.best_checkpoints:
+            return None
+        keyed = [
+            (m.get(metric), c)
+            for c, m in self.best_checkpoints
+            if m and m.get(metric) is not None
+        ]
+        if not keyed:
+            return None
+        keyed.sort(key=lambda t: t[0], reverse=(mode == "max"))
+        return keyed[0][1]

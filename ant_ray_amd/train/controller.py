@@ -1,0 +1,175 @@
+"""Train controller: drives worker groups, failure policy, checkpoints.
+
+Role parity: reference python/ray/train/v2/_internal/execution/controller/
+controller.py:101 (TrainController control loop :505-527 — poll worker
+group, apply failure/scaling decisions) and checkpoint/checkpoint_manager.py
+(retention by CheckpointConfig). Runs in the driver process here (the
+reference pins it to the driver node as an actor; same failure domain —
+driver death ends the run either way on one node).
+"""
+from __future__ import annotations
+
+import logging
+import os
+import shutil
+import time
+from typing import Any, Callable, Dict, List, Optional
+
+from ant_ray_amd.train._checkpoint import Checkpoint
+from ant_ray_amd.train.config import (
+    Result,
+    RunConfig,
+    ScalingConfig,
+    TorchConfig,
+)
+from ant_ray_amd.train.worker_group import WorkerGroup
+
+logger = logging.getLogger("antray.train")
+
+
+class CheckpointManager:
+    """Tracks reported checkpoints; enforces num_to_keep retention."""
+
+    def __init__(self, checkpoint_config):
+        self.cfg = checkpoint_config
+        self.checkpoints: List[tuple] = []  # (path, metrics)
+        self.latest: Optional[str] = None
+
+    def register(self, path: str, metrics: Dict[str, Any]):
+        if self.latest == path:
+            # same checkpoint dir reported by another rank: merge metrics
+            for i, (p, m) in enumerate(self.checkpoints):
+                if p == path:
+                    m.update(metrics)
+            return
+        self.latest = path
+        self.checkpoints.append((path, dict(metrics)))
+        self._enforce_retention()
+
+    def _enforce_retention(self):
+        keep = self.cfg.num_to_keep
+        if keep is None or len(self.checkpoints) <= keep:
+            return
+        attr = self.cfg.checkpoint_score_attribute
+        if attr:
+            order = self.cfg.checkpoint_score_order == "max"
+            ranked = sorted(
+                self.checkpoints,
+                key=lambda t: (t[1].get(attr) is not None,
+                               t[1].get(attr) if t[1].get(attr) is not None else 0),
+                reverse=order,
+            )
+            doomed = [c for c in ranked[keep:] if c[0] != self.latest]
+        else:
+            doomed = [c for c in self.checkpoints[:-keep]]
+        for path, _ in doomed:
+            self.checkpoints.remove((path, _))
+            try:
+                shutil.rmtree(path, ignore_errors=True)
+            except Exception:
+                pass
+
+    def best_checkpoints(self):
+        return [(Checkpoint(p), m) for p, m in self.checkpoints]
+
+
+class TrainController:
+    def __init__(
+        self,
+        train_fn: Callable,
+        train_loop_config: Optional[dict],
+        scaling_config: ScalingConfig,
+        run_config: RunConfig,
+        torch_config: Optional[TorchConfig] = None,
+        datasets: Optional[Dict[str, Any]] = None,
+        resume_from_checkpoint: Optional[Checkpoint] = None,
+    ):
+        self.train_fn = train_fn
+        self.train_loop_config = train_loop_config
+        self.scaling = scaling_config
+        self.run_config = run_config
+        self.torch_config = torch_config or TorchConfig()
+        self.datasets = datasets or {}
+        self.name = run_config.name or f"train_{int(time.time())}"
+        self.experiment_path = os.path.join(run_config.storage_path, self.name)
+        os.makedirs(self.experiment_path, exist_ok=True)
+        self.ckpt_manager = CheckpointManager(run_config.checkpoint_config)
+        self.resume_from_checkpoint = resume_from_checkpoint
+        self.latest_metrics: Dict[int, Dict[str, Any]] = {}
+
+    def _split_datasets(self, n: int):
+        """Per-worker dataset shards: ant_ray_amd.data datasets stream-split;
+        plain iterables are handed to every worker whole."""
+        if not self.datasets:
+            return None
+        shards = [dict() for _ in range(n)]
+        for name, ds in self.datasets.items():
+            splits = None
+            if hasattr(ds, "streaming_split"):
+                try:
+                    splits = ds.streaming_split(n, equal=True)
+                except Exception:
+                    splits = None
+            for i in range(n):
+                shards[i][name] = splits[i] if splits else ds
+        return shards
+
+    def run(self) -> Result:
+        max_failures = self.run_config.failure_config.max_failures
+        attempt = 0
+        error: Optional[BaseException] = None
+        while True:
+            group = WorkerGroup(self.scaling, self.torch_config, self.name,
+                                self.experiment_path)
+            try:
+                group.start()
+                restore = None
+                if self.ckpt_manager.latest:
+                    restore = self.ckpt_manager.latest
+                elif self.resume_from_checkpoint:
+                    restore = self.resume_from_checkpoint.path
+                group.start_training(
+                    self.train_fn, self.train_loop_config, restore,
+                    self._split_datasets(self.scaling.num_workers),
+                )
+                failed = self._poll_until_done(group)
+            except Exception as e:  # actor/scheduling level failure
+                logger.exception("worker group failed")
+                failed = e
+            finally:
+                group.shutdown()
+            if failed is None:
+                error = None
+                break
+            error = failed if isinstance(failed, BaseException) else RuntimeError(failed)
+            attempt += 1
+            if max_failures >= 0 and attempt > max_failures:
+                break
+            logger.warning("restarting worker group (attempt %d/%d): %s",
+                           attempt, max_failures, error)
+        latest_ckpt = (Checkpoint(self.ckpt_manager.latest)
+                       if self.ckpt_manager.latest else None)
+        return Result(
+            metrics=self.latest_metrics.get(0),
+            checkpoint=latest_ckpt,
+            path=self.experiment_path,
+            error=error,
+            best_checkpoints=self.ckpt_manager.best_checkpoints(),
+        )
+
+    def _poll_until_done(self, group: WorkerGroup):
+        """Returns None on clean finish, error string on worker failure."""
+        while True:
+            statuses = group.poll()
+            for st in statuses:
+                for rep in st["reports"]:
+                    self.latest_metrics[rep["rank"]] = rep["metrics"]
+                    if rep["checkpoint_path"]:
+                        self.ckpt_manager.register(rep["checkpoint_path"],
+                                                   rep["metrics"])
+            errs = [st["error"] for st in statuses if st["status"] == "errored"]
+            if errs:
+                return errs[0]
+            if all(st["status"] == "finished" for st in statuses):
+                return None
+            time.sleep(0.2)

@@ -1,0 +1,164 @@
+"""ray.train.torch — TorchTrainer + in-loop utilities.
+
+Role parity: reference python/ray/train/torch/ (TorchTrainer
+torch_trainer.py, train_loop_utils.py: prepare_model :374,
+prepare_data_loader, get_device, backward). MI355X-first difference: where
+the reference wraps torch DDP (C++ reducer, 25 MiB copy-packed buckets),
+prepare_model here wraps FlatDDP — all grads land directly in ONE flat
+communication buffer, bucketed RCCL all-reduce overlapped with backward,
+no pack/unpack copies (ant_ray_amd/parallel/flat.py).
+"""
+from __future__ import annotations
+
+import os
+from typing import Any, Optional
+
+import torch
+
+from ant_ray_amd.train.base_trainer import DataParallelTrainer
+from ant_ray_amd.train.config import TorchConfig
+from ant_ray_amd.train.session import get_context
+
+__all__ = [
+    "TorchTrainer", "TorchConfig", "prepare_model", "prepare_data_loader",
+    "prepare_optimizer", "get_device", "get_devices", "backward",
+    "enable_reproducibility",
+]
+
+
+class TorchTrainer(DataParallelTrainer):
+    def __init__(self, train_loop_per_worker, *, torch_config: Optional[TorchConfig] = None,
+                 **kwargs):
+        kwargs.setdefault("backend_config", torch_config or TorchConfig())
+        super().__init__(train_loop_per_worker, **kwargs)
+
+
+def get_device() -> torch.device:
+    """The device this worker rank owns (HIP_VISIBLE_DEVICES scoped by the
+    raylet, so local index 0..n-1; parity ray.train.torch.get_device)."""
+    if torch.cuda.is_available():
+        ctx = get_context()
+        idx = ctx.get_local_rank() % max(torch.cuda.device_count(), 1)
+        return torch.device(f"cuda:{idx}")
+    return torch.device("cpu")
+
+
+def get_devices():
+    return [get_device()]
+
+
+def prepare_model(
+    model: torch.nn.Module,
+    move_to_device: bool = True,
+    parallel_strategy: Optional[str] = "ddp",
+    parallel_strategy_kwargs: Optional[dict] = None,
+) -> torch.nn.Module:
+    """Move the model to this rank's device and wrap for data parallelism.
+
+    world_size>1 + "ddp" → FlatDDP with auto grad sync at backward end (the
+    returned module is a drop-in for a torch-DDP-style train loop: forward,
+    loss.backward(), optimizer.step()). "fsdp" → torch-ROCm FSDP passthrough
+    (parity train_loop_utils.py:171)."""
+    import torch.distributed as dist
+
+    from ant_ray_amd.parallel import FlatDDP, FlatParamManager
+
+    device = get_device()
+    if move_to_device:
+        model = model.to(device)
+    world = dist.get_world_size() if dist.is_initialized() else 1
+    if world <= 1 or parallel_strategy is None:
+        return model
+    kwargs = dict(parallel_strategy_kwargs or {})
+    if parallel_strategy == "fsdp":
+        from torch.distributed.fsdp import FullyShardedDataParallel
+
+        return FullyShardedDataParallel(model, **kwargs)
+    dtype = next(model.parameters()).dtype
+    mgr = FlatParamManager(model, device=device, dtype=dtype)
+    bucket_mb = kwargs.pop("bucket_mb", 64)
+    return FlatDDP(model, manager=mgr, bucket_mb=bucket_mb, auto_sync=True,
+                   average_grads=True)
+
+
+def prepare_optimizer(optimizer):
+    return optimizer
+
+
+def backward(tensor: torch.Tensor):
+    tensor.backward()
+
+
+def prepare_data_loader(
+    data_loader,
+    add_dist_sampler: bool = True,
+    move_to_device: bool = True,
+    auto_transfer: bool = True,
+):
+    """Shard a torch DataLoader across ranks and move batches to device.
+    Parity: train_loop_utils.py prepare_data_loader."""
+    import torch.distributed as dist
+    from torch.utils.data import DataLoader, DistributedSampler, IterableDataset
+
+    world = dist.get_world_size() if dist.is_initialized() else 1
+    if (world > 1 and add_dist_sampler
+            and not isinstance(data_loader.dataset, IterableDataset)
+            and not isinstance(data_loader.sampler, DistributedSampler)):
+        ctx = get_context()
+        sampler = DistributedSampler(
+            data_loader.dataset, num_replicas=world, rank=ctx.get_world_rank(),
+            shuffle=isinstance(
+                data_loader.sampler, torch.utils.data.RandomSampler
+            ),
+        )
+        data_loader = DataLoader(
+            data_loader.dataset,
+            batch_size=data_loader.batch_size,
+            sampler=sampler,
+            num_workers=data_loader.num_workers,
+            collate_fn=data_loader.collate_fn,
+            pin_memory=data_loader.pin_memory,
+            drop_last=data_loader.drop_last,
+        )
+    if not move_to_device:
+        return data_loader
+    return _DeviceDataLoader(data_loader, get_device())
+
+
+class _DeviceDataLoader:
+    """Iterates a DataLoader, moving each batch to `device` (non_blocking)."""
+
+    def __init__(self, loader, device):
+        self._loader = loader
+        self.device = device
+
+    def __len__(self):
+        return len(self._loader)
+
+    def __getattr__(self, name):
+        return getattr(self._loader, name)
+
+    def __iter__(self):
+        for batch in self._loader:
+            yield _move(batch, self.device)
+
+
+def _move(obj: Any, device):
+    if isinstance(obj, torch.Tensor):
+        return obj.to(device, non_blocking=True)
+    if isinstance(obj, (list, tuple)):
+        return type(obj)(_move(o, device) for o in obj)
+    if isinstance(obj, dict):
+        return {k: _move(v, device) for k, v in obj.items()}
+    return obj
+
+
+def enable_reproducibility(seed: int = 0):
+    import random
+
+    import numpy as np
+
+    torch.manual_seed(seed)
+    np.random.seed(seed)
+    random.seed(seed)
+    os.environ["PYTHONHASHSEED"] = str(seed)

@@ -1,0 +1,197 @@
+"""Ray Train parity tests on CPU: TorchTrainer over 2 gloo workers,
+report/checkpoint flow, failure restart, checkpoint directory format."""
+import json
+import os
+
+import pytest
+import torch
+
+
+@pytest.fixture(scope="module")
+def ray_mod():
+    import ant_ray_amd as ray
+
+    if not ray.is_initialized():
+        ray.init(num_cpus=6)
+    yield ray
+    ray.shutdown()
+
+
+def test_torch_trainer_2workers(ray_mod, tmp_path_factory):
+    from ant_ray_amd.train import (
+        Checkpoint,
+        RunConfig,
+        ScalingConfig,
+    )
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    storage = str(tmp_path_factory.mktemp("storage"))
+
+    def train_fn(config):
+        import tempfile
+
+        import torch.distributed as dist
+
+        from ant_ray_amd import train
+
+        ctx = train.get_context()
+        assert ctx.get_world_size() == 2
+        assert dist.is_initialized() and dist.get_world_size() == 2
+        rank = ctx.get_world_rank()
+
+        model = torch.nn.Linear(8, 4)
+        model = train.torch.prepare_model(model)
+        opt = torch.optim.SGD(model.parameters(), lr=0.1)
+        torch.manual_seed(100 + rank)
+        for step in range(3):
+            x = torch.randn(16, 8)
+            loss = model(x).pow(2).mean()
+            opt.zero_grad()
+            loss.backward()
+            opt.step()
+            if rank == 0:
+                with tempfile.TemporaryDirectory() as d:
+                    torch.save({"step": step}, os.path.join(d, "model.pt"))
+                    ckpt = Checkpoint.from_directory(d)
+                    ckpt.set_metadata({"step": step})
+                    train.report({"loss": float(loss), "step": step}, checkpoint=ckpt)
+            else:
+                train.report({"loss": float(loss), "step": step})
+
+    trainer = TorchTrainer(
+        train_fn,
+        scaling_config=ScalingConfig(num_workers=2),
+        run_config=RunConfig(name="t2w", storage_path=storage),
+    )
+    result = trainer.fit()
+    assert result.error is None
+    assert result.metrics["step"] == 2
+    # checkpoint directory format: storage/name/checkpoint_NNNNNN + .metadata.json
+    assert result.checkpoint is not None
+    assert os.path.basename(result.checkpoint.path) == "checkpoint_000002"
+    assert os.path.isfile(os.path.join(result.checkpoint.path, "model.pt"))
+    md = json.load(open(os.path.join(result.checkpoint.path, ".metadata.json")))
+    assert md == {"step": 2}
+    assert result.checkpoint.get_metadata() == {"step": 2}
+    sd = torch.load(os.path.join(result.checkpoint.path, "model.pt"))
+    assert sd["step"] == 2
+
+
+def test_trainer_failure_restart_restores_checkpoint(ray_mod, tmp_path_factory):
+    from ant_ray_amd.train import RunConfig, FailureConfig, ScalingConfig, Checkpoint
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    storage = str(tmp_path_factory.mktemp("storage"))
+    marker = os.path.join(storage, "fail_once")
+
+    def train_fn(config):
+        import tempfile
+
+        from ant_ray_amd import train
+
+        restored = train.get_checkpoint()
+        start = 0
+        if restored is not None:
+            start = restored.get_metadata()["step"] + 1
+        for step in range(start, 4):
+            if step == 2 and not os.path.exists(marker):
+                open(marker, "w").close()
+                raise RuntimeError("boom")
+            if train.get_context().get_world_rank() == 0:
+                with tempfile.TemporaryDirectory() as d:
+                    ckpt = Checkpoint.from_directory(d)
+                    ckpt.set_metadata({"step": step})
+                    train.report({"step": step, "resumed": start > 0},
+                                 checkpoint=ckpt)
+            else:
+                train.report({"step": step})
+
+    trainer = TorchTrainer(
+        train_fn,
+        scaling_config=ScalingConfig(num_workers=2),
+        run_config=RunConfig(
+            name="restart", storage_path=storage,
+            failure_config=FailureConfig(max_failures=1),
+        ),
+    )
+    result = trainer.fit()
+    assert result.error is None
+    # resumed from step-1 checkpoint after the injected failure
+    assert result.metrics["step"] == 3
+    assert result.metrics["resumed"] is True
+
+
+def test_trainer_raises_after_retries_exhausted(ray_mod, tmp_path_factory):
+    from ant_ray_amd.train import RunConfig, ScalingConfig
+    from ant_ray_amd.train.base_trainer import TrainingFailedError
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    storage = str(tmp_path_factory.mktemp("storage"))
+
+    def train_fn(config):
+        raise ValueError("always fails")
+
+    trainer = TorchTrainer(
+        train_fn,
+        scaling_config=ScalingConfig(num_workers=1),
+        run_config=RunConfig(name="fails", storage_path=storage),
+    )
+    with pytest.raises(TrainingFailedError):
+        trainer.fit()
+
+
+def test_checkpoint_num_to_keep(ray_mod, tmp_path_factory):
+    from ant_ray_amd.train import (
+        Checkpoint,
+        CheckpointConfig,
+        RunConfig,
+        ScalingConfig,
+    )
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    storage = str(tmp_path_factory.mktemp("storage"))
+
+    def train_fn(config):
+        import tempfile
+
+        from ant_ray_amd import train
+
+        for step in range(5):
+            with tempfile.TemporaryDirectory() as d:
+                ckpt = Checkpoint.from_directory(d)
+                ckpt.set_metadata({"step": step})
+                train.report({"step": step}, checkpoint=ckpt)
+
+    trainer = TorchTrainer(
+        train_fn,
+        scaling_config=ScalingConfig(num_workers=1),
+        run_config=RunConfig(
+            name="keep2", storage_path=storage,
+            checkpoint_config=CheckpointConfig(num_to_keep=2),
+        ),
+    )
+    result = trainer.fit()
+    exp = os.path.join(storage, "keep2")
+    kept = sorted(d for d in os.listdir(exp) if d.startswith("checkpoint_"))
+    assert kept == ["checkpoint_000003", "checkpoint_000004"]
+    assert result.checkpoint.path.endswith("checkpoint_000004")
+
+
+def test_prepare_data_loader_shards(ray_mod):
+    from ant_ray_amd.train import ScalingConfig
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    def train_fn(config):
+        from torch.utils.data import DataLoader, TensorDataset
+
+        from ant_ray_amd import train
+
+        ds = TensorDataset(torch.arange(20).float().unsqueeze(1))
+        dl = DataLoader(ds, batch_size=2)
+        dl = train.torch.prepare_data_loader(dl)
+        seen = sum(len(b[0]) for b in dl)
+        # DistributedSampler pads to equal shards: 10 rows per rank
+        assert seen == 10
+        train.report({"seen": seen})
+
+    TorchTrainer(train_fn, scaling_config=ScalingConfig(num_workers=2)).fit()
