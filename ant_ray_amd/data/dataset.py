@@ -1,0 +1,433 @@
+"""Dataset: lazy logical plan over blocks + consumption APIs.
+
+Role parity: reference python/ray/data/dataset.py (map_batches :467,
+streaming_split :1881, iter_batches via iterator.py). Laziness: transforms
+append logical ops; consumption (iter/take/write/materialize) runs the
+streaming executor.
+"""
+from __future__ import annotations
+
+import itertools
+from typing import Any, Callable, Dict, Iterator, List, Optional, Union
+
+import numpy as np
+import pyarrow as pa
+
+import ant_ray_amd as ray
+from ant_ray_amd.data.block import Block, BlockAccessor, block_from_dict
+from ant_ray_amd.data.context import DataContext
+from ant_ray_amd.data.executor import execute_plan
+from ant_ray_amd.data.plan import (
+    ActorPoolStrategy,
+    AllToAllOp,
+    LimitOp,
+    MapOp,
+    ReadOp,
+)
+
+
+class Dataset:
+    def __init__(self, ops: List[Any]):
+        self._ops = ops
+
+    # ------------------------------------------------------------ transforms
+
+    def _with(self, op) -> "Dataset":
+        return Dataset(self._ops + [op])
+
+    def map_batches(
+        self,
+        fn: Union[Callable, type],
+        *,
+        batch_size: Optional[int] = "default",
+        batch_format: Optional[str] = "default",
+        compute: Optional[Any] = None,
+        fn_args: tuple = (),
+        fn_kwargs: Optional[dict] = None,
+        fn_constructor_args: tuple = (),
+        fn_constructor_kwargs: Optional[dict] = None,
+        num_cpus: Optional[float] = None,
+        num_gpus: Optional[float] = None,
+        concurrency: Optional[Union[int, tuple]] = None,
+        zero_copy_batch: bool = False,
+        **_ignored,
+    ) -> "Dataset":
+        if batch_size == "default":
+            batch_size = 1024
+        import inspect
+
+        if compute is None and concurrency is not None and inspect.isclass(fn):
+            size = concurrency if isinstance(concurrency, int) else concurrency[-1]
+            compute = ActorPoolStrategy(size=size)
+        return self._with(MapOp(
+            name=f"MapBatches({getattr(fn, '__name__', type(fn).__name__)})",
+            kind="map_batches", fn=fn, batch_size=batch_size,
+            batch_format=batch_format, fn_args=fn_args,
+            fn_kwargs=fn_kwargs or {}, fn_constructor_args=fn_constructor_args,
+            fn_constructor_kwargs=fn_constructor_kwargs or {},
+            compute=compute, num_cpus=num_cpus, num_gpus=num_gpus,
+            concurrency=concurrency,
+        ))
+
+    def map(self, fn, *, compute=None, num_cpus=None, num_gpus=None, **_):
+        return self._with(MapOp(name="Map", kind="map_rows", fn=fn,
+                                compute=compute, num_cpus=num_cpus,
+                                num_gpus=num_gpus))
+
+    def flat_map(self, fn, **kw):
+        return self._with(MapOp(name="FlatMap", kind="flat_map", fn=fn))
+
+    def filter(self, fn, **kw):
+        return self._with(MapOp(name="Filter", kind="filter", fn=fn))
+
+    def add_column(self, col: str, fn, **kw):
+        def _add(batch):
+            batch[col] = fn(batch)
+            return batch
+
+        return self.map_batches(_add, batch_format="pandas")
+
+    def drop_columns(self, cols: List[str], **kw):
+        def _drop(t: pa.Table):
+            keep = [c for c in t.column_names if c not in cols]
+            return t.select(keep)
+
+        return self.map_batches(_drop, batch_format="pyarrow", batch_size=None)
+
+    def select_columns(self, cols: List[str], **kw):
+        return self.map_batches(lambda t: t.select(cols),
+                                batch_format="pyarrow", batch_size=None)
+
+    def rename_columns(self, mapping: Dict[str, str], **kw):
+        def _ren(t: pa.Table):
+            return t.rename_columns([mapping.get(c, c) for c in t.column_names])
+
+        return self.map_batches(_ren, batch_format="pyarrow", batch_size=None)
+
+    def limit(self, n: int) -> "Dataset":
+        return self._with(LimitOp(name=f"Limit[{n}]", limit=n))
+
+    def repartition(self, num_blocks: int, **kw) -> "Dataset":
+        def _repart(refs: List[Any]) -> List[Any]:
+            blocks = [ray.get(r) for r in refs]
+            if not blocks:
+                return []
+            table = pa.concat_tables(blocks, promote_options="default")
+            n = table.num_rows
+            per = max(1, -(-n // num_blocks))
+            out = []
+            for s in range(0, n, per):
+                out.append(ray.put(table.slice(s, per)))
+            return out
+
+        return self._with(AllToAllOp(name="Repartition", fn=_repart))
+
+    def random_shuffle(self, *, seed: Optional[int] = None, **kw) -> "Dataset":
+        def _shuffle(refs: List[Any]) -> List[Any]:
+            blocks = [ray.get(r) for r in refs]
+            if not blocks:
+                return []
+            table = pa.concat_tables(blocks, promote_options="default")
+            rng = np.random.RandomState(seed)
+            idx = rng.permutation(table.num_rows)
+            shuffled = table.take(pa.array(idx))
+            k = max(len(refs), 1)
+            per = max(1, -(-shuffled.num_rows // k))
+            return [ray.put(shuffled.slice(s, per))
+                    for s in range(0, shuffled.num_rows, per)]
+
+        return self._with(AllToAllOp(name="RandomShuffle", fn=_shuffle))
+
+    def sort(self, key: Union[str, List[str]], descending: bool = False) -> "Dataset":
+        keys = [key] if isinstance(key, str) else list(key)
+
+        def _sort(refs: List[Any]) -> List[Any]:
+            blocks = [ray.get(r) for r in refs]
+            if not blocks:
+                return []
+            table = pa.concat_tables(blocks, promote_options="default")
+            order = "descending" if descending else "ascending"
+            table = table.sort_by([(k, order) for k in keys])
+            k = max(len(refs), 1)
+            per = max(1, -(-table.num_rows // k))
+            return [ray.put(table.slice(s, per))
+                    for s in range(0, table.num_rows, per)]
+
+        return self._with(AllToAllOp(name="Sort", fn=_sort))
+
+    def union(self, *others: "Dataset") -> "Dataset":
+        me = self
+
+        def _union(refs: List[Any]) -> List[Any]:
+            out = list(refs)
+            for o in others:
+                out.extend(o.iter_internal_ref_bundles())
+            return out
+
+        return self._with(AllToAllOp(name="Union", fn=_union))
+
+    def groupby(self, key: str) -> "GroupedData":
+        return GroupedData(self, key)
+
+    def random_sample(self, fraction: float, *, seed=None) -> "Dataset":
+        rng = np.random.RandomState(seed)
+
+        def _sample(t: pa.Table):
+            mask = rng.rand(t.num_rows) < fraction
+            return t.filter(pa.array(mask))
+
+        return self.map_batches(_sample, batch_format="pyarrow", batch_size=None)
+
+    # ----------------------------------------------------------- consumption
+
+    def iter_internal_ref_bundles(self) -> Iterator[Any]:
+        return execute_plan(self._ops)
+
+    def iter_blocks(self) -> Iterator[Block]:
+        for ref in self.iter_internal_ref_bundles():
+            yield ray.get(ref)
+
+    def take(self, limit: int = 20) -> List[Dict[str, Any]]:
+        out = []
+        for block in self.limit(limit).iter_blocks():
+            out.extend(BlockAccessor(block).iter_rows())
+            if len(out) >= limit:
+                break
+        return out[:limit]
+
+    def take_all(self) -> List[Dict[str, Any]]:
+        out = []
+        for block in self.iter_blocks():
+            out.extend(BlockAccessor(block).iter_rows())
+        return out
+
+    def take_batch(self, batch_size: int = 20, *, batch_format="default"):
+        for batch in self.iter_batches(batch_size=batch_size,
+                                       batch_format=batch_format):
+            return batch
+        return {}
+
+    def show(self, limit: int = 20):
+        for row in self.take(limit):
+            print(row)
+
+    def count(self) -> int:
+        ops = self._ops
+        if len(ops) == 1 and isinstance(ops[0], ReadOp) and ops[0].num_rows is not None:
+            return ops[0].num_rows
+        return sum(BlockAccessor(b).num_rows() for b in self.iter_blocks())
+
+    def schema(self):
+        for block in self.iter_blocks():
+            return BlockAccessor(block).schema()
+        return None
+
+    def columns(self) -> List[str]:
+        s = self.schema()
+        return list(s.names) if s is not None else []
+
+    def iter_rows(self) -> Iterator[Dict[str, Any]]:
+        for block in self.iter_blocks():
+            yield from BlockAccessor(block).iter_rows()
+
+    def iter_batches(self, *, batch_size: Optional[int] = 256,
+                     batch_format: str = "default", drop_last: bool = False,
+                     prefetch_batches: int = 1, **_) -> Iterator[Any]:
+        from ant_ray_amd.data.iterator import rebatch_blocks
+
+        yield from rebatch_blocks(self.iter_blocks(), batch_size, batch_format,
+                                  drop_last)
+
+    def iter_torch_batches(self, *, batch_size: Optional[int] = 256,
+                           dtypes=None, device: Optional[str] = None,
+                           collate_fn: Optional[Callable] = None,
+                           drop_last: bool = False, **kw) -> Iterator[Any]:
+        from ant_ray_amd.data.iterator import to_torch_batch
+
+        for batch in self.iter_batches(batch_size=batch_size, drop_last=drop_last):
+            yield to_torch_batch(batch, dtypes, device, collate_fn)
+
+    def to_pandas(self, limit: Optional[int] = None):
+        import pandas as pd
+
+        blocks = list((self.limit(limit) if limit else self).iter_blocks())
+        if not blocks:
+            return pd.DataFrame()
+        return pa.concat_tables(blocks, promote_options="default").to_pandas()
+
+    def to_arrow_refs(self) -> List[Any]:
+        return list(self.iter_internal_ref_bundles())
+
+    def materialize(self) -> "MaterializedDataset":
+        refs = list(self.iter_internal_ref_bundles())
+        return MaterializedDataset(refs)
+
+    def stats(self) -> str:
+        return f"Dataset(ops={[getattr(o, 'name', o) for o in self._ops]})"
+
+    def num_blocks(self) -> int:
+        return len(list(self.iter_internal_ref_bundles()))
+
+    def size_bytes(self) -> int:
+        return sum(BlockAccessor(b).size_bytes() for b in self.iter_blocks())
+
+    # ----------------------------------------------------------------- split
+
+    def split(self, n: int, *, equal: bool = False, locality_hints=None
+              ) -> List["MaterializedDataset"]:
+        refs = list(self.iter_internal_ref_bundles())
+        if equal:
+            blocks = [ray.get(r) for r in refs]
+            table = (pa.concat_tables(blocks, promote_options="default")
+                     if blocks else pa.table({}))
+            per = table.num_rows // n  # equal shards; remainder rows dropped
+            return [
+                MaterializedDataset([ray.put(table.slice(i * per, per))])
+                for i in range(n)
+            ]
+        shards: List[List[Any]] = [[] for _ in range(n)]
+        for i, r in enumerate(refs):
+            shards[i % n].append(r)
+        return [MaterializedDataset(s) for s in shards]
+
+    def train_test_split(self, test_size: float, *, shuffle: bool = False,
+                         seed=None):
+        ds = self.random_shuffle(seed=seed) if shuffle else self
+        rows = ds.take_all()
+        n_test = int(len(rows) * test_size)
+        from ant_ray_amd.data import from_items
+
+        return from_items(rows[: len(rows) - n_test]), from_items(
+            rows[len(rows) - n_test:])
+
+    def streaming_split(self, n: int, *, equal: bool = False,
+                        locality_hints=None) -> List[Any]:
+        """N coordinated iterators over one execution (parity dataset.py:1881;
+        used by Train's get_dataset_shard)."""
+        from ant_ray_amd.data.iterator import SplitCoordinator, DataIterator
+
+        coord = ray.remote(max_concurrency=2 * n + 2)(SplitCoordinator).remote(
+            self, n, equal
+        )
+        return [DataIterator(coord, i) for i in range(n)]
+
+    # ----------------------------------------------------------------- write
+
+    def write_parquet(self, path: str, **kw):
+        self._write(path, "parquet")
+
+    def write_csv(self, path: str, **kw):
+        self._write(path, "csv")
+
+    def write_json(self, path: str, **kw):
+        self._write(path, "json")
+
+    def _write(self, path: str, fmt: str):
+        import os
+
+        os.makedirs(path, exist_ok=True)
+        refs = []
+        write_remote = ray.remote(_write_block)
+        for i, ref in enumerate(self.iter_internal_ref_bundles()):
+            refs.append(write_remote.remote(ref, path, fmt, i))
+        ray.get(refs)
+
+    def __repr__(self):
+        return self.stats()
+
+
+def _write_block(block, path, fmt, index):
+    import os
+
+    import pyarrow.csv as pcsv
+    import pyarrow.parquet as pq
+
+    f = os.path.join(path, f"part-{index:06d}.{fmt}")
+    if fmt == "parquet":
+        pq.write_table(block, f)
+    elif fmt == "csv":
+        pcsv.write_csv(block, f)
+    elif fmt == "json":
+        import json
+
+        rows = list(BlockAccessor(block).iter_rows())
+        with open(f, "w") as fh:
+            for r in rows:
+                fh.write(json.dumps({k: _json_safe(v) for k, v in r.items()}) + "\n")
+    return f
+
+
+def _json_safe(v):
+    if isinstance(v, (np.integer,)):
+        return int(v)
+    if isinstance(v, (np.floating,)):
+        return float(v)
+    if isinstance(v, np.ndarray):
+        return v.tolist()
+    return v
+
+
+class MaterializedDataset(Dataset):
+    """A Dataset whose blocks are already in the object store."""
+
+    def __init__(self, refs: List[Any]):
+        super().__init__([
+            ReadOp(name="Materialized", read_tasks=[], num_rows=None),
+            AllToAllOp(name="Materialized", fn=lambda _refs, refs=refs: refs),
+        ])
+        self._refs = refs
+
+
+class GroupedData:
+    """Minimal groupby: aggregates + map_groups (parity data/grouped_data.py)."""
+
+    def __init__(self, ds: Dataset, key: str):
+        self.ds = ds
+        self.key = key
+
+    def _agg(self, aggs: List[tuple]) -> Dataset:
+        key = self.key
+
+        def _do(refs: List[Any]) -> List[Any]:
+            blocks = [ray.get(r) for r in refs]
+            if not blocks:
+                return []
+            table = pa.concat_tables(blocks, promote_options="default")
+            res = table.group_by(key).aggregate(aggs)
+            # arrow names output "<col>_<agg>"; keep as-is
+            return [ray.put(res)]
+
+        return self.ds._with(AllToAllOp(name="GroupByAgg", fn=_do))
+
+    def count(self) -> Dataset:
+        return self._agg([(self.key, "count")])
+
+    def sum(self, col: str) -> Dataset:
+        return self._agg([(col, "sum")])
+
+    def mean(self, col: str) -> Dataset:
+        return self._agg([(col, "mean")])
+
+    def min(self, col: str) -> Dataset:
+        return self._agg([(col, "min")])
+
+    def max(self, col: str) -> Dataset:
+        return self._agg([(col, "max")])
+
+    def map_groups(self, fn, *, batch_format="pandas") -> Dataset:
+        key = self.key
+
+        def _do(refs: List[Any]) -> List[Any]:
+            blocks = [ray.get(r) for r in refs]
+            if not blocks:
+                return []
+            table = pa.concat_tables(blocks, promote_options="default")
+            df = table.to_pandas()
+            outs = []
+            for _, grp in df.groupby(key):
+                res = fn(grp if batch_format == "pandas" else
+                         {c: grp[c].to_numpy() for c in grp.columns})
+                outs.append(BlockAccessor.for_block(res).to_arrow())
+            return [ray.put(pa.concat_tables(outs, promote_options="default"))] if outs else []
+
+        return self.ds._with(AllToAllOp(name="MapGroups", fn=_do))

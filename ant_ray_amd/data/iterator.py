@@ -1,0 +1,157 @@
+"""Batch iteration + streaming_split coordination.
+
+Role parity: reference python/ray/data/iterator.py (iter_batches/
+iter_torch_batches) and _internal/execution/streaming_executor's
+streaming_split output splitter (output_splitter.py). The SplitCoordinator
+actor drives ONE execution of the dataset and deals blocks to N consumer
+shards on demand (Train workers each pull their shard).
+"""
+from __future__ import annotations
+
+import threading
+from typing import Any, Callable, Dict, Iterator, List, Optional
+
+import numpy as np
+import pyarrow as pa
+
+from ant_ray_amd.data.block import BlockAccessor
+
+
+def rebatch_blocks(blocks: Iterator[Any], batch_size: Optional[int],
+                   batch_format: str, drop_last: bool) -> Iterator[Any]:
+    """Slice a stream of arrow blocks into exact-size batches."""
+    buf: List[pa.Table] = []
+    buffered = 0
+    for block in blocks:
+        if block.num_rows == 0:
+            continue
+        if batch_size is None:
+            yield BlockAccessor(block).to_batch(batch_format)
+            continue
+        buf.append(block)
+        buffered += block.num_rows
+        while buffered >= batch_size:
+            table = pa.concat_tables(buf, promote_options="default")
+            out = table.slice(0, batch_size)
+            rest = table.slice(batch_size)
+            buf = [rest] if rest.num_rows else []
+            buffered = rest.num_rows
+            yield BlockAccessor(out).to_batch(batch_format)
+    if buffered and not (drop_last and batch_size and buffered < batch_size):
+        table = pa.concat_tables(buf, promote_options="default")
+        yield BlockAccessor(table).to_batch(batch_format)
+
+
+def to_torch_batch(batch: Dict[str, np.ndarray], dtypes=None,
+                   device: Optional[str] = None,
+                   collate_fn: Optional[Callable] = None):
+    import torch
+
+    if collate_fn is not None:
+        return collate_fn(batch)
+    out = {}
+    for k, v in batch.items():
+        if isinstance(v, np.ndarray) and v.dtype != object:
+            t = torch.from_numpy(np.ascontiguousarray(v))
+        else:
+            t = v
+        if isinstance(t, torch.Tensor):
+            if isinstance(dtypes, dict) and k in dtypes:
+                t = t.to(dtypes[k])
+            elif dtypes is not None and not isinstance(dtypes, dict):
+                t = t.to(dtypes)
+            if device:
+                t = t.to(device, non_blocking=True)
+        out[k] = t
+    return out
+
+
+class SplitCoordinator:
+    """Actor: runs the dataset's executor once; shards pull blocks.
+
+    equal=True → blocks are dealt strictly round-robin starting from shard 0
+    and the stream ends for everyone when the source ends (row-equality is
+    approximate at block granularity, parity with the reference's
+    per-bundle splitter)."""
+
+    def __init__(self, dataset, n: int, equal: bool):
+        self.n = n
+        self.equal = equal
+        self._lock = threading.Lock()
+        self._iter = dataset.iter_internal_ref_bundles()
+        self._queues: List[List[Any]] = [[] for _ in range(n)]
+        self._done = False
+        self._next_shard = 0
+        self._epoch_ids = [0] * n
+
+    def next_block(self, shard: int):
+        """Returns a block ref or None when exhausted."""
+        with self._lock:
+            while not self._queues[shard] and not self._done:
+                try:
+                    ref = next(self._iter)
+                except StopIteration:
+                    self._done = True
+                    break
+                self._queues[self._next_shard].append(ref)
+                self._next_shard = (self._next_shard + 1) % self.n
+            if self._queues[shard]:
+                return self._queues[shard].pop(0)
+            return None
+
+
+class DataIterator:
+    """Per-shard iterator handle (parity ray.data.DataIterator — what
+    train.get_dataset_shard returns)."""
+
+    def __init__(self, coordinator, shard: int):
+        self._coord = coordinator
+        self._shard = shard
+
+    def _iter_blocks(self) -> Iterator[Any]:
+        import ant_ray_amd as ray
+
+        while True:
+            ref = ray.get(self._coord.next_block.remote(self._shard))
+            if ref is None:
+                return
+            yield ray.get(ref)
+
+    def iter_batches(self, *, batch_size: Optional[int] = 256,
+                     batch_format: str = "default", drop_last: bool = False,
+                     prefetch_batches: int = 1, **_) -> Iterator[Any]:
+        yield from rebatch_blocks(self._iter_blocks(), batch_size,
+                                  batch_format, drop_last)
+
+    def iter_torch_batches(self, *, batch_size: Optional[int] = 256,
+                           dtypes=None, device: Optional[str] = None,
+                           collate_fn: Optional[Callable] = None,
+                           drop_last: bool = False, **kw) -> Iterator[Any]:
+        if device is None:
+            import torch
+
+            if torch.cuda.is_available():
+                from ant_ray_amd.train.session import get_context
+
+                idx = get_context().get_local_rank() % max(
+                    torch.cuda.device_count(), 1)
+                device = f"cuda:{idx}"
+        for batch in self.iter_batches(batch_size=batch_size,
+                                       drop_last=drop_last):
+            yield to_torch_batch(batch, dtypes, device, collate_fn)
+
+    def iter_rows(self) -> Iterator[Dict[str, Any]]:
+        for block in self._iter_blocks():
+            yield from BlockAccessor(block).iter_rows()
+
+    def materialize(self):
+        from ant_ray_amd.data.dataset import MaterializedDataset
+        import ant_ray_amd as ray
+
+        refs = []
+        while True:
+            ref = ray.get(self._coord.next_block.remote(self._shard))
+            if ref is None:
+                break
+            refs.append(ref)
+        return MaterializedDataset(refs)

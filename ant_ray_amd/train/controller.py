@@ -128,9 +128,12 @@ class TrainController:
                     restore = self.ckpt_manager.latest
                 elif self.resume_from_checkpoint:
                     restore = self.resume_from_checkpoint.path
+                # keep shard iterators referenced for the whole attempt: they
+                # hold the SplitCoordinator actor handle alive
+                self._dataset_shards = self._split_datasets(self.scaling.num_workers)
                 group.start_training(
                     self.train_fn, self.train_loop_config, restore,
-                    self._split_datasets(self.scaling.num_workers),
+                    self._dataset_shards,
                 )
                 failed = self._poll_until_done(group)
             except Exception as e:  # actor/scheduling level failure

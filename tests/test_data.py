@@ -1,0 +1,186 @@
+"""Ray Data parity tests on CPU: creation, transforms, execution, splits,
+writes, and Train integration (streaming_split shards)."""
+import os
+
+import numpy as np
+import pytest
+
+
+@pytest.fixture(scope="module")
+def ray_mod():
+    import ant_ray_amd as ray
+
+    if not ray.is_initialized():
+        ray.init(num_cpus=8)
+    yield ray
+    ray.shutdown()
+
+
+@pytest.fixture(scope="module")
+def data(ray_mod):
+    import ant_ray_amd.data as data
+
+    return data
+
+
+def test_range_count_take(data):
+    ds = data.range(100)
+    assert ds.count() == 100
+    rows = ds.take(5)
+    assert rows == [{"id": i} for i in range(5)]
+
+
+def test_map_batches_numpy(data):
+    ds = data.range(1000).map_batches(lambda b: {"id": b["id"] * 2})
+    out = ds.take_all()
+    assert len(out) == 1000
+    assert sorted(r["id"] for r in out) == [2 * i for i in range(1000)]
+
+
+def test_map_filter_flatmap_fusion(data):
+    ds = (data.range(50)
+          .map(lambda r: {"id": r["id"] + 1})
+          .filter(lambda r: r["id"] % 2 == 0)
+          .flat_map(lambda r: [{"v": r["id"]}, {"v": -r["id"]}]))
+    vals = sorted(r["v"] for r in ds.take_all())
+    expect = sorted(v for i in range(50) if (i + 1) % 2 == 0 for v in ((i + 1), -(i + 1)))
+    assert vals == expect
+
+
+def test_map_batches_actor_class(data):
+    class AddConst:
+        def __init__(self, c):
+            self.c = c
+
+        def __call__(self, batch):
+            return {"id": batch["id"] + self.c}
+
+    ds = data.range(100).map_batches(
+        AddConst, fn_constructor_args=(7,), concurrency=2, batch_size=32,
+    )
+    out = sorted(r["id"] for r in ds.take_all())
+    assert out == [i + 7 for i in range(100)]
+
+
+def test_batch_formats(data):
+    import pandas as pd
+    import pyarrow as pa
+
+    ds = data.range(10)
+
+    def check_pandas(df):
+        assert isinstance(df, pd.DataFrame), type(df)
+        df["is_df"] = 1
+        return df
+
+    def check_arrow(t):
+        assert isinstance(t, pa.Table), type(t)
+        return t.append_column("is_tbl", pa.array([1] * t.num_rows))
+
+    assert all(r["is_df"] == 1 for r in
+               ds.map_batches(check_pandas, batch_format="pandas").take_all())
+    assert all(r["is_tbl"] == 1 for r in
+               ds.map_batches(check_arrow, batch_format="pyarrow").take_all())
+
+
+def test_sort_shuffle_repartition(data):
+    ds = data.range(200).random_shuffle(seed=4)
+    ids = [r["id"] for r in ds.take_all()]
+    assert ids != list(range(200)) and sorted(ids) == list(range(200))
+    ds2 = data.from_items([{"k": i % 5, "v": i} for i in range(100)]).sort(
+        "v", descending=True)
+    vs = [r["v"] for r in ds2.take_all()]
+    assert vs == sorted(vs, reverse=True)
+    assert data.range(100).repartition(7).num_blocks() == 7
+
+
+def test_groupby(data):
+    ds = data.from_items([{"k": i % 3, "v": float(i)} for i in range(30)])
+    res = {r["k"]: r["v_sum"] for r in ds.groupby("k").sum("v").take_all()}
+    expect = {}
+    for i in range(30):
+        expect[i % 3] = expect.get(i % 3, 0.0) + i
+    assert res == expect
+
+
+def test_iter_batches_exact_sizes(data):
+    ds = data.range(100, override_num_blocks=7)
+    sizes = [len(b["id"]) for b in ds.iter_batches(batch_size=32)]
+    assert sizes == [32, 32, 32, 4]
+    sizes = [len(b["id"]) for b in ds.iter_batches(batch_size=32, drop_last=True)]
+    assert sizes == [32, 32, 32]
+
+
+def test_iter_torch_batches(data):
+    import torch
+
+    ds = data.range(16)
+    batches = list(ds.iter_torch_batches(batch_size=8, dtypes=torch.float32))
+    assert len(batches) == 2
+    assert batches[0]["id"].dtype == torch.float32
+    assert batches[0]["id"].shape == (8,)
+
+
+def test_tensor_columns(data):
+    arr = np.random.rand(32, 4, 4).astype(np.float32)
+    ds = data.from_numpy(arr)
+    got = ds.take_batch(32)["data"]
+    np.testing.assert_allclose(got.reshape(32, 4, 4), arr)
+
+
+def test_write_read_roundtrip(data, tmp_path):
+    p = str(tmp_path / "pq")
+    data.range(100).map_batches(lambda b: {"id": b["id"], "sq": b["id"] ** 2}).write_parquet(p)
+    back = data.read_parquet(p)
+    rows = back.take_all()
+    assert len(rows) == 100
+    assert all(r["sq"] == r["id"] ** 2 for r in rows)
+    c = str(tmp_path / "csv")
+    data.range(10).write_csv(c)
+    assert data.read_csv(c).count() == 10
+
+
+def test_split_and_streaming_split(data):
+    ds = data.range(100, override_num_blocks=10)
+    shards = ds.split(4)
+    assert sum(s.count() for s in shards) == 100
+    its = ds.streaming_split(2, equal=True)
+    seen = [sorted(r["id"] for r in it.iter_rows()) for it in its]
+    assert len(seen[0]) + len(seen[1]) == 100
+    assert not (set(seen[0]) & set(seen[1]))
+
+
+def test_limit_streams_early(data):
+    ds = data.range(10_000, override_num_blocks=100)
+    assert len(ds.take(30)) == 30
+
+
+def test_train_get_dataset_shard(data, ray_mod):
+    from ant_ray_amd.train import ScalingConfig
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    ds = data.range(64, override_num_blocks=8)
+
+    def train_fn(config):
+        import torch
+        import torch.distributed as dist
+
+        from ant_ray_amd import train
+
+        shard = train.get_dataset_shard("train")
+        total = 0
+        for batch in shard.iter_torch_batches(batch_size=8):
+            total += len(batch["id"])
+        t = torch.tensor([total])
+        dist.all_reduce(t)
+        train.report({"rows": total, "total_rows": int(t.item())})
+
+    trainer = TorchTrainer(
+        train_fn,
+        scaling_config=ScalingConfig(num_workers=2),
+        datasets={"train": ds},
+    )
+    result = trainer.fit()
+    # every row lands on exactly one worker across the split
+    assert result.metrics["total_rows"] == 64
+    assert result.metrics["rows"] == 32  # 8 equal blocks dealt round-robin
