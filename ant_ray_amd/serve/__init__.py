@@ -1,0 +1,296 @@
+"""ant_ray_amd.serve — Ray Serve parity: scalable model serving on actors.
+
+Role parity: reference python/ray/serve/ (~102k LoC).
+Surface: @serve.deployment / .bind() / serve.run / serve.start /
+serve.shutdown / serve.delete / serve.status / get_app_handle /
+get_deployment_handle / @serve.batch, DeploymentHandle composition.
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Any, Callable, Dict, List, Optional, Union
+
+from ant_ray_amd.serve._private.common import (
+    CONTROLLER_NAME,
+    DEFAULT_APP_NAME,
+    AutoscalingConfig,
+)
+from ant_ray_amd.serve.batching import batch
+from ant_ray_amd.serve.handle import DeploymentHandle, DeploymentResponse
+
+PROXY_NAME = "SERVE_PROXY_ACTOR"
+
+__all__ = [
+    "Application", "AutoscalingConfig", "Deployment", "DeploymentHandle",
+    "DeploymentResponse", "batch", "delete", "deployment",
+    "get_app_handle", "get_deployment_handle", "ingress", "run", "shutdown",
+    "start", "status",
+]
+
+
+@dataclass
+class Application:
+    """A deployment bound to its constructor args (possibly nested)."""
+
+    deployment: "Deployment"
+    args: tuple = ()
+    kwargs: dict = field(default_factory=dict)
+
+
+class Deployment:
+    def __init__(self, target: Union[Callable, type], name: str,
+                 num_replicas: int = 1, ray_actor_options: Optional[dict] = None,
+                 max_ongoing_requests: int = 100, user_config: Any = None,
+                 autoscaling_config: Optional[Union[dict, AutoscalingConfig]] = None,
+                 health_check_period_s: float = 10.0,
+                 graceful_shutdown_timeout_s: float = 20.0, **_extra):
+        self._target = target
+        self.name = name
+        self.num_replicas = num_replicas
+        self.ray_actor_options = ray_actor_options or {}
+        self.max_ongoing_requests = max_ongoing_requests
+        self.user_config = user_config
+        self.autoscaling_config = autoscaling_config
+        self.health_check_period_s = health_check_period_s
+        self.graceful_shutdown_timeout_s = graceful_shutdown_timeout_s
+
+    def options(self, **kwargs) -> "Deployment":
+        merged = dict(
+            name=self.name, num_replicas=self.num_replicas,
+            ray_actor_options=self.ray_actor_options,
+            max_ongoing_requests=self.max_ongoing_requests,
+            user_config=self.user_config,
+            autoscaling_config=self.autoscaling_config,
+            health_check_period_s=self.health_check_period_s,
+            graceful_shutdown_timeout_s=self.graceful_shutdown_timeout_s,
+        )
+        merged.update(kwargs)
+        return Deployment(self._target, **merged)
+
+    def bind(self, *args, **kwargs) -> Application:
+        return Application(self, args, kwargs)
+
+    def __call__(self, *a, **kw):
+        raise RuntimeError(
+            "deployments are not callable directly; use .bind() + serve.run "
+            "then handle.remote()"
+        )
+
+    def __repr__(self):
+        return f"Deployment(name={self.name!r})"
+
+
+def deployment(_target=None, **kwargs):
+    """@serve.deployment decorator (parity serve/api.py:321)."""
+
+    def wrap(target):
+        name = kwargs.pop("name", None) or target.__name__
+        return Deployment(target, name=name, **kwargs)
+
+    if _target is not None:
+        return wrap(_target)
+    return wrap
+
+
+def ingress(asgi_app):
+    """@serve.ingress(fastapi_app): class decorator mounting the deployment's
+    methods behind a FastAPI/starlette app (parity serve/api.py ingress)."""
+
+    def wrap(cls):
+        class ASGIWrapped(cls):
+            __name__ = cls.__name__
+
+            async def __call__(self, request):
+                # run one request through the ASGI app
+                scope = dict(request.scope)
+                body = await request.body()
+                scope["app_root_path"] = ""
+                messages = [{"type": "http.request", "body": body,
+                             "more_body": False}]
+                sent: List[dict] = []
+
+                async def receive():
+                    return messages.pop(0) if messages else {
+                        "type": "http.disconnect"}
+
+                async def send(msg):
+                    sent.append(msg)
+
+                # make `self` reachable from route functions via app state
+                asgi_app.state.serve_self = self
+                await asgi_app(scope, receive, send)
+                status = 200
+                headers: List = []
+                chunks = []
+                for m in sent:
+                    if m["type"] == "http.response.start":
+                        status = m["status"]
+                        headers = m.get("headers", [])
+                    elif m["type"] == "http.response.body":
+                        chunks.append(m.get("body", b""))
+                from starlette.responses import Response
+
+                return Response(
+                    content=b"".join(chunks), status_code=status,
+                    headers={k.decode(): v.decode() for k, v in headers},
+                )
+
+        return ASGIWrapped
+
+    return wrap
+
+
+# ------------------------------------------------------------- lifecycle
+
+
+def _get_or_create_controller(http_host="127.0.0.1", http_port=8000):
+    import ant_ray_amd as ray
+
+    from ant_ray_amd.serve._private.controller import ServeController
+
+    if not ray.is_initialized():
+        ray.init()
+    try:
+        return ray.get_actor(CONTROLLER_NAME)
+    except Exception:
+        pass
+    Controller = ray.remote(ServeController)
+    c = Controller.options(
+        name=CONTROLLER_NAME, lifetime="detached", num_cpus=0,
+        max_concurrency=1000,
+    ).remote(http_port, http_host)
+    return c
+
+
+def start(detached: bool = True, http_options: Optional[dict] = None, **_):
+    """Start Serve system actors (controller + HTTP proxy)."""
+    import ant_ray_amd as ray
+
+    http_options = http_options or {}
+    host = http_options.get("host", "127.0.0.1")
+    port = http_options.get("port", 8000)
+    controller = _get_or_create_controller(host, port)
+    try:
+        proxy = ray.get_actor(PROXY_NAME)
+    except Exception:
+        from ant_ray_amd.serve._private.proxy import HTTPProxy
+
+        Proxy = ray.remote(HTTPProxy)
+        proxy = Proxy.options(
+            name=PROXY_NAME, lifetime="detached", num_cpus=0,
+            max_concurrency=1000,
+        ).remote(host, port)
+        ray.get(proxy.ready.remote(), timeout=60)
+    return controller
+
+
+def _collect_deployments(app: Application, out: Dict[str, dict],
+                         app_name: str):
+    """Topological flatten: nested bound apps become handles."""
+    import cloudpickle
+
+    def resolve(v):
+        if isinstance(v, Application):
+            _collect_deployments(v, out, app_name)
+            return DeploymentHandle(v.deployment.name, app_name)
+        return v
+
+    d = app.deployment
+    init_args = tuple(resolve(a) for a in app.args)
+    init_kwargs = {k: resolve(v) for k, v in app.kwargs.items()}
+    auto = d.autoscaling_config
+    if isinstance(auto, AutoscalingConfig):
+        auto = auto.__dict__
+    out[d.name] = {
+        "name": d.name,
+        "callable_bytes": cloudpickle.dumps(d._target),
+        "init_args": init_args,
+        "init_kwargs": init_kwargs,
+        "num_replicas": d.num_replicas,
+        "max_ongoing_requests": d.max_ongoing_requests,
+        "ray_actor_options": d.ray_actor_options,
+        "autoscaling_config": auto,
+        "user_config": d.user_config,
+    }
+
+
+def run(target: Application, *, name: str = DEFAULT_APP_NAME,
+        route_prefix: str = "/", blocking: bool = False,
+        _local_testing_mode: bool = False, **_) -> DeploymentHandle:
+    """Deploy an application; returns the ingress DeploymentHandle
+    (parity serve/api.py:686)."""
+    import ant_ray_amd as ray
+
+    if not isinstance(target, Application):
+        raise TypeError("serve.run expects a bound deployment "
+                        "(Deployment.bind(...))")
+    controller = start()
+    deployments: Dict[str, dict] = {}
+    _collect_deployments(target, deployments, name)
+    ingress_name = target.deployment.name
+    ray.get(controller.deploy_application.remote(
+        name, route_prefix, list(deployments.values()), ingress_name,
+    ), timeout=300)
+    handle = DeploymentHandle(ingress_name, name)
+    if blocking:
+        try:
+            while True:
+                time.sleep(1)
+        except KeyboardInterrupt:
+            pass
+    return handle
+
+
+def delete(name: str, _blocking: bool = True):
+    import ant_ray_amd as ray
+
+    try:
+        controller = ray.get_actor(CONTROLLER_NAME)
+    except Exception:
+        return
+    ray.get(controller.delete_application.remote(name), timeout=120)
+
+
+def status() -> dict:
+    import ant_ray_amd as ray
+
+    try:
+        controller = ray.get_actor(CONTROLLER_NAME)
+    except Exception:
+        return {"applications": {}}
+    return {"applications": ray.get(controller.list_applications.remote(),
+                                    timeout=60)}
+
+
+def shutdown():
+    import ant_ray_amd as ray
+
+    try:
+        controller = ray.get_actor(CONTROLLER_NAME)
+    except Exception:
+        return
+    try:
+        ray.get(controller.graceful_shutdown.remote(), timeout=120)
+    except Exception:
+        pass
+    for actor_name in (PROXY_NAME, CONTROLLER_NAME):
+        try:
+            ray.kill(ray.get_actor(actor_name))
+        except Exception:
+            pass
+
+
+def get_app_handle(name: str = DEFAULT_APP_NAME) -> DeploymentHandle:
+    import ant_ray_amd as ray
+
+    controller = ray.get_actor(CONTROLLER_NAME)
+    cfg = ray.get(controller.get_app_config.remote(name), timeout=60)
+    if cfg is None:
+        raise RuntimeError(f"no application named {name!r}")
+    return DeploymentHandle(cfg["ingress"], name)
+
+
+def get_deployment_handle(deployment_name: str,
+                          app_name: str = DEFAULT_APP_NAME) -> DeploymentHandle:
+    return DeploymentHandle(deployment_name, app_name)

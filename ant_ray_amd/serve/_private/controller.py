@@ -1,0 +1,221 @@
+"""ServeController actor: app/deployment/replica state reconciliation.
+
+Role parity: reference python/ray/serve/_private/controller.py:105
+(ServeController singleton actor), application_state.py /
+deployment_state.py (state machines), autoscaling_policy.py:16
+(_calculate_desired_num_replicas). Config push is pull-based here (handles/
+proxies refresh replica lists from the controller, parity long_poll.py in
+spirit), replica fan-out and health checks are asyncio background loops in
+this actor.
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import pickle
+import time
+from typing import Any, Dict, List, Optional
+
+import ant_ray_amd as ray
+from ant_ray_amd.serve._private.common import AutoscalingConfig
+from ant_ray_amd.serve._private.replica import Replica
+
+logger = logging.getLogger("antray.serve.controller")
+
+
+def _calculate_desired_num_replicas(cfg: AutoscalingConfig,
+                                    total_ongoing: float,
+                                    current: int) -> int:
+    """Parity: serve/autoscaling_policy.py:16 — replicas sized so each sees
+    ~target_ongoing_requests."""
+    if current == 0:
+        return cfg.min_replicas
+    desired = total_ongoing / max(cfg.target_ongoing_requests, 1e-9)
+    import math
+
+    desired = math.ceil(desired)
+    return max(cfg.min_replicas, min(cfg.max_replicas, desired))
+
+
+class DeploymentReplicas:
+    def __init__(self):
+        self.replicas: List[Any] = []  # actor handles
+        self.target: int = 0
+        self.cfg: Dict[str, Any] = {}
+        self.last_scale_up = 0.0
+        self.last_scale_down = 0.0
+        self.ongoing_history: List[float] = []
+
+
+class ServeController:
+    """Detached named actor; all methods are async (max_concurrency high)."""
+
+    def __init__(self, http_port: int = 8000, http_host: str = "127.0.0.1"):
+        # {app: {deployment: DeploymentReplicas}}
+        self.apps: Dict[str, Dict[str, DeploymentReplicas]] = {}
+        self.ingress: Dict[str, str] = {}  # app -> ingress deployment name
+        self.route_prefixes: Dict[str, str] = {}  # app -> route prefix
+        self.http_port = http_port
+        self.http_host = http_host
+        self._proxy = None
+        self._bg = None
+        self._shutdown = False
+
+    def _ensure_bg(self):
+        if self._bg is None:
+            loop = asyncio.get_event_loop()
+            self._bg = loop.create_task(self._reconcile_loop())
+
+    # ------------------------------------------------------------- deploy
+
+    async def deploy_application(self, name: str, route_prefix: str,
+                                 deployments: List[dict], ingress: str):
+        """deployments: [{name, callable_bytes, init_args, init_kwargs,
+        num_replicas, max_ongoing_requests, ray_actor_options,
+        autoscaling_config, user_config}]"""
+        self._ensure_bg()
+        app = self.apps.setdefault(name, {})
+        wanted = {d["name"] for d in deployments}
+        for dep_name in list(app):
+            if dep_name not in wanted:
+                await self._scale_to(app[dep_name], 0, name, dep_name)
+                del app[dep_name]
+        for d in deployments:
+            dr = app.setdefault(d["name"], DeploymentReplicas())
+            dr.cfg = d
+            auto = AutoscalingConfig.coerce(d.get("autoscaling_config"))
+            dr.cfg["autoscaling_config"] = auto
+            if auto is not None:
+                target = max(auto.min_replicas, min(auto.max_replicas,
+                                                    dr.target or auto.min_replicas))
+            else:
+                target = d.get("num_replicas", 1)
+            await self._scale_to(dr, target, name, d["name"])
+        self.ingress[name] = ingress
+        self.route_prefixes[name] = route_prefix
+        return True
+
+    async def delete_application(self, name: str):
+        app = self.apps.pop(name, {})
+        for dep_name, dr in app.items():
+            await self._scale_to(dr, 0, name, dep_name)
+        self.ingress.pop(name, None)
+        self.route_prefixes.pop(name, None)
+        return True
+
+    async def _scale_to(self, dr: DeploymentReplicas, target: int,
+                        app: str, dep_name: str):
+        dr.target = target
+        d = dr.cfg
+        while len(dr.replicas) < target:
+            opts = dict(d.get("ray_actor_options") or {})
+            opts.setdefault("num_cpus", 1)
+            opts["max_concurrency"] = max(d.get("max_ongoing_requests", 100), 8)
+            ReplicaCls = ray.remote(Replica)
+            h = ReplicaCls.options(**opts).remote(
+                d["callable_bytes"], d.get("init_args") or (),
+                d.get("init_kwargs") or {}, d.get("user_config"),
+            )
+            dr.replicas.append(h)
+            logger.info("started replica %d of %s/%s", len(dr.replicas), app,
+                        dep_name)
+        while len(dr.replicas) > target:
+            h = dr.replicas.pop()
+            try:
+                ray.kill(h)
+            except Exception:
+                pass
+        # wait until new replicas construct (first health check)
+        if dr.replicas:
+            await asyncio.gather(*[
+                _aw(h.check_health.remote()) for h in dr.replicas
+            ], return_exceptions=True)
+
+    # ------------------------------------------------------------ queries
+
+    async def get_replicas(self, app: str, deployment: str):
+        dr = self.apps.get(app, {}).get(deployment)
+        if dr is None:
+            return None
+        return list(dr.replicas)
+
+    async def get_deployment_info(self, app: str, deployment: str):
+        dr = self.apps.get(app, {}).get(deployment)
+        if dr is None:
+            return None
+        return {"target": dr.target, "num_replicas": len(dr.replicas),
+                "max_ongoing_requests": dr.cfg.get("max_ongoing_requests")}
+
+    async def get_app_config(self, app: str):
+        if app not in self.apps:
+            return None
+        return {"ingress": self.ingress.get(app),
+                "route_prefix": self.route_prefixes.get(app),
+                "deployments": list(self.apps[app])}
+
+    async def list_applications(self):
+        return {
+            app: {
+                "route_prefix": self.route_prefixes.get(app),
+                "ingress": self.ingress.get(app),
+                "deployments": {
+                    name: {"replicas": len(dr.replicas), "target": dr.target}
+                    for name, dr in deps.items()
+                },
+            }
+            for app, deps in self.apps.items()
+        }
+
+    async def graceful_shutdown(self):
+        self._shutdown = True
+        for name in list(self.apps):
+            await self.delete_application(name)
+        return True
+
+    # -------------------------------------------------- reconcile/autoscale
+
+    async def _reconcile_loop(self):
+        while not self._shutdown:
+            try:
+                await self._reconcile_once()
+            except Exception:
+                logger.exception("reconcile failed")
+            await asyncio.sleep(1.0)
+
+    async def _reconcile_once(self):
+        for app, deps in self.apps.items():
+            for dep_name, dr in deps.items():
+                # replace dead replicas
+                alive = []
+                for h in dr.replicas:
+                    try:
+                        await _aw(h.num_ongoing_requests.remote())
+                        alive.append(h)
+                    except Exception:
+                        logger.warning("replica of %s/%s died; replacing",
+                                       app, dep_name)
+                dr.replicas = alive
+                if len(dr.replicas) < dr.target:
+                    await self._scale_to(dr, dr.target, app, dep_name)
+                auto: Optional[AutoscalingConfig] = dr.cfg.get("autoscaling_config")
+                if auto is None or not dr.replicas:
+                    continue
+                counts = await asyncio.gather(*[
+                    _aw(h.num_ongoing_requests.remote()) for h in dr.replicas
+                ], return_exceptions=True)
+                total = sum(c for c in counts if isinstance(c, int))
+                desired = _calculate_desired_num_replicas(auto, total,
+                                                          len(dr.replicas))
+                now = time.monotonic()
+                if desired > dr.target and now - dr.last_scale_up > auto.upscale_delay_s:
+                    dr.last_scale_up = now
+                    await self._scale_to(dr, desired, app, dep_name)
+                elif (desired < dr.target
+                      and now - dr.last_scale_down > auto.downscale_delay_s):
+                    dr.last_scale_down = now
+                    await self._scale_to(dr, desired, app, dep_name)
+
+
+async def _aw(ref):
+    """Await an ObjectRef inside the controller's event loop."""
+    return await ref

@@ -1,0 +1,76 @@
+"""Router + DeploymentHandle plumbing.
+
+Role parity: reference python/ray/serve/_private/router.py:496
+(AsyncioRouter) and request_router/pow_2_router.py:52 — power-of-two-choices
+replica selection on queue length. The replica list is pulled from the
+controller and cached (refreshed periodically or when every replica looks
+dead), instead of the reference's long-poll push.
+"""
+from __future__ import annotations
+
+import random
+import time
+from typing import Any, Dict, List, Optional
+
+from ant_ray_amd.serve._private.common import CONTROLLER_NAME
+
+_REFRESH_S = 5.0
+
+
+class Router:
+    def __init__(self, app: str, deployment: str):
+        self.app = app
+        self.deployment = deployment
+        self._replicas: List[Any] = []
+        self._last_refresh = 0.0
+        self._ongoing: Dict[int, int] = {}  # index -> in-flight from THIS router
+
+    def _controller(self):
+        import ant_ray_amd as ray
+
+        return ray.get_actor(CONTROLLER_NAME)
+
+    def _refresh(self, force=False):
+        now = time.monotonic()
+        if not force and self._replicas and now - self._last_refresh < _REFRESH_S:
+            return
+        import ant_ray_amd as ray
+
+        replicas = ray.get(
+            self._controller().get_replicas.remote(self.app, self.deployment)
+        )
+        if replicas is None:
+            raise RuntimeError(
+                f"deployment '{self.deployment}' not found in app '{self.app}'"
+            )
+        self._replicas = replicas
+        self._ongoing = {i: self._ongoing.get(i, 0) for i in range(len(replicas))}
+        self._last_refresh = now
+
+    def choose_replica(self):
+        """Power-of-two-choices on locally tracked in-flight counts."""
+        self._refresh()
+        n = len(self._replicas)
+        if n == 0:
+            self._refresh(force=True)
+            n = len(self._replicas)
+            if n == 0:
+                raise RuntimeError(
+                    f"no replicas for {self.app}/{self.deployment}")
+        if n == 1:
+            i = 0
+        else:
+            a, b = random.sample(range(n), 2)
+            i = a if self._ongoing.get(a, 0) <= self._ongoing.get(b, 0) else b
+        return i, self._replicas[i]
+
+    def submit(self, method_name: Optional[str], args, kwargs):
+        """Returns (ObjectRef, replica_index). The DeploymentResponse calls
+        complete(index) when the result is consumed, closing the in-flight
+        accounting the pow-2 choice reads."""
+        i, replica = self.choose_replica()
+        self._ongoing[i] = self._ongoing.get(i, 0) + 1
+        return replica.handle_request.remote(method_name, args, kwargs), i
+
+    def complete(self, i: int):
+        self._ongoing[i] = max(0, self._ongoing.get(i, 1) - 1)

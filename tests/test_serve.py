@@ -1,0 +1,162 @@
+"""Ray Serve parity tests: deployments, handles, composition, HTTP proxy,
+batching, autoscaling policy math, status/delete."""
+import asyncio
+import time
+
+import pytest
+
+
+@pytest.fixture(scope="module")
+def serve_mod():
+    import ant_ray_amd as ray
+    from ant_ray_amd import serve
+
+    if not ray.is_initialized():
+        ray.init(num_cpus=8)
+    yield serve
+    serve.shutdown()
+    ray.shutdown()
+
+
+def test_deploy_and_handle_call(serve_mod):
+    serve = serve_mod
+
+    @serve.deployment(num_replicas=2)
+    class Doubler:
+        def __call__(self, x):
+            return 2 * x
+
+        def triple(self, x):
+            return 3 * x
+
+    h = serve.run(Doubler.bind(), name="doubler", route_prefix="/doubler")
+    assert h.remote(21).result(timeout_s=30) == 42
+    assert h.triple.remote(5).result(timeout_s=30) == 15
+    # load-balances over 2 replicas without error
+    assert [h.remote(i).result() for i in range(10)] == [2 * i for i in range(10)]
+
+
+def test_function_deployment_and_composition(serve_mod):
+    serve = serve_mod
+
+    @serve.deployment
+    def magnitude(x):
+        return abs(x)
+
+    @serve.deployment
+    class Pipeline:
+        def __init__(self, mag_handle):
+            self.mag = mag_handle
+
+        async def __call__(self, x):
+            return await self.mag.remote(x) + 1
+
+    h = serve.run(Pipeline.bind(magnitude.bind()), name="pipe",
+                  route_prefix="/pipe")
+    assert h.remote(-41).result(timeout_s=30) == 42
+
+
+def test_http_proxy(serve_mod):
+    serve = serve_mod
+    import ant_ray_amd as ray
+
+    @serve.deployment
+    class Echo:
+        async def __call__(self, request):
+            body = await request.body()
+            return {"path": request.url.path, "body": body.decode(),
+                    "method": request.method}
+
+    serve.run(Echo.bind(), name="echo", route_prefix="/echo")
+    port = ray.get(ray.get_actor("SERVE_PROXY_ACTOR").ready.remote())
+    import urllib.request
+
+    req = urllib.request.Request(f"http://127.0.0.1:{port}/echo/sub?q=1",
+                                 data=b"hello", method="POST")
+    with urllib.request.urlopen(req, timeout=30) as resp:
+        import json
+
+        out = json.loads(resp.read())
+    assert out == {"path": "/sub", "body": "hello", "method": "POST"}
+    # healthz + routes endpoints
+    with urllib.request.urlopen(f"http://127.0.0.1:{port}/-/healthz",
+                                timeout=10) as r:
+        assert r.read() == b"ok"
+
+
+def test_starlette_response_passthrough(serve_mod):
+    serve = serve_mod
+    import ant_ray_amd as ray
+
+    @serve.deployment
+    class Custom:
+        def __call__(self, request):
+            from starlette.responses import PlainTextResponse
+
+            return PlainTextResponse("teapot", status_code=418)
+
+    serve.run(Custom.bind(), name="custom", route_prefix="/custom")
+    port = ray.get(ray.get_actor("SERVE_PROXY_ACTOR").ready.remote())
+    import urllib.error
+    import urllib.request
+
+    try:
+        urllib.request.urlopen(f"http://127.0.0.1:{port}/custom", timeout=30)
+        assert False, "expected 418"
+    except urllib.error.HTTPError as e:
+        assert e.code == 418
+        assert e.read() == b"teapot"
+
+
+def test_batching(serve_mod):
+    serve = serve_mod
+
+    @serve.deployment
+    class Batched:
+        def __init__(self):
+            self.batch_sizes = []
+
+        @serve.batch(max_batch_size=8, batch_wait_timeout_s=0.2)
+        async def __call__(self, items):
+            self.batch_sizes.append(len(items))
+            return [i * 10 for i in items]
+
+        def get_sizes(self):
+            return self.batch_sizes
+
+    h = serve.run(Batched.bind(), name="batched", route_prefix="/batched")
+    responses = [h.remote(i) for i in range(8)]
+    results = [r.result(timeout_s=30) for r in responses]
+    assert sorted(results) == [i * 10 for i in range(8)]
+    sizes = h.get_sizes.remote().result(timeout_s=30)
+    assert max(sizes) > 1  # at least some calls were actually batched
+
+
+def test_status_and_delete(serve_mod):
+    serve = serve_mod
+
+    @serve.deployment
+    def f():
+        return "ok"
+
+    serve.run(f.bind(), name="tmp_app", route_prefix="/tmp")
+    st = serve.status()["applications"]
+    assert "tmp_app" in st
+    assert st["tmp_app"]["deployments"]["f"]["replicas"] == 1
+    serve.delete("tmp_app")
+    st = serve.status()["applications"]
+    assert "tmp_app" not in st
+
+
+def test_autoscaling_policy_math():
+    from ant_ray_amd.serve._private.common import AutoscalingConfig
+    from ant_ray_amd.serve._private.controller import (
+        _calculate_desired_num_replicas,
+    )
+
+    cfg = AutoscalingConfig(min_replicas=1, max_replicas=10,
+                            target_ongoing_requests=2.0)
+    assert _calculate_desired_num_replicas(cfg, total_ongoing=0, current=2) == 1
+    assert _calculate_desired_num_replicas(cfg, total_ongoing=8, current=2) == 4
+    assert _calculate_desired_num_replicas(cfg, total_ongoing=100, current=2) == 10
+    assert _calculate_desired_num_replicas(cfg, total_ongoing=3, current=2) == 2
