@@ -1,0 +1,255 @@
+"""ant_ray_amd.tune — Ray Tune parity: hyperparameter search over trials.
+
+Role parity: reference python/ray/tune/ (~58k LoC): Tuner
+(tune/tune.py), trial execution (execution/tune_controller.py) on actors,
+search (search/), schedulers (schedulers/), ResultGrid. Trials run as
+actors (one per trial, resources via with_resources); tune.report shares
+the Train session so a function trainable works under both.
+"""
+from __future__ import annotations
+
+import os
+import time
+import traceback
+from dataclasses import dataclass
+from typing import Any, Callable, Dict, List, Optional
+
+from ant_ray_amd.train._checkpoint import Checkpoint
+from ant_ray_amd.train.config import Result, RunConfig
+from ant_ray_amd.train.session import get_checkpoint  # noqa: F401
+from ant_ray_amd.train.session import report  # noqa: F401 (tune.report parity)
+from ant_ray_amd.tune.schedulers import (  # noqa: F401
+    CONTINUE,
+    STOP,
+    ASHAScheduler,
+    FIFOScheduler,
+    MedianStoppingRule,
+)
+from ant_ray_amd.tune.search import (  # noqa: F401
+    BasicVariantGenerator,
+    choice,
+    grid_search,
+    loguniform,
+    randint,
+    uniform,
+)
+
+__all__ = [
+    "ASHAScheduler", "BasicVariantGenerator", "FIFOScheduler",
+    "MedianStoppingRule", "ResultGrid", "TuneConfig", "Tuner", "choice",
+    "grid_search", "loguniform", "randint", "report", "run", "uniform",
+    "with_parameters", "with_resources",
+]
+
+
+@dataclass
+class TuneConfig:
+    metric: Optional[str] = None
+    mode: Optional[str] = "max"
+    num_samples: int = 1
+    max_concurrent_trials: Optional[int] = None
+    scheduler: Optional[Any] = None
+    search_alg: Optional[Any] = None
+    seed: Optional[int] = None
+
+
+class ResultGrid:
+    def __init__(self, results: List[Result]):
+        self._results = results
+
+    def __len__(self):
+        return len(self._results)
+
+    def __getitem__(self, i) -> Result:
+        return self._results[i]
+
+    def __iter__(self):
+        return iter(self._results)
+
+    @property
+    def errors(self):
+        return [r.error for r in self._results if r.error]
+
+    def get_best_result(self, metric: Optional[str] = None,
+                        mode: str = "max") -> Result:
+        scored = [r for r in self._results
+                  if r.metrics and r.metrics.get(metric) is not None]
+        if not scored:
+            raise RuntimeError(f"no trial reported metric {metric!r}")
+        return (max if mode == "max" else min)(
+            scored, key=lambda r: r.metrics[metric])
+
+    def get_dataframe(self):
+        import pandas as pd
+
+        return pd.DataFrame([r.metrics or {} for r in self._results])
+
+
+def with_parameters(fn: Callable, **params):
+    """Bind large objects by value (reference puts them in the object store;
+    our serializer ships them with the trainable)."""
+    import functools
+
+    @functools.wraps(fn)
+    def inner(config):
+        return fn(config, **params)
+
+    return inner
+
+
+def with_resources(fn: Callable, resources: Dict[str, float]):
+    fn = _copy_fn(fn)
+    fn._tune_resources = dict(resources)
+    return fn
+
+
+def _copy_fn(fn):
+    import functools
+
+    @functools.wraps(fn)
+    def inner(config):
+        return fn(config)
+
+    return inner
+
+
+class _TrialRunner:
+    """Actor hosting one trial's function trainable on a thread."""
+
+    def __init__(self):
+        import queue
+        import threading
+
+        self._q = queue.Queue()
+        self._thread = None
+        self._error = None
+        self._done = False
+        self._threading = threading
+
+    def start(self, fn, config, trial_id: str, experiment_path: str):
+        from ant_ray_amd.train.session import TrainContext, set_train_context
+
+        ctx = TrainContext(experiment_name=trial_id,
+                           experiment_path=experiment_path)
+        ctx.report_queue = self._q
+
+        def run():
+            set_train_context(ctx)
+            try:
+                fn(config)
+            except BaseException:
+                self._error = traceback.format_exc()
+            finally:
+                self._done = True
+
+        self._thread = self._threading.Thread(target=run, daemon=True)
+        self._thread.start()
+        return True
+
+    def poll(self):
+        reports = []
+        while True:
+            try:
+                reports.append(self._q.get_nowait())
+            except Exception:
+                break
+        status = ("errored" if self._error else
+                  "finished" if self._done else "running")
+        return {"status": status, "reports": reports, "error": self._error}
+
+
+class Tuner:
+    def __init__(self, trainable: Callable, *, param_space: Optional[dict] = None,
+                 tune_config: Optional[TuneConfig] = None,
+                 run_config: Optional[RunConfig] = None):
+        self.trainable = trainable
+        self.param_space = param_space or {}
+        self.tune_config = tune_config or TuneConfig()
+        self.run_config = run_config or RunConfig()
+
+    def fit(self) -> ResultGrid:
+        import ant_ray_amd as ray
+
+        if not ray.is_initialized():
+            ray.init()
+        tc = self.tune_config
+        variants = BasicVariantGenerator(
+            self.param_space, tc.num_samples, seed=tc.seed).variants()
+        scheduler = tc.scheduler or FIFOScheduler()
+        scheduler.set_objective(tc.metric, tc.mode or "max")
+        name = self.run_config.name or f"tune_{int(time.time())}"
+        exp_path = os.path.join(self.run_config.storage_path, name)
+        os.makedirs(exp_path, exist_ok=True)
+        resources = getattr(self.trainable, "_tune_resources", {"CPU": 1})
+        max_conc = tc.max_concurrent_trials or 8
+
+        RunnerCls = ray.remote(_TrialRunner)
+        opts = {"num_cpus": resources.get("CPU", 1), "max_concurrency": 2}
+        if resources.get("GPU"):
+            opts["num_gpus"] = resources["GPU"]
+
+        pending = list(enumerate(variants))
+        running: Dict[int, dict] = {}
+        results: Dict[int, Result] = {}
+
+        def launch(idx, config):
+            trial_id = f"{name}_{idx:05d}"
+            trial_path = os.path.join(exp_path, trial_id)
+            os.makedirs(trial_path, exist_ok=True)
+            actor = RunnerCls.options(**opts).remote()
+            ray.get(actor.start.remote(self.trainable, config, trial_id,
+                                       trial_path))
+            running[idx] = {"actor": actor, "config": config,
+                            "trial_id": trial_id, "path": trial_path,
+                            "iter": 0, "last_metrics": None, "ckpt": None}
+
+        while pending or running:
+            while pending and len(running) < max_conc:
+                idx, cfg = pending.pop(0)
+                launch(idx, cfg)
+            polls = ray.get([t["actor"].poll.remote()
+                             for t in running.values()])
+            finished_idx = []
+            for (idx, t), st in zip(list(running.items()), polls):
+                stop = False
+                for rep in st["reports"]:
+                    t["iter"] += 1
+                    metrics = dict(rep["metrics"])
+                    metrics.setdefault("training_iteration", t["iter"])
+                    metrics["config"] = t["config"]
+                    t["last_metrics"] = metrics
+                    if rep.get("checkpoint_path"):
+                        t["ckpt"] = rep["checkpoint_path"]
+                    if scheduler.on_trial_result(t["trial_id"], metrics) == STOP:
+                        stop = True
+                if st["status"] in ("finished", "errored") or stop:
+                    err = None
+                    if st["status"] == "errored":
+                        err = RuntimeError(st["error"])
+                    results[idx] = Result(
+                        metrics=t["last_metrics"],
+                        checkpoint=Checkpoint(t["ckpt"]) if t["ckpt"] else None,
+                        path=t["path"], error=err,
+                    )
+                    try:
+                        ray.kill(t["actor"])
+                    except Exception:
+                        pass
+                    finished_idx.append(idx)
+            for idx in finished_idx:
+                del running[idx]
+            if running:
+                time.sleep(0.1)
+        return ResultGrid([results[i] for i in sorted(results)])
+
+
+def run(trainable, *, config: Optional[dict] = None, num_samples: int = 1,
+        metric: Optional[str] = None, mode: str = "max", scheduler=None,
+        **kwargs) -> ResultGrid:
+    """Legacy tune.run API (parity tune/tune.py:run)."""
+    tuner = Tuner(
+        trainable, param_space=config or {},
+        tune_config=TuneConfig(metric=metric, mode=mode,
+                               num_samples=num_samples, scheduler=scheduler),
+    )
+    return tuner.fit()

@@ -1,0 +1,112 @@
+"""Ray Tune parity tests: grid/random search, ResultGrid, ASHA early
+stopping, checkpoint reporting."""
+import pytest
+
+
+@pytest.fixture(scope="module")
+def ray_mod():
+    import ant_ray_amd as ray
+
+    if not ray.is_initialized():
+        ray.init(num_cpus=8)
+    yield ray
+    ray.shutdown()
+
+
+def test_grid_and_random_search(ray_mod, tmp_path_factory):
+    from ant_ray_amd import tune
+    from ant_ray_amd.train import RunConfig
+
+    def objective(config):
+        score = -((config["x"] - 3) ** 2) + config["b"]
+        tune.report({"score": score})
+
+    tuner = tune.Tuner(
+        objective,
+        param_space={"x": tune.grid_search([1, 2, 3, 4]),
+                     "b": tune.choice([10])},
+        tune_config=tune.TuneConfig(metric="score", mode="max"),
+        run_config=RunConfig(name="grid",
+                             storage_path=str(tmp_path_factory.mktemp("t"))),
+    )
+    grid = tuner.fit()
+    assert len(grid) == 4
+    best = grid.get_best_result(metric="score", mode="max")
+    assert best.metrics["config"]["x"] == 3
+    assert best.metrics["score"] == 10
+
+
+def test_num_samples_random(ray_mod):
+    from ant_ray_amd import tune
+
+    def objective(config):
+        tune.report({"v": config["u"]})
+
+    grid = tune.Tuner(
+        objective,
+        param_space={"u": tune.uniform(0, 1)},
+        tune_config=tune.TuneConfig(num_samples=5, metric="v", seed=42),
+    ).fit()
+    vals = [r.metrics["v"] for r in grid]
+    assert len(vals) == 5 and len(set(vals)) == 5
+    assert all(0 <= v <= 1 for v in vals)
+
+
+def test_asha_stops_bad_trials(ray_mod):
+    from ant_ray_amd import tune
+
+    def objective(config):
+        import time
+
+        for i in range(20):
+            time.sleep(0.05)
+            tune.report({"acc": config["q"] * (i + 1)})
+
+    # strong trials run first (concurrency 2) and populate the rungs, so the
+    # weak trials get compared -- and stopped -- at their first milestones
+    grid = tune.Tuner(
+        objective,
+        param_space={"q": tune.grid_search([2.0, 1.0, 0.01, 0.02])},
+        tune_config=tune.TuneConfig(
+            metric="acc", mode="max", max_concurrent_trials=2,
+            scheduler=tune.ASHAScheduler(max_t=20, grace_period=2,
+                                         reduction_factor=2),
+        ),
+    ).fit()
+    best = grid.get_best_result(metric="acc", mode="max")
+    assert best.metrics["config"]["q"] == 2.0
+    # at least one weak trial was stopped before max_t
+    iters = [r.metrics["training_iteration"] for r in grid]
+    assert min(iters) < 20
+    assert max(iters) == 20
+
+
+def test_checkpoint_and_errors(ray_mod, tmp_path_factory):
+    from ant_ray_amd import tune
+    from ant_ray_amd.train import Checkpoint, RunConfig
+
+    def objective(config):
+        import os
+        import tempfile
+
+        if config["x"] == 13:
+            raise ValueError("unlucky")
+        with tempfile.TemporaryDirectory() as d:
+            with open(os.path.join(d, "state.txt"), "w") as f:
+                f.write(str(config["x"]))
+            tune.report({"x2": config["x"] ** 2},
+                        checkpoint=Checkpoint.from_directory(d))
+
+    grid = tune.Tuner(
+        objective,
+        param_space={"x": tune.grid_search([2, 13])},
+        tune_config=tune.TuneConfig(metric="x2", mode="max"),
+        run_config=RunConfig(name="ck",
+                             storage_path=str(tmp_path_factory.mktemp("t"))),
+    ).fit()
+    assert len(grid.errors) == 1
+    ok = [r for r in grid if r.error is None][0]
+    import os
+
+    with open(os.path.join(ok.checkpoint.path, "state.txt")) as f:
+        assert f.read() == "2"
