@@ -47,6 +47,14 @@ class ActorMethod:
     def remote(self, *args, **kwargs):
         return self._handle._actor_method_call(self._name, args, kwargs, self._opts)
 
+    def bind(self, *args, **kwargs):
+        """DAG node calling this method on the EXISTING actor (parity:
+        actor method .bind in python/ray/dag)."""
+        from ant_ray_amd.dag.node import ClassMethodNode, _ExistingActorShim
+
+        return ClassMethodNode(_ExistingActorShim(self._handle), self._name,
+                               args, kwargs)
+
     def __call__(self, *args, **kwargs):
         raise TypeError(
             f"Actor method '{self._name}' cannot be called directly; use "
