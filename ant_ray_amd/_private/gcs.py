@@ -490,6 +490,22 @@ class GcsServer:
     async def rpc_list_actors(self, conn, p):
         return [a.view() for a in self.actors.values()]
 
+    # --------------------------------------------------------- task events
+    # parity: GcsTaskManager (src/ray/gcs/gcs_task_manager.h) fed by the
+    # CoreWorker TaskEventBuffer; ring buffer, newest wins
+    async def rpc_task_events(self, conn, p):
+        import collections
+
+        if not hasattr(self, "task_events"):
+            self.task_events = collections.deque(maxlen=20000)
+        self.task_events.extend(p["events"])
+        return {"ok": True}
+
+    async def rpc_list_task_events(self, conn, p):
+        limit = p.get("limit", 1000)
+        evs = list(getattr(self, "task_events", []))
+        return evs[-limit:]
+
     async def rpc_list_workers(self, conn, p):
         return list(self.workers.values())
 

@@ -145,9 +145,30 @@ class _Pickler(cloudpickle.CloudPickler):
                 r = _torch_cpu_tensor_reducer(obj)
                 if r is not NotImplemented:
                     return r
+            # torch.ops / torch.ops.aten are pseudo-modules (_Ops/_OpNamespace)
+            # cloudpickle's submodule heuristic drags in whenever a function
+            # names both "torch" and "ops"; pickle them by import reference
+            import torch._ops as _t_ops
+
+            if isinstance(obj, _t_ops._Ops):
+                return (_reimport_torch_ops, ())
+            if isinstance(obj, _t_ops._OpNamespace):
+                return (_reimport_torch_op_namespace, (obj.name,))
         except ImportError:
             pass
         return super().reducer_override(obj)
+
+
+def _reimport_torch_ops():
+    import torch
+
+    return torch.ops
+
+
+def _reimport_torch_op_namespace(name):
+    import torch
+
+    return getattr(torch.ops, name)
 
 
 _BUFFER_THRESHOLD = 512  # buffers below this get pickled in-band
@@ -220,6 +241,17 @@ def _make_dispatch(contained_refs):
     except ImportError:
         pass
     return table
+
+
+def dumps_by_value(obj: Any) -> bytes:
+    """cloudpickle-by-value with our reducers (torch.ops workaround, refs);
+    for shipping user callables outside the task-arg path (e.g. Serve
+    deployment bodies)."""
+    import io
+
+    f = io.BytesIO()
+    _Pickler(f).dump(obj)
+    return f.getvalue()
 
 
 def serialize(value: Any, metadata: bytes = META_PICKLE) -> SerializedObject:

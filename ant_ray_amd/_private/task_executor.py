@@ -22,6 +22,7 @@ import logging
 import os
 import queue
 import threading
+import time
 from typing import Any, Dict, Optional
 
 from ant_ray_amd._private import serialization
@@ -180,10 +181,46 @@ class TaskExecutor:
             reply = self._error_reply(payload, e)
         done(reply)
 
+    def _record_event(self, payload, t0, t1, ok: bool):
+        """Task-event pipeline (parity: core_worker/task_event_buffer.cc ->
+        GcsTaskManager). Fire-and-forget batched notify to the GCS."""
+        try:
+            gcs = getattr(self.cw, "gcs", None)
+            if gcs is None:
+                return
+            name = payload.get("name") or payload.get("method") or ""
+            if not name and payload.get("fn_id"):
+                fn = self._fn_cache.get(payload["fn_id"])
+                name = getattr(fn, "__name__", "")
+            buf = getattr(self, "_event_buf", None)
+            if buf is None:
+                buf = self._event_buf = []
+                self._event_flush_t = 0.0
+            buf.append({
+                "task_id": payload["task_id"].hex(),
+                "type": payload["type"],
+                "name": name,
+                "actor_id": (self.actor_id.hex() if self.actor_id else None),
+                "state": "FINISHED" if ok else "FAILED",
+                "start_ts": t0,
+                "end_ts": t1,
+                "pid": os.getpid(),
+                "worker_id": getattr(self.cw, "worker_id", b"").hex()
+                if isinstance(getattr(self.cw, "worker_id", None), bytes) else "",
+            })
+            now = time.monotonic()
+            if len(buf) >= 100 or now - self._event_flush_t > 1.0:
+                self._event_flush_t = now
+                events, self._event_buf = buf, []
+                self.cw.io.submit(gcs.notify("task_events", {"events": events}))
+        except Exception:
+            pass
+
     def _execute(self, payload) -> dict:
         ttype = payload["type"]
         task_id = payload["task_id"]
         self.cw.current_task_id = task_id
+        _t0 = self._last_t0 = time.time()
         try:
             if ttype == "normal":
                 fn = self._resolve_fn(payload)
@@ -226,6 +263,7 @@ class TaskExecutor:
                 return self._reply_results(payload, result)
             raise ValueError(f"unknown task type {ttype}")
         except BaseException as e:  # noqa: BLE001
+            self._record_event(payload, _t0, time.time(), ok=False)
             return self._error_reply(payload, e)
         finally:
             self.cw.current_task_id = None
@@ -248,6 +286,8 @@ class TaskExecutor:
 
     # ---------------------------------------------------------------- replies
     def _reply_results(self, payload, result) -> dict:
+        self._record_event(payload, getattr(self, '_last_t0', time.time()),
+                           time.time(), ok=True)
         n = payload.get("n_returns", 1)
         if n == 0:
             return {"status": "ok", "results": []}

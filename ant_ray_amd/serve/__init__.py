@@ -188,7 +188,7 @@ def start(detached: bool = True, http_options: Optional[dict] = None, **_):
 def _collect_deployments(app: Application, out: Dict[str, dict],
                          app_name: str):
     """Topological flatten: nested bound apps become handles."""
-    import cloudpickle
+    from ant_ray_amd._private.serialization import dumps_by_value as _dumps
 
     def resolve(v):
         if isinstance(v, Application):
@@ -204,7 +204,7 @@ def _collect_deployments(app: Application, out: Dict[str, dict],
         auto = auto.__dict__
     out[d.name] = {
         "name": d.name,
-        "callable_bytes": cloudpickle.dumps(d._target),
+        "callable_bytes": _dumps(d._target),
         "init_args": init_args,
         "init_kwargs": init_kwargs,
         "num_replicas": d.num_replicas,

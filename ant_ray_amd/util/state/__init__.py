@@ -1,0 +1,139 @@
+"""State API: programmatic cluster introspection.
+
+Role parity: reference python/ray/util/state/ (list_actors/list_tasks/
+list_nodes/... against GCS + agents; state_head.py aggregator). Here every
+query goes straight to the GCS over the connected worker's channel.
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional
+
+
+def _gcs_call(method: str, payload: Optional[dict] = None, timeout: float = 30):
+    import ant_ray_amd as ray
+    from ant_ray_amd._private.worker import global_worker
+
+    if not ray.is_initialized():
+        ray.init(ignore_reinit_error=True)
+    cw = global_worker.core_worker
+    return cw.io.run(cw.gcs.call(method, payload or {}, timeout=timeout),
+                     timeout=timeout + 5)
+
+
+def list_actors(filters=None, limit: int = 1000, **_) -> List[Dict[str, Any]]:
+    out = []
+    for a in _gcs_call("list_actors"):
+        row = {
+            "actor_id": a["actor_id"].hex() if isinstance(a.get("actor_id"), bytes) else a.get("actor_id"),
+            "state": a.get("state"),
+            "name": a.get("name") or "",
+            "class_name": a.get("class_name") or "",
+            "pid": a.get("pid"),
+            "node_id": a.get("node_id").hex() if isinstance(a.get("node_id"), bytes) else a.get("node_id"),
+        }
+        if _match(row, filters):
+            out.append(row)
+    return out[:limit]
+
+
+def list_tasks(filters=None, limit: int = 1000, **_) -> List[Dict[str, Any]]:
+    evs = _gcs_call("list_task_events", {"limit": limit})
+    out = [e for e in evs if _match(e, filters)]
+    return out[:limit]
+
+
+def list_nodes(filters=None, limit: int = 1000, **_) -> List[Dict[str, Any]]:
+    out = []
+    for n in _gcs_call("node_table"):
+        row = dict(n)
+        if isinstance(row.get("node_id"), bytes):
+            row["node_id"] = row["node_id"].hex()
+        if _match(row, filters):
+            out.append(row)
+    return out[:limit]
+
+
+def list_workers(filters=None, limit: int = 1000, **_) -> List[Dict[str, Any]]:
+    rows = [_hexify(w) for w in _gcs_call("list_workers")]
+    return [r for r in rows if _match(r, filters)][:limit]
+
+
+def list_jobs(filters=None, limit: int = 1000, **_) -> List[Dict[str, Any]]:
+    rows = [_hexify(j) for j in _gcs_call("list_jobs")]
+    return [r for r in rows if _match(r, filters)][:limit]
+
+
+def list_placement_groups(filters=None, limit: int = 1000, **_):
+    rows = []
+    for pg in _gcs_call("list_placement_groups"):
+        rows.append({
+            "placement_group_id": pg["pg_id"].hex(),
+            "state": pg.get("state"),
+            "strategy": pg.get("strategy"),
+            "bundles": pg.get("bundles"),
+            "name": pg.get("name", ""),
+        })
+    return [r for r in rows if _match(r, filters)][:limit]
+
+
+def list_objects(filters=None, limit: int = 1000, **_):
+    """Summarized view: per-node object-store stats (the reference lists
+    per-object entries from all core workers; our shm store keeps only
+    aggregate counters)."""
+    out = []
+    for n in _gcs_call("node_table"):
+        out.append({
+            "node_id": n["node_id"].hex() if isinstance(n.get("node_id"), bytes) else n.get("node_id"),
+            "object_store_memory_total": n.get("resources_total", {}).get("object_store_memory"),
+            "object_store_memory_available": n.get("resources_available", {}).get("object_store_memory"),
+        })
+    return out[:limit]
+
+
+def summarize_tasks():
+    from collections import Counter
+
+    c = Counter((e.get("name"), e.get("state")) for e in list_tasks(limit=20000))
+    return [{"name": k[0], "state": k[1], "count": v} for k, v in c.items()]
+
+
+def _hexify(d: dict) -> dict:
+    return {k: (v.hex() if isinstance(v, bytes) else v) for k, v in d.items()}
+
+
+def _match(row: dict, filters) -> bool:
+    if not filters:
+        return True
+    for f in filters:
+        key, op, val = f
+        have = row.get(key)
+        if op in ("=", "=="):
+            if str(have) != str(val):
+                return False
+        elif op == "!=":
+            if str(have) == str(val):
+                return False
+    return True
+
+
+def get_timeline(limit: int = 20000) -> List[dict]:
+    """Chrome-trace events from the GCS task-event buffer
+    (parity: `ray timeline` / GcsTaskManager export)."""
+    evs = _gcs_call("list_task_events", {"limit": limit})
+    trace = []
+    for e in evs:
+        start = e.get("start_ts")
+        end = e.get("end_ts")
+        if start is None or end is None:
+            continue
+        trace.append({
+            "name": e.get("name") or e.get("type"),
+            "cat": e.get("type"),
+            "ph": "X",
+            "ts": start * 1e6,
+            "dur": max((end - start) * 1e6, 1),
+            "pid": e.get("pid"),
+            "tid": e.get("pid"),
+            "args": {"task_id": e.get("task_id"), "state": e.get("state")},
+        })
+    return trace
