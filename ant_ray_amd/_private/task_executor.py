@@ -196,6 +196,19 @@ class TaskExecutor:
             if buf is None:
                 buf = self._event_buf = []
                 self._event_flush_t = 0.0
+
+                async def _periodic_flush():
+                    while True:
+                        await asyncio.sleep(1.0)
+                        if self._event_buf:
+                            events, self._event_buf = self._event_buf, []
+                            try:
+                                await gcs.notify("task_events",
+                                                 {"events": events})
+                            except Exception:
+                                return
+
+                self.cw.io.submit(_periodic_flush())
             buf.append({
                 "task_id": payload["task_id"].hex(),
                 "type": payload["type"],

@@ -1,0 +1,235 @@
+"""`ray` CLI.
+
+Role parity: reference python/ray/scripts/scripts.py (`ray start` :728,
+`ray stop` :1290, status/memory/timeline), python/ray/util/state/state_cli.py
+(`ray list ...`), and dashboard/modules/job/cli.py (`ray job submit/...`).
+Run as `python -m ant_ray_amd.scripts.cli <cmd>` (a `ray` console script when
+pip-installed; tools/ray shim in-tree).
+"""
+from __future__ import annotations
+
+import json
+import os
+import sys
+import time
+
+import click
+
+
+@click.group()
+def cli():
+    """ant-ray-amd cluster CLI."""
+
+
+# ------------------------------------------------------------------ start/stop
+
+
+@cli.command()
+@click.option("--head", is_flag=True, help="start a head node (GCS + raylet)")
+@click.option("--address", default=None, help="GCS address to join as worker node")
+@click.option("--num-cpus", type=int, default=None)
+@click.option("--num-gpus", type=int, default=None)
+@click.option("--port", type=int, default=6379)
+@click.option("--object-store-memory", type=int, default=None)
+@click.option("--block", is_flag=True, help="stay in the foreground")
+def start(head, address, num_cpus, num_gpus, port, object_store_memory, block):
+    """Start head or worker node processes on this machine."""
+    if head:
+        from ant_ray_amd._private.node import start_head
+
+        hp = start_head(num_cpus=num_cpus, num_gpus=num_gpus, port=port,
+                        object_store_memory=object_store_memory)
+        info = hp.info
+        click.echo(f"Started head: GCS at {info['gcs_addr']}")
+        click.echo(f"session dir: {info['session_dir']}")
+        click.echo("To connect: ray.init(address="
+                   f"\"{info['gcs_addr']}\") or RAY_ADDRESS={info['gcs_addr']}")
+        if block:
+            try:
+                hp.proc.wait()
+            except KeyboardInterrupt:
+                hp.terminate()
+        else:
+            # detach: the head subprocess keeps running after the CLI exits
+            hp.proc.stdout.close()
+    elif address:
+        import subprocess
+
+        cmd = [sys.executable, "-m", "ant_ray_amd._private.raylet",
+               "--gcs", address]
+        if num_cpus is not None:
+            cmd += ["--num-cpus", str(num_cpus)]
+        if num_gpus is not None:
+            cmd += ["--num-gpus", str(num_gpus)]
+        proc = subprocess.Popen(cmd, start_new_session=True)
+        click.echo(f"Started worker raylet (pid {proc.pid}) joining {address}")
+        if block:
+            proc.wait()
+    else:
+        raise click.UsageError("pass --head or --address")
+
+
+@cli.command()
+def stop():
+    """Stop the most recent local cluster (GCS shutdown broadcast)."""
+    import ant_ray_amd as ray
+
+    try:
+        ray.init(address="auto", ignore_reinit_error=True)
+    except ConnectionError:
+        click.echo("no running cluster found")
+        return
+    from ant_ray_amd._private.worker import global_worker
+
+    cw = global_worker.core_worker
+    try:
+        cw.io.run(cw.gcs.call("shutdown", {}, timeout=5), timeout=6)
+        click.echo("cluster shutdown requested")
+    except Exception as e:
+        click.echo(f"shutdown failed: {e}")
+
+
+@cli.command()
+def status():
+    """Cluster resource overview (parity `ray status`)."""
+    import ant_ray_amd as ray
+
+    if not ray.is_initialized():
+        ray.init(address=os.environ.get("RAY_ADDRESS", "auto"),
+                 ignore_reinit_error=True)
+    total = ray.cluster_resources()
+    avail = ray.available_resources()
+    nodes = ray.nodes()
+    click.echo(f"Nodes: {len(nodes)}")
+    click.echo("Resources")
+    click.echo("  total:    " + json.dumps({k: total[k] for k in sorted(total)}))
+    click.echo("  available:" + json.dumps({k: round(avail.get(k, 0), 2) for k in sorted(total)}))
+
+
+# ----------------------------------------------------------------- state list
+
+
+@cli.command("list")
+@click.argument("kind", type=click.Choice(
+    ["actors", "tasks", "nodes", "workers", "jobs", "placement-groups",
+     "objects"]))
+@click.option("--limit", type=int, default=100)
+def list_cmd(kind, limit):
+    """List cluster state (parity `ray list ...` state CLI)."""
+    from ant_ray_amd.util import state as S
+
+    fn = {
+        "actors": S.list_actors, "tasks": S.list_tasks, "nodes": S.list_nodes,
+        "workers": S.list_workers, "jobs": S.list_jobs,
+        "placement-groups": S.list_placement_groups, "objects": S.list_objects,
+    }[kind]
+    rows = fn(limit=limit)
+    if not rows:
+        click.echo(f"(no {kind})")
+        return
+    import tabulate
+
+    cols = list(rows[0].keys())
+    click.echo(tabulate.tabulate(
+        [[str(r.get(c))[:40] for c in cols] for r in rows], headers=cols))
+
+
+@cli.command()
+@click.option("--output", "-o", default="timeline.json")
+def timeline(output):
+    """Export task events as a Chrome trace (parity `ray timeline`)."""
+    from ant_ray_amd.util.state import get_timeline
+
+    trace = get_timeline()
+    with open(output, "w") as f:
+        json.dump(trace, f)
+    click.echo(f"wrote {len(trace)} events to {output} "
+               "(open in chrome://tracing or perfetto)")
+
+
+@cli.command()
+@click.argument("kind", type=click.Choice(["tasks"]))
+def summary(kind):
+    from ant_ray_amd.util.state import summarize_tasks
+
+    for row in summarize_tasks():
+        click.echo(f"{row['name'] or '(anon)':<40} {row['state']:<10} {row['count']}")
+
+
+# ----------------------------------------------------------------------- job
+
+
+@cli.group()
+def job():
+    """Job submission (parity `ray job ...`)."""
+
+
+@job.command()
+@click.option("--address", default=None)
+@click.option("--working-dir", default=None)
+@click.option("--submission-id", default=None)
+@click.option("--no-wait", is_flag=True)
+@click.argument("entrypoint", nargs=-1, required=True)
+def submit(address, working_dir, submission_id, no_wait, entrypoint):
+    import ant_ray_amd as ray
+    from ant_ray_amd.job_submission import JobSubmissionClient
+
+    ray.init(address=address or os.environ.get("RAY_ADDRESS", "auto"),
+             ignore_reinit_error=True)
+    client = JobSubmissionClient()
+    job_id = client.submit_job(entrypoint=" ".join(entrypoint),
+                               submission_id=submission_id,
+                               runtime_env={"working_dir": working_dir}
+                               if working_dir else None)
+    click.echo(f"submitted job {job_id}")
+    if no_wait:
+        return
+    while True:
+        st = client.get_job_status(job_id)
+        if st in ("SUCCEEDED", "FAILED", "STOPPED"):
+            break
+        time.sleep(1)
+    click.echo(f"job {job_id} finished: {st}")
+    click.echo(client.get_job_logs(job_id))
+    sys.exit(0 if st == "SUCCEEDED" else 1)
+
+
+@job.command("list")
+def job_list():
+    import ant_ray_amd as ray
+    from ant_ray_amd.job_submission import JobSubmissionClient
+
+    ray.init(address=os.environ.get("RAY_ADDRESS", "auto"),
+             ignore_reinit_error=True)
+    for j in JobSubmissionClient().list_jobs():
+        click.echo(json.dumps(j))
+
+
+@job.command("status")
+@click.argument("job_id")
+def job_status(job_id):
+    import ant_ray_amd as ray
+    from ant_ray_amd.job_submission import JobSubmissionClient
+
+    ray.init(address=os.environ.get("RAY_ADDRESS", "auto"),
+             ignore_reinit_error=True)
+    click.echo(JobSubmissionClient().get_job_status(job_id))
+
+
+@job.command("logs")
+@click.argument("job_id")
+def job_logs(job_id):
+    import ant_ray_amd as ray
+    from ant_ray_amd.job_submission import JobSubmissionClient
+
+    ray.init(address=os.environ.get("RAY_ADDRESS", "auto"),
+             ignore_reinit_error=True)
+    click.echo(JobSubmissionClient().get_job_logs(job_id))
+
+
+def main():
+    cli()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,97 @@
+"""State API, CLI, and job submission tests."""
+import time
+
+import pytest
+
+
+@pytest.fixture(scope="module")
+def ray_mod():
+    import ant_ray_amd as ray
+
+    if not ray.is_initialized():
+        ray.init(num_cpus=8)
+    yield ray
+    ray.shutdown()
+
+
+def test_state_lists(ray_mod):
+    ray = ray_mod
+    from ant_ray_amd.util import state
+
+    @ray.remote
+    def noop(i):
+        return i
+
+    @ray.remote
+    class Named:
+        def ping(self):
+            return "pong"
+
+    a = Named.options(name="state_test_actor").remote()
+    assert ray.get(a.ping.remote()) == "pong"
+    ray.get([noop.remote(i) for i in range(20)])
+
+    actors = state.list_actors()
+    assert any(x["name"] == "state_test_actor" for x in actors)
+    nodes = state.list_nodes()
+    assert len(nodes) >= 1
+    # task events are flushed in batches (<=1s cadence)
+    deadline = time.time() + 10
+    tasks = []
+    while time.time() < deadline:
+        tasks = state.list_tasks(limit=5000)
+        if sum(1 for t in tasks if t.get("name") == "noop") >= 20:
+            break
+        time.sleep(0.3)
+    assert sum(1 for t in tasks if t.get("name") == "noop") >= 20
+    assert all(t["state"] in ("FINISHED", "FAILED") for t in tasks)
+
+    trace = state.get_timeline()
+    assert len(trace) >= 20
+    ev = next(e for e in trace if e["name"] == "noop")
+    assert ev["ph"] == "X" and ev["dur"] >= 1
+
+
+def test_cli_status_and_list(ray_mod):
+    from click.testing import CliRunner
+
+    from ant_ray_amd.scripts.cli import cli
+
+    r = CliRunner().invoke(cli, ["status"])
+    assert r.exit_code == 0, r.output
+    assert "Nodes: " in r.output
+    r = CliRunner().invoke(cli, ["list", "actors"])
+    assert r.exit_code == 0, r.output
+    assert "state_test_actor" in r.output
+
+
+def test_cli_timeline(ray_mod, tmp_path):
+    from click.testing import CliRunner
+
+    from ant_ray_amd.scripts.cli import cli
+
+    out = str(tmp_path / "trace.json")
+    r = CliRunner().invoke(cli, ["timeline", "-o", out])
+    assert r.exit_code == 0, r.output
+    import json
+
+    trace = json.load(open(out))
+    assert isinstance(trace, list) and trace
+
+
+def test_job_submission(ray_mod):
+    from ant_ray_amd.job_submission import JobSubmissionClient
+
+    client = JobSubmissionClient()
+    job_id = client.submit_job(
+        entrypoint="python -c \"print('hello from job 12321')\"")
+    deadline = time.time() + 60
+    while time.time() < deadline:
+        st = client.get_job_status(job_id)
+        if st in ("SUCCEEDED", "FAILED"):
+            break
+        time.sleep(0.5)
+    assert st == "SUCCEEDED"
+    assert "hello from job 12321" in client.get_job_logs(job_id)
+    jobs = client.list_jobs()
+    assert any(j["submission_id"] == job_id for j in jobs)
