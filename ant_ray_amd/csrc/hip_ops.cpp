@@ -21,6 +21,9 @@ void launch_cross_entropy(void*, const int32_t*, float*, int64_t, int64_t,
 void launch_adamw(float*, void*, const void*, float*, float*, int64_t, float,
                   float, float, float, float, float, float, float, hipStream_t);
 void launch_bf16_scale(void*, float, int64_t, hipStream_t);
+void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
+                     int, int, int, int, long, long, long, long, long, long,
+                     long, long, long, float, int, void*);
 void launch_bf16_to_f32(const void*, float*, int64_t, hipStream_t);
 void launch_f32_to_bf16(const float*, void*, int64_t, hipStream_t);
 }
@@ -185,6 +188,44 @@ void f32_to_bf16_(torch::Tensor x, torch::Tensor y) {
 
 }  // namespace
 
+
+// Flash attention forward. q: [B, Hq, S, D], k/v: [B, Hk, S, D] (strided ok
+// along the head dim so fused-qkv views work; the S,D inner block must be
+// row-contiguous). Returns (o, lse) — lse in log2 domain, empty unless
+// need_lse.
+std::tuple<torch::Tensor, torch::Tensor> attn_fwd(torch::Tensor q,
+                                                  torch::Tensor k,
+                                                  torch::Tensor v,
+                                                  double scale, bool causal,
+                                                  bool need_lse) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16, "q bf16 gpu");
+  TORCH_CHECK(q.dim() == 4 && k.dim() == 4 && v.dim() == 4, "q/k/v 4-D");
+  int B = q.size(0), Hq = q.size(1), S = q.size(2), D = q.size(3);
+  int Hk = k.size(1);
+  TORCH_CHECK(D == 128, "attn_fwd: D must be 128");
+  TORCH_CHECK(Hq % Hk == 0, "attn_fwd: Hq % Hk != 0");
+  TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1,
+              "innermost dim must be contiguous");
+  TORCH_CHECK(k.strides() == v.strides() && k.sizes() == v.sizes(),
+              "k and v must share layout");
+  auto o = torch::empty({B, Hq, S, D},
+                        q.options().memory_format(torch::MemoryFormat::Contiguous));
+  torch::Tensor lse;
+  float* lse_ptr = nullptr;
+  if (need_lse) {
+    lse = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat32));
+    lse_ptr = lse.data_ptr<float>();
+  } else {
+    lse = torch::empty({0}, q.options().dtype(torch::kFloat32));
+  }
+  launch_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                  lse_ptr, B, S, Hq, Hk, q.stride(0), q.stride(1), q.stride(2),
+                  k.stride(0), k.stride(1), k.stride(2), o.stride(0),
+                  o.stride(1), o.stride(2), (float)scale, causal ? 1 : 0,
+                  (void*)cur_stream());
+  return {o, lse};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused (add+)rmsnorm forward",
         py::arg("x"), py::arg("w"), py::arg("residual") = py::none(),
@@ -198,6 +239,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("targets"), py::arg("grad_scale") = 1.0,
         py::arg("ignore_index") = -100, py::arg("write_dlogits") = true);
   m.def("adamw_", &adamw_);
+  m.def("attn_fwd", &attn_fwd, py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("scale"), py::arg("causal") = true,
+        py::arg("need_lse") = false);
   m.def("bf16_scale_", &bf16_scale_);
   m.def("bf16_to_f32", &bf16_to_f32);
   m.def("f32_to_bf16_", &f32_to_bf16_);

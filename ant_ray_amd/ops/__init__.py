@@ -12,6 +12,7 @@ import torch
 from ant_ray_amd.ops import reference
 from ant_ray_amd.ops.functional import (  # noqa: F401
     adamw_step,
+    attention,
     fused_add_rmsnorm,
     rmsnorm,
     rope_qkv,

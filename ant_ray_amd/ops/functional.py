@@ -146,6 +146,27 @@ def swiglu(gate_up: torch.Tensor) -> torch.Tensor:
     return (torch.nn.functional.silu(g) * u).to(gate_up.dtype)
 
 
+# -------------------------------------------------------------- attention
+
+
+def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+              causal: bool = True, scale=None,
+              need_lse: bool = False):
+    """Flash attention forward on the hand-written CDNA4 MFMA kernel
+    (csrc/kernels/attention.hip). q: [B,Hq,S,D], k/v: [B,Hk,S,D] (GQA by
+    head-count ratio), D=128, bf16. Returns o (and lse in log2 domain when
+    need_lse). CPU fallback: exact fp32 reference."""
+    if scale is None:
+        scale = q.shape[-1] ** -0.5
+    if _use_hip(q):
+        o, lse = _hip().attn_fwd(q, k, v, float(scale), causal, need_lse)
+        return (o, lse) if need_lse else o
+    o = ref.attention_ref(q, k, v, causal, float(scale))
+    if need_lse:
+        raise NotImplementedError("need_lse is GPU-only")
+    return o
+
+
 # ------------------------------------------------------------------ adamw
 
 

@@ -73,3 +73,22 @@ def adamw_step(p32, p_bf16, g, m, v, lr, b1, b2, eps, wd, step, grad_scale):
     update = (m * bc1) / ((v * bc2).sqrt() + eps) + wd * p32
     p32.add_(update, alpha=-lr)
     p_bf16.copy_(p32.to(torch.bfloat16))
+
+
+def attention_ref(q, k, v, causal=True, scale=None):
+    """fp32 reference attention with GQA (q:[B,Hq,S,D], k/v:[B,Hk,S,D])."""
+    import torch
+
+    B, Hq, S, D = q.shape
+    Hk = k.shape[1]
+    if scale is None:
+        scale = D ** -0.5
+    rep = Hq // Hk
+    kf = k.float().repeat_interleave(rep, dim=1)
+    vf = v.float().repeat_interleave(rep, dim=1)
+    s = torch.einsum("bhid,bhjd->bhij", q.float(), kf) * scale
+    if causal:
+        mask = torch.ones(S, S, dtype=torch.bool, device=q.device).tril()
+        s = s.masked_fill(~mask, float("-inf"))
+    p = torch.softmax(s, dim=-1)
+    return torch.einsum("bhij,bhjd->bhid", p, vf).to(q.dtype)
