@@ -106,6 +106,13 @@ def init(
 
     host, port = gcs_addr.rsplit(":", 1)
     cw = CoreWorker(DRIVER_MODE, node_ip=_node_ip_address, session_dir=global_worker.session_dir)
+    # ant-fork parity: pin this driver's job to a virtual cluster
+    # (reference: job submission carries virtual_cluster_id; scheduling is
+    # then restricted to that cluster's nodes)
+    vc = kwargs.pop("_virtual_cluster_id", None) or os.environ.get(
+        "ANTRAY_VIRTUAL_CLUSTER")
+    if vc:
+        cw.virtual_cluster_id = vc
     try:
         cw.connect((host, int(port)), is_driver=True)
     except Exception:

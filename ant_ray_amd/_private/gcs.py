@@ -130,6 +130,13 @@ class GcsServer:
         self._server = None
         self.port = None
         self._shutdown = asyncio.Event()
+        # virtual clusters (ant fork feature — reference
+        # src/ray/gcs/gcs_virtual_cluster.h:29-610): named partitions of the
+        # node set; jobs pinned to a cluster schedule only on its nodes, and
+        # node-count-based clusters replenish from the unassigned pool on
+        # node death (gcs_virtual_cluster_manager.cc:730)
+        self.virtual_clusters: Dict[str, Dict[str, Any]] = {}
+        self.worker_vc: Dict[bytes, str] = {}  # worker_id -> vc id
 
     # ------------------------------------------------------------------ serve
     async def start(self, host="127.0.0.1", port=0):
@@ -176,6 +183,18 @@ class GcsServer:
     async def _on_node_dead(self, node: NodeInfo):
         node.alive = False
         await self._publish("NODE", node.node_id, node.view())
+        # virtual-cluster replenishment (gcs_virtual_cluster_manager.cc:730):
+        # a dead member of a count-based cluster is replaced from the
+        # unassigned pool when possible
+        for vc in self.virtual_clusters.values():
+            if node.node_id in vc["nodes"]:
+                vc["nodes"].discard(node.node_id)
+                free = [n.node_id for n in self.nodes.values()
+                        if n.alive and
+                        n.node_id not in self._vc_assigned_nodes()]
+                if free and len(vc["nodes"]) < vc["node_count"]:
+                    vc["nodes"].add(free[0])
+                vc["revision"] += 1
         # fail actors on that node
         for actor in list(self.actors.values()):
             if actor.node_id == node.node_id and actor.state in (ALIVE, PENDING_CREATION, RESTARTING):
@@ -222,6 +241,13 @@ class GcsServer:
             "job_id": job_id,
             "pid": p.get("pid"),
         }
+        vc = p.get("virtual_cluster_id")
+        if vc:
+            if vc not in self.virtual_clusters:
+                return {"error": f"virtual cluster {vc!r} does not exist"}
+            self.worker_vc[wid] = vc
+            if is_driver and job_id in self.jobs:
+                self.jobs[job_id]["virtual_cluster_id"] = vc
         conn.session["worker_id"] = wid
         return {"job_id": job_id}
 
@@ -232,6 +258,74 @@ class GcsServer:
             if actor.worker_id == wid and actor.state in (ALIVE, PENDING_CREATION, RESTARTING):
                 await self._on_actor_worker_died(actor, p.get("reason", "worker died"))
         return {"ok": True}
+
+
+    # -------------------------------------------------- virtual clusters
+    # parity: ant fork GcsVirtualClusterManager (gcs_virtual_cluster.h,
+    # doc/source/virtual-cluster/design-overview.rst)
+
+    def _vc_assigned_nodes(self):
+        out = set()
+        for vc in self.virtual_clusters.values():
+            out |= vc["nodes"]
+        return out
+
+    def _vc_allowed(self, vc_id):
+        """Node-id set a vc may schedule on, or None = unrestricted."""
+        if not vc_id:
+            return None
+        vc = self.virtual_clusters.get(vc_id)
+        return vc["nodes"] if vc is not None else set()
+
+    async def rpc_create_or_update_virtual_cluster(self, conn, p):
+        vc_id = p["virtual_cluster_id"]
+        vc = self.virtual_clusters.get(vc_id)
+        if vc is None:
+            vc = {"id": vc_id, "divisible": bool(p.get("divisible", False)),
+                  "nodes": set(), "node_count": 0, "revision": 0}
+            self.virtual_clusters[vc_id] = vc
+        if "node_ids" in p and p["node_ids"] is not None:
+            wanted = {bytes.fromhex(n) if isinstance(n, str) else n
+                      for n in p["node_ids"]}
+            taken = self._vc_assigned_nodes() - vc["nodes"]
+            conflict = wanted & taken
+            if conflict:
+                return {"ok": False, "error": "nodes already assigned: "
+                        + ",".join(n.hex()[:8] for n in conflict)}
+            vc["nodes"] = wanted
+            vc["node_count"] = len(wanted)
+        else:
+            want = int(p.get("node_count", 0))
+            vc["node_count"] = want
+            free = [n.node_id for n in self.nodes.values()
+                    if n.alive and n.node_id not in self._vc_assigned_nodes()]
+            while len(vc["nodes"]) < want and free:
+                vc["nodes"].add(free.pop(0))
+            while len(vc["nodes"]) > want:
+                vc["nodes"].pop()
+            if len(vc["nodes"]) < want:
+                vc["revision"] += 1
+                return {"ok": False,
+                        "error": f"only {len(vc['nodes'])} of {want} nodes "
+                                 "available", "view": self._vc_view(vc)}
+        vc["revision"] += 1
+        return {"ok": True, "view": self._vc_view(vc)}
+
+    def _vc_view(self, vc):
+        return {"virtual_cluster_id": vc["id"], "divisible": vc["divisible"],
+                "node_ids": [n.hex() for n in vc["nodes"]],
+                "node_count": vc["node_count"], "revision": vc["revision"]}
+
+    async def rpc_remove_virtual_cluster(self, conn, p):
+        vc = self.virtual_clusters.pop(p["virtual_cluster_id"], None)
+        return {"ok": vc is not None}
+
+    async def rpc_list_virtual_clusters(self, conn, p):
+        return [self._vc_view(vc) for vc in self.virtual_clusters.values()]
+
+    async def rpc_get_virtual_cluster(self, conn, p):
+        vc = self.virtual_clusters.get(p["virtual_cluster_id"])
+        return self._vc_view(vc) if vc else None
 
     # --------------------------------------------------------------------- kv
     async def rpc_kv_put(self, conn, p):
@@ -308,7 +402,8 @@ class GcsServer:
         return {"existing": False, "actor_id": actor_id}
 
     def _pick_node(self, resources: Dict[str, float], pg: Optional[dict] = None,
-                   node_affinity: Optional[bytes] = None) -> Optional[NodeInfo]:
+                   node_affinity: Optional[bytes] = None,
+                   allowed: Optional[set] = None) -> Optional[NodeInfo]:
         if pg:
             pg_info = self.pgs.get(pg["pg_id"])
             if pg_info and pg_info.state == "CREATED":
@@ -327,6 +422,8 @@ class GcsServer:
         best, best_score = None, -1.0
         for node in self.nodes.values():
             if not node.alive:
+                continue
+            if allowed is not None and node.node_id not in allowed:
                 continue
             feasible = all(
                 node.resources_total.get(k, 0) >= v for k, v in resources.items()
@@ -354,8 +451,10 @@ class GcsServer:
             resources["GPU"] = float(opts["num_gpus"])
         pg = opts.get("placement_group")
         deadline = time.monotonic() + float(opts.get("_scheduling_timeout", 3600.0))
+        vc_id = opts.get("virtual_cluster_id") or self.worker_vc.get(info.owner)
         while True:
-            node = self._pick_node(resources, pg, opts.get("_node_affinity"))
+            node = self._pick_node(resources, pg, opts.get("_node_affinity"),
+                                   self._vc_allowed(vc_id))
             if node is not None and node.conn is not None and not node.conn.closed:
                 try:
                     lease = await node.conn.call(
@@ -515,6 +614,7 @@ class GcsServer:
     # ------------------------------------------------------------ placement
     async def rpc_create_placement_group(self, conn, p):
         pg = PlacementGroupInfo(p["pg_id"], p["bundles"], p.get("strategy", "PACK"), p.get("name", ""))
+        pg.vc_id = self.worker_vc.get(conn.session.get("worker_id"))
         self.pgs[pg.pg_id] = pg
         asyncio.get_running_loop().create_task(self._schedule_pg(pg))
         return {"ok": True}
@@ -558,6 +658,9 @@ class GcsServer:
     def _plan_pg(self, pg: PlacementGroupInfo):
         """Return [(bundle_index, NodeInfo)] or None if not placeable now."""
         alive = [n for n in self.nodes.values() if n.alive]
+        allowed = self._vc_allowed(getattr(pg, "vc_id", None))
+        if allowed is not None:
+            alive = [n for n in alive if n.node_id in allowed]
         if not alive:
             return None
         avail = {n.node_id: dict(n.resources_available) for n in alive}

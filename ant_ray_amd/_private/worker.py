@@ -177,8 +177,12 @@ class CoreWorker:
                 "node_id": self.node_id,
                 "is_driver": is_driver,
                 "pid": os.getpid(),
+                "virtual_cluster_id": os.environ.get("ANTRAY_VIRTUAL_CLUSTER")
+                or getattr(self, "virtual_cluster_id", None),
             },
         )
+        if reply.get("error"):
+            raise ConnectionError(reply["error"])
         self.job_id = reply.get("job_id") or 0
         await self.gcs.call("subscribe", {"channels": ["ACTOR", "NODE"]})
 

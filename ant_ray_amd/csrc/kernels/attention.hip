@@ -54,7 +54,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   const int hk = hq / (Hq / Hk);
 
   const int q0 = m_block * BLOCK_M + wave * 32;   // this wave's first q row
-  if (q0 >= S) return;
+  // NOTE: no early return — the K/V staging barrier below is workgroup-wide,
+  // so every wave must stay in the tile loop (tail waves just mask out)
   const int q_row = q0 + (lane & 31);             // this lane's q row
   const bool q_valid = q_row < S;
 
@@ -80,8 +81,13 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   float m_run = -1e30f, l_run = 0.f;
   floatx16 o_acc[4] = {};  // 4 d-tiles of 32; C[m=q? no: m spread, n=d]
 
-  const int n_end_row = causal ? (q0 + 31) : (S - 1);  // last key this wave needs
-  const int n_tiles = min((n_end_row / BLOCK_N) + 1, (S + BLOCK_N - 1) / BLOCK_N);
+  // tile count must be UNIFORM across the workgroup (staging barriers):
+  // loop to the LAST wave's causal bound; earlier waves skip compute on
+  // tiles past their own bound
+  const int wg_last_row = m_block * BLOCK_M + (BLOCK_M - 1);
+  const int n_end_row = causal ? min(wg_last_row, S - 1) : (S - 1);
+  const int n_tiles = (n_end_row / BLOCK_N) + 1;
+  const int my_last_tile = causal ? ((q0 + 31) / BLOCK_N) : (n_tiles - 1);
 
   for (int t = 0; t < n_tiles; ++t) {
     const int n0 = t * BLOCK_N;
@@ -105,6 +111,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       }
     }
     __syncthreads();
+    if (t > my_last_tile || q0 >= S) continue;  // masked-out wave: stage only
 
     // ---- S^T = K . Q^T  (C: n = q = lane&31, m = key offset)
     floatx16 st_acc = {};

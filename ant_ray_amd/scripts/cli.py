@@ -233,3 +233,48 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+@cli.group()
+def vcluster():
+    """Virtual clusters (ant-fork parity: partition nodes into named slices)."""
+
+
+@vcluster.command("create")
+@click.argument("vc_id")
+@click.option("--node-count", type=int, default=None)
+@click.option("--node-ids", default=None, help="comma-separated hex node ids")
+@click.option("--divisible", is_flag=True)
+def vcluster_create(vc_id, node_count, node_ids, divisible):
+    import ant_ray_amd as ray
+    from ant_ray_amd.util import virtual_cluster as vc
+
+    ray.init(address=os.environ.get("RAY_ADDRESS", "auto"),
+             ignore_reinit_error=True)
+    view = vc.create_or_update_virtual_cluster(
+        vc_id, node_count=node_count,
+        node_ids=node_ids.split(",") if node_ids else None,
+        divisible=divisible)
+    click.echo(json.dumps(view))
+
+
+@vcluster.command("remove")
+@click.argument("vc_id")
+def vcluster_remove(vc_id):
+    import ant_ray_amd as ray
+    from ant_ray_amd.util import virtual_cluster as vc
+
+    ray.init(address=os.environ.get("RAY_ADDRESS", "auto"),
+             ignore_reinit_error=True)
+    click.echo(json.dumps({"removed": vc.remove_virtual_cluster(vc_id)}))
+
+
+@vcluster.command("list")
+def vcluster_list():
+    import ant_ray_amd as ray
+    from ant_ray_amd.util import virtual_cluster as vc
+
+    ray.init(address=os.environ.get("RAY_ADDRESS", "auto"),
+             ignore_reinit_error=True)
+    for v in vc.list_virtual_clusters():
+        click.echo(json.dumps(v))

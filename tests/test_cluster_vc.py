@@ -1,0 +1,116 @@
+"""Multi-node tests: cluster_utils.Cluster (fake multi-raylet cluster),
+node death handling, and virtual clusters (ant-fork feature parity)."""
+import time
+
+import pytest
+
+
+@pytest.fixture()
+def cluster():
+    from ant_ray_amd.cluster_utils import Cluster
+
+    c = Cluster(initialize_head=True, head_node_args={"num_cpus": 2})
+    yield c
+    c.shutdown()
+
+
+def test_add_remove_node_and_spread(cluster):
+    import ant_ray_amd as ray
+
+    cluster.connect()
+    n1 = cluster.add_node(num_cpus=2, resources={"tag_a": 1})
+    cluster.add_node(num_cpus=2, resources={"tag_b": 1})
+    assert len(ray.nodes()) == 3
+    total = ray.cluster_resources()
+    assert total["CPU"] == 6.0
+    assert total.get("tag_a") == 1.0
+
+    # actors land on the custom-resource nodes
+    @ray.remote(num_cpus=1, resources={"tag_a": 0.1})
+    class A:
+        def where(self):
+            import os
+
+            return os.getpid()
+
+    a = A.remote()
+    assert ray.get(a.where.remote(), timeout=60) > 0
+
+    # node removal -> resources drop, node marked dead
+    cluster.remove_node(n1)
+    deadline = time.time() + 30
+    while time.time() < deadline:
+        alive = [n for n in ray.nodes() if n["Alive"]]
+        if len(alive) == 2:
+            break
+        time.sleep(0.3)
+    assert len([n for n in ray.nodes() if n["Alive"]]) == 2
+
+
+def test_virtual_cluster_scheduling(cluster):
+    import ant_ray_amd as ray
+    from ant_ray_amd.util import virtual_cluster as vc
+
+    cluster.connect()
+    cluster.add_node(num_cpus=2)
+    cluster.add_node(num_cpus=2)
+
+    nodes = ray.nodes()
+    head_id = min(n["NodeID"] for n in nodes)  # arbitrary but stable pick
+    view = vc.create_or_update_virtual_cluster("vc1", node_ids=[head_id])
+    assert view["node_ids"] == [head_id]
+    assert any(v["virtual_cluster_id"] == "vc1"
+               for v in vc.list_virtual_clusters())
+
+    # a second vc cannot steal vc1's node
+    with pytest.raises(RuntimeError):
+        vc.create_or_update_virtual_cluster("vc2", node_ids=[head_id])
+
+    # count-based vc takes nodes from the unassigned pool
+    view2 = vc.create_or_update_virtual_cluster("vc2", node_count=1)
+    assert len(view2["node_ids"]) == 1 and view2["node_ids"] != [head_id]
+
+    # actors of a vc1-pinned driver land only on vc1's node
+    import subprocess
+    import sys
+    import textwrap
+
+    script = textwrap.dedent(f"""
+        import os
+        import ant_ray_amd as ray
+        ray.init(address="{cluster.address}", _virtual_cluster_id="vc1")
+
+        @ray.remote(num_cpus=1)
+        class W:
+            def node(self):
+                return ray.get_runtime_context().get_node_id()
+
+        ws = [W.remote() for _ in range(2)]
+        nodes = set(ray.get([w.node.remote() for w in ws], timeout=60))
+        assert nodes == {{"{head_id}"}}, nodes
+        print("VC_OK")
+    """)
+    out = subprocess.run([sys.executable, "-c", script], capture_output=True,
+                         text=True, timeout=120)
+    assert "VC_OK" in out.stdout, out.stdout + out.stderr
+
+    assert vc.remove_virtual_cluster("vc1")
+    assert vc.get_virtual_cluster("vc1") is None
+
+
+def test_unknown_virtual_cluster_rejected(cluster):
+    import subprocess
+    import sys
+    import textwrap
+
+    script = textwrap.dedent(f"""
+        import ant_ray_amd as ray
+        try:
+            ray.init(address="{cluster.address}", _virtual_cluster_id="nope")
+            print("CONNECTED")
+        except ConnectionError as e:
+            print("REJECTED", e)
+    """)
+    out = subprocess.run([sys.executable, "-c", script], capture_output=True,
+                         text=True, timeout=120)
+    assert "REJECTED" in out.stdout, out.stdout + out.stderr

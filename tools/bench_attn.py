@@ -2,7 +2,11 @@
 
 Run on the GPU box:  python tools/bench_attn.py
 """
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
