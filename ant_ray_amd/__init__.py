@@ -123,6 +123,10 @@ def init(
     global_worker.mode = DRIVER_MODE
     global_worker._head_proc = head
     atexit.register(shutdown)
+    if include_dashboard:
+        from ant_ray_amd.dashboard import start_dashboard
+
+        start_dashboard(port=dashboard_port or 8265, host=dashboard_host)
     return RuntimeContext(global_worker)
 
 

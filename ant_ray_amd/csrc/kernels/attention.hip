@@ -1,22 +1,27 @@
 // Flash-attention forward, hand-written for CDNA4 (gfx950).
 //
-// Replaces torch SDPA's AOTriton path (measured 369 TF/s fwd on this
-// workload) with an MFMA 32x32x16 bf16 kernel using the CDNA4 idioms from
-// the playbook: swapped QK^T (S^T = K.Q^T) so each lane owns a full P row
-// (q = lane&31), online softmax in exp2 domain, K staged row-major in LDS,
-// V staged TRANSPOSED in LDS so the P.V B-fragment is one ds_read_b128.
+// Replaces torch SDPA's AOTriton path (measured 369-399 TF/s fwd on the
+// Llama-3-8B shape) with an MFMA 32x32x16 bf16 kernel built from the CDNA4
+// playbook idioms:
+//   * swapped QK^T (S^T = K.Q^T) so each lane owns one q row (q = lane&31)
+//     and the softmax row reduction needs only a lane<->lane^32 exchange,
+//   * online softmax in the exp2 domain with defer-max rescale (T13):
+//     the O/l rescale pass runs only when the row max grows > THR,
+//   * P A-fragments built by v_cvt_pk_bf16_f32 + v_permlane32_swap (T12):
+//     4 permlanes + 8 cvt_pk replace 16 ds_bpermute + scalar bf16 packing,
+//   * -inf masking (exp2(-inf - m) == 0) so the exp path has NO branches,
+//   * K staged row-major with the T2 XOR swizzle (block ^= row&7) and V
+//     staged TRANSPOSED (VROW=40, k ^= 8*((d>>4)&3)): conflict-reduced
+//     ds_read_b128 fragment reads on both,
+//   * T14 issue-early staging: tile t+1's global loads are in flight while
+//     tile t computes.
 //
 // Layout contract (one workgroup = 4 waves = one 128-row Q block):
 //   wave w handles q rows [m0 + 32w, m0 + 32w + 32)
-//   K-tile loop: 32 keys/tile, causal upper bound at the wave's last row.
 //   mfma_f32_32x32x16_bf16 fragment maps (cdna4_isa.md §10):
 //     A[m][k]: m = l&31, k = (l>>5)*8 + j            (j = 0..7)
 //     B[k][n]: n = l&31, k = (l>>5)*8 + j
 //     C[m][n]: n = l&31, m = (j&3) + 8*(j>>2) + 4*(l>>5)  (j = 0..15)
-//
-// Swapped S^T = mfma(A=K, B=Q): C cols n = q (lane-local), rows m = key.
-// Lane pair (l, l^32) splits the 32 keys of one q row; __shfl_xor(.,32)
-// closes row reductions and builds the P A-fragment for P.V.
 //
 // Shapes: D = 128 fixed; Hq % Hk == 0 (GQA); any S, causal or not.
 #include <hip/hip_runtime.h>
@@ -25,14 +30,35 @@
 
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float floatx16 __attribute__((ext_vector_type(16)));
+typedef unsigned int uint32x4 __attribute__((ext_vector_type(4)));
 
 #define ATTN_D 128
 #define BLOCK_M 128   // q rows per workgroup (32 per wave)
 #define BLOCK_N 32    // keys per tile
-#define KROW 136      // K LDS row stride in elems (16B aligned, de-banked)
-#define VROW 48       // V^T LDS row stride in elems (16B aligned)
+#define KROW 128      // K LDS row stride (T2 swizzle instead of padding)
+#define VROW 40       // V^T LDS row stride (odd multiple of 8)
+#define RESCALE_THR 8.0f  // T13: defer O-rescale until max grows > 2^8
+
+#define NEG_INF (-__builtin_inff())
 
 __device__ __forceinline__ float xor32(float v) { return __shfl_xor(v, 32); }
+
+// v_cvt_pk_bf16_f32: packs (lo, hi) f32 -> one u32 of 2 bf16 (HW RNE)
+__device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+// K LDS: element offset of (row, 8-elem block blk), T2 XOR swizzle
+__device__ __forceinline__ int k_lds_off(int row, int blk) {
+  return row * KROW + 8 * (blk ^ (row & 7));
+}
+
+// V^T LDS: element offset of (d row, k element), k-block swizzled by d
+__device__ __forceinline__ int v_lds_off(int d, int k) {
+  return d * VROW + (k ^ (8 * ((d >> 4) & 3)));
+}
 
 extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     const ushort_t* __restrict__ Q,  // [B, Hq, S, D] via strides
@@ -41,7 +67,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     ushort_t* __restrict__ O,        // [B, Hq, S, D]
     float* __restrict__ LSE,         // [B, Hq, S] log2-domain lse (for bwd)
     int S, int Hq, int Hk,
-    long qb, long qh, long qs,       // Q strides (elements)
+    long qb, long qh, long qs,       // strides (elements)
     long kb, long kh, long ks,
     long ob, long oh, long os,
     float scale_log2,                // softmax_scale * log2(e)
@@ -54,9 +80,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   const int hk = hq / (Hq / Hk);
 
   const int q0 = m_block * BLOCK_M + wave * 32;   // this wave's first q row
-  // NOTE: no early return — the K/V staging barrier below is workgroup-wide,
-  // so every wave must stay in the tile loop (tail waves just mask out)
-  const int q_row = q0 + (lane & 31);             // this lane's q row
+  // no early return: the staging barriers are workgroup-wide
+  const int q_row = q0 + (lane & 31);
   const bool q_valid = q_row < S;
 
   const ushort_t* Qp = Q + (long)b * qb + (long)hq * qh;
@@ -66,7 +91,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   __shared__ ushort_t k_lds[32 * KROW];
   __shared__ ushort_t v_lds[ATTN_D * VROW];
 
-  // ---- Q fragments: B[k=d][n=q], lane holds Q[q_row][step*16+(l>>5)*8+0..7]
+  // ---- Q fragments: B[k=d][n=q]
   bf16x8 qf[8];
   {
     const ushort_t* qrow = Qp + (long)(q_valid ? q_row : 0) * qs;
@@ -77,139 +102,135 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
           bf16x8, *(const ushortx8*)(qrow + st * 16 + dbase));
   }
 
-  // ---- online-softmax state (per lane == per q row) + O accumulators
   float m_run = -1e30f, l_run = 0.f;
-  floatx16 o_acc[4] = {};  // 4 d-tiles of 32; C[m=q? no: m spread, n=d]
+  floatx16 o_acc[4] = {};
 
-  // tile count must be UNIFORM across the workgroup (staging barriers):
-  // loop to the LAST wave's causal bound; earlier waves skip compute on
-  // tiles past their own bound
   const int wg_last_row = m_block * BLOCK_M + (BLOCK_M - 1);
   const int n_end_row = causal ? min(wg_last_row, S - 1) : (S - 1);
   const int n_tiles = (n_end_row / BLOCK_N) + 1;
   const int my_last_tile = causal ? ((q0 + 31) / BLOCK_N) : (n_tiles - 1);
 
+  // staging geometry: thread covers (row = tid>>3, 16 elems at (tid&7)*16)
+  const int st_row = threadIdx.x >> 3;
+  const int st_col = (threadIdx.x & 7) * 16;
+
+  // T14: tile 0 loads issued before the loop
+  ushortx8 kv0, kv1, vv0, vv1;
+  {
+    const int krow_g = min(st_row, S - 1);
+    kv0 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
+    kv1 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col + 8);
+    vv0 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col);
+    vv1 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col + 8);
+  }
+
   for (int t = 0; t < n_tiles; ++t) {
     const int n0 = t * BLOCK_N;
-    // ---- stage K tile row-major + V tile transposed
-    __syncthreads();
-    {
-      // 256 threads, 32x128 elems: thread tid covers (row=tid>>3, 16 elems)
-      const int row = threadIdx.x >> 3;
-      const int col = (threadIdx.x & 7) * 16;
-      const int krow_g = min(n0 + row, S - 1);
-      const ushortx8 kv0 = *(const ushortx8*)(Kp + (long)krow_g * ks + col);
-      const ushortx8 kv1 = *(const ushortx8*)(Kp + (long)krow_g * ks + col + 8);
-      *(ushortx8*)(&k_lds[row * KROW + col]) = kv0;
-      *(ushortx8*)(&k_lds[row * KROW + col + 8]) = kv1;
-      const ushortx8 vv0 = *(const ushortx8*)(Vp + (long)krow_g * ks + col);
-      const ushortx8 vv1 = *(const ushortx8*)(Vp + (long)krow_g * ks + col + 8);
+    __syncthreads();  // everyone done reading the previous tile's LDS
+    *(ushortx8*)(&k_lds[k_lds_off(st_row, st_col / 8)]) = kv0;
+    *(ushortx8*)(&k_lds[k_lds_off(st_row, st_col / 8 + 1)]) = kv1;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        v_lds[(col + j) * VROW + row] = vv0[j];
-        v_lds[(col + 8 + j) * VROW + row] = vv1[j];
-      }
+    for (int j = 0; j < 8; ++j) {
+      v_lds[v_lds_off(st_col + j, st_row)] = vv0[j];
+      v_lds[v_lds_off(st_col + 8 + j, st_row)] = vv1[j];
     }
-    __syncthreads();
-    if (t > my_last_tile || q0 >= S) continue;  // masked-out wave: stage only
+    __syncthreads();  // tile t fully staged
+    if (t + 1 < n_tiles) {
+      const int krow_g = min((t + 1) * BLOCK_N + st_row, S - 1);
+      kv0 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
+      kv1 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col + 8);
+      vv0 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col);
+      vv1 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col + 8);
+    }
+    if (t > my_last_tile || q0 >= S) continue;  // masked wave: stage only
 
     // ---- S^T = K . Q^T  (C: n = q = lane&31, m = key offset)
     floatx16 st_acc = {};
-    {
-      const int dbase = (lane >> 5) * 8;
 #pragma unroll
-      for (int stp = 0; stp < 8; ++stp) {
-        bf16x8 kf = __builtin_bit_cast(
-            bf16x8,
-            *(const ushortx8*)(&k_lds[(lane & 31) * KROW + stp * 16 + dbase]));
-        st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[stp], st_acc,
-                                                         0, 0, 0);
-      }
+    for (int stp = 0; stp < 8; ++stp) {
+      bf16x8 kf = __builtin_bit_cast(
+          bf16x8, *(const ushortx8*)(
+                      &k_lds[k_lds_off(lane & 31, 2 * stp + (lane >> 5))]));
+      st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[stp], st_acc,
+                                                       0, 0, 0);
     }
 
-    // ---- mask + scale into exp2 domain
+    // ---- mask + scale; dead scores are -inf so exp2 underflows to 0
     const int mrow_base = 4 * (lane >> 5);
     float s_val[16];
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
       const int key = n0 + (j & 3) + 8 * (j >> 2) + mrow_base;
-      float s = st_acc[j] * scale_log2;
       const bool dead = (causal && key > q_row) || key >= S || !q_valid;
-      s_val[j] = dead ? -1e30f : s;
+      s_val[j] = dead ? NEG_INF : st_acc[j] * scale_log2;
     }
 
-    // ---- online softmax (lane pair l, l^32 shares q row)
-    float tmax = s_val[0];
+    // ---- online softmax with defer-max (T13)
+    float tmax = fmaxf(s_val[0], s_val[1]);
 #pragma unroll
-    for (int j = 1; j < 16; ++j) tmax = fmaxf(tmax, s_val[j]);
+    for (int j = 2; j < 16; ++j) tmax = fmaxf(tmax, s_val[j]);
     tmax = fmaxf(tmax, xor32(tmax));
-    const float m_new = fmaxf(m_run, tmax);
-    const float alpha = (m_run <= -1e30f) ? 0.f : __builtin_exp2f(m_run - m_new);
+    if (__any(tmax > m_run + RESCALE_THR)) {
+      const float m_new = fmaxf(m_run, tmax);
+      const float alpha = __builtin_exp2f(m_run - m_new);  // 0 on 1st tile
+      l_run *= alpha;
+      m_run = m_new;
+      float alpha_j[16];
+#pragma unroll
+      for (int j = 0; j < 16; ++j)
+        alpha_j[j] = __shfl(alpha, (j & 3) + 8 * (j >> 2) + mrow_base);
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+        for (int j = 0; j < 16; ++j) o_acc[dt][j] *= alpha_j[j];
+    }
     float psum = 0.f;
     float p_val[16];
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
-      p_val[j] = (s_val[j] <= -1e30f) ? 0.f : __builtin_exp2f(s_val[j] - m_new);
+      p_val[j] = __builtin_exp2f(s_val[j] - m_run);  // bounded by 2^THR
       psum += p_val[j];
     }
     psum += xor32(psum);
-    l_run = l_run * alpha + psum;
-    m_run = m_new;
+    l_run += psum;
 
-    // ---- build P A-fragments: A[m=q][k=key], lane needs keys
-    //      (l>>5)*8 + 0..7 (+16 for kstep 1); own regs cover keys
-    //      {0..3,8..11,16..19,24..27} + mrow_base; partner has the rest.
-    float p_part[16];
+    // ---- P A-fragments via cvt_pk + permlane32_swap (T12).
+    // own_pk[2b+c] packs this lane-half's keys (8b + 4h + 2c, +1);
+    // permlane32_swap(D, S) exchanges D's hi lanes with S's lo lanes, so
+    // swap(own_pk[x], own_pk[x+2]) returns the pf words for BOTH halves.
+    unsigned own_pk[8];
 #pragma unroll
-    for (int j = 0; j < 16; ++j) p_part[j] = xor32(p_val[j]);
-    // Lane half h = lane>>5 wants keys 8h+16*kstep+i (i<8). Inverting the
-    // C map gives reg j = (i&3) + 8*kstep + 4h, from p_val when
-    // (i>=4)==(h==1) else p_part. All selects below are between
-    // CONSTANT-indexed registers (rule #20: runtime-indexed arrays spill).
-    const bool hi_half = (lane >> 5) != 0;
-    float own_sh[8], par_sh[8];  // c = (i&3) + 4*kstep -> j = (c&3)+8*(c>>2)+4h
-#pragma unroll
-    for (int c = 0; c < 8; ++c) {
-      const int j0 = (c & 3) + 8 * (c >> 2);
-      own_sh[c] = hi_half ? p_val[j0 + 4] : p_val[j0];
-      par_sh[c] = hi_half ? p_part[j0 + 4] : p_part[j0];
+    for (int b2 = 0; b2 < 4; ++b2) {
+      own_pk[2 * b2] = cvt_pk_bf16(p_val[4 * b2], p_val[4 * b2 + 1]);
+      own_pk[2 * b2 + 1] = cvt_pk_bf16(p_val[4 * b2 + 2], p_val[4 * b2 + 3]);
     }
-    bf16x8 pf[2];
-#pragma unroll
-    for (int kstep = 0; kstep < 2; ++kstep) {
-      ushortx8 pk;
-#pragma unroll
-      for (int i = 0; i < 8; ++i) {
-        const int c = (i & 3) + 4 * kstep;
-        const bool own_lo = (i < 4);  // own iff (i>=4)==hi_half
-        const float v = hi_half ? (own_lo ? par_sh[c] : own_sh[c])
-                                : (own_lo ? own_sh[c] : par_sh[c]);
-        pk[i] = f2bf(v);
-      }
-      pf[kstep] = __builtin_bit_cast(bf16x8, pk);
+    uint32x4 pw0, pw1;
+    {
+      auto r0 = __builtin_amdgcn_permlane32_swap(own_pk[0], own_pk[2], false,
+                                                 false);
+      auto r1 = __builtin_amdgcn_permlane32_swap(own_pk[1], own_pk[3], false,
+                                                 false);
+      pw0[0] = r0[0]; pw0[2] = r0[1];
+      pw0[1] = r1[0]; pw0[3] = r1[1];
+      auto r2 = __builtin_amdgcn_permlane32_swap(own_pk[4], own_pk[6], false,
+                                                 false);
+      auto r3 = __builtin_amdgcn_permlane32_swap(own_pk[5], own_pk[7], false,
+                                                 false);
+      pw1[0] = r2[0]; pw1[2] = r2[1];
+      pw1[1] = r3[0]; pw1[3] = r3[1];
     }
-
-    // ---- O rescale by alpha (broadcast alpha from lane q to C rows)
-    float alpha_j[16];
-#pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      const int qrow_j = (j & 3) + 8 * (j >> 2) + mrow_base;
-      alpha_j[j] = __shfl(alpha, qrow_j);
-    }
-#pragma unroll
-    for (int dt = 0; dt < 4; ++dt)
-#pragma unroll
-      for (int j = 0; j < 16; ++j) o_acc[dt][j] *= alpha_j[j];
+    bf16x8 pf[2] = {__builtin_bit_cast(bf16x8, pw0),
+                    __builtin_bit_cast(bf16x8, pw1)};
 
     // ---- O += P . V   (B[k][n=d] = one b128 from transposed V)
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt) {
-      const int d0 = dt * 32;
+      const int d0 = dt * 32 + (lane & 31);
 #pragma unroll
       for (int kstep = 0; kstep < 2; ++kstep) {
         bf16x8 vf = __builtin_bit_cast(
-            bf16x8, *(const ushortx8*)(&v_lds[(d0 + (lane & 31)) * VROW +
-                                              kstep * 16 + (lane >> 5) * 8]));
+            bf16x8, *(const ushortx8*)(
+                        &v_lds[v_lds_off(d0, kstep * 16 + (lane >> 5) * 8)]));
         o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf[kstep], vf,
                                                             o_acc[dt], 0, 0, 0);
       }
@@ -234,10 +255,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       Op[(long)qr * os + dt * 32 + (lane & 31)] = f2bf(o_acc[dt][j] * linv_j[j]);
     }
   }
-  if (LSE != nullptr && q_valid && lane < 32) {
-    // lane l<32 and partner hold identical (m, l) after the xor reduction
-    LSE[((long)b * Hq + hq) * S + q_row] =
-        m_run + __builtin_log2f(l_safe);
+  if (LSE != nullptr && q_valid && q0 < S && lane < 32) {
+    LSE[((long)b * Hq + hq) * S + q_row] = m_run + __builtin_log2f(l_safe);
   }
 }
 

@@ -1,0 +1,123 @@
+"""Tests: ray.util.queue / ActorPool / metrics, and the dashboard REST head."""
+import json
+import time
+import urllib.request
+
+import pytest
+
+
+@pytest.fixture(scope="module")
+def ray_mod():
+    import ant_ray_amd as ray
+
+    if not ray.is_initialized():
+        ray.init(num_cpus=8)
+    yield ray
+    ray.shutdown()
+
+
+def test_queue(ray_mod):
+    from ant_ray_amd.util.queue import Empty, Queue
+
+    q = Queue(maxsize=3)
+    q.put(1)
+    q.put(2)
+    assert q.qsize() == 2
+    assert q.get() == 1
+    assert q.get(timeout=5) == 2
+    with pytest.raises(Empty):
+        q.get(block=False)
+    q.shutdown()
+
+
+def test_queue_producer_consumer(ray_mod):
+    ray = ray_mod
+    from ant_ray_amd.util.queue import Queue
+
+    q = Queue()
+
+    @ray.remote
+    def producer(q, n):
+        for i in range(n):
+            q.put(i)
+        return n
+
+    @ray.remote
+    def consumer(q, n):
+        return sum(q.get(timeout=30) for _ in range(n))
+
+    p = producer.remote(q, 10)
+    c = consumer.remote(q, 10)
+    assert ray.get(c, timeout=60) == 45
+    assert ray.get(p) == 10
+    q.shutdown()
+
+
+def test_actor_pool(ray_mod):
+    ray = ray_mod
+    from ant_ray_amd.util.actor_pool import ActorPool
+
+    @ray.remote
+    class Sq:
+        def sq(self, x):
+            return x * x
+
+    pool = ActorPool([Sq.remote() for _ in range(3)])
+    out = list(pool.map(lambda a, v: a.sq.remote(v), range(10)))
+    assert out == [i * i for i in range(10)]
+    out = sorted(pool.map_unordered(lambda a, v: a.sq.remote(v), range(6)))
+    assert out == [i * i for i in range(6)]
+
+
+def test_metrics_and_dashboard(ray_mod):
+    import ant_ray_amd as ray
+    from ant_ray_amd.dashboard import start_dashboard
+    from ant_ray_amd.util.metrics import Counter, Gauge, Histogram
+
+    c = Counter("test_requests_total", tag_keys=("route",))
+    c.inc(1, {"route": "/a"})
+    c.inc(2, {"route": "/a"})
+    g = Gauge("test_temp", tag_keys=())
+    g.set(42.5)
+    h = Histogram("test_lat", boundaries=[0.1, 1.0], tag_keys=())
+    h.observe(0.05)
+    h.observe(0.5)
+    h.observe(5.0)
+
+    port = start_dashboard(port=0 or 8277)
+    base = f"http://127.0.0.1:{port}"
+    time.sleep(1.2)  # metrics publish is fire-and-forget
+
+    with urllib.request.urlopen(f"{base}/healthz", timeout=10) as r:
+        assert r.read() == b"ok"
+    with urllib.request.urlopen(f"{base}/api/cluster_status", timeout=10) as r:
+        st = json.loads(r.read())
+    assert st["nodes"] and "resources" in st
+    with urllib.request.urlopen(f"{base}/api/actors", timeout=10) as r:
+        assert isinstance(json.loads(r.read()), list)
+    with urllib.request.urlopen(f"{base}/metrics", timeout=10) as r:
+        text = r.read().decode()
+    assert 'test_requests_total{route="/a"} 3.0' in text
+    assert "test_temp 42.5" in text
+    assert 'test_lat_bucket{le="0.1"} 1' in text
+    assert 'test_lat_bucket{le="+Inf"} 3' in text
+
+    # job submission via REST
+    req = urllib.request.Request(
+        f"{base}/api/jobs",
+        data=json.dumps({"entrypoint": "echo rest_job_ok"}).encode(),
+        method="POST", headers={"Content-Type": "application/json"})
+    with urllib.request.urlopen(req, timeout=30) as r:
+        sub = json.loads(r.read())
+    job_id = sub["submission_id"]
+    deadline = time.time() + 30
+    while time.time() < deadline:
+        with urllib.request.urlopen(f"{base}/api/jobs/{job_id}", timeout=10) as r:
+            info = json.loads(r.read())
+        if info.get("status") in ("SUCCEEDED", "FAILED"):
+            break
+        time.sleep(0.5)
+    assert info["status"] == "SUCCEEDED"
+    with urllib.request.urlopen(f"{base}/api/jobs/{job_id}/logs",
+                                timeout=10) as r:
+        assert b"rest_job_ok" in r.read()
