@@ -88,8 +88,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   const ushort_t* Kp = K + (long)b * kb + (long)hk * kh;
   const ushort_t* Vp = V + (long)b * kb + (long)hk * kh;
 
-  __shared__ ushort_t k_lds[32 * KROW];
-  __shared__ ushort_t v_lds[ATTN_D * VROW];
+  __shared__ ushort_t k_lds[2][32 * KROW];
+  __shared__ ushort_t v_lds[2][ATTN_D * VROW];
 
   // ---- Q fragments: B[k=d][n=q]
   bf16x8 qf[8];
@@ -126,15 +126,17 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 
   for (int t = 0; t < n_tiles; ++t) {
     const int n0 = t * BLOCK_N;
-    __syncthreads();  // everyone done reading the previous tile's LDS
-    *(ushortx8*)(&k_lds[k_lds_off(st_row, st_col / 8)]) = kv0;
-    *(ushortx8*)(&k_lds[k_lds_off(st_row, st_col / 8 + 1)]) = kv1;
+    const int buf = t & 1;
+    // double buffer: write tile t into buf while tile t-1 (other buf) may
+    // still be read; ONE barrier per tile orders write(t) vs read(t)
+    *(ushortx8*)(&k_lds[buf][k_lds_off(st_row, st_col / 8)]) = kv0;
+    *(ushortx8*)(&k_lds[buf][k_lds_off(st_row, st_col / 8 + 1)]) = kv1;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      v_lds[v_lds_off(st_col + j, st_row)] = vv0[j];
-      v_lds[v_lds_off(st_col + 8 + j, st_row)] = vv1[j];
+      v_lds[buf][v_lds_off(st_col + j, st_row)] = vv0[j];
+      v_lds[buf][v_lds_off(st_col + 8 + j, st_row)] = vv1[j];
     }
-    __syncthreads();  // tile t fully staged
+    __syncthreads();  // tile t staged in buf; tile t-1 reads are also done
     if (t + 1 < n_tiles) {
       const int krow_g = min((t + 1) * BLOCK_N + st_row, S - 1);
       kv0 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
@@ -146,23 +148,32 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 
     // ---- S^T = K . Q^T  (C: n = q = lane&31, m = key offset)
     floatx16 st_acc = {};
+    __builtin_amdgcn_s_setprio(1);  // T5: keep the matrix pipe fed
 #pragma unroll
     for (int stp = 0; stp < 8; ++stp) {
       bf16x8 kf = __builtin_bit_cast(
           bf16x8, *(const ushortx8*)(
-                      &k_lds[k_lds_off(lane & 31, 2 * stp + (lane >> 5))]));
+                      &k_lds[buf][k_lds_off(lane & 31, 2 * stp + (lane >> 5))]));
       st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[stp], st_acc,
                                                        0, 0, 0);
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // ---- mask + scale; dead scores are -inf so exp2 underflows to 0
     const int mrow_base = 4 * (lane >> 5);
     float s_val[16];
+    const bool needs_mask =
+        (causal && n0 + BLOCK_N - 1 > q0) || (n0 + BLOCK_N > S) || !q_valid;
+    if (needs_mask) {
 #pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      const int key = n0 + (j & 3) + 8 * (j >> 2) + mrow_base;
-      const bool dead = (causal && key > q_row) || key >= S || !q_valid;
-      s_val[j] = dead ? NEG_INF : st_acc[j] * scale_log2;
+      for (int j = 0; j < 16; ++j) {
+        const int key = n0 + (j & 3) + 8 * (j >> 2) + mrow_base;
+        const bool dead = (causal && key > q_row) || key >= S || !q_valid;
+        s_val[j] = dead ? NEG_INF : st_acc[j] * scale_log2;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 16; ++j) s_val[j] = st_acc[j] * scale_log2;
     }
 
     // ---- online softmax with defer-max (T13)
@@ -223,6 +234,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
                     __builtin_bit_cast(bf16x8, pw1)};
 
     // ---- O += P . V   (B[k][n=d] = one b128 from transposed V)
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt) {
       const int d0 = dt * 32 + (lane & 31);
@@ -230,11 +242,12 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       for (int kstep = 0; kstep < 2; ++kstep) {
         bf16x8 vf = __builtin_bit_cast(
             bf16x8, *(const ushortx8*)(
-                        &v_lds[v_lds_off(d0, kstep * 16 + (lane >> 5) * 8)]));
+                        &v_lds[buf][v_lds_off(d0, kstep * 16 + (lane >> 5) * 8)]));
         o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf[kstep], vf,
                                                             o_acc[dt], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
   }
 
   // ---- epilogue: O /= l, write bf16; LSE in log2 domain
