@@ -51,7 +51,8 @@ async def run_head(args):
     await raylet.start(0)
     # warm the worker pool
     try:
-        await raylet.rpc_prestart_workers(None, {"n": min(int(ncpu), args.prestart)})
+        n_pre = args.prestart if args.prestart > 0 else int(ncpu)
+        await raylet.rpc_prestart_workers(None, {"n": min(int(ncpu), n_pre)})
     except Exception:
         pass
 
@@ -78,7 +79,7 @@ def main():
     ap.add_argument("--resources", default="")
     ap.add_argument("--object-store-memory", type=int, default=0)
     ap.add_argument("--session-dir", default="")
-    ap.add_argument("--prestart", type=int, default=4)
+    ap.add_argument("--prestart", type=int, default=0)
     args = ap.parse_args()
     if not args.session_dir:
         args.session_dir = os.path.join(

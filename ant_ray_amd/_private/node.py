@@ -38,7 +38,7 @@ def start_head(
     host: str = "127.0.0.1",
     port: int = 0,
     session_dir: str = "",
-    prestart: int = 4,
+    prestart: int = 0,  # 0 = one per CPU
     timeout: float = 60.0,
 ) -> HeadProcess:
     if not session_dir:

@@ -52,7 +52,7 @@ LOCAL_MODE = "local"
 DRIVER_MODE = "driver"
 WORKER_MODE = "worker"
 
-PIPELINE_DEPTH = 4          # tasks pushed per leased worker before waiting
+PIPELINE_DEPTH = 8          # tasks pushed per leased worker before waiting
 LEASE_IDLE_RELEASE_S = 2.0  # return leased workers after this idle time
 
 
@@ -487,7 +487,7 @@ class CoreWorker:
         if opts.get("num_gpus"):
             resources["GPU"] = float(opts["num_gpus"])
         key = self._scheduling_key(fn_id, {"resources": resources})
-        self.io.submit(self._enqueue_task(key, payload, resources, opts)).result()
+        self.io.submit(self._enqueue_task(key, payload, resources, opts))  # fire-and-forget: refs are pre-created, failures land on them
         return refs
 
     async def _enqueue_task(self, key, payload, resources, opts):
@@ -706,7 +706,7 @@ class CoreWorker:
             "concurrency_group": opts.get("concurrency_group"),
             "tensor_transport": opts.get("tensor_transport"),
         }
-        self.io.submit(self._submit_actor_async(st, payload)).result()
+        self.io.submit(self._submit_actor_async(st, payload))  # fire-and-forget (per-actor seq assigned before this call keeps ordering)
         return refs
 
     async def _submit_actor_async(self, st: ActorHandleState, payload):
