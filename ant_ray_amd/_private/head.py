@@ -51,7 +51,9 @@ async def run_head(args):
     await raylet.start(0)
     # warm the worker pool
     try:
-        n_pre = args.prestart if args.prestart > 0 else int(ncpu)
+        # one worker per CPU up to a spawn-storm cap (large hosts fork the
+        # rest on demand)
+        n_pre = args.prestart if args.prestart > 0 else min(int(ncpu), 16)
         await raylet.rpc_prestart_workers(None, {"n": min(int(ncpu), n_pre)})
     except Exception:
         pass

@@ -24,6 +24,16 @@ void launch_bf16_scale(void*, float, int64_t, hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
                      int, int, int, int, long, long, long, long, long, long,
                      long, long, long, float, int, void*);
+void launch_attn_bwd_preprocess(const void*, const void*, float*, long, void*);
+void launch_attn_bwd_dq(const void*, const void*, const void*, const void*,
+                        const float*, const float*, void*, int, int, int, int,
+                        long, long, long, long, long, long, float, int, void*);
+void launch_attn_bwd_dkv(const void*, const void*, const void*, const void*,
+                         const float*, const float*, void*, void*, int, int,
+                         int, int, long, long, long, long, long, long, float,
+                         int, void*);
+void launch_attn_bwd_reduce_kv(const void*, const void*, void*, void*, int,
+                               int, int, int, void*);
 void launch_bf16_to_f32(const void*, float*, int64_t, hipStream_t);
 void launch_f32_to_bf16(const float*, void*, int64_t, hipStream_t);
 }
@@ -226,6 +236,64 @@ std::tuple<torch::Tensor, torch::Tensor> attn_fwd(torch::Tensor q,
   return {o, lse};
 }
 
+
+// Flash-attention backward. Inputs as attn_fwd plus o (fwd output), lse
+// (log2 domain from attn_fwd(need_lse=True)) and grad dO. Returns
+// (dq, dk, dv) in [B,Hq,S,D]/[B,Hk,S,D] contiguous bf16.
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> attn_bwd(
+    torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
+    torch::Tensor o, torch::Tensor lse, double scale, bool causal) {
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == 128, "attn_bwd: q [B,Hq,S,128]");
+  int B = q.size(0), Hq = q.size(1), S = q.size(2);
+  int Hk = k.size(1);
+  TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1,
+              "innermost dim must be contiguous");
+  TORCH_CHECK(k.strides() == v.strides(), "k/v must share layout");
+  auto dc = dout.contiguous();
+  auto oc = o.contiguous();
+  auto stream = cur_stream();
+  long rows = (long)B * Hq * S;
+  auto delta = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat32));
+  launch_attn_bwd_preprocess(dc.data_ptr(), oc.data_ptr(),
+                             delta.data_ptr<float>(), rows, (void*)stream);
+  auto dq = torch::empty({B, Hq, S, 128}, q.options());
+  // dO/o are contiguous [B,Hq,S,D]; q may be strided (fused-qkv view), so
+  // the dq kernel takes q's strides but reads dO at the same logical index
+  // through contiguous strides. For simplicity require q contiguous layout
+  // == dO layout when strided; else fall back to a contiguous copy of q.
+  torch::Tensor qq = q, kk = k, vv = v;
+  if (!(q.stride(0) == dc.stride(0) && q.stride(1) == dc.stride(1) &&
+        q.stride(2) == dc.stride(2))) {
+    qq = q.contiguous();
+  }
+  if (qq.strides() != dc.strides()) qq = q.contiguous();
+  launch_attn_bwd_dq(qq.data_ptr(), kk.data_ptr(), vv.data_ptr(),
+                     dc.data_ptr(), lse.data_ptr<float>(),
+                     delta.data_ptr<float>(), dq.data_ptr(), B, S, Hq, Hk,
+                     qq.stride(0), qq.stride(1), qq.stride(2), kk.stride(0),
+                     kk.stride(1), kk.stride(2), (float)scale, causal ? 1 : 0,
+                     (void*)stream);
+  auto dkp = torch::empty({B, Hq, S, 128}, q.options());
+  auto dvp = torch::empty({B, Hq, S, 128}, q.options());
+  launch_attn_bwd_dkv(qq.data_ptr(), kk.data_ptr(), vv.data_ptr(),
+                      dc.data_ptr(), lse.data_ptr<float>(),
+                      delta.data_ptr<float>(), dkp.data_ptr(), dvp.data_ptr(),
+                      B, S, Hq, Hk, qq.stride(0), qq.stride(1), qq.stride(2),
+                      kk.stride(0), kk.stride(1), kk.stride(2), (float)scale,
+                      causal ? 1 : 0, (void*)stream);
+  torch::Tensor dk, dv;
+  if (Hq == Hk) {
+    dk = dkp;
+    dv = dvp;
+  } else {
+    dk = torch::empty({B, Hk, S, 128}, q.options());
+    dv = torch::empty({B, Hk, S, 128}, q.options());
+    launch_attn_bwd_reduce_kv(dkp.data_ptr(), dvp.data_ptr(), dk.data_ptr(),
+                              dv.data_ptr(), B, Hq, Hk, S, (void*)stream);
+  }
+  return {dq, dk, dv};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused (add+)rmsnorm forward",
         py::arg("x"), py::arg("w"), py::arg("residual") = py::none(),
@@ -242,6 +310,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd, py::arg("q"), py::arg("k"), py::arg("v"),
         py::arg("scale"), py::arg("causal") = true,
         py::arg("need_lse") = false);
+  m.def("attn_bwd", &attn_bwd, py::arg("dout"), py::arg("q"), py::arg("k"),
+        py::arg("v"), py::arg("o"), py::arg("lse"), py::arg("scale"),
+        py::arg("causal") = true);
   m.def("bf16_scale_", &bf16_scale_);
   m.def("bf16_to_f32", &bf16_to_f32);
   m.def("f32_to_bf16_", &f32_to_bf16_);

@@ -149,6 +149,26 @@ def swiglu(gate_up: torch.Tensor) -> torch.Tensor:
 # -------------------------------------------------------------- attention
 
 
+class _AttentionFn(torch.autograd.Function):
+    """Training path: custom flash fwd saving (q,k,v,o,lse); bwd runs the
+    hand-written dq/dkv kernels (csrc/kernels/attention_bwd.hip)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        o, lse = _hip().attn_fwd(q, k, v, float(scale), causal, True)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal = causal
+        ctx.scale = float(scale)
+        return o
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = _hip().attn_bwd(dout, q, k, v, o, lse, ctx.scale,
+                                     ctx.causal)
+        return dq, dk, dv, None, None
+
+
 def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
               causal: bool = True, scale=None,
               need_lse: bool = False):
@@ -159,6 +179,11 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     if scale is None:
         scale = q.shape[-1] ** -0.5
     if _use_hip(q):
+        if torch.is_grad_enabled() and (q.requires_grad or k.requires_grad
+                                        or v.requires_grad):
+            if need_lse:
+                raise ValueError("need_lse not supported on the grad path")
+            return _AttentionFn.apply(q, k, v, causal, float(scale))
         o, lse = _hip().attn_fwd(q, k, v, float(scale), causal, need_lse)
         return (o, lse) if need_lse else o
     o = ref.attention_ref(q, k, v, causal, float(scale))

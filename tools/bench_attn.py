@@ -39,7 +39,32 @@ def main():
     t_sdpa = bench(lambda: torch.nn.functional.scaled_dot_product_attention(
         q, k, v, is_causal=True, enable_gqa=True))
     print(f"torch sdpa: {t_sdpa*1e3:8.2f} ms  {flops/t_sdpa/1e12:7.1f} TF/s")
-    print(f"speedup: {t_sdpa/t_ours:.2f}x")
+    print(f"fwd speedup: {t_sdpa/t_ours:.2f}x")
+
+    # ---- fwd+bwd
+    bwd_flops = flops * 3.5  # S recomputed twice + 4 bwd gemms + fwd
+    qg = q.clone().requires_grad_()
+    kg = k.clone().requires_grad_()
+    vg = v.clone().requires_grad_()
+    dout = torch.randn_like(q)
+
+    def ours_fb():
+        out = ops.attention(qg, kg, vg, causal=True)
+        out.backward(dout)
+        qg.grad = kg.grad = vg.grad = None
+
+    t_ours_fb = bench(ours_fb, iters=10)
+    print(f"ours f+b  : {t_ours_fb*1e3:8.2f} ms  {bwd_flops/t_ours_fb/1e12:7.1f} TF/s")
+
+    def sdpa_fb():
+        out = torch.nn.functional.scaled_dot_product_attention(
+            qg, kg, vg, is_causal=True, enable_gqa=True)
+        out.backward(dout)
+        qg.grad = kg.grad = vg.grad = None
+
+    t_sdpa_fb = bench(sdpa_fb, iters=10)
+    print(f"sdpa f+b  : {t_sdpa_fb*1e3:8.2f} ms  {bwd_flops/t_sdpa_fb/1e12:7.1f} TF/s")
+    print(f"f+b speedup: {t_sdpa_fb/t_ours_fb:.2f}x")
 
 
 if __name__ == "__main__":

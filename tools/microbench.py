@@ -92,6 +92,8 @@ def main():
     timeit("1:1 actor calls concurrent",
            lambda: ray.get([conc.ping.remote() for _ in range(1000)]), 1000, dur)
 
+    ray.kill(a)
+    ray.kill(conc)
     n_cpu = min(os.cpu_count() or 4, 8)
     actors = [Actor.remote() for _ in range(n_cpu)]
     ray.get([x.ping.remote() for x in actors])
@@ -105,6 +107,8 @@ def main():
         async def ping(self):
             return b"ok"
 
+    for x in actors:
+        ray.kill(x)
     aa = AsyncActor.remote()
     ray.get(aa.ping.remote())
     timeit("1:1 async-actor calls sync",
@@ -113,6 +117,9 @@ def main():
            lambda: ray.get([aa.ping.remote() for _ in range(1000)]), 1000, dur)
 
     # ---- placement groups
+    ray.kill(aa)
+    import time as _t
+    _t.sleep(1.0)  # let killed actors release CPUs
     from ant_ray_amd.util.placement_group import (
         placement_group,
         remove_placement_group,
