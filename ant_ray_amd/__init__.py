@@ -31,7 +31,7 @@ __all__ = [
     "kill", "cancel", "get_actor", "get_gpu_ids", "get_runtime_context",
     "nodes", "cluster_resources", "available_resources", "method",
     "ObjectRef", "ActorHandle", "exceptions", "actor", "remote_function",
-    "available_resources_per_node", "util", "train", "serve", "data", "tune",
+    "util", "train", "serve", "data", "tune",
 ]
 
 

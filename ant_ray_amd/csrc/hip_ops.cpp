@@ -27,11 +27,12 @@ void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
 void launch_attn_bwd_preprocess(const void*, const void*, float*, long, void*);
 void launch_attn_bwd_dq(const void*, const void*, const void*, const void*,
                         const float*, const float*, void*, int, int, int, int,
-                        long, long, long, long, long, long, float, int, void*);
+                        long, long, long, long, long, long, long, long, long,
+                        float, int, void*);
 void launch_attn_bwd_dkv(const void*, const void*, const void*, const void*,
                          const float*, const float*, void*, void*, int, int,
-                         int, int, long, long, long, long, long, long, float,
-                         int, void*);
+                         int, int, long, long, long, long, long, long, long,
+                         long, long, float, int, void*);
 void launch_attn_bwd_reduce_kv(const void*, const void*, void*, void*, int,
                                int, int, int, void*);
 void launch_bf16_to_f32(const void*, float*, int64_t, hipStream_t);
@@ -257,29 +258,21 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> attn_bwd(
   launch_attn_bwd_preprocess(dc.data_ptr(), oc.data_ptr(),
                              delta.data_ptr<float>(), rows, (void*)stream);
   auto dq = torch::empty({B, Hq, S, 128}, q.options());
-  // dO/o are contiguous [B,Hq,S,D]; q may be strided (fused-qkv view), so
-  // the dq kernel takes q's strides but reads dO at the same logical index
-  // through contiguous strides. For simplicity require q contiguous layout
-  // == dO layout when strided; else fall back to a contiguous copy of q.
-  torch::Tensor qq = q, kk = k, vv = v;
-  if (!(q.stride(0) == dc.stride(0) && q.stride(1) == dc.stride(1) &&
-        q.stride(2) == dc.stride(2))) {
-    qq = q.contiguous();
-  }
-  if (qq.strides() != dc.strides()) qq = q.contiguous();
-  launch_attn_bwd_dq(qq.data_ptr(), kk.data_ptr(), vv.data_ptr(),
+  launch_attn_bwd_dq(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                      dc.data_ptr(), lse.data_ptr<float>(),
                      delta.data_ptr<float>(), dq.data_ptr(), B, S, Hq, Hk,
-                     qq.stride(0), qq.stride(1), qq.stride(2), kk.stride(0),
-                     kk.stride(1), kk.stride(2), (float)scale, causal ? 1 : 0,
+                     q.stride(0), q.stride(1), q.stride(2), k.stride(0),
+                     k.stride(1), k.stride(2), dc.stride(0), dc.stride(1),
+                     dc.stride(2), (float)scale, causal ? 1 : 0,
                      (void*)stream);
   auto dkp = torch::empty({B, Hq, S, 128}, q.options());
   auto dvp = torch::empty({B, Hq, S, 128}, q.options());
-  launch_attn_bwd_dkv(qq.data_ptr(), kk.data_ptr(), vv.data_ptr(),
+  launch_attn_bwd_dkv(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                       dc.data_ptr(), lse.data_ptr<float>(),
                       delta.data_ptr<float>(), dkp.data_ptr(), dvp.data_ptr(),
-                      B, S, Hq, Hk, qq.stride(0), qq.stride(1), qq.stride(2),
-                      kk.stride(0), kk.stride(1), kk.stride(2), (float)scale,
+                      B, S, Hq, Hk, q.stride(0), q.stride(1), q.stride(2),
+                      k.stride(0), k.stride(1), k.stride(2), dc.stride(0),
+                      dc.stride(1), dc.stride(2), (float)scale,
                       causal ? 1 : 0, (void*)stream);
   torch::Tensor dk, dv;
   if (Hq == Hk) {

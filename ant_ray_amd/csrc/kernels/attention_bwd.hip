@@ -81,8 +81,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     const float* __restrict__ Delta,  // [B,Hq,S]
     ushort_t* __restrict__ dQ,        // [B,Hq,S,D] contiguous
     int S, int Hq, int Hk,
-    long qb, long qh, long qs,        // Q/dO/dQ strides (elements)
+    long qb, long qh, long qs,        // Q strides (elements)
     long kb, long kh, long ks,        // K/V strides
+    long ob, long oh, long os,        // dO/dQ (contiguous) strides
     float scale, int causal) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -97,7 +98,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   const bool q_valid = q_row < S;
 
   const ushort_t* Qp = Q + (long)b * qb + (long)hq * qh;
-  const ushort_t* dOp = dO + (long)b * qb + (long)hq * qh;
+  const ushort_t* dOp = dO + (long)b * ob + (long)hq * oh;
   const ushort_t* Kp = K + (long)b * kb + (long)hk * kh;
   const ushort_t* Vp = V + (long)b * kb + (long)hk * kh;
   const float* Lp = LSE + ((long)b * Hq + hq) * S;
@@ -111,13 +112,14 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   bf16x8 qf[8], dof[8];
   {
     const long r = (long)(q_valid ? q_row : 0) * qs;
+    const long ro = (long)(q_valid ? q_row : 0) * os;
     const int dbase = (lane >> 5) * 8;
 #pragma unroll
     for (int st = 0; st < 8; ++st) {
       qf[st] = __builtin_bit_cast(bf16x8,
                                   *(const ushortx8*)(Qp + r + st * 16 + dbase));
       dof[st] = __builtin_bit_cast(
-          bf16x8, *(const ushortx8*)(dOp + r + st * 16 + dbase));
+          bf16x8, *(const ushortx8*)(dOp + ro + st * 16 + dbase));
     }
   }
   const float L_q = q_valid ? Lp[q_row] : 0.f;
@@ -221,57 +223,191 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   }
 
   // epilogue: write dQ (C: m = q reg-spread, n = d lane)
-  ushort_t* dQp = dQ + (long)b * qb + (long)hq * qh;
+  ushort_t* dQp = dQ + (long)b * ob + (long)hq * oh;
 #pragma unroll
   for (int j = 0; j < 16; ++j) {
     const int qr = q0 + (j & 3) + 8 * (j >> 2) + 4 * (lane >> 5);
     if (qr >= S) continue;
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt)
-      dQp[(long)qr * qs + dt * 32 + (lane & 31)] = f2bf(dq_acc[dt][j]);
+      dQp[(long)qr * os + dt * 32 + (lane & 31)] = f2bf(dq_acc[dt][j]);
   }
 }
 
-// ----------------------------------------------------------- dKV kernel
+// ------------------------------------------------------ dV / dK kernels
+//
+// Split (instead of one fused dKV kernel) so each stays under the 256-VGPR
+// occupancy-2 line: the fused version needed 128 accumulator VGPRs + two
+// C tiles and spilled. dV re-derives P from (S^T, lse); dK re-derives S^T
+// and dP^T. +25% MFMA vs fused, but 2 waves/SIMD instead of 1 hides the
+// LDS/HBM latency that dominated the fused kernel (PMC: 37% WAIT_ANY).
 
-extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dkv_kernel(
+extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
     const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
-    const ushort_t* __restrict__ V, const ushort_t* __restrict__ dO,
-    const float* __restrict__ LSE, const float* __restrict__ Delta,
-    ushort_t* __restrict__ dKp_out,  // [B,Hq,S,D] per-Q-head partials
-    ushort_t* __restrict__ dVp_out,  // [B,Hq,S,D]
+    const ushort_t* __restrict__ dO,
+    const float* __restrict__ LSE,
+    ushort_t* __restrict__ dVp_out,  // [B,Hq,S,D] per-Q-head partials
     int S, int Hq, int Hk,
     long qb, long qh, long qs,
     long kb, long kh, long ks,
+    long ob, long oh, long os,       // dO (contiguous) strides
     float scale, int causal) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int k_block = blockIdx.x;   // 128 keys per workgroup
+  const int k_block = blockIdx.x;
   const int hq = blockIdx.y;
   const int b = blockIdx.z;
   const int hk = hq / (Hq / Hk);
   const float c_log2 = scale * 1.4426950408889634f;
 
-  const int key0 = k_block * 128 + wave * 32;  // this wave's first key
+  const int key0 = k_block * 128 + wave * 32;
   const int key_row = key0 + (lane & 31);
 
   const ushort_t* Qp = Q + (long)b * qb + (long)hq * qh;
-  const ushort_t* dOp = dO + (long)b * qb + (long)hq * qh;
+  const ushort_t* dOp = dO + (long)b * ob + (long)hq * oh;
+  const ushort_t* Kp = K + (long)b * kb + (long)hk * kh;
+  const float* Lp = LSE + ((long)b * Hq + hq) * S;
+
+  __shared__ ushort_t q_rm[32 * KROW];
+  __shared__ ushort_t do_tr[ATTN_D * TROW];
+  __shared__ ushort_t scr_p[4][32 * SCR_ROW];
+
+  // own K rows as A-fragments in registers
+  bf16x8 kfr[8];
+  {
+    const long kg = (long)min(key_row, S - 1) * ks;
+    const int dbase = (lane >> 5) * 8;
+#pragma unroll
+    for (int st = 0; st < 8; ++st)
+      kfr[st] = __builtin_bit_cast(
+          bf16x8, *(const ushortx8*)(Kp + kg + st * 16 + dbase));
+  }
+
+  floatx16 dv_acc[4] = {};
+  const int q_start_tile = causal ? (k_block * 128) / BLOCK : 0;
+  const int n_q_tiles = (S + BLOCK - 1) / BLOCK;
+  const int st_row = threadIdx.x >> 3;
+  const int st_col = (threadIdx.x & 7) * 16;
+
+  ushortx8 qa, qa2, da, da2;
+  {
+    const int qg = min(q_start_tile * BLOCK + st_row, S - 1);
+    qa = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
+    qa2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
+    da = *(const ushortx8*)(dOp + (long)qg * os + st_col);
+    da2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
+  }
+
+  for (int t = q_start_tile; t < n_q_tiles; ++t) {
+    const int tq0 = t * BLOCK;
+    __syncthreads();
+    *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8)]) = qa;
+    *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8 + 1)]) = qa2;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      do_tr[tr_off(st_col + j, st_row)] = da[j];
+      do_tr[tr_off(st_col + 8 + j, st_row)] = da2[j];
+    }
+    __syncthreads();
+    if (t + 1 < n_q_tiles) {
+      const int qg = min((t + 1) * BLOCK + st_row, S - 1);
+      qa = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
+      qa2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
+      da = *(const ushortx8*)(dOp + (long)qg * os + st_col);
+      da2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
+    }
+
+    const int q_here = tq0 + (lane & 31);
+    const bool qv = q_here < S;
+    const float L_q = qv ? Lp[q_here] : 0.f;
+
+    floatx16 st_acc = {};
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int stp = 0; stp < 8; ++stp) {
+      bf16x8 qfr = __builtin_bit_cast(
+          bf16x8,
+          *(const ushortx8*)(&q_rm[rm_off(lane & 31, 2 * stp + (lane >> 5))]));
+      st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr[stp], qfr, st_acc,
+                                                       0, 0, 0);
+    }
+    __builtin_amdgcn_s_setprio(0);
+
+    const int mrow_base = 4 * (lane >> 5);
+    ushort_t* my_scr = (ushort_t*)scr_p[wave];
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      const int key = key0 + (j & 3) + 8 * (j >> 2) + mrow_base;
+      const bool dead = (causal && key > q_here) || key >= S || !qv;
+      const float p = dead ? 0.f : __builtin_exp2f(st_acc[j] * c_log2 - L_q);
+      const int krow = (j & 3) + 8 * (j >> 2) + mrow_base;
+      my_scr[krow * SCR_ROW + (lane & 31)] = f2bf(p);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt) {
+      const int d0 = dt * 32 + (lane & 31);
+#pragma unroll
+      for (int kstep = 0; kstep < 2; ++kstep) {
+        bf16x8 pa = __builtin_bit_cast(
+            bf16x8, *(const ushortx8*)(
+                        &my_scr[(lane & 31) * SCR_ROW + kstep * 16 +
+                                (lane >> 5) * 8]));
+        bf16x8 dob = __builtin_bit_cast(
+            bf16x8, *(const ushortx8*)(
+                        &do_tr[tr_off(d0, kstep * 16 + (lane >> 5) * 8)]));
+        dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob,
+                                                             dv_acc[dt], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+  }
+
+  ushort_t* dVh = dVp_out + ((long)b * Hq + hq) * (long)S * ATTN_D;
+#pragma unroll
+  for (int j = 0; j < 16; ++j) {
+    const int kr = key0 + (j & 3) + 8 * (j >> 2) + 4 * (lane >> 5);
+    if (kr >= S) continue;
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt)
+      dVh[(long)kr * ATTN_D + dt * 32 + (lane & 31)] = f2bf(dv_acc[dt][j]);
+  }
+}
+
+extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
+    const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
+    const ushort_t* __restrict__ V, const ushort_t* __restrict__ dO,
+    const float* __restrict__ LSE, const float* __restrict__ Delta,
+    ushort_t* __restrict__ dKp_out,  // [B,Hq,S,D] per-Q-head partials
+    int S, int Hq, int Hk,
+    long qb, long qh, long qs,
+    long kb, long kh, long ks,
+    long ob, long oh, long os,       // dO (contiguous) strides
+    float scale, int causal) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int k_block = blockIdx.x;
+  const int hq = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hk = hq / (Hq / Hk);
+  const float c_log2 = scale * 1.4426950408889634f;
+
+  const int key0 = k_block * 128 + wave * 32;
+  const int key_row = key0 + (lane & 31);
+
+  const ushort_t* Qp = Q + (long)b * qb + (long)hq * qh;
+  const ushort_t* dOp = dO + (long)b * ob + (long)hq * oh;
   const ushort_t* Kp = K + (long)b * kb + (long)hk * kh;
   const ushort_t* Vp = V + (long)b * kb + (long)hk * kh;
   const float* Lp = LSE + ((long)b * Hq + hq) * S;
   const float* Dp = Delta + ((long)b * Hq + hq) * S;
 
-  __shared__ ushort_t q_rm[32 * KROW];    // per q-tile
+  __shared__ ushort_t q_rm[32 * KROW];
   __shared__ ushort_t do_rm[32 * KROW];
   __shared__ ushort_t q_tr[ATTN_D * TROW];
-  __shared__ ushort_t do_tr[ATTN_D * TROW];
-  __shared__ ushort_t scr_p[4][32 * SCR_ROW];   // wave-private P round-trip
-  __shared__ ushort_t scr_ds[4][32 * SCR_ROW];  // wave-private dS round-trip
+  __shared__ ushort_t scr_ds[4][32 * SCR_ROW];
 
-  // the wave's 32 K and V rows live as A-fragments in REGISTERS for the
-  // whole kernel (A[m=own key = l&31][kdim = d slices]): 32+32 VGPRs buys
-  // back 64 KB of LDS -> 2 workgroups/CU instead of 1
   bf16x8 kfr[8], vfr[8];
   {
     const long kg = (long)min(key_row, S - 1) * ks;
@@ -285,55 +421,38 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dkv_kernel(
     }
   }
 
-  floatx16 dv_acc[4] = {}, dk_acc[4] = {};
-
+  floatx16 dk_acc[4] = {};
   const int q_start_tile = causal ? (k_block * 128) / BLOCK : 0;
   const int n_q_tiles = (S + BLOCK - 1) / BLOCK;
-
   const int st_row = threadIdx.x >> 3;
   const int st_col = (threadIdx.x & 7) * 16;
-
-  // T14: first tile's Q/dO loads issued before the loop
-  ushortx8 qa, qa2, da, da2;
-  {
-    const int qg = min(q_start_tile * BLOCK + st_row, S - 1);
-    qa = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
-    qa2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
-    da = *(const ushortx8*)(dOp + (long)qg * qs + st_col);
-    da2 = *(const ushortx8*)(dOp + (long)qg * qs + st_col + 8);
-  }
 
   for (int t = q_start_tile; t < n_q_tiles; ++t) {
     const int tq0 = t * BLOCK;
     __syncthreads();
     {
-      *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8)]) = qa;
-      *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8 + 1)]) = qa2;
-      *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8)]) = da;
-      *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8 + 1)]) = da2;
+      const int qg = min(tq0 + st_row, S - 1);
+      const ushortx8 a = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
+      const ushortx8 a2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
+      const ushortx8 d = *(const ushortx8*)(dOp + (long)qg * os + st_col);
+      const ushortx8 d2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
+      *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8)]) = a;
+      *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8 + 1)]) = a2;
+      *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8)]) = d;
+      *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8 + 1)]) = d2;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        q_tr[tr_off(st_col + j, st_row)] = qa[j];
-        q_tr[tr_off(st_col + 8 + j, st_row)] = qa2[j];
-        do_tr[tr_off(st_col + j, st_row)] = da[j];
-        do_tr[tr_off(st_col + 8 + j, st_row)] = da2[j];
+        q_tr[tr_off(st_col + j, st_row)] = a[j];
+        q_tr[tr_off(st_col + 8 + j, st_row)] = a2[j];
       }
     }
     __syncthreads();
-    if (t + 1 < n_q_tiles) {
-      const int qg = min((t + 1) * BLOCK + st_row, S - 1);
-      qa = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
-      qa2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
-      da = *(const ushortx8*)(dOp + (long)qg * qs + st_col);
-      da2 = *(const ushortx8*)(dOp + (long)qg * qs + st_col + 8);
-    }
 
     const int q_here = tq0 + (lane & 31);
     const bool qv = q_here < S;
     const float L_q = qv ? Lp[q_here] : 0.f;
     const float D_q = qv ? Dp[q_here] : 0.f;
 
-    // S^T = K.Q^T, dP^T = V.dO^T  (C: n = q lane-local, m = key offset)
     floatx16 st_acc = {}, dp_acc = {};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -350,48 +469,31 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dkv_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // P^T and dS^T in C layout; write both to the wave scratch as bf16
-    // ([key][q] rows so the re-read A-frag is key-lane-local)
     const int mrow_base = 4 * (lane >> 5);
-    float p_val[16];
+    ushort_t* my_scr = (ushort_t*)scr_ds[wave];
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
       const int key = key0 + (j & 3) + 8 * (j >> 2) + mrow_base;
       const bool dead = (causal && key > q_here) || key >= S || !qv;
-      p_val[j] = dead ? 0.f : __builtin_exp2f(st_acc[j] * c_log2 - L_q);
-    }
-    // scratch layout: rows = key, cols = q. Both tiles written up front
-    // into separate buffers; ONE lgkm drain covers both MFMA blocks below.
-    ushort_t* my_p = (ushort_t*)scr_p[wave];
-    ushort_t* my_ds = (ushort_t*)scr_ds[wave];
-#pragma unroll
-    for (int j = 0; j < 16; ++j) {
+      const float p = dead ? 0.f : __builtin_exp2f(st_acc[j] * c_log2 - L_q);
       const int krow = (j & 3) + 8 * (j >> 2) + mrow_base;
-      my_p[krow * SCR_ROW + (lane & 31)] = f2bf(p_val[j]);
-      my_ds[krow * SCR_ROW + (lane & 31)] =
-          f2bf(scale * p_val[j] * (dp_acc[j] - D_q));
+      my_scr[krow * SCR_ROW + (lane & 31)] =
+          f2bf(scale * p * (dp_acc[j] - D_q));
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-    // dV += P^T . dO ; dK += dS^T . Q   (A = scratch rows, B = transposed)
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt) {
       const int d0 = dt * 32 + (lane & 31);
 #pragma unroll
       for (int kstep = 0; kstep < 2; ++kstep) {
-        const int scr_off = (lane & 31) * SCR_ROW + kstep * 16 +
-                            (lane >> 5) * 8;
-        const int tr = tr_off(d0, kstep * 16 + (lane >> 5) * 8);
-        bf16x8 pa = __builtin_bit_cast(bf16x8,
-                                       *(const ushortx8*)(&my_p[scr_off]));
-        bf16x8 dob = __builtin_bit_cast(bf16x8,
-                                        *(const ushortx8*)(&do_tr[tr]));
-        dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob,
-                                                             dv_acc[dt], 0, 0, 0);
-        bf16x8 dsa = __builtin_bit_cast(bf16x8,
-                                        *(const ushortx8*)(&my_ds[scr_off]));
-        bf16x8 qtb = __builtin_bit_cast(bf16x8,
-                                        *(const ushortx8*)(&q_tr[tr]));
+        bf16x8 dsa = __builtin_bit_cast(
+            bf16x8, *(const ushortx8*)(
+                        &my_scr[(lane & 31) * SCR_ROW + kstep * 16 +
+                                (lane >> 5) * 8]));
+        bf16x8 qtb = __builtin_bit_cast(
+            bf16x8, *(const ushortx8*)(
+                        &q_tr[tr_off(d0, kstep * 16 + (lane >> 5) * 8)]));
         dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, qtb,
                                                              dk_acc[dt], 0, 0, 0);
       }
@@ -399,18 +501,14 @@ extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dkv_kernel(
     __builtin_amdgcn_s_setprio(0);
   }
 
-  // epilogue: write per-head dK/dV partials (C: m = key reg, n = d lane)
   ushort_t* dKh = dKp_out + ((long)b * Hq + hq) * (long)S * ATTN_D;
-  ushort_t* dVh = dVp_out + ((long)b * Hq + hq) * (long)S * ATTN_D;
 #pragma unroll
   for (int j = 0; j < 16; ++j) {
     const int kr = key0 + (j & 3) + 8 * (j >> 2) + 4 * (lane >> 5);
     if (kr >= S) continue;
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
+    for (int dt = 0; dt < 4; ++dt)
       dKh[(long)kr * ATTN_D + dt * 32 + (lane & 31)] = f2bf(dk_acc[dt][j]);
-      dVh[(long)kr * ATTN_D + dt * 32 + (lane & 31)] = f2bf(dv_acc[dt][j]);
-    }
   }
 }
 
@@ -459,14 +557,15 @@ extern "C" void launch_attn_bwd_dq(const void* q, const void* k, const void* v,
                                    const void* dO, const float* lse,
                                    const float* delta, void* dq, int B, int S,
                                    int Hq, int Hk, long qb, long qh, long qs,
-                                   long kb, long kh, long ks, float scale,
+                                   long kb, long kh, long ks, long ob,
+                                   long oh, long os, float scale,
                                    int causal, void* stream) {
   dim3 grid((S + 127) / 128, Hq, B);
   hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, dim3(256), 0,
                      (hipStream_t)stream, (const ushort_t*)q,
                      (const ushort_t*)k, (const ushort_t*)v,
                      (const ushort_t*)dO, lse, delta, (ushort_t*)dq, S, Hq,
-                     Hk, qb, qh, qs, kb, kh, ks, scale, causal);
+                     Hk, qb, qh, qs, kb, kh, ks, ob, oh, os, scale, causal);
 }
 
 extern "C" void launch_attn_bwd_dkv(const void* q, const void* k,
@@ -474,15 +573,20 @@ extern "C" void launch_attn_bwd_dkv(const void* q, const void* k,
                                     const float* lse, const float* delta,
                                     void* dkp, void* dvp, int B, int S,
                                     int Hq, int Hk, long qb, long qh, long qs,
-                                    long kb, long kh, long ks, float scale,
+                                    long kb, long kh, long ks, long ob,
+                                    long oh, long os, float scale,
                                     int causal, void* stream) {
   dim3 grid((S + 127) / 128, Hq, B);
-  hipLaunchKernelGGL(attn_bwd_dkv_kernel, grid, dim3(256), 0,
+  hipLaunchKernelGGL(attn_bwd_dv_kernel, grid, dim3(256), 0,
+                     (hipStream_t)stream, (const ushort_t*)q,
+                     (const ushort_t*)k, (const ushort_t*)dO, lse,
+                     (ushort_t*)dvp, S, Hq, Hk, qb, qh, qs, kb, kh, ks,
+                     ob, oh, os, scale, causal);
+  hipLaunchKernelGGL(attn_bwd_dk_kernel, grid, dim3(256), 0,
                      (hipStream_t)stream, (const ushort_t*)q,
                      (const ushort_t*)k, (const ushort_t*)v,
-                     (const ushort_t*)dO, lse, delta, (ushort_t*)dkp,
-                     (ushort_t*)dvp, S, Hq, Hk, qb, qh, qs, kb, kh, ks, scale,
-                     causal);
+                     (const ushort_t*)dO, lse, delta, (ushort_t*)dkp, S, Hq,
+                     Hk, qb, qh, qs, kb, kh, ks, ob, oh, os, scale, causal);
 }
 
 extern "C" void launch_attn_bwd_reduce_kv(const void* dkp, const void* dvp,

@@ -8,7 +8,8 @@ Design (not a port of any reference model code):
       - SwiGLU on the packed gate_up projection
       - chunked fused vocab-projection + cross-entropy (logits never
         materialize: 4.2 GB saved at B4xS4096xV128256)
-  * attention: torch SDPA (GQA-aware) — fused flash-style HIP attention is the
+  * attention: hand-written CDNA4 flash attention (ops.attention) on GPU for
+    D=128; torch SDPA elsewhere — was: the
     next kernel on the roadmap
   * parameters live as views into flat buffers when wrapped by
     parallel.FlatParamManager (optimizer = one fused kernel pass).
@@ -66,7 +67,13 @@ class LlamaAttention(nn.Module):
         q = qkv[..., : Hq * D].view(B, S, Hq, D).transpose(1, 2)
         k = qkv[..., Hq * D : (Hq + Hk) * D].view(B, S, Hk, D).transpose(1, 2)
         v = qkv[..., (Hq + Hk) * D :].view(B, S, Hk, D).transpose(1, 2)
-        o = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
+        if qkv.is_cuda and D == 128:
+            # hand-written CDNA4 flash attention (fwd: attention.hip,
+            # bwd: attention_bwd.hip) — takes the strided fused-qkv views
+            o = ops.attention(q, k, v, causal=True)
+        else:
+            o = F.scaled_dot_product_attention(q, k, v, is_causal=True,
+                                               enable_gqa=True)
         o = o.transpose(1, 2).reshape(B, S, Hq * D)
         return self.wo(o)
 
