@@ -19,6 +19,8 @@ from __future__ import annotations
 import math
 from dataclasses import dataclass
 
+import os
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
@@ -67,7 +69,7 @@ class LlamaAttention(nn.Module):
         q = qkv[..., : Hq * D].view(B, S, Hq, D).transpose(1, 2)
         k = qkv[..., Hq * D : (Hq + Hk) * D].view(B, S, Hk, D).transpose(1, 2)
         v = qkv[..., (Hq + Hk) * D :].view(B, S, Hk, D).transpose(1, 2)
-        if qkv.is_cuda and D == 128:
+        if qkv.is_cuda and D == 128 and os.environ.get("ANTRAY_FLASH", "1") != "0":
             # hand-written CDNA4 flash attention (fwd: attention.hip,
             # bwd: attention_bwd.hip) — takes the strided fused-qkv views
             o = ops.attention(q, k, v, causal=True)
