@@ -178,7 +178,7 @@ class Raylet:
         return await fn(conn, p or {})
 
     # --------------------------------------------------------------- workers
-    def _spawn_worker(self) -> WorkerProc:
+    def _spawn_worker(self, runtime_env: Optional[dict] = None) -> WorkerProc:
         worker_id = WorkerID.from_random().binary()
         log_dir = os.path.join(self.session_dir, "logs")
         os.makedirs(log_dir, exist_ok=True)
@@ -190,15 +190,28 @@ class Raylet:
         env["ANTRAY_NODE_ID"] = self.node_id.hex()
         env["ANTRAY_STORE"] = self.store_path
         env["ANTRAY_SESSION_DIR"] = self.session_dir
+        cmd = [sys.executable, "-m",
+               "ant_ray_amd._private.workers.default_worker"]
+        cwd = None
+        if runtime_env:
+            # runtime-env plugins applied at spawn (parity with the
+            # runtime-env agent's worker-command mutation,
+            # _private/runtime_env/: env_vars, working_dir, py_modules,
+            # rocprof_sys/nsight profiler prefixes)
+            from ant_ray_amd._private.runtime_env import build_worker_spawn
+
+            cmd, env, cwd = build_worker_spawn(cmd, env, runtime_env)
         proc = subprocess.Popen(
-            [sys.executable, "-m", "ant_ray_amd._private.workers.default_worker"],
+            cmd,
             env=env,
+            cwd=cwd,
             stdout=out,
             stderr=subprocess.STDOUT,
             start_new_session=True,
         )
         out.close()
         w = WorkerProc(worker_id, proc)
+        w.dedicated = bool(runtime_env)
         self.workers[worker_id] = w
         return w
 
@@ -351,16 +364,18 @@ class Raylet:
         res = dict(p.get("resources") or {})
         pg = p.get("pg")
         gpu_ids = self._take(res, pg)
+        renv = p.get("runtime_env") or {}
+        needs_dedicated = any(k != "env_vars" for k in renv)
         w = None
         while w is None:
-            if self.idle_workers:
+            if self.idle_workers and not needs_dedicated:
                 w = self.idle_workers.pop()
                 if w.proc.poll() is not None:
                     self.workers.pop(w.worker_id, None)
                     w = None
                     continue
             else:
-                w = self._spawn_worker()
+                w = self._spawn_worker(renv if needs_dedicated else None)
             try:
                 await asyncio.wait_for(w.registered.wait(), timeout=60)
             except asyncio.TimeoutError:
@@ -409,7 +424,8 @@ class Raylet:
         if w is None:
             return {"ok": False}
         self._release_resources(w)
-        kill = p.get("kill", False) or w.is_actor or w.gpu_ids
+        kill = (p.get("kill", False) or w.is_actor or w.gpu_ids
+                or getattr(w, "dedicated", False))
         if kill or w.proc.poll() is not None:
             self._kill_worker(w)
         else:
