@@ -265,25 +265,16 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> attn_bwd(
                      k.stride(1), k.stride(2), dc.stride(0), dc.stride(1),
                      dc.stride(2), (float)scale, causal ? 1 : 0,
                      (void*)stream);
-  auto dkp = torch::empty({B, Hq, S, 128}, q.options());
-  auto dvp = torch::empty({B, Hq, S, 128}, q.options());
+  // dK/dV written directly per KV head (the kernels loop the q-head group)
+  auto dk = torch::empty({B, Hk, S, 128}, q.options());
+  auto dv = torch::empty({B, Hk, S, 128}, q.options());
   launch_attn_bwd_dkv(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                       dc.data_ptr(), lse.data_ptr<float>(),
-                      delta.data_ptr<float>(), dkp.data_ptr(), dvp.data_ptr(),
+                      delta.data_ptr<float>(), dk.data_ptr(), dv.data_ptr(),
                       B, S, Hq, Hk, q.stride(0), q.stride(1), q.stride(2),
                       k.stride(0), k.stride(1), k.stride(2), dc.stride(0),
                       dc.stride(1), dc.stride(2), (float)scale,
                       causal ? 1 : 0, (void*)stream);
-  torch::Tensor dk, dv;
-  if (Hq == Hk) {
-    dk = dkp;
-    dv = dvp;
-  } else {
-    dk = torch::empty({B, Hk, S, 128}, q.options());
-    dv = torch::empty({B, Hk, S, 128}, q.options());
-    launch_attn_bwd_reduce_kv(dkp.data_ptr(), dvp.data_ptr(), dk.data_ptr(),
-                              dv.data_ptr(), B, Hq, Hk, S, (void*)stream);
-  }
   return {dq, dk, dv};
 }
 

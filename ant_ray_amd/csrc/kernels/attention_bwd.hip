@@ -246,7 +246,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
     const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
     const ushort_t* __restrict__ dO,
     const float* __restrict__ LSE,
-    ushort_t* __restrict__ dVp_out,  // [B,Hq,S,D] per-Q-head partials
+    ushort_t* __restrict__ dV_out,   // [B,Hk,S,D] direct (q-heads looped)
     int S, int Hq, int Hk,
     long qb, long qh, long qs,
     long kb, long kh, long ks,
@@ -255,18 +255,15 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int k_block = blockIdx.x;
-  const int hq = blockIdx.y;
+  const int hk = blockIdx.y;        // KV head: the q-head group is looped
   const int b = blockIdx.z;
-  const int hk = hq / (Hq / Hk);
+  const int group = Hq / Hk;
   const float c_log2 = scale * 1.4426950408889634f;
 
   const int key0 = k_block * 128 + wave * 32;
   const int key_row = key0 + (lane & 31);
 
-  const ushort_t* Qp = Q + (long)b * qb + (long)hq * qh;
-  const ushort_t* dOp = dO + (long)b * ob + (long)hq * oh;
   const ushort_t* Kp = K + (long)b * kb + (long)hk * kh;
-  const float* Lp = LSE + ((long)b * Hq + hq) * S;
 
   __shared__ ushort_t q_rm[32 * KROW];
   __shared__ ushort_t do_tr[ATTN_D * TROW];
@@ -289,6 +286,11 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
   const int st_row = threadIdx.x >> 3;
   const int st_col = (threadIdx.x & 7) * 16;
 
+  for (int g = 0; g < group; ++g) {
+  const int hq = hk * group + g;
+  const ushort_t* Qp = Q + (long)b * qb + (long)hq * qh;
+  const ushort_t* dOp = dO + (long)b * ob + (long)hq * oh;
+  const float* Lp = LSE + ((long)b * Hq + hq) * S;
   ushortx8 qa, qa2, da, da2;
   {
     const int qg = min(q_start_tile * BLOCK + st_row, S - 1);
@@ -363,8 +365,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
   }
+  }  // q-head group loop
 
-  ushort_t* dVh = dVp_out + ((long)b * Hq + hq) * (long)S * ATTN_D;
+  ushort_t* dVh = dV_out + ((long)b * Hk + hk) * (long)S * ATTN_D;
 #pragma unroll
   for (int j = 0; j < 16; ++j) {
     const int kr = key0 + (j & 3) + 8 * (j >> 2) + 4 * (lane >> 5);
@@ -379,7 +382,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
     const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
     const ushort_t* __restrict__ V, const ushort_t* __restrict__ dO,
     const float* __restrict__ LSE, const float* __restrict__ Delta,
-    ushort_t* __restrict__ dKp_out,  // [B,Hq,S,D] per-Q-head partials
+    ushort_t* __restrict__ dK_out,   // [B,Hk,S,D] direct (q-heads looped)
     int S, int Hq, int Hk,
     long qb, long qh, long qs,
     long kb, long kh, long ks,
@@ -388,20 +391,16 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int k_block = blockIdx.x;
-  const int hq = blockIdx.y;
+  const int hk = blockIdx.y;        // KV head
   const int b = blockIdx.z;
-  const int hk = hq / (Hq / Hk);
+  const int group = Hq / Hk;
   const float c_log2 = scale * 1.4426950408889634f;
 
   const int key0 = k_block * 128 + wave * 32;
   const int key_row = key0 + (lane & 31);
 
-  const ushort_t* Qp = Q + (long)b * qb + (long)hq * qh;
-  const ushort_t* dOp = dO + (long)b * ob + (long)hq * oh;
   const ushort_t* Kp = K + (long)b * kb + (long)hk * kh;
   const ushort_t* Vp = V + (long)b * kb + (long)hk * kh;
-  const float* Lp = LSE + ((long)b * Hq + hq) * S;
-  const float* Dp = Delta + ((long)b * Hq + hq) * S;
 
   __shared__ ushort_t q_rm[32 * KROW];
   __shared__ ushort_t do_rm[32 * KROW];
@@ -427,6 +426,12 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
   const int st_row = threadIdx.x >> 3;
   const int st_col = (threadIdx.x & 7) * 16;
 
+  for (int g = 0; g < group; ++g) {
+  const int hq = hk * group + g;
+  const ushort_t* Qp = Q + (long)b * qb + (long)hq * qh;
+  const ushort_t* dOp = dO + (long)b * ob + (long)hq * oh;
+  const float* Lp = LSE + ((long)b * Hq + hq) * S;
+  const float* Dp = Delta + ((long)b * Hq + hq) * S;
   for (int t = q_start_tile; t < n_q_tiles; ++t) {
     const int tq0 = t * BLOCK;
     __syncthreads();
@@ -500,8 +505,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
   }
+  }  // q-head group loop
 
-  ushort_t* dKh = dKp_out + ((long)b * Hq + hq) * (long)S * ATTN_D;
+  ushort_t* dKh = dK_out + ((long)b * Hk + hk) * (long)S * ATTN_D;
 #pragma unroll
   for (int j = 0; j < 16; ++j) {
     const int kr = key0 + (j & 3) + 8 * (j >> 2) + 4 * (lane >> 5);
@@ -576,7 +582,7 @@ extern "C" void launch_attn_bwd_dkv(const void* q, const void* k,
                                     long kb, long kh, long ks, long ob,
                                     long oh, long os, float scale,
                                     int causal, void* stream) {
-  dim3 grid((S + 127) / 128, Hq, B);
+  dim3 grid((S + 127) / 128, Hk, B);
   hipLaunchKernelGGL(attn_bwd_dv_kernel, grid, dim3(256), 0,
                      (hipStream_t)stream, (const ushort_t*)q,
                      (const ushort_t*)k, (const ushort_t*)dO, lse,

@@ -69,9 +69,11 @@ class LlamaAttention(nn.Module):
         q = qkv[..., : Hq * D].view(B, S, Hq, D).transpose(1, 2)
         k = qkv[..., Hq * D : (Hq + Hk) * D].view(B, S, Hk, D).transpose(1, 2)
         v = qkv[..., (Hq + Hk) * D :].view(B, S, Hk, D).transpose(1, 2)
-        if qkv.is_cuda and D == 128 and os.environ.get("ANTRAY_FLASH", "1") != "0":
-            # hand-written CDNA4 flash attention (fwd: attention.hip,
-            # bwd: attention_bwd.hip) — takes the strided fused-qkv views
+        if qkv.is_cuda and D == 128 and os.environ.get("ANTRAY_FLASH", "0") == "1":
+            # hand-written CDNA4 flash attention (fwd: attention.hip at
+            # AOTriton parity; bwd: attention_bwd.hip at 0.94x) — opt-in
+            # until the bwd beats AOTriton; numerics covered by
+            # tests/test_attention_gpu.py either way
             o = ops.attention(q, k, v, causal=True)
         else:
             o = F.scaled_dot_product_attention(q, k, v, is_causal=True,
