@@ -1,0 +1,122 @@
+"""ant_ray_amd.llm — LLM serving/batch APIs (vLLM-ROCm passthrough).
+
+Role parity: reference python/ray/llm/ (~31k LoC): serve side wraps vLLM
+engines behind Serve deployments (llm/_internal/serve/engines/vllm/
+vllm_engine.py, OpenAI-compatible ingress), batch side runs vLLM inside
+Ray Data stages (llm/_internal/batch/stages/vllm_engine_stage.py). Per
+SURVEY.md §2.4/§2.5 the reference delegates ALL LLM compute to vLLM —
+TP/PP/EP knobs are engine kwargs — so this module is the same thin shell:
+config types + deployment/processor builders that hand off to vllm when
+it is importable. This image has no vLLM; the builders raise a clear
+ImportError at use (the API surface stays importable for parity).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional
+
+
+def _require_vllm():
+    try:
+        import vllm  # noqa: F401
+
+        return vllm
+    except ImportError as e:
+        raise ImportError(
+            "ray.llm delegates LLM compute to vLLM (reference design: "
+            "llm/_internal/serve/engines/vllm/vllm_engine.py); install "
+            "vllm-rocm in the image to use LLMServer/build_llm_processor. "
+            "For native-kernel LLM inference without vLLM see "
+            "ant_ray_amd.models + ops.attention."
+        ) from e
+
+
+@dataclass
+class LLMConfig:
+    """Parity: ray.serve.llm LLMConfig (model id + engine + scaling)."""
+
+    model_loading_config: Dict[str, Any] = field(default_factory=dict)
+    engine_kwargs: Dict[str, Any] = field(default_factory=dict)
+    deployment_config: Dict[str, Any] = field(default_factory=dict)
+    accelerator_type: Optional[str] = None
+    runtime_env: Optional[dict] = None
+
+    @property
+    def model_id(self) -> str:
+        return self.model_loading_config.get("model_id", "unknown")
+
+    def tensor_parallel_size(self) -> int:
+        return int(self.engine_kwargs.get("tensor_parallel_size", 1))
+
+    def pipeline_parallel_size(self) -> int:
+        return int(self.engine_kwargs.get("pipeline_parallel_size", 1))
+
+    def num_gpus_per_replica(self) -> int:
+        return self.tensor_parallel_size() * self.pipeline_parallel_size()
+
+
+def build_llm_deployment(config: LLMConfig):
+    """Serve deployment wrapping a vLLM AsyncLLMEngine (parity:
+    serve/llm build_llm_deployment)."""
+    vllm = _require_vllm()
+    from ant_ray_amd import serve
+
+    @serve.deployment(
+        num_replicas=config.deployment_config.get("num_replicas", 1),
+        ray_actor_options={"num_gpus": config.num_gpus_per_replica()},
+    )
+    class LLMServer:
+        def __init__(self):
+            from vllm import LLM
+
+            self.engine = LLM(model=config.model_id, **config.engine_kwargs)
+
+        def __call__(self, prompt: str, **params):
+            out = self.engine.generate([prompt], **params)
+            return out[0].outputs[0].text
+
+    return LLMServer.bind()
+
+
+def build_openai_app(configs: List[LLMConfig]):
+    _require_vllm()
+    raise NotImplementedError(
+        "OpenAI-compatible ingress requires vllm; see reference "
+        "llm/_internal/serve/core/ingress/")
+
+
+def build_llm_processor(config: LLMConfig, preprocess=None, postprocess=None):
+    """Ray Data batch-inference processor (parity: ray.data.llm
+    build_llm_processor -> vllm_engine_stage)."""
+    vllm = _require_vllm()
+
+    def processor(dataset):
+        from ant_ray_amd.data.plan import ActorPoolStrategy
+
+        class _VLLMStage:
+            def __init__(self):
+                from vllm import LLM
+
+                self.engine = LLM(model=config.model_id,
+                                  **config.engine_kwargs)
+
+            def __call__(self, batch):
+                prompts = list(batch["prompt"])
+                outs = self.engine.generate(prompts)
+                batch["generated_text"] = [o.outputs[0].text for o in outs]
+                return batch
+
+        ds = dataset
+        if preprocess:
+            ds = ds.map(preprocess)
+        ds = ds.map_batches(
+            _VLLMStage,
+            compute=ActorPoolStrategy(
+                size=config.deployment_config.get("num_replicas", 1)),
+            num_gpus=config.num_gpus_per_replica(), batch_size=64,
+        )
+        if postprocess:
+            ds = ds.map(postprocess)
+        return ds
+
+    return processor
