@@ -57,6 +57,8 @@ def test_autoscaler_with_local_provider():
     )
     from ant_ray_amd.cluster_utils import Cluster
 
+    if ray.is_initialized():
+        ray.shutdown()  # isolate from any prior module's session
     c = Cluster(initialize_head=True, head_node_args={"num_cpus": 1})
     try:
         c.connect()
@@ -88,8 +90,9 @@ def test_insight_flow_graph():
     import ant_ray_amd as ray
     from ant_ray_amd.util import insight
 
-    if not ray.is_initialized():
-        ray.init(num_cpus=4)
+    if ray.is_initialized():
+        ray.shutdown()
+    ray.init(num_cpus=4)
 
     @insight.insight_monitor
     def local_step():

@@ -146,11 +146,12 @@ class TestDistributed:
 
         @ray.remote
         def slow():
-            time.sleep(5)
+            time.sleep(20)
             return 2
 
+        ray.get(fast.remote())  # warm the lease so timing is deterministic
         a, b = fast.remote(), slow.remote()
-        ready, not_ready = ray.wait([a, b], num_returns=1, timeout=3)
+        ready, not_ready = ray.wait([a, b], num_returns=1, timeout=10)
         assert ready == [a] and not_ready == [b]
 
     def test_get_timeout(self, ray_start_regular_module):
