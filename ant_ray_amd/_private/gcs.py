@@ -92,6 +92,7 @@ class ActorInfo:
             "num_restarts": self.num_restarts,
             "death_cause": self.death_cause,
             "max_restarts": int(self.opts.get("max_restarts", 0)),
+            "max_task_retries": int(self.opts.get("max_task_retries", 0)),
         }
 
 
@@ -456,6 +457,14 @@ class GcsServer:
             node = self._pick_node(resources, pg, opts.get("_node_affinity"),
                                    self._vc_allowed(vc_id))
             if node is not None and node.conn is not None and not node.conn.closed:
+                # optimistic accounting: concurrent schedulings must not all
+                # pile onto the same node while its heartbeat is stale (the
+                # reference's ClusterResourceManager debits its view the
+                # same way); the next heartbeat trues it up
+                if not pg:
+                    for k, v in resources.items():
+                        node.resources_available[k] = (
+                            node.resources_available.get(k, 0) - v)
                 try:
                     lease = await node.conn.call(
                         "lease_worker",

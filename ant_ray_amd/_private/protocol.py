@@ -17,6 +17,7 @@ Message format (msgpack arrays):
 from __future__ import annotations
 
 import asyncio
+import os
 import logging
 import struct
 import threading
@@ -39,6 +40,40 @@ def pack(msg) -> bytes:
 
 def unpack(data: bytes):
     return msgpack.unpackb(data, raw=False, strict_map_key=False)
+
+
+# Chaos injection (parity: src/ray/common/asio/asio_chaos.cc —
+# RAY_testing_asio_delay_us="method=min:max,method2=...,*=min:max" injects a
+# uniform-random handler delay; used by the fault-injection test suite).
+_chaos_spec = None
+
+
+def _parse_chaos():
+    global _chaos_spec
+    raw = os.environ.get("RAY_testing_asio_delay_us", "")
+    spec = {}
+    for part in raw.split(","):
+        if "=" not in part:
+            continue
+        name, rng = part.split("=", 1)
+        try:
+            lo, hi = rng.split(":")
+            spec[name.strip()] = (int(lo), int(hi))
+        except ValueError:
+            continue
+    _chaos_spec = spec
+
+
+async def _chaos_delay(method: str):
+    if _chaos_spec is None:
+        _parse_chaos()
+    if not _chaos_spec:
+        return
+    rng = _chaos_spec.get(method) or _chaos_spec.get("*")
+    if rng:
+        import random
+
+        await asyncio.sleep(random.uniform(rng[0], rng[1]) / 1e6)
 
 
 class RpcError(Exception):
@@ -118,6 +153,7 @@ class Connection:
 
     async def _handle_request(self, seq, method, payload):
         try:
+            await _chaos_delay(method)
             result = await self.handler(self, method, payload)
             await self.send([REPLY, seq, result])
         except Exception as e:  # noqa: BLE001 - forwarded to caller

@@ -320,6 +320,7 @@ class Raylet:
     async def rpc_lease_worker(self, conn, p):
         """Grant a worker lease; queue if resources busy; spillback hint if
         infeasible on this node."""
+        logger.info("lease request: %s", p.get("resources"))
         res = dict(p.get("resources") or {})
         pg = p.get("pg")
         if not pg and not self._feasible(res):
@@ -355,7 +356,10 @@ class Raylet:
                     free_gpu = sum(self.gpu_pool.values())
                     if free_gpu + 1e-9 < res["GPU"]:
                         continue
-                asyncio.get_running_loop().create_task(self._grant_lease(item))
+                t = asyncio.get_running_loop().create_task(self._grant_lease(item))
+                t.add_done_callback(
+                    lambda t: t.exception() and logger.error(
+                        "grant_lease failed: %r", t.exception()))
                 self.pending_leases.remove(item)
                 granted_any = True
 

@@ -127,6 +127,10 @@ class _Pickler(cloudpickle.CloudPickler):
             self.contained_refs.append(obj)
             return (ObjectRef._rehydrate, (obj.binary(), obj.owner_addr))
         try:
+            import sys
+
+            if "torch" not in sys.modules:
+                raise ImportError("torch not loaded")  # don't pay the import
             import torch
 
             if isinstance(obj, torch.Tensor):
@@ -209,6 +213,10 @@ def _make_dispatch(contained_refs):
 
     table[ObjectRef] = reduce_ref
     try:
+        import sys as _sys
+
+        if "torch" not in _sys.modules:
+            raise ImportError("torch not loaded in this process")
         import torch
 
         def reduce_tensor(t):

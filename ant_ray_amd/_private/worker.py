@@ -98,6 +98,7 @@ class ActorHandleState:
         self.handle_count = 0
         self.is_owner = False
         self.detached = False
+        self.restart_gen = 0  # mirrors GCS num_restarts; seq resets with it
 
 
 class CoreWorker:
@@ -217,10 +218,44 @@ class CoreWorker:
             if new_addr != st.addr:
                 st.addr = new_addr
                 st.conn = None  # reconnect lazily
+            self._apply_restart_locked(st, view)
         if view["state"] == "DEAD":
             self._fail_actor_tasks(st)
         elif view["state"] == "ALIVE":
             self.io.loop.create_task(self._drain_actor_queue(st))
+
+    def _apply_restart_locked(self, st, view):
+        """A restarted actor runs a FRESH executor whose per-caller ordering
+        restarts at seq 1; every caller must renumber its queued tasks
+        (reference parity: ActorTaskSubmitter resets the sequence and
+        resubmits retryable in-flight tasks on restart)."""
+        gen = int(view.get("num_restarts", 0))
+        if gen == st.restart_gen:
+            return
+        st.restart_gen = gen
+        retries_allowed = int(view.get("max_task_retries", 0))
+        retry, fail = [], []
+        for seq in sorted(st.inflight):
+            p = st.inflight[seq]
+            left = p.get("_retries_left", retries_allowed)
+            if left > 0:
+                p["_retries_left"] = left - 1
+                retry.append(p)
+            else:
+                fail.append(p)
+        st.inflight.clear()
+        st.pending = retry + st.pending
+        st.seq = 0
+        for p in st.pending:
+            st.seq += 1
+            p["seq"] = st.seq
+        for p in fail:
+            self._fail_task(
+                p,
+                RayActorError(
+                    f"actor {st.actor_id.hex()[:8]} restarted; task lost "
+                    f"(set max_task_retries to retry)"),
+            )
 
     async def rpc_set_lease(self, conn, p):
         gpu_ids = p.get("gpu_ids") or []
@@ -799,20 +834,11 @@ class CoreWorker:
                 self._fail_actor_tasks(st, include_inflight=True)
                 return
             if state == "ALIVE" and view.get("addr"):
-                addr = tuple(view["addr"])
                 with st.lock:
-                    st.addr = addr
+                    st.addr = tuple(view["addr"])
                     st.state = "ALIVE"
-                    inflight = list(st.inflight.values())
-                    st.inflight.clear()
-                # default semantics: in-flight tasks on a restarted actor fail
-                for payload in inflight:
-                    self._fail_task(
-                        payload,
-                        RayActorError(
-                            f"actor {st.actor_id.hex()[:8]} restarted; task lost"
-                        ),
-                    )
+                    st.conn = None
+                    self._apply_restart_locked(st, view)
                 await self._drain_actor_queue(st)
                 return
             await protocol.asyncio.sleep(0.2)

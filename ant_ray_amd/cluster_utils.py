@@ -108,6 +108,7 @@ class Cluster:
         deadline = time.monotonic() + timeout
         if not ray.is_initialized():
             ray.init(address=self.gcs_address)
+
         while time.monotonic() < deadline:
             alive = [n for n in ray.nodes() if n.get("alive", n.get("Alive"))]
             if len(alive) >= expect:
@@ -118,6 +119,8 @@ class Cluster:
     def connect(self):
         import ant_ray_amd as ray
 
+        if ray.is_initialized():
+            ray.shutdown()  # a stale session from another test/cluster
         return ray.init(address=self.gcs_address)
 
     def shutdown(self):
