@@ -18,6 +18,7 @@ from ant_ray_amd.data.block import Block, BlockAccessor, block_from_dict
 from ant_ray_amd.data.context import DataContext
 from ant_ray_amd.data.dataset import Dataset, MaterializedDataset
 from ant_ray_amd.data.iterator import DataIterator
+from ant_ray_amd.data.expressions import col, lit
 from ant_ray_amd.data.plan import ActorPoolStrategy, ReadOp
 
 __all__ = [
@@ -25,7 +26,7 @@ __all__ = [
     "MaterializedDataset", "range", "range_tensor", "from_items",
     "from_numpy", "from_pandas", "from_arrow", "from_torch",
     "from_huggingface", "read_parquet", "read_csv", "read_json", "read_text",
-    "read_binary_files", "read_numpy",
+    "read_binary_files", "read_numpy", "col", "lit",
 ]
 
 from builtins import range as _builtin_range  # noqa: E402 (public `range` below shadows it)

@@ -104,6 +104,33 @@ class Dataset:
 
         return self.map_batches(_ren, batch_format="pyarrow", batch_size=None)
 
+    def with_columns(self, exprs: Dict[str, Any]) -> "Dataset":
+        """Add/replace columns from expressions (parity dataset.with_columns
+        + data/expressions.py)."""
+        from ant_ray_amd.data.expressions import Expr, eval_expr_to_column
+
+        def _apply(t: pa.Table):
+            for name, e in exprs.items():
+                colv = eval_expr_to_column(t, e) if isinstance(e, Expr) else e
+                if name in t.column_names:
+                    t = t.set_column(t.column_names.index(name), name, colv)
+                else:
+                    t = t.append_column(name, colv)
+            return t
+
+        return self.map_batches(_apply, batch_format="pyarrow", batch_size=None)
+
+    def filter_expr(self, expr) -> "Dataset":
+        """Vectorized filter by expression (the reference overloads
+        Dataset.filter with Expr; kept as a separate method here)."""
+        from ant_ray_amd.data.expressions import eval_expr_to_column
+
+        def _apply(t: pa.Table):
+            mask = eval_expr_to_column(t, expr)
+            return t.filter(mask)
+
+        return self.map_batches(_apply, batch_format="pyarrow", batch_size=None)
+
     def limit(self, n: int) -> "Dataset":
         return self._with(LimitOp(name=f"Limit[{n}]", limit=n))
 

@@ -184,3 +184,17 @@ def test_train_get_dataset_shard(data, ray_mod):
     # every row lands on exactly one worker across the split
     assert result.metrics["total_rows"] == 64
     assert result.metrics["rows"] == 32  # 8 equal blocks dealt round-robin
+
+
+def test_expressions(data):
+    from ant_ray_amd.data import col, lit
+
+    ds = data.range(20).with_columns({
+        "double": col("id") * 2,
+        "shifted": col("id") + lit(100),
+    })
+    rows = ds.take_all()
+    assert all(r["double"] == 2 * r["id"] and r["shifted"] == r["id"] + 100
+               for r in rows)
+    kept = ds.filter_expr((col("id") > 5) & (col("id") <= 10)).take_all()
+    assert sorted(r["id"] for r in kept) == [6, 7, 8, 9, 10]
