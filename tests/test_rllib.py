@@ -17,6 +17,8 @@ def test_ppo_learns_cartpole():
     import ant_ray_amd as ray
     from ant_ray_amd.rllib import CartPoleEnv, PPOConfig
 
+    if ray.is_initialized():
+        ray.shutdown()  # never inherit another module's (possibly dying) session
     if not ray.is_initialized():
         ray.init(num_cpus=6)
     algo = (PPOConfig()

@@ -9,6 +9,8 @@ import pytest
 def ray_mod():
     import ant_ray_amd as ray
 
+    if ray.is_initialized():
+        ray.shutdown()  # never inherit another module's (possibly dying) session
     if not ray.is_initialized():
         ray.init(num_cpus=8)
     yield ray

@@ -11,6 +11,8 @@ def serve_mod():
     import ant_ray_amd as ray
     from ant_ray_amd import serve
 
+    if ray.is_initialized():
+        ray.shutdown()  # never inherit another module's (possibly dying) session
     if not ray.is_initialized():
         ray.init(num_cpus=8)
     yield serve
