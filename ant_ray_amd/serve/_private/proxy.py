@@ -55,9 +55,9 @@ class HTTPProxy:
 
         asyncio.new_event_loop().run_until_complete(run())
 
-    async def _refresh_routes(self):
+    async def _refresh_routes(self, force: bool = False):
         now = time.monotonic()
-        if self._routes and now - self._last_refresh < _ROUTE_REFRESH_S:
+        if not force and self._routes and now - self._last_refresh < _ROUTE_REFRESH_S:
             return
         import ant_ray_amd as ray
 
@@ -110,6 +110,10 @@ class HTTPProxy:
             await _send_simple(send, 503, f"controller unavailable: {e}".encode())
             return
         prefix = self._match(path)
+        if prefix is None:
+            # the app may have been deployed within the refresh window
+            await self._refresh_routes(force=True)
+            prefix = self._match(path)
         if prefix is None:
             await _send_simple(send, 404, b"no app at this route")
             return
