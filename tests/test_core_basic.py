@@ -257,3 +257,26 @@ class TestDistributed:
 
         wid, tid = ray.get(ctx.remote())
         assert wid and tid
+
+
+class TestObjectStorePressure:
+    def test_eviction_and_object_lost(self):
+        """Sealed unreferenced-in-store objects are LRU-evicted under
+        pressure; getting an evicted ray.put object raises ObjectLostError
+        (no lineage), newest objects stay readable."""
+        import numpy as np
+
+        import ant_ray_amd as ray
+        from ant_ray_amd.exceptions import ObjectLostError
+
+        if ray.is_initialized():
+            ray.shutdown()
+        ray.init(num_cpus=2, object_store_memory=200 * 1024 * 1024)
+        try:
+            refs = [ray.put(np.zeros(10 * 1024 * 1024, dtype=np.uint8))
+                    for _ in range(30)]
+            assert ray.get(refs[-1]).nbytes == 10 * 1024 * 1024
+            with pytest.raises(ObjectLostError):
+                ray.get(refs[0], timeout=15)
+        finally:
+            ray.shutdown()
