@@ -17,6 +17,7 @@ from ant_ray_amd.serve._private.common import (
     AutoscalingConfig,
 )
 from ant_ray_amd.serve.batching import batch
+from ant_ray_amd.serve.multiplex import get_multiplexed_model_id, multiplexed
 from ant_ray_amd.serve.handle import DeploymentHandle, DeploymentResponse
 
 PROXY_NAME = "SERVE_PROXY_ACTOR"
@@ -24,7 +25,8 @@ PROXY_NAME = "SERVE_PROXY_ACTOR"
 __all__ = [
     "Application", "AutoscalingConfig", "Deployment", "DeploymentHandle",
     "DeploymentResponse", "batch", "delete", "deployment",
-    "get_app_handle", "get_deployment_handle", "ingress", "run", "shutdown",
+    "get_app_handle", "get_deployment_handle", "get_multiplexed_model_id",
+    "ingress", "multiplexed", "run", "shutdown",
     "start", "status",
 ]
 

@@ -162,3 +162,30 @@ def test_autoscaling_policy_math():
     assert _calculate_desired_num_replicas(cfg, total_ongoing=8, current=2) == 4
     assert _calculate_desired_num_replicas(cfg, total_ongoing=100, current=2) == 10
     assert _calculate_desired_num_replicas(cfg, total_ongoing=3, current=2) == 2
+
+
+def test_multiplexed(serve_mod):
+    serve = serve_mod
+
+    @serve.deployment
+    class MultiModel:
+        def __init__(self):
+            self.loads = []
+
+        @serve.multiplexed(max_num_models_per_replica=2)
+        async def get_model(self, model_id: str):
+            self.loads.append(model_id)
+            return f"model:{model_id}"
+
+        async def __call__(self, model_id: str):
+            m = await self.get_model(model_id)
+            return {"model": m, "loads": list(self.loads)}
+
+    h = serve.run(MultiModel.bind(), name="mux", route_prefix="/mux")
+    r1 = h.remote("a").result(timeout_s=30)
+    r2 = h.remote("a").result(timeout_s=30)
+    assert r2["model"] == "model:a" and r2["loads"].count("a") == 1  # cached
+    h.remote("b").result(timeout_s=30)
+    out = h.remote("c").result(timeout_s=30)  # evicts LRU ("a")
+    out = h.remote("a").result(timeout_s=30)  # reloads a
+    assert out["loads"].count("a") == 2
