@@ -162,6 +162,11 @@ class Raylet:
             self.idle_workers.remove(w)
         if w.leased:
             self._release_resources(w)
+        # reclaim leases this worker held as LESSEE (e.g. an actor that was
+        # submitting tasks when it died): without this its leased workers
+        # stay marked busy forever and the node starves
+        if w.conn is not None:
+            self._reclaim_leases_of(w.conn)
         try:
             await self.gcs_conn.call(
                 "report_worker_failure", {"worker_id": w.worker_id, "reason": reason}, timeout=5
@@ -307,6 +312,19 @@ class Raylet:
                 back = min(1.0, remaining)
                 self.gpu_pool[gid] = min(1.0, self.gpu_pool[gid] + back)
                 remaining -= back
+
+    def _reclaim_leases_of(self, owner_conn):
+        for w2 in list(self.workers.values()):
+            if w2.leased and w2.lease_owner is owner_conn:
+                logger.info("reclaiming lease of dead lessee: worker %s",
+                            w2.worker_id.hex()[:8])
+                self._release_resources(w2)
+                if w2.is_actor or w2.gpu_ids or getattr(w2, "dedicated", False):
+                    self._kill_worker(w2)
+                elif w2 not in self.idle_workers:
+                    w2.idle_since = time.monotonic()
+                    self.idle_workers.append(w2)
+        self._pump_leases()
 
     def _release_resources(self, w: WorkerProc):
         if not w.leased:

@@ -201,6 +201,37 @@ class gpu_transport_context:
         _transport_ctx.ctx = None
 
 
+_PLAIN_TYPES = (type(None), bool, int, float, str, bytes, bytearray)
+
+
+def _is_plain_data(v, depth=0) -> bool:
+    """True when the C-pickler fast path is SAFE: only data types whose plain
+    pickling round-trips in any process. Functions/classes/arbitrary objects
+    force the cloudpickle path (plain pickle would serialize a __main__
+    function by reference, which a worker whose __main__ is default_worker
+    cannot resolve)."""
+    if isinstance(v, _PLAIN_TYPES):
+        return True
+    t = type(v)
+    mod = t.__module__
+    if mod == "numpy" or mod.startswith("numpy."):
+        return True
+    if mod == "torch" or mod.startswith("torch."):
+        return t.__name__ in ("Tensor", "Parameter", "dtype", "Size")
+    from ant_ray_amd._private.object_ref import ObjectRef
+
+    if t is ObjectRef:
+        return True
+    if depth > 6:
+        return False
+    if t in (list, tuple, set, frozenset):
+        return all(_is_plain_data(x, depth + 1) for x in v)
+    if t is dict:
+        return all(_is_plain_data(k, depth + 1) and _is_plain_data(x, depth + 1)
+                   for k, x in v.items())
+    return False
+
+
 def _make_dispatch(contained_refs):
     """copyreg-style dispatch table for the C pickler fast path."""
     from ant_ray_amd._private.object_ref import ObjectRef
@@ -274,17 +305,21 @@ def serialize(value: Any, metadata: bytes = META_PICKLE) -> SerializedObject:
         buffers.append(mv)
         return False
 
-    # Fast path: C pickler (ObjectRef / plain torch tensors via dispatch
-    # table). Functions/lambdas/local classes raise -> cloudpickle fallback.
+    # Fast path: C pickler for verified plain-data payloads (ObjectRef /
+    # numpy / torch tensors / basic containers). Anything else — functions,
+    # classes, user objects — goes through cloudpickle so __main__-defined
+    # code ships by value.
     contained_refs: list = []
     f = io.BytesIO()
-    try:
-        p = pickle.Pickler(f, protocol=5, buffer_callback=cb)
-        p.dispatch_table = _make_dispatch(contained_refs)
-        p.dump(value)
-        return SerializedObject(f.getvalue(), buffers, contained_refs, metadata)
-    except (pickle.PicklingError, TypeError, AttributeError):
-        pass
+    if _is_plain_data(value):
+        try:
+            p = pickle.Pickler(f, protocol=5, buffer_callback=cb)
+            p.dispatch_table = _make_dispatch(contained_refs)
+            p.dump(value)
+            return SerializedObject(f.getvalue(), buffers, contained_refs,
+                                    metadata)
+        except (pickle.PicklingError, TypeError, AttributeError):
+            pass
     buffers.clear()
     f = io.BytesIO()
     p = _Pickler(f, buffer_callback=cb)
