@@ -93,6 +93,12 @@ class ActorInfo:
             "death_cause": self.death_cause,
             "max_restarts": int(self.opts.get("max_restarts", 0)),
             "max_task_retries": int(self.opts.get("max_task_retries", 0)),
+            "required_resources": {
+                "CPU": float(self.opts.get("num_cpus", 1)),
+                **({"GPU": float(self.opts["num_gpus"])}
+                   if self.opts.get("num_gpus") else {}),
+                **(self.opts.get("resources") or {}),
+            },
         }
 
 

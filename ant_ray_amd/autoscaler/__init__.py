@@ -121,7 +121,7 @@ class StandardAutoscaler:
         demands: List[Dict[str, float]] = []
         for a in _gcs_call("list_actors"):
             if a.get("state") in ("PENDING_CREATION", "DEPENDENCIES_UNREADY"):
-                demands.append({"CPU": 1})
+                demands.append(a.get("required_resources") or {"CPU": 1})
         for pg in _gcs_call("list_placement_groups"):
             if pg.get("state") == "PENDING":
                 demands.extend(pg.get("bundles") or [])
