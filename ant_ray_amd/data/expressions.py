@@ -40,6 +40,17 @@ class Expr:
     def __truediv__(self, other):
         return BinaryExpr(pc.divide, self, _wrap(other))
 
+    def __mod__(self, other):
+        return BinaryExpr(lambda a, b: pc.subtract(
+            a, pc.multiply(pc.floor(pc.divide(pc.cast(a, pa.float64()),
+                                              pc.cast(b, pa.float64()))), b)),
+            self, _wrap(other))
+
+    def __floordiv__(self, other):
+        return BinaryExpr(lambda a, b: pc.floor(
+            pc.divide(pc.cast(a, pa.float64()), pc.cast(b, pa.float64()))),
+            self, _wrap(other))
+
     # comparisons
     def __gt__(self, other):
         return BinaryExpr(pc.greater, self, _wrap(other))
