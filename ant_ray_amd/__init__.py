@@ -103,6 +103,10 @@ def init(
         global_worker.session_dir = info["session_dir"]
     else:
         gcs_addr = address
+    # Ray-Client scheme (parity ray://host:port — util/client): a remote
+    # driver that proxies big objects through the raylet data plane
+    if gcs_addr.startswith("ray://"):
+        gcs_addr = gcs_addr[len("ray://"):]
 
     host, port = gcs_addr.rsplit(":", 1)
     cw = CoreWorker(DRIVER_MODE, node_ip=_node_ip_address, session_dir=global_worker.session_dir)

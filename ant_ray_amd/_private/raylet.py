@@ -499,6 +499,28 @@ class Raylet:
                     self.resources_available[k] = self.resources_available.get(k, 0.0) + v
         return {"ok": True}
 
+    # ------------------------------------------------- client-mode data plane
+    # parity: the Ray Client server's object plane (util/client/server) —
+    # remote drivers have no shm mapping, so the raylet stores/serves bytes
+    async def rpc_store_put(self, conn, p):
+        oid, data, meta = p["oid"], p["data"], p.get("meta", b"py")
+        try:
+            off = self._store.create_object(oid, len(data), meta)
+            mv = self._store.view_at(off, len(data), True)
+            mv[:] = data
+            del mv
+            self._store.seal(oid)
+            self._store.release(oid)
+        except ValueError:
+            pass  # already exists
+        return {"ok": True}
+
+    async def rpc_pull_object(self, conn, p):
+        buf, meta = self._store.get_buffer(p["oid"], 0.0)
+        if buf is None:
+            return {"data": None}
+        return {"data": bytes(memoryview(buf)), "meta": bytes(meta)}
+
     # ------------------------------------------------------------------ misc
     async def rpc_node_info(self, conn, p):
         return {

@@ -157,10 +157,21 @@ class CoreWorker:
             node_id = node["node_id"]
         self.node_id = node_id
         self.raylet_addr = tuple(raylet_addr)
-        # object store attach
-        from ant_ray_amd._shm_store import ShmStore
+        # object store attach; a REMOTE driver (Ray-Client mode) has no
+        # local shm mapping — it proxies big objects through the raylet
+        self.client_mode = False
+        try:
+            if is_driver and os.environ.get("ANTRAY_FORCE_CLIENT") == "1":
+                raise OSError("client mode forced")
+            from ant_ray_amd._shm_store import ShmStore
 
-        self.store = ObjectStore(ShmStore.open(store_path))
+            self.store = ObjectStore(ShmStore.open(store_path))
+        except Exception:
+            if is_driver:
+                self.client_mode = True
+                self.store = ObjectStore(None)
+            else:
+                raise
         # our RPC server (direct task pushes / object pulls)
         self._server, port = await protocol.serve(self._handle_rpc, self.node_ip, 0)
         self.addr = (self.node_ip, port)
@@ -332,6 +343,17 @@ class CoreWorker:
         self._register_escapes(sobj)
         if _owner_inline and sobj.total_size <= INLINE_OBJECT_MAX:
             self.store.memory.put(oid, _RawResult(sobj.to_bytes(), sobj.metadata))
+        elif getattr(self, "client_mode", False):
+            if sobj.total_size <= INLINE_OBJECT_MAX:
+                self.store.memory.put(oid, _RawResult(sobj.to_bytes(),
+                                                      sobj.metadata))
+            else:
+                # Ray-Client data plane: bytes live in the raylet's store
+                self.io.run(self.raylet.call("store_put", {
+                    "oid": oid, "data": sobj.to_bytes(),
+                    "meta": sobj.metadata}, timeout=60), timeout=65)
+                self.store.memory.put(oid, IN_PLASMA)
+                self._object_locations[oid] = tuple(self.raylet_addr)
         else:
             self.store.put_serialized_to_shm(oid, sobj)
             self.store.memory.put(oid, IN_PLASMA)
