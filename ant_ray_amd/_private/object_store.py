@@ -102,7 +102,13 @@ class ObjectStore:
 
     # --- writes ---------------------------------------------------------------
 
-    def put_serialized_to_shm(self, oid: bytes, sobj: serialization.SerializedObject):
+    def put_serialized_to_shm(self, oid: bytes,
+                              sobj: serialization.SerializedObject,
+                              pin: bool = False):
+        """pin=True keeps the creator's refcount (the owner holds the
+        primary copy pinned while local refs exist — parity with the
+        reference's primary-copy pinning; unpinned objects are LRU-evictable
+        the moment readers drop them)."""
         off = self.shm.create_object(oid, sobj.total_size, sobj.metadata)
         try:
             mv = self.shm.view_at(off, sobj.total_size, writable=True)
@@ -111,8 +117,9 @@ class ObjectStore:
             self.shm.seal(oid)
         except Exception:
             self.shm.abort(oid)
+            self.shm.release(oid)
             raise
-        finally:
+        if not pin:
             self.shm.release(oid)
 
     def put_value(self, oid: bytes, value: Any, *, force_shm: bool = False) -> int:

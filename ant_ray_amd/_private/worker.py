@@ -43,6 +43,7 @@ from ant_ray_amd.exceptions import (
     ObjectLostError,
     RayActorError,
     RayTaskError,
+    ObjectStoreFullError,
     RaySystemError,
 )
 
@@ -116,7 +117,8 @@ class CoreWorker:
         self._put_index = 0
         self._lock = threading.RLock()
         self._local_refs: Dict[bytes, int] = defaultdict(int)
-        self._owned: Dict[bytes, dict] = {}  # oid -> {escaped, size}
+        self._owned: Dict[bytes, dict] = {}
+        self._spilled: Dict[bytes, str] = {}  # oid -> spill file path  # oid -> {escaped, size}
         self._object_locations: Dict[bytes, Tuple[str, int]] = {}
         self._actors: Dict[bytes, ActorHandleState] = {}
         self._actor_results: Dict[bytes, dict] = {}
@@ -308,6 +310,9 @@ class CoreWorker:
             buf, meta = self.store.shm.get_buffer(oid, 0.0)
             if buf is not None:
                 return {"data": bytes(memoryview(buf)), "meta": bytes(meta)}
+        spilled = self._read_spilled(oid)
+        if spilled is not None:
+            return {"data": spilled[0], "meta": spilled[1]}
         holder = self._object_locations.get(oid)
         if holder is not None and tuple(holder) != self.addr:
             return {"redirect": list(holder)}
@@ -343,7 +348,7 @@ class CoreWorker:
         self._register_escapes(sobj)
         if _owner_inline and sobj.total_size <= INLINE_OBJECT_MAX:
             self.store.memory.put(oid, _RawResult(sobj.to_bytes(), sobj.metadata))
-        elif getattr(self, "client_mode", False):
+        elif getattr(self, "client_mode", False):  # noqa: SIM114
             if sobj.total_size <= INLINE_OBJECT_MAX:
                 self.store.memory.put(oid, _RawResult(sobj.to_bytes(),
                                                       sobj.metadata))
@@ -355,11 +360,75 @@ class CoreWorker:
                 self.store.memory.put(oid, IN_PLASMA)
                 self._object_locations[oid] = tuple(self.raylet_addr)
         else:
-            self.store.put_serialized_to_shm(oid, sobj)
+            self._put_pinned_with_spill(oid, sobj)
             self.store.memory.put(oid, IN_PLASMA)
         with self._lock:
-            self._owned[oid] = {"escaped": False, "size": sobj.total_size}
+            self._owned[oid] = {"escaped": False, "size": sobj.total_size,
+                                "pinned": True}
         return ObjectRef(oid, self.addr, worker=self)
+
+    def _put_pinned_with_spill(self, oid, sobj):
+        """Primary copies are pinned; when the store cannot hold a new one
+        even after LRU eviction, this owner SPILLS its oldest pinned objects
+        to disk and retries (parity: raylet LocalObjectManager spill —
+        local_object_manager.h:44)."""
+        for _ in range(8):
+            try:
+                self.store.put_serialized_to_shm(oid, sobj, pin=True)
+                return
+            except RuntimeError as e:
+                if "out of memory" not in str(e):
+                    raise
+            # over-spill to beat fragmentation/alignment overheads
+            freed = self._spill_owned(max(sobj.total_size * 2,
+                                          32 * 1024 * 1024))
+            if freed <= 0:
+                raise ObjectStoreFullError(
+                    f"object of {sobj.total_size} bytes does not fit and "
+                    "nothing is spillable")
+        raise ObjectStoreFullError(
+            f"object of {sobj.total_size} bytes does not fit after spilling")
+
+    def _spill_owned(self, need_bytes: int) -> int:
+        spill_dir = os.path.join(self.session_dir or "/tmp/antray", "spill")
+        os.makedirs(spill_dir, exist_ok=True)
+        freed = 0
+        with self._lock:
+            candidates = [o for o, info in self._owned.items()
+                          if info.get("pinned") and o not in self._spilled]
+        for o in candidates:
+            if freed >= need_bytes:
+                break
+            buf, meta = self.store.shm.get_buffer(o, 0.0)
+            if buf is None:
+                continue
+            path = os.path.join(spill_dir, o.hex())
+            meta_b = bytes(meta)
+            with open(path, "wb") as f:
+                f.write(len(meta_b).to_bytes(4, "little"))
+                f.write(meta_b)
+                f.write(memoryview(buf))
+            del buf
+            self._spilled[o] = path
+            self.store.shm.release(o)   # drop the owner pin
+            self.store.shm.delete(o)
+            with self._lock:
+                info = self._owned.get(o)
+                if info:
+                    info["pinned"] = False
+                    freed += info.get("size", 0)
+        logger.info("spilled %d bytes to %s", freed, spill_dir)
+        return freed
+
+    def _read_spilled(self, oid):
+        path = self._spilled.get(oid)
+        if path is None:
+            return None
+        with open(path, "rb") as f:
+            mlen = int.from_bytes(f.read(4), "little")
+            meta = f.read(mlen)
+            data = f.read()
+        return data, meta
 
     def _register_escapes(self, sobj: serialization.SerializedObject):
         # refs serialized into a stored value may be read anywhere -> pin them
@@ -436,6 +505,10 @@ class CoreWorker:
             # under memory pressure; ray.put objects have no lineage, so fail
             # loudly instead of spinning (reference raises ObjectLostError;
             # task outputs are re-driven by task retries upstream)
+            spilled = self._read_spilled(oid)
+            if spilled is not None:
+                data, meta = spilled
+                return self._deserialize_buffer(memoryview(data), meta)
             no_remote = pull_addr is None or tuple(pull_addr) == self.addr
             if (attempt >= 3 and no_remote
                     and not self.store.memory.is_pending(oid)
@@ -957,6 +1030,17 @@ class CoreWorker:
                 if info is None or info.get("escaped"):
                     return
             # owned, unreferenced, never escaped -> free storage
+            if info.get("pinned"):
+                try:
+                    self.store.shm.release(oid)
+                except Exception:
+                    pass
+            path = self._spilled.pop(oid, None)
+            if path:
+                try:
+                    os.unlink(path)
+                except OSError:
+                    pass
             self.store.free([oid])
         except Exception:
             pass

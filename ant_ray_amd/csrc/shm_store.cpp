@@ -312,6 +312,13 @@ class ShmStore : public std::enable_shared_from_this<ShmStore> {
     if (prev == 1 && (slot->flags & 1)) FreeSlotLocked(slot);
   }
 
+  void AddRef(const std::string& oid) {
+    Guard g(this);
+    Slot* slot = FindSlot(oid.data());
+    if (slot && slot->state != SLOT_TOMBSTONE)
+      slot->refcount.fetch_add(1, std::memory_order_relaxed);
+  }
+
   // Owner-driven delete (refcount-zero objects freed now, pinned ones marked).
   void Delete(const std::string& oid) {
     Guard g(this);
@@ -527,7 +534,23 @@ class PinnedBuffer {
   PinnedBuffer(std::shared_ptr<ShmStore> store, std::string oid, uint64_t off,
                uint64_t size)
       : store_(std::move(store)), oid_(std::move(oid)), off_(off), size_(size) {}
+  // copying takes its OWN pin; moving transfers it. (The default copy ctor
+  // previously let the pybind return-tuple's temporary release the pin on
+  // destruction, leaving live zero-copy readers unprotected and corrupting
+  // the refcount when callers released explicitly.)
+  PinnedBuffer(const PinnedBuffer& o)
+      : store_(o.store_), oid_(o.oid_), off_(o.off_), size_(o.size_) {
+    if (store_) store_->AddRef(oid_);
+  }
+  PinnedBuffer(PinnedBuffer&& o) noexcept
+      : store_(std::move(o.store_)), oid_(std::move(o.oid_)), off_(o.off_),
+        size_(o.size_) {
+    o.store_.reset();
+  }
+  PinnedBuffer& operator=(const PinnedBuffer&) = delete;
+  PinnedBuffer& operator=(PinnedBuffer&&) = delete;
   ~PinnedBuffer() {
+    if (!store_) return;
     try {
       store_->Release(oid_);
     } catch (...) {
