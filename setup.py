@@ -3,10 +3,9 @@
 Two tiers:
   * `_shm_store` — plain pybind11/C++ (no torch, no GPU): the shared-memory
     object store used by the core runtime.
-  * `_hip_ops`, `_gpu_ipc` — HIP/CDNA4 extensions built via
-    torch.utils.cpp_extension with PYTORCH_ROCM_ARCH=gfx950 (cross-compiles
-    without a GPU). Built by `python setup.py build_ext --inplace` or by
-    `__graft_entry__.build()`.
+  * `_hip_ops`, `_gpu_ipc` — HIP/CDNA4 extensions: built ONLY by
+    `ant_ray_amd/csrc/build.py` (hipcc --offload-arch=gfx950), called from
+    `__graft_entry__.build()`. Not declared here.
 """
 import os
 import sys
@@ -31,54 +30,11 @@ ext_modules = [
     ),
 ]
 
-# HIP extensions (torch): compile only when requested, since importing torch
-# costs ~2s and the HIP toolchain path is separate.
-if os.environ.get("ANTRAY_BUILD_HIP", "1") == "1":
-    try:
-        from torch.utils.cpp_extension import CppExtension  # noqa: F401
-
-        from torch.utils import cpp_extension as tce
-
-        hip_sources = [
-            "ant_ray_amd/csrc/hip_ops.cpp",
-            "ant_ray_amd/csrc/kernels/elementwise.hip",
-            "ant_ray_amd/csrc/kernels/norms.hip",
-            "ant_ray_amd/csrc/kernels/rope.hip",
-            "ant_ray_amd/csrc/kernels/cross_entropy.hip",
-            "ant_ray_amd/csrc/kernels/adamw.hip",
-            "ant_ray_amd/csrc/kernels/pack.hip",
-        ]
-        hip_sources = [s for s in hip_sources if os.path.exists(os.path.join(ROOT, s))]
-        if hip_sources:
-            ext_modules.append(
-                tce.CUDAExtension(
-                    name="ant_ray_amd._hip_ops",
-                    sources=hip_sources,
-                    extra_compile_args={
-                        "cxx": ["-O3", "-std=c++17"],
-                        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
-                    },
-                )
-            )
-        ipc_src = "ant_ray_amd/csrc/gpu_ipc.cpp"
-        if os.path.exists(os.path.join(ROOT, ipc_src)):
-            ext_modules.append(
-                tce.CUDAExtension(
-                    name="ant_ray_amd._gpu_ipc",
-                    sources=[ipc_src],
-                    extra_compile_args={
-                        "cxx": ["-O3", "-std=c++17"],
-                        "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
-                    },
-                )
-            )
-        from torch.utils.cpp_extension import BuildExtension
-
-        cmdclass = {"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)}
-    except ImportError:
-        cmdclass = {}
-else:
-    cmdclass = {}
+# HIP extensions (_hip_ops, _gpu_ipc) are built EXCLUSIVELY by
+# ant_ray_amd/csrc/build.py (hipcc --offload-arch=gfx950, explicit source
+# list, incremental objects under csrc/_build). They are deliberately NOT
+# setuptools extensions: an earlier stale source list here once clobbered
+# the good in-tree .so with one missing the attention kernels.
 
 setup(
     name="ant_ray_amd",
@@ -86,5 +42,4 @@ setup(
     description="MI355X-native distributed actor/task runtime with Ray-compatible APIs",
     packages=["ant_ray_amd"],
     ext_modules=ext_modules,
-    cmdclass=cmdclass if "cmdclass" in dir() else {},
 )

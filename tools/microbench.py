@@ -51,6 +51,17 @@ def main():
     timeit("single client put calls (Plasma Store)",
            lambda: [ray.put(small) for _ in range(100)], 100, dur)
 
+    # multi client put: 10 worker processes each doing 100 puts
+    # (parity ray_perf.py:113 do_put_small / put_multi_small)
+    @ray.remote
+    def do_put_small():
+        for _ in range(100):
+            ray.put(0)
+
+    timeit("multi client put calls (Plasma Store)",
+           lambda: ray.get([do_put_small.remote() for _ in range(10)]),
+           1000, dur)
+
     def put_gb():
         for _ in range(8):
             ray.put(arr_1mb)
@@ -65,6 +76,26 @@ def main():
     print(json.dumps({"name": "single client put gigabytes",
                       "value": round(gbps, 2), "unit": "GB/s"}), flush=True)
 
+    # multi client put gigabytes: 10 workers x 10 puts of 80 MB
+    # (parity ray_perf.py:140 do_put / put_multi)
+    @ray.remote
+    def do_put():
+        for _ in range(10):
+            ray.put(np.zeros(10 * 1024 * 1024, dtype=np.int64))
+
+    def put_multi():
+        ray.get([do_put.remote() for _ in range(10)])
+
+    put_multi()  # warmup
+    start = time.time()
+    cycles = 0
+    while time.time() - start < dur:
+        put_multi()
+        cycles += 1
+    gbps = cycles * 10 * 8 * 0.1 / (time.time() - start)
+    print(json.dumps({"name": "multi client put gigabytes",
+                      "value": round(gbps, 2), "unit": "GB/s"}), flush=True)
+
     # ---- tasks
     @ray.remote
     def tiny():
@@ -75,10 +106,20 @@ def main():
     timeit("single client tasks async",
            lambda: ray.get([tiny.remote() for _ in range(1000)]), 1000, dur)
 
+    def wait_multiple_refs():
+        not_ready = [tiny.remote() for _ in range(1000)]
+        while not_ready:
+            _ready, not_ready = ray.wait(not_ready)
+
+    timeit("single client wait 1k refs", wait_multiple_refs, 1, dur)
+
     # ---- actors
     @ray.remote
     class Actor:
         def ping(self):
+            return b"ok"
+
+        def ping_arg(self, x):
             return b"ok"
 
     a = Actor.remote()
@@ -94,12 +135,57 @@ def main():
 
     ray.kill(a)
     ray.kill(conc)
+
+    # multi client tasks async: 4 worker-actors each submitting a batch of
+    # tasks from inside the worker (parity ray_perf.py:183 multi_task)
+    @ray.remote(num_cpus=0)
+    class Batcher:
+        def ping(self):
+            return b"ok"
+
+        def task_batch(self, k):
+            ray.get([tiny.remote() for _ in range(k)])
+
+        def call_batch(self, servers, k):
+            ray.get([s.ping.remote() for s in servers for _ in range(k)])
+
+        def call_batch_arg(self, servers, k):
+            x = ray.put(0)
+            ray.get([s.ping_arg.remote(x) for s in servers for _ in range(k)])
+
+    m, nb = 4, (500 if args.quick else 2000)
+    batchers = [Batcher.remote() for _ in range(m)]
+    ray.get([b.ping.remote() for b in batchers])
+    timeit("multi client tasks async",
+           lambda: ray.get([b.task_batch.remote(nb) for b in batchers]),
+           nb * m, dur)
+
     n_cpu = min(os.cpu_count() or 4, 8)
-    actors = [Actor.remote() for _ in range(n_cpu)]
-    ray.get([x.ping.remote() for x in actors])
+    servers = [Actor.options(num_cpus=0).remote() for _ in range(n_cpu)]
+    ray.get([x.ping.remote() for x in servers])
+
+    # 1:n — one client actor fanning out to n_cpu server actors
+    # (parity ray_perf.py:222 actor_async_direct)
+    nk = 200 if args.quick else 600
+    client = batchers[0]
+    timeit("1:n actor calls async",
+           lambda: ray.get(client.call_batch.remote(servers, nk)),
+           nk * n_cpu, dur)
+
+    # n:n — m client actors each fanning out to all servers
+    # (parity ray_perf.py:238 actor_multi2)
     timeit("n:n actor calls async",
-           lambda: ray.get([x.ping.remote() for x in actors
-                            for _ in range(200)]), 200 * n_cpu, dur)
+           lambda: ray.get([b.call_batch.remote(servers, nk // m)
+                            for b in batchers]), (nk // m) * n_cpu * m, dur)
+
+    # n:n with a shared put-arg per batch
+    # (parity ray_perf.py:243 actor_multi2_direct_arg)
+    timeit("n:n actor calls with arg async",
+           lambda: ray.get([b.call_batch_arg.remote(servers, nk // m)
+                            for b in batchers]), (nk // m) * n_cpu * m, dur)
+    for b in batchers:
+        ray.kill(b)
+    actors = servers
 
     # ---- async actors
     @ray.remote
