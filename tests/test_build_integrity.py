@@ -1,0 +1,29 @@
+"""The in-tree native extensions must be complete and loadable on CPU.
+
+Guards against the failure mode where a partial rebuild relinks
+`_hip_ops` without some kernel objects: the .so links fine but dies at
+import time on the GPU box with `undefined symbol: launch_*`. Importing
+here (torch first, so libtorch is resolvable) catches that on every
+driver CPU run.
+"""
+import pytest
+
+
+def test_hip_ops_complete():
+    torch = pytest.importorskip("torch")  # noqa: F841
+
+    import ant_ray_amd._hip_ops as m
+
+    for fn in ("rmsnorm_fwd", "rmsnorm_bwd", "rope_", "swiglu_fwd",
+               "swiglu_bwd", "adamw_", "cross_entropy_fwd_bwd",
+               "attn_fwd", "attn_bwd"):
+        assert hasattr(m, fn), f"_hip_ops is missing {fn} — stale/partial link"
+
+
+def test_shm_store_loads():
+    from ant_ray_amd._shm_store import ShmStore  # noqa: F401
+
+
+def test_gpu_ipc_loads():
+    pytest.importorskip("torch")
+    import ant_ray_amd._gpu_ipc  # noqa: F401
