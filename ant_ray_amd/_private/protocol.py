@@ -294,7 +294,16 @@ class EventLoopThread:
 
     def stop(self):
         try:
-            self.loop.call_soon_threadsafe(self.loop.stop)
+            def _cancel_then_stop():
+                # cancel recv loops etc. BEFORE stopping, so tasks unwind
+                # inside the loop instead of warning at GC time
+                for t in asyncio.all_tasks(self.loop):
+                    t.cancel()
+                self.loop.call_soon(self.loop.stop)
+
+            self.loop.call_soon_threadsafe(_cancel_then_stop)
             self._thread.join(timeout=2)
+            if not self._thread.is_alive():
+                self.loop.close()
         except Exception:
             pass
