@@ -258,6 +258,14 @@ class GcsServer:
         conn.session["worker_id"] = wid
         return {"job_id": job_id}
 
+    async def rpc_pick_raylet(self, conn, p):
+        """Name a feasible raylet for a task lease that is infeasible on the
+        requester's node (spillback target)."""
+        node = self._pick_node(dict(p.get("resources") or {}))
+        if node is None:
+            return {"addr": None}
+        return {"addr": list(node.addr), "node_id": node.node_id}
+
     async def rpc_report_worker_failure(self, conn, p):
         wid = p["worker_id"]
         self.workers.pop(wid, None)

@@ -18,6 +18,8 @@ from ant_ray_amd.exceptions import GetTimeoutError, ObjectLostError
 # Objects <= this are inlined in RPC replies / memory store (parity with
 # reference max_direct_call_object_size, ray_config_def.h:200).
 INLINE_OBJECT_MAX = 100 * 1024
+PULL_CHUNK_BYTES = 5 << 20  # cross-node pull slice size (parity: reference
+                            # ObjectManagerConfig object_chunk_size = 5 MiB)
 
 
 class _InPlasma:

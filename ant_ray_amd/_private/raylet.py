@@ -342,7 +342,29 @@ class Raylet:
         res = dict(p.get("resources") or {})
         pg = p.get("pg")
         if not pg and not self._feasible(res):
-            return {"granted": False, "infeasible": True}
+            reply = {"granted": False, "infeasible": True}
+            try:
+                # ask the GCS for a node whose TOTAL resources fit -> the
+                # client re-leases there (spillback parity)
+                r = await self.gcs_conn.call(
+                    "pick_raylet", {"resources": res}, timeout=5)
+                if r.get("addr"):
+                    reply["spillback"] = r["addr"]
+            except Exception:
+                pass
+            return reply
+        # make sure a dying lessee's leases come back even when the lessee
+        # is a remote driver (spillback) that never registered as a worker
+        if not getattr(conn, "_lease_reclaim_hooked", False):
+            conn._lease_reclaim_hooked = True
+            prev = conn.on_close
+
+            def _reclaim(c, prev=prev):
+                if prev:
+                    prev(c)
+                self._reclaim_leases_of(c)
+
+            conn.on_close = _reclaim
         fut = asyncio.get_running_loop().create_future()
         self.pending_leases.append(((p, conn), fut))
         self._pump_leases()
@@ -519,7 +541,18 @@ class Raylet:
         buf, meta = self._store.get_buffer(p["oid"], 0.0)
         if buf is None:
             return {"data": None}
-        return {"data": bytes(memoryview(buf)), "meta": bytes(meta)}
+        from ant_ray_amd._private.object_store import PULL_CHUNK_BYTES
+
+        mv = memoryview(buf)
+        n = len(mv)
+        off = p.get("offset")
+        if off is not None:
+            end = min(n, off + p.get("length", PULL_CHUNK_BYTES))
+            return {"data": bytes(mv[off:end])}
+        if n <= PULL_CHUNK_BYTES:
+            return {"data": bytes(mv), "meta": bytes(meta)}
+        return {"data": bytes(mv[:PULL_CHUNK_BYTES]),
+                "meta": bytes(meta), "size": n}
 
     # ------------------------------------------------------------------ misc
     async def rpc_node_info(self, conn, p):

@@ -36,7 +36,12 @@ from typing import Any, Dict, List, Optional, Sequence, Tuple
 from ant_ray_amd._private import protocol, serialization
 from ant_ray_amd._private.ids import ActorID, JobID, ObjectID, TaskID, WorkerID
 from ant_ray_amd._private.object_ref import ObjectRef
-from ant_ray_amd._private.object_store import IN_PLASMA, INLINE_OBJECT_MAX, ObjectStore
+from ant_ray_amd._private.object_store import (
+    IN_PLASMA,
+    INLINE_OBJECT_MAX,
+    PULL_CHUNK_BYTES,
+    ObjectStore,
+)
 from ant_ray_amd.exceptions import (
     ActorDiedError,
     GetTimeoutError,
@@ -118,7 +123,8 @@ class CoreWorker:
         self._lock = threading.RLock()
         self._local_refs: Dict[bytes, int] = defaultdict(int)
         self._owned: Dict[bytes, dict] = {}
-        self._spilled: Dict[bytes, str] = {}  # oid -> spill file path  # oid -> {escaped, size}
+        self._spilled: Dict[bytes, str] = {}  # oid -> spill file path
+        self._pull_cache: Dict[bytes, tuple] = {}  # big-object pull slices
         self._object_locations: Dict[bytes, Tuple[str, int]] = {}
         self._actors: Dict[bytes, ActorHandleState] = {}
         self._actor_results: Dict[bytes, dict] = {}
@@ -297,26 +303,65 @@ class CoreWorker:
     async def rpc_ping(self, conn, p):
         return {"ok": True, "worker_id": self.worker_id}
 
-    async def rpc_pull_object(self, conn, p):
-        """Owner-mediated object fetch (cross-node data plane)."""
-        oid = p["oid"]
+    def _pull_source(self, oid):
+        """Resolve an object's serialized bytes for the pull data plane.
+
+        Returns (bytes_like, meta) or None. Live in-memory values are
+        serialized once and kept in a tiny cache so a chunked pull does
+        not re-serialize per chunk.
+        """
+        cached = self._pull_cache.get(oid)
+        if cached is not None:
+            return cached
         found, v = self.store.memory.get_now(oid)
+        out = None
         if found and isinstance(v, _RawResult):
-            return {"data": v.data, "meta": v.meta}
-        if found and v is not IN_PLASMA and not isinstance(v, _ErrorResult):
+            out = (v.data, v.meta)
+        elif found and v is not IN_PLASMA and not isinstance(v, _ErrorResult):
             sobj = serialization.serialize(v)
-            return {"data": sobj.to_bytes(), "meta": sobj.metadata}
-        if self.store.shm is not None:
+            out = (sobj.to_bytes(), sobj.metadata)
+        elif self.store.shm is not None:
             buf, meta = self.store.shm.get_buffer(oid, 0.0)
             if buf is not None:
-                return {"data": bytes(memoryview(buf)), "meta": bytes(meta)}
-        spilled = self._read_spilled(oid)
-        if spilled is not None:
-            return {"data": spilled[0], "meta": spilled[1]}
-        holder = self._object_locations.get(oid)
-        if holder is not None and tuple(holder) != self.addr:
-            return {"redirect": list(holder)}
-        return {"missing": True}
+                out = (bytes(memoryview(buf)), bytes(meta))
+        if out is None:
+            spilled = self._read_spilled(oid)
+            if spilled is not None:
+                out = spilled
+        if out is not None and len(out[0]) > PULL_CHUNK_BYTES:
+            # keep big objects around for the follow-up chunk requests
+            self._pull_cache[oid] = out
+            while len(self._pull_cache) > 4:
+                self._pull_cache.pop(next(iter(self._pull_cache)))
+        return out
+
+    async def rpc_pull_object(self, conn, p):
+        """Owner-mediated object fetch (cross-node data plane).
+
+        Objects larger than PULL_CHUNK_BYTES are served in 5 MiB slices
+        (role parity: reference object_manager.cc pushes objects as
+        config_.object_chunk_size chunks): the first reply carries
+        chunk 0 + total size, the puller pipelines the remaining chunk
+        requests on the same multiplexed connection.
+        """
+        oid = p["oid"]
+        off = p.get("offset")
+        src = self._pull_source(oid)
+        if src is None:
+            holder = self._object_locations.get(oid)
+            if holder is not None and tuple(holder) != self.addr:
+                return {"redirect": list(holder)}
+            return {"missing": True}
+        data, meta = src
+        n = len(data)
+        if off is not None:
+            end = min(n, off + p.get("length", PULL_CHUNK_BYTES))
+            if end >= n:
+                self._pull_cache.pop(oid, None)
+            return {"data": bytes(data[off:end])}
+        if n <= PULL_CHUNK_BYTES:
+            return {"data": bytes(data), "meta": meta}
+        return {"data": bytes(data[:PULL_CHUNK_BYTES]), "meta": meta, "size": n}
 
     async def rpc_free_objects(self, conn, p):
         self.store.free(p["oids"])
@@ -541,6 +586,26 @@ class CoreWorker:
             if reply.get("data") is not None:
                 data = reply["data"]
                 meta = reply.get("meta", serialization.META_PICKLE)
+                size = reply.get("size")
+                if size is not None and size > len(data):
+                    # big object: pipeline the remaining 5 MiB chunk
+                    # requests on the same multiplexed connection
+                    buf = bytearray(size)
+                    buf[: len(data)] = data
+                    futs = [
+                        (off, self.io.submit(conn.call(
+                            "pull_object",
+                            {"oid": oid, "offset": off,
+                             "length": PULL_CHUNK_BYTES}, timeout=60)))
+                        for off in range(len(data), size, PULL_CHUNK_BYTES)
+                    ]
+                    try:
+                        for off, f in futs:
+                            chunk = f.result(70)["data"]
+                            buf[off: off + len(chunk)] = chunk
+                    except Exception:
+                        return None, False
+                    data = buf
                 value = serialization.deserialize(memoryview(data), meta)
                 if meta in (serialization.META_ERROR, serialization.META_ACTOR_DIED):
                     if isinstance(value, RayTaskError):
@@ -571,26 +636,39 @@ class CoreWorker:
     # =================================================================== wait
     def wait(self, refs: Sequence[ObjectRef], num_returns=1, timeout=None, fetch_local=True):
         deadline = None if timeout is None else time.monotonic() + timeout
-        pending = list(refs)
+        mem = self.store.memory
+        shm = self.store.shm
+
+        def _is_ready(oid: bytes) -> bool:
+            found, v = mem.get_now(oid)
+            if found and v is not IN_PLASMA:
+                return True
+            return shm is not None and shm.contains(oid)
+
+        def _scan(candidates, ready, pending):
+            """Partition; stop checking once num_returns are ready (the
+            rest go to pending unchecked) so each pass is O(first-hits),
+            not O(all pending) — ray.wait over a draining 1k-ref list is
+            otherwise quadratic."""
+            for i, ref in enumerate(candidates):
+                if len(ready) >= num_returns:
+                    pending.extend(candidates[i:])
+                    return
+                (ready if _is_ready(ref.binary()) else pending).append(ref)
+
         ready: List[ObjectRef] = []
-        while True:
-            still = []
-            for ref in pending:
-                oid = ref.binary()
-                found, v = self.store.memory.get_now(oid)
-                ok = found and v is not IN_PLASMA
-                if not ok and self.store.shm is not None and self.store.shm.contains(oid):
-                    ok = True
-                (ready if ok else still).append(ref)
-            pending = still
-            if len(ready) >= num_returns or not pending:
-                break
+        pending: List[ObjectRef] = []
+        _scan(list(refs), ready, pending)
+        while len(ready) < num_returns and pending:
             if deadline is not None and time.monotonic() >= deadline:
                 break
-            time.sleep(0.005)
-        if len(ready) > num_returns:
-            pending = ready[num_returns:] + pending
-            ready = ready[:num_returns]
+            # task completions land in the memory store and notify its CV;
+            # the 5ms cap covers shm-only arrivals from other processes
+            with mem._cv:
+                mem._cv.wait(0.005)
+            still: List[ObjectRef] = []
+            _scan(pending, ready, still)
+            pending = still
         return ready, pending
 
     # ============================================================ task submit
@@ -659,24 +737,30 @@ class CoreWorker:
             self.io.loop.create_task(self._request_lease(key, queue[0]))
 
     async def _request_lease(self, key, sample_item):
+        req = {
+            "resources": sample_item["resources"],
+            "pg": sample_item["opts"].get("placement_group"),
+            "runtime_env": sample_item["opts"].get("runtime_env"),
+        }
+        granting_raylet = self.raylet
         try:
-            reply = await self.raylet.call(
-                "lease_worker",
-                {
-                    "resources": sample_item["resources"],
-                    "pg": sample_item["opts"].get("placement_group"),
-                    "runtime_env": sample_item["opts"].get("runtime_env"),
-                },
-                timeout=None,
-            )
+            reply = await granting_raylet.call("lease_worker", req, timeout=None)
+            # spillback (parity: raylet local_task_manager spillback — the
+            # local raylet names a feasible remote node, we lease there)
+            hops = 0
+            while (not reply.get("granted") and reply.get("spillback")
+                   and hops < 3):
+                hops += 1
+                granting_raylet = await self._get_worker_conn_async_cached(
+                    tuple(reply["spillback"]))
+                reply = await granting_raylet.call(
+                    "lease_worker", req, timeout=None)
         except Exception as e:
             logger.warning("lease request failed: %s", e)
             self._lease_requests[key] -= 1
             return
         self._lease_requests[key] -= 1
         if not reply.get("granted"):
-            # infeasible locally: fail queued tasks for now (spillback in gcs
-            # handles actors; cross-node normal tasks arrive in round 2)
             err = RaySystemError(
                 f"no feasible node for resources {sample_item['resources']}"
             )
@@ -687,6 +771,7 @@ class CoreWorker:
         conn = self._get_worker_conn_async_cached(tuple(reply["addr"]))
         conn = await conn
         lw = LeasedWorker(reply["worker_id"], reply["addr"], reply.get("node_id"), conn)
+        lw.raylet = granting_raylet
         self._leases[key].append(lw)
         await self._pump_tasks(key)
         self.io.loop.create_task(self._lease_idle_watch(key, lw))
@@ -701,7 +786,9 @@ class CoreWorker:
                     except ValueError:
                         pass
                     try:
-                        await self.raylet.call("return_worker", {"worker_id": lw.worker_id}, timeout=10)
+                        await getattr(lw, "raylet", self.raylet).call(
+                            "return_worker",
+                            {"worker_id": lw.worker_id}, timeout=10)
                     except Exception:
                         pass
                     return

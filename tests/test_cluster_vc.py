@@ -114,3 +114,48 @@ def test_unknown_virtual_cluster_rejected(cluster):
     out = subprocess.run([sys.executable, "-c", script], capture_output=True,
                          text=True, timeout=120)
     assert "REJECTED" in out.stdout, out.stdout + out.stderr
+
+
+def test_cross_node_chunked_pull(cluster):
+    """A >5MiB object created on one node is pulled by a worker on another
+    node through the chunked data plane (PULL_CHUNK_BYTES slices pipelined
+    on one connection)."""
+    import numpy as np
+
+    import ant_ray_amd as ray
+
+    cluster.connect()
+    cluster.add_node(num_cpus=2, resources={"far": 1})
+    deadline = time.time() + 30
+    while time.time() < deadline and not ray.cluster_resources().get("far"):
+        time.sleep(0.2)
+
+    arr = np.arange(3 * 1024 * 1024, dtype=np.float64)  # 24 MB -> 5 chunks
+    ref = ray.put(arr)
+
+    @ray.remote(num_cpus=1, resources={"far": 0.1})
+    def consume(x):
+        return float(x.sum()), x.shape[0]
+
+    total, n = ray.get(consume.remote(ref), timeout=120)
+    assert n == arr.shape[0]
+    assert total == pytest.approx(float(arr.sum()))
+
+
+def test_wait_many_refs_drain(cluster):
+    """ray.wait drains 300 refs one at a time (the ray_perf wait-1k shape)."""
+    import ant_ray_amd as ray
+
+    cluster.connect()
+
+    @ray.remote
+    def unit():
+        return 1
+
+    not_ready = [unit.remote() for _ in range(300)]
+    got = 0
+    deadline = time.time() + 90
+    while not_ready and time.time() < deadline:
+        ready, not_ready = ray.wait(not_ready)
+        got += len(ready)
+    assert got == 300 and not not_ready
