@@ -106,6 +106,19 @@ def main():
     timeit("single client tasks async",
            lambda: ray.get([tiny.remote() for _ in range(1000)]), 1000, dur)
 
+    timeit("single client tasks and get batch",
+           lambda: ray.get([tiny.remote() for _ in range(1000)]), 1, dur)
+
+    # an object whose payload is 10k ObjectRefs (parity ray_perf.py:149)
+    @ray.remote
+    def create_object_containing_ref():
+        return [ray.put(1) for _ in range(10000)]
+
+    obj_containing_ref = create_object_containing_ref.remote()
+    ray.get(obj_containing_ref)
+    timeit("single client get object containing 10k refs",
+           lambda: ray.get(obj_containing_ref), 1, dur)
+
     def wait_multiple_refs():
         not_ready = [tiny.remote() for _ in range(1000)]
         while not_ready:
