@@ -199,6 +199,10 @@ def get(
     timeout: Optional[float] = None,
 ):
     _check_connected()
+    from ant_ray_amd.dag.node import CompiledDAGRef
+
+    if isinstance(object_refs, CompiledDAGRef):
+        return object_refs.get(timeout)
     single = isinstance(object_refs, ObjectRef)
     refs = [object_refs] if single else list(object_refs)
     for r in refs:

@@ -270,6 +270,9 @@ class TaskExecutor:
             if ttype == "actor_task":
                 if self.actor_instance is None:
                     raise RuntimeError("no actor instance in this worker")
+                if payload["method"] == "__adag_loop__":
+                    (spec,), _ = self._deserialize_args(payload)
+                    return self._reply_results(payload, self._adag_loop(spec))
                 method = getattr(self.actor_instance, payload["method"])
                 args, kwargs = self._deserialize_args(payload)
                 result = method(*args, **kwargs)
@@ -280,6 +283,63 @@ class TaskExecutor:
             return self._error_reply(payload, e)
         finally:
             self.cw.current_task_id = None
+
+    def _adag_loop(self, spec) -> int:
+        """Resident compiled-DAG node loop (parity: the reference's
+        do_exec_tasks loop in compiled_dag_node.py). Reads every input
+        channel once per iteration, applies the bound method, writes the
+        output channel; exits when an upstream channel closes and cascades
+        the close downstream. Returns the iteration count."""
+        from ant_ray_amd.experimental.channel import (
+            ChannelClosedError,
+            _project_input,
+            _WrappedError,
+        )
+
+        ops = spec["ops"]
+        methods = {op["key"]: getattr(self.actor_instance, op["method"])
+                   for op in ops}
+        iters = 0
+        try:
+            while True:
+                values: dict = {}      # node key -> result (same-actor edges)
+                chan_cache: dict = {}  # id(chan) -> value this iteration
+                for op in ops:
+                    args = []
+                    for desc in op["ins"]:
+                        if desc[0] == "local":
+                            args.append(values[desc[1]])
+                        elif desc[0] in ("chan", "chan_key"):
+                            ch = desc[1]
+                            if id(ch) not in chan_cache:
+                                chan_cache[id(ch)] = ch.read(unwrap=False)
+                            v = chan_cache[id(ch)]
+                            if desc[0] == "chan_key" \
+                                    and not isinstance(v, _WrappedError):
+                                v = _project_input(v, desc[2])
+                            args.append(v)
+                        else:
+                            args.append(desc[1])
+                    bad = next((a for a in args
+                                if isinstance(a, _WrappedError)), None)
+                    if bad is not None:
+                        result = bad  # forward upstream errors unchanged
+                    else:
+                        try:
+                            result = methods[op["key"]](*args)
+                        except ChannelClosedError:
+                            raise
+                        except BaseException as e:  # noqa: BLE001
+                            result = _WrappedError(e)
+                    values[op["key"]] = result
+                    if op["out"] is not None:
+                        op["out"].write(result)
+                iters += 1
+        except ChannelClosedError:
+            for op in ops:
+                if op["out"] is not None:
+                    op["out"].close()
+            return iters
 
     def _run_async_actor_task(self, payload, done):
         async def runner():

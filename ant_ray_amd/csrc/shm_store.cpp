@@ -355,6 +355,107 @@ class ShmStore : public std::enable_shared_from_this<ShmStore> {
 
   const std::string& path() const { return path_; }
 
+  // --- mutable channels -------------------------------------------------------
+  // A channel is a pinned, sealed store object whose payload is
+  // [ChanHeader | capacity bytes]. Single writer, num_readers readers.
+  // Writer blocks until every reader consumed the previous message
+  // (reference parity: experimental/channel shared-memory channels used by
+  // compiled graphs — python/ray/experimental/channel/shared_memory_channel.py);
+  // readers block on a version futex, so a hop costs a memcpy + futex wake
+  // instead of an RPC round trip.
+  struct ChanHeader {
+    std::atomic<uint32_t> version;     // futex: bumped on every write/close
+    std::atomic<uint32_t> reads_done;  // futex: readers finished this version
+    uint32_t num_readers;
+    std::atomic<uint32_t> closed;
+    uint64_t msg_size;  // UINT64_MAX = closed sentinel
+    uint8_t _pad[40];
+  };
+  static_assert(sizeof(ChanHeader) == 64, "chan header one cacheline");
+
+  uint64_t ChannelCreate(const std::string& oid, uint64_t capacity,
+                         uint32_t num_readers) {
+    uint64_t doff = CreateObject(oid, sizeof(ChanHeader) + capacity, "chan");
+    auto* h = reinterpret_cast<ChanHeader*>(Ptr(doff));
+    h->version.store(0, std::memory_order_relaxed);
+    h->reads_done.store(num_readers, std::memory_order_relaxed);  // "consumed"
+    h->num_readers = num_readers;
+    h->closed.store(0, std::memory_order_relaxed);
+    h->msg_size = 0;
+    Seal(oid);  // creator's pin (refcount 1) keeps it resident
+    return capacity;
+  }
+
+  void ChannelWrite(const std::string& oid, py::bytes data, double timeout_s) {
+    uint64_t cap;
+    ChanHeader* h = ChanPtr(oid, &cap);
+    char* buf;
+    ssize_t n;
+    if (PYBIND11_BYTES_AS_STRING_AND_SIZE(data.ptr(), &buf, &n) != 0)
+      throw py::value_error("bad bytes");
+    if ((uint64_t)n > cap) throw py::value_error("message exceeds channel capacity");
+    {
+      py::gil_scoped_release nogil;
+      struct timespec slice = {0, 50000000};
+      double waited = 0;
+      while (true) {
+        if (h->closed.load(std::memory_order_acquire))
+          throw std::runtime_error("channel closed");
+        uint32_t rd = h->reads_done.load(std::memory_order_acquire);
+        if (rd >= h->num_readers) break;
+        if (timeout_s >= 0 && waited >= timeout_s)
+          throw std::runtime_error("channel write timed out");
+        futex_wait(&h->reads_done, rd, &slice);
+        waited += 0.05;
+      }
+      std::memcpy(reinterpret_cast<char*>(h) + sizeof(ChanHeader), buf, n);
+      h->msg_size = (uint64_t)n;
+      h->reads_done.store(0, std::memory_order_release);
+      h->version.fetch_add(1, std::memory_order_release);
+      futex_wake_all(&h->version);
+    }
+  }
+
+  // Returns (payload_bytes, new_version); raises on close/timeout.
+  py::tuple ChannelRead(const std::string& oid, uint32_t last_version,
+                        double timeout_s) {
+    uint64_t cap;
+    ChanHeader* h = ChanPtr(oid, &cap);
+    std::string out;
+    uint32_t v;
+    {
+      py::gil_scoped_release nogil;
+      struct timespec slice = {0, 50000000};
+      double waited = 0;
+      while (true) {
+        v = h->version.load(std::memory_order_acquire);
+        if (v != last_version) break;
+        if (h->closed.load(std::memory_order_acquire))
+          throw std::runtime_error("channel closed");
+        if (timeout_s >= 0 && waited >= timeout_s)
+          throw std::runtime_error("channel read timed out");
+        futex_wait(&h->version, v, &slice);
+        waited += 0.05;
+      }
+      uint64_t n = h->msg_size;
+      if (n == UINT64_MAX) throw std::runtime_error("channel closed");
+      out.assign(reinterpret_cast<char*>(h) + sizeof(ChanHeader), n);
+      h->reads_done.fetch_add(1, std::memory_order_acq_rel);
+      futex_wake_all(&h->reads_done);
+    }
+    return py::make_tuple(py::bytes(out), v);
+  }
+
+  void ChannelClose(const std::string& oid) {
+    uint64_t cap;
+    ChanHeader* h = ChanPtr(oid, &cap);
+    h->closed.store(1, std::memory_order_release);
+    h->msg_size = UINT64_MAX;
+    h->version.fetch_add(1, std::memory_order_release);
+    futex_wake_all(&h->version);
+    futex_wake_all(&h->reads_done);
+  }
+
  private:
   ShmStore(void* base, uint64_t size, std::string path)
       : base_(base), size_(size), path_(std::move(path)) {}
@@ -364,6 +465,17 @@ class ShmStore : public std::enable_shared_from_this<ShmStore> {
 
   static void CheckId(const std::string& oid) {
     if (oid.size() != kIdLen) throw py::value_error("object id must be 20 bytes");
+  }
+
+  // Resolve a channel's header pointer (pinned objects never move).
+  ChanHeader* ChanPtr(const std::string& oid, uint64_t* cap_out) {
+    CheckId(oid);
+    Guard g(this);
+    Slot* s = FindSlot(oid.data());
+    if (!s || s->state != SLOT_SEALED) throw py::key_error("no such channel");
+    if (s->data_size < sizeof(ChanHeader)) throw py::value_error("not a channel");
+    *cap_out = s->data_size - sizeof(ChanHeader);
+    return reinterpret_cast<ChanHeader*>(Ptr(s->offset + align_up(s->meta_size, 8)));
   }
 
   struct Guard {
@@ -603,6 +715,12 @@ PYBIND11_MODULE(_shm_store, m) {
       .def("release", &ShmStore::Release)
       .def("delete", &ShmStore::Delete)
       .def("stats", &ShmStore::Stats)
+      .def("channel_create", &ShmStore::ChannelCreate)
+      .def("channel_write", &ShmStore::ChannelWrite, py::arg("oid"),
+           py::arg("data"), py::arg("timeout_s") = -1.0)
+      .def("channel_read", &ShmStore::ChannelRead, py::arg("oid"),
+           py::arg("last_version"), py::arg("timeout_s") = -1.0)
+      .def("channel_close", &ShmStore::ChannelClose)
       .def("view_at", &ShmStore::ViewAt, py::arg("offset"), py::arg("size"),
            py::arg("writable") = false)
       .def("address_at", &ShmStore::AddressAt)

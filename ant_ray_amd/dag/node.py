@@ -43,7 +43,9 @@ class DAGNode:
                 {k: r(v) for k, v in self._bound_kwargs.items()})
 
     def experimental_compile(self, **kwargs) -> "CompiledDAG":
-        return CompiledDAG(self)
+        return CompiledDAG(
+            self,
+            _buffer_size_bytes=kwargs.get("_buffer_size_bytes", 1 << 20))
 
     # --------------------------------------------------------- introspection
 
@@ -197,14 +199,33 @@ class MultiOutputNode(DAGNode):
         return list(args)
 
 
-class CompiledDAG:
-    """Parity: compiled_dag_node.py:805. Pre-creates every ClassNode's actor
-    and freezes the node topology; execute() then only submits actor calls /
-    tasks in topological order. Returns a single ObjectRef (or list for
-    MultiOutputNode) like the reference's CompiledDAGRef."""
+class CompiledDAGRef:
+    """Result handle of a channel-mode execute (parity: CompiledDAGRef).
+    ray.get() on it drains the output channel in execution order."""
 
-    def __init__(self, root: DAGNode):
+    def __init__(self, dag: "CompiledDAG", seq: int):
+        self._dag = dag
+        self._seq = seq
+
+    def get(self, timeout=None):
+        return self._dag._result_for(self._seq, timeout)
+
+
+class CompiledDAG:
+    """Parity: compiled_dag_node.py:805 CompiledDAG.
+
+    Channel mode (the aDAG fast path): when the graph is actor-method
+    nodes on one node, every edge becomes a mutable shm channel
+    (`experimental/channel.py`) and each actor runs a resident
+    `__adag_loop__` (task_executor._adag_loop) — execute() is then one
+    channel write + one channel read instead of per-node RPCs. Falls back
+    to re-submitting actor calls per execute (the pre-compile path) for
+    graphs with task nodes, multi-node clusters, or no local shm.
+    """
+
+    def __init__(self, root: DAGNode, _buffer_size_bytes: int = 1 << 20):
         self._root = root
+        self._buffer_size = _buffer_size_bytes
         # materialize all actors up front
         seen = set()
         stack = [root]
@@ -218,13 +239,203 @@ class CompiledDAG:
             if isinstance(n, ClassMethodNode):
                 n._class_node._get_actor()
             stack.extend(n._upstream())
+        self._channel_mode = False
+        self._torn_down = False
+        self._seq = 0
+        self._read_seq = 0
+        self._results: dict = {}
+        try:
+            self._channel_mode = self._compile_channels()
+        except Exception:
+            self._channel_mode = False
 
+    # ------------------------------------------------------- channel compile
+    def _compile_channels(self) -> bool:
+        import ant_ray_amd as ray
+        from ant_ray_amd._private.worker import global_worker
+        from ant_ray_amd.experimental.channel import Channel
+
+        cw = global_worker.core_worker
+        if cw is None or cw.store.shm is None:
+            return False
+        if len([n for n in ray.nodes() if n.get("Alive")]) != 1:
+            return False  # shm channels are single-node
+
+        # classify the graph: method nodes + input (+ MultiOutputNode root)
+        outputs: List[DAGNode]
+        if isinstance(self._root, MultiOutputNode):
+            outputs = [a for a in self._root._bound_args]
+        else:
+            outputs = [self._root]
+        method_nodes: List[ClassMethodNode] = []
+        input_chan_nodes: list = []  # InputNode / InputAttributeNode
+        seen = set()
+        stack = list(outputs)
+        while stack:
+            n = stack.pop()
+            if id(n) in seen:
+                continue
+            seen.add(id(n))
+            if isinstance(n, ClassMethodNode):
+                method_nodes.append(n)
+                for a in n._bound_args:
+                    if isinstance(a, DAGNode):
+                        stack.append(a)
+                if n._bound_kwargs and any(
+                        isinstance(v, DAGNode)
+                        for v in n._bound_kwargs.values()):
+                    return False
+            elif isinstance(n, (InputNode, InputAttributeNode)):
+                input_chan_nodes.append(n)
+            else:
+                return False
+        if not method_nodes:
+            return False
+        if any(not isinstance(o, ClassMethodNode) for o in outputs):
+            return False
+
+        # every node must consume at least one channel/upstream value, else
+        # its resident loop would produce unboundedly — leave such on RPC
+        for m in method_nodes:
+            if not any(isinstance(a, DAGNode) for a in m._bound_args):
+                return False
+
+        # topological order of method nodes (DFS postorder from outputs)
+        order: List[ClassMethodNode] = []
+        marked: set = set()
+
+        def visit(n):
+            if id(n) in marked or not isinstance(n, ClassMethodNode):
+                return
+            marked.add(id(n))
+            for a in n._bound_args:
+                if isinstance(a, DAGNode):
+                    visit(a)
+            order.append(n)
+
+        for o in outputs:
+            visit(o)
+
+        def actor_key(m: ClassMethodNode):
+            return bytes(m._class_node._get_actor()._ray_actor_id)
+
+        # which ACTORS (or the driver) consume each produced value — a
+        # channel is only needed for cross-actor/driver edges; same-actor
+        # edges pass through the loop's local value map
+        out_ids = {id(o) for o in outputs}
+        readers: dict = {}  # "input" | id(node) -> set(actor_key | b"driver")
+        for m in order:
+            ak = actor_key(m)
+            for a in m._bound_args:
+                if isinstance(a, (InputNode, InputAttributeNode)):
+                    readers.setdefault("input", set()).add(ak)
+                elif isinstance(a, ClassMethodNode):
+                    if actor_key(a) != ak:
+                        readers.setdefault(id(a), set()).add(ak)
+        for o in outputs:
+            readers.setdefault(id(o), set()).add(b"driver")
+
+        n_input_readers = len(readers.get("input", ()))
+        self._in_chan = (Channel(self._buffer_size, n_input_readers)
+                         if n_input_readers else None)
+        chan_of: dict = {}
+        for m in order:
+            n_readers = len(readers.get(id(m), ()))
+            chan_of[id(m)] = (Channel(self._buffer_size, n_readers)
+                              if n_readers else None)
+
+        # ONE resident loop per actor running all its ops in topo order
+        # (parity: compiled_dag_node do_exec_tasks — an actor is dedicated
+        # to the DAG and executes its bound tasks every iteration)
+        specs_by_actor: dict = {}
+        handles: dict = {}
+        for m in order:
+            ak = actor_key(m)
+            handles[ak] = m._class_node._get_actor()
+            ins = []
+            for a in m._bound_args:
+                if isinstance(a, InputNode):
+                    ins.append(("chan", self._in_chan))
+                elif isinstance(a, InputAttributeNode):
+                    ins.append(("chan_key", self._in_chan, a._key))
+                elif isinstance(a, ClassMethodNode):
+                    if actor_key(a) == ak:
+                        ins.append(("local", id(a)))
+                    else:
+                        ins.append(("chan", chan_of[id(a)]))
+                else:
+                    ins.append(("const", a))
+            specs_by_actor.setdefault(ak, []).append(
+                {"method": m._method, "key": id(m), "ins": ins,
+                 "out": chan_of[id(m)]})
+
+        self._loop_refs = []
+        for ak, ops in specs_by_actor.items():
+            refs = handles[ak]._actor_method_call(
+                "__adag_loop__", ({"ops": ops},), {}, {})
+            self._loop_refs.append(refs[0] if isinstance(refs, list) else refs)
+        self._out_chans = [chan_of[id(o)] for o in outputs]
+        self._multi = isinstance(self._root, MultiOutputNode)
+        self._all_chans = [c for c in chan_of.values() if c is not None]
+        if self._in_chan is not None:
+            self._all_chans.append(self._in_chan)
+        return True
+
+    # --------------------------------------------------------------- execute
     def execute(self, *args, **kwargs):
-        return self._root.execute(*args, **kwargs)
+        if not self._channel_mode:
+            return self._root.execute(*args, **kwargs)
+        if self._torn_down:
+            raise RuntimeError("compiled DAG was torn down")
+        if len(args) == 1 and not kwargs:
+            value = args[0]
+        elif not args and not kwargs:
+            value = None
+        else:
+            value = _DagInput(args, kwargs)
+        if self._in_chan is not None:
+            self._in_chan.write(value, timeout=60)
+        self._seq += 1
+        return CompiledDAGRef(self, self._seq)
+
+    def _result_for(self, seq: int, timeout=None):
+        from ant_ray_amd.experimental.channel import _WrappedError
+
+        while self._read_seq < seq:
+            # read unwrapped: an error result must still advance the read
+            # cursor, else the next get() desynchronizes from the channel
+            vals = [c.read(timeout, unwrap=False) for c in self._out_chans]
+            self._read_seq += 1
+            self._results[self._read_seq] = (
+                vals if self._multi else vals[0])
+            # only the latest few results are retained
+            self._results.pop(self._read_seq - 8, None)
+        try:
+            out = self._results[seq]
+        except KeyError:
+            raise RuntimeError("compiled DAG result no longer buffered")
+        if isinstance(out, _WrappedError):
+            raise out.exc
+        if isinstance(out, list):
+            for v in out:
+                if isinstance(v, _WrappedError):
+                    raise v.exc
+        return out
 
     def teardown(self):
         import ant_ray_amd as ray
 
+        if self._channel_mode and not self._torn_down:
+            self._torn_down = True
+            if self._in_chan is not None:
+                self._in_chan.close()
+            for ref in self._loop_refs:
+                try:
+                    ray.get(ref, timeout=10)
+                except Exception:
+                    pass
+            for c in self._all_chans:
+                c.destroy()
         seen = set()
         stack = [self._root]
         while stack:

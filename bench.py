@@ -25,6 +25,10 @@ def _setup_tunableop(local_rank: int):
     pre-tuned table for the model's GEMM shapes (profiles/tunableop_gfx950
     .csv, measured +4-8% over the default heuristics on gfx950); set
     ANTRAY_TUNE=1 to (re)tune and write a fresh table."""
+    if os.environ.get("ANTRAY_TUNEOP") == "0":
+        # TunableOp's hipblasLtCreate aborts under rocprofv3 counter
+        # collection — profiling runs disable it
+        return
     try:
         import torch.cuda.tunable as tun
     except ImportError:

@@ -126,3 +126,79 @@ def test_input_attribute_node(ray_mod):
     with InputNode() as inp:
         dag = f.bind(inp["k"])
     assert ray.get(dag.execute({"k": 21})) == 42
+
+
+class TestCompiledChannelDAG:
+    """aDAG channel mode: edges are mutable shm channels, actors run
+    resident __adag_loop__ fibers (csrc ChanHeader + experimental/channel)."""
+
+    def test_linear_chain(self, ray_mod):
+        import ant_ray_amd as ray
+        from ant_ray_amd.dag import InputNode
+
+        @ray.remote
+        class Adder:
+            def __init__(self, k):
+                self.k = k
+
+            def add(self, x):
+                return x + self.k
+
+        a = Adder.remote(1)
+        b = Adder.remote(100)
+        with InputNode() as inp:
+            dag = b.add.bind(a.add.bind(inp)).experimental_compile()
+        assert dag._channel_mode, "single-node actor chain must use channels"
+        for i in range(20):
+            assert ray.get(dag.execute(i)) == i + 101
+        dag.teardown()
+        # existing handles survive teardown and are usable again
+        assert ray.get(a.add.remote(5), timeout=30) == 6
+
+    def test_diamond_multi_output_and_errors(self, ray_mod):
+        import pytest as _pytest
+
+        import ant_ray_amd as ray
+        from ant_ray_amd.dag import InputNode, MultiOutputNode
+
+        @ray.remote
+        class W:
+            def __init__(self, k):
+                self.k = k
+
+            def add(self, x):
+                return x + self.k
+
+            def mul2(self, x):
+                return x * 2
+
+            def combine(self, a, b):
+                return a + b
+
+            def boom(self, x):
+                if x == 13:
+                    raise ValueError("boom13")
+                return x
+
+        a = W.remote(1)
+        b = W.remote(100)
+        with InputNode() as inp:
+            p = a.add.bind(inp["v"])
+            q = a.mul2.bind(p)
+            r = b.combine.bind(p, q)
+            dag = MultiOutputNode([q, r]).experimental_compile()
+        assert dag._channel_mode
+        out = ray.get(dag.execute({"v": 5}))
+        assert out == [12, 18]
+        dag.teardown()
+
+        # node exceptions surface at get() and the loop keeps running
+        c = W.remote(0)
+        with InputNode() as inp:
+            d2 = c.boom.bind(inp).experimental_compile()
+        assert d2._channel_mode
+        assert ray.get(d2.execute(1)) == 1
+        with _pytest.raises(ValueError, match="boom13"):
+            ray.get(d2.execute(13))
+        assert ray.get(d2.execute(2)) == 2
+        d2.teardown()
