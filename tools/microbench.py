@@ -8,7 +8,9 @@ Prints one JSON line per metric: {"name", "value", "unit"}.
 import argparse
 import json
 import os
+import subprocess
 import sys
+import textwrap
 import time
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
@@ -232,6 +234,47 @@ def main():
             remove_placement_group(pg)
 
     timeit("placement group create/removal", pg_cycle, 10, dur)
+
+    # ---- Ray Client plane (parity: ray_client_microbenchmark.py) — a
+    # separate driver process WITHOUT local shm, objects proxied through
+    # the raylet data plane
+    addr = ray.get_runtime_context().gcs_address
+    client_script = textwrap.dedent(f"""
+        import json, os, sys, time
+        os.environ["ANTRAY_FORCE_CLIENT"] = "1"
+        sys.path.insert(0, {os.path.dirname(os.path.dirname(os.path.abspath(__file__)))!r})
+        import ant_ray_amd as ray
+        ray.init(address="ray://{addr}")
+
+        def timeit(name, fn, mult=1, duration={dur}):
+            fn()
+            start = time.time(); count = 0
+            while time.time() - start < duration:
+                fn(); count += 1
+            print(json.dumps({{"name": name,
+                               "value": round(count * mult / (time.time() - start), 1),
+                               "unit": "ops/s"}}), flush=True)
+
+        ref = ray.put(b"ok")
+        timeit("Ray Client: get calls", lambda: [ray.get(ref) for _ in range(20)], 20)
+        timeit("Ray Client: put calls", lambda: [ray.put(0) for _ in range(20)], 20)
+
+        @ray.remote
+        class A:
+            def ping(self):
+                return b"ok"
+
+        a = A.remote(); ray.get(a.ping.remote())
+        timeit("Ray Client: 1:1 actor calls sync",
+               lambda: [ray.get(a.ping.remote()) for _ in range(20)], 20)
+        ray.shutdown()
+    """)
+    r = subprocess.run([sys.executable, "-c", client_script],
+                       capture_output=True, text=True, timeout=120)
+    sys.stdout.write(r.stdout)
+    if r.returncode != 0:
+        print(json.dumps({"name": "Ray Client", "error": r.stderr[-500:]}),
+              flush=True)
 
     ray.shutdown()
 
