@@ -22,6 +22,11 @@ class ScalingConfig:
     trainer_resources: Optional[Dict[str, float]] = None
     placement_strategy: str = "PACK"
     accelerator_type: Optional[str] = None
+    # elastic training (ant-fork parity): when set, a (re)started worker
+    # group may run with fewer than num_workers — down to min_workers —
+    # sized to the resources actually available, resuming from the latest
+    # checkpoint; it grows back toward num_workers on later restarts.
+    min_workers: Optional[int] = None
 
     @property
     def _resources_per_worker_not_none(self) -> Dict[str, float]:

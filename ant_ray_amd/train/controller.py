@@ -13,6 +13,7 @@ import logging
 import os
 import shutil
 import time
+from dataclasses import replace
 from typing import Any, Callable, Dict, List, Optional
 
 from ant_ray_amd.train._checkpoint import Checkpoint
@@ -114,12 +115,37 @@ class TrainController:
                 shards[i][name] = splits[i] if splits else ds
         return shards
 
+    def _elastic_size(self) -> int:
+        """Pick the attempt's world size: num_workers when resources allow,
+        else the largest feasible size >= min_workers (elastic training —
+        each attempt restarts from the latest checkpoint, so shrinking or
+        regrowing between attempts is safe)."""
+        sc = self.scaling
+        if not sc.min_workers or sc.min_workers >= sc.num_workers:
+            return sc.num_workers
+        try:
+            import ant_ray_amd as ray
+
+            per = sc._resources_per_worker_not_none
+            avail = ray.available_resources()
+            n_fit = min(int(avail.get(k, 0.0) / v)
+                        for k, v in per.items() if v > 0)
+        except Exception:
+            return sc.num_workers
+        return max(sc.min_workers, min(sc.num_workers, n_fit))
+
     def run(self) -> Result:
         max_failures = self.run_config.failure_config.max_failures
         attempt = 0
         error: Optional[BaseException] = None
         while True:
-            group = WorkerGroup(self.scaling, self.torch_config, self.name,
+            n = self._elastic_size()
+            scaling = (self.scaling if n == self.scaling.num_workers
+                       else replace(self.scaling, num_workers=n))
+            if n != self.scaling.num_workers:
+                logger.warning("elastic: running with %d/%d workers",
+                               n, self.scaling.num_workers)
+            group = WorkerGroup(scaling, self.torch_config, self.name,
                                 self.experiment_path)
             try:
                 group.start()
@@ -130,7 +156,7 @@ class TrainController:
                     restore = self.resume_from_checkpoint.path
                 # keep shard iterators referenced for the whole attempt: they
                 # hold the SplitCoordinator actor handle alive
-                self._dataset_shards = self._split_datasets(self.scaling.num_workers)
+                self._dataset_shards = self._split_datasets(n)
                 group.start_training(
                     self.train_fn, self.train_loop_config, restore,
                     self._dataset_shards,

@@ -197,3 +197,32 @@ def test_prepare_data_loader_shards(ray_mod):
         train.report({"seen": seen})
 
     TorchTrainer(train_fn, scaling_config=ScalingConfig(num_workers=2)).fit()
+
+
+def test_elastic_shrinks_to_available(ray_mod, tmp_path_factory):
+    """Elastic training (ScalingConfig.min_workers): with only ~4 free CPUs
+    a num_workers=16 job still runs, shrunk to whatever fits."""
+    from ant_ray_amd.train import RunConfig, ScalingConfig
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    storage = str(tmp_path_factory.mktemp("storage"))
+
+    def train_fn(config):
+        import torch.distributed as dist
+
+        from ant_ray_amd import train
+
+        ctx = train.get_context()
+        world = ctx.get_world_size()
+        assert dist.get_world_size() == world
+        train.report({"world": world})
+
+    trainer = TorchTrainer(
+        train_fn,
+        scaling_config=ScalingConfig(num_workers=16, min_workers=1),
+        run_config=RunConfig(name="elastic", storage_path=storage),
+    )
+    result = trainer.fit()
+    assert result.error is None
+    # shrunk below 16, at least min_workers, bounded by the 6-CPU session
+    assert 1 <= result.metrics["world"] <= 6
