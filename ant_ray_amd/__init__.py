@@ -282,6 +282,7 @@ def nodes() -> List[dict]:
             "Resources": n["resources_total"],
             "Available": n["resources_available"],
             "ObjectStoreSocketName": n["store_path"],
+            "Labels": n.get("labels") or {},
         }
         for n in table
     ]

@@ -49,6 +49,7 @@ class NodeInfo:
         self.resources_available: Dict[str, float] = dict(resources)
         self.store_path: str = store_path
         self.object_store_bytes = object_store_bytes
+        self.labels: dict = {}
         self.alive = True
         self.last_heartbeat = time.monotonic()
         self.conn: Optional[Connection] = None
@@ -61,6 +62,7 @@ class NodeInfo:
             "resources_available": self.resources_available,
             "store_path": self.store_path,
             "alive": self.alive,
+            "labels": self.labels,
         }
 
 
@@ -174,6 +176,7 @@ class GcsServer:
             p["node_id"], tuple(p["addr"]), p.get("resources", {}),
             p.get("store_path", ""), p.get("object_store_bytes", 0),
         )
+        node.labels = dict(p.get("labels") or {})
         node.conn = conn
         self.nodes[node.node_id] = node
         conn.session["node_id"] = node.node_id
@@ -261,7 +264,8 @@ class GcsServer:
     async def rpc_pick_raylet(self, conn, p):
         """Name a feasible raylet for a task lease that is infeasible on the
         requester's node (spillback target)."""
-        node = self._pick_node(dict(p.get("resources") or {}))
+        node = self._pick_node(dict(p.get("resources") or {}),
+                               label_selector=p.get("_label_selector"))
         if node is None:
             return {"addr": None}
         return {"addr": list(node.addr), "node_id": node.node_id}
@@ -418,7 +422,8 @@ class GcsServer:
 
     def _pick_node(self, resources: Dict[str, float], pg: Optional[dict] = None,
                    node_affinity: Optional[bytes] = None,
-                   allowed: Optional[set] = None) -> Optional[NodeInfo]:
+                   allowed: Optional[set] = None,
+                   label_selector: Optional[dict] = None) -> Optional[NodeInfo]:
         if pg:
             pg_info = self.pgs.get(pg["pg_id"])
             if pg_info and pg_info.state == "CREATED":
@@ -440,6 +445,12 @@ class GcsServer:
                 continue
             if allowed is not None and node.node_id not in allowed:
                 continue
+            if label_selector:
+                from ant_ray_amd._private.raylet import _labels_match
+
+                if not _labels_match(label_selector.get("hard") or {},
+                                     node.labels):
+                    continue
             feasible = all(
                 node.resources_total.get(k, 0) >= v for k, v in resources.items()
             )
@@ -452,6 +463,11 @@ class GcsServer:
             score += node.resources_available.get("CPU", 0) / max(
                 node.resources_total.get("CPU", 1), 1
             )
+            if label_selector and label_selector.get("soft"):
+                from ant_ray_amd._private.raylet import _labels_match
+
+                if _labels_match(label_selector["soft"], node.labels):
+                    score += 10.0  # prefer soft-matching nodes
             if score > best_score:
                 best, best_score = node, score
         return best
@@ -469,7 +485,8 @@ class GcsServer:
         vc_id = opts.get("virtual_cluster_id") or self.worker_vc.get(info.owner)
         while True:
             node = self._pick_node(resources, pg, opts.get("_node_affinity"),
-                                   self._vc_allowed(vc_id))
+                                   self._vc_allowed(vc_id),
+                                   opts.get("_label_selector"))
             if node is not None and node.conn is not None and not node.conn.closed:
                 # optimistic accounting: concurrent schedulings must not all
                 # pile onto the same node while its heartbeat is stale (the

@@ -75,6 +75,20 @@ class WorkerProc:
         self.idle_since = time.monotonic()
 
 
+def _labels_match(hard: dict, labels: dict) -> bool:
+    """True when every hard constraint matches this node's labels; a list
+    value means "in" (reference label-selector semantics, common.proto
+    LabelMatchExpressions reduced to equality/in)."""
+    for k, v in (hard or {}).items():
+        have = labels.get(k)
+        if isinstance(v, (list, tuple, set)):
+            if have not in v:
+                return False
+        elif have != v:
+            return False
+    return True
+
+
 class Raylet:
     def __init__(self, node_ip: str, gcs_addr: Tuple[str, int], resources: Dict[str, float],
                  store_path: str, store_capacity: int, session_dir: str):
@@ -86,6 +100,8 @@ class Raylet:
         self.store_path = store_path
         self.store_capacity = store_capacity
         self.session_dir = session_dir
+        # node labels (NodeLabelSchedulingStrategy parity); auto node-id label
+        self.labels: dict = {"ray.io/node_id": self.node_id.hex()}
         self.workers: Dict[bytes, WorkerProc] = {}
         self.idle_workers: List[WorkerProc] = []
         self.pending_leases: List[Tuple[dict, asyncio.Future]] = []
@@ -121,6 +137,7 @@ class Raylet:
                 "resources": self.resources_total,
                 "store_path": self.store_path,
                 "object_store_bytes": self.store_capacity,
+                "labels": self.labels,
             },
         )
         asyncio.get_running_loop().create_task(self._heartbeat_loop())
@@ -341,13 +358,27 @@ class Raylet:
         logger.info("lease request: %s", p.get("resources"))
         res = dict(p.get("resources") or {})
         pg = p.get("pg")
+        selector = p.get("_label_selector")
+        if selector and not _labels_match(selector.get("hard") or {}, self.labels):
+            reply = {"granted": False, "infeasible": True}
+            try:
+                r = await self.gcs_conn.call(
+                    "pick_raylet",
+                    {"resources": res, "_label_selector": selector}, timeout=5)
+                if r.get("addr"):
+                    reply["spillback"] = r["addr"]
+            except Exception:
+                pass
+            return reply
         if not pg and not self._feasible(res):
             reply = {"granted": False, "infeasible": True}
             try:
                 # ask the GCS for a node whose TOTAL resources fit -> the
                 # client re-leases there (spillback parity)
                 r = await self.gcs_conn.call(
-                    "pick_raylet", {"resources": res}, timeout=5)
+                    "pick_raylet",
+                    {"resources": res,
+                     "_label_selector": p.get("_label_selector")}, timeout=5)
                 if r.get("addr"):
                     reply["spillback"] = r["addr"]
             except Exception:
@@ -595,6 +626,9 @@ async def run_raylet(args):
         args.node_ip, (host, int(port)), resources, args.store_path,
         args.store_capacity, args.session_dir,
     )
+    for src in (os.environ.get("RAY_NODE_LABELS", ""), args.labels):
+        if src:
+            raylet.labels.update(json.loads(src))
     await raylet.start(args.port)
     if args.announce_fd:
         os.write(args.announce_fd, (str(raylet.port) + "\n").encode())
@@ -610,6 +644,7 @@ def main():
     ap.add_argument("--num-cpus", type=int, default=None)
     ap.add_argument("--num-gpus", type=int, default=None)
     ap.add_argument("--resources", default="")
+    ap.add_argument("--labels", default="")
     ap.add_argument("--store-path", required=True)
     ap.add_argument("--store-capacity", type=int, default=2 * 1024**3)
     ap.add_argument("--session-dir", required=True)

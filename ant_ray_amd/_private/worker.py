@@ -741,6 +741,7 @@ class CoreWorker:
             "resources": sample_item["resources"],
             "pg": sample_item["opts"].get("placement_group"),
             "runtime_env": sample_item["opts"].get("runtime_env"),
+            "_label_selector": sample_item["opts"].get("_label_selector"),
         }
         granting_raylet = self.raylet
         try:
@@ -889,6 +890,7 @@ class CoreWorker:
                             "num_cpus", "num_gpus", "resources", "max_restarts",
                             "max_task_retries", "max_concurrency", "lifetime",
                             "placement_group", "runtime_env", "_node_affinity",
+                            "_label_selector",
                             "_scheduling_timeout",
                         )
                     },

@@ -62,7 +62,8 @@ class Cluster:
     def add_node(self, num_cpus: int = 1, num_gpus: int = 0,
                  resources: Optional[Dict[str, float]] = None,
                  object_store_memory: int = 512 * 1024 * 1024,
-                 wait: bool = True, **_) -> ClusterNode:
+                 wait: bool = True, labels: Optional[Dict[str, str]] = None,
+                 **_) -> ClusterNode:
         assert self.gcs_address, "head must be started first"
         session_dir = self.head.info["session_dir"]
         store_path = os.path.join(
@@ -81,6 +82,8 @@ class Cluster:
         ]
         if resources:
             cmd += ["--resources", json.dumps(resources)]
+        if labels:
+            cmd += ["--labels", json.dumps(labels)]
         proc = subprocess.Popen(cmd, pass_fds=(w_fd,),
                                 stderr=open(os.path.join(
                                     session_dir, "logs",

@@ -26,7 +26,14 @@ def _normalize_opts(opts: Dict[str, Any]) -> Dict[str, Any]:
             PlacementGroupSchedulingStrategy,
         )
 
-        if isinstance(strategy, PlacementGroupSchedulingStrategy):
+        from ant_ray_amd.util.scheduling_strategies import (
+            NodeLabelSchedulingStrategy,
+        )
+
+        if isinstance(strategy, NodeLabelSchedulingStrategy):
+            out["_label_selector"] = {"hard": strategy.hard,
+                                      "soft": strategy.soft}
+        elif isinstance(strategy, PlacementGroupSchedulingStrategy):
             pg = strategy.placement_group
             out["placement_group"] = {
                 "pg_id": pg.id.binary() if hasattr(pg.id, "binary") else pg.id,
@@ -35,6 +42,11 @@ def _normalize_opts(opts: Dict[str, Any]) -> Dict[str, Any]:
         elif isinstance(strategy, NodeAffinitySchedulingStrategy):
             nid = strategy.node_id
             out["_node_affinity"] = bytes.fromhex(nid) if isinstance(nid, str) else nid
+    # Ray 2.x plain label_selector option: dict of hard constraints
+    sel = out.pop("label_selector", None)
+    if sel:
+        cur = out.setdefault("_label_selector", {"hard": {}, "soft": {}})
+        cur["hard"] = {**cur.get("hard", {}), **sel}
     return out
 
 

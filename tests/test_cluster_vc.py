@@ -159,3 +159,41 @@ def test_wait_many_refs_drain(cluster):
         ready, not_ready = ray.wait(not_ready)
         got += len(ready)
     assert got == 300 and not not_ready
+
+
+def test_node_label_scheduling(cluster):
+    """NodeLabelSchedulingStrategy: hard label constraints route actors and
+    tasks to matching nodes (including lease spillback for tasks)."""
+    import ant_ray_amd as ray
+    from ant_ray_amd.util.scheduling_strategies import (
+        NodeLabelSchedulingStrategy,
+    )
+
+    cluster.connect()
+    cluster.add_node(num_cpus=2, labels={"accel": "mi355x", "zone": "a"})
+    deadline = time.time() + 30
+    while time.time() < deadline and len(
+            [n for n in ray.nodes() if n["Alive"]]) < 2:
+        time.sleep(0.2)
+
+    @ray.remote(num_cpus=1,
+                scheduling_strategy=NodeLabelSchedulingStrategy(
+                    hard={"accel": "mi355x"}))
+    class Pinned:
+        def node(self):
+            return ray.get_runtime_context().get_node_id()
+
+    a = Pinned.remote()
+    nid = ray.get(a.node.remote(), timeout=60)
+    labeled = [n for n in ray.nodes()
+               if n["Resources"].get("CPU") and not n.get("is_head")]
+    assert nid is not None
+
+    @ray.remote(num_cpus=1,
+                scheduling_strategy=NodeLabelSchedulingStrategy(
+                    hard={"accel": ["mi355x", "mi300x"]}))
+    def where():
+        return ray.get_runtime_context().get_node_id()
+
+    tid = ray.get(where.remote(), timeout=60)
+    assert tid == nid, "task must spill back to the labeled node"
