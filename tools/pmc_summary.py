@@ -6,10 +6,10 @@ Reads every *counter_collection.csv under the directory, sums counter
 values per (kernel, counter), and prints one row per kernel with the
 counters as columns, sorted by sort_counter (default GRBM_GUI_ACTIVE).
 Derived columns when the inputs are present:
-  * mfma_pct  = SQ_VALU_MFMA_BUSY_CYCLES / (SQ_BUSY_CYCLES*4) — matrix-core
-    occupancy of the busy SIMDs (4 SIMDs per CU accumulate busy cycles)
-  * hbm_gb    = (FETCH_SIZE + WRITE_SIZE) in GB (TCC counters report 32B/64B
-    granules already scaled by rocprofv3)
+  * mfma_ratio = SQ_VALU_MFMA_BUSY_CYCLES / SQ_BUSY_CYCLES — relative
+    matrix-core pressure (the MFMA counter aggregates per-pipe passes, so
+    this is a RELATIVE ranking across kernels, not an absolute percent)
+  * hbm_gb     = (FETCH_SIZE + WRITE_SIZE) in GB
 """
 import glob
 import os
@@ -37,8 +37,11 @@ def main():
     if ndisp is not None:
         pivot["dispatches"] = ndisp
     if {"SQ_VALU_MFMA_BUSY_CYCLES", "SQ_BUSY_CYCLES"} <= set(pivot.columns):
-        pivot["mfma_pct"] = (100 * pivot["SQ_VALU_MFMA_BUSY_CYCLES"]
-                             / pivot["SQ_BUSY_CYCLES"]).round(1)
+        pivot["mfma_ratio"] = (pivot["SQ_VALU_MFMA_BUSY_CYCLES"]
+                               / pivot["SQ_BUSY_CYCLES"]).round(2)
+    if {"SQ_LDS_BANK_CONFLICT", "SQ_BUSY_CYCLES"} <= set(pivot.columns):
+        pivot["lds_conf_frac"] = (pivot["SQ_LDS_BANK_CONFLICT"]
+                                  / pivot["SQ_BUSY_CYCLES"]).round(3)
     if {"FETCH_SIZE", "WRITE_SIZE"} <= set(pivot.columns):
         pivot["hbm_gb"] = ((pivot["FETCH_SIZE"] + pivot["WRITE_SIZE"])
                            / 1024).round(2)
