@@ -109,13 +109,19 @@ def test_compiled_dag_gpu_channels(ray_gpu):
         def scale(self, x, k):
             return x * k
 
+        def ready(self):
+            return True
+
     s = GpuStage.remote()
+    # gate on actor liveness: earlier tests' actors release their GPU
+    # fractions via handle-drop GC; a clean timeout here beats a wedge
+    assert ray.get(s.ready.remote(), timeout=180)
     with InputNode() as inp:
         dag = s.scale.bind(s.rms.bind(inp), 2.0).experimental_compile()
     assert dag._channel_mode, "single-node GPU actor chain must use channels"
     x = np.random.rand(64, 256).astype(np.float32)
     for _ in range(3):
-        y = ray.get(dag.execute(x))
+        y = ray.get(dag.execute(x), timeout=120)
     ref = x / np.sqrt((x ** 2).mean(-1, keepdims=True) + 1e-5) * 2.0
     assert np.abs(y - ref).mean() < 2e-2
     dag.teardown()
