@@ -142,6 +142,7 @@ class Raylet:
         )
         asyncio.get_running_loop().create_task(self._heartbeat_loop())
         asyncio.get_running_loop().create_task(self._reap_loop())
+        asyncio.get_running_loop().create_task(self._memory_monitor_loop())
         logger.info(
             "raylet %s on %s:%s (%s)", self.node_id.hex()[:8], self.node_ip, self.port,
             {k: v for k, v in self.resources_total.items()},
@@ -172,6 +173,84 @@ class Raylet:
             if len(self.idle_workers) > max_idle:
                 for w in [w for w in self.idle_workers if now - w.idle_since > IDLE_WORKER_KILL_S][: len(self.idle_workers) - max_idle]:
                     self._kill_worker(w)
+
+    # ------------------------------------------------------- memory monitor
+    # Parity: src/ray/common/memory_monitor.h (RAY_memory_usage_threshold
+    # 0.95, refresh-interval loop) + raylet worker_killing_policy.h: when
+    # node memory crosses the threshold, kill the largest leased worker,
+    # preferring retriable normal-task workers over actors.
+    @staticmethod
+    def _node_memory_fraction() -> float:
+        total = avail = None
+        with open("/proc/meminfo") as f:
+            for line in f:
+                if line.startswith("MemTotal:"):
+                    total = float(line.split()[1])
+                elif line.startswith("MemAvailable:"):
+                    avail = float(line.split()[1])
+                if total is not None and avail is not None:
+                    break
+        if not total:
+            return 0.0
+        return 1.0 - (avail or 0.0) / total
+
+    @staticmethod
+    def _rss_bytes(pid: int) -> int:
+        try:
+            with open(f"/proc/{pid}/statm") as f:
+                return int(f.read().split()[1]) * os.sysconf("SC_PAGE_SIZE")
+        except Exception:
+            return 0
+
+    def _pick_oom_victim(self) -> Optional[WorkerProc]:
+        leased = [w for w in self.workers.values()
+                  if w.leased and w.proc.poll() is None]
+        if not leased:
+            return None
+        # retriable task workers first (group), then actors; largest RSS
+        tasks = [w for w in leased if not w.is_actor]
+        pool = tasks or [w for w in leased if not w.detached_actor] or leased
+        return max(pool, key=lambda w: self._rss_bytes(w.pid))
+
+    async def _memory_monitor_loop(self):
+        threshold = float(os.environ.get("RAY_memory_usage_threshold", "0.95"))
+        period = float(os.environ.get(
+            "RAY_memory_monitor_refresh_ms", "1000")) / 1000.0
+        if period <= 0:
+            return  # monitor disabled (parity: refresh_ms = 0)
+        while not self._shutdown.is_set():
+            await asyncio.sleep(period)
+            try:
+                frac = self._node_memory_fraction()
+            except Exception:
+                continue
+            if frac <= threshold:
+                continue
+            victim = self._pick_oom_victim()
+            if victim is None:
+                continue
+            logger.warning(
+                "memory monitor: node memory %.1f%% > %.1f%% — killing "
+                "worker %s (rss %.1f MB)", frac * 100, threshold * 100,
+                victim.worker_id.hex()[:8],
+                self._rss_bytes(victim.pid) / 1e6)
+            victim.oom_killed = True
+            if victim.lease_owner is not None and not victim.lease_owner.closed:
+                try:
+                    await asyncio.wait_for(victim.lease_owner.call(
+                        "worker_killed_notice",
+                        {"worker_id": victim.worker_id,
+                         "reason": f"worker killed by the node memory "
+                                   f"monitor: node memory usage {frac:.2f} > "
+                                   f"threshold {threshold:.2f} "
+                                   f"(OutOfMemoryError)"}, timeout=2), 3)
+                except Exception:
+                    pass
+            self._kill_worker(victim)
+            await self._on_worker_dead(
+                victim,
+                f"killed by the memory monitor: node memory usage {frac:.2f} "
+                f"exceeds threshold {threshold:.2f} (OutOfMemoryError)")
 
     async def _on_worker_dead(self, w: WorkerProc, reason: str):
         self.workers.pop(w.worker_id, None)

@@ -125,6 +125,7 @@ class CoreWorker:
         self._owned: Dict[bytes, dict] = {}
         self._spilled: Dict[bytes, str] = {}  # oid -> spill file path
         self._pull_cache: Dict[bytes, tuple] = {}  # big-object pull slices
+        self._killed_workers: Dict[bytes, str] = {}  # worker_id -> kill reason
         self._object_locations: Dict[bytes, Tuple[str, int]] = {}
         self._actors: Dict[bytes, ActorHandleState] = {}
         self._actor_results: Dict[bytes, dict] = {}
@@ -362,6 +363,14 @@ class CoreWorker:
         if n <= PULL_CHUNK_BYTES:
             return {"data": bytes(data), "meta": meta}
         return {"data": bytes(data[:PULL_CHUNK_BYTES]), "meta": meta, "size": n}
+
+    async def rpc_worker_killed_notice(self, conn, p):
+        """Raylet tells us WHY a worker we lease is about to die (e.g. the
+        memory monitor), so the push failure surfaces that cause."""
+        self._killed_workers[p["worker_id"]] = p.get("reason", "worker killed")
+        while len(self._killed_workers) > 64:
+            self._killed_workers.pop(next(iter(self._killed_workers)))
+        return {"ok": True}
 
     async def rpc_free_objects(self, conn, p):
         self.store.free(p["oids"])
@@ -807,6 +816,9 @@ class CoreWorker:
                 reply = await lw.conn.call("push_task", payload, timeout=None)
             self._handle_task_reply(payload, reply)
         except Exception as e:  # worker died mid-task
+            killed_reason = self._killed_workers.pop(lw.worker_id, None)
+            if killed_reason is not None:
+                e = RaySystemError(killed_reason)
             retries = payload.get("max_retries", 3)
             if retries > 0 and not isinstance(e, protocol.RpcError):
                 payload["max_retries"] = retries - 1

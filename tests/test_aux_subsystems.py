@@ -117,3 +117,68 @@ def test_insight_flow_graph():
     assert node["calls"] >= 1 and node["total_s"] > 0
     assert any(e["to"] == "traced_task" for e in g["edges"])
     ray.shutdown()
+
+
+class TestMemoryMonitor:
+    """Raylet memory monitor (parity: common/memory_monitor.h +
+    worker_killing_policy: kill the largest leased worker, tasks first)."""
+
+    def test_victim_policy_prefers_tasks(self):
+        from ant_ray_amd._private.raylet import Raylet
+
+        class FakeProc:
+            def poll(self):
+                return None
+
+        class W:
+            def __init__(self, pid, is_actor, detached=False):
+                self.pid = pid
+                self.is_actor = is_actor
+                self.detached_actor = detached
+                self.leased = True
+                self.proc = FakeProc()
+                self.worker_id = bytes(20)
+
+        r = object.__new__(Raylet)
+        me = __import__("os").getpid()
+        r.workers = {
+            b"a": W(me, is_actor=True),
+            b"t": W(me, is_actor=False),
+        }
+        v = Raylet._pick_oom_victim(r)
+        assert v is not None and not v.is_actor  # task worker chosen first
+        r.workers = {b"a": W(me, is_actor=True)}
+        assert Raylet._pick_oom_victim(r).is_actor
+
+    def test_oom_kill_fails_running_task(self):
+        """threshold=0 forces the monitor to kill the leased task worker;
+        the task surfaces a memory-monitor error after retries."""
+        import os
+        import subprocess
+        import sys
+        import textwrap
+
+        script = textwrap.dedent("""
+            import ant_ray_amd as ray
+
+            ray.init(num_cpus=2)
+
+            @ray.remote(max_retries=0)
+            def hog():
+                import time
+                time.sleep(300)
+
+            try:
+                ray.get(hog.remote(), timeout=120)
+                print("NO_ERROR")
+            except Exception as e:
+                msg = str(e)
+                print("GOT_ERROR", type(e).__name__, msg[:200])
+        """)
+        env = dict(os.environ,
+                   RAY_memory_usage_threshold="0.0",
+                   RAY_memory_monitor_refresh_ms="200")
+        out = subprocess.run([sys.executable, "-c", script], env=env,
+                             capture_output=True, text=True, timeout=180)
+        assert "GOT_ERROR" in out.stdout, out.stdout[-800:] + out.stderr[-800:]
+        assert "memory" in out.stdout.lower(), out.stdout[-800:]
