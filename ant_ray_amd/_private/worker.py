@@ -187,6 +187,10 @@ class CoreWorker:
         self.raylet = await protocol.connect(self.raylet_addr, self._handle_rpc, name="->raylet")
         if self.mode == WORKER_MODE:
             self.raylet.on_close = lambda c: os._exit(0)  # fate-share with raylet
+            # leasable the moment registration lands: mark connected BEFORE
+            # any further await, or a fast lease+push can execute user code
+            # that still sees connected=False (get_runtime_context raises)
+            self.connected = True
             await self.raylet.call(
                 "register_worker", {"worker_id": self.worker_id, "addr": list(self.addr), "pid": os.getpid()}
             )
