@@ -70,6 +70,22 @@ def start(head, address, num_cpus, num_gpus, port, object_store_memory, block):
 
 
 @cli.command()
+@click.option("--quick", is_flag=True, help="1s per metric instead of 2s")
+def microbenchmark(quick):
+    """Run the core-runtime microbenchmark (parity: `ray microbenchmark`,
+    reference ray_perf.py metric definitions)."""
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    cmd = [sys.executable, os.path.join(root, "tools", "microbench.py")]
+    if quick:
+        cmd.append("--quick")
+    raise SystemExit(subprocess.call(cmd))
+
+
+@cli.command()
 def stop():
     """Stop the most recent local cluster (GCS shutdown broadcast)."""
     import ant_ray_amd as ray
