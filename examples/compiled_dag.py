@@ -9,7 +9,12 @@ chain on CPU; the win compounds with depth).
 Run: python examples/compiled_dag.py  (works on CPU; uses the HIP rmsnorm
 kernel when a GPU is visible)
 """
+import os
+import sys
+
 import numpy as np
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import ant_ray_amd as ray
 from ant_ray_amd.dag import InputNode
