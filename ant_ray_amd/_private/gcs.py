@@ -181,6 +181,8 @@ class GcsServer:
         self.nodes[node.node_id] = node
         conn.session["node_id"] = node.node_id
         conn.on_close = self._make_node_close_cb(node)
+        self._record_event("NODE", "REGISTERED", node.node_id,
+                           f"{node.addr[0]}:{node.addr[1]}")
         await self._publish("NODE", node.node_id, node.view())
         return {"ok": True}
 
@@ -192,6 +194,7 @@ class GcsServer:
 
     async def _on_node_dead(self, node: NodeInfo):
         node.alive = False
+        self._record_event("NODE", "DEAD", node.node_id)
         await self._publish("NODE", node.node_id, node.view())
         # virtual-cluster replenishment (gcs_virtual_cluster_manager.cc:730):
         # a dead member of a count-based cluster is replaced from the
@@ -381,6 +384,30 @@ class GcsServer:
                 subs.append(conn)
         return {"ok": True}
 
+    def _record_event(self, source: str, event: str, entity_id, message: str = ""):
+        """Structured cluster lifecycle events (parity: observability/
+        ray_event_recorder.cc definition+lifecycle events; surfaced by
+        `ray list cluster-events` and state.list_cluster_events)."""
+        import collections
+
+        if not hasattr(self, "cluster_events"):
+            self.cluster_events = collections.deque(maxlen=10000)
+        self.cluster_events.append({
+            "timestamp": time.time(),
+            "source": source,          # NODE | ACTOR | JOB | WORKER
+            "event": event,            # REGISTERED | ALIVE | DEAD | ...
+            "entity_id": entity_id.hex() if isinstance(entity_id, bytes)
+            else str(entity_id),
+            "message": message,
+        })
+
+    async def rpc_list_cluster_events(self, conn, p):
+        limit = p.get("limit", 1000)
+        evs = list(getattr(self, "cluster_events", []))
+        if p.get("source"):
+            evs = [e for e in evs if e["source"] == p["source"]]
+        return evs[-limit:]
+
     async def _publish(self, channel: str, key: bytes, data: Any):
         subs = self._subscribers.get(channel, [])
         dead = []
@@ -516,6 +543,7 @@ class GcsServer:
                     return
             if time.monotonic() > deadline or self._shutdown.is_set():
                 info.state = DEAD
+                self._record_event("ACTOR", "DEAD", info.actor_id, info.death_cause)
                 info.death_cause = "scheduling timed out (insufficient resources)"
                 await self._publish("ACTOR", info.actor_id, info.view())
                 self._wake_waiters(info)
@@ -538,11 +566,13 @@ class GcsServer:
             return
         if reply.get("status") == "ok":
             info.state = ALIVE
+            self._record_event("ACTOR", "ALIVE", info.actor_id, info.name)
             info.addr = addr
             await self._publish("ACTOR", info.actor_id, info.view())
             self._wake_waiters(info)
         else:
             info.state = DEAD
+            self._record_event("ACTOR", "DEAD", info.actor_id, info.death_cause)
             info.death_cause = reply.get("error", "actor __init__ failed")
             info.creation_error = reply.get("error_payload")
             await self._publish("ACTOR", info.actor_id, info.view())
@@ -568,12 +598,14 @@ class GcsServer:
         if max_restarts == -1 or info.num_restarts < max_restarts:
             info.num_restarts += 1
             info.state = RESTARTING
+            self._record_event("ACTOR", "RESTARTING", info.actor_id)
             info.addr = None
             info.worker_id = None
             await self._publish("ACTOR", info.actor_id, info.view())
             asyncio.get_running_loop().create_task(self._schedule_actor(info, delay=0.1))
         else:
             info.state = DEAD
+            self._record_event("ACTOR", "DEAD", info.actor_id, info.death_cause)
             info.death_cause = reason
             info.addr = None
             await self._publish("ACTOR", info.actor_id, info.view())

@@ -128,6 +128,7 @@ def status():
 @cli.command("list")
 @click.argument("kind", type=click.Choice(
     ["actors", "tasks", "nodes", "workers", "jobs", "placement-groups",
+     "cluster-events",
      "objects"]))
 @click.option("--limit", type=int, default=100)
 def list_cmd(kind, limit):
@@ -138,6 +139,7 @@ def list_cmd(kind, limit):
         "actors": S.list_actors, "tasks": S.list_tasks, "nodes": S.list_nodes,
         "workers": S.list_workers, "jobs": S.list_jobs,
         "placement-groups": S.list_placement_groups, "objects": S.list_objects,
+        "cluster-events": S.list_cluster_events,
     }[kind]
     rows = fn(limit=limit)
     if not rows:

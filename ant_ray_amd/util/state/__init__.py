@@ -53,6 +53,13 @@ def list_nodes(filters=None, limit: int = 1000, **_) -> List[Dict[str, Any]]:
     return out[:limit]
 
 
+def list_cluster_events(source: str = None, limit: int = 1000, **_):
+    """Structured lifecycle events (NODE/ACTOR REGISTERED/ALIVE/DEAD/...),
+    parity: reference export events / RayEventRecorder."""
+    return _gcs_call("list_cluster_events",
+                     {"limit": limit, "source": source})
+
+
 def list_workers(filters=None, limit: int = 1000, **_) -> List[Dict[str, Any]]:
     rows = [_hexify(w) for w in _gcs_call("list_workers")]
     return [r for r in rows if _match(r, filters)][:limit]

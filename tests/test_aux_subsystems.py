@@ -182,3 +182,38 @@ class TestMemoryMonitor:
                              capture_output=True, text=True, timeout=180)
         assert "GOT_ERROR" in out.stdout, out.stdout[-800:] + out.stderr[-800:]
         assert "memory" in out.stdout.lower(), out.stdout[-800:]
+
+
+def test_cluster_lifecycle_events():
+    """GCS records structured NODE/ACTOR lifecycle events (observability
+    parity: RayEventRecorder definition+lifecycle events)."""
+    import ant_ray_amd as ray
+    from ant_ray_amd.util.state import list_cluster_events
+
+    if ray.is_initialized():
+        ray.shutdown()
+    ray.init(num_cpus=2)
+
+    @ray.remote
+    class A:
+        def ping(self):
+            return 1
+
+    a = A.remote()
+    assert ray.get(a.ping.remote(), timeout=60) == 1
+    ray.kill(a)
+    import time as _t
+
+    deadline = _t.time() + 30
+    while _t.time() < deadline:
+        evs = list_cluster_events()
+        kinds = {(e["source"], e["event"]) for e in evs}
+        if {("NODE", "REGISTERED"), ("ACTOR", "ALIVE"),
+                ("ACTOR", "DEAD")} <= kinds:
+            break
+        _t.sleep(0.3)
+    assert ("NODE", "REGISTERED") in kinds
+    assert ("ACTOR", "ALIVE") in kinds
+    assert ("ACTOR", "DEAD") in kinds
+    assert all("timestamp" in e and "entity_id" in e for e in evs)
+    ray.shutdown()
