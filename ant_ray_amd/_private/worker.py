@@ -602,8 +602,25 @@ class CoreWorker:
                 size = reply.get("size")
                 if size is not None and size > len(data):
                     # big object: pipeline the remaining 5 MiB chunk
-                    # requests on the same multiplexed connection
-                    buf = bytearray(size)
+                    # requests on the same multiplexed connection. When a
+                    # local shm store exists, chunks land DIRECTLY in a
+                    # freshly created shm buffer (no intermediate heap
+                    # copy) and the sealed object is shared with every
+                    # other worker on this node (parity: reference
+                    # ObjectBufferPool writes Push chunks straight into
+                    # the plasma create buffer, object_buffer_pool.cc).
+                    buf, in_shm = None, False
+                    if self.store.shm is not None:
+                        try:
+                            soff = self.store.shm.create_object(
+                                oid, size, bytes(meta))
+                            buf = self.store.shm.view_at(
+                                soff, size, writable=True)
+                            in_shm = True
+                        except Exception:
+                            buf = None  # exists already / OOM -> heap path
+                    if buf is None:
+                        buf = memoryview(bytearray(size))
                     buf[: len(data)] = data
                     futs = [
                         (off, self.io.submit(conn.call(
@@ -617,6 +634,25 @@ class CoreWorker:
                             chunk = f.result(70)["data"]
                             buf[off: off + len(chunk)] = chunk
                     except Exception:
+                        if in_shm:
+                            del buf
+                            try:
+                                self.store.shm.abort(oid)
+                                self.store.shm.release(oid)
+                            except Exception:
+                                pass
+                        return None, False
+                    if in_shm:
+                        del buf
+                        self.store.shm.seal(oid)
+                        self.store.shm.release(oid)
+                        shm_buf, shm_meta = self.store.shm.get_buffer(oid, 0.0)
+                        if shm_buf is not None:
+                            self.store.memory.put(oid, IN_PLASMA)
+                            return self._deserialize_buffer(
+                                shm_buf, bytes(shm_meta)), True
+                        # sealed copy evicted between seal and read (tiny
+                        # arena under pressure): refetch via the slow path
                         return None, False
                     data = buf
                 value = serialization.deserialize(memoryview(data), meta)

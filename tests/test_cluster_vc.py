@@ -141,6 +141,21 @@ def test_cross_node_chunked_pull(cluster):
     assert n == arr.shape[0]
     assert total == pytest.approx(float(arr.sum()))
 
+    # the pulled copy must land in the consumer node's shm store (the
+    # shm-destination write path), so a second task on that node reads
+    # it locally without a cross-node pull
+    @ray.remote(num_cpus=1, resources={"far": 0.1})
+    def consume_nested(refs):
+        x = ray.get(refs[0])
+        from ant_ray_amd._private.worker import global_worker
+
+        shm = global_worker.core_worker.store.shm
+        return float(x.sum()), shm.contains(refs[0].binary())
+
+    total2, in_shm = ray.get(consume_nested.remote([ref]), timeout=120)
+    assert total2 == pytest.approx(float(arr.sum()))
+    assert in_shm, "pulled big object should be cached in local shm"
+
 
 def test_wait_many_refs_drain(cluster):
     """ray.wait drains 300 refs one at a time (the ray_perf wait-1k shape)."""
