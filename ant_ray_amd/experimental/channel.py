@@ -51,6 +51,12 @@ class Channel:
             self._num_readers = num_readers
             self._store.channel_create(self._oid, capacity, num_readers)
         self._last_version = 0
+        # hipIpc producer-side pin window: tensors exported for version N
+        # stay pinned until write(N+2) is admitted. Admission of write(N)
+        # means every (single-threaded) reader returned from read(N-1),
+        # which strictly follows its fenced clone of version N-2 — so N-2's
+        # device memory is provably unreferenced (see read()).
+        self._gpu_pins: list = []
 
     def __reduce__(self):
         return (_attach, (self._path, self._oid, self._num_readers))
@@ -59,7 +65,18 @@ class Channel:
     def write(self, value: Any, timeout: Optional[float] = None):
         from ant_ray_amd._private import serialization
 
-        sobj = serialization.serialize(value)
+        pins = None
+        if _may_hold_gpu(value):
+            # GPU payloads ride the hipIpc tier: only the 64-byte handle
+            # descriptor crosses the shm channel; the reader maps the
+            # producer's HBM over xGMI and clones device-to-device — zero
+            # host copies (vs the reference's CPU serialization through
+            # shared_memory_channel.py for non-NCCL-annotated tensors).
+            with serialization.gpu_transport_context("hip_ipc") as gctx:
+                sobj = serialization.serialize(value)
+            pins = gctx.pinned
+        else:
+            sobj = serialization.serialize(value)
         meta = sobj.metadata
         payload = struct.pack("<I", len(meta)) + meta + sobj.to_bytes()
         try:
@@ -67,6 +84,9 @@ class Channel:
                 self._oid, payload, -1.0 if timeout is None else timeout)
         except RuntimeError as e:
             raise _map_err(e) from None
+        if pins is not None or self._gpu_pins:
+            self._gpu_pins.append(pins or [])
+            del self._gpu_pins[:-2]  # keep versions N and N-1 pinned
 
     def read(self, timeout: Optional[float] = None, unwrap: bool = True) -> Any:
         """unwrap=False returns _WrappedError values instead of raising —
@@ -81,12 +101,26 @@ class Channel:
             raise _map_err(e) from None
         (mlen,) = struct.unpack_from("<I", data)
         meta = data[4:4 + mlen]
-        value = serialization.deserialize(memoryview(data)[4 + mlen:], meta)
+        from ant_ray_amd.experimental.gpu_object_manager.gpu_object_store import (
+            gpu_import_clone,
+        )
+
+        with gpu_import_clone() as imp:
+            value = serialization.deserialize(
+                memoryview(data)[4 + mlen:], meta)
+        if imp.cloned:
+            # fence the async device clones BEFORE the next channel_read
+            # acks a newer version (the producer's pin-window proof relies
+            # on clone-complete-before-next-ack)
+            import torch
+
+            torch.cuda.current_stream().synchronize()
         if unwrap and isinstance(value, _WrappedError):
             raise value.exc
         return value
 
     def close(self):
+        self._gpu_pins.clear()
         try:
             self._store.channel_close(self._oid)
         except KeyError:
@@ -100,6 +134,31 @@ class Channel:
             self._store.delete(self._oid)
         except Exception:
             pass
+
+
+def _may_hold_gpu(value: Any) -> bool:
+    """Cheap check whether a channel payload can contain CUDA tensors
+    (walks one container level; deeper nesting still works — the
+    serializer just takes the default CPU-copy path for those)."""
+    import sys
+
+    if "torch" not in sys.modules:
+        return False
+    import torch
+
+    if not torch.cuda.is_available():
+        return False
+
+    def is_gpu(v):
+        return isinstance(v, torch.Tensor) and v.is_cuda
+
+    if is_gpu(value):
+        return True
+    if isinstance(value, (list, tuple, set)):
+        return any(is_gpu(v) for v in value)
+    if isinstance(value, dict):
+        return any(is_gpu(v) for v in value.values())
+    return False
 
 
 def _attach(path: str, oid: bytes, num_readers: int) -> Channel:

@@ -98,5 +98,33 @@ def import_tensor(meta: dict) -> torch.Tensor:
     )
 
 
+_import_mode = threading.local()
+
+
+class gpu_import_clone:
+    """While active, hip_ipc descriptors deserialized in this thread
+    materialize as LOCAL device clones (and the peer mapping is released
+    immediately) instead of zero-copy views. Compiled-DAG channel reads
+    use this: the producer's buffer is mutable (rotating pin window), so
+    the reader must own its copy. Sets `.cloned` when any import happened
+    so the caller can fence the async device copies."""
+
+    def __enter__(self):
+        _import_mode.clone = True
+        _import_mode.cloned = False
+        return self
+
+    def __exit__(self, *a):
+        _import_mode.clone = False
+        self.cloned = getattr(_import_mode, "cloned", False)
+        _import_mode.cloned = False
+
+
 def _rebuild_gpu_tensor(meta: dict):
-    return import_tensor(meta)
+    view = import_tensor(meta)
+    if getattr(_import_mode, "clone", False):
+        out = view.clone()
+        del view  # drop the peer mapping refcount now
+        _import_mode.cloned = True
+        return out
+    return view

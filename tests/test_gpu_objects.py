@@ -131,3 +131,57 @@ def test_compiled_dag_gpu_channels():
     out = subprocess.run([sys.executable, "-u", "-c", script],
                          capture_output=True, text=True, timeout=300)
     assert "DAG_GPU_OK" in out.stdout, out.stdout[-1500:] + out.stderr[-1500:]
+
+
+def test_compiled_dag_channel_hipipc_tensors():
+    """GPU tensors crossing a compiled-DAG channel ride the hipIpc tier:
+    only the 64-byte handle descriptor goes through shm, the consumer
+    stage clones the producer's HBM device-to-device (experimental/
+    channel.py write/_may_hold_gpu + gpu_import_clone). The produce stage
+    keeps tensors on device end-to-end; the sink converts to numpy."""
+    import subprocess
+    import sys
+    import textwrap
+
+    script = textwrap.dedent("""
+        import numpy as np
+        import torch
+        import ant_ray_amd as ray
+        from ant_ray_amd.dag import InputNode
+
+        ray.init(num_cpus=4, num_gpus=1)
+
+        @ray.remote(num_gpus=0.25)
+        class Produce:
+            def up(self, x):
+                import torch
+                # returns a DEVICE tensor -> channel must take the hipIpc path
+                return torch.from_numpy(x).to("cuda") * 2.0
+
+        @ray.remote(num_gpus=0.25)
+        class Sink:
+            def __init__(self):
+                self.saw_cuda = False
+
+            def down(self, t):
+                import torch
+                assert isinstance(t, torch.Tensor) and t.is_cuda, (
+                    "expected the device tensor to arrive on device")
+                self.saw_cuda = True
+                return (t + 1.0).cpu().numpy()
+
+        p = Produce.remote()
+        s = Sink.remote()
+        with InputNode() as inp:
+            dag = s.down.bind(p.up.bind(inp)).experimental_compile()
+        assert dag._channel_mode
+        for i in range(4):
+            x = np.full((128, 64), float(i), dtype=np.float32)
+            y = ray.get(dag.execute(x), timeout=120)
+            assert np.allclose(y, x * 2.0 + 1.0), (i, y[0, 0])
+        dag.teardown()
+        print("DAG_HIPIPC_OK")
+    """)
+    out = subprocess.run([sys.executable, "-u", "-c", script],
+                         capture_output=True, text=True, timeout=300)
+    assert "DAG_HIPIPC_OK" in out.stdout, out.stdout[-1500:] + out.stderr[-1500:]
