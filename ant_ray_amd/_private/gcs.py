@@ -80,6 +80,7 @@ class ActorInfo:
         self.node_id: Optional[bytes] = None
         self.num_restarts = 0
         self.death_cause = ""
+        self.killed = False  # ray.kill / out-of-scope while still scheduling
         self.pending_waiters: List[asyncio.Future] = []
 
     def view(self):
@@ -511,6 +512,13 @@ class GcsServer:
         deadline = time.monotonic() + float(opts.get("_scheduling_timeout", 3600.0))
         vc_id = opts.get("virtual_cluster_id") or self.worker_vc.get(info.owner)
         while True:
+            if info.killed:
+                info.state = DEAD
+                info.death_cause = "actor killed before creation completed"
+                self._record_event("ACTOR", "DEAD", info.actor_id, info.death_cause)
+                await self._publish("ACTOR", info.actor_id, info.view())
+                self._wake_waiters(info)
+                return
             node = self._pick_node(resources, pg, opts.get("_node_affinity"),
                                    self._vc_allowed(vc_id),
                                    opts.get("_label_selector"))
@@ -552,6 +560,20 @@ class GcsServer:
 
     async def _push_actor_creation(self, info: ActorInfo, node: NodeInfo, lease):
         addr = tuple(lease["addr"])
+        if info.killed:
+            # killed between lease grant and creation push: hand the
+            # leased worker back (its death returns the resources)
+            try:
+                wconn = await self._get_worker_conn(addr)
+                await wconn.notify("exit_worker", {"reason": "ray.kill"})
+            except Exception:
+                pass
+            info.state = DEAD
+            info.death_cause = "actor killed before creation completed"
+            self._record_event("ACTOR", "DEAD", info.actor_id, info.death_cause)
+            await self._publish("ACTOR", info.actor_id, info.view())
+            self._wake_waiters(info)
+            return
         info.worker_id = lease["worker_id"]
         info.node_id = node.node_id
         try:
@@ -570,6 +592,12 @@ class GcsServer:
             info.addr = addr
             await self._publish("ACTOR", info.actor_id, info.view())
             self._wake_waiters(info)
+            if info.killed:  # kill raced the creation push: finish it now
+                try:
+                    wconn = await self._get_worker_conn(addr)
+                    await wconn.notify("exit_worker", {"reason": "ray.kill"})
+                except Exception:
+                    pass
         else:
             info.state = DEAD
             self._record_event("ACTOR", "DEAD", info.actor_id, info.death_cause)
@@ -645,6 +673,12 @@ class GcsServer:
         no_restart = p.get("no_restart", True)
         if no_restart:
             info.opts["max_restarts"] = 0
+            # an actor killed while STILL SCHEDULING must not become a
+            # zombie: the _schedule_actor loop checks this flag before
+            # leasing and before pushing the creation task (otherwise a
+            # handle GC'd during creation leaves an unkillable actor
+            # holding its GPU/CPU fraction and starving later leases)
+            info.killed = True
         if info.addr:
             try:
                 wconn = await self._get_worker_conn(tuple(info.addr))

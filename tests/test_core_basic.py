@@ -173,7 +173,9 @@ class TestDistributed:
             def who(self):
                 return "reg"
 
-        Reg.options(name="registry").remote()
+        keep = Reg.options(name="registry").remote()  # keep a handle: a
+        # non-detached actor whose handles all drop is terminated (the
+        # out-of-scope kill now also covers still-scheduling actors)
         h = ray.get_actor("registry")
         assert ray.get(h.who.remote()) == "reg"
 
@@ -301,3 +303,35 @@ class TestObjectStorePressure:
             assert not global_worker.core_worker._spilled
         finally:
             ray.shutdown()
+
+
+def test_kill_actor_while_scheduling(ray_start_regular):
+    """ray.kill (or the last handle dropping) on an actor that is still
+    WAITING for resources must terminate the scheduling loop — not leave
+    a zombie that later grabs the lease and holds its resources forever
+    (the shared-GPU-session wedge: GC'd fractional-GPU actors kept
+    starving the next test's lease)."""
+    import time
+
+    import ant_ray_amd as ray
+    from ant_ray_amd.util import state
+
+    @ray.remote(resources={"no_such_resource": 1})
+    class Stuck:
+        def ping(self):
+            return 1
+
+    a = Stuck.remote()
+    aid = a._ray_actor_id
+    time.sleep(0.3)  # let it enter the scheduling loop
+    ray.kill(a)
+
+    deadline = time.time() + 15
+    dead = False
+    while time.time() < deadline:
+        rows = [r for r in state.list_actors() if r["actor_id"] == aid.hex()]
+        if rows and rows[0]["state"] == "DEAD":
+            dead = True
+            break
+        time.sleep(0.2)
+    assert dead, "killed-while-scheduling actor should transition to DEAD"
