@@ -432,26 +432,37 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
   const ushort_t* dOp = dO + (long)b * ob + (long)hq * oh;
   const float* Lp = LSE + ((long)b * Hq + hq) * S;
   const float* Dp = Delta + ((long)b * Hq + hq) * S;
+  // double-buffered q/do tile prefetch (same pattern as attn_bwd_dv):
+  // next tile's global loads issue OUTSIDE the LDS critical section so
+  // the barrier never waits on HBM latency
+  ushortx8 qa, qa2, da, da2;
+  {
+    const int qg = min(q_start_tile * BLOCK + st_row, S - 1);
+    qa = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
+    qa2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
+    da = *(const ushortx8*)(dOp + (long)qg * os + st_col);
+    da2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
+  }
   for (int t = q_start_tile; t < n_q_tiles; ++t) {
     const int tq0 = t * BLOCK;
     __syncthreads();
-    {
-      const int qg = min(tq0 + st_row, S - 1);
-      const ushortx8 a = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
-      const ushortx8 a2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
-      const ushortx8 d = *(const ushortx8*)(dOp + (long)qg * os + st_col);
-      const ushortx8 d2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
-      *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8)]) = a;
-      *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8 + 1)]) = a2;
-      *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8)]) = d;
-      *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8 + 1)]) = d2;
+    *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8)]) = qa;
+    *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8 + 1)]) = qa2;
+    *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8)]) = da;
+    *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8 + 1)]) = da2;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        q_tr[tr_off(st_col + j, st_row)] = a[j];
-        q_tr[tr_off(st_col + 8 + j, st_row)] = a2[j];
-      }
+    for (int j = 0; j < 8; ++j) {
+      q_tr[tr_off(st_col + j, st_row)] = qa[j];
+      q_tr[tr_off(st_col + 8 + j, st_row)] = qa2[j];
     }
     __syncthreads();
+    if (t + 1 < n_q_tiles) {
+      const int qg = min((t + 1) * BLOCK + st_row, S - 1);
+      qa = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
+      qa2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
+      da = *(const ushortx8*)(dOp + (long)qg * os + st_col);
+      da2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
+    }
 
     const int q_here = tq0 + (lane & 31);
     const bool qv = q_here < S;
