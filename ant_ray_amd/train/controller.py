@@ -27,6 +27,8 @@ from ant_ray_amd.train.worker_group import WorkerGroup
 
 logger = logging.getLogger("antray.train")
 
+_RESIZE = object()  # poll sentinel: regrow the group (not a failure)
+
 
 class CheckpointManager:
     """Tracks reported checkpoints; enforces num_to_keep retention."""
@@ -115,6 +117,18 @@ class TrainController:
                 shards[i][name] = splits[i] if splits else ds
         return shards
 
+    def _fit_in_available(self) -> int:
+        """How many additional workers fit in currently-free resources."""
+        try:
+            import ant_ray_amd as ray
+
+            per = self.scaling._resources_per_worker_not_none
+            avail = ray.available_resources()
+            return min(int(avail.get(k, 0.0) / v)
+                       for k, v in per.items() if v > 0)
+        except Exception:
+            return 0
+
     def _elastic_size(self) -> int:
         """Pick the attempt's world size: num_workers when resources allow,
         else the largest feasible size >= min_workers (elastic training —
@@ -161,7 +175,7 @@ class TrainController:
                     self.train_fn, self.train_loop_config, restore,
                     self._dataset_shards,
                 )
-                failed = self._poll_until_done(group)
+                failed = self._poll_until_done(group, n)
             except Exception as e:  # actor/scheduling level failure
                 logger.exception("worker group failed")
                 failed = e
@@ -170,6 +184,15 @@ class TrainController:
             if failed is None:
                 error = None
                 break
+            if failed is _RESIZE:
+                # scaling decision, not a failure (parity: reference
+                # controller applies ScalingPolicy resize decisions by
+                # restarting the group from the latest checkpoint,
+                # train/v2/.../controller.py:180-191) — don't burn an
+                # attempt
+                logger.warning("elastic: resources recovered, regrowing "
+                               "worker group from checkpoint")
+                continue
             error = failed if isinstance(failed, BaseException) else RuntimeError(failed)
             attempt += 1
             if max_failures >= 0 and attempt > max_failures:
@@ -186,8 +209,10 @@ class TrainController:
             best_checkpoints=self.ckpt_manager.best_checkpoints(),
         )
 
-    def _poll_until_done(self, group: WorkerGroup):
-        """Returns None on clean finish, error string on worker failure."""
+    def _poll_until_done(self, group: WorkerGroup, current_n: int):
+        """Returns None on clean finish, error string on worker failure, or
+        _RESIZE when a shrunk group can regrow (upscale decision)."""
+        last_scale_check = time.monotonic()
         while True:
             statuses = group.poll()
             for st in statuses:
@@ -201,4 +226,18 @@ class TrainController:
                 return errs[0]
             if all(st["status"] == "finished" for st in statuses):
                 return None
+            # upscale decision: a group running below num_workers regrows
+            # once resources free up AND a checkpoint exists to resume from
+            # (without one, restarting would lose more progress than the
+            # extra workers regain)
+            if (current_n < self.scaling.num_workers
+                    and self.ckpt_manager.latest
+                    and time.monotonic() - last_scale_check > 2.0):
+                last_scale_check = time.monotonic()
+                # the running group holds current_n workers' resources;
+                # after teardown those return, so the regrown size is
+                # current_n + whatever fits in what's free NOW
+                if min(self.scaling.num_workers,
+                       current_n + self._fit_in_available()) > current_n:
+                    return _RESIZE
             time.sleep(0.2)

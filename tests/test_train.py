@@ -226,3 +226,80 @@ def test_elastic_shrinks_to_available(ray_mod, tmp_path_factory):
     assert result.error is None
     # shrunk below 16, at least min_workers, bounded by the 6-CPU session
     assert 1 <= result.metrics["world"] <= 6
+
+
+def test_elastic_regrows_when_resources_return(ray_mod, tmp_path_factory):
+    """Upscale decision: a group shrunk by resource pressure regrows (via
+    checkpoint restart, parity with the reference controller's scaling
+    decisions) once the hogged resources free up mid-run."""
+    import time as _time
+
+    from ant_ray_amd.train import Checkpoint, RunConfig, ScalingConfig
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    ray = ray_mod
+    storage = str(tmp_path_factory.mktemp("storage"))
+
+    # hog 4 of the 6 session CPUs; released only once the SHRUNK first
+    # attempt has checkpointed (marker file), so the sequence is
+    # deterministic: shrink -> checkpoint -> resources return -> regrow
+    @ray.remote(num_cpus=4)
+    class Hog:
+        def ping(self):
+            return True
+
+    hog = Hog.remote()
+    ray.get(hog.ping.remote(), timeout=30)
+    marker = os.path.join(storage, "first_attempt_reported")
+
+    import threading
+
+    def release_when_marked():
+        deadline = _time.time() + 60
+        while _time.time() < deadline and not os.path.exists(marker):
+            _time.sleep(0.2)
+        ray.kill(hog)
+
+    watcher = threading.Thread(target=release_when_marked, daemon=True)
+    watcher.start()
+
+    def train_fn(config):
+        import os
+        import tempfile
+
+        from ant_ray_amd import train
+
+        ctx = train.get_context()
+        world = ctx.get_world_size()
+        if train.get_checkpoint() is None:
+            # first (shrunk) attempt: persist a checkpoint so the
+            # controller may regrow, then linger until it does
+            with tempfile.TemporaryDirectory() as d:
+                with open(os.path.join(d, "state.json"), "w") as f:
+                    f.write("{}")
+                train.report({"world": world, "phase": "first"},
+                             checkpoint=Checkpoint.from_directory(d))
+            if ctx.world_rank == 0:
+                with open(os.path.join(os.path.dirname(ctx.experiment_path),
+                                       "first_attempt_reported"), "w") as f:
+                    f.write("1")
+            _t = 0.0
+            while _t < 60.0:
+                import time
+
+                time.sleep(0.5)
+                _t += 0.5
+        else:
+            train.report({"world": world, "phase": "resumed"})
+
+    trainer = TorchTrainer(
+        train_fn,
+        scaling_config=ScalingConfig(num_workers=4, min_workers=1),
+        run_config=RunConfig(name="elastic_up", storage_path=storage),
+    )
+    t0 = _time.time()
+    result = trainer.fit()
+    assert result.error is None, result.error
+    assert result.metrics["phase"] == "resumed"
+    assert result.metrics["world"] == 4, result.metrics
+    assert _time.time() - t0 < 90
