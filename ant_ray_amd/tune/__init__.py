@@ -24,6 +24,7 @@ from ant_ray_amd.tune.schedulers import (  # noqa: F401
     ASHAScheduler,
     FIFOScheduler,
     MedianStoppingRule,
+    PopulationBasedTraining,
 )
 from ant_ray_amd.tune.search import (  # noqa: F401
     BasicVariantGenerator,
@@ -126,12 +127,16 @@ class _TrialRunner:
         self._done = False
         self._threading = threading
 
-    def start(self, fn, config, trial_id: str, experiment_path: str):
+    def start(self, fn, config, trial_id: str, experiment_path: str,
+              restore_path: str = None):
+        from ant_ray_amd.train._checkpoint import Checkpoint
         from ant_ray_amd.train.session import TrainContext, set_train_context
 
         ctx = TrainContext(experiment_name=trial_id,
                            experiment_path=experiment_path)
         ctx.report_queue = self._q
+        if restore_path:
+            ctx.restore_checkpoint = Checkpoint(restore_path)
 
         def run():
             set_train_context(ctx)
@@ -192,16 +197,20 @@ class Tuner:
         running: Dict[int, dict] = {}
         results: Dict[int, Result] = {}
 
-        def launch(idx, config):
+        def launch(idx, config, restore=None, prev=None):
             trial_id = f"{name}_{idx:05d}"
             trial_path = os.path.join(exp_path, trial_id)
             os.makedirs(trial_path, exist_ok=True)
             actor = RunnerCls.options(**opts).remote()
             ray.get(actor.start.remote(self.trainable, config, trial_id,
-                                       trial_path))
+                                       trial_path, restore))
             running[idx] = {"actor": actor, "config": config,
                             "trial_id": trial_id, "path": trial_path,
-                            "iter": 0, "last_metrics": None, "ckpt": None}
+                            "iter": prev["iter"] if prev else 0,
+                            "last_metrics": prev["last_metrics"] if prev else None,
+                            "ckpt": prev["ckpt"] if prev else None}
+            if hasattr(scheduler, "on_trial_start"):
+                scheduler.on_trial_start(trial_id, config)
 
         while pending or running:
             while pending and len(running) < max_conc:
@@ -212,6 +221,7 @@ class Tuner:
             finished_idx = []
             for (idx, t), st in zip(list(running.items()), polls):
                 stop = False
+                perturb = False
                 for rep in st["reports"]:
                     t["iter"] += 1
                     metrics = dict(rep["metrics"])
@@ -220,8 +230,26 @@ class Tuner:
                     t["last_metrics"] = metrics
                     if rep.get("checkpoint_path"):
                         t["ckpt"] = rep["checkpoint_path"]
-                    if scheduler.on_trial_result(t["trial_id"], metrics) == STOP:
+                        if hasattr(scheduler, "on_checkpoint"):
+                            scheduler.on_checkpoint(t["trial_id"],
+                                                    rep["checkpoint_path"])
+                    decision = scheduler.on_trial_result(t["trial_id"], metrics)
+                    if decision == STOP:
                         stop = True
+                    elif decision == "PERTURB":
+                        perturb = True
+                if perturb and st["status"] == "running":
+                    # PBT exploit/explore: clone a top trial's checkpoint,
+                    # mutate hyperparams, relaunch this slot
+                    ex = scheduler.exploit(t["trial_id"])
+                    if ex is not None:
+                        restore_path, new_cfg = ex
+                        try:
+                            ray.kill(t["actor"])
+                        except Exception:
+                            pass
+                        launch(idx, new_cfg, restore=restore_path, prev=t)
+                        continue
                 if st["status"] in ("finished", "errored") or stop:
                     err = None
                     if st["status"] == "errored":

@@ -12,6 +12,7 @@ from typing import Dict, List, Optional
 
 CONTINUE = "CONTINUE"
 STOP = "STOP"
+PERTURB = "PERTURB"
 
 
 class FIFOScheduler:
@@ -105,3 +106,101 @@ class MedianStoppingRule:
         if self.mode == "max":
             return CONTINUE if best >= median else STOP
         return CONTINUE if best <= median else STOP
+
+
+class PopulationBasedTraining:
+    """PBT (reference tune/schedulers/pbt.py): at every
+    perturbation_interval, trials in the bottom quantile EXPLOIT a top-
+    quantile trial (clone its latest checkpoint + config) and EXPLORE by
+    mutating hyperparameters (resample from the mutation space with
+    probability resample_probability, else multiply by 1.2 / 0.8).
+
+    Protocol with the Tuner: on_trial_result may return PERTURB; the
+    Tuner then calls exploit(trial_id) -> (restore_path, new_config) and
+    relaunches the trial. The Tuner feeds checkpoints and configs in via
+    on_checkpoint()/on_trial_start().
+    """
+
+    def __init__(self, metric: Optional[str] = None, mode: Optional[str] = None,
+                 perturbation_interval: int = 4,
+                 hyperparam_mutations: Optional[Dict] = None,
+                 quantile_fraction: float = 0.25,
+                 resample_probability: float = 0.25,
+                 time_attr: str = "training_iteration", seed: Optional[int] = None):
+        import random
+
+        self.metric = metric
+        self.mode = mode
+        self.interval = perturbation_interval
+        self.mutations = hyperparam_mutations or {}
+        self.quantile = quantile_fraction
+        self.resample_p = resample_probability
+        self.time_attr = time_attr
+        self._rng = random.Random(seed)
+        self._score: Dict[str, float] = {}
+        self._ckpt: Dict[str, str] = {}
+        self._config: Dict[str, dict] = {}
+        self._last_perturb: Dict[str, int] = defaultdict(int)
+
+    def set_objective(self, metric, mode):
+        self.metric = self.metric or metric
+        self.mode = self.mode or mode
+
+    # -- Tuner feed-in hooks
+    def on_trial_start(self, trial_id: str, config: dict):
+        self._config[trial_id] = dict(config)
+
+    def on_checkpoint(self, trial_id: str, path: str):
+        self._ckpt[trial_id] = path
+
+    # -- decisions
+    def on_trial_result(self, trial_id: str, result: dict) -> str:
+        v = result.get(self.metric)
+        t = result.get(self.time_attr)
+        if v is None or t is None:
+            return CONTINUE
+        self._score[trial_id] = float(v)
+        if t - self._last_perturb[trial_id] < self.interval:
+            return CONTINUE
+        self._last_perturb[trial_id] = t
+        lower, upper = self._quantiles()
+        if trial_id in lower and any(u in self._ckpt for u in upper):
+            return PERTURB
+        return CONTINUE
+
+    def _quantiles(self):
+        trials = [tid for tid in self._score]
+        if len(trials) < 2:
+            return [], []
+        trials.sort(key=lambda tid: self._score[tid],
+                    reverse=(self.mode != "max"))  # worst first
+        n = max(1, int(len(trials) * self.quantile))
+        if n > len(trials) // 2:
+            n = len(trials) // 2
+        return trials[:n], trials[-n:]
+
+    def exploit(self, trial_id: str):
+        """Returns (restore_checkpoint_path, mutated_config) cloning a
+        random top-quantile trial, or None if none has a checkpoint."""
+        _, upper = self._quantiles()
+        donors = [u for u in upper if u in self._ckpt and u != trial_id]
+        if not donors:
+            return None
+        donor = self._rng.choice(donors)
+        cfg = dict(self._config.get(donor, {}))
+        for key, space in self.mutations.items():
+            if self._rng.random() < self.resample_p or key not in cfg:
+                cfg[key] = self._sample(space)
+            else:
+                cfg[key] = cfg[key] * self._rng.choice([0.8, 1.2])
+        self._config[trial_id] = dict(cfg)
+        return self._ckpt[donor], cfg
+
+    def _sample(self, space):
+        if callable(space):
+            return space()
+        if isinstance(space, (list, tuple)):
+            return self._rng.choice(list(space))
+        if hasattr(space, "sample"):
+            return space.sample(self._rng)
+        return space

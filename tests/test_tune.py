@@ -112,3 +112,63 @@ def test_checkpoint_and_errors(ray_mod, tmp_path_factory):
 
     with open(os.path.join(ok.checkpoint.path, "state.txt")) as f:
         assert f.read() == "2"
+
+
+def test_pbt_exploits_and_perturbs(ray_mod, tmp_path_factory):
+    """PopulationBasedTraining: a trial started with a bad hyperparameter
+    gets relaunched from a good trial's checkpoint with a mutated config
+    (reference tune/schedulers/pbt.py exploit/explore)."""
+    import json
+    import os
+
+    from ant_ray_amd import tune
+    from ant_ray_amd.train import Checkpoint, RunConfig
+    from ant_ray_amd.tune import PopulationBasedTraining, Tuner
+
+    storage = str(tmp_path_factory.mktemp("pbt"))
+
+    def trainable(config):
+        import tempfile
+
+        from ant_ray_amd import train
+
+        # resume: score continues from the donor's checkpointed score
+        base, start_it = 0.0, 0
+        ck = train.get_checkpoint()
+        if ck is not None:
+            with open(os.path.join(ck.path, "state.json")) as f:
+                st = json.load(f)
+            base, start_it = st["score"], st["it"]
+        lr = config["lr"]
+        import time as _t
+
+        for it in range(start_it + 1, 21):
+            _t.sleep(0.15)  # let the Tuner poll between iterations
+            base += lr  # good lr climbs faster
+            with tempfile.TemporaryDirectory() as d:
+                with open(os.path.join(d, "state.json"), "w") as f:
+                    json.dump({"score": base, "it": it, "lr": lr}, f)
+                train.report({"score": base, "training_iteration": it,
+                              "lr": lr},
+                             checkpoint=Checkpoint.from_directory(d))
+
+    pbt = PopulationBasedTraining(
+        perturbation_interval=5,
+        hyperparam_mutations={"lr": [0.1, 1.0]},
+        quantile_fraction=0.5,
+        resample_probability=1.0,
+        seed=7,
+    )
+    from ant_ray_amd.tune import TuneConfig
+    tuner = Tuner(
+        trainable,
+        param_space={"lr": tune.grid_search([0.001, 1.0])},
+        tune_config=TuneConfig(metric="score", mode="max", scheduler=pbt),
+        run_config=RunConfig(name="pbt", storage_path=storage),
+    )
+    grid = tuner.fit()
+    scores = sorted(r.metrics["score"] for r in grid)
+    # the bad-lr trial must have been perturbed onto a useful lr: its
+    # final score is far above what 20 iterations of lr=0.001 (0.02) give
+    assert scores[0] > 1.0, scores
+    assert grid.get_best_result(metric="score", mode="max").metrics["score"] >= 19.0
