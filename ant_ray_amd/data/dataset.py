@@ -193,6 +193,72 @@ class Dataset:
 
         return self._with(AllToAllOp(name="Union", fn=_union))
 
+    def join(self, other: "Dataset", *, on: Union[str, List[str]],
+             join_type: str = "inner",
+             right_on: Optional[Union[str, List[str]]] = None,
+             suffixes: tuple = ("", "_r")) -> "Dataset":
+        """Hash join with another dataset on key column(s) (parity:
+        reference data/_internal/execution/operators/join.py; arrow hash
+        join inside; join_type: inner/left outer/right outer/full outer)."""
+        keys = [on] if isinstance(on, str) else list(on)
+        rkeys = ([right_on] if isinstance(right_on, str)
+                 else list(right_on) if right_on else keys)
+        jt = {"inner": "inner", "left": "left outer", "right": "right outer",
+              "full": "full outer"}.get(join_type, join_type)
+
+        def _join(refs: List[Any]) -> List[Any]:
+            left_blocks = [ray.get(r) for r in refs]
+            right_blocks = [ray.get(r)
+                            for r in other.iter_internal_ref_bundles()]
+            if not left_blocks or not right_blocks:
+                if jt == "inner":
+                    return []
+                # outer joins with one empty side degrade to the other side
+                blocks = left_blocks or right_blocks
+                return [ray.put(b) for b in blocks]
+            lt = pa.concat_tables(left_blocks, promote_options="default")
+            rt = pa.concat_tables(right_blocks, promote_options="default")
+            res = lt.join(rt, keys=keys, right_keys=rkeys, join_type=jt,
+                          left_suffix=suffixes[0], right_suffix=suffixes[1])
+            k = max(len(refs), 1)
+            per = max(1, -(-res.num_rows // k))
+            return [ray.put(res.slice(s, per))
+                    for s in range(0, res.num_rows, per)]
+
+        return self._with(AllToAllOp(name=f"Join[{jt}]", fn=_join))
+
+    def zip(self, other: "Dataset") -> "Dataset":
+        """Column-wise zip of two same-length datasets (parity:
+        Dataset.zip — duplicate column names from the right side get a
+        _1 suffix)."""
+
+        def _zip(refs: List[Any]) -> List[Any]:
+            lt = pa.concat_tables([ray.get(r) for r in refs],
+                                  promote_options="default")
+            rblocks = [ray.get(r) for r in other.iter_internal_ref_bundles()]
+            rt = pa.concat_tables(rblocks, promote_options="default")
+            if lt.num_rows != rt.num_rows:
+                raise ValueError(
+                    f"zip needs equal row counts: {lt.num_rows} vs {rt.num_rows}")
+            cols = {name: lt.column(name) for name in lt.column_names}
+            for name in rt.column_names:
+                out = name if name not in cols else f"{name}_1"
+                cols[out] = rt.column(name)
+            res = pa.table(cols)
+            k = max(len(refs), 1)
+            per = max(1, -(-res.num_rows // k))
+            return [ray.put(res.slice(s, per))
+                    for s in range(0, res.num_rows, per)]
+
+        return self._with(AllToAllOp(name="Zip", fn=_zip))
+
+    def unique(self, column: str) -> List[Any]:
+        """Distinct values of a column (parity: Dataset.unique)."""
+        vals = set()
+        for block in self.iter_blocks():
+            vals.update(block.column(column).to_pylist())
+        return sorted(vals, key=lambda v: (v is None, v))
+
     def groupby(self, key: str) -> "GroupedData":
         return GroupedData(self, key)
 

@@ -200,3 +200,21 @@ def test_expressions(data):
                for r in rows)
     kept = ds.filter_expr((col("id") > 5) & (col("id") <= 10)).take_all()
     assert sorted(r["id"] for r in kept) == [6, 7, 8, 9, 10]
+
+
+def test_join_and_zip(data):
+    left = data.from_items([{"id": i, "a": i * 10} for i in range(6)])
+    right = data.from_items([{"id": i, "b": i * 100} for i in range(3, 9)])
+
+    joined = left.join(right, on="id").sort("id").take_all()
+    assert [r["id"] for r in joined] == [3, 4, 5]
+    assert joined[0]["a"] == 30 and joined[0]["b"] == 300
+
+    outer = left.join(right, on="id", join_type="left").count()
+    assert outer == 6
+
+    z = left.zip(data.from_items([{"c": i} for i in range(6)])).take_all()
+    assert z[0]["a"] == 0 and "c" in z[0]
+    assert len(z) == 6
+
+    assert left.unique("id") == [0, 1, 2, 3, 4, 5]
