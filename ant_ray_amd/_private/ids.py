@@ -92,9 +92,29 @@ class ActorID(BaseID):
 class TaskID(BaseID):
     SIZE = 16
 
+    # process-unique 6-byte prefix + per-process counter: os.urandom is a
+    # ~37us syscall and for_task sits on the task-submission hot path
+    # (ray_perf "tasks async" shape); collision across processes needs a
+    # 48-bit prefix clash AND the same counter value
+    _prefix = os.urandom(6)
+    _counter = 0
+    _lock = None  # lazily bound (threading import kept off the module top)
+
+    @staticmethod
+    def _refresh_prefix():
+        TaskID._prefix = os.urandom(6)
+        TaskID._counter = 0
+
     @classmethod
     def for_task(cls, job_id: JobID):
-        return cls(os.urandom(12) + job_id.binary())
+        if cls._lock is None:
+            import threading
+
+            cls._lock = threading.Lock()
+        with cls._lock:
+            cls._counter += 1
+            c = cls._counter
+        return cls(cls._prefix + c.to_bytes(6, "little") + job_id.binary())
 
     @classmethod
     def for_actor_task(cls, actor_id: ActorID, seq: int):
@@ -129,3 +149,9 @@ class ObjectID(BaseID):
 
 class PlacementGroupID(BaseID):
     SIZE = 16
+
+
+# a forked child inherits TaskID._prefix/_counter; without a refresh its
+# task ids would collide with the parent's
+if hasattr(os, "register_at_fork"):
+    os.register_at_fork(after_in_child=TaskID._refresh_prefix)
