@@ -695,6 +695,18 @@ class GcsServer:
     async def rpc_list_actors(self, conn, p):
         return [a.view() for a in self.actors.values()]
 
+    async def rpc_store_stats(self, conn, p):
+        """Fan out per-node object-store stats (for `ray memory`)."""
+        out = []
+        for node in list(self.nodes.values()):
+            if node.conn is None or node.conn.closed:
+                continue
+            try:
+                out.append(await node.conn.call("store_stats", {}, timeout=5))
+            except Exception:
+                pass
+        return out
+
     # --------------------------------------------------------- task events
     # parity: GcsTaskManager (src/ray/gcs/gcs_task_manager.h) fed by the
     # CoreWorker TaskEventBuffer; ring buffer, newest wins

@@ -675,6 +675,17 @@ class Raylet:
             "num_workers": len(self.workers),
         }
 
+    async def rpc_store_stats(self, conn, p):
+        """Object-store usage for `ray memory` / state API (parity:
+        reference raylet GetNodeStats -> memory summary)."""
+        try:
+            stats = dict(self._store.stats())
+        except Exception:
+            stats = {}
+        stats["node_id"] = self.node_id
+        stats["spilled_files"] = len(getattr(self, "_spilled", {}) or {})
+        return stats
+
     async def rpc_shutdown(self, conn, p):
         await self._do_shutdown()
         return {"ok": True}

@@ -106,6 +106,30 @@ def stop():
 
 
 @cli.command()
+def memory():
+    """Object-store memory usage per node (parity `ray memory`)."""
+    import ant_ray_amd as ray
+    from ant_ray_amd.util import state as S
+
+    if not ray.is_initialized():
+        ray.init(address=os.environ.get("RAY_ADDRESS", "auto"),
+                 ignore_reinit_error=True)
+    rows = S.store_stats()
+    if not rows:
+        click.echo("no object store stats available")
+        return
+    for s in rows:
+        used = s.get("used_bytes", 0)
+        cap = s.get("arena_size", 0) or 1
+        click.echo(f"node {str(s.get('node_id'))[:16]}: "
+                   f"{used / (1 << 20):.1f}/{cap / (1 << 20):.1f} MiB used, "
+                   f"{s.get('num_objects', 0)} objects, "
+                   f"{s.get('total_created', 0)} created, "
+                   f"{s.get('total_evicted', 0)} evicted, "
+                   f"{s.get('spilled_files', 0)} spilled")
+
+
+@cli.command()
 def status():
     """Cluster resource overview (parity `ray status`)."""
     import ant_ray_amd as ray

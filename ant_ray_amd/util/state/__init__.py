@@ -97,6 +97,14 @@ def list_objects(filters=None, limit: int = 1000, **_):
     return out[:limit]
 
 
+def store_stats() -> List[Dict[str, Any]]:
+    """Per-node shm object-store usage (backs `ray memory`)."""
+    out = []
+    for s in _gcs_call("store_stats"):
+        out.append(_hexify(dict(s)))
+    return out
+
+
 def summarize_tasks():
     from collections import Counter
 

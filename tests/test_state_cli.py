@@ -97,3 +97,27 @@ def test_job_submission(ray_mod):
     assert "hello from job 12321" in client.get_job_logs(job_id)
     jobs = client.list_jobs()
     assert any(j["submission_id"] == job_id for j in jobs)
+
+
+def test_cli_memory(ray_mod):
+    """`ray memory` prints per-node object-store usage from the raylet's
+    shm stats (parity: reference ray memory / raylet GetNodeStats)."""
+    import numpy as np
+
+    from click.testing import CliRunner
+
+    from ant_ray_amd.scripts.cli import cli
+    from ant_ray_amd.util import state
+
+    ray = ray_mod
+    keep = ray.put(np.zeros(1 << 20, dtype=np.uint8))  # 1 MiB in shm
+
+    rows = state.store_stats()
+    assert rows, "expected at least the head node's store stats"
+    assert rows[0]["arena_size"] > 0
+    assert any(r["total_created"] >= 1 for r in rows)
+
+    r = CliRunner().invoke(cli, ["memory"])
+    assert r.exit_code == 0, r.output
+    assert "MiB used" in r.output
+    del keep
