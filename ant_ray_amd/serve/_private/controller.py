@@ -125,11 +125,27 @@ class ServeController:
                 ray.kill(h)
             except Exception:
                 pass
-        # wait until new replicas construct (first health check)
+        # wait until new replicas construct (first health check) — bounded:
+        # an unplaceable replica (cluster out of CPUs/GPUs) must surface as
+        # a deploy error, not an indefinite hang (parity: the reference's
+        # deploy timeout / "1 replica pending allocation" status)
         if dr.replicas:
-            await asyncio.gather(*[
-                _aw(h.check_health.remote()) for h in dr.replicas
-            ], return_exceptions=True)
+            import os
+
+            deploy_timeout = float(
+                os.environ.get("ANTRAY_SERVE_DEPLOY_TIMEOUT_S", "60"))
+            try:
+                await asyncio.wait_for(
+                    asyncio.gather(*[
+                        _aw(h.check_health.remote()) for h in dr.replicas
+                    ], return_exceptions=True),
+                    timeout=deploy_timeout)
+            except asyncio.TimeoutError:
+                raise RuntimeError(
+                    f"deployment {app}/{dep_name}: replicas not ready after "
+                    f"{deploy_timeout:.0f}s — likely pending allocation "
+                    "(insufficient cluster resources for "
+                    f"{len(dr.replicas)} replicas)") from None
 
     # ------------------------------------------------------------ queries
 

@@ -1,0 +1,114 @@
+"""`serve` CLI — run/status/shutdown/delete an application.
+
+Role parity: reference python/ray/serve/scripts.py (`serve run
+module:app`, `serve status`, `serve shutdown`). The import-path target
+follows the reference convention: "pkg.module:app_object" where the
+object is a bound Application (Deployment.bind(...)).
+"""
+from __future__ import annotations
+
+import importlib
+import json
+import os
+import sys
+import time
+
+import click
+
+
+def _load_target(import_path: str):
+    if ":" not in import_path:
+        raise click.ClickException(
+            f"target {import_path!r} must look like 'module.sub:app'")
+    mod_name, _, attr = import_path.partition(":")
+    sys.path.insert(0, os.getcwd())
+    mod = importlib.import_module(mod_name)
+    try:
+        return getattr(mod, attr)
+    except AttributeError:
+        raise click.ClickException(
+            f"module {mod_name!r} has no attribute {attr!r}") from None
+
+
+@click.group()
+def cli():
+    """Manage Ray Serve applications."""
+
+
+@cli.command()
+@click.argument("import_path")
+@click.option("--name", default="default", help="Application name.")
+@click.option("--route-prefix", default="/", help="HTTP route prefix.")
+@click.option("--host", default="127.0.0.1")
+@click.option("--port", default=8000, type=int)
+@click.option("--blocking/--non-blocking", default=True)
+def run(import_path, name, route_prefix, host, port, blocking):
+    """Deploy and (by default) block: serve run my_module:app"""
+    import ant_ray_amd as ray
+    from ant_ray_amd import serve
+
+    if not ray.is_initialized():
+        addr = os.environ.get("RAY_ADDRESS")
+        ray.init(address=addr) if addr else ray.init()
+    app = _load_target(import_path)
+    serve.start(http_options={"host": host, "port": port})
+    serve.run(app, name=name, route_prefix=route_prefix)
+    click.echo(f"application {name!r} deployed at "
+               f"http://{host}:{port}{route_prefix}")
+    if blocking:
+        try:
+            while True:
+                time.sleep(3600)
+        except KeyboardInterrupt:
+            click.echo("shutting down")
+            serve.shutdown()
+
+
+@cli.command()
+def status():
+    """Show application/deployment status."""
+    import ant_ray_amd as ray
+    from ant_ray_amd import serve
+
+    if not ray.is_initialized():
+        ray.init(address=os.environ.get("RAY_ADDRESS", "auto"),
+                 ignore_reinit_error=True)
+    click.echo(json.dumps(serve.status(), indent=2, default=str))
+
+
+@cli.command()
+@click.argument("name")
+def delete(name):
+    """Delete one application."""
+    import ant_ray_amd as ray
+    from ant_ray_amd import serve
+
+    if not ray.is_initialized():
+        ray.init(address=os.environ.get("RAY_ADDRESS", "auto"),
+                 ignore_reinit_error=True)
+    serve.delete(name)
+    click.echo(f"deleted application {name!r}")
+
+
+@cli.command()
+@click.option("--yes", "-y", is_flag=True, default=False)
+def shutdown(yes):
+    """Tear down Serve (controller, proxies, all replicas)."""
+    import ant_ray_amd as ray
+    from ant_ray_amd import serve
+
+    if not yes:
+        click.confirm("Shut down Serve and all applications?", abort=True)
+    if not ray.is_initialized():
+        ray.init(address=os.environ.get("RAY_ADDRESS", "auto"),
+                 ignore_reinit_error=True)
+    serve.shutdown()
+    click.echo("serve shut down")
+
+
+def main():
+    cli()
+
+
+if __name__ == "__main__":
+    main()

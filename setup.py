@@ -42,4 +42,10 @@ setup(
     description="MI355X-native distributed actor/task runtime with Ray-compatible APIs",
     packages=["ant_ray_amd"],
     ext_modules=ext_modules,
+    entry_points={
+        "console_scripts": [
+            "ray=ant_ray_amd.scripts.cli:main",
+            "serve=ant_ray_amd.serve.scripts:main",
+        ],
+    },
 )

@@ -189,3 +189,50 @@ def test_multiplexed(serve_mod):
     out = h.remote("c").result(timeout_s=30)  # evicts LRU ("a")
     out = h.remote("a").result(timeout_s=30)  # reloads a
     assert out["loads"].count("a") == 2
+
+
+def test_serve_cli(serve_mod, tmp_path, monkeypatch):
+    """serve CLI: run (non-blocking) an import-path app, status, delete."""
+    from click.testing import CliRunner
+
+    from ant_ray_amd.serve.scripts import cli
+
+    app_mod = tmp_path / "cli_app_mod.py"
+    app_mod.write_text(
+        "from ant_ray_amd import serve\n"
+        "@serve.deployment\n"
+        "def hello(request=None):\n"
+        "    return 'hi-from-cli'\n"
+        "app = hello.bind()\n"
+    )
+    monkeypatch.chdir(tmp_path)
+    import sys
+
+    # earlier module tests leave their apps (and 1-CPU replicas) running;
+    # clear them so this app's replica can schedule in the 8-CPU session
+    serve = serve_mod
+    for app_name in list(serve.status().get("applications", {})):
+        serve.delete(app_name)
+
+    sys.path.insert(0, str(tmp_path))
+    try:
+        r = CliRunner().invoke(cli, [
+            "run", "cli_app_mod:app", "--name", "cliapp",
+            "--route-prefix", "/cliapp", "--non-blocking"])
+        assert r.exit_code == 0, r.output
+        assert "deployed" in r.output
+
+        r = CliRunner().invoke(cli, ["status"])
+        assert r.exit_code == 0, r.output
+        assert "cliapp" in r.output
+
+        import urllib.request
+
+        body = urllib.request.urlopen(
+            "http://127.0.0.1:8000/cliapp", timeout=30).read().decode()
+        assert "hi-from-cli" in body
+
+        r = CliRunner().invoke(cli, ["delete", "cliapp"])
+        assert r.exit_code == 0, r.output
+    finally:
+        sys.path.remove(str(tmp_path))
