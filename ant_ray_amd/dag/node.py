@@ -244,6 +244,11 @@ class CompiledDAG:
         self._seq = 0
         self._read_seq = 0
         self._results: dict = {}
+        import threading
+
+        # serializes the output-channel drain: concurrent ray.get on two
+        # CompiledDAGRefs (execute_async) must not interleave channel reads
+        self._result_lock = threading.Lock()
         try:
             self._channel_mode = self._compile_channels()
         except Exception:
@@ -398,22 +403,40 @@ class CompiledDAG:
         self._seq += 1
         return CompiledDAGRef(self, self._seq)
 
+    async def execute_async(self, *args, **kwargs):
+        """Asyncio variant (parity: CompiledDAG.execute_async returning an
+        awaitable future): returns an awaitable that resolves to the
+        execution's result without blocking the event loop."""
+        import asyncio
+
+        ref = self.execute(*args, **kwargs)
+        loop = asyncio.get_running_loop()
+
+        async def _await_result():
+            import ant_ray_amd as ray
+
+            return await loop.run_in_executor(None, lambda: ray.get(ref))
+
+        return asyncio.ensure_future(_await_result())
+
     def _result_for(self, seq: int, timeout=None):
         from ant_ray_amd.experimental.channel import _WrappedError
 
-        while self._read_seq < seq:
-            # read unwrapped: an error result must still advance the read
-            # cursor, else the next get() desynchronizes from the channel
-            vals = [c.read(timeout, unwrap=False) for c in self._out_chans]
-            self._read_seq += 1
-            self._results[self._read_seq] = (
-                vals if self._multi else vals[0])
-            # only the latest few results are retained
-            self._results.pop(self._read_seq - 8, None)
-        try:
-            out = self._results[seq]
-        except KeyError:
-            raise RuntimeError("compiled DAG result no longer buffered")
+        with self._result_lock:
+            while self._read_seq < seq:
+                # read unwrapped: an error result must still advance the
+                # read cursor, else the next get() desynchronizes from the
+                # channel
+                vals = [c.read(timeout, unwrap=False) for c in self._out_chans]
+                self._read_seq += 1
+                self._results[self._read_seq] = (
+                    vals if self._multi else vals[0])
+                # only the latest few results are retained
+                self._results.pop(self._read_seq - 8, None)
+            try:
+                out = self._results[seq]
+            except KeyError:
+                raise RuntimeError("compiled DAG result no longer buffered")
         if isinstance(out, _WrappedError):
             raise out.exc
         if isinstance(out, list):

@@ -202,3 +202,36 @@ class TestCompiledChannelDAG:
             ray.get(d2.execute(13))
         assert ray.get(d2.execute(2)) == 2
         d2.teardown()
+
+
+def test_execute_async(ray_mod):
+    """CompiledDAG.execute_async: awaitable results in submission order
+    (parity compiled_dag_node.py execute_async / CompiledDAGFuture)."""
+    import asyncio
+
+    import ant_ray_amd as ray
+    from ant_ray_amd.dag import InputNode
+
+    @ray.remote
+    class Adder:
+        def add(self, x):
+            return x + 100
+
+    a = Adder.remote()
+    with InputNode() as inp:
+        dag = a.add.bind(inp).experimental_compile()
+
+    async def main():
+        out = []
+        # channels are depth-1: keep at most two executions in flight
+        fut = await dag.execute_async(0)
+        for i in range(1, 4):
+            nxt = await dag.execute_async(i)
+            out.append(await fut)
+            fut = nxt
+        out.append(await fut)
+        return out
+
+    out = asyncio.run(main())
+    assert out == [100, 101, 102, 103]
+    dag.teardown()
