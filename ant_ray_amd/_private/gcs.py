@@ -147,13 +147,121 @@ class GcsServer:
         # node death (gcs_virtual_cluster_manager.cc:730)
         self.virtual_clusters: Dict[str, Dict[str, Any]] = {}
         self.worker_vc: Dict[bytes, str] = {}  # worker_id -> vc id
+        # fault tolerance: durable tables are snapshotted to this file and
+        # restored by a restarted GCS (role parity with the reference's
+        # Redis-backed gcs_table_storage / redis_store_client.cc — this
+        # air-gapped image has no Redis, so the store client is a file)
+        self.persist_path: Optional[str] = None
+        self._persist_task: Optional[asyncio.Task] = None
 
     # ------------------------------------------------------------------ serve
-    async def start(self, host="127.0.0.1", port=0):
+    async def start(self, host="127.0.0.1", port=0, persist_path=None):
+        self.persist_path = persist_path or os.environ.get("ANTRAY_GCS_PERSIST")
+        if self.persist_path and os.path.exists(self.persist_path):
+            try:
+                self._restore_tables()
+                logger.warning("GCS restored %d actors, %d kv namespaces, "
+                               "%d jobs from %s", len(self.actors),
+                               len(self.kv), len(self.jobs), self.persist_path)
+            except Exception:
+                logger.exception("GCS table restore failed; starting fresh")
         self._server, self.port = await protocol.serve(self._handle, host, port)
         asyncio.get_running_loop().create_task(self._health_loop())
         logger.info("GCS listening on %s:%s", host, self.port)
         return self.port
+
+    # ------------------------------------------------------- fault tolerance
+    def _persist_soon(self):
+        """Debounced snapshot: coalesce bursts of mutations into one write."""
+        if not self.persist_path or (
+                self._persist_task and not self._persist_task.done()):
+            return
+        try:
+            loop = asyncio.get_running_loop()
+        except RuntimeError:
+            return
+        self._persist_task = loop.create_task(self._persist_later())
+
+    async def _persist_later(self):
+        await asyncio.sleep(0.1)
+        try:
+            self._persist_now()
+        except Exception:
+            logger.exception("GCS table persist failed")
+
+    def _persist_now(self):
+        import msgpack
+
+        snap = {
+            "kv": {ns: dict(d) for ns, d in self.kv.items()},
+            "named_actors": [[list(k), v] for k, v in self.named_actors.items()],
+            "jobs": self.jobs,
+            "job_counter": self._job_counter,
+            "virtual_clusters": {
+                k: {**v, "nodes": list(v["nodes"])}
+                for k, v in self.virtual_clusters.items()
+            },
+            "actors": [
+                {
+                    "actor_id": a.actor_id, "owner": a.owner, "name": a.name,
+                    "namespace": a.namespace, "opts": a.opts,
+                    "create_payload": a.create_payload, "state": a.state,
+                    "addr": list(a.addr) if a.addr else None,
+                    "worker_id": a.worker_id, "node_id": a.node_id,
+                    "num_restarts": a.num_restarts,
+                    "death_cause": a.death_cause,
+                }
+                for a in self.actors.values()
+            ],
+            "pgs": [
+                {
+                    "pg_id": g.pg_id, "bundles": g.bundles,
+                    "strategy": g.strategy, "name": g.name, "state": g.state,
+                    "bundle_nodes": g.bundle_nodes,
+                }
+                for g in self.pgs.values()
+            ],
+        }
+        tmp = self.persist_path + ".tmp"
+        with open(tmp, "wb") as f:
+            f.write(msgpack.packb(snap, use_bin_type=True))
+        os.replace(tmp, self.persist_path)
+
+    def _restore_tables(self):
+        import msgpack
+
+        with open(self.persist_path, "rb") as f:
+            snap = msgpack.unpackb(f.read(), raw=False, strict_map_key=False)
+        self.kv = {ns: dict(d) for ns, d in snap.get("kv", {}).items()}
+        self.named_actors = {tuple(k): v
+                             for k, v in snap.get("named_actors", [])}
+        self.jobs = {int(k): v for k, v in snap.get("jobs", {}).items()}
+        self._job_counter = snap.get("job_counter", 0)
+        self.virtual_clusters = {
+            k: {**v, "nodes": set(v.get("nodes") or [])}
+            for k, v in snap.get("virtual_clusters", {}).items()
+        }
+        for a in snap.get("actors", []):
+            info = ActorInfo(a["actor_id"], a["owner"], a["name"],
+                             a["namespace"], a["opts"], a["create_payload"])
+            info.state = a["state"]
+            info.addr = tuple(a["addr"]) if a["addr"] else None
+            info.worker_id = a["worker_id"]
+            info.node_id = a["node_id"]
+            info.num_restarts = a["num_restarts"]
+            info.death_cause = a["death_cause"]
+            # in-flight creations did not survive the GCS death; their
+            # owners resubmit (non-detached) or the actor is simply gone
+            if info.state in (PENDING_CREATION, RESTARTING):
+                info.state = DEAD
+                info.death_cause = "GCS restarted during actor creation"
+            self.actors[a["actor_id"]] = info
+        for g in snap.get("pgs", []):
+            pg = PlacementGroupInfo(g["pg_id"], g["bundles"], g["strategy"],
+                                    g.get("name", ""))
+            pg.state = g["state"]
+            pg.bundle_nodes = g["bundle_nodes"]
+            self.pgs[g["pg_id"]] = pg
 
     async def _health_loop(self):
         while not self._shutdown.is_set():
@@ -307,6 +415,7 @@ class GcsServer:
             vc = {"id": vc_id, "divisible": bool(p.get("divisible", False)),
                   "nodes": set(), "node_count": 0, "revision": 0}
             self.virtual_clusters[vc_id] = vc
+            self._persist_soon()
         if "node_ids" in p and p["node_ids"] is not None:
             wanted = {bytes.fromhex(n) if isinstance(n, str) else n
                       for n in p["node_ids"]}
@@ -341,6 +450,8 @@ class GcsServer:
 
     async def rpc_remove_virtual_cluster(self, conn, p):
         vc = self.virtual_clusters.pop(p["virtual_cluster_id"], None)
+        if vc is not None:
+            self._persist_soon()
         return {"ok": vc is not None}
 
     async def rpc_list_virtual_clusters(self, conn, p):
@@ -357,6 +468,7 @@ class GcsServer:
         if not p.get("overwrite", True) and key in ns:
             return {"added": False}
         ns[key] = p["value"]
+        self._persist_soon()
         return {"added": True}
 
     async def rpc_kv_get(self, conn, p):
@@ -366,6 +478,8 @@ class GcsServer:
     async def rpc_kv_del(self, conn, p):
         ns = self.kv.get(p.get("ns", ""), {})
         existed = ns.pop(p["key"], None) is not None
+        if existed:
+            self._persist_soon()
         return {"deleted": existed}
 
     async def rpc_kv_keys(self, conn, p):
@@ -375,6 +489,7 @@ class GcsServer:
 
     async def rpc_next_job_id(self, conn, p):
         self._job_counter += 1
+        self._persist_soon()
         return {"job_id": self._job_counter}
 
     # ----------------------------------------------------------------- pubsub
@@ -410,6 +525,8 @@ class GcsServer:
         return evs[-limit:]
 
     async def _publish(self, channel: str, key: bytes, data: Any):
+        if channel == "ACTOR":
+            self._persist_soon()
         subs = self._subscribers.get(channel, [])
         dead = []
         for c in subs:
@@ -540,6 +657,10 @@ class GcsServer:
                             "runtime_env": opts.get("runtime_env"),
                             "pg": pg,
                             "detached": bool(opts.get("lifetime") == "detached"),
+                            # deny-don't-queue: a full node must not hold
+                            # this scheduling loop while other nodes are
+                            # free (it retries every 0.2s)
+                            "no_wait": True,
                         },
                         timeout=120,
                     )
@@ -875,9 +996,10 @@ class GcsServer:
         return {"ok": True}
 
 
-async def run_gcs(host="127.0.0.1", port=0, announce_fd: int = None):
+async def run_gcs(host="127.0.0.1", port=0, announce_fd: int = None,
+                  persist_path: str = None):
     gcs = GcsServer()
-    bound = await gcs.start(host, port)
+    bound = await gcs.start(host, port, persist_path=persist_path)
     if announce_fd is not None:
         os.write(announce_fd, (str(bound) + "\n").encode())
         os.close(announce_fd)
@@ -888,9 +1010,11 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--host", default="127.0.0.1")
     ap.add_argument("--port", type=int, default=0)
+    ap.add_argument("--persist", default=None,
+                    help="snapshot/restore durable tables at this path")
     args = ap.parse_args()
     logging.basicConfig(level=logging.INFO)
-    asyncio.run(run_gcs(args.host, args.port))
+    asyncio.run(run_gcs(args.host, args.port, persist_path=args.persist))
 
 
 if __name__ == "__main__":

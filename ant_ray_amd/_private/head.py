@@ -34,7 +34,9 @@ async def run_head(args):
     os.makedirs(os.path.join(session_dir, "logs"), exist_ok=True)
 
     gcs = GcsServer()
-    gcs_port = await gcs.start(args.host, args.port)
+    gcs_port = await gcs.start(
+        args.host, args.port,
+        persist_path=os.path.join(session_dir, "gcs_tables.msgpack"))
 
     resources = json.loads(args.resources) if args.resources else {}
     ncpu = args.num_cpus if args.num_cpus >= 0 else os.cpu_count()

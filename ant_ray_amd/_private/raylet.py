@@ -128,18 +128,8 @@ class Raylet:
             self._store = ShmStore.open(self.store_path)
         self._server, self.port = await protocol.serve(self._handle, self.node_ip, port)
         self.gcs_conn = await protocol.connect(self.gcs_addr, self._handle, name="raylet->gcs")
-        self.gcs_conn.on_close = lambda c: os._exit(1)  # fate-share with GCS
-        await self.gcs_conn.call(
-            "register_node",
-            {
-                "node_id": self.node_id,
-                "addr": [self.node_ip, self.port],
-                "resources": self.resources_total,
-                "store_path": self.store_path,
-                "object_store_bytes": self.store_capacity,
-                "labels": self.labels,
-            },
-        )
+        self.gcs_conn.on_close = self._on_gcs_conn_closed
+        await self.gcs_conn.call("register_node", self._register_payload())
         asyncio.get_running_loop().create_task(self._heartbeat_loop())
         asyncio.get_running_loop().create_task(self._reap_loop())
         asyncio.get_running_loop().create_task(self._memory_monitor_loop())
@@ -148,6 +138,50 @@ class Raylet:
             {k: v for k, v in self.resources_total.items()},
         )
         return self.port
+
+    def _register_payload(self):
+        return {
+            "node_id": self.node_id,
+            "addr": [self.node_ip, self.port],
+            "resources": self.resources_total,
+            "store_path": self.store_path,
+            "object_store_bytes": self.store_capacity,
+            "labels": self.labels,
+        }
+
+    def _on_gcs_conn_closed(self, conn):
+        """GCS fault tolerance (parity: reference raylets survive a GCS
+        restart and re-register via NotifyGCSRestart, node_manager.proto
+        :446): reconnect with backoff instead of fate-sharing; exit only
+        after the reconnect window expires."""
+        if self._shutdown.is_set():
+            return
+        try:
+            asyncio.get_running_loop().create_task(self._reconnect_gcs())
+        except RuntimeError:
+            os._exit(1)  # event loop gone: process is tearing down anyway
+
+    async def _reconnect_gcs(self):
+        window = float(os.environ.get("ANTRAY_GCS_RECONNECT_TIMEOUT_S", "60"))
+        deadline = time.monotonic() + window
+        logger.warning("GCS connection lost; reconnecting for up to %.0fs",
+                       window)
+        while time.monotonic() < deadline and not self._shutdown.is_set():
+            try:
+                conn = await protocol.connect(self.gcs_addr, self._handle,
+                                              name="raylet->gcs")
+                conn.on_close = self._on_gcs_conn_closed
+                await conn.call("register_node", self._register_payload(),
+                                timeout=10)
+                self.gcs_conn = conn
+                logger.warning("re-registered with restarted GCS")
+                return
+            except Exception:
+                await asyncio.sleep(1.0)
+        if self._shutdown.is_set():
+            return
+        logger.error("GCS unreachable after %.0fs; raylet exiting", window)
+        os._exit(1)
 
     async def _heartbeat_loop(self):
         while not self._shutdown.is_set():
@@ -410,6 +444,12 @@ class Raylet:
                 remaining -= back
 
     def _reclaim_leases_of(self, owner_conn):
+        if owner_conn is self.gcs_conn:
+            # the GCS connection dropping is a GCS restart, not a lessee
+            # death: actor workers it leased (incl. detached actors) keep
+            # running; the restarted GCS reconciles from its persisted
+            # actor table
+            return
         for w2 in list(self.workers.values()):
             if w2.leased and w2.lease_owner is owner_conn:
                 logger.info("reclaiming lease of dead lessee: worker %s",
@@ -478,6 +518,18 @@ class Raylet:
         fut = asyncio.get_running_loop().create_future()
         self.pending_leases.append(((p, conn), fut))
         self._pump_leases()
+        if p.get("no_wait") and any(x[1] is fut for x in self.pending_leases):
+            # GCS actor scheduling: the pump left us queued (resources
+            # unavailable NOW) — deny instead of holding the RPC so the
+            # scheduler can try another (possibly empty) node right away;
+            # blocking here serializes actor placement behind whichever
+            # node was picked first (parity: reference raylets reply with
+            # spillback rather than holding the lease). A request the pump
+            # already picked up (async worker spawn) is NOT denied.
+            self.pending_leases = [x for x in self.pending_leases
+                                   if x[1] is not fut]
+            fut.cancel()
+            return {"granted": False, "reason": "resources unavailable now"}
         return await fut
 
     def _pump_leases(self):

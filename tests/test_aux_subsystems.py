@@ -76,8 +76,13 @@ def test_autoscaler_with_local_provider():
                 return "ok"
 
         actors = [Fat.remote() for _ in range(2)]
-        time.sleep(1.0)
-        scaler.update()
+        # demand may take a moment to reach the GCS under load: keep
+        # running autoscaler rounds until a worker node exists
+        deadline = time.time() + 60
+        while (provider.non_terminated_nodes().get("worker", 0) < 1
+               and time.time() < deadline):
+            time.sleep(0.5)
+            scaler.update()
         assert provider.non_terminated_nodes().get("worker", 0) >= 1
         # once nodes join, the pending actors schedule
         assert ray.get([a.ping.remote() for a in actors], timeout=120) == \
