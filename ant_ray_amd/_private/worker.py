@@ -25,6 +25,7 @@ reference's io_service threads.
 """
 from __future__ import annotations
 
+import asyncio
 import atexit
 import logging
 import os
@@ -211,6 +212,48 @@ class CoreWorker:
             raise ConnectionError(reply["error"])
         self.job_id = reply.get("job_id") or 0
         await self.gcs.call("subscribe", {"channels": ["ACTOR", "NODE"]})
+        # GCS fault tolerance: reconnect + re-register + resubscribe when
+        # the GCS restarts (raylet does the same; see raylet._reconnect_gcs)
+        self._is_driver = is_driver
+        self.gcs.on_close = self._on_gcs_conn_closed
+
+    def _on_gcs_conn_closed(self, conn):
+        if not self.connected:
+            return
+        try:
+            asyncio.get_running_loop().create_task(self._reconnect_gcs())
+        except RuntimeError:
+            pass  # io loop gone: process shutting down
+
+    async def _reconnect_gcs(self):
+        import asyncio as _a
+
+        window = float(os.environ.get("ANTRAY_GCS_RECONNECT_TIMEOUT_S", "60"))
+        deadline = time.monotonic() + window
+        logger.warning("GCS connection lost; reconnecting for up to %.0fs",
+                       window)
+        while time.monotonic() < deadline and self.connected:
+            try:
+                conn = await protocol.connect(tuple(self.gcs_addr),
+                                              self._handle_rpc, name="->gcs")
+                await conn.call("register_worker", {
+                    "worker_id": self.worker_id,
+                    "addr": list(self.addr),
+                    "node_id": self.node_id,
+                    "is_driver": getattr(self, "_is_driver", False),
+                    "job_id": self.job_id,
+                    "pid": os.getpid(),
+                }, timeout=10)
+                await conn.call("subscribe", {"channels": ["ACTOR", "NODE"]})
+                conn.on_close = self._on_gcs_conn_closed
+                self.gcs = conn
+                logger.warning("reconnected to restarted GCS")
+                return
+            except Exception:
+                await _a.sleep(1.0)
+        if self.connected:
+            logger.error("GCS unreachable after %.0fs; GCS-dependent calls "
+                         "will fail until it returns", window)
 
     def connect_local_mode(self):
         self.job_id = 1

@@ -92,13 +92,25 @@ def test_gcs_restart_with_persistence(tmp_path):
         assert ray.get(k.get.remote(), timeout=60) == "alive-across-gcs-restart"
         time.sleep(0.5)  # let the debounced snapshot land
         assert os.path.exists(persist)
-        ray.shutdown()
 
-        # ---- kill the GCS hard, restart it on the same port
+        # ---- kill the GCS hard, restart it on the same port; the LIVE
+        # driver must reconnect on its own (worker._reconnect_gcs)
         gcs.send_signal(signal.SIGKILL)
         gcs.wait(timeout=10)
         gcs = _spawn_gcs(port, persist)
         assert _wait_port(port)
+
+        deadline = time.time() + 60
+        while True:
+            try:
+                assert internal_kv._internal_kv_get(b"durable_key") == \
+                    b"durable_value"
+                break
+            except Exception:
+                if time.time() > deadline:
+                    raise
+                time.sleep(0.5)
+        ray.shutdown()
 
         # fresh driver: KV restored, named detached actor reachable, and
         # the raylet re-registered (its worker kept the actor alive)
