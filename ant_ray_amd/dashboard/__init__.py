@@ -18,6 +18,7 @@ dashboard.start_dashboard(port).
 from __future__ import annotations
 
 import json
+import os
 import threading
 from typing import Optional
 
@@ -105,6 +106,31 @@ class DashboardHead:
             return js(await self._gcs("list_task_events", {"limit": 1000}))
         if path == "/api/placement_groups":
             return js(await self._gcs("list_placement_groups"))
+        if path == "/api/cluster_events":
+            return js(await self._gcs("list_cluster_events", {"limit": 1000}))
+        if path == "/api/memory":
+            return js(await self._gcs("store_stats"))
+        if path == "/api/logs" or path.startswith("/api/logs/"):
+            # session log serving (reference dashboard log module:
+            # log_agent.py file listing + tail)
+            from ant_ray_amd._private.worker import global_worker
+
+            cw = global_worker.core_worker
+            logs_dir = os.path.join(
+                getattr(cw, "session_dir", "") or "", "logs")
+            if not os.path.isdir(logs_dir):
+                return js([])
+            if path == "/api/logs":
+                return js(sorted(os.listdir(logs_dir)))
+            name = os.path.basename(path.split("/api/logs/", 1)[1])
+            fp = os.path.join(logs_dir, name)
+            if not os.path.isfile(fp):
+                return 404, b"text/plain", b"no such log"
+            with open(fp, "rb") as f:
+                f.seek(0, 2)
+                size = f.tell()
+                f.seek(max(0, size - 64 * 1024))
+                return 200, b"text/plain", f.read()
         if path == "/api/virtual_clusters":
             return js(await self._gcs("list_virtual_clusters"))
         if path == "/metrics":

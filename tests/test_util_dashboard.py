@@ -97,6 +97,18 @@ def test_metrics_and_dashboard(ray_mod):
     assert st["nodes"] and "resources" in st
     with urllib.request.urlopen(f"{base}/api/actors", timeout=10) as r:
         assert isinstance(json.loads(r.read()), list)
+    with urllib.request.urlopen(f"{base}/api/memory", timeout=10) as r:
+        mem = json.loads(r.read())
+    assert mem and "arena_size" in mem[0]
+    with urllib.request.urlopen(f"{base}/api/cluster_events", timeout=10) as r:
+        assert isinstance(json.loads(r.read()), list)
+    with urllib.request.urlopen(f"{base}/api/logs", timeout=10) as r:
+        logs = json.loads(r.read())
+    assert isinstance(logs, list)
+    if logs:
+        with urllib.request.urlopen(f"{base}/api/logs/{logs[0]}",
+                                    timeout=10) as r:
+            r.read()
     with urllib.request.urlopen(f"{base}/metrics", timeout=10) as r:
         text = r.read().decode()
     assert 'test_requests_total{route="/a"} 3.0' in text
