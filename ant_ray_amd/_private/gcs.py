@@ -670,6 +670,14 @@ class GcsServer:
                 if lease and lease.get("granted"):
                     await self._push_actor_creation(info, node, lease)
                     return
+                if not pg:
+                    # denied/failed: credit the optimistic debit back now
+                    # instead of waiting ~1s for the next heartbeat to
+                    # true it up (the retry loop would otherwise see a
+                    # phantom-full node)
+                    for k, v in resources.items():
+                        node.resources_available[k] = (
+                            node.resources_available.get(k, 0) + v)
             if time.monotonic() > deadline or self._shutdown.is_set():
                 info.state = DEAD
                 self._record_event("ACTOR", "DEAD", info.actor_id, info.death_cause)
