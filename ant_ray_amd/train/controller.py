@@ -223,6 +223,20 @@ class TrainController:
                                                    rep["metrics"])
             errs = [st["error"] for st in statuses if st["status"] == "errored"]
             if errs:
+                # drain the other ranks' queued reports before tearing the
+                # group down: a checkpoint reported just before a peer's
+                # failure must be registered, or the restart resumes from
+                # an older (or no) checkpoint
+                time.sleep(0.3)
+                try:
+                    for st in group.poll():
+                        for rep in st["reports"]:
+                            self.latest_metrics[rep["rank"]] = rep["metrics"]
+                            if rep["checkpoint_path"]:
+                                self.ckpt_manager.register(
+                                    rep["checkpoint_path"], rep["metrics"])
+                except Exception:
+                    pass
                 return errs[0]
             if all(st["status"] == "finished" for st in statuses):
                 return None

@@ -95,9 +95,15 @@ def test_trainer_failure_restart_restores_checkpoint(ray_mod, tmp_path_factory):
         start = 0
         if restored is not None:
             start = restored.get_metadata()["step"] + 1
+        rank = train.get_context().get_world_rank()
+        my_marker = f"{marker}.rank{rank}"
         for step in range(start, 4):
-            if step == 2 and not os.path.exists(marker):
-                open(marker, "w").close()
+            if step == 2 and not os.path.exists(my_marker):
+                # EVERY rank fails its first attempt at step 2: no rank can
+                # race through to completion before the group is torn down
+                # (a global marker let the surviving rank checkpoint step 3
+                # with resumed=False and made the restart a no-op)
+                open(my_marker, "w").close()
                 raise RuntimeError("boom")
             if train.get_context().get_world_rank() == 0:
                 with tempfile.TemporaryDirectory() as d:
