@@ -14,6 +14,7 @@ from typing import Any, List, Optional, Sequence, Union
 
 from ant_ray_amd import exceptions
 from ant_ray_amd._private.object_ref import ObjectRef
+from ant_ray_amd._private.worker import ObjectRefGenerator  # noqa: F401
 from ant_ray_amd._private.worker import (
     DRIVER_MODE,
     LOCAL_MODE,
@@ -30,7 +31,7 @@ __all__ = [
     "init", "shutdown", "is_initialized", "remote", "get", "put", "wait",
     "kill", "cancel", "get_actor", "get_gpu_ids", "get_runtime_context",
     "nodes", "cluster_resources", "available_resources", "method",
-    "ObjectRef", "ActorHandle", "exceptions", "actor", "remote_function",
+    "ObjectRef", "ObjectRefGenerator", "ActorHandle", "exceptions", "actor", "remote_function",
     "util", "train", "serve", "data", "tune",
 ]
 

@@ -55,7 +55,33 @@ class LocalModeExecutor:
     def submit_task(self, fn, args, kwargs, opts):
         task_id = TaskID.for_task(JobID.from_int(self.cw.job_id or 0)).binary()
         n = opts.get("num_returns", 1)
+        if n in ("streaming", "dynamic"):
+            return self._store_stream(task_id, fn, args, kwargs)
         return self._store_result(task_id, n, fn, args, kwargs, getattr(fn, "__name__", "task"))
+
+    def _store_stream(self, task_id, fn, args, kwargs):
+        """local_mode streaming: materialize the generator eagerly and hand
+        back an ObjectRefGenerator over the stored items."""
+        from ant_ray_amd._private.worker import ObjectRefGenerator, _StreamState
+
+        st = _StreamState()
+        self.cw._streams = getattr(self.cw, "_streams", {})
+        self.cw._streams[task_id] = st
+        try:
+            args = tuple(self._resolve(a) for a in args)
+            kwargs = {k: self._resolve(v) for k, v in kwargs.items()}
+            count = 0
+            for v in fn(*args, **kwargs):
+                oid = ObjectID.for_return(TaskID(task_id), count).binary()
+                self.cw.store.memory.put(oid, v)
+                st.refs.append(oid)
+                count += 1
+            st.total = count
+        except Exception as e:  # noqa: BLE001
+            st.error = (e if isinstance(e, RayTaskError)
+                        else RayTaskError.from_exception(
+                            e, getattr(fn, "__name__", "task")))
+        return ObjectRefGenerator(task_id, self.cw)
 
     def create_actor(self, cls, actor_id, args, kwargs, opts):
         args = tuple(self._resolve(a) for a in args)

@@ -358,42 +358,72 @@ class TaskExecutor:
         self._actor_loop.submit(runner())
 
     # ---------------------------------------------------------------- replies
+    def _store_one_return(self, payload, i: int, v) -> dict:
+        """Serialize + place return value i; returns the result entry."""
+        transport = payload.get("tensor_transport")
+        oid = ObjectID.for_return(TaskID(payload["task_id"]), i).binary()
+        if transport == "hip_ipc":
+            with serialization.gpu_transport_context("hip_ipc") as gctx:
+                sobj = serialization.serialize(v)
+            if gctx.pinned:
+                from ant_ray_amd.experimental.gpu_object_manager import (
+                    gpu_object_store,
+                )
+
+                gpu_object_store.add(oid, gctx.pinned)
+        else:
+            sobj = serialization.serialize(v)
+        self.cw._register_escapes(sobj)
+        if sobj.total_size <= INLINE_OBJECT_MAX:
+            entry = {"oid": oid, "inline": sobj.to_bytes(), "meta": sobj.metadata}
+            if transport == "hip_ipc":
+                # the device tensors stay pinned here; record us as holder
+                # so frees reach this process
+                entry["holder"] = list(self.cw.addr)
+            return entry
+        self.cw.store.put_serialized_to_shm(oid, sobj)
+        return {"oid": oid, "inline": None, "holder": list(self.cw.addr)}
+
     def _reply_results(self, payload, result) -> dict:
         self._record_event(payload, getattr(self, '_last_t0', time.time()),
                            time.time(), ok=True)
+        if payload.get("streaming"):
+            return self._stream_results(payload, result)
         n = payload.get("n_returns", 1)
         if n == 0:
             return {"status": "ok", "results": []}
         values = (result,) if n == 1 else tuple(result)
         if n > 1 and len(values) != n:
             raise ValueError(f"task returned {len(values)} values, expected {n}")
-        transport = payload.get("tensor_transport")
-        out = []
-        for i, v in enumerate(values):
-            oid = ObjectID.for_return(TaskID(payload["task_id"]), i).binary()
-            if transport == "hip_ipc":
-                with serialization.gpu_transport_context("hip_ipc") as gctx:
-                    sobj = serialization.serialize(v)
-                if gctx.pinned:
-                    from ant_ray_amd.experimental.gpu_object_manager import (
-                        gpu_object_store,
-                    )
-
-                    gpu_object_store.add(oid, gctx.pinned)
-            else:
-                sobj = serialization.serialize(v)
-            self.cw._register_escapes(sobj)
-            if sobj.total_size <= INLINE_OBJECT_MAX:
-                entry = {"oid": oid, "inline": sobj.to_bytes(), "meta": sobj.metadata}
-                if transport == "hip_ipc":
-                    # the device tensors stay pinned here; record us as holder
-                    # so frees reach this process
-                    entry["holder"] = list(self.cw.addr)
-                out.append(entry)
-            else:
-                self.cw.store.put_serialized_to_shm(oid, sobj)
-                out.append({"oid": oid, "inline": None, "holder": list(self.cw.addr)})
+        out = [self._store_one_return(payload, i, v)
+               for i, v in enumerate(values)]
         return {"status": "ok", "results": out}
+
+    def _stream_results(self, payload, result) -> dict:
+        """num_returns='streaming' (parity: reference streaming generators /
+        ObjectRefGenerator, core_worker TaskManager dynamic returns): the
+        task body is a generator; each yielded value becomes return object
+        i, pushed to the caller the moment it is produced via a
+        stream_item notify, so the consumer overlaps with the producer."""
+        if not hasattr(result, "__next__"):
+            result = iter(
+                result if hasattr(result, "__iter__") else (result,))
+        caller_addr = tuple(payload["caller_addr"])
+        task_id = payload["task_id"]
+        count = 0
+        for v in result:
+            entry = self._store_one_return(payload, count, v)
+            count += 1
+
+            async def _send(entry=entry, idx=count - 1):
+                conn = await self.cw._get_worker_conn_async_cached(caller_addr)
+                await conn.notify("stream_item", {
+                    "task_id": task_id, "index": idx, "entry": entry})
+
+            # sent in order on the io loop; failures surface on the final
+            # reply path (caller gone -> push reply fails too)
+            self.cw.io.submit(_send()).result(60)
+        return {"status": "ok", "results": [], "streaming_done": count}
 
     def _error_reply(self, payload, exc: BaseException) -> dict:
         if isinstance(exc, RayTaskError):

@@ -81,6 +81,77 @@ class _ErrorResult:
         self.exc = exc
 
 
+class _StreamState:
+    """Owner-side state of one streaming task (generator returns)."""
+
+    __slots__ = ("refs", "received", "total", "error", "cv")
+
+    def __init__(self):
+        self.refs: List[bytes] = []   # oids in yield order
+        self.received = 0
+        self.total: Optional[int] = None  # set by the final reply
+        self.error: Optional[BaseException] = None
+        self.cv = threading.Condition()
+
+
+class ObjectRefGenerator:
+    """Iterator over a streaming task's yielded ObjectRefs (parity:
+    reference ObjectRefGenerator for num_returns='streaming'): each
+    __next__ blocks until the worker has produced the next item and
+    returns its ObjectRef; StopIteration once the generator task finished
+    and every yielded ref was handed out."""
+
+    def __init__(self, task_id: bytes, worker: "CoreWorker"):
+        self._task_id = task_id
+        self._worker = worker
+        self._cursor = 0
+
+    def __iter__(self):
+        return self
+
+    def __next__(self):
+        st = self._worker._streams.get(self._task_id)
+        if st is None:
+            raise StopIteration
+        with st.cv:
+            while True:
+                if self._cursor < len(st.refs):
+                    oid = st.refs[self._cursor]
+                    self._cursor += 1
+                    return ObjectRef(oid, self._worker.addr,
+                                     worker=self._worker)
+                if st.error is not None:
+                    raise st.error
+                if st.total is not None and self._cursor >= st.total:
+                    self._worker._streams.pop(self._task_id, None)
+                    raise StopIteration
+                st.cv.wait(0.5)
+
+    async def __anext__(self):
+        import asyncio as _a
+
+        _END = object()
+
+        def step():
+            # StopIteration cannot cross a Future boundary (PEP 479)
+            try:
+                return self.__next__()
+            except StopIteration:
+                return _END
+
+        out = await _a.get_running_loop().run_in_executor(None, step)
+        if out is _END:
+            raise StopAsyncIteration
+        return out
+
+    def __aiter__(self):
+        return self
+
+    def completed(self):
+        """Drain-and-count convenience (non-standard helper)."""
+        return list(self)
+
+
 class LeasedWorker:
     def __init__(self, worker_id, addr, node_id, conn):
         self.worker_id = worker_id
@@ -127,6 +198,7 @@ class CoreWorker:
         self._owned: Dict[bytes, dict] = {}
         self._spilled: Dict[bytes, str] = {}  # oid -> spill file path
         self._pull_cache: Dict[bytes, tuple] = {}  # big-object pull slices
+        self._streams: Dict[bytes, "_StreamState"] = {}  # streaming tasks
         self._killed_workers: Dict[bytes, str] = {}  # worker_id -> kill reason
         self._object_locations: Dict[bytes, Tuple[str, int]] = {}
         self._actors: Dict[bytes, ActorHandleState] = {}
@@ -351,6 +423,26 @@ class CoreWorker:
 
     async def rpc_ping(self, conn, p):
         return {"ok": True, "worker_id": self.worker_id}
+
+    async def rpc_stream_item(self, conn, p):
+        """A streaming task produced its next yielded value (executor
+        task_executor._stream_results); store it and wake the
+        ObjectRefGenerator."""
+        st = self._streams.get(p["task_id"])
+        r = p["entry"]
+        oid = r["oid"]
+        if r.get("holder"):
+            self._object_locations[oid] = tuple(r["holder"])
+        if r.get("inline") is not None:
+            self.store.memory.put(oid, _RawResult(r["inline"], r.get("meta", b"py")))
+        else:
+            self.store.memory.put(oid, IN_PLASMA)
+        if st is not None:
+            with st.cv:
+                st.refs.append(oid)
+                st.received += 1
+                st.cv.notify_all()
+        return {"ok": True}
 
     def _pull_source(self, oid):
         """Resolve an object's serialized bytes for the pull data plane.
@@ -779,15 +871,21 @@ class CoreWorker:
         res = opts.get("resources") or {}
         return fn_id.hex() + "|" + ",".join(f"{k}={v}" for k, v in sorted(res.items()))
 
-    def submit_task(self, fn, fn_id: bytes, args, kwargs, opts: dict) -> List[ObjectRef]:
+    def submit_task(self, fn, fn_id: bytes, args, kwargs, opts: dict):
         n_returns = opts.get("num_returns", 1)
+        streaming = n_returns in ("streaming", "dynamic")
         task_id = TaskID.for_task(JobID.from_int(self.job_id or 0)).binary()
-        refs = [
-            ObjectRef(ObjectID.for_return(TaskID(task_id), i).binary(), self.addr, worker=self)
-            for i in range(n_returns)
-        ]
-        for r in refs:
-            self.store.memory.mark_pending(r.binary())
+        if streaming:
+            n_returns = 0
+            self._streams[task_id] = _StreamState()
+            refs = ObjectRefGenerator(task_id, self)
+        else:
+            refs = [
+                ObjectRef(ObjectID.for_return(TaskID(task_id), i).binary(), self.addr, worker=self)
+                for i in range(n_returns)
+            ]
+            for r in refs:
+                self.store.memory.mark_pending(r.binary())
         sobj = serialization.serialize((args, kwargs))
         self._register_escapes(sobj)
         if fn_id not in self._pushed_fns:
@@ -802,10 +900,11 @@ class CoreWorker:
             "fn": None,
             "args": sobj.to_bytes(),
             "n_returns": n_returns,
+            "streaming": streaming,
             "caller": self.worker_id,
             "caller_addr": list(self.addr),
             "name": opts.get("name", ""),
-            "max_retries": opts.get("max_retries", 3),
+            "max_retries": 0 if streaming else opts.get("max_retries", 3),
         }
         resources = dict(opts.get("resources") or {})
         resources.setdefault("CPU", float(opts.get("num_cpus", 1)))
@@ -933,11 +1032,38 @@ class CoreWorker:
 
     def _fail_task(self, payload, exc: BaseException):
         task_id = payload["task_id"]
+        st = self._streams.get(task_id)
+        if st is not None:
+            with st.cv:
+                st.error = exc
+                st.cv.notify_all()
         for i in range(payload.get("n_returns", 1)):
             oid = ObjectID.for_return(TaskID(task_id), i).binary()
             self.store.memory.put(oid, _ErrorResult(exc))
 
     def _handle_task_reply(self, payload, reply):
+        if payload.get("streaming"):
+            st = self._streams.get(payload["task_id"])
+            if st is not None:
+                with st.cv:
+                    if reply.get("status") == "ok":
+                        st.total = reply.get("streaming_done", 0)
+                    else:
+                        data = reply.get("error_payload")
+                        if data is not None:
+                            v = serialization.deserialize(
+                                memoryview(data),
+                                reply.get("error_meta",
+                                          serialization.META_ERROR))
+                            st.error = (v.as_instanceof_cause()
+                                        if isinstance(v, RayTaskError) else
+                                        v if isinstance(v, BaseException) else
+                                        RaySystemError(str(v)))
+                        else:
+                            st.error = RaySystemError(
+                                reply.get("error", "streaming task failed"))
+                    st.cv.notify_all()
+            return
         if reply.get("status") == "ok":
             for r in reply.get("results", []):
                 oid = r["oid"]
@@ -1019,17 +1145,23 @@ class CoreWorker:
 
     def submit_actor_task(self, actor_id: bytes, method_name: str, args, kwargs, opts: dict):
         n_returns = opts.get("num_returns", 1)
+        streaming = n_returns in ("streaming", "dynamic")
         st = self._get_actor_state(actor_id)
         with st.lock:
             st.seq += 1
             seq = st.seq
         task_id = TaskID.for_actor_task(ActorID(actor_id), seq).binary()
-        refs = [
-            ObjectRef(ObjectID.for_return(TaskID(task_id), i).binary(), self.addr, worker=self)
-            for i in range(n_returns)
-        ]
-        for r in refs:
-            self.store.memory.mark_pending(r.binary())
+        if streaming:
+            n_returns = 0
+            self._streams[task_id] = _StreamState()
+            refs = ObjectRefGenerator(task_id, self)
+        else:
+            refs = [
+                ObjectRef(ObjectID.for_return(TaskID(task_id), i).binary(), self.addr, worker=self)
+                for i in range(n_returns)
+            ]
+            for r in refs:
+                self.store.memory.mark_pending(r.binary())
         sobj = serialization.serialize((args, kwargs))
         self._register_escapes(sobj)
         payload = {
@@ -1039,6 +1171,7 @@ class CoreWorker:
             "method": method_name,
             "args": sobj.to_bytes(),
             "n_returns": n_returns,
+            "streaming": streaming,
             "seq": seq,
             "caller": self.worker_id,
             "caller_addr": list(self.addr),

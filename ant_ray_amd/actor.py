@@ -94,6 +94,8 @@ class ActorHandle:
             )
         else:
             refs = cw.submit_actor_task(self._ray_actor_id, method_name, args, kwargs, opts)
+        if n_returns in ("streaming", "dynamic"):
+            return refs  # ObjectRefGenerator
         if n_returns == 1:
             return refs[0]
         return refs

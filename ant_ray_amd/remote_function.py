@@ -93,6 +93,8 @@ class RemoteFunction:
             refs = cw.executor.submit_task(self._function, args, kwargs, opts)
         else:
             refs = cw.submit_task(self._function, self._fn_id, args, kwargs, opts)
+        if n_returns in ("streaming", "dynamic"):
+            return refs  # ObjectRefGenerator
         if n_returns == 1:
             return refs[0]
         return refs
