@@ -204,6 +204,14 @@ def _collect_deployments(app: Application, out: Dict[str, dict],
     auto = d.autoscaling_config
     if isinstance(auto, AutoscalingConfig):
         auto = auto.__dict__
+    import inspect as _inspect
+
+    t = d._target
+    call = t if (_inspect.isfunction(t) or _inspect.ismethod(t)) \
+        else getattr(t, "__call__", None)
+    is_streaming = bool(call) and (
+        _inspect.isgeneratorfunction(call)
+        or _inspect.isasyncgenfunction(call))
     out[d.name] = {
         "name": d.name,
         "callable_bytes": _dumps(d._target),
@@ -214,6 +222,7 @@ def _collect_deployments(app: Application, out: Dict[str, dict],
         "ray_actor_options": d.ray_actor_options,
         "autoscaling_config": auto,
         "user_config": d.user_config,
+        "is_streaming": is_streaming,
     }
 
 

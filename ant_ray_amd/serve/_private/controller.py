@@ -170,10 +170,16 @@ class ServeController:
                 "deployments": list(self.apps[app])}
 
     async def list_applications(self):
+        def _ing_streaming(app, deps):
+            ing = self.ingress.get(app)
+            dr = deps.get(ing) if ing else None
+            return bool(dr and dr.cfg.get("is_streaming"))
+
         return {
             app: {
                 "route_prefix": self.route_prefixes.get(app),
                 "ingress": self.ingress.get(app),
+                "ingress_streaming": _ing_streaming(app, deps),
                 "deployments": {
                     name: {"replicas": len(dr.replicas), "target": dr.target}
                     for name, dr in deps.items()

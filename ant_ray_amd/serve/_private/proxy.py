@@ -65,6 +65,7 @@ class HTTPProxy:
         apps = await controller.list_applications.remote()
         routes = {}
         handles = {}
+        streaming = {}
         for app, info in apps.items():
             prefix = info.get("route_prefix") or "/"
             ingress = info.get("ingress")
@@ -72,8 +73,10 @@ class HTTPProxy:
                 routes[prefix] = app
                 handles[app] = self._handles.get(app) or DeploymentHandle(
                     ingress, app)
+                streaming[app] = bool(info.get("ingress_streaming"))
         self._routes = routes
         self._handles = handles
+        self._streaming = streaming
         self._last_refresh = now
 
     def _match(self, path: str) -> Optional[str]:
@@ -134,12 +137,61 @@ class HTTPProxy:
             headers=[(k.decode(), v.decode()) for k, v in scope.get("headers", [])],
             body=body, route_prefix=prefix,
         )
+        if getattr(self, "_streaming", {}).get(app):
+            await self._proxy_stream(send, app, req)
+            return
         try:
             result = await self._handles[app].remote(req)
         except Exception as e:
             await _send_simple(send, 500, f"error: {e}".encode())
             return
         await _send_result(send, result)
+
+    async def _proxy_stream(self, send, app: str, req):
+        """Chunked transfer of a generator ingress: each yielded item is
+        flushed to the client as it arrives from the replica."""
+        import ant_ray_amd as ray
+
+        handle = self._handles[app]
+        router = handle._get_router()
+        try:
+            gen, idx = router.submit_stream(None, (req,), {})
+        except Exception as e:
+            await _send_simple(send, 503, f"error: {e}".encode())
+            return
+        started = False
+        try:
+            async for ref in gen:
+                chunk = _encode_chunk(ray.get(ref))
+                if not started:
+                    started = True
+                    await send({"type": "http.response.start", "status": 200,
+                                "headers": [(b"content-type",
+                                             b"text/plain; charset=utf-8"),
+                                            (b"transfer-encoding",
+                                             b"chunked")]})
+                await send({"type": "http.response.body", "body": chunk,
+                            "more_body": True})
+            if not started:
+                await send({"type": "http.response.start", "status": 200,
+                            "headers": [(b"content-type", b"text/plain")]})
+            await send({"type": "http.response.body", "body": b""})
+        except Exception as e:
+            if not started:
+                await _send_simple(send, 500, f"error: {e}".encode())
+            else:
+                # mid-stream failure: terminate the body
+                await send({"type": "http.response.body", "body": b""})
+        finally:
+            router.complete(idx)
+
+
+def _encode_chunk(v) -> bytes:
+    if isinstance(v, (bytes, bytearray)):
+        return bytes(v)
+    if isinstance(v, str):
+        return v.encode()
+    return (json.dumps(v) + "\n").encode()
 
 
 async def _send_simple(send, status: int, body: bytes,

@@ -81,6 +81,40 @@ class Replica:
     def num_ongoing_requests(self) -> int:
         return self._num_ongoing
 
+    def handle_request_streaming(self, method_name: Optional[str], args,
+                                 kwargs):
+        """Streaming entry: the user callable is a (sync or async)
+        generator; each produced chunk rides the runtime's streaming
+        return path (num_returns='streaming' actor method) so the proxy /
+        caller consumes chunks as they are generated (parity: reference
+        Serve response streaming over ObjectRefGenerator)."""
+        self._num_ongoing += 1
+        try:
+            if self._is_function:
+                fn = self._callable
+            else:
+                fn = getattr(self._callable, method_name or "__call__")
+            args = tuple(
+                a.to_starlette() if isinstance(a, HTTPRequestData) else a
+                for a in args
+            )
+            out = fn(*args, **(kwargs or {}))
+            if hasattr(out, "__anext__"):
+                loop = asyncio.new_event_loop()
+                try:
+                    while True:
+                        try:
+                            yield loop.run_until_complete(out.__anext__())
+                        except StopAsyncIteration:
+                            break
+                finally:
+                    loop.close()
+            else:
+                for chunk in out:
+                    yield chunk
+        finally:
+            self._num_ongoing -= 1
+
     async def handle_request(self, method_name: Optional[str], args, kwargs):
         """Entry for handle calls AND HTTP (args[0] is HTTPRequestData)."""
         self._num_ongoing += 1
@@ -119,3 +153,7 @@ class _HTTPResponseData:
         self.status = status
         self.raw_headers = raw_headers
         self.body = body
+
+
+Replica.handle_request_streaming.__ray_method_opts__ = {
+    "num_returns": "streaming"}

@@ -72,5 +72,12 @@ class Router:
         self._ongoing[i] = self._ongoing.get(i, 0) + 1
         return replica.handle_request.remote(method_name, args, kwargs), i
 
+    def submit_stream(self, method_name: Optional[str], args, kwargs):
+        """Streaming variant: returns (ObjectRefGenerator, replica_index)."""
+        i, replica = self.choose_replica()
+        self._ongoing[i] = self._ongoing.get(i, 0) + 1
+        return replica.handle_request_streaming.remote(
+            method_name, args, kwargs), i
+
     def complete(self, i: int):
         self._ongoing[i] = max(0, self._ongoing.get(i, 1) - 1)

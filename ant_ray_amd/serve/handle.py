@@ -54,6 +54,42 @@ class DeploymentResponse:
             pass
 
 
+class DeploymentResponseGenerator:
+    """Iterator over a streaming deployment call's chunk values (parity:
+    reference DeploymentResponseGenerator from handle.options(stream=True);
+    here iteration yields the chunk VALUES directly)."""
+
+    def __init__(self, gen, router, index):
+        self._gen = gen
+        self._router = router
+        self._index = index
+
+    def __iter__(self):
+        return self
+
+    def __next__(self):
+        import ant_ray_amd as ray
+
+        try:
+            return ray.get(next(self._gen))
+        except StopIteration:
+            self._router.complete(self._index)
+            raise
+
+    def __aiter__(self):
+        return self
+
+    async def __anext__(self):
+        import ant_ray_amd as ray
+
+        try:
+            ref = await self._gen.__anext__()
+        except StopAsyncIteration:
+            self._router.complete(self._index)
+            raise
+        return ray.get(ref)
+
+
 class DeploymentHandle:
     def __init__(self, deployment_name: str, app_name: str = "default",
                  method_name: Optional[str] = None):
@@ -67,17 +103,24 @@ class DeploymentHandle:
             self._router = Router(self.app_name, self.deployment_name)
         return self._router
 
-    def options(self, *, method_name: Optional[str] = None, **_):
-        return DeploymentHandle(self.deployment_name, self.app_name,
-                                method_name or self._method_name)
+    def options(self, *, method_name: Optional[str] = None,
+                stream: bool = False, **_):
+        h = DeploymentHandle(self.deployment_name, self.app_name,
+                             method_name or self._method_name)
+        h._stream = stream
+        h._router = self._router
+        return h
 
     def __getattr__(self, name):
         if name.startswith("_"):
             raise AttributeError(name)
         return DeploymentHandle(self.deployment_name, self.app_name, name)
 
-    def remote(self, *args, **kwargs) -> DeploymentResponse:
+    def remote(self, *args, **kwargs):
         router = self._get_router()
+        if getattr(self, "_stream", False):
+            gen, idx = router.submit_stream(self._method_name, args, kwargs)
+            return DeploymentResponseGenerator(gen, router, idx)
         ref, i = router.submit(self._method_name, args, kwargs)
         return DeploymentResponse(ref, router, i)
 

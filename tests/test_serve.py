@@ -236,3 +236,41 @@ def test_serve_cli(serve_mod, tmp_path, monkeypatch):
         assert r.exit_code == 0, r.output
     finally:
         sys.path.remove(str(tmp_path))
+
+
+def test_streaming_response(serve_mod):
+    """A generator ingress streams chunks: via handle.options(stream=True)
+    AND over HTTP with chunked transfer — first chunk arrives while the
+    replica is still producing."""
+    import time as _t
+    import urllib.request
+
+    import ant_ray_amd as ray
+
+    serve = serve_mod
+
+    @serve.deployment
+    def streamer(request=None):
+        for i in range(4):
+            yield f"chunk{i}|"
+            _t.sleep(0.3)
+
+    serve.run(streamer.bind(), name="streamapp", route_prefix="/stream")
+
+    # handle-level streaming
+    h = serve.get_app_handle("streamapp")
+    chunks = list(h.options(stream=True).remote())
+    assert chunks == [f"chunk{i}|" for i in range(4)]
+
+    # HTTP chunked: read incrementally, first chunk must arrive early
+    port = ray.get(ray.get_actor("SERVE_PROXY_ACTOR").ready.remote())
+    t0 = _t.time()
+    with urllib.request.urlopen(f"http://127.0.0.1:{port}/stream",
+                                timeout=30) as r:
+        first = r.read(7)
+        first_latency = _t.time() - t0
+        rest = r.read()
+    assert first == b"chunk0|"
+    assert rest == b"chunk1|chunk2|chunk3|"
+    assert first_latency < 1.0, f"first chunk took {first_latency:.2f}s"
+    serve.delete("streamapp")
