@@ -238,8 +238,18 @@ def kill(actor: ActorHandle, *, no_restart: bool = True):
 
 
 def cancel(object_ref: ObjectRef, *, force: bool = False, recursive: bool = True):
-    # Round-1: best effort no-op (task cancellation protocol in round 2).
+    """Cancel the task that produces object_ref (parity: ray.cancel /
+    CoreWorker::CancelTask). Queued tasks are dropped; running tasks get
+    KeyboardInterrupt at their next Python bytecode boundary (force=False
+    — a task blocked in one long C call needs force) or their worker
+    killed outright (force=True);
+    ray.get on the ref then raises TaskCancelledError. recursive is
+    accepted for API parity (children are not chased yet)."""
     _check_connected()
+    cw = global_worker.core_worker
+    if cw.mode == LOCAL_MODE:
+        return  # local-mode tasks run synchronously at submit
+    cw.cancel_task(object_ref, force=force)
 
 
 def get_actor(name: str, namespace: Optional[str] = None) -> ActorHandle:

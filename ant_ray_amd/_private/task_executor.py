@@ -42,6 +42,8 @@ class TaskExecutor:
         self.actor_id: Optional[bytes] = None
         self.max_concurrency = 1
         self._serial_q: "queue.Queue" = queue.Queue()
+        self._cancelled: set = set()        # task_ids cancelled pre-start
+        self._running_serial: bytes = None  # task running on the serial thread
         self._serial_thread = threading.Thread(target=self._serial_loop, daemon=True)
         self._serial_thread.start()
         self._pool: Optional[concurrent.futures.ThreadPoolExecutor] = None
@@ -116,7 +118,33 @@ class TaskExecutor:
             payload, done = self._serial_q.get()
             if payload is None:
                 return
-            self._run_and_reply(payload, done)
+            self._running_serial = payload.get("task_id")
+            try:
+                self._run_and_reply(payload, done)
+            finally:
+                self._running_serial = None
+
+    def cancel_task(self, task_id: bytes, force: bool):
+        """Cancel a queued or running task in this worker (parity:
+        CoreWorkerService CancelTask). Queued -> dropped before start;
+        running on the serial thread -> KeyboardInterrupt raised into it
+        via PyThreadState_SetAsyncExc (the reference sends SIGINT to the
+        worker's main thread); force -> the whole worker exits (reference
+        force kill), the owner maps the dead push to TaskCancelledError."""
+        self._cancelled.add(task_id)
+        if force:
+            import threading as _t
+
+            _t.Timer(0.05, lambda: os._exit(1)).start()  # let the RPC reply
+            return {"cancelled": "force-killing worker"}
+        if self._running_serial == task_id:
+            import ctypes
+
+            ctypes.pythonapi.PyThreadState_SetAsyncExc(
+                ctypes.c_long(self._serial_thread.ident),
+                ctypes.py_object(KeyboardInterrupt))
+            return {"cancelled": "interrupted running task"}
+        return {"cancelled": "queued"}
 
     def request_exit(self):
         self._exit_requested = True
@@ -232,6 +260,11 @@ class TaskExecutor:
     def _execute(self, payload) -> dict:
         ttype = payload["type"]
         task_id = payload["task_id"]
+        if task_id in self._cancelled:
+            from ant_ray_amd.exceptions import TaskCancelledError
+
+            return self._error_reply(payload, TaskCancelledError(
+                "task was cancelled before it started"))
         self.cw.current_task_id = task_id
         _t0 = self._last_t0 = time.time()
         try:
@@ -280,6 +313,12 @@ class TaskExecutor:
             raise ValueError(f"unknown task type {ttype}")
         except BaseException as e:  # noqa: BLE001
             self._record_event(payload, _t0, time.time(), ok=False)
+            if (isinstance(e, KeyboardInterrupt)
+                    and task_id in self._cancelled):
+                from ant_ray_amd.exceptions import TaskCancelledError
+
+                return self._error_reply(payload, TaskCancelledError(
+                    "task was cancelled while running"))
             return self._error_reply(payload, e)
         finally:
             self.cw.current_task_id = None

@@ -199,6 +199,9 @@ class CoreWorker:
         self._spilled: Dict[bytes, str] = {}  # oid -> spill file path
         self._pull_cache: Dict[bytes, tuple] = {}  # big-object pull slices
         self._streams: Dict[bytes, "_StreamState"] = {}  # streaming tasks
+        self._task_of_oid: Dict[bytes, tuple] = {}  # oid -> (task_id, key)
+        self._inflight_tasks: Dict[bytes, "LeasedWorker"] = {}
+        self._cancelled_tasks: set = set()
         self._killed_workers: Dict[bytes, str] = {}  # worker_id -> kill reason
         self._object_locations: Dict[bytes, Tuple[str, int]] = {}
         self._actors: Dict[bytes, ActorHandleState] = {}
@@ -423,6 +426,11 @@ class CoreWorker:
 
     async def rpc_ping(self, conn, p):
         return {"ok": True, "worker_id": self.worker_id}
+
+    async def rpc_cancel_task(self, conn, p):
+        if self.executor is None:
+            return {"cancelled": False}
+        return self.executor.cancel_task(p["task_id"], p.get("force", False))
 
     async def rpc_stream_item(self, conn, p):
         """A streaming task produced its next yielded value (executor
@@ -911,6 +919,11 @@ class CoreWorker:
         if opts.get("num_gpus"):
             resources["GPU"] = float(opts["num_gpus"])
         key = self._scheduling_key(fn_id, {"resources": resources})
+        if not streaming:
+            for r in refs:
+                self._task_of_oid[r.binary()] = (task_id, key)
+            while len(self._task_of_oid) > 20000:  # bound the cancel index
+                self._task_of_oid.pop(next(iter(self._task_of_oid)))
         self.io.submit(self._enqueue_task(key, payload, resources, opts))  # fire-and-forget: refs are pre-created, failures land on them
         return refs
 
@@ -1003,12 +1016,21 @@ class CoreWorker:
             if fn_id is not None and fn_id not in lw.sent_fns:
                 payload = dict(payload, fn=self._fn_cache.get(fn_id))
                 lw.sent_fns.add(fn_id)
+            self._inflight_tasks[payload["task_id"]] = lw
             reply = await lw.conn.call("push_task", payload, timeout=None)
             if reply.get("status") == "need_fn":
                 payload = dict(payload, fn=self._fn_cache.get(fn_id))
                 reply = await lw.conn.call("push_task", payload, timeout=None)
             self._handle_task_reply(payload, reply)
         except Exception as e:  # worker died mid-task
+            if payload["task_id"] in self._cancelled_tasks:
+                # force-cancel killed the worker: this is the cancellation
+                # outcome, not a failure to retry
+                from ant_ray_amd.exceptions import TaskCancelledError
+
+                self._fail_task(payload, TaskCancelledError(
+                    "task was force-cancelled (ray.cancel(force=True))"))
+                return
             killed_reason = self._killed_workers.pop(lw.worker_id, None)
             if killed_reason is not None:
                 e = RaySystemError(killed_reason)
@@ -1026,9 +1048,45 @@ class CoreWorker:
                 self._fail_task(payload, RaySystemError(f"task push failed: {e}"))
             return
         finally:
+            self._inflight_tasks.pop(payload["task_id"], None)
             lw.in_flight -= 1
             lw.idle_since = time.monotonic()
         await self._pump_tasks(key)
+
+    def cancel_task(self, ref: ObjectRef, force: bool = False) -> bool:
+        """ray.cancel (parity: CoreWorker::CancelTask): a queued task is
+        dropped before execution; a running task gets KeyboardInterrupt
+        raised in its thread (force=False) or its worker process killed
+        (force=True). Returns True if a cancellation was delivered."""
+        from ant_ray_amd.exceptions import TaskCancelledError
+
+        rec = self._task_of_oid.get(ref.binary())
+        if rec is None:
+            return False
+        task_id, key = rec
+        self._cancelled_tasks.add(task_id)
+        while len(self._cancelled_tasks) > 10000:
+            self._cancelled_tasks.pop()
+
+        async def do():
+            q = self._lease_queue.get(key) or []
+            for item in list(q):
+                if item["payload"]["task_id"] == task_id:
+                    q.remove(item)
+                    self._fail_task(item["payload"], TaskCancelledError(
+                        "task was cancelled before it started"))
+                    return True
+            lw = self._inflight_tasks.get(task_id)
+            if lw is not None:
+                try:
+                    await lw.conn.call("cancel_task", {
+                        "task_id": task_id, "force": force}, timeout=10)
+                    return True
+                except Exception:
+                    return False
+            return False
+
+        return bool(self.io.run(do(), timeout=30))
 
     def _fail_task(self, payload, exc: BaseException):
         task_id = payload["task_id"]
