@@ -312,6 +312,35 @@ class TaskExecutor:
                 return self._reply_results(payload, result)
             raise ValueError(f"unknown task type {ttype}")
         except BaseException as e:  # noqa: BLE001
+            from ant_ray_amd.actor import ActorExit
+
+            if isinstance(e, ActorExit):
+                # graceful self-exit (ray.actor.exit_actor): ack this call,
+                # mark the death intentional (no restart), then leave once
+                # the reply has drained
+                self._record_event(payload, _t0, time.time(), ok=True)
+                import threading as _t
+
+                def _graceful_exit():
+                    # after the reply has drained: mark the death
+                    # intentional (no restart), then exit
+                    if self.actor_id:
+                        try:
+                            self.cw.io.submit(self.cw.gcs.call(
+                                "kill_actor",
+                                {"actor_id": self.actor_id,
+                                 "no_restart": True})).result(5)
+                        except Exception:
+                            pass
+                    self.request_exit()
+
+                _t.Timer(0.2, _graceful_exit).start()
+                return {"status": "ok", "results": [
+                    {"oid": ObjectID.for_return(
+                        TaskID(task_id), i).binary(),
+                     "inline": serialization.serialize(None).to_bytes(),
+                     "meta": serialization.META_PICKLE}
+                    for i in range(payload.get("n_returns", 1))]}
             self._record_event(payload, _t0, time.time(), ok=False)
             if (isinstance(e, KeyboardInterrupt)
                     and task_id in self._cancelled):

@@ -23,6 +23,24 @@ _VALID_ACTOR_OPTIONS = {
 }
 
 
+class ActorExit(SystemExit):
+    """Raised by exit_actor(); the executor exits the worker gracefully
+    after replying to the in-flight call."""
+
+
+def exit_actor():
+    """Terminate the current actor from inside one of its methods
+    (parity: ray.actor.exit_actor) — pending queued calls fail with
+    the actor-died error, the reply for THIS call still reaches the
+    caller."""
+    from ant_ray_amd._private.worker import global_worker
+
+    cw = global_worker.core_worker
+    if cw is None or cw.executor is None or cw.executor.actor_instance is None:
+        raise RuntimeError("exit_actor() called outside an actor method")
+    raise ActorExit(0)
+
+
 def method(**kwargs):
     """@ray.method decorator (num_returns, concurrency_group...)."""
 

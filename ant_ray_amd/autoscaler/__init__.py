@@ -125,6 +125,22 @@ class StandardAutoscaler:
         for pg in _gcs_call("list_placement_groups"):
             if pg.get("state") == "PENDING":
                 demands.extend(pg.get("bundles") or [])
+        # standing requests from autoscaler.sdk.request_resources: only
+        # the part NOT already satisfiable by free resources adds demand
+        try:
+            from ant_ray_amd.autoscaler.sdk import get_requested_resources
+
+            import ant_ray_amd as ray
+
+            avail = dict(ray.available_resources())
+            for b in get_requested_resources():
+                if all(avail.get(k, 0) >= v for k, v in b.items()):
+                    for k, v in b.items():
+                        avail[k] = avail.get(k, 0) - v
+                else:
+                    demands.append(dict(b))
+        except Exception:
+            pass
         return demands
 
     def update(self):

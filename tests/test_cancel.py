@@ -77,3 +77,54 @@ def test_cancel_force_kills_worker(ray_mod):
     ray.cancel(ref, force=True)
     with pytest.raises(TaskCancelledError):
         ray.get(ref, timeout=60)
+
+
+def test_exit_actor(ray_mod):
+    """ray.actor.exit_actor(): the call completes, the actor then dies;
+    later calls raise the actor-died error (no restart: graceful exits
+    don't count as failures)."""
+    import ant_ray_amd as ray_
+
+    ray = ray_mod
+
+    @ray.remote(max_restarts=3)
+    class Quitter:
+        def ping(self):
+            return "pong"
+
+        def quit(self):
+            ray_.actor.exit_actor()
+
+    q = Quitter.remote()
+    assert ray.get(q.ping.remote(), timeout=60) == "pong"
+    ray.get(q.quit.remote(), timeout=60)  # reply arrives, then actor exits
+    time.sleep(1.0)
+    with pytest.raises(Exception):
+        ray.get(q.ping.remote(), timeout=30)
+
+
+def test_request_resources_sdk(ray_mod):
+    """autoscaler.sdk.request_resources: standing demand visible to the
+    autoscaler beyond free capacity."""
+    from ant_ray_amd.autoscaler import NodeTypeConfig, StandardAutoscaler
+    from ant_ray_amd.autoscaler.sdk import (
+        get_requested_resources,
+        request_resources,
+    )
+
+    class NullProvider:
+        def non_terminated_nodes(self):
+            return {}
+
+        def create_node(self, cfg):
+            pass
+
+    request_resources(bundles=[{"CPU": 64}])  # far beyond the 2-CPU session
+    assert get_requested_resources() == [{"CPU": 64}]
+    scaler = StandardAutoscaler(
+        {"big": NodeTypeConfig("big", {"CPU": 64}, max_workers=2)},
+        NullProvider())
+    demands = scaler.pending_demands()
+    assert {"CPU": 64} in demands
+    request_resources()  # clear
+    assert get_requested_resources() == []
