@@ -258,6 +258,18 @@ def job_status(job_id):
     click.echo(JobSubmissionClient().get_job_status(job_id))
 
 
+@job.command("stop")
+@click.argument("job_id")
+def job_stop(job_id):
+    import ant_ray_amd as ray
+    from ant_ray_amd.job_submission import JobSubmissionClient
+
+    ray.init(address=os.environ.get("RAY_ADDRESS", "auto"),
+             ignore_reinit_error=True)
+    ok = JobSubmissionClient().stop_job(job_id)
+    click.echo(f"stopped {job_id}" if ok else f"could not stop {job_id}")
+
+
 @job.command("logs")
 @click.argument("job_id")
 def job_logs(job_id):
