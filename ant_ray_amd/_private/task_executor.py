@@ -102,13 +102,18 @@ class TaskExecutor:
                 self._maybe_exit()
 
         method = payload.get("method", "")
+        group = payload.get("concurrency_group")
         if (
             payload.get("type") == "actor_task"
             and self.actor_instance is not None
             and self._is_async_method(method)
         ):
             self._run_async_actor_task(payload, done)
-        elif payload.get("type") == "actor_task" and self.max_concurrency > 1:
+        elif (payload.get("type") == "actor_task" and group
+                and group in getattr(self, "_groups", {})):
+            self._groups[group].submit(self._run_and_reply, payload, done)
+        elif payload.get("type") == "actor_task" and (
+                self.max_concurrency > 1 and self._pool is not None):
             self._pool.submit(self._run_and_reply, payload, done)
         else:
             self._serial_q.put((payload, done))
@@ -285,6 +290,24 @@ class TaskExecutor:
                     self._pool = concurrent.futures.ThreadPoolExecutor(
                         max_workers=self.max_concurrency
                     )
+                # named concurrency groups (parity: reference
+                # concurrency_group_manager.cc — per-group executors so a
+                # saturated group never blocks another's methods)
+                self._groups = {}
+                self._group_sems = {}
+                for gname, gmax in (payload.get("concurrency_groups")
+                                    or {}).items():
+                    self._groups[gname] = concurrent.futures.ThreadPoolExecutor(
+                        max_workers=int(gmax),
+                        thread_name_prefix=f"cg-{gname}")
+                    self._group_sem_sizes = getattr(
+                        self, "_group_sem_sizes", {})
+                    self._group_sem_sizes[gname] = int(gmax)
+                if self._groups and self._pool is None:
+                    # default group for un-annotated methods
+                    self._pool = concurrent.futures.ThreadPoolExecutor(
+                        max_workers=max(self.max_concurrency, 1))
+                    self.max_concurrency = max(self.max_concurrency, 2)
                 self.actor_id = payload.get("actor_id")
                 self.cw.actor_id = self.actor_id
                 self.actor_instance = cls(*args, **kwargs)
@@ -410,10 +433,21 @@ class TaskExecutor:
             return iters
 
     def _run_async_actor_task(self, payload, done):
+        group = payload.get("concurrency_group")
+
         async def runner():
             if self._async_sem is None:
                 self._async_sem = asyncio.Semaphore(self._async_sem_size)
-            async with self._async_sem:
+            sem = self._async_sem
+            sizes = getattr(self, "_group_sem_sizes", {})
+            if group in sizes:
+                sems = getattr(self, "_async_group_sems", None)
+                if sems is None:
+                    sems = self._async_group_sems = {}
+                if group not in sems:
+                    sems[group] = asyncio.Semaphore(sizes[group])
+                sem = sems[group]
+            async with sem:
                 try:
                     method = getattr(self.actor_instance, payload["method"])
                     args, kwargs = self._deserialize_args(payload)

@@ -1157,6 +1157,7 @@ class CoreWorker:
             "caller": self.worker_id,
             "caller_addr": list(self.addr),
             "max_concurrency": opts.get("max_concurrency", 1),
+            "concurrency_groups": opts.get("concurrency_groups"),
             "n_returns": 0,
         }
         st = ActorHandleState(actor_id)
