@@ -328,7 +328,16 @@ class GcsServer:
             node.last_heartbeat = time.monotonic()
             if "resources_available" in p:
                 node.resources_available = p["resources_available"]
+            node.pending_demands = p.get("pending_demands") or []
         return {"ok": True}
+
+    async def rpc_pending_resource_demands(self, conn, p):
+        """Queued lease demand across nodes (autoscaler input)."""
+        out = []
+        for node in self.nodes.values():
+            if node.alive:
+                out.extend(getattr(node, "pending_demands", []))
+        return out
 
     async def rpc_node_table(self, conn, p):
         return [n.view() for n in self.nodes.values()]

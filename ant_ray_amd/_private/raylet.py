@@ -188,7 +188,16 @@ class Raylet:
             try:
                 await self.gcs_conn.call(
                     "heartbeat",
-                    {"node_id": self.node_id, "resources_available": self.resources_available},
+                    {"node_id": self.node_id,
+                     "resources_available": self.resources_available,
+                     # queued lease demand (parity: reference resource_load
+                     # in the syncer messages — what the autoscaler scales
+                     # for beyond pending actors/PGs)
+                     "pending_demands": [
+                         dict(p.get("resources") or {})
+                         for (p, _conn), _fut in self.pending_leases[:64]
+                     ] + [r for t, r in getattr(self, "_infeasible", [])
+                          if time.monotonic() - t < 10.0]},
                     timeout=5,
                 )
             except Exception:
@@ -490,6 +499,12 @@ class Raylet:
                 pass
             return reply
         if not pg and not self._feasible(res):
+            # remember the demand so the autoscaler sees it (reference
+            # reports infeasible resource load through the syncer)
+            now = time.monotonic()
+            self._infeasible = [(t, r) for t, r in
+                                getattr(self, "_infeasible", [])
+                                if now - t < 10.0][-63:] + [(now, dict(res))]
             reply = {"granted": False, "infeasible": True}
             try:
                 # ask the GCS for a node whose TOTAL resources fit -> the

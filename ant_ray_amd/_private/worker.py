@@ -977,8 +977,27 @@ class CoreWorker:
             return
         self._lease_requests[key] -= 1
         if not reply.get("granted"):
+            # cluster-infeasible NOW: keep the tasks queued and retry —
+            # the autoscaler may add a node (reference keeps infeasible
+            # tasks pending and warns; ANTRAY_SCHED_TIMEOUT_S bounds it)
+            window = float(os.environ.get("ANTRAY_SCHED_TIMEOUT_S", "3600"))
+            first = sample_item.setdefault("_infeasible_since",
+                                           time.monotonic())
+            if time.monotonic() - first < window:
+                if time.monotonic() - first < 3.0:  # warn once, early
+                    logger.warning(
+                        "no node can currently satisfy resources %s; task(s)"
+                        " stay queued (autoscaler may provision)",
+                        sample_item["resources"])
+                await protocol.asyncio.sleep(2.0)
+                if self._lease_queue[key]:
+                    self._lease_requests[key] += 1
+                    self.io.loop.create_task(
+                        self._request_lease(key, sample_item))
+                return
             err = RaySystemError(
-                f"no feasible node for resources {sample_item['resources']}"
+                f"no feasible node for resources {sample_item['resources']} "
+                f"after {window:.0f}s"
             )
             for item in self._lease_queue[key]:
                 self._fail_task(item["payload"], err)

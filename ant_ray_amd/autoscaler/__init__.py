@@ -125,6 +125,11 @@ class StandardAutoscaler:
         for pg in _gcs_call("list_placement_groups"):
             if pg.get("state") == "PENDING":
                 demands.extend(pg.get("bundles") or [])
+        # queued TASK leases reported by raylets via heartbeat
+        try:
+            demands.extend(_gcs_call("pending_resource_demands"))
+        except Exception:
+            pass
         # standing requests from autoscaler.sdk.request_resources: only
         # the part NOT already satisfiable by free resources adds demand
         try:
@@ -156,24 +161,38 @@ class StandardAutoscaler:
             for _ in range(n):
                 logger.info("autoscaler: launching node type %s", t)
                 self.provider.create_node(self.node_types[t])
-        # idle downscale: nodes beyond min with zero used resources
-        if not demands:
-            import ant_ray_amd as ray
+        # idle downscale: after idle_timeout_s with no pending demand AND
+        # fully-unused worker capacity, terminate one launched node per
+        # tick (LIFO via the provider) until back at min_workers
+        now = time.time()
+        if demands or to_launch:
+            self._no_demand_since = None
+        else:
+            if getattr(self, "_no_demand_since", None) is None:
+                self._no_demand_since = now
+            idle_for = now - self._no_demand_since
+            if idle_for >= self.idle_timeout_s:
+                import ant_ray_amd as ray
 
-            now = time.time()
-            for node in ray.nodes():
-                if not node.get("Alive"):
-                    continue
-                used = {
-                    k: node["Resources"].get(k, 0) - node["Available"].get(k, 0)
-                    for k in node["Resources"]
-                }
-                busy = any(v > 1e-9 for k, v in used.items()
-                           if k not in ("memory", "object_store_memory")
-                           and not k.startswith("node:"))
-                nid = node["NodeID"]
-                if busy:
-                    self._idle_since.pop(nid, None)
-                else:
-                    self._idle_since.setdefault(nid, now)
+                idle_nodes = 0
+                for node in ray.nodes():
+                    if not node.get("Alive"):
+                        continue
+                    used = {
+                        k: node["Resources"].get(k, 0)
+                        - node["Available"].get(k, 0)
+                        for k in node["Resources"]
+                    }
+                    busy = any(v > 1e-9 for k, v in used.items()
+                               if k not in ("memory", "object_store_memory")
+                               and not k.startswith("node:"))
+                    if not busy:
+                        idle_nodes += 1
+                if idle_nodes:
+                    for t, cfg in self.node_types.items():
+                        if existing.get(t, 0) > cfg.min_workers:
+                            logger.info(
+                                "autoscaler: terminating one idle %s node", t)
+                            self.provider.terminate_node(t)
+                            break
         return to_launch

@@ -222,3 +222,48 @@ def test_cluster_lifecycle_events():
     assert ("ACTOR", "DEAD") in kinds
     assert all("timestamp" in e and "entity_id" in e for e in evs)
     ray.shutdown()
+
+
+def test_autoscaler_idle_downscale():
+    """Nodes launched for demand are terminated again once the cluster has
+    been idle past idle_timeout_s (down to min_workers)."""
+    import ant_ray_amd as ray
+    from ant_ray_amd.autoscaler import (
+        LocalNodeProvider,
+        NodeTypeConfig,
+        StandardAutoscaler,
+    )
+    from ant_ray_amd.cluster_utils import Cluster
+
+    if ray.is_initialized():
+        ray.shutdown()
+    c = Cluster(initialize_head=True, head_node_args={"num_cpus": 1})
+    try:
+        c.connect()
+        provider = LocalNodeProvider(c)
+        scaler = StandardAutoscaler(
+            {"worker": NodeTypeConfig("worker", {"CPU": 2}, max_workers=3)},
+            provider, idle_timeout_s=1.0)
+
+        @ray.remote(num_cpus=2)
+        def burst():
+            return "done"
+
+        ref = burst.remote()
+        deadline = time.time() + 60
+        while (provider.non_terminated_nodes().get("worker", 0) < 1
+               and time.time() < deadline):
+            time.sleep(0.5)
+            scaler.update()
+        assert provider.non_terminated_nodes().get("worker", 0) >= 1
+        assert ray.get(ref, timeout=60) == "done"
+
+        # idle: after the timeout, the worker nodes drain away
+        deadline = time.time() + 60
+        while (provider.non_terminated_nodes().get("worker", 0) > 0
+               and time.time() < deadline):
+            time.sleep(0.5)
+            scaler.update()
+        assert provider.non_terminated_nodes().get("worker", 0) == 0
+    finally:
+        c.shutdown()
