@@ -309,3 +309,76 @@ def test_elastic_regrows_when_resources_return(ray_mod, tmp_path_factory):
     assert result.metrics["phase"] == "resumed"
     assert result.metrics["world"] == 4, result.metrics
     assert _time.time() - t0 < 90
+
+
+def test_transformers_integration(ray_mod, tmp_path_factory):
+    """HF Trainer inside TorchTrainer with RayTrainReportCallback: logs and
+    checkpoints flow into ray.train.report (reference
+    train/huggingface/transformers parity)."""
+    import pytest
+
+    pytest.importorskip("transformers")
+    storage = str(tmp_path_factory.mktemp("hf"))
+
+    def train_fn(config):
+        import torch as _torch
+        from torch.utils.data import Dataset as TorchDataset
+
+        from transformers import Trainer, TrainingArguments
+        from transformers.modeling_utils import PreTrainedModel
+        from transformers.configuration_utils import PretrainedConfig
+
+        from ant_ray_amd import train
+        from ant_ray_amd.train.huggingface import prepare_trainer
+
+        class TinyConfig(PretrainedConfig):
+            model_type = "tiny_test"
+
+        class TinyModel(PreTrainedModel):
+            config_class = TinyConfig
+
+            def __init__(self, config):
+                super().__init__(config)
+                self.lin = _torch.nn.Linear(4, 2)
+
+            def forward(self, x=None, labels=None, **kw):
+                logits = self.lin(x)
+                loss = _torch.nn.functional.cross_entropy(logits, labels)
+                return {"loss": loss, "logits": logits}
+
+        class Ds(TorchDataset):
+            def __len__(self):
+                return 32
+
+            def __getitem__(self, i):
+                return {"x": _torch.randn(4),
+                        "labels": _torch.tensor(i % 2)}
+
+        ctx = train.get_context()
+        args = TrainingArguments(
+            output_dir=f"/tmp/hf_out_{ctx.world_rank}",
+            per_device_train_batch_size=8,
+            num_train_epochs=1,
+            logging_steps=1,
+            save_steps=2,
+            save_strategy="steps",
+            report_to=[],
+            use_cpu=True,
+            disable_tqdm=True,
+        )
+        trainer = Trainer(model=TinyModel(TinyConfig()), args=args,
+                          train_dataset=Ds())
+        trainer = prepare_trainer(trainer)
+        trainer.train()
+
+    from ant_ray_amd.train import RunConfig, ScalingConfig
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    result = TorchTrainer(
+        train_fn,
+        scaling_config=ScalingConfig(num_workers=1),
+        run_config=RunConfig(name="hf", storage_path=storage),
+    ).fit()
+    assert result.error is None, result.error
+    assert result.metrics and result.metrics.get("step", 0) >= 2
+    assert result.checkpoint is not None  # HF checkpoint dir persisted
