@@ -65,6 +65,40 @@ def run(import_path, name, route_prefix, host, port, blocking):
 
 
 @cli.command()
+@click.argument("config_file")
+def deploy(config_file):
+    """Deploy applications from a YAML config (parity: serve deploy).
+
+    Schema (reference multi-app config, reduced):
+        applications:
+          - name: app1
+            route_prefix: /app1
+            import_path: my_module:app
+    """
+    import yaml
+
+    import ant_ray_amd as ray
+    from ant_ray_amd import serve
+
+    with open(config_file) as f:
+        cfg = yaml.safe_load(f) or {}
+    apps = cfg.get("applications") or []
+    if not apps:
+        raise click.ClickException("config has no applications")
+    if not ray.is_initialized():
+        addr = os.environ.get("RAY_ADDRESS")
+        ray.init(address=addr) if addr else ray.init()
+    http = cfg.get("http_options") or {}
+    serve.start(http_options=http)
+    for app in apps:
+        target = _load_target(app["import_path"])
+        serve.run(target, name=app.get("name", "default"),
+                  route_prefix=app.get("route_prefix", "/"))
+        click.echo(f"deployed {app.get('name', 'default')} at "
+                   f"{app.get('route_prefix', '/')}")
+
+
+@cli.command()
 def status():
     """Show application/deployment status."""
     import ant_ray_amd as ray

@@ -274,3 +274,53 @@ def test_streaming_response(serve_mod):
     assert rest == b"chunk1|chunk2|chunk3|"
     assert first_latency < 1.0, f"first chunk took {first_latency:.2f}s"
     serve.delete("streamapp")
+
+
+def test_serve_deploy_yaml(serve_mod, tmp_path, monkeypatch):
+    """serve deploy <config.yaml>: multi-app declarative deploy."""
+    from click.testing import CliRunner
+
+    from ant_ray_amd.serve.scripts import cli
+
+    serve = serve_mod
+    for app_name in list(serve.status().get("applications", {})):
+        serve.delete(app_name)
+
+    (tmp_path / "yaml_apps.py").write_text(
+        "from ant_ray_amd import serve\n"
+        "@serve.deployment\n"
+        "def alpha(request=None):\n"
+        "    return 'from-alpha'\n"
+        "@serve.deployment\n"
+        "def beta(request=None):\n"
+        "    return 'from-beta'\n"
+        "app_a = alpha.bind()\n"
+        "app_b = beta.bind()\n"
+    )
+    (tmp_path / "serve_config.yaml").write_text(
+        "applications:\n"
+        "  - name: appa\n"
+        "    route_prefix: /a\n"
+        "    import_path: yaml_apps:app_a\n"
+        "  - name: appb\n"
+        "    route_prefix: /b\n"
+        "    import_path: yaml_apps:app_b\n"
+    )
+    monkeypatch.chdir(tmp_path)
+    import sys
+
+    sys.path.insert(0, str(tmp_path))
+    try:
+        r = CliRunner().invoke(cli, ["deploy", "serve_config.yaml"])
+        assert r.exit_code == 0, r.output
+        import urllib.request
+
+        a = urllib.request.urlopen("http://127.0.0.1:8000/a",
+                                   timeout=30).read().decode()
+        b = urllib.request.urlopen("http://127.0.0.1:8000/b",
+                                   timeout=30).read().decode()
+        assert a == "from-alpha" and b == "from-beta"
+        serve.delete("appa")
+        serve.delete("appb")
+    finally:
+        sys.path.remove(str(tmp_path))
