@@ -383,10 +383,26 @@ class GcsServer:
         return {"job_id": job_id}
 
     async def rpc_pick_raylet(self, conn, p):
-        """Name a feasible raylet for a task lease that is infeasible on the
-        requester's node (spillback target)."""
-        node = self._pick_node(dict(p.get("resources") or {}),
-                               label_selector=p.get("_label_selector"))
+        """Name a feasible raylet for a task lease that is infeasible on
+        the requester's node (spillback target), or — when
+        preferred_worker names a worker holding the task's biggest arg —
+        the raylet co-located with that data (locality-aware lease
+        policy, parity: core_worker lease_policy.cc LocalityAwareLeasePolicy)."""
+        res = dict(p.get("resources") or {})
+        pref = p.get("preferred_worker")
+        if pref is not None:
+            pref = tuple(pref)
+            for w in self.workers.values():
+                if w.get("addr") and tuple(w["addr"]) == pref:
+                    node = self.nodes.get(w.get("node_id"))
+                    if (node is not None and node.alive
+                            and node.conn is not None and not node.conn.closed
+                            and all(node.resources_total.get(k, 0) >= v
+                                    for k, v in res.items())):
+                        return {"addr": list(node.addr),
+                                "node_id": node.node_id, "locality": True}
+                    break
+        node = self._pick_node(res, label_selector=p.get("_label_selector"))
         if node is None:
             return {"addr": None}
         return {"addr": list(node.addr), "node_id": node.node_id}

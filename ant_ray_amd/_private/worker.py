@@ -924,6 +924,20 @@ class CoreWorker:
                 self._task_of_oid[r.binary()] = (task_id, key)
             while len(self._task_of_oid) > 20000:  # bound the cancel index
                 self._task_of_oid.pop(next(iter(self._task_of_oid)))
+        # locality hint: an arg ref held by a worker on ANOTHER node pulls
+        # the lease request toward that node (reference locality-aware
+        # LeasePolicy); first remote-held ref wins
+        for cref in getattr(sobj, "contained_refs", []):
+            coid = cref.binary()
+            if self.store.memory.get_now(coid)[0] or (
+                    self.store.shm is not None
+                    and self.store.shm.contains(coid)):
+                continue  # value already on this node: no pull to avoid
+            loc = (self._object_locations.get(coid)
+                   or (tuple(cref.owner_addr) if cref.owner_addr else None))
+            if loc and tuple(loc) != self.addr:
+                payload["_locality"] = list(loc)
+                break
         self.io.submit(self._enqueue_task(key, payload, resources, opts))  # fire-and-forget: refs are pre-created, failures land on them
         return refs
 
@@ -959,6 +973,21 @@ class CoreWorker:
             "_label_selector": sample_item["opts"].get("_label_selector"),
         }
         granting_raylet = self.raylet
+        locality = sample_item["payload"].get("_locality")
+        if locality is not None and not req["pg"]:
+            # data-locality first: lease on the node holding the task's
+            # remote arg if the GCS confirms it fits the resources
+            try:
+                r = await self.gcs.call("pick_raylet", {
+                    "resources": req["resources"],
+                    "preferred_worker": locality,
+                    "_label_selector": req["_label_selector"]}, timeout=5)
+                if (r.get("locality") and r.get("addr")
+                        and tuple(r["addr"]) != tuple(self.raylet_addr)):
+                    granting_raylet = await self._get_worker_conn_async_cached(
+                        tuple(r["addr"]))
+            except Exception:
+                pass
         try:
             reply = await granting_raylet.call("lease_worker", req, timeout=None)
             # spillback (parity: raylet local_task_manager spillback — the

@@ -218,3 +218,44 @@ def test_node_label_scheduling(cluster):
         except Exception:
             time.sleep(1.0)  # worker churn from the previous cluster teardown
     assert tid == nid, "task must spill back to the labeled node"
+
+
+def test_locality_aware_leasing(cluster):
+    """A task whose (big) argument lives on another node leases THERE
+    (reference locality-aware LeasePolicy) instead of pulling the bytes
+    to a random node."""
+    import numpy as np
+
+    import ant_ray_amd as ray
+
+    cluster.connect()
+    cluster.add_node(num_cpus=2, resources={"datanode": 1})
+    deadline = time.time() + 30
+    while time.time() < deadline and not ray.cluster_resources().get("datanode"):
+        time.sleep(0.2)
+
+    @ray.remote(num_cpus=1, resources={"datanode": 0.1})
+    def produce():
+        import ant_ray_amd as ray2
+
+        data = np.ones(2 * 1024 * 1024, dtype=np.float64)  # 16 MB
+        # inner put: the bytes stay OWNED by this datanode worker; only
+        # the tiny ref wrapper travels to the driver
+        return [ray2.put(data)]
+
+    inner = ray.get(produce.remote(), timeout=60)[0]
+
+    @ray.remote(num_cpus=1)
+    def consume_where(x):
+        from ant_ray_amd._private.worker import global_worker
+
+        cw = global_worker.core_worker
+        return float(x.sum()), cw.node_id.hex()
+
+    # no resource pin: the locality hint must route the lease to the
+    # node owning the 16 MB argument
+    total, consumer_node = ray.get(consume_where.remote(inner), timeout=120)
+    assert total == 2 * 1024 * 1024
+    rows = {n["NodeID"]: n["Resources"] for n in ray.nodes()}
+    assert "datanode" in rows.get(consumer_node, {}), (
+        "consumer should be scheduled on the node holding the argument")
