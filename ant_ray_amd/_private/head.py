@@ -34,6 +34,24 @@ async def run_head(args):
     os.makedirs(os.path.join(session_dir, "logs"), exist_ok=True)
 
     gcs = GcsServer()
+    if getattr(args, "owner_pid", 0):
+        async def _watch_owner():
+            # the driver that started this head is gone (crash/SIGKILL)
+            # -> tear the whole local cluster down instead of leaking
+            # orphaned head+worker processes
+            while True:
+                await asyncio.sleep(2.0)
+                try:
+                    os.kill(args.owner_pid, 0)
+                except ProcessLookupError:
+                    logging.getLogger("antray.head").warning(
+                        "owner pid %d gone; shutting down session",
+                        args.owner_pid)
+                    os._exit(0)  # raylet workers fate-share via conn close
+                except PermissionError:
+                    pass  # pid exists under another uid: still alive
+
+        asyncio.get_running_loop().create_task(_watch_owner())
     gcs_port = await gcs.start(
         args.host, args.port,
         persist_path=os.path.join(session_dir, "gcs_tables.msgpack"))
@@ -84,6 +102,8 @@ def main():
     ap.add_argument("--object-store-memory", type=int, default=0)
     ap.add_argument("--session-dir", default="")
     ap.add_argument("--prestart", type=int, default=0)
+    ap.add_argument("--owner-pid", type=int, default=0,
+                    help="fate-share: exit when this pid dies (0=detached)")
     args = ap.parse_args()
     if not args.session_dir:
         args.session_dir = os.path.join(

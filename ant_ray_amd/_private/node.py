@@ -40,6 +40,7 @@ def start_head(
     session_dir: str = "",
     prestart: int = 0,  # 0 = one per CPU
     timeout: float = 60.0,
+    owner_pid: Optional[int] = -1,  # -1 = this process; 0/None = detached
 ) -> HeadProcess:
     if not session_dir:
         session_dir = os.path.join("/tmp/antray", f"session_{int(time.time()*1000)}_{os.getpid()}")
@@ -51,6 +52,14 @@ def start_head(
         "--session-dir", session_dir,
         "--prestart", str(prestart),
     ]
+    if owner_pid == -1:
+        owner_pid = os.getpid()
+    if owner_pid:
+        # a head started by ray.init fate-shares with its driver: if the
+        # driver dies without ray.shutdown (SIGKILL, crash), the cluster
+        # must not linger as orphaned processes. `ray start --head`
+        # passes owner_pid=0 and stays detached.
+        cmd += ["--owner-pid", str(owner_pid)]
     if num_cpus is not None:
         cmd += ["--num-cpus", str(num_cpus)]
     if num_gpus is not None:

@@ -38,7 +38,8 @@ def start(head, address, num_cpus, num_gpus, port, object_store_memory, block):
         from ant_ray_amd._private.node import start_head
 
         hp = start_head(num_cpus=num_cpus, num_gpus=num_gpus, port=port,
-                        object_store_memory=object_store_memory)
+                        object_store_memory=object_store_memory,
+                        owner_pid=0)  # ray start --head: detached
         info = hp.info
         click.echo(f"Started head: GCS at {info['gcs_addr']}")
         click.echo(f"session dir: {info['session_dir']}")

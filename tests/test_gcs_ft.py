@@ -141,3 +141,57 @@ def test_gcs_restart_with_persistence(tmp_path):
         for proc in (gcs, raylet):
             if proc is not None and proc.poll() is None:
                 proc.kill()
+
+
+def test_head_fate_shares_with_driver(tmp_path):
+    """A head started by ray.init must die when its driver is SIGKILLed
+    (no orphaned head+worker process trees); `ray start --head` stays
+    detached (owner_pid=0)."""
+    import textwrap
+
+    script = textwrap.dedent("""
+        import os, sys, time
+        import ant_ray_amd as ray
+
+        ray.init(num_cpus=2)
+        from ant_ray_amd._private.worker import global_worker
+
+        # print the head pid for the parent test, then hang until killed
+        import subprocess
+        print("SESSION", global_worker.core_worker.session_dir, flush=True)
+        sys.stdout.flush()
+        time.sleep(300)
+    """)
+    proc = subprocess.Popen([sys.executable, "-c", script],
+                            stdout=subprocess.PIPE, text=True)
+    try:
+        line = proc.stdout.readline()
+        assert line.startswith("SESSION"), line
+        session_dir = line.split(" ", 1)[1].strip()
+
+        # find the head process serving that session
+        out = subprocess.run(["ps", "-ww", "-eo", "pid,args"], capture_output=True,
+                             text=True).stdout
+        head_pid = None
+        for row in out.splitlines():
+            if "_private.head" in row and session_dir in row:
+                head_pid = int(row.split()[0])
+                break
+        assert head_pid is not None, f"no head for {session_dir}"
+
+        proc.kill()  # SIGKILL the driver: no ray.shutdown runs
+        proc.wait(timeout=10)
+
+        deadline = time.time() + 30
+        alive = True
+        while time.time() < deadline:
+            try:
+                os.kill(head_pid, 0)
+            except ProcessLookupError:
+                alive = False
+                break
+            time.sleep(0.5)
+        assert not alive, "orphaned head survived its driver's death"
+    finally:
+        if proc.poll() is None:
+            proc.kill()
