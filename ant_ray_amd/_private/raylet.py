@@ -596,7 +596,13 @@ class Raylet:
                     w = None
                     continue
             else:
-                w = self._spawn_worker(renv if needs_dedicated else None)
+                # Popen + log-file open cost ~5-15ms of syscalls: off the
+                # event loop, or a 200-wide actor burst serializes its
+                # spawns behind the loop and starves the registration
+                # RPCs those same spawns are waiting on
+                w = await asyncio.get_running_loop().run_in_executor(
+                    None, self._spawn_worker,
+                    renv if needs_dedicated else None)
             try:
                 await asyncio.wait_for(w.registered.wait(), timeout=60)
             except asyncio.TimeoutError:
