@@ -111,7 +111,16 @@ class ObjectStore:
         primary copy pinned while local refs exist — parity with the
         reference's primary-copy pinning; unpinned objects are LRU-evictable
         the moment readers drop them)."""
-        off = self.shm.create_object(oid, sobj.total_size, sobj.metadata)
+        try:
+            off = self.shm.create_object(oid, sobj.total_size, sobj.metadata)
+        except ValueError:
+            # a re-executed task (lineage reconstruction / retry) rewrites
+            # its own return object: drop the stale copy and recreate
+            try:
+                self.shm.delete(oid)
+            except Exception:
+                pass
+            off = self.shm.create_object(oid, sobj.total_size, sobj.metadata)
         try:
             mv = self.shm.view_at(off, sobj.total_size, writable=True)
             sobj.write_into(mv)

@@ -200,6 +200,11 @@ class CoreWorker:
         self._pull_cache: Dict[bytes, tuple] = {}  # big-object pull slices
         self._streams: Dict[bytes, "_StreamState"] = {}  # streaming tasks
         self._task_of_oid: Dict[bytes, tuple] = {}  # oid -> (task_id, key)
+        # lineage: task return oid -> (key, payload, resources, opts) for
+        # reconstruction of lost objects (parity: reference TaskManager
+        # lineage + ObjectRecoveryManager, bounded by bytes)
+        self._lineage: Dict[bytes, tuple] = {}
+        self._lineage_bytes = 0
         self._inflight_tasks: Dict[bytes, "LeasedWorker"] = {}
         self._cancelled_tasks: set = set()
         self._killed_workers: Dict[bytes, str] = {}  # worker_id -> kill reason
@@ -711,6 +716,46 @@ class CoreWorker:
             if spilled is not None:
                 data, meta = spilled
                 return self._deserialize_buffer(memoryview(data), meta)
+            # lineage reconstruction (reference object_recovery_manager
+            # .h:41): a task output whose holder is unreachable is
+            # recomputed by resubmitting the producing task — the return
+            # ids are deterministic, so the recomputed value lands under
+            # this same oid
+            if (attempt >= 3 and oid in self._lineage
+                    and not self.store.memory.is_pending(oid)):
+                lkey, lpayload, lres, lopts = self._lineage[oid]
+                retries_left = lpayload.get("max_retries", 3)
+                if retries_left > 0:
+                    lpayload = dict(lpayload, max_retries=retries_left - 1)
+                    self._lineage[oid] = (lkey, lpayload, lres, lopts)
+                    logger.warning(
+                        "object %s lost (holder unreachable); resubmitting "
+                        "its producing task (lineage reconstruction)",
+                        oid.hex()[:12])
+                    for i in range(lpayload.get("n_returns", 1)):
+                        roid = ObjectID.for_return(
+                            TaskID(lpayload["task_id"]), i).binary()
+                        # clear the stale IN_PLASMA marker or mark_pending
+                        # no-ops and the wait below returns immediately
+                        self.store.memory.delete(roid)
+                        self.store.memory.mark_pending(roid)
+                    self._object_locations.pop(oid, None)
+                    pull_addr = None
+                    attempt = 0
+                    self.io.submit(
+                        self._enqueue_task(lkey, lpayload, lres, lopts))
+                    ok = self.store.memory.wait(
+                        oid, None if deadline is None
+                        else max(0.0, deadline - time.monotonic()))
+                    if ok:
+                        found, v = self.store.memory.get_now(oid)
+                        if found and v is not IN_PLASMA:
+                            value = self._resolve_memory_entry(v)
+                            if isinstance(value, RayTaskError):
+                                raise value.as_instanceof_cause()
+                            return value
+                        pull_addr = self._object_locations.get(oid)
+                    continue
             no_remote = pull_addr is None or tuple(pull_addr) == self.addr
             if (attempt >= 3 and no_remote
                     and not self.store.memory.is_pending(oid)
@@ -924,6 +969,15 @@ class CoreWorker:
                 self._task_of_oid[r.binary()] = (task_id, key)
             while len(self._task_of_oid) > 20000:  # bound the cancel index
                 self._task_of_oid.pop(next(iter(self._task_of_oid)))
+        if not streaming:
+            cost = len(payload.get("args") or b"") + 512
+            for r in refs:
+                self._lineage[r.binary()] = (key, payload, resources, opts)
+            self._lineage_bytes += cost
+            while self._lineage_bytes > 256 * 1024 * 1024 and self._lineage:
+                old_oid = next(iter(self._lineage))
+                _, old_p, _, _ = self._lineage.pop(old_oid)
+                self._lineage_bytes -= len(old_p.get("args") or b"") + 512
         # locality hint: an arg ref held by a worker on ANOTHER node pulls
         # the lease request toward that node (reference locality-aware
         # LeasePolicy); first remote-held ref wins
