@@ -896,12 +896,15 @@ class CoreWorker:
                     if len(ready) >= num_returns:
                         pending.extend(candidates[i:])
                         return
-                    v = objs.get(ref.binary())
-                    ok = (v is not None and v is not IN_PLASMA
-                          and not isinstance(v, _MemPending))
-                    if not ok and shm is not None and (
-                            v is None or v is IN_PLASMA):
-                        ok = shm.contains(ref.binary())
+                    oid = ref.binary()
+                    v = objs.get(oid)
+                    # IN_PLASMA = the owner saw the task complete; that IS
+                    # ready (reference semantics: wait readiness means the
+                    # object exists, wherever it lives — fetch happens at
+                    # ray.get). Only a truly unknown oid consults shm.
+                    ok = v is not None and not isinstance(v, _MemPending)
+                    if not ok and shm is not None and v is None:
+                        ok = shm.contains(oid)
                     (ready if ok else pending).append(ref)
 
         ready: List[ObjectRef] = []

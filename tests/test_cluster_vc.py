@@ -259,3 +259,26 @@ def test_locality_aware_leasing(cluster):
     rows = {n["NodeID"]: n["Resources"] for n in ray.nodes()}
     assert "datanode" in rows.get(consumer_node, {}), (
         "consumer should be scheduled on the node holding the argument")
+
+
+def test_wait_remote_held_object(cluster):
+    """ray.wait must report a remotely-held (un-fetched) object as ready:
+    readiness means 'exists', not 'local' (reference semantics)."""
+    import numpy as np
+
+    import ant_ray_amd as ray
+
+    cluster.connect()
+    cluster.add_node(num_cpus=2, resources={"remote": 1})
+    deadline = time.time() + 30
+    while time.time() < deadline and not ray.cluster_resources().get("remote"):
+        time.sleep(0.2)
+
+    @ray.remote(num_cpus=1, resources={"remote": 0.1})
+    def make_big():
+        return np.zeros(1024 * 1024)  # 8 MB: stays on the remote node
+
+    ref = make_big.remote()
+    ready, not_ready = ray.wait([ref], timeout=60)
+    assert ready == [ref] and not_ready == []
+    assert float(ray.get(ref, timeout=60).sum()) == 0.0
