@@ -137,6 +137,14 @@ class TaskExecutor:
         worker's main thread); force -> the whole worker exits (reference
         force kill), the owner maps the dead push to TaskCancelledError."""
         self._cancelled.add(task_id)
+        t = getattr(self, "_async_tasks", {}).get(task_id)
+        if t is not None and not force:
+            # running async actor method: cancel its coroutine
+            try:
+                t.cancel()
+            except Exception:
+                pass
+            return {"cancelled": "async task cancelled"}
         if force:
             import threading as _t
 
@@ -434,6 +442,7 @@ class TaskExecutor:
 
     def _run_async_actor_task(self, payload, done):
         group = payload.get("concurrency_group")
+        task_id = payload.get("task_id")
 
         async def runner():
             if self._async_sem is None:
@@ -449,15 +458,29 @@ class TaskExecutor:
                 sem = sems[group]
             async with sem:
                 try:
+                    if task_id in self._cancelled:
+                        raise asyncio.CancelledError
                     method = getattr(self.actor_instance, payload["method"])
                     args, kwargs = self._deserialize_args(payload)
                     result = await method(*args, **kwargs)
                     reply = self._reply_results(payload, result)
+                except asyncio.CancelledError:
+                    from ant_ray_amd.exceptions import TaskCancelledError
+
+                    reply = self._error_reply(payload, TaskCancelledError(
+                        "actor task was cancelled"))
                 except BaseException as e:  # noqa: BLE001
                     reply = self._error_reply(payload, e)
+                finally:
+                    if not hasattr(self, "_async_tasks"):
+                        self._async_tasks = {}
+                    self._async_tasks.pop(task_id, None)
                 done(reply)
 
-        self._actor_loop.submit(runner())
+        t = self._actor_loop.submit(runner())
+        if not hasattr(self, "_async_tasks"):
+            self._async_tasks = {}
+        self._async_tasks[task_id] = t
 
     # ---------------------------------------------------------------- replies
     def _store_one_return(self, payload, i: int, v) -> dict:

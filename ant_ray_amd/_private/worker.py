@@ -1168,6 +1168,8 @@ class CoreWorker:
         rec = self._task_of_oid.get(ref.binary())
         if rec is None:
             return False
+        if rec[0] == "actor":
+            return self._cancel_actor_task(rec[1], rec[2])
         task_id, key = rec
         self._cancelled_tasks.add(task_id)
         while len(self._cancelled_tasks) > 10000:
@@ -1186,6 +1188,38 @@ class CoreWorker:
                 try:
                     await lw.conn.call("cancel_task", {
                         "task_id": task_id, "force": force}, timeout=10)
+                    return True
+                except Exception:
+                    return False
+            return False
+
+        return bool(self.io.run(do(), timeout=30))
+
+    def _cancel_actor_task(self, actor_id: bytes, task_id: bytes) -> bool:
+        """Cancel an actor task: dropped if still queued owner-side,
+        else the actor's worker cancels it (queued there, or a running
+        ASYNC method gets its asyncio task cancelled — a running sync
+        actor method is not interruptible, matching the reference)."""
+        from ant_ray_amd.exceptions import TaskCancelledError
+
+        st = self._actors.get(actor_id)
+        if st is None:
+            return False
+
+        async def do():
+            with st.lock:
+                for pl in list(st.pending):
+                    if pl["task_id"] == task_id:
+                        st.pending.remove(pl)
+                        self._fail_task(pl, TaskCancelledError(
+                            "actor task cancelled before it was sent"))
+                        return True
+            if st.addr is not None:
+                try:
+                    conn = await self._get_worker_conn_async_cached(
+                        tuple(st.addr))
+                    await conn.call("cancel_task", {
+                        "task_id": task_id, "force": False}, timeout=10)
                     return True
                 except Exception:
                     return False
@@ -1328,6 +1362,9 @@ class CoreWorker:
                 self.store.memory.mark_pending(r.binary())
         sobj = serialization.serialize((args, kwargs))
         self._register_escapes(sobj)
+        if not streaming:
+            for r in refs:
+                self._task_of_oid[r.binary()] = ("actor", actor_id, task_id)
         payload = {
             "type": "actor_task",
             "task_id": task_id,

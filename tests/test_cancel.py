@@ -159,3 +159,32 @@ def test_concurrency_groups(ray_mod):
     assert fast == [f"fetched-{i}" for i in range(4)]
     assert io_latency < 1.5, f"io group blocked behind compute ({io_latency:.2f}s)"
     assert ray.get(slow, timeout=60) == "crunched"
+
+
+def test_cancel_async_actor_task(ray_mod):
+    """ray.cancel on a RUNNING async actor method cancels its coroutine;
+    queued calls behind it cancel before starting."""
+    ray = ray_mod
+    from ant_ray_amd.exceptions import TaskCancelledError
+
+    @ray.remote
+    class AsyncWorker:
+        async def long(self):
+            import asyncio
+
+            await asyncio.sleep(300)
+            return "never"
+
+        async def quick(self):
+            return "quick"
+
+    a = AsyncWorker.remote()
+    ref = a.long.remote()
+    time.sleep(1.0)
+    ray.cancel(ref)
+    t0 = time.time()
+    with pytest.raises(TaskCancelledError):
+        ray.get(ref, timeout=60)
+    assert time.time() - t0 < 30
+    # the actor stays healthy for later calls
+    assert ray.get(a.quick.remote(), timeout=60) == "quick"
