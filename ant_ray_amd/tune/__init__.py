@@ -171,6 +171,54 @@ class Tuner:
         self.param_space = param_space or {}
         self.tune_config = tune_config or TuneConfig()
         self.run_config = run_config or RunConfig()
+        self._restore_state: Optional[dict] = None
+
+    @classmethod
+    def restore(cls, path: str, trainable: Callable,
+                resume_errored: bool = False,
+                restart_errored: bool = False) -> "Tuner":
+        """Resume an interrupted experiment from its directory (parity:
+        Tuner.restore — finished trials keep their results, unfinished
+        ones re-run from their latest checkpoint; errored trials re-run
+        from checkpoint with resume_errored=True or from scratch with
+        restart_errored=True)."""
+        import pickle
+
+        state_file = os.path.join(path, "tuner_state.pkl")
+        with open(state_file, "rb") as f:
+            state = pickle.load(f)
+        t = cls(trainable, tune_config=state["tune_config"],
+                run_config=state["run_config"])
+        t.run_config.name = os.path.basename(path.rstrip("/"))
+        t.run_config.storage_path = os.path.dirname(path.rstrip("/"))
+        state["resume_errored"] = resume_errored
+        state["restart_errored"] = restart_errored
+        t._restore_state = state
+        return t
+
+    def _save_state(self, exp_path, variants, results, running):
+        import pickle
+
+        trials = []
+        for idx, cfg in enumerate(variants):
+            row = {"idx": idx, "config": cfg}
+            if idx in results:
+                r = results[idx]
+                row["status"] = "errored" if r.error else "finished"
+                row["metrics"] = r.metrics
+                row["ckpt"] = r.checkpoint.path if r.checkpoint else None
+            else:
+                t = running.get(idx)
+                row["status"] = "running" if t else "pending"
+                row["ckpt"] = t["ckpt"] if t else None
+            trials.append(row)
+        tmp = os.path.join(exp_path, ".tuner_state.tmp")
+        with open(tmp, "wb") as f:
+            pickle.dump({"tune_config": self.tune_config,
+                         "run_config": self.run_config,
+                         "variants": variants,
+                         "trials": trials}, f)
+        os.replace(tmp, os.path.join(exp_path, "tuner_state.pkl"))
 
     def fit(self) -> ResultGrid:
         import ant_ray_amd as ray
@@ -178,8 +226,11 @@ class Tuner:
         if not ray.is_initialized():
             ray.init()
         tc = self.tune_config
-        variants = BasicVariantGenerator(
-            self.param_space, tc.num_samples, seed=tc.seed).variants()
+        if self._restore_state is not None:
+            variants = self._restore_state["variants"]
+        else:
+            variants = BasicVariantGenerator(
+                self.param_space, tc.num_samples, seed=tc.seed).variants()
         scheduler = tc.scheduler or FIFOScheduler()
         scheduler.set_objective(tc.metric, tc.mode or "max")
         name = self.run_config.name or f"tune_{int(time.time())}"
@@ -196,8 +247,33 @@ class Tuner:
         pending = list(enumerate(variants))
         running: Dict[int, dict] = {}
         results: Dict[int, Result] = {}
+        restore_ckpts: Dict[int, str] = {}
+        if self._restore_state is not None:
+            pending = []
+            resume_err = self._restore_state.get("resume_errored")
+            restart_err = self._restore_state.get("restart_errored")
+            for row in self._restore_state["trials"]:
+                idx = row["idx"]
+                if row["status"] == "errored" and (resume_err or restart_err):
+                    pending.append((idx, row["config"]))
+                    if resume_err and row.get("ckpt"):
+                        restore_ckpts[idx] = row["ckpt"]
+                elif row["status"] in ("finished", "errored"):
+                    results[idx] = Result(
+                        metrics=row.get("metrics"),
+                        checkpoint=Checkpoint(row["ckpt"]) if row.get("ckpt")
+                        else None,
+                        path=os.path.join(exp_path, f"{name}_{idx:05d}"),
+                        error=None if row["status"] == "finished"
+                        else RuntimeError("errored before restore"),
+                    )
+                else:
+                    pending.append((idx, row["config"]))
+                    if row.get("ckpt"):
+                        restore_ckpts[idx] = row["ckpt"]
 
         def launch(idx, config, restore=None, prev=None):
+            restore = restore or restore_ckpts.pop(idx, None)
             trial_id = f"{name}_{idx:05d}"
             trial_path = os.path.join(exp_path, trial_id)
             os.makedirs(trial_path, exist_ok=True)
@@ -212,6 +288,10 @@ class Tuner:
             if hasattr(scheduler, "on_trial_start"):
                 scheduler.on_trial_start(trial_id, config)
 
+        try:
+            self._save_state(exp_path, variants, results, running)
+        except Exception:
+            pass
         while pending or running:
             while pending and len(running) < max_conc:
                 idx, cfg = pending.pop(0)
@@ -266,6 +346,11 @@ class Tuner:
                     finished_idx.append(idx)
             for idx in finished_idx:
                 del running[idx]
+            if finished_idx or not running:
+                try:
+                    self._save_state(exp_path, variants, results, running)
+                except Exception:
+                    pass
             if running:
                 time.sleep(0.1)
         return ResultGrid([results[i] for i in sorted(results)])

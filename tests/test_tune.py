@@ -172,3 +172,58 @@ def test_pbt_exploits_and_perturbs(ray_mod, tmp_path_factory):
     # final score is far above what 20 iterations of lr=0.001 (0.02) give
     assert scores[0] > 1.0, scores
     assert grid.get_best_result(metric="score", mode="max").metrics["score"] >= 19.0
+
+
+def test_tuner_restore(ray_mod, tmp_path_factory):
+    """Tuner.restore resumes an interrupted experiment: finished trials
+    keep results, unfinished ones re-run from their checkpoints."""
+    import json
+    import os
+
+    from ant_ray_amd import tune
+    from ant_ray_amd.train import Checkpoint, RunConfig
+    from ant_ray_amd.tune import TuneConfig, Tuner
+
+    storage = str(tmp_path_factory.mktemp("resume"))
+    poison = os.path.join(storage, "poison")
+    open(poison, "w").close()  # first run: trial with x>=2 dies
+
+    def trainable(config):
+        import time as _t
+
+        from ant_ray_amd import train
+
+        start = 0
+        ck = train.get_checkpoint()
+        if ck is not None:
+            with open(os.path.join(ck.path, "s.json")) as f:
+                start = json.load(f)["it"] + 1
+        for it in range(start, 4):
+            _t.sleep(0.1)
+            if config["x"] >= 2 and os.path.exists(poison):
+                raise RuntimeError("interrupted")
+            import tempfile
+
+            with tempfile.TemporaryDirectory() as d:
+                with open(os.path.join(d, "s.json"), "w") as f:
+                    json.dump({"it": it}, f)
+                train.report({"it": it, "x": config["x"],
+                              "resumed_from": start},
+                             checkpoint=Checkpoint.from_directory(d))
+
+    tuner = Tuner(trainable,
+                  param_space={"x": tune.grid_search([1, 2])},
+                  tune_config=TuneConfig(metric="it", mode="max"),
+                  run_config=RunConfig(name="resume_exp",
+                                       storage_path=storage))
+    grid = tuner.fit()
+    assert len(grid.errors) == 1  # the poisoned trial failed
+
+    os.unlink(poison)
+    restored = Tuner.restore(os.path.join(storage, "resume_exp"),
+                             trainable, resume_errored=True)
+    grid2 = restored.fit()
+    assert not grid2.errors
+    by_x = {r.metrics["x"]: r.metrics for r in grid2}
+    assert by_x[1]["it"] == 3          # finished trial carried over
+    assert by_x[2]["it"] == 3          # interrupted trial completed
