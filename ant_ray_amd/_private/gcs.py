@@ -389,6 +389,13 @@ class GcsServer:
         the raylet co-located with that data (locality-aware lease
         policy, parity: core_worker lease_policy.cc LocalityAwareLeasePolicy)."""
         res = dict(p.get("resources") or {})
+        if p.get("spread"):
+            node = self._pick_node(res, label_selector=p.get("_label_selector"),
+                                   spread=True)
+            if node is None:
+                return {"addr": None}
+            return {"addr": list(node.addr), "node_id": node.node_id,
+                    "locality": True}
         pref = p.get("preferred_worker")
         if pref is not None:
             pref = tuple(pref)
@@ -593,7 +600,8 @@ class GcsServer:
     def _pick_node(self, resources: Dict[str, float], pg: Optional[dict] = None,
                    node_affinity: Optional[bytes] = None,
                    allowed: Optional[set] = None,
-                   label_selector: Optional[dict] = None) -> Optional[NodeInfo]:
+                   label_selector: Optional[dict] = None,
+                   spread: bool = False) -> Optional[NodeInfo]:
         if pg:
             pg_info = self.pgs.get(pg["pg_id"])
             if pg_info and pg_info.state == "CREATED":
@@ -638,6 +646,17 @@ class GcsServer:
 
                 if _labels_match(label_selector["soft"], node.labels):
                     score += 10.0  # prefer soft-matching nodes
+            if spread:
+                # SPREAD strategy: round-robin across feasible nodes
+                # (reference spread_scheduling_policy) — rotate by a
+                # global counter so consecutive picks alternate nodes
+                self._spread_seq = getattr(self, "_spread_seq", 0) + 1
+                import hashlib as _h
+
+                rot = int(_h.blake2b(
+                    node.node_id + self._spread_seq.to_bytes(4, "little"),
+                    digest_size=2).hexdigest(), 16)
+                score += rot / 65536.0
             if score > best_score:
                 best, best_score = node, score
         return best
@@ -663,7 +682,8 @@ class GcsServer:
                 return
             node = self._pick_node(resources, pg, opts.get("_node_affinity"),
                                    self._vc_allowed(vc_id),
-                                   opts.get("_label_selector"))
+                                   opts.get("_label_selector"),
+                                   spread=bool(opts.get("_spread")))
             if node is not None and node.conn is not None and not node.conn.closed:
                 # optimistic accounting: concurrent schedulings must not all
                 # pile onto the same node while its heartbeat is stale (the

@@ -1030,6 +1030,16 @@ class CoreWorker:
             "_label_selector": sample_item["opts"].get("_label_selector"),
         }
         granting_raylet = self.raylet
+        if sample_item["opts"].get("_spread") and not req["pg"]:
+            try:
+                r = await self.gcs.call("pick_raylet", {
+                    "resources": req["resources"], "spread": True,
+                    "_label_selector": req["_label_selector"]}, timeout=5)
+                if r.get("addr") and tuple(r["addr"]) != tuple(self.raylet_addr):
+                    granting_raylet = await self._get_worker_conn_async_cached(
+                        tuple(r["addr"]))
+            except Exception:
+                pass
         locality = sample_item["payload"].get("_locality")
         if locality is not None and not req["pg"]:
             # data-locality first: lease on the node holding the task's

@@ -20,6 +20,10 @@ _VALID_OPTIONS = {
 def _normalize_opts(opts: Dict[str, Any]) -> Dict[str, Any]:
     out = dict(opts)
     strategy = out.pop("scheduling_strategy", None)
+    if isinstance(strategy, str):
+        if strategy == "SPREAD":
+            out["_spread"] = True
+        # "DEFAULT" falls through
     if strategy is not None and not isinstance(strategy, str):
         from ant_ray_amd.util.scheduling_strategies import (
             NodeAffinitySchedulingStrategy,

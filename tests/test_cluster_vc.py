@@ -282,3 +282,26 @@ def test_wait_remote_held_object(cluster):
     ready, not_ready = ray.wait([ref], timeout=60)
     assert ready == [ref] and not_ready == []
     assert float(ray.get(ref, timeout=60).sum()) == 0.0
+
+
+def test_spread_scheduling_strategy(cluster):
+    """scheduling_strategy="SPREAD": actors fan out across feasible nodes
+    instead of packing the first one."""
+    import ant_ray_amd as ray
+
+    cluster.connect()
+    cluster.add_node(num_cpus=4)
+    cluster.add_node(num_cpus=4)
+    deadline = time.time() + 30
+    while time.time() < deadline and len(
+            [n for n in ray.nodes() if n["Alive"]]) < 3:
+        time.sleep(0.2)
+
+    @ray.remote(num_cpus=0.1, scheduling_strategy="SPREAD")
+    class S:
+        def node(self):
+            return ray.get_runtime_context().get_node_id()
+
+    actors = [S.remote() for _ in range(9)]
+    nodes = set(ray.get([a.node.remote() for a in actors], timeout=120))
+    assert len(nodes) >= 2, f"SPREAD placed everything on {nodes}"
