@@ -99,6 +99,22 @@ class TrainController:
         self.ckpt_manager = CheckpointManager(run_config.checkpoint_config)
         self.resume_from_checkpoint = resume_from_checkpoint
         self.latest_metrics: Dict[int, Dict[str, Any]] = {}
+        self.callbacks = list(run_config.callbacks or [])
+
+    def _fire(self, hook: str, *a, **kw):
+        """Invoke a user callback hook if defined (parity: reference
+        train/v2 UserCallback — on_report / on_checkpoint /
+        on_worker_group_start / on_worker_group_shutdown /
+        on_failure)."""
+        for cb in self.callbacks:
+            fn = getattr(cb, hook, None)
+            if fn is None:
+                continue
+            try:
+                fn(*a, **kw)
+            except Exception:
+                logger.exception("user callback %s.%s failed",
+                                 type(cb).__name__, hook)
 
     def _split_datasets(self, n: int):
         """Per-worker dataset shards: ant_ray_amd.data datasets stream-split;
@@ -163,6 +179,7 @@ class TrainController:
                                 self.experiment_path)
             try:
                 group.start()
+                self._fire("on_worker_group_start", num_workers=n)
                 restore = None
                 if self.ckpt_manager.latest:
                     restore = self.ckpt_manager.latest
@@ -181,6 +198,7 @@ class TrainController:
                 failed = e
             finally:
                 group.shutdown()
+                self._fire("on_worker_group_shutdown")
             if failed is None:
                 error = None
                 break
@@ -218,11 +236,17 @@ class TrainController:
             for st in statuses:
                 for rep in st["reports"]:
                     self.latest_metrics[rep["rank"]] = rep["metrics"]
+                    self._fire("on_report", metrics=rep["metrics"],
+                               rank=rep["rank"])
                     if rep["checkpoint_path"]:
                         self.ckpt_manager.register(rep["checkpoint_path"],
                                                    rep["metrics"])
+                        self._fire("on_checkpoint",
+                                   checkpoint_path=rep["checkpoint_path"],
+                                   metrics=rep["metrics"], rank=rep["rank"])
             errs = [st["error"] for st in statuses if st["status"] == "errored"]
             if errs:
+                self._fire("on_failure", error=errs[0])
                 # drain the other ranks' queued reports before tearing the
                 # group down: a checkpoint reported just before a peer's
                 # failure must be registered, or the restart resumes from

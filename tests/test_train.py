@@ -382,3 +382,50 @@ def test_transformers_integration(ray_mod, tmp_path_factory):
     assert result.error is None, result.error
     assert result.metrics and result.metrics.get("step", 0) >= 2
     assert result.checkpoint is not None  # HF checkpoint dir persisted
+
+
+def test_user_callbacks(ray_mod, tmp_path_factory):
+    """RunConfig(callbacks=...): controller fires on_report/on_checkpoint/
+    worker-group lifecycle hooks (reference Train v2 UserCallback)."""
+    from ant_ray_amd.train import Checkpoint, RunConfig, ScalingConfig
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    events = []
+
+    class CB:
+        def on_worker_group_start(self, num_workers):
+            events.append(("start", num_workers))
+
+        def on_report(self, metrics, rank):
+            events.append(("report", metrics["step"], rank))
+
+        def on_checkpoint(self, checkpoint_path, metrics, rank):
+            events.append(("ckpt", metrics["step"]))
+
+        def on_worker_group_shutdown(self):
+            events.append(("shutdown",))
+
+    def train_fn(config):
+        import tempfile
+
+        from ant_ray_amd import train
+
+        for step in range(2):
+            if train.get_context().get_world_rank() == 0:
+                with tempfile.TemporaryDirectory() as d:
+                    train.report({"step": step},
+                                 checkpoint=Checkpoint.from_directory(d))
+            else:
+                train.report({"step": step})
+
+    storage = str(tmp_path_factory.mktemp("cb"))
+    res = TorchTrainer(
+        train_fn, scaling_config=ScalingConfig(num_workers=2),
+        run_config=RunConfig(name="cb", storage_path=storage,
+                             callbacks=[CB()]),
+    ).fit()
+    assert res.error is None
+    kinds = [e[0] for e in events]
+    assert kinds[0] == "start" and kinds[-1] == "shutdown"
+    assert kinds.count("ckpt") == 2
+    assert ("report", 1, 0) in events
