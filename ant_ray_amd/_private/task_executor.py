@@ -287,7 +287,16 @@ class TaskExecutor:
                     return {"status": "need_fn"}
                 args, kwargs = self._deserialize_args(payload)
                 result = fn(*args, **kwargs)
-                return self._reply_results(payload, result)
+                reply = self._reply_results(payload, result)
+                mc = int(payload.get("max_calls") or 0)
+                if mc:
+                    self._normal_calls = getattr(self, "_normal_calls", 0) + 1
+                    if self._normal_calls >= mc:
+                        reply["recycle"] = True
+                        import threading as _t
+
+                        _t.Timer(0.2, self.request_exit).start()
+                return reply
             if ttype == "actor_create":
                 import cloudpickle
 

@@ -961,6 +961,7 @@ class CoreWorker:
             "caller_addr": list(self.addr),
             "name": opts.get("name", ""),
             "max_retries": 0 if streaming else opts.get("max_retries", 3),
+            "max_calls": opts.get("max_calls", 0),
         }
         resources = dict(opts.get("resources") or {})
         resources.setdefault("CPU", float(opts.get("num_cpus", 1)))
@@ -1137,6 +1138,14 @@ class CoreWorker:
                 payload = dict(payload, fn=self._fn_cache.get(fn_id))
                 reply = await lw.conn.call("push_task", payload, timeout=None)
             self._handle_task_reply(payload, reply)
+            if reply.get("recycle"):
+                # max_calls reached: the worker exits once idle — stop
+                # pushing to it (parity: reference worker recycling for
+                # leaky native libs)
+                try:
+                    self._leases[key].remove(lw)
+                except ValueError:
+                    pass
         except Exception as e:  # worker died mid-task
             if payload["task_id"] in self._cancelled_tasks:
                 # force-cancel killed the worker: this is the cancellation

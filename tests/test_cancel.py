@@ -188,3 +188,20 @@ def test_cancel_async_actor_task(ray_mod):
     assert time.time() - t0 < 30
     # the actor stays healthy for later calls
     assert ray.get(a.quick.remote(), timeout=60) == "quick"
+
+
+def test_max_calls_recycles_worker(ray_mod):
+    """@ray.remote(max_calls=N): the worker process exits after N calls
+    (reference worker recycling for leaky native libs) and later tasks
+    run in a fresh process."""
+    ray = ray_mod
+
+    @ray.remote(num_cpus=0.1, max_calls=2)
+    def pid():
+        import os
+
+        return os.getpid()
+
+    pids = [ray.get(pid.remote(), timeout=60) for _ in range(6)]
+    # 6 calls at max_calls=2 -> at least 3 distinct processes
+    assert len(set(pids)) >= 3, pids
