@@ -360,6 +360,21 @@ class RuntimeContext:
         return "{}"
 
 
+def timeline(filename: Optional[str] = None):
+    """Chrome-trace of recorded task events (parity: ray.timeline);
+    returns the event list, and writes JSON when filename is given."""
+    _check_connected()
+    from ant_ray_amd.util.state import get_timeline
+
+    trace = get_timeline()
+    if filename:
+        import json as _json
+
+        with open(filename, "w") as f:
+            _json.dump(trace, f)
+    return trace
+
+
 def get_runtime_context() -> RuntimeContext:
     _check_connected()
     return RuntimeContext(global_worker)

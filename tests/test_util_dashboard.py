@@ -135,3 +135,23 @@ def test_metrics_and_dashboard(ray_mod):
     with urllib.request.urlopen(f"{base}/api/jobs/{job_id}/logs",
                                 timeout=10) as r:
         assert b"rest_job_ok" in r.read()
+
+
+def test_multiprocessing_pool(ray_mod):
+    """ray.util.multiprocessing.Pool parity: map/starmap/apply_async/
+    imap_unordered on actor-backed workers."""
+    from ant_ray_amd.util.multiprocessing import Pool
+
+    def sq(x):
+        return x * x
+
+    def add(a, b):
+        return a + b
+
+    with Pool(processes=3) as p:
+        assert p.map(sq, range(10)) == [x * x for x in range(10)]
+        assert p.starmap(add, [(1, 2), (3, 4)]) == [3, 7]
+        ar = p.apply_async(sq, (9,))
+        assert ar.get(timeout=60) == 81
+        assert sorted(p.imap_unordered(sq, range(5))) == [0, 1, 4, 9, 16]
+        assert list(p.imap(sq, range(5))) == [0, 1, 4, 9, 16]
