@@ -191,12 +191,20 @@ def timeline(output):
 
 
 @cli.command()
-@click.argument("kind", type=click.Choice(["tasks"]))
+@click.argument("kind", type=click.Choice(["tasks", "actors", "objects"]))
 def summary(kind):
-    from ant_ray_amd.util.state import summarize_tasks
+    from ant_ray_amd.util import state as S
 
-    for row in summarize_tasks():
-        click.echo(f"{row['name'] or '(anon)':<40} {row['state']:<10} {row['count']}")
+    if kind == "tasks":
+        for row in S.summarize_tasks():
+            click.echo(f"{row['name'] or '(anon)':<40} "
+                       f"{row['state']:<10} {row['count']}")
+    elif kind == "actors":
+        for row in S.summarize_actors():
+            click.echo(f"{row['state']:<20} {row['count']}")
+    else:
+        for row in S.summarize_objects():
+            click.echo(json.dumps(row))
 
 
 # ----------------------------------------------------------------------- job

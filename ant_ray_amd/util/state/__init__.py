@@ -6,6 +6,7 @@ query goes straight to the GCS over the connected worker's channel.
 """
 from __future__ import annotations
 
+import os
 from typing import Any, Dict, List, Optional
 
 
@@ -95,6 +96,42 @@ def list_objects(filters=None, limit: int = 1000, **_):
             "object_store_memory_available": n.get("resources_available", {}).get("object_store_memory"),
         })
     return out[:limit]
+
+
+def list_logs(node_id: str = None) -> List[str]:
+    """Session log files (parity: ray.util.state.list_logs)."""
+    from ant_ray_amd._private.worker import global_worker
+
+    cw = global_worker.core_worker
+    logs_dir = os.path.join(getattr(cw, "session_dir", "") or "", "logs")
+    if not os.path.isdir(logs_dir):
+        return []
+    return sorted(os.listdir(logs_dir))
+
+
+def get_log(filename: str, tail: int = 1000) -> str:
+    """Tail of one session log file (parity: ray.util.state.get_log)."""
+    from ant_ray_amd._private.worker import global_worker
+
+    cw = global_worker.core_worker
+    fp = os.path.join(getattr(cw, "session_dir", "") or "", "logs",
+                      os.path.basename(filename))
+    if not os.path.isfile(fp):
+        raise FileNotFoundError(filename)
+    with open(fp, errors="replace") as f:
+        lines = f.readlines()
+    return "".join(lines[-tail:])
+
+
+def summarize_actors():
+    from collections import Counter
+
+    c = Counter(a.get("state") for a in list_actors(limit=20000))
+    return [{"state": k, "count": v} for k, v in c.items()]
+
+
+def summarize_objects():
+    return store_stats()
 
 
 def store_stats() -> List[Dict[str, Any]]:

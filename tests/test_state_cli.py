@@ -121,3 +121,26 @@ def test_cli_memory(ray_mod):
     assert r.exit_code == 0, r.output
     assert "MiB used" in r.output
     del keep
+
+
+def test_logs_and_summaries(ray_mod):
+    from click.testing import CliRunner
+
+    from ant_ray_amd.scripts.cli import cli
+    from ant_ray_amd.util import state
+
+    logs = state.list_logs()
+    assert isinstance(logs, list)
+    if logs:
+        assert isinstance(state.get_log(logs[0], tail=10), str)
+    assert isinstance(state.summarize_actors(), list)
+    r = CliRunner().invoke(cli, ["summary", "actors"])
+    assert r.exit_code == 0, r.output
+
+
+def test_llm_alias_imports():
+    import ant_ray_amd.data.llm as dllm
+    import ant_ray_amd.serve.llm as sllm
+
+    assert hasattr(sllm, "build_openai_app")
+    assert hasattr(dllm, "build_llm_processor")
