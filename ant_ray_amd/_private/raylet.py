@@ -794,7 +794,8 @@ async def run_raylet(args):
             raylet.labels.update(json.loads(src))
     await raylet.start(args.port)
     if args.announce_fd:
-        os.write(args.announce_fd, (str(raylet.port) + "\n").encode())
+        os.write(args.announce_fd,
+                 f"{raylet.port} {raylet.node_id.hex()}\n".encode())
         os.close(args.announce_fd)
     await raylet._shutdown.wait()
 

@@ -91,8 +91,11 @@ class Cluster:
                                     "wb"))
         os.close(w_fd)
         with os.fdopen(r_fd) as f:
-            port = int(f.readline().strip())
+            parts = f.readline().split()
+            port = int(parts[0])
         node = ClusterNode(proc, port, store_path)
+        if len(parts) > 1:
+            node._node_id = parts[1]
         self.worker_nodes.append(node)
         if wait:
             self.wait_for_nodes()
