@@ -176,6 +176,34 @@ class DQN:
         ]
         self.iteration = 0
 
+
+    def save(self, path: str) -> str:
+        """Persist the algorithm state (parity: Algorithm.save_checkpoint)."""
+        import os
+
+        import torch as _torch
+
+        os.makedirs(path, exist_ok=True)
+        _torch.save({
+            "policy": self.q.state_dict(),
+            "optimizer": self.opt.state_dict(),
+            "iteration": self.iteration,
+        }, os.path.join(path, "algorithm_state.pt"))
+        return path
+
+    def restore(self, path: str):
+        """Load state saved by save() (parity: Algorithm.restore)."""
+        import os
+
+        import torch as _torch
+
+        st = _torch.load(os.path.join(path, "algorithm_state.pt"),
+                         weights_only=False)
+        self.q.load_state_dict(st["policy"])
+        self.target.load_state_dict(st["policy"])
+        self.opt.load_state_dict(st["optimizer"])
+        self.iteration = st["iteration"]
+
     def _epsilon(self) -> float:
         c = self.cfg
         frac = min(1.0, self.iteration / max(1, c.eps_decay_iters))

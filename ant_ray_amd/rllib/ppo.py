@@ -152,6 +152,33 @@ class PPO:
         ]
         self.iteration = 0
 
+
+    def save(self, path: str) -> str:
+        """Persist the algorithm state (parity: Algorithm.save_checkpoint)."""
+        import os
+
+        import torch as _torch
+
+        os.makedirs(path, exist_ok=True)
+        _torch.save({
+            "policy": self.policy.state_dict(),
+            "optimizer": self.opt.state_dict(),
+            "iteration": self.iteration,
+        }, os.path.join(path, "algorithm_state.pt"))
+        return path
+
+    def restore(self, path: str):
+        """Load state saved by save() (parity: Algorithm.restore)."""
+        import os
+
+        import torch as _torch
+
+        st = _torch.load(os.path.join(path, "algorithm_state.pt"),
+                         weights_only=False)
+        self.policy.load_state_dict(st["policy"])
+        self.opt.load_state_dict(st["optimizer"])
+        self.iteration = st["iteration"]
+
     def train(self) -> Dict:
         """One iteration: parallel rollouts -> PPO update -> broadcast."""
         import ant_ray_amd as ray

@@ -85,3 +85,28 @@ def test_dqn_learns_bandit(ray_start_regular_module):
     assert last["loss"] is not None
     # with eps=0 the greedy policy must pick the rewarding arm
     assert last["episode_return_mean"] > 0.9, last
+
+
+def test_algorithm_save_restore(tmp_path):
+    import ant_ray_amd as ray
+    from ant_ray_amd.rllib import CartPoleEnv, PPOConfig
+
+    if ray.is_initialized():
+        ray.shutdown()
+    ray.init(num_cpus=4)
+    algo = (PPOConfig().environment(lambda seed: CartPoleEnv(seed=seed))
+            .env_runners(num_env_runners=1, rollout_fragment_length=64)
+            .build())
+    algo.train()
+    path = algo.save(str(tmp_path / "ckpt"))
+
+    algo2 = (PPOConfig().environment(lambda seed: CartPoleEnv(seed=seed))
+             .env_runners(num_env_runners=1, rollout_fragment_length=64)
+             .build())
+    algo2.restore(path)
+    assert algo2.iteration == 1
+    import torch
+
+    for a, b in zip(algo.policy.parameters(), algo2.policy.parameters()):
+        assert torch.equal(a, b)
+    ray.shutdown()
