@@ -1,1 +1,5 @@
-from ant_ray_amd._common.usage import usage_lib  # noqa: F401
+from ant_ray_amd._common.usage.usage_lib import (  # noqa: F401
+    TagKey,
+    record_extra_usage_tag,
+    usage_stats_enabled,
+)

@@ -1,57 +1,61 @@
-"""Usage stats (opt-out telemetry) — collection parity, no egress.
+"""Usage-stats lib (parity: reference python/ray/_common/usage/usage_lib.py).
 
-Role parity: reference python/ray/_common/usage/usage_lib.py: collects
-cluster metadata + feature-usage tags and reports to a telemetry endpoint
-unless RAY_USAGE_STATS_ENABLED=0. This deployment is air-gapped: the same
-report payload is assembled and written to the session dir instead of
-being POSTed anywhere.
+The reference collects opt-out telemetry and reports it to a public
+endpoint; this air-gapped build keeps the same recording API but only
+persists locally (session_dir/usage_stats.json) and NEVER transmits.
+Disabled entirely with RAY_USAGE_STATS_ENABLED=0 (same knob).
 """
 from __future__ import annotations
 
 import json
 import os
+import threading
 import time
-from typing import Any, Dict
 
-_tags: Dict[str, str] = {}
+_lock = threading.Lock()
+_tags: dict = {}
+
+
+class TagKey:
+    """Well-known extra-usage tag names (subset of usage_pb2.TagKey)."""
+
+    RLLIB_FRAMEWORK = "rllib_framework"
+    TRAIN_TRAINER = "train_trainer"
+    SERVE_API_VERSION = "serve_api_version"
+    DATA_API = "data_api"
+    CORE_STATE_API = "core_state_api"
 
 
 def usage_stats_enabled() -> bool:
-    return os.environ.get("RAY_USAGE_STATS_ENABLED", "1") == "1"
-
-
-def record_library_usage(library: str):
-    _tags[f"library_{library}"] = "1"
+    return os.environ.get("RAY_USAGE_STATS_ENABLED", "1") not in ("0", "false")
 
 
 def record_extra_usage_tag(key: str, value: str):
-    _tags[str(key)] = str(value)
-
-
-def generate_report() -> Dict[str, Any]:
-    import platform
-
-    import ant_ray_amd
-
-    return {
-        "schema_version": "0.1",
-        "source": "ant_ray_amd",
-        "session_start_timestamp_ms": int(time.time() * 1000),
-        "os": platform.system().lower(),
-        "python_version": platform.python_version(),
-        "version": ant_ray_amd.__version__,
-        "extra_usage_tags": dict(_tags),
-        "total_num_cpus": os.cpu_count(),
-    }
-
-
-def write_report(session_dir: str) -> str:
+    """Record a library-usage tag (no-op when disabled; local-only)."""
     if not usage_stats_enabled():
-        return ""
-    path = os.path.join(session_dir, "usage_stats.json")
+        return
+    with _lock:
+        _tags[str(key)] = str(value)
+    _flush()
+
+
+def get_extra_usage_tags() -> dict:
+    with _lock:
+        return dict(_tags)
+
+
+def _flush():
     try:
-        with open(path, "w") as f:
-            json.dump(generate_report(), f)
-    except OSError:
-        return ""
-    return path
+        from ant_ray_amd._private.worker import global_worker
+
+        cw = global_worker.core_worker
+        sd = getattr(cw, "session_dir", "") if cw else ""
+        if not sd:
+            return
+        with _lock:
+            snap = {"tags": dict(_tags), "ts": time.time(),
+                    "transmitted": False}  # air-gapped: local record only
+        with open(os.path.join(sd, "usage_stats.json"), "w") as f:
+            json.dump(snap, f)
+    except Exception:
+        pass
