@@ -978,7 +978,8 @@ class CoreWorker:
             for r in refs:
                 self._lineage[r.binary()] = (key, payload, resources, opts)
             self._lineage_bytes += cost
-            while self._lineage_bytes > 256 * 1024 * 1024 and self._lineage:
+            while ((self._lineage_bytes > 256 * 1024 * 1024
+                    or len(self._lineage) > 50_000) and self._lineage):
                 old_oid = next(iter(self._lineage))
                 _, old_p, _, _ = self._lineage.pop(old_oid)
                 self._lineage_bytes -= len(old_p.get("args") or b"") + 512
