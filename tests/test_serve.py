@@ -324,3 +324,66 @@ def test_serve_deploy_yaml(serve_mod, tmp_path, monkeypatch):
         serve.delete("appb")
     finally:
         sys.path.remove(str(tmp_path))
+
+
+def test_autoscaling_e2e(serve_mod):
+    """Request-rate autoscaling end to end: concurrent slow requests grow
+    the replica set above min; draining traffic shrinks it back."""
+    import threading
+
+    serve = serve_mod
+    for app_name in list(serve.status().get("applications", {})):
+        serve.delete(app_name)
+
+    @serve.deployment(autoscaling_config={
+        "min_replicas": 1, "max_replicas": 3,
+        "target_ongoing_requests": 1.0,
+        "upscale_delay_s": 0.2, "downscale_delay_s": 0.5,
+    }, max_ongoing_requests=10)
+    class Slow:
+        async def __call__(self, x):
+            import asyncio as _a
+
+            await _a.sleep(0.4)
+            return x
+
+    h = serve.run(Slow.bind(), name="auto", route_prefix="/auto")
+
+    stop = threading.Event()
+
+    def hammer():
+        while not stop.is_set():
+            try:
+                h.remote(1).result(timeout_s=30)
+            except Exception:
+                pass
+
+    threads = [threading.Thread(target=hammer, daemon=True)
+               for _ in range(8)]
+    for t in threads:
+        t.start()
+    try:
+        deadline = time.time() + 60
+        grew = False
+        while time.time() < deadline:
+            st = serve.status()["applications"]["auto"]["deployments"]["Slow"]
+            if st["replicas"] > 1:
+                grew = True
+                break
+            time.sleep(0.5)
+        assert grew, "autoscaler never scaled up under load"
+    finally:
+        stop.set()
+        for t in threads:
+            t.join(timeout=10)
+
+    deadline = time.time() + 60
+    shrunk = False
+    while time.time() < deadline:
+        st = serve.status()["applications"]["auto"]["deployments"]["Slow"]
+        if st["replicas"] == 1:
+            shrunk = True
+            break
+        time.sleep(0.5)
+    assert shrunk, "autoscaler never scaled back down"
+    serve.delete("auto")
