@@ -127,9 +127,19 @@ class Replica:
                 a.to_starlette() if isinstance(a, HTTPRequestData) else a
                 for a in args
             )
-            out = fn(*args, **(kwargs or {}))
-            if inspect.iscoroutine(out):
-                out = await out
+            if inspect.iscoroutinefunction(fn):
+                out = await fn(*args, **(kwargs or {}))
+            else:
+                # sync callables run on a thread pool so one blocking
+                # handler never stalls the replica's event loop (health
+                # probes, concurrent requests) — parity with the
+                # reference's sync-handler thread execution
+                import functools
+
+                out = await asyncio.get_event_loop().run_in_executor(
+                    None, functools.partial(fn, *args, **(kwargs or {})))
+                if inspect.iscoroutine(out):
+                    out = await out
             return _encode_response(out)
         finally:
             self._num_ongoing -= 1

@@ -387,3 +387,29 @@ def test_autoscaling_e2e(serve_mod):
         time.sleep(0.5)
     assert shrunk, "autoscaler never scaled back down"
     serve.delete("auto")
+
+
+def test_sync_handlers_run_concurrently(serve_mod):
+    """A BLOCKING sync handler must not serialize the replica: concurrent
+    requests overlap on the replica's thread pool (the reference runs
+    sync handlers off the event loop)."""
+    serve = serve_mod
+    for app_name in list(serve.status().get("applications", {})):
+        serve.delete(app_name)
+
+    @serve.deployment(max_ongoing_requests=8)
+    class Blocking:
+        def __call__(self, x):
+            import time as _t
+
+            _t.sleep(0.5)
+            return x
+
+    h = serve.run(Blocking.bind(), name="blocky", route_prefix="/blocky")
+    t0 = time.time()
+    rs = [h.remote(i) for i in range(6)]
+    out = sorted(r.result(timeout_s=60) for r in rs)
+    dt = time.time() - t0
+    assert out == list(range(6))
+    assert dt < 2.0, f"6x0.5s blocking requests took {dt:.2f}s (serialized)"
+    serve.delete("blocky")
