@@ -420,7 +420,27 @@ class GcsServer:
         for actor in list(self.actors.values()):
             if actor.worker_id == wid and actor.state in (ALIVE, PENDING_CREATION, RESTARTING):
                 await self._on_actor_worker_died(actor, p.get("reason", "worker died"))
+        await self._kill_actors_owned_by(wid, p.get("reason", "owner died"))
         return {"ok": True}
+
+    async def _kill_actors_owned_by(self, owner_wid: bytes, reason: str):
+        """Owner fate-sharing (parity: reference non-detached actors die
+        with their owner): when the creating worker/driver dies, its
+        non-detached actors are torn down — e.g. a crashed Serve
+        controller's replicas, so its restart rebuilds a clean set."""
+        for actor in list(self.actors.values()):
+            if (actor.owner == owner_wid
+                    and actor.state in (ALIVE, PENDING_CREATION, RESTARTING)
+                    and actor.opts.get("lifetime") != "detached"):
+                actor.opts["max_restarts"] = 0
+                actor.killed = True
+                if actor.addr:
+                    try:
+                        wconn = await self._get_worker_conn(tuple(actor.addr))
+                        await wconn.notify("exit_worker", {
+                            "reason": f"owner died: {reason}"})
+                    except Exception:
+                        pass
 
 
     # -------------------------------------------------- virtual clusters

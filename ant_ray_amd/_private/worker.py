@@ -178,6 +178,10 @@ class ActorHandleState:
         self.is_owner = False
         self.detached = False
         self.restart_gen = 0  # mirrors GCS num_restarts; seq resets with it
+        # task-id counter: NEVER resets (unlike seq) — a post-restart call
+        # must not reuse a pre-restart call's deterministic task id, or
+        # ray.get would return the old call's cached result
+        self.task_counter = 0
 
 
 class CoreWorker:
@@ -1368,7 +1372,9 @@ class CoreWorker:
         with st.lock:
             st.seq += 1
             seq = st.seq
-        task_id = TaskID.for_actor_task(ActorID(actor_id), seq).binary()
+            st.task_counter += 1
+            tc = st.task_counter
+        task_id = TaskID.for_actor_task(ActorID(actor_id), tc).binary()
         if streaming:
             n_returns = 0
             self._streams[task_id] = _StreamState()

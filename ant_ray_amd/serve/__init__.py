@@ -160,7 +160,7 @@ def _get_or_create_controller(http_host="127.0.0.1", http_port=8000):
     Controller = ray.remote(ServeController)
     c = Controller.options(
         name=CONTROLLER_NAME, lifetime="detached", num_cpus=0,
-        max_concurrency=1000,
+        max_concurrency=1000, max_restarts=-1,
     ).remote(http_port, http_host)
     return c
 

@@ -65,6 +65,61 @@ class ServeController:
         if self._bg is None:
             loop = asyncio.get_event_loop()
             self._bg = loop.create_task(self._reconcile_loop())
+            loop.create_task(self._recover_state_from_checkpoint())
+
+    # ------------------------------------------------- crash recovery
+    def _checkpoint_state(self):
+        """Persist the app configs to the GCS KV (parity: the reference
+        controller's KV checkpoint, controller.py:586) so a restarted
+        controller (max_restarts=-1) redeploys everything."""
+        try:
+            snap = pickle.dumps({
+                "apps": {
+                    app: {
+                        "route_prefix": self.route_prefixes.get(app),
+                        "ingress": self.ingress.get(app),
+                        "deployments": [dr.cfg for dr in deps.values()],
+                    }
+                    for app, deps in self.apps.items()
+                },
+                "http": (self.http_host, self.http_port),
+            })
+            from ant_ray_amd.experimental import internal_kv
+
+            internal_kv._internal_kv_put(b"controller_ckpt", snap,
+                                         namespace=b"serve")
+        except Exception:
+            logger.exception("serve controller checkpoint failed")
+
+    async def _recover_state_from_checkpoint(self):
+        if self.apps:
+            return  # fresh deploys already arrived
+        try:
+            from ant_ray_amd.experimental import internal_kv
+
+            raw = internal_kv._internal_kv_get(b"controller_ckpt",
+                                               namespace=b"serve")
+        except Exception:
+            return
+        if not raw:
+            return
+        try:
+            snap = pickle.loads(raw)
+        except Exception:
+            logger.exception("bad serve controller checkpoint")
+            return
+        for app, info in snap.get("apps", {}).items():
+            if app in self.apps:
+                continue
+            logger.warning("serve controller: recovering app %r from "
+                           "checkpoint", app)
+            try:
+                await self.deploy_application(
+                    app, info.get("route_prefix") or "/",
+                    info.get("deployments") or [],
+                    info.get("ingress"))
+            except Exception:
+                logger.exception("recovery of app %r failed", app)
 
     # ------------------------------------------------------------- deploy
 
@@ -93,6 +148,7 @@ class ServeController:
             await self._scale_to(dr, target, name, d["name"])
         self.ingress[name] = ingress
         self.route_prefixes[name] = route_prefix
+        self._checkpoint_state()
         return True
 
     async def delete_application(self, name: str):
@@ -101,6 +157,7 @@ class ServeController:
             await self._scale_to(dr, 0, name, dep_name)
         self.ingress.pop(name, None)
         self.route_prefixes.pop(name, None)
+        self._checkpoint_state()
         return True
 
     async def _scale_to(self, dr: DeploymentReplicas, target: int,
@@ -147,6 +204,12 @@ class ServeController:
                     "(insufficient cluster resources for "
                     f"{len(dr.replicas)} replicas)") from None
 
+    async def getpid(self) -> int:
+        import os
+
+        self._ensure_bg()
+        return os.getpid()
+
     # ------------------------------------------------------------ queries
 
     async def get_replicas(self, app: str, deployment: str):
@@ -170,6 +233,7 @@ class ServeController:
                 "deployments": list(self.apps[app])}
 
     async def list_applications(self):
+        self._ensure_bg()  # first call after a restart triggers recovery
         def _ing_streaming(app, deps):
             ing = self.ingress.get(app)
             dr = deps.get(ing) if ing else None
