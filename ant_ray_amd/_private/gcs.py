@@ -380,7 +380,38 @@ class GcsServer:
             if is_driver and job_id in self.jobs:
                 self.jobs[job_id]["virtual_cluster_id"] = vc
         conn.session["worker_id"] = wid
+        self.workers[wid]["conn"] = conn
+        prev = conn.on_close
+
+        def _closed(c, wid=wid, is_driver=is_driver, job_id=job_id, prev=prev):
+            if prev:
+                prev(c)
+            asyncio.get_running_loop().create_task(
+                self._on_worker_conn_closed(wid, is_driver, job_id, c))
+
+        conn.on_close = _closed
         return {"job_id": job_id}
+
+    async def _on_worker_conn_closed(self, wid, is_driver, job_id, conn):
+        """A registered worker/driver's GCS connection dropped. Grace
+        period covers live reconnects (transient TCP drops); if the
+        process is really gone and it was a DRIVER, its non-detached
+        actors die with it (owner fate-sharing, reference job cleanup)
+        and the job completes."""
+        await asyncio.sleep(10.0)
+        cur = self.workers.get(wid)
+        if cur is None or cur.get("conn") is not conn:
+            return  # re-registered (or already cleaned up)
+        self.workers.pop(wid, None)
+        if is_driver:
+            logger.warning("driver %s disconnected; cleaning up its actors",
+                           wid.hex()[:8] if isinstance(wid, bytes) else wid)
+            await self._kill_actors_owned_by(wid, "driver exited")
+            job = self.jobs.get(job_id)
+            if job is not None and job.get("state") == "RUNNING":
+                job["state"] = "FINISHED"
+                job["end_time"] = time.time()
+                self._persist_soon()
 
     async def rpc_pick_raylet(self, conn, p):
         """Name a feasible raylet for a task lease that is infeasible on
@@ -918,7 +949,8 @@ class GcsServer:
         return evs[-limit:]
 
     async def rpc_list_workers(self, conn, p):
-        return list(self.workers.values())
+        return [{k: v for k, v in w.items() if k != "conn"}
+                for w in self.workers.values()]
 
     async def rpc_list_jobs(self, conn, p):
         return list(self.jobs.values())

@@ -195,3 +195,65 @@ def test_head_fate_shares_with_driver(tmp_path):
     finally:
         if proc.poll() is None:
             proc.kill()
+
+
+def test_driver_death_cleans_up_actors(tmp_path):
+    """A SIGKILLed driver's non-detached actors die with it (owner
+    fate-sharing on driver disconnect); detached actors survive."""
+    import textwrap
+
+    from ant_ray_amd.cluster_utils import Cluster
+
+    import ant_ray_amd as ray
+
+    if ray.is_initialized():
+        ray.shutdown()
+    c = Cluster(initialize_head=True, head_node_args={"num_cpus": 4})
+    try:
+        script = textwrap.dedent(f"""
+            import time
+            import ant_ray_amd as ray
+            ray.init(address="{c.address}")
+
+            @ray.remote
+            class A:
+                def ping(self):
+                    return 1
+
+            plain = A.remote()
+            kept = A.options(name="survivor", lifetime="detached").remote()
+            ray.get([plain.ping.remote(), kept.ping.remote()], timeout=60)
+            print("READY", flush=True)
+            time.sleep(300)
+        """)
+        proc = subprocess.Popen([sys.executable, "-c", script],
+                                stdout=subprocess.PIPE, text=True,
+                                cwd="/root/repo")
+        assert proc.stdout.readline().startswith("READY")
+        proc.kill()
+        proc.wait(timeout=10)
+
+        c.connect()
+        from ant_ray_amd.util import state
+
+        deadline = time.time() + 60
+        ok = False
+        while time.time() < deadline:
+            rows = state.list_actors()
+            named = {r["name"]: r["state"] for r in rows}
+            anon_alive = [r for r in rows
+                          if not r["name"] and r["state"] == "ALIVE"]
+            if (named.get("survivor") == "ALIVE" and not anon_alive):
+                ok = True
+                break
+            time.sleep(1.0)
+        assert ok, rows
+        # the detached actor still answers
+        s = ray.get_actor("survivor")
+        assert ray.get(s.ping.remote(), timeout=60) == 1
+    finally:
+        try:
+            ray.shutdown()
+        except Exception:
+            pass
+        c.shutdown()
