@@ -181,7 +181,7 @@ def start(detached: bool = True, http_options: Optional[dict] = None, **_):
         Proxy = ray.remote(HTTPProxy)
         proxy = Proxy.options(
             name=PROXY_NAME, lifetime="detached", num_cpus=0,
-            max_concurrency=1000,
+            max_concurrency=1000, max_restarts=-1,
         ).remote(host, port)
         ray.get(proxy.ready.remote(), timeout=60)
     return controller
