@@ -165,7 +165,8 @@ def _get_or_create_controller(http_host="127.0.0.1", http_port=8000):
     return c
 
 
-def start(detached: bool = True, http_options: Optional[dict] = None, **_):
+def start(detached: bool = True, http_options: Optional[dict] = None,
+          grpc_options: Optional[dict] = None, **_):
     """Start Serve system actors (controller + HTTP proxy)."""
     import ant_ray_amd as ray
 
@@ -184,6 +185,11 @@ def start(detached: bool = True, http_options: Optional[dict] = None, **_):
             max_concurrency=1000, max_restarts=-1,
         ).remote(host, port)
         ray.get(proxy.ready.remote(), timeout=60)
+    if grpc_options:
+        from ant_ray_amd.serve._private.grpc_proxy import start_grpc_proxy
+
+        start_grpc_proxy(grpc_options.get("host", "127.0.0.1"),
+                         grpc_options.get("port", 9000))
     return controller
 
 

@@ -457,3 +457,44 @@ def test_controller_crash_recovery(serve_mod):
         time.sleep(1.0)
     assert ok, "controller did not recover the app after SIGKILL"
     serve.delete("recov")
+
+
+def test_grpc_ingress(serve_mod):
+    """gRPC proxy parity: a stub-less unary call to
+    /ray.serve.UserApplicationService/<app> reaches the deployment."""
+    import pickle
+
+    import grpc
+
+    import ant_ray_amd as ray
+
+    serve = serve_mod
+    for app_name in list(serve.status().get("applications", {})):
+        serve.delete(app_name)
+
+    @serve.deployment
+    def echo_upper(payload):
+        return {"upper": payload["text"].upper()}
+
+    serve.run(echo_upper.bind(), name="grpcapp", route_prefix="/grpcapp")
+    from ant_ray_amd.serve._private.grpc_proxy import start_grpc_proxy
+
+    port = start_grpc_proxy(port=0)
+    assert port > 0
+
+    ch = grpc.insecure_channel(f"127.0.0.1:{port}")
+    call = ch.unary_unary(
+        "/ray.serve.UserApplicationService/grpcapp",
+        request_serializer=None, response_deserializer=None)
+    reply = call(pickle.dumps({"text": "grpc works"}), timeout=60)
+    assert pickle.loads(reply) == {"upper": "GRPC WORKS"}
+
+    # unknown app -> NOT_FOUND
+    bad = ch.unary_unary("/ray.serve.UserApplicationService/nope",
+                         request_serializer=None, response_deserializer=None)
+    import pytest as _pt
+
+    with _pt.raises(grpc.RpcError):
+        bad(b"x", timeout=30)
+    ch.close()
+    serve.delete("grpcapp")
