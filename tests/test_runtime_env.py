@@ -72,3 +72,29 @@ def test_pip_rejected_offline():
         build_worker_spawn(["python"], {}, {"pip": ["requests"]})
     with pytest.raises(RuntimeEnvSetupError, match="unknown"):
         build_worker_spawn(["python"], {}, {"bogus_plugin": 1})
+
+
+def test_runtime_env_class(ray_mod):
+    """ray.runtime_env.RuntimeEnv public type: validates fields, passes
+    through the normal runtime_env plumbing as a dict."""
+    import pytest as _pt
+
+    import ant_ray_amd as ray
+    from ant_ray_amd.runtime_env import RuntimeEnv
+
+    with _pt.raises(ValueError):
+        RuntimeEnv(pip=["requests"])  # cloud-only: rejected up front
+    with _pt.raises(ValueError):
+        RuntimeEnv(bogus_field=1)
+    with _pt.raises(TypeError):
+        RuntimeEnv(env_vars={"A": 1})
+
+    env = RuntimeEnv(env_vars={"MARKER_VAR": "via-class"})
+
+    @ray.remote(runtime_env=env)
+    def read():
+        import os
+
+        return os.environ.get("MARKER_VAR")
+
+    assert ray.get(read.remote(), timeout=60) == "via-class"
