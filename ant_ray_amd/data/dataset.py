@@ -151,17 +151,9 @@ class Dataset:
 
     def random_shuffle(self, *, seed: Optional[int] = None, **kw) -> "Dataset":
         def _shuffle(refs: List[Any]) -> List[Any]:
-            blocks = [ray.get(r) for r in refs]
-            if not blocks:
-                return []
-            table = pa.concat_tables(blocks, promote_options="default")
-            rng = np.random.RandomState(seed)
-            idx = rng.permutation(table.num_rows)
-            shuffled = table.take(pa.array(idx))
-            k = max(len(refs), 1)
-            per = max(1, -(-shuffled.num_rows // k))
-            return [ray.put(shuffled.slice(s, per))
-                    for s in range(0, shuffled.num_rows, per)]
+            from ant_ray_amd.data.exchange import shuffle_exchange
+
+            return shuffle_exchange(refs, seed)
 
         return self._with(AllToAllOp(name="RandomShuffle", fn=_shuffle))
 
@@ -169,16 +161,9 @@ class Dataset:
         keys = [key] if isinstance(key, str) else list(key)
 
         def _sort(refs: List[Any]) -> List[Any]:
-            blocks = [ray.get(r) for r in refs]
-            if not blocks:
-                return []
-            table = pa.concat_tables(blocks, promote_options="default")
-            order = "descending" if descending else "ascending"
-            table = table.sort_by([(k, order) for k in keys])
-            k = max(len(refs), 1)
-            per = max(1, -(-table.num_rows // k))
-            return [ray.put(table.slice(s, per))
-                    for s in range(0, table.num_rows, per)]
+            from ant_ray_amd.data.exchange import sort_exchange
+
+            return sort_exchange(refs, keys, descending)
 
         return self._with(AllToAllOp(name="Sort", fn=_sort))
 
@@ -482,13 +467,17 @@ class GroupedData:
         key = self.key
 
         def _do(refs: List[Any]) -> List[Any]:
-            blocks = [ray.get(r) for r in refs]
-            if not blocks:
-                return []
-            table = pa.concat_tables(blocks, promote_options="default")
-            res = table.group_by(key).aggregate(aggs)
-            # arrow names output "<col>_<agg>"; keep as-is
-            return [ray.put(res)]
+            from ant_ray_amd.data.exchange import groupby_exchange
+
+            def finalize(t: pa.Table):
+                if t.num_rows == 0:
+                    return t
+                # arrow names output "<col>_<agg>"; keep as-is
+                return t.group_by(key).aggregate(aggs)
+
+            # hash exchange: a group lands wholly in one reducer, so the
+            # per-partition aggregates concatenate into the global answer
+            return groupby_exchange(refs, key, finalize)
 
         return self.ds._with(AllToAllOp(name="GroupByAgg", fn=_do))
 
@@ -511,16 +500,20 @@ class GroupedData:
         key = self.key
 
         def _do(refs: List[Any]) -> List[Any]:
-            blocks = [ray.get(r) for r in refs]
-            if not blocks:
-                return []
-            table = pa.concat_tables(blocks, promote_options="default")
-            df = table.to_pandas()
-            outs = []
-            for _, grp in df.groupby(key):
-                res = fn(grp if batch_format == "pandas" else
-                         {c: grp[c].to_numpy() for c in grp.columns})
-                outs.append(BlockAccessor.for_block(res).to_arrow())
-            return [ray.put(pa.concat_tables(outs, promote_options="default"))] if outs else []
+            from ant_ray_amd.data.exchange import groupby_exchange
+
+            def finalize(t: pa.Table):
+                if t.num_rows == 0:
+                    return t
+                df = t.to_pandas()
+                outs = []
+                for _, grp in df.groupby(key):
+                    res = fn(grp if batch_format == "pandas" else
+                             {c: grp[c].to_numpy() for c in grp.columns})
+                    outs.append(BlockAccessor.for_block(res).to_arrow())
+                return (pa.concat_tables(outs, promote_options="default")
+                        if outs else t.slice(0, 0))
+
+            return groupby_exchange(refs, key, finalize)
 
         return self.ds._with(AllToAllOp(name="MapGroups", fn=_do))
