@@ -192,23 +192,15 @@ class Dataset:
               "full": "full outer"}.get(join_type, join_type)
 
         def _join(refs: List[Any]) -> List[Any]:
-            left_blocks = [ray.get(r) for r in refs]
-            right_blocks = [ray.get(r)
-                            for r in other.iter_internal_ref_bundles()]
-            if not left_blocks or not right_blocks:
+            from ant_ray_amd.data.exchange import join_exchange
+
+            right_refs = list(other.iter_internal_ref_bundles())
+            if not refs or not right_refs:
                 if jt == "inner":
                     return []
                 # outer joins with one empty side degrade to the other side
-                blocks = left_blocks or right_blocks
-                return [ray.put(b) for b in blocks]
-            lt = pa.concat_tables(left_blocks, promote_options="default")
-            rt = pa.concat_tables(right_blocks, promote_options="default")
-            res = lt.join(rt, keys=keys, right_keys=rkeys, join_type=jt,
-                          left_suffix=suffixes[0], right_suffix=suffixes[1])
-            k = max(len(refs), 1)
-            per = max(1, -(-res.num_rows // k))
-            return [ray.put(res.slice(s, per))
-                    for s in range(0, res.num_rows, per)]
+                return list(refs) or right_refs
+            return join_exchange(refs, right_refs, keys, rkeys, jt, suffixes)
 
         return self._with(AllToAllOp(name=f"Join[{jt}]", fn=_join))
 

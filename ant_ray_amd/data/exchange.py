@@ -166,3 +166,48 @@ def groupby_exchange(refs: List[Any], key: str,
         return _part_hash(block, key, P)
 
     return exchange(refs, P, partitioner, finalize)
+
+
+def join_exchange(left_refs: List[Any], right_refs: List[Any],
+                  keys: List[str], right_keys: List[str], join_type: str,
+                  suffixes: tuple, P: Optional[int] = None) -> List[Any]:
+    """Distributed hash join: both sides are hash-partitioned on their
+    key so matching rows meet in the same reducer; per-partition arrow
+    joins concatenate into the global result (partitions are disjoint
+    in key space, so inner AND outer joins compose)."""
+    if not left_refs and not right_refs:
+        return []
+    P = P or max(1, min(max(len(left_refs), len(right_refs)), 16))
+
+    @ray.remote(num_cpus=1)
+    def part_left(block):
+        parts = _part_hash(block, keys[0], P)
+        return parts[0] if P == 1 else tuple(parts)
+
+    @ray.remote(num_cpus=1)
+    def part_right(block):
+        parts = _part_hash(block, right_keys[0], P)
+        return parts[0] if P == 1 else tuple(parts)
+
+    @ray.remote(num_cpus=1)
+    def reduce_join(n_left, *pieces):
+        lt = _concat(list(pieces[:n_left]))
+        rt = _concat(list(pieces[n_left:]))
+        if lt.num_rows == 0 and rt.num_rows == 0:
+            return lt
+        if lt.num_rows == 0 and join_type == "inner":
+            return lt
+        if rt.num_rows == 0 and join_type == "inner":
+            return rt.select([]) if lt.num_rows == 0 else lt.slice(0, 0)
+        return lt.join(rt, keys=keys, right_keys=right_keys,
+                       join_type=join_type, left_suffix=suffixes[0],
+                       right_suffix=suffixes[1])
+
+    louts = [part_left.options(num_returns=P).remote(r) for r in left_refs]
+    routs = [part_right.options(num_returns=P).remote(r) for r in right_refs]
+    out = []
+    for p in range(P):
+        lcol = [m if P == 1 else m[p] for m in louts]
+        rcol = [m if P == 1 else m[p] for m in routs]
+        out.append(reduce_join.remote(len(lcol), *(lcol + rcol)))
+    return out
