@@ -136,16 +136,16 @@ class Dataset:
 
     def repartition(self, num_blocks: int, **kw) -> "Dataset":
         def _repart(refs: List[Any]) -> List[Any]:
-            blocks = [ray.get(r) for r in refs]
-            if not blocks:
-                return []
-            table = pa.concat_tables(blocks, promote_options="default")
-            n = table.num_rows
-            per = max(1, -(-n // num_blocks))
-            out = []
-            for s in range(0, n, per):
-                out.append(ray.put(table.slice(s, per)))
-            return out
+            from ant_ray_amd.data.exchange import exchange
+
+            P = max(1, num_blocks)
+
+            def partitioner(block):
+                n = block.num_rows
+                per = -(-n // P) if n else 0
+                return [block.slice(min(p * per, n), per) for p in range(P)]
+
+            return exchange(refs, P, partitioner, lambda t: t)
 
         return self._with(AllToAllOp(name="Repartition", fn=_repart))
 
