@@ -484,6 +484,8 @@ class Raylet:
         """Grant a worker lease; queue if resources busy; spillback hint if
         infeasible on this node."""
         logger.info("lease request: %s", p.get("resources"))
+        if getattr(self, "_draining", False):
+            return {"granted": False, "reason": "node draining"}
         res = dict(p.get("resources") or {})
         pg = p.get("pg")
         selector = p.get("_label_selector")
@@ -758,6 +760,31 @@ class Raylet:
         stats["node_id"] = self.node_id
         stats["spilled_files"] = len(getattr(self, "_spilled", {}) or {})
         return stats
+
+    async def rpc_drain(self, conn, p):
+        """Graceful drain (parity: reference DrainNode / autoscaler
+        graceful termination): stop accepting leases, deny the queue,
+        wait for leased workers to finish (bounded), then exit."""
+        self._draining = True
+        deadline = time.monotonic() + float(p.get("timeout_s", 60.0))
+
+        async def _drain():
+            # deny queued leases so owners re-lease elsewhere
+            for (pl, _c), fut in list(self.pending_leases):
+                if not fut.done():
+                    fut.set_result({"granted": False,
+                                    "reason": "node draining"})
+            self.pending_leases.clear()
+            while time.monotonic() < deadline:
+                busy = [w for w in self.workers.values()
+                        if w.leased and w.proc.poll() is None]
+                if not busy:
+                    break
+                await asyncio.sleep(0.5)
+            await self._do_shutdown()
+
+        asyncio.get_running_loop().create_task(_drain())
+        return {"ok": True, "draining": True}
 
     async def rpc_shutdown(self, conn, p):
         await self._do_shutdown()

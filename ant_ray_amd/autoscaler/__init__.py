@@ -96,7 +96,7 @@ class LocalNodeProvider:
     def terminate_node(self, node_type: str):
         lst = self.nodes.get(node_type) or []
         if lst:
-            self.cluster.remove_node(lst.pop())
+            self.cluster.remove_node(lst.pop(), allow_graceful=True)
 
     def non_terminated_nodes(self) -> Dict[str, int]:
         return {t: len(v) for t, v in self.nodes.items()}

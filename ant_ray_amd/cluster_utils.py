@@ -101,8 +101,28 @@ class Cluster:
             self.wait_for_nodes()
         return node
 
-    def remove_node(self, node: ClusterNode, allow_graceful: bool = True):
-        node.kill()
+    def remove_node(self, node: ClusterNode, allow_graceful: bool = False):
+        """allow_graceful=True drains the raylet first (running leases
+        finish, queued ones spill elsewhere) — the autoscaler's idle
+        termination path; default stays the hard kill used by failure
+        tests."""
+        if allow_graceful:
+            try:
+                import asyncio
+
+                from ant_ray_amd._private import protocol
+
+                async def _drain():
+                    conn = await protocol.connect(("127.0.0.1", node.port),
+                                                  None, name="drain")
+                    await conn.call("drain", {"timeout_s": 30.0}, timeout=5)
+
+                asyncio.run(_drain())
+                node.proc.wait(timeout=45)
+            except Exception:
+                node.kill()
+        else:
+            node.kill()
         if node in self.worker_nodes:
             self.worker_nodes.remove(node)
 

@@ -305,3 +305,37 @@ def test_spread_scheduling_strategy(cluster):
     actors = [S.remote() for _ in range(9)]
     nodes = set(ray.get([a.node.remote() for a in actors], timeout=120))
     assert len(nodes) >= 2, f"SPREAD placed everything on {nodes}"
+
+
+def test_graceful_drain(cluster):
+    """remove_node(allow_graceful=True): the raylet finishes its running
+    lease before exiting (no task failure, no retry)."""
+    import os
+
+    import ant_ray_amd as ray
+
+    cluster.connect()
+    n = cluster.add_node(num_cpus=2, resources={"drainme": 1})
+    deadline = time.time() + 30
+    while time.time() < deadline and not ray.cluster_resources().get("drainme"):
+        time.sleep(0.2)
+
+    attempts = "/tmp/antray_drain_attempts"
+    open(attempts, "w").close()
+
+    @ray.remote(num_cpus=1, resources={"drainme": 0.1}, max_retries=0)
+    def slowish():
+        import os as _os
+        import time as _t
+
+        with open(attempts, "a") as f:
+            f.write(f"{_os.getpid()}\n")
+        _t.sleep(2.0)
+        return "finished"
+
+    ref = slowish.remote()
+    time.sleep(0.8)  # task is running
+    cluster.remove_node(n, allow_graceful=True)  # drain waits for it
+    assert ray.get(ref, timeout=60) == "finished"
+    with open(attempts) as f:
+        assert len(f.read().splitlines()) == 1  # ran exactly once
