@@ -64,11 +64,30 @@ class Router:
             i = a if self._ongoing.get(a, 0) <= self._ongoing.get(b, 0) else b
         return i, self._replicas[i]
 
-    def submit(self, method_name: Optional[str], args, kwargs):
+    def submit(self, method_name: Optional[str], args, kwargs,
+               multiplexed_model_id: Optional[str] = None):
         """Returns (ObjectRef, replica_index). The DeploymentResponse calls
         complete(index) when the result is consumed, closing the in-flight
-        accounting the pow-2 choice reads."""
-        i, replica = self.choose_replica()
+        accounting the pow-2 choice reads. multiplexed_model_id biases
+        routing to the replica that served that model last (parity:
+        the reference's model-multiplex-aware router — avoids reloading
+        an LRU-cached model on a cold replica)."""
+        i = None
+        if multiplexed_model_id is not None:
+            cache = getattr(self, "_model_affinity", None)
+            if cache is None:
+                cache = self._model_affinity = {}
+            self._refresh()
+            j = cache.get(multiplexed_model_id)
+            if j is not None and j < len(self._replicas):
+                i = j
+        if i is None:
+            i, _ = self.choose_replica()
+        if multiplexed_model_id is not None:
+            self._model_affinity[multiplexed_model_id] = i
+            while len(self._model_affinity) > 1024:
+                self._model_affinity.pop(next(iter(self._model_affinity)))
+        replica = self._replicas[i]
         self._ongoing[i] = self._ongoing.get(i, 0) + 1
         return replica.handle_request.remote(method_name, args, kwargs), i
 

@@ -104,11 +104,15 @@ class DeploymentHandle:
         return self._router
 
     def options(self, *, method_name: Optional[str] = None,
-                stream: bool = False, **_):
+                stream: bool = False,
+                multiplexed_model_id: Optional[str] = None, **_):
         h = DeploymentHandle(self.deployment_name, self.app_name,
                              method_name or self._method_name)
         h._stream = stream
-        h._router = self._router
+        h._model_id = multiplexed_model_id
+        # share ONE router across derived handles: in-flight accounting
+        # and model-affinity state must not reset per .options() call
+        h._router = self._get_router()
         return h
 
     def __getattr__(self, name):
@@ -121,7 +125,9 @@ class DeploymentHandle:
         if getattr(self, "_stream", False):
             gen, idx = router.submit_stream(self._method_name, args, kwargs)
             return DeploymentResponseGenerator(gen, router, idx)
-        ref, i = router.submit(self._method_name, args, kwargs)
+        ref, i = router.submit(
+            self._method_name, args, kwargs,
+            multiplexed_model_id=getattr(self, "_model_id", None))
         return DeploymentResponse(ref, router, i)
 
     def __reduce__(self):

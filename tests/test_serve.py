@@ -498,3 +498,37 @@ def test_grpc_ingress(serve_mod):
         bad(b"x", timeout=30)
     ch.close()
     serve.delete("grpcapp")
+
+
+def test_multiplexed_model_affinity_routing(serve_mod):
+    """handle.options(multiplexed_model_id=...): repeated calls for one
+    model stick to one replica (each model loads on exactly one of the
+    replicas instead of everywhere)."""
+    serve = serve_mod
+    for app_name in list(serve.status().get("applications", {})):
+        serve.delete(app_name)
+
+    @serve.deployment(num_replicas=2)
+    class MultiModel:
+        def __init__(self):
+            self.loaded = set()
+
+        @serve.multiplexed(max_num_models_per_replica=4)
+        async def get_model(self, model_id: str):
+            self.loaded.add(model_id)
+            return f"model:{model_id}"
+
+        async def __call__(self, model_id: str):
+            import os
+
+            await self.get_model(model_id)
+            return (os.getpid(), sorted(self.loaded))
+
+    h = serve.run(MultiModel.bind(), name="muxaff", route_prefix="/muxaff")
+    pids_a = {h.options(multiplexed_model_id="a").remote("a")
+              .result(timeout_s=30)[0] for _ in range(6)}
+    pids_b = {h.options(multiplexed_model_id="b").remote("b")
+              .result(timeout_s=30)[0] for _ in range(6)}
+    assert len(pids_a) == 1, "model 'a' requests should stick to one replica"
+    assert len(pids_b) == 1
+    serve.delete("muxaff")
