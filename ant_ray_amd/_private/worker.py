@@ -211,6 +211,7 @@ class CoreWorker:
         self._lineage_bytes = 0
         self._inflight_tasks: Dict[bytes, "LeasedWorker"] = {}
         self._cancelled_tasks: set = set()
+        self._retry_exceptions: Dict[bytes, Any] = {}
         self._killed_workers: Dict[bytes, str] = {}  # worker_id -> kill reason
         self._object_locations: Dict[bytes, Tuple[str, int]] = {}
         self._actors: Dict[bytes, ActorHandleState] = {}
@@ -977,6 +978,14 @@ class CoreWorker:
                 self._task_of_oid[r.binary()] = (task_id, key)
             while len(self._task_of_oid) > 20000:  # bound the cancel index
                 self._task_of_oid.pop(next(iter(self._task_of_oid)))
+        rexc = opts.get("retry_exceptions")
+        if rexc and not streaming:
+            # owner-side application-retry policy (classes never travel
+            # over the wire); True retries ANY exception
+            self._retry_exceptions[task_id] = (
+                True if rexc is True else tuple(rexc))
+            while len(self._retry_exceptions) > 20000:
+                self._retry_exceptions.pop(next(iter(self._retry_exceptions)))
         if not streaming:
             cost = len(payload.get("args") or b"") + 512
             for r in refs:
@@ -1251,6 +1260,46 @@ class CoreWorker:
 
         return bool(self.io.run(do(), timeout=30))
 
+    def _should_retry_app_error(self, payload, data, meta) -> bool:
+        """retry_exceptions (reference task option): an APPLICATION error
+        matching the policy consumes one of max_retries and resubmits the
+        task instead of surfacing the exception."""
+        task_id = payload.get("task_id")
+        policy = self._retry_exceptions.get(task_id)
+        if policy is None or payload.get("max_retries", 0) <= 0 \
+                or data is None:
+            return False
+        if policy is not True:
+            try:
+                err = serialization.deserialize(memoryview(data), meta)
+                cause = getattr(err, "cause", err)
+                cause_type = type(cause).__name__
+                names = {getattr(c, "__name__", str(c)) for c in policy}
+                # compare by name: the deserialized cause may be a
+                # RayTaskError-wrapped dynamic subclass
+                if cause_type not in names and not any(
+                        isinstance(cause, c) for c in policy
+                        if isinstance(c, type)):
+                    return False
+            except Exception:
+                return False
+        rec = self._lineage.get(
+            ObjectID.for_return(TaskID(task_id), 0).binary())
+        if rec is None:
+            return False
+        key, lpayload, res, opts = rec
+        lpayload = dict(lpayload, max_retries=lpayload.get("max_retries", 0) - 1)
+        logger.warning("task %s raised a retryable exception; retrying "
+                       "(%d retries left)", task_id.hex()[:8],
+                       lpayload["max_retries"])
+        for i in range(lpayload.get("n_returns", 1)):
+            roid = ObjectID.for_return(TaskID(task_id), i).binary()
+            self._lineage[roid] = (key, lpayload, res, opts)
+            self.store.memory.delete(roid)
+            self.store.memory.mark_pending(roid)
+        self.io.loop.create_task(self._enqueue_task(key, lpayload, res, opts))
+        return True
+
     def _fail_task(self, payload, exc: BaseException):
         task_id = payload["task_id"]
         st = self._streams.get(task_id)
@@ -1297,6 +1346,8 @@ class CoreWorker:
         else:
             data = reply.get("error_payload")
             meta = reply.get("error_meta", serialization.META_ERROR)
+            if self._should_retry_app_error(payload, data, meta):
+                return
             for i in range(payload.get("n_returns", 1)):
                 oid = ObjectID.for_return(TaskID(payload["task_id"]), i).binary()
                 if data is not None:

@@ -205,3 +205,47 @@ def test_max_calls_recycles_worker(ray_mod):
     pids = [ray.get(pid.remote(), timeout=60) for _ in range(6)]
     # 6 calls at max_calls=2 -> at least 3 distinct processes
     assert len(set(pids)) >= 3, pids
+
+
+def test_retry_exceptions(ray_mod):
+    """retry_exceptions: application errors matching the policy consume
+    max_retries and resubmit; non-matching errors surface immediately."""
+    import os
+
+    ray = ray_mod
+    marker = "/tmp/antray_retry_exc"
+    for f in (marker, marker + "2"):
+        try:
+            os.unlink(f)
+        except FileNotFoundError:
+            pass
+
+    @ray.remote(max_retries=3, retry_exceptions=[ValueError])
+    def flaky():
+        if not os.path.exists(marker):
+            open(marker, "w").close()
+            raise ValueError("transient")
+        return "recovered"
+
+    assert ray.get(flaky.remote(), timeout=60) == "recovered"
+
+    @ray.remote(max_retries=3, retry_exceptions=[KeyError])
+    def wrong_type():
+        raise ValueError("not retryable for this policy")
+
+    import pytest as _pt
+
+    with _pt.raises(Exception) as ei:
+        ray.get(wrong_type.remote(), timeout=60)
+    assert "not retryable" in str(ei.value)
+
+    @ray.remote(max_retries=2, retry_exceptions=True)
+    def always_fails():
+        with open(marker + "2", "a") as f:
+            f.write("x")
+        raise RuntimeError("permanent")
+
+    with _pt.raises(Exception):
+        ray.get(always_fails.remote(), timeout=60)
+    with open(marker + "2") as f:
+        assert len(f.read()) == 3  # initial try + 2 retries
