@@ -285,7 +285,14 @@ class Dataset:
         ops = self._ops
         if len(ops) == 1 and isinstance(ops[0], ReadOp) and ops[0].num_rows is not None:
             return ops[0].num_rows
-        return sum(BlockAccessor(b).num_rows() for b in self.iter_blocks())
+
+        @ray.remote(num_cpus=0.25)
+        def _rows(b):
+            return BlockAccessor(b).num_rows()
+
+        # counts happen task-side: only ints travel to the driver
+        return sum(ray.get([_rows.remote(r)
+                            for r in self.iter_internal_ref_bundles()]))
 
     def schema(self):
         for block in self.iter_blocks():
@@ -339,7 +346,12 @@ class Dataset:
         return len(list(self.iter_internal_ref_bundles()))
 
     def size_bytes(self) -> int:
-        return sum(BlockAccessor(b).size_bytes() for b in self.iter_blocks())
+        @ray.remote(num_cpus=0.25)
+        def _sz(b):
+            return BlockAccessor(b).size_bytes()
+
+        return sum(ray.get([_sz.remote(r)
+                            for r in self.iter_internal_ref_bundles()]))
 
     # ----------------------------------------------------------------- split
 
