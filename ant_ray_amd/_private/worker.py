@@ -212,6 +212,7 @@ class CoreWorker:
         self._inflight_tasks: Dict[bytes, "LeasedWorker"] = {}
         self._cancelled_tasks: set = set()
         self._retry_exceptions: Dict[bytes, Any] = {}
+        self._actor_retry_payloads: Dict[bytes, bytes] = {}  # task -> actor
         self._killed_workers: Dict[bytes, str] = {}  # worker_id -> kill reason
         self._object_locations: Dict[bytes, Tuple[str, int]] = {}
         self._actors: Dict[bytes, ActorHandleState] = {}
@@ -1283,6 +1284,25 @@ class CoreWorker:
                     return False
             except Exception:
                 return False
+        if payload.get("type") == "actor_task":
+            aid = self._actor_retry_payloads.get(task_id)
+            st = self._actors.get(aid) if aid else None
+            if st is None:
+                return False
+            lpayload = dict(payload,
+                            max_retries=payload.get("max_retries", 0) - 1)
+            logger.warning("actor task %s raised a retryable exception; "
+                           "retrying (%d left)", task_id.hex()[:8],
+                           lpayload["max_retries"])
+            for i in range(lpayload.get("n_returns", 1)):
+                roid = ObjectID.for_return(TaskID(task_id), i).binary()
+                self.store.memory.delete(roid)
+                self.store.memory.mark_pending(roid)
+            with st.lock:
+                st.seq += 1
+                lpayload["seq"] = st.seq
+            self.io.loop.create_task(self._submit_actor_async(st, lpayload))
+            return True
         rec = self._lineage.get(
             ObjectID.for_return(TaskID(task_id), 0).binary())
         if rec is None:
@@ -1442,6 +1462,11 @@ class CoreWorker:
         if not streaming:
             for r in refs:
                 self._task_of_oid[r.binary()] = ("actor", actor_id, task_id)
+            rexc = opts.get("retry_exceptions")
+            if rexc:
+                self._retry_exceptions[task_id] = (
+                    True if rexc is True else tuple(rexc))
+                self._actor_retry_payloads[task_id] = actor_id
         payload = {
             "type": "actor_task",
             "task_id": task_id,
@@ -1450,6 +1475,8 @@ class CoreWorker:
             "args": sobj.to_bytes(),
             "n_returns": n_returns,
             "streaming": streaming,
+            "max_retries": int(opts.get("max_task_retries",
+                                        opts.get("max_retries", 0)) or 0),
             "seq": seq,
             "caller": self.worker_id,
             "caller_addr": list(self.addr),

@@ -249,3 +249,24 @@ def test_retry_exceptions(ray_mod):
         ray.get(always_fails.remote(), timeout=60)
     with open(marker + "2") as f:
         assert len(f.read()) == 3  # initial try + 2 retries
+
+
+def test_actor_method_retry_exceptions(ray_mod):
+    """@ray.method(retry_exceptions=[...], max_task_retries=N): a failing
+    actor METHOD retries on the live actor without restarting it."""
+    ray = ray_mod
+
+    @ray.remote
+    class Flaky:
+        def __init__(self):
+            self.calls = 0
+
+        @ray.method(retry_exceptions=[ValueError], max_task_retries=3)
+        def work(self):
+            self.calls += 1
+            if self.calls < 3:
+                raise ValueError("transient")
+            return f"ok after {self.calls}"
+
+    f = Flaky.remote()
+    assert ray.get(f.work.remote(), timeout=60) == "ok after 3"
