@@ -131,6 +131,14 @@ class DashboardHead:
                 size = f.tell()
                 f.seek(max(0, size - 64 * 1024))
                 return 200, b"text/plain", f.read()
+        if path == "/api/serve/applications":
+            # serve observability (reference dashboard serve head)
+            try:
+                from ant_ray_amd import serve as _serve
+
+                return js(_serve.status())
+            except Exception as e:
+                return js({"error": str(e)}, 500)
         if path == "/api/virtual_clusters":
             return js(await self._gcs("list_virtual_clusters"))
         if path == "/metrics":

@@ -77,16 +77,76 @@ class SplitCoordinator:
     def __init__(self, dataset, n: int, equal: bool):
         self.n = n
         self.equal = equal
+        self._dataset = dataset
+        self._epoch = 0
         self._lock = threading.Lock()
         self._iter = dataset.iter_internal_ref_bundles()
         self._queues: List[List[Any]] = [[] for _ in range(n)]
         self._done = False
         self._next_shard = 0
         self._epoch_ids = [0] * n
+        self._planned = False
 
-    def next_block(self, shard: int):
-        """Returns a block ref or None when exhausted."""
+    def _plan_equal_locked(self):
+        """equal=True must be ROW-exact, not block-granular: shards that
+        differ by even one training step deadlock synchronous DDP (each
+        backward is a collective). Drain the ref stream (refs are tiny;
+        blocks stay in shm), count rows with tasks, deal contiguous
+        per-shard row ranges, and slice the boundary blocks via tasks.
+        Rows beyond n*floor(total/n) are dropped (the equalization)."""
+        import ant_ray_amd as ray
+
+        refs = list(self._iter)
+        self._done = True
+        if not refs:
+            self._planned = True
+            return
+
+        @ray.remote(num_cpus=0.25)
+        def nrows(b):
+            return b.num_rows
+
+        @ray.remote(num_cpus=0.25)
+        def slice_block(b, start, stop):
+            return b.slice(start, stop - start)
+
+        counts = ray.get([nrows.remote(r) for r in refs], timeout=600)
+        per = sum(counts) // self.n
+        shard, filled = 0, 0
+        for ref, cnt in zip(refs, counts):
+            off = 0
+            while off < cnt and shard < self.n:
+                take = min(cnt - off, per - filled)
+                if take <= 0:
+                    break
+                if off == 0 and take == cnt:
+                    self._queues[shard].append(ref)
+                else:
+                    self._queues[shard].append(
+                        slice_block.remote(ref, off, off + take))
+                filled += take
+                off += take
+                if filled == per:
+                    shard += 1
+                    filled = 0
+        self._planned = True
+
+    def next_block(self, shard: int, epoch: int = 0):
+        """Returns a block ref or None when this epoch is exhausted.
+        A request for epoch N+1 after epoch N drained re-executes the
+        dataset (reference DataIterator: each iter_batches pass re-runs
+        the pipeline) — shards advance epochs in lockstep under DDP."""
         with self._lock:
+            if (epoch > self._epoch and self._done
+                    and not any(self._queues)):
+                self._epoch = epoch
+                self._iter = self._dataset.iter_internal_ref_bundles()
+                self._done = False
+                self._planned = False
+            elif epoch < self._epoch:
+                return None  # straggler from a finished epoch
+            if self.equal and not self._planned:
+                self._plan_equal_locked()
             while not self._queues[shard] and not self._done:
                 try:
                     ref = next(self._iter)
@@ -111,9 +171,11 @@ class DataIterator:
     def _iter_blocks(self) -> Iterator[Any]:
         import ant_ray_amd as ray
 
+        epoch = getattr(self, "_epoch", 0)
         while True:
-            ref = ray.get(self._coord.next_block.remote(self._shard))
+            ref = ray.get(self._coord.next_block.remote(self._shard, epoch))
             if ref is None:
+                self._epoch = epoch + 1  # next pass re-executes
                 return
             yield ray.get(ref)
 
