@@ -1440,6 +1440,19 @@ class CoreWorker:
         n_returns = opts.get("num_returns", 1)
         streaming = n_returns in ("streaming", "dynamic")
         st = self._get_actor_state(actor_id)
+        mpc = int(opts.get("max_pending_calls",
+                           getattr(st, "max_pending_calls", -1)) or -1)
+        if mpc > 0:
+            with st.lock:
+                outstanding = len(st.pending) + len(st.inflight)
+            if outstanding >= mpc:
+                from ant_ray_amd.exceptions import (
+                    PendingCallsLimitExceeded,
+                )
+
+                raise PendingCallsLimitExceeded(
+                    f"actor {actor_id.hex()[:8]} has {outstanding} pending "
+                    f"calls (max_pending_calls={mpc})")
         with st.lock:
             st.seq += 1
             seq = st.seq

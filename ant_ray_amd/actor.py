@@ -212,4 +212,6 @@ class ActorClass:
         st = cw._get_actor_state(actor_id)
         st.is_owner = True
         st.detached = opts.get("lifetime") == "detached"
+        if opts.get("max_pending_calls"):
+            st.max_pending_calls = int(opts["max_pending_calls"])
         return handle

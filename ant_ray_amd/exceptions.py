@@ -103,3 +103,9 @@ class RaySystemError(RayError):
 
 class CrossLanguageError(RayError):
     pass
+
+
+class PendingCallsLimitExceeded(RayError):
+    """Raised when an actor handle with max_pending_calls set already has
+    that many calls outstanding (parity: ray.exceptions
+    .PendingCallsLimitExceeded)."""

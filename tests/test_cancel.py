@@ -270,3 +270,23 @@ def test_actor_method_retry_exceptions(ray_mod):
 
     f = Flaky.remote()
     assert ray.get(f.work.remote(), timeout=60) == "ok after 3"
+
+
+def test_max_pending_calls(ray_mod):
+    """max_pending_calls: client-side backpressure raises
+    PendingCallsLimitExceeded instead of queueing without bound."""
+    ray = ray_mod
+    from ant_ray_amd.exceptions import PendingCallsLimitExceeded
+
+    @ray.remote
+    class Slow:
+        def work(self):
+            time.sleep(1.0)
+            return 1
+
+    a = Slow.options(max_pending_calls=3).remote()
+    refs = [a.work.remote() for _ in range(3)]
+    with pytest.raises(PendingCallsLimitExceeded):
+        for _ in range(20):  # the 4th+ submission must trip the limit
+            refs.append(a.work.remote())
+    assert ray.get(refs[:3], timeout=120) == [1, 1, 1]
