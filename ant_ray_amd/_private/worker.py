@@ -171,13 +171,14 @@ class ActorHandleState:
         self.seq = 0
         self.state = "PENDING"
         self.death_cause = ""
-        self.lock = threading.Lock()
+        self.lock = threading.RLock()  # re-entrant: _fail_task runs under it
         self.pending: List[dict] = []  # queued before addr known
         self.inflight: Dict[int, dict] = {}  # seq -> task payload (for errors)
         self.handle_count = 0
         self.is_owner = False
         self.detached = False
         self.restart_gen = 0  # mirrors GCS num_restarts; seq resets with it
+        self.outstanding = 0  # client-side in-flight count (max_pending_calls)
         # task-id counter: NEVER resets (unlike seq) — a post-restart call
         # must not reuse a pre-restart call's deterministic task id, or
         # ray.get would return the old call's cached result
@@ -1322,6 +1323,11 @@ class CoreWorker:
 
     def _fail_task(self, payload, exc: BaseException):
         task_id = payload["task_id"]
+        if payload.get("type") == "actor_task":
+            st = self._actors.get(payload.get("actor_id"))
+            if st is not None:
+                with st.lock:
+                    st.outstanding = max(0, st.outstanding - 1)
         st = self._streams.get(task_id)
         if st is not None:
             with st.cv:
@@ -1444,7 +1450,7 @@ class CoreWorker:
                            getattr(st, "max_pending_calls", -1)) or -1)
         if mpc > 0:
             with st.lock:
-                outstanding = len(st.pending) + len(st.inflight)
+                outstanding = st.outstanding
             if outstanding >= mpc:
                 from ant_ray_amd.exceptions import (
                     PendingCallsLimitExceeded,
@@ -1454,6 +1460,7 @@ class CoreWorker:
                     f"actor {actor_id.hex()[:8]} has {outstanding} pending "
                     f"calls (max_pending_calls={mpc})")
         with st.lock:
+            st.outstanding += 1
             st.seq += 1
             seq = st.seq
             st.task_counter += 1
@@ -1568,6 +1575,7 @@ class CoreWorker:
             self._handle_task_reply(payload, reply)
             with st.lock:
                 st.inflight.pop(payload["seq"], None)
+                st.outstanding = max(0, st.outstanding - 1)
         except Exception:
             # connection to actor lost: wait for GCS verdict (restart/dead)
             with st.lock:
