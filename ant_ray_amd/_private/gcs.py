@@ -185,7 +185,10 @@ class GcsServer:
     async def _persist_later(self):
         await asyncio.sleep(0.1)
         try:
-            self._persist_now()
+            # serialize + write OFF the event loop: with thousands of
+            # actors the snapshot is tens of MB and would stall heartbeats
+            await asyncio.get_running_loop().run_in_executor(
+                None, self._persist_now)
         except Exception:
             logger.exception("GCS table persist failed")
 
