@@ -185,16 +185,20 @@ class GcsServer:
     async def _persist_later(self):
         await asyncio.sleep(0.1)
         try:
-            # serialize + write OFF the event loop: with thousands of
-            # actors the snapshot is tens of MB and would stall heartbeats
+            # build the snapshot ON the loop (cheap dict-of-refs walk, no
+            # concurrent-mutation hazard), then pack+write OFF the loop:
+            # with thousands of actors the bytes are tens of MB and would
+            # stall heartbeats
+            snap = self._snapshot()
             await asyncio.get_running_loop().run_in_executor(
-                None, self._persist_now)
+                None, self._write_snapshot, snap)
         except Exception:
             logger.exception("GCS table persist failed")
 
     def _persist_now(self):
-        import msgpack
+        self._write_snapshot(self._snapshot())
 
+    def _snapshot(self):
         snap = {
             "kv": {ns: dict(d) for ns, d in self.kv.items()},
             "named_actors": [[list(k), v] for k, v in self.named_actors.items()],
@@ -225,6 +229,11 @@ class GcsServer:
                 for g in self.pgs.values()
             ],
         }
+        return snap
+
+    def _write_snapshot(self, snap):
+        import msgpack
+
         tmp = self.persist_path + ".tmp"
         with open(tmp, "wb") as f:
             f.write(msgpack.packb(snap, use_bin_type=True))
