@@ -151,6 +151,13 @@ class ObjectRefGenerator:
         """Drain-and-count convenience (non-standard helper)."""
         return list(self)
 
+    def __del__(self):
+        # an abandoned generator must not leak its stream state
+        try:
+            self._worker._streams.pop(self._task_id, None)
+        except Exception:
+            pass
+
 
 class LeasedWorker:
     def __init__(self, worker_id, addr, node_id, conn):
