@@ -80,6 +80,11 @@ def test_gcs_restart_with_persistence(tmp_path):
 
         internal_kv._internal_kv_put(b"durable_key", b"durable_value")
 
+        from ant_ray_amd.util import virtual_cluster as vc
+
+        node_id = [n["NodeID"] for n in ray.nodes()][0]
+        vc.create_or_update_virtual_cluster("vc_persist", node_ids=[node_id])
+
         @ray.remote
         class Keeper:
             def __init__(self):
@@ -124,6 +129,11 @@ def test_gcs_restart_with_persistence(tmp_path):
                     raise
                 time.sleep(0.5)
         assert internal_kv._internal_kv_get(b"durable_key") == b"durable_value"
+
+        from ant_ray_amd.util import virtual_cluster as vc
+
+        vcs = {v["virtual_cluster_id"] for v in vc.list_virtual_clusters()}
+        assert "vc_persist" in vcs, "virtual cluster lost across GCS restart"
 
         deadline = time.time() + 60
         nodes = []
