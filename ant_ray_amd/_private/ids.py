@@ -117,9 +117,14 @@ class TaskID(BaseID):
         return cls(cls._prefix + c.to_bytes(6, "little") + job_id.binary())
 
     @classmethod
-    def for_actor_task(cls, actor_id: ActorID, seq: int):
+    def for_actor_task(cls, actor_id: ActorID, seq: int,
+                       caller: bytes = b""):
+        # caller MUST be mixed in: two processes each count their own
+        # calls from 1, and colliding task ids would collide the
+        # deterministic return-object ids in the node-shared shm store
         h = hashlib.blake2b(
-            actor_id.binary() + seq.to_bytes(8, "little"), digest_size=12
+            actor_id.binary() + caller + seq.to_bytes(8, "little"),
+            digest_size=12,
         ).digest()
         return cls(h + actor_id.job_id().binary())
 

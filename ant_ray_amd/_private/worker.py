@@ -1472,7 +1472,8 @@ class CoreWorker:
             seq = st.seq
             st.task_counter += 1
             tc = st.task_counter
-        task_id = TaskID.for_actor_task(ActorID(actor_id), tc).binary()
+        task_id = TaskID.for_actor_task(ActorID(actor_id), tc,
+                                        caller=self.worker_id).binary()
         if streaming:
             n_returns = 0
             self._streams[task_id] = _StreamState()

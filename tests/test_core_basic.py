@@ -335,3 +335,42 @@ def test_kill_actor_while_scheduling(ray_start_regular):
             break
         time.sleep(0.2)
     assert dead, "killed-while-scheduling actor should transition to DEAD"
+
+
+def test_actor_task_ids_unique_across_callers(ray_start_regular):
+    """Two different callers of one actor must not collide return-object
+    ids: big (shm-resident) results from distinct callers stay distinct
+    even at identical per-caller call counts."""
+    import numpy as np
+
+    ray = ray_start_regular
+
+    @ray.remote
+    class Producer:
+        def big(self, fill):
+            return np.full(200_000, float(fill))  # 1.6 MB -> shm
+
+    @ray.remote
+    class CallerA:
+        def run(self, p):
+            import numpy as _np
+
+            out = ray.get(p.big.remote(1.0))
+            return float(out.sum())
+
+    @ray.remote
+    class CallerB:
+        def run(self, p):
+            import numpy as _np
+
+            out = ray.get(p.big.remote(2.0))
+            return float(out.sum())
+
+    p = Producer.remote()
+    a, b = CallerA.remote(), CallerB.remote()
+    # same call-count (1) from both callers, nearly simultaneous
+    ra = a.run.remote(p)
+    rb = b.run.remote(p)
+    sa, sb = ray.get([ra, rb], timeout=120)
+    assert sa == 200_000 * 1.0, sa
+    assert sb == 200_000 * 2.0, sb
