@@ -142,8 +142,10 @@ def test_pbt_exploits_and_perturbs(ray_mod, tmp_path_factory):
         lr = config["lr"]
         import time as _t
 
-        for it in range(start_it + 1, 21):
-            _t.sleep(0.15)  # let the Tuner poll between iterations
+        for it in range(start_it + 1, 31):
+            _t.sleep(0.3)  # let the Tuner poll between iterations even
+            # when the whole machine is under load (a trial finishing
+            # inside ONE poll window can never be perturbed)
             base += lr  # good lr climbs faster
             with tempfile.TemporaryDirectory() as d:
                 with open(os.path.join(d, "state.json"), "w") as f:
@@ -171,7 +173,7 @@ def test_pbt_exploits_and_perturbs(ray_mod, tmp_path_factory):
     # the bad-lr trial must have been perturbed onto a useful lr: its
     # final score is far above what 20 iterations of lr=0.001 (0.02) give
     assert scores[0] > 1.0, scores
-    assert grid.get_best_result(metric="score", mode="max").metrics["score"] >= 19.0
+    assert grid.get_best_result(metric="score", mode="max").metrics["score"] >= 29.0
 
 
 def test_tuner_restore(ray_mod, tmp_path_factory):
