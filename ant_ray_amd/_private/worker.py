@@ -963,6 +963,12 @@ class CoreWorker:
 
             self._fn_cache[fn_id] = cloudpickle.dumps(fn)
             self._pushed_fns.add(fn_id)
+            # closure-capturing remote fns (e.g. Data exchanges) mint a
+            # fresh fn_id per call site: bound the pickled-blob cache
+            while len(self._fn_cache) > 4096:
+                old_id = next(iter(self._fn_cache))
+                self._fn_cache.pop(old_id, None)
+                self._pushed_fns.discard(old_id)
         payload = {
             "type": "normal",
             "task_id": task_id,
@@ -1669,6 +1675,11 @@ class CoreWorker:
     async def _get_worker_conn_async_cached(self, addr):
         conn = self._worker_conns.get(addr)
         if conn is None or conn.closed:
+            if len(self._worker_conns) > 512:
+                # drop dead entries: worker churn mints new addrs forever
+                for k in [k for k, c in self._worker_conns.items()
+                          if c.closed]:
+                    self._worker_conns.pop(k, None)
             conn = await protocol.connect(addr, self._handle_rpc, name=f"->worker{addr}")
             self._worker_conns[addr] = conn
         return conn
