@@ -1715,6 +1715,10 @@ class CoreWorker:
                 except OSError:
                     pass
             self.store.free([oid])
+            # bookkeeping tied to this oid goes with it
+            self._object_locations.pop(oid, None)
+            self._lineage.pop(oid, None)
+            self._task_of_oid.pop(oid, None)
         except Exception:
             pass
 
