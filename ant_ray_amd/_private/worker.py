@@ -783,6 +783,30 @@ class CoreWorker:
             if self.store.shm is None:
                 time.sleep(0.05)
 
+    def _prefetch_remote(self, refs):
+        """ray.wait(fetch_local=True): ready objects held on another node
+        start pulling into local shm in the background so the follow-up
+        ray.get is a local read (reference wait semantics)."""
+        if not hasattr(self, "_prefetching"):
+            self._prefetching = set()
+        for ref in refs:
+            oid = ref.binary()
+            loc = self._object_locations.get(oid)
+            if (loc is None or tuple(loc) == self.addr
+                    or self.store.shm.contains(oid)
+                    or oid in self._prefetching):
+                continue
+            self._prefetching.add(oid)
+
+            def run(oid=oid, loc=loc):
+                try:
+                    self._try_pull(oid, loc)
+                finally:
+                    self._prefetching.discard(oid)
+
+            threading.Thread(target=run, daemon=True,
+                             name="prefetch").start()
+
     def _deserialize_buffer(self, buf, meta: bytes):
         value = serialization.deserialize(memoryview(buf), meta)
         if meta in (serialization.META_ERROR, serialization.META_ACTOR_DIED):
@@ -934,6 +958,8 @@ class CoreWorker:
             still: List[ObjectRef] = []
             _scan(pending, ready, still)
             pending = still
+        if fetch_local and shm is not None:
+            self._prefetch_remote(ready)
         return ready, pending
 
     # ============================================================ task submit

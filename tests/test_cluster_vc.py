@@ -339,3 +339,34 @@ def test_graceful_drain(cluster):
     assert ray.get(ref, timeout=60) == "finished"
     with open(attempts) as f:
         assert len(f.read().splitlines()) == 1  # ran exactly once
+
+
+def test_wait_fetch_local_prefetches(cluster):
+    """ray.wait(fetch_local=True) pulls ready remote objects into local
+    shm in the background (reference wait semantics)."""
+    import numpy as np
+
+    import ant_ray_amd as ray
+    from ant_ray_amd._private.worker import global_worker
+
+    cluster.connect()
+    cluster.add_node(num_cpus=2, resources={"src": 1})
+    deadline = time.time() + 30
+    while time.time() < deadline and not ray.cluster_resources().get("src"):
+        time.sleep(0.2)
+
+    @ray.remote(num_cpus=1, resources={"src": 0.1})
+    def make():
+        return np.ones(1024 * 1024)  # 8 MB task output, stays remote
+
+    inner = make.remote()
+    ready, _ = ray.wait([inner], timeout=60, fetch_local=True)
+    assert ready == [inner]
+    cw = global_worker.core_worker
+    deadline = time.time() + 60
+    while time.time() < deadline:
+        if cw.store.shm.contains(inner.binary()):
+            break
+        time.sleep(0.2)
+    assert cw.store.shm.contains(inner.binary()), "prefetch never landed"
+    assert float(ray.get(inner, timeout=30).sum()) == 1024 * 1024
