@@ -451,6 +451,22 @@ class CoreWorker:
             return {"cancelled": False}
         return self.executor.cancel_task(p["task_id"], p.get("force", False))
 
+    async def rpc_probe_object(self, conn, p):
+        """Does this object exist yet? (borrowed-ref ray.wait readiness;
+        parity: wait on borrowed refs asks the owner in the reference's
+        ownership protocol)."""
+        oid = p["oid"]
+        found, v = self.store.memory.get_now(oid)
+        if found and not isinstance(v, _ErrorResult):
+            return {"exists": True}
+        if found:
+            return {"exists": True}  # error results are 'ready' too
+        if self.store.shm is not None and self.store.shm.contains(oid):
+            return {"exists": True}
+        if oid in self._spilled:
+            return {"exists": True}
+        return {"exists": self.store.memory.is_pending(oid) and False}
+
     async def rpc_stream_item(self, conn, p):
         """A streaming task produced its next yielded value (executor
         task_executor._stream_results); store it and wake the
@@ -939,11 +955,44 @@ class CoreWorker:
                     # IN_PLASMA = the owner saw the task complete; that IS
                     # ready (reference semantics: wait readiness means the
                     # object exists, wherever it lives — fetch happens at
-                    # ray.get). Only a truly unknown oid consults shm.
+                    # ray.get). Only a truly unknown oid consults shm,
+                    # then the borrowed-ready cache.
                     ok = v is not None and not isinstance(v, _MemPending)
                     if not ok and shm is not None and v is None:
                         ok = shm.contains(oid)
+                    if not ok and v is None and oid in borrow_cache:
+                        ok = True
                     (ready if ok else pending).append(ref)
+
+        # borrowed refs (owner is another worker): the local stores know
+        # nothing — ask the owner, throttled, and cache positives
+        borrow_cache = getattr(self, "_borrow_ready", None)
+        if borrow_cache is None:
+            borrow_cache = self._borrow_ready = set()
+            self._borrow_last_probe = {}
+
+        def _probe_borrowed(candidates):
+            now = time.monotonic()
+            for ref in candidates:
+                oid = ref.binary()
+                if oid in borrow_cache or not ref.owner_addr:
+                    continue
+                owner = tuple(ref.owner_addr)
+                if owner == self.addr:
+                    continue
+                if now - self._borrow_last_probe.get(oid, 0.0) < 0.2:
+                    continue
+                self._borrow_last_probe[oid] = now
+                try:
+                    conn = self._get_worker_conn(owner)
+                    r = self.io.run(conn.call("probe_object", {"oid": oid},
+                                              timeout=5), timeout=6)
+                    if r.get("exists"):
+                        borrow_cache.add(oid)
+                        while len(borrow_cache) > 10000:
+                            borrow_cache.pop()
+                except Exception:
+                    pass
 
         ready: List[ObjectRef] = []
         pending: List[ObjectRef] = []
@@ -956,6 +1005,7 @@ class CoreWorker:
             with mem._cv:
                 mem._cv.wait(0.005)
             still: List[ObjectRef] = []
+            _probe_borrowed(pending)
             _scan(pending, ready, still)
             pending = still
         if fetch_local and shm is not None:

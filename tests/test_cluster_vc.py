@@ -370,3 +370,31 @@ def test_wait_fetch_local_prefetches(cluster):
         time.sleep(0.2)
     assert cw.store.shm.contains(inner.binary()), "prefetch never landed"
     assert float(ray.get(inner, timeout=30).sum()) == 1024 * 1024
+
+
+def test_wait_on_borrowed_ref(cluster):
+    """ray.wait works on BORROWED refs (owned by a remote worker): the
+    waiter probes the owner for existence."""
+    import numpy as np
+
+    import ant_ray_amd as ray
+
+    cluster.connect()
+    cluster.add_node(num_cpus=2, resources={"own": 1})
+    deadline = time.time() + 30
+    while time.time() < deadline and not ray.cluster_resources().get("own"):
+        time.sleep(0.2)
+
+    @ray.remote(num_cpus=1, resources={"own": 0.1})
+    def owner_task():
+        import time as _t
+
+        import ant_ray_amd as ray2
+
+        slow = ray2.put(np.ones(700_000))  # 5.6 MB owned by this worker
+        return [slow]
+
+    inner = ray.get(owner_task.remote(), timeout=60)[0]
+    ready, not_ready = ray.wait([inner], timeout=60)
+    assert ready == [inner], (ready, not_ready)
+    assert float(ray.get(inner, timeout=60).sum()) == 700_000
