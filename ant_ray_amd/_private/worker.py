@@ -983,6 +983,9 @@ class CoreWorker:
                 if now - self._borrow_last_probe.get(oid, 0.0) < 0.2:
                     continue
                 self._borrow_last_probe[oid] = now
+                while len(self._borrow_last_probe) > 20000:
+                    self._borrow_last_probe.pop(
+                        next(iter(self._borrow_last_probe)))
                 try:
                     conn = self._get_worker_conn(owner)
                     r = self.io.run(conn.call("probe_object", {"oid": oid},
