@@ -167,8 +167,37 @@ class GcsServer:
                 logger.exception("GCS table restore failed; starting fresh")
         self._server, self.port = await protocol.serve(self._handle, host, port)
         asyncio.get_running_loop().create_task(self._health_loop())
+        if self.actors:
+            asyncio.get_running_loop().create_task(
+                self._verify_restored_actors())
         logger.info("GCS listening on %s:%s", host, self.port)
         return self.port
+
+    async def _verify_restored_actors(self):
+        """A worker that died DURING a GCS outage was reported to the
+        dead GCS: restored ALIVE actors must be liveness-checked or they
+        stay reachable-on-paper forever. Ping each restored actor's
+        worker; unreachable -> normal worker-death handling (restart per
+        max_restarts or DEAD)."""
+        await asyncio.sleep(3.0)  # give raylets/workers time to reconnect
+        for info in list(self.actors.values()):
+            if info.state != ALIVE or not info.addr:
+                continue
+            alive = False
+            for _ in range(3):
+                try:
+                    conn = await self._get_worker_conn(tuple(info.addr))
+                    r = await conn.call("ping", {}, timeout=5)
+                    alive = bool(r.get("ok"))
+                    break
+                except Exception:
+                    await asyncio.sleep(1.0)
+            if not alive:
+                logger.warning("restored actor %s unreachable at %s; "
+                               "declaring its worker dead",
+                               info.actor_id.hex()[:8], info.addr)
+                await self._on_actor_worker_died(
+                    info, "worker died while the GCS was down")
 
     # ------------------------------------------------------- fault tolerance
     def _persist_soon(self):

@@ -267,3 +267,73 @@ def test_driver_death_cleans_up_actors(tmp_path):
         except Exception:
             pass
         c.shutdown()
+
+
+def test_restored_actor_with_dead_worker_recovers(tmp_path):
+    """An actor whose WORKER dies while the GCS is down must be detected
+    by the restarted GCS's liveness verification and restarted (its
+    max_restarts budget), not left ALIVE-with-stale-addr forever."""
+    port = _free_port()
+    persist = str(tmp_path / "gcs.msgpack")
+    session = str(tmp_path / "session")
+    os.makedirs(session, exist_ok=True)
+
+    gcs = _spawn_gcs(port, persist)
+    raylet = None
+    try:
+        assert _wait_port(port)
+        raylet = subprocess.Popen(
+            [sys.executable, "-m", "ant_ray_amd._private.raylet",
+             "--gcs", f"127.0.0.1:{port}", "--num-cpus", "2",
+             "--num-gpus", "0", "--store-path", str(tmp_path / "store"),
+             "--store-capacity", str(256 * 1024 * 1024),
+             "--session-dir", session],
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
+
+        import ant_ray_amd as ray
+
+        if ray.is_initialized():
+            ray.shutdown()
+        deadline = time.time() + 30
+        while True:
+            try:
+                ray.init(address=f"127.0.0.1:{port}")
+                break
+            except Exception:
+                if time.time() > deadline:
+                    raise
+                time.sleep(0.5)
+
+        @ray.remote(max_restarts=2)
+        class Phoenix:
+            def pid(self):
+                import os as _os
+
+                return _os.getpid()
+
+        p = Phoenix.options(name="phoenix", lifetime="detached").remote()
+        pid1 = ray.get(p.pid.remote(), timeout=60)
+        time.sleep(0.5)  # snapshot lands
+
+        gcs.send_signal(signal.SIGKILL)
+        gcs.wait(timeout=10)
+        os.kill(pid1, signal.SIGKILL)  # the actor dies DURING the outage
+        time.sleep(1.0)
+        gcs = _spawn_gcs(port, persist)
+        assert _wait_port(port)
+
+        deadline = time.time() + 90
+        pid2 = None
+        while time.time() < deadline:
+            try:
+                p2 = ray.get_actor("phoenix")
+                pid2 = ray.get(p2.pid.remote(), timeout=20)
+                break
+            except Exception:
+                time.sleep(1.0)
+        assert pid2 is not None and pid2 != pid1, (pid1, pid2)
+        ray.shutdown()
+    finally:
+        for proc in (gcs, raylet):
+            if proc is not None and proc.poll() is None:
+                proc.kill()
