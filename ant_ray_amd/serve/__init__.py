@@ -229,6 +229,8 @@ def _collect_deployments(app: Application, out: Dict[str, dict],
         "autoscaling_config": auto,
         "user_config": d.user_config,
         "is_streaming": is_streaming,
+        "graceful_shutdown_timeout_s": d.graceful_shutdown_timeout_s,
+        "health_check_period_s": d.health_check_period_s,
     }
 
 

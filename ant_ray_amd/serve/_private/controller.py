@@ -67,6 +67,23 @@ class ServeController:
             self._bg = loop.create_task(self._reconcile_loop())
             loop.create_task(self._recover_state_from_checkpoint())
 
+    async def _graceful_stop_replica(self, h, timeout_s: float):
+        """Drain in-flight requests before killing a removed replica
+        (parity: reference graceful_shutdown_timeout_s). The replica is
+        already out of the routing set, so its queue only shrinks."""
+        deadline = asyncio.get_event_loop().time() + max(timeout_s, 0.0)
+        while asyncio.get_event_loop().time() < deadline:
+            try:
+                if await _aw(h.num_ongoing_requests.remote()) == 0:
+                    break
+            except Exception:
+                break
+            await asyncio.sleep(0.2)
+        try:
+            ray.kill(h)
+        except Exception:
+            pass
+
     # ------------------------------------------------- crash recovery
     def _checkpoint_state(self):
         """Persist the app configs to the GCS KV (parity: the reference
@@ -178,10 +195,10 @@ class ServeController:
                         dep_name)
         while len(dr.replicas) > target:
             h = dr.replicas.pop()
-            try:
-                ray.kill(h)
-            except Exception:
-                pass
+            asyncio.get_event_loop().create_task(
+                self._graceful_stop_replica(
+                    h, float(d.get("graceful_shutdown_timeout_s", 20.0)
+                             if isinstance(d, dict) else 20.0)))
         # wait until new replicas construct (first health check) — bounded:
         # an unplaceable replica (cluster out of CPUs/GPUs) must surface as
         # a deploy error, not an indefinite hang (parity: the reference's

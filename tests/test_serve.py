@@ -532,3 +532,25 @@ def test_multiplexed_model_affinity_routing(serve_mod):
     assert len(pids_a) == 1, "model 'a' requests should stick to one replica"
     assert len(pids_b) == 1
     serve.delete("muxaff")
+
+
+def test_graceful_replica_shutdown(serve_mod):
+    """Scaling down / deleting drains in-flight requests before killing
+    replicas: a slow request completes across the delete."""
+    serve = serve_mod
+    for app_name in list(serve.status().get("applications", {})):
+        serve.delete(app_name)
+
+    @serve.deployment(graceful_shutdown_timeout_s=30)
+    class SlowFinish:
+        async def __call__(self, x):
+            import asyncio as _a
+
+            await _a.sleep(2.0)
+            return f"done-{x}"
+
+    h = serve.run(SlowFinish.bind(), name="drainapp", route_prefix="/drain")
+    resp = h.remote(1)  # in flight...
+    time.sleep(0.3)
+    serve.delete("drainapp", _blocking=False)  # removal starts mid-request
+    assert resp.result(timeout_s=60) == "done-1"
