@@ -288,15 +288,33 @@ class ServeController:
     async def _reconcile_once(self):
         for app, deps in self.apps.items():
             for dep_name, dr in deps.items():
-                # replace dead replicas
+                # liveness + periodic USER health check (reference
+                # health_check_period_s: an unhealthy replica — raise or
+                # falsy return from check_health — is replaced like a
+                # dead one)
+                period = float(dr.cfg.get("health_check_period_s", 10.0))
+                due = (time.monotonic()
+                       - getattr(dr, "last_health_check", 0.0)) >= period
+                if due:
+                    dr.last_health_check = time.monotonic()
                 alive = []
                 for h in dr.replicas:
                     try:
-                        await _aw(h.num_ongoing_requests.remote())
+                        if due:
+                            healthy = await _aw(h.check_health.remote())
+                            if healthy is False:
+                                raise RuntimeError("check_health falsy")
+                        else:
+                            await _aw(h.num_ongoing_requests.remote())
                         alive.append(h)
                     except Exception:
-                        logger.warning("replica of %s/%s died; replacing",
-                                       app, dep_name)
+                        logger.warning(
+                            "replica of %s/%s dead or unhealthy; replacing",
+                            app, dep_name)
+                        try:
+                            ray.kill(h)
+                        except Exception:
+                            pass
                 dr.replicas = alive
                 if len(dr.replicas) < dr.target:
                     await self._scale_to(dr, dr.target, app, dep_name)

@@ -554,3 +554,51 @@ def test_graceful_replica_shutdown(serve_mod):
     time.sleep(0.3)
     serve.delete("drainapp", _blocking=False)  # removal starts mid-request
     assert resp.result(timeout_s=60) == "done-1"
+
+
+def test_unhealthy_replica_replaced(serve_mod):
+    """A replica whose user check_health starts failing is killed and
+    replaced by the controller's periodic health loop."""
+    import os
+
+    serve = serve_mod
+    for app_name in list(serve.status().get("applications", {})):
+        serve.delete(app_name)
+
+    poison = "/tmp/antray_unhealthy_marker"
+    try:
+        os.unlink(poison)
+    except FileNotFoundError:
+        pass
+
+    @serve.deployment(health_check_period_s=1.0)
+    class Fragile:
+        def __call__(self, x):
+            import os as _os
+
+            return _os.getpid()
+
+        def check_health(self):
+            import os as _os
+
+            if _os.path.exists(poison):
+                raise RuntimeError("simulated sickness")
+            return True
+
+    h = serve.run(Fragile.bind(), name="fragile", route_prefix="/fragile")
+    pid1 = h.remote(1).result(timeout_s=60)
+    open(poison, "w").close()  # every current replica now reports sick
+    time.sleep(2.5)
+    os.unlink(poison)  # replacements come up healthy
+    deadline = time.time() + 60
+    pid2 = None
+    while time.time() < deadline:
+        try:
+            pid2 = h.remote(1).result(timeout_s=20)
+            if pid2 != pid1:
+                break
+        except Exception:
+            pass
+        time.sleep(0.5)
+    assert pid2 is not None and pid2 != pid1, (pid1, pid2)
+    serve.delete("fragile")
