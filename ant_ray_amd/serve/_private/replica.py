@@ -70,12 +70,15 @@ class Replica:
             self._callable.reconfigure(user_config)
         return True
 
-    def check_health(self):
+    async def check_health(self):
         fn = getattr(self._callable, "check_health", None)
         if fn is not None:
             r = fn()
             if inspect.iscoroutine(r):
-                return asyncio.get_event_loop().run_until_complete(r)
+                # run_until_complete would blow up inside the actor's
+                # already-running loop; await instead
+                return await r
+            return r if r is not None else True
         return True
 
     def num_ongoing_requests(self) -> int:

@@ -602,3 +602,30 @@ def test_unhealthy_replica_replaced(serve_mod):
         time.sleep(0.5)
     assert pid2 is not None and pid2 != pid1, (pid1, pid2)
     serve.delete("fragile")
+
+
+def test_async_check_health(serve_mod):
+    """An ASYNC user check_health must be awaited, not run_until_complete
+    (which explodes inside the replica's running loop)."""
+    serve = serve_mod
+    for app_name in list(serve.status().get("applications", {})):
+        serve.delete(app_name)
+
+    @serve.deployment(health_check_period_s=0.5)
+    class AsyncHealthy:
+        async def check_health(self):
+            import asyncio as _a
+
+            await _a.sleep(0.01)
+            return True
+
+        def __call__(self, x):
+            return x + 1
+
+    h = serve.run(AsyncHealthy.bind(), name="ahealth", route_prefix="/ah")
+    assert h.remote(1).result(timeout_s=60) == 2
+    time.sleep(2.0)  # several health periods pass without replica churn
+    assert h.remote(2).result(timeout_s=60) == 3
+    st = serve.status()["applications"]["ahealth"]["deployments"]
+    assert st["AsyncHealthy"]["replicas"] == 1
+    serve.delete("ahealth")
