@@ -16,3 +16,7 @@ def __getattr__(name):
     if name == "ActorPool":
         return importlib.import_module("ant_ray_amd.util.actor_pool").ActorPool
     raise AttributeError(name)
+
+from ant_ray_amd.util.inspect_serializability import (  # noqa: F401,E402
+    inspect_serializability,
+)
