@@ -51,6 +51,13 @@ def _check(obj, path: str, depth: int, failures) -> bool:
             if not _try(inner):
                 found_cause = True
                 _check(inner, f"{path}.<closure>.{nm}", depth - 1, failures)
+    code = getattr(obj, "__code__", None)
+    gl = getattr(obj, "__globals__", None)
+    if code is not None and isinstance(gl, dict):
+        for nm in code.co_names:
+            if nm in gl and not _try(gl[nm]):
+                found_cause = True
+                _check(gl[nm], f"{path}.<global>.{nm}", depth - 1, failures)
     d = getattr(obj, "__dict__", None)
     if isinstance(d, dict):
         for k, v in list(d.items())[:64]:
