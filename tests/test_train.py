@@ -429,3 +429,45 @@ def test_user_callbacks(ray_mod, tmp_path_factory):
     assert kinds[0] == "start" and kinds[-1] == "shutdown"
     assert kinds.count("ckpt") == 2
     assert ("report", 1, 0) in events
+
+
+def test_accelerate_inside_torchtrainer(ray_mod, tmp_path_factory):
+    """HF accelerate works inside a TorchTrainer train_fn (reference
+    train.accelerate posture: accelerate rides the torch process group
+    Ray Train sets up)."""
+    import pytest
+
+    pytest.importorskip("accelerate")
+    storage = str(tmp_path_factory.mktemp("acc"))
+
+    def train_fn(config):
+        import torch
+        import torch.nn as nn
+        from accelerate import Accelerator
+
+        from ant_ray_amd import train
+
+        acc = Accelerator(cpu=True)
+        model = nn.Linear(4, 1)
+        opt = torch.optim.SGD(model.parameters(), lr=0.1)
+        model, opt = acc.prepare(model, opt)
+        x = torch.randn(16, 4)
+        y = torch.randn(16, 1)
+        for _ in range(3):
+            loss = ((model(x) - y) ** 2).mean()
+            acc.backward(loss)
+            opt.step()
+            opt.zero_grad()
+        train.report({"loss": float(loss),
+                      "procs": acc.num_processes,
+                      "rank": acc.process_index})
+
+    from ant_ray_amd.train import RunConfig, ScalingConfig
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    res = TorchTrainer(
+        train_fn, scaling_config=ScalingConfig(num_workers=2),
+        run_config=RunConfig(name="acc", storage_path=storage),
+    ).fit()
+    assert res.error is None, res.error
+    assert res.metrics["procs"] == 2  # accelerate picked up the group

@@ -155,3 +155,19 @@ def test_multiprocessing_pool(ray_mod):
         assert ar.get(timeout=60) == 81
         assert sorted(p.imap_unordered(sq, range(5))) == [0, 1, 4, 9, 16]
         assert list(p.imap(sq, range(5))) == [0, 1, 4, 9, 16]
+
+
+def test_inspect_serializability():
+    import threading
+
+    from ant_ray_amd.util import inspect_serializability
+
+    lk = threading.Lock()
+
+    def bad():
+        return lk
+
+    ok, fails = inspect_serializability(bad)
+    assert not ok and any("lk" in f for f in fails)
+    ok, fails = inspect_serializability({"x": 1})
+    assert ok and not fails
