@@ -33,8 +33,18 @@ if __name__ == "__main__":
     ray.init()
     serve.run(Summarizer.bind(Tokenizer.bind()), name="demo",
               route_prefix="/summarize")
-    print("serving on :8000/summarize — ctrl-c to exit")
-    import time
+    import sys
+    import urllib.request
 
-    while True:
-        time.sleep(5)
+    req = urllib.request.Request("http://127.0.0.1:8000/summarize",
+                                 data=b"hello ray serve demo", method="POST")
+    with urllib.request.urlopen(req, timeout=30) as r:
+        print("self-test:", r.read().decode())
+    if "--serve" in sys.argv:
+        print("serving on :8000/summarize — ctrl-c to exit")
+        import time
+
+        while True:
+            time.sleep(5)
+    serve.shutdown()
+    ray.shutdown()
