@@ -43,9 +43,39 @@ class Router:
             raise RuntimeError(
                 f"deployment '{self.deployment}' not found in app '{self.app}'"
             )
+        try:
+            info = ray.get(self._controller().get_deployment_info.remote(
+                self.app, self.deployment))
+            self._max_ongoing = int(
+                (info or {}).get("max_ongoing_requests") or 0) or None
+        except Exception:
+            self._max_ongoing = None
         self._replicas = replicas
         self._ongoing = {i: self._ongoing.get(i, 0) for i in range(len(replicas))}
         self._last_refresh = now
+
+    def _admit(self, i: int, deadline_s: float = 60.0) -> int:
+        """Per-replica max_ongoing_requests backpressure (reference
+        router honors the deployment cap): when the chosen replica is at
+        capacity, spill to the least-loaded one; when EVERY replica is
+        at capacity, wait until one drains."""
+        cap = getattr(self, "_max_ongoing", None)
+        if not cap:
+            return i
+        deadline = time.monotonic() + deadline_s
+        while True:
+            if self._ongoing.get(i, 0) < cap:
+                return i
+            j = min(range(len(self._replicas)),
+                    key=lambda k: self._ongoing.get(k, 0))
+            if self._ongoing.get(j, 0) < cap:
+                return j
+            if time.monotonic() > deadline:
+                raise RuntimeError(
+                    f"all {len(self._replicas)} replicas of "
+                    f"{self.app}/{self.deployment} at max_ongoing_requests="
+                    f"{cap} for {deadline_s:.0f}s")
+            time.sleep(0.005)
 
     def choose_replica(self):
         """Power-of-two-choices on locally tracked in-flight counts."""
@@ -83,6 +113,7 @@ class Router:
                 i = j
         if i is None:
             i, _ = self.choose_replica()
+        i = self._admit(i)
         if multiplexed_model_id is not None:
             self._model_affinity[multiplexed_model_id] = i
             while len(self._model_affinity) > 1024:
@@ -94,6 +125,8 @@ class Router:
     def submit_stream(self, method_name: Optional[str], args, kwargs):
         """Streaming variant: returns (ObjectRefGenerator, replica_index)."""
         i, replica = self.choose_replica()
+        i = self._admit(i)
+        replica = self._replicas[i]
         self._ongoing[i] = self._ongoing.get(i, 0) + 1
         return replica.handle_request_streaming.remote(
             method_name, args, kwargs), i

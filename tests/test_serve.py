@@ -629,3 +629,44 @@ def test_async_check_health(serve_mod):
     st = serve.status()["applications"]["ahealth"]["deployments"]
     assert st["AsyncHealthy"]["replicas"] == 1
     serve.delete("ahealth")
+
+
+def test_router_max_ongoing_backpressure(serve_mod):
+    """max_ongoing_requests is enforced at the ROUTER: with one replica
+    capped at 2, a burst of 6 requests never has more than 2 in flight
+    replica-side."""
+    serve = serve_mod
+    for app_name in list(serve.status().get("applications", {})):
+        serve.delete(app_name)
+
+    @serve.deployment(max_ongoing_requests=2)
+    class Gauged:
+        def __init__(self):
+            self.inflight = 0
+            self.peak = 0
+
+        async def __call__(self, x):
+            import asyncio as _a
+
+            self.inflight += 1
+            self.peak = max(self.peak, self.inflight)
+            await _a.sleep(0.3)
+            self.inflight -= 1
+            return self.peak
+
+    import threading
+
+    h = serve.run(Gauged.bind(), name="gauged", route_prefix="/gauged")
+    results = []
+
+    def call():
+        results.append(h.remote(0).result(timeout_s=60))
+
+    ts = [threading.Thread(target=call) for _ in range(6)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join(timeout=90)
+    assert len(results) == 6
+    assert max(results) <= 2, f"router admitted beyond the cap: {results}"
+    serve.delete("gauged")
