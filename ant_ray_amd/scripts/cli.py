@@ -145,6 +145,15 @@ def status():
     click.echo("Resources")
     click.echo("  total:    " + json.dumps({k: total[k] for k in sorted(total)}))
     click.echo("  available:" + json.dumps({k: round(avail.get(k, 0), 2) for k in sorted(total)}))
+    try:
+        from ant_ray_amd.util.state import _gcs_call
+
+        demands = _gcs_call("pending_resource_demands")
+        if demands:
+            click.echo(f"Pending demands: {len(demands)} "
+                       f"(e.g. {json.dumps(demands[0])})")
+    except Exception:
+        pass
 
 
 # ----------------------------------------------------------------- state list
