@@ -134,6 +134,28 @@ def summarize_objects():
     return store_stats()
 
 
+def get_actor(actor_id: str) -> Optional[Dict[str, Any]]:
+    """Single-actor view by hex id (parity: state.get_actor)."""
+    for a in list_actors(limit=100000):
+        if a["actor_id"] == actor_id:
+            return a
+    return None
+
+
+def get_node(node_id: str) -> Optional[Dict[str, Any]]:
+    for n in list_nodes(limit=100000):
+        if n["node_id"] == node_id:
+            return n
+    return None
+
+
+def get_placement_group(pg_id: str) -> Optional[Dict[str, Any]]:
+    for g in list_placement_groups(limit=100000):
+        if g["placement_group_id"] == pg_id:
+            return g
+    return None
+
+
 def store_stats() -> List[Dict[str, Any]]:
     """Per-node shm object-store usage (backs `ray memory`)."""
     out = []
