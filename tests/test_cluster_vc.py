@@ -398,3 +398,41 @@ def test_wait_on_borrowed_ref(cluster):
     ready, not_ready = ray.wait([inner], timeout=60)
     assert ready == [inner], (ready, not_ready)
     assert float(ray.get(inner, timeout=60).sum()) == 700_000
+
+
+def test_serve_replicas_on_worker_node(cluster):
+    """Serve replicas placed on a second node serve traffic through the
+    head-node proxy (cross-node actor data plane)."""
+    import urllib.request
+
+    import ant_ray_amd as ray
+    from ant_ray_amd import serve
+
+    cluster.connect()
+    cluster.add_node(num_cpus=4, resources={"edge": 1})
+    deadline = time.time() + 30
+    while time.time() < deadline and not ray.cluster_resources().get("edge"):
+        time.sleep(0.2)
+
+    @serve.deployment(num_replicas=2,
+                      ray_actor_options={"num_cpus": 1,
+                                         "resources": {"edge": 0.1}})
+    class EdgeApp:
+        def __call__(self, request):
+            import ant_ray_amd as ray2
+
+            return {"node": ray2.get_runtime_context().get_node_id()}
+
+    serve.run(EdgeApp.bind(), name="edgeapp", route_prefix="/edge")
+    try:
+        port = ray.get(ray.get_actor("SERVE_PROXY_ACTOR").ready.remote())
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/edge", timeout=30).read().decode()
+        import json as _json
+
+        node = _json.loads(body)["node"]
+        edge_nodes = {n["NodeID"] for n in ray.nodes()
+                      if n["Resources"].get("edge")}
+        assert node in edge_nodes, (node, edge_nodes)
+    finally:
+        serve.shutdown()
