@@ -12,6 +12,12 @@ from __future__ import annotations
 from typing import Any, Dict, List, Optional, Tuple
 
 
+def _now():
+    import time
+
+    return time.monotonic()
+
+
 class DAGNode:
     def __init__(self, args: tuple, kwargs: dict):
         self._bound_args = args
@@ -384,7 +390,23 @@ class CompiledDAG:
         self._all_chans = [c for c in chan_of.values() if c is not None]
         if self._in_chan is not None:
             self._all_chans.append(self._in_chan)
+        self._dag_actor_ids = {actor_key(m) for m in order}
         return True
+
+    def _check_dag_actors_alive(self):
+        """Raise if a participating actor died (its resident loop is
+        gone, so channel reads would block forever)."""
+        from ant_ray_amd._private.worker import global_worker
+        from ant_ray_amd.exceptions import ActorDiedError
+
+        cw = global_worker.core_worker
+        for aid in getattr(self, "_dag_actor_ids", ()):  # pubsub-updated
+            st = cw._actors.get(aid)
+            if st is not None and st.state == "DEAD":
+                self.teardown()
+                raise ActorDiedError(
+                    f"compiled DAG actor {aid.hex()[:8]} died: "
+                    f"{st.death_cause}")
 
     # --------------------------------------------------------------- execute
     def execute(self, *args, **kwargs):
@@ -426,8 +448,26 @@ class CompiledDAG:
             while self._read_seq < seq:
                 # read unwrapped: an error result must still advance the
                 # read cursor, else the next get() desynchronizes from the
-                # channel
-                vals = [c.read(timeout, unwrap=False) for c in self._out_chans]
+                # channel. Reads poll in 1s slices so a DEAD actor (whose
+                # loop can never write) surfaces as ActorDiedError instead
+                # of an indefinite hang.
+                vals = []
+                for c in self._out_chans:
+                    deadline = (None if timeout is None
+                                else _now() + timeout)
+                    while True:
+                        slice_t = 1.0
+                        if deadline is not None:
+                            slice_t = min(slice_t, max(0.0, deadline - _now()))
+                        try:
+                            vals.append(c.read(slice_t, unwrap=False))
+                            break
+                        except Exception as e:
+                            if "timed out" not in str(e):
+                                raise
+                            self._check_dag_actors_alive()
+                            if deadline is not None and _now() >= deadline:
+                                raise
                 self._read_seq += 1
                 self._results[self._read_seq] = (
                     vals if self._multi else vals[0])

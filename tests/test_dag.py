@@ -1,5 +1,7 @@
 """Ray DAG tests: bind/execute for tasks + actors, InputNode, diamond
 dependencies, MultiOutputNode, compiled DAG reuse."""
+import time
+
 import pytest
 
 
@@ -235,3 +237,32 @@ def test_execute_async(ray_mod):
     out = asyncio.run(main())
     assert out == [100, 101, 102, 103]
     dag.teardown()
+
+
+def test_compiled_dag_actor_death_surfaces(ray_mod):
+    """A compiled-DAG actor dying mid-run raises ActorDiedError at get()
+    instead of hanging on the channel forever."""
+    import os
+
+    import ant_ray_amd as ray
+    from ant_ray_amd.dag import InputNode
+    from ant_ray_amd.exceptions import ActorDiedError
+
+    @ray.remote
+    class Stage:
+        def step(self, x):
+            if x == "die":
+                os._exit(1)
+            return x + 1
+
+    s = Stage.remote()
+    with InputNode() as inp:
+        dag = s.step.bind(inp).experimental_compile()
+    assert dag._channel_mode
+    assert ray.get(dag.execute(1), timeout=60) == 2
+    ref = dag.execute("die")
+    t0 = time.time()
+    with pytest.raises(Exception) as ei:
+        ray.get(ref, timeout=90)
+    assert time.time() - t0 < 60, "actor death took too long to surface"
+    assert isinstance(ei.value, (ActorDiedError, RuntimeError)), ei.value
