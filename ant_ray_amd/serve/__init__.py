@@ -183,7 +183,7 @@ def start(detached: bool = True, http_options: Optional[dict] = None,
         proxy = Proxy.options(
             name=PROXY_NAME, lifetime="detached", num_cpus=0,
             max_concurrency=1000, max_restarts=-1,
-        ).remote(host, port)
+        ).remote(host, port, http_options.get("request_timeout_s"))
         ray.get(proxy.ready.remote(), timeout=60)
     if grpc_options:
         from ant_ray_amd.serve._private.grpc_proxy import start_grpc_proxy

@@ -21,9 +21,11 @@ _ROUTE_REFRESH_S = 2.0
 
 
 class HTTPProxy:
-    def __init__(self, host: str = "127.0.0.1", port: int = 8000):
+    def __init__(self, host: str = "127.0.0.1", port: int = 8000,
+                 request_timeout_s: float = None):
         self.host = host
         self.port = port
+        self.request_timeout_s = request_timeout_s  # None = no limit
         self._routes: Dict[str, str] = {}  # route_prefix -> app name
         self._handles: Dict[str, DeploymentHandle] = {}
         self._last_refresh = 0.0
@@ -141,8 +143,25 @@ class HTTPProxy:
             await self._proxy_stream(send, app, req)
             return
         try:
-            result = await self._handles[app].remote(req)
+            coro = self._handles[app].remote(req)
+            if self.request_timeout_s:
+                import asyncio as _a
+
+                result = await _a.wait_for(
+                    _aw_response(coro), timeout=self.request_timeout_s)
+            else:
+                result = await coro
+        except TimeoutError:
+            await _send_simple(send, 408,
+                               b"request timed out (request_timeout_s)")
+            return
         except Exception as e:
+            import asyncio as _a
+
+            if isinstance(e, _a.TimeoutError):
+                await _send_simple(send, 408,
+                                   b"request timed out (request_timeout_s)")
+                return
             await _send_simple(send, 500, f"error: {e}".encode())
             return
         await _send_result(send, result)
@@ -184,6 +203,10 @@ class HTTPProxy:
                 await send({"type": "http.response.body", "body": b""})
         finally:
             router.complete(idx)
+
+
+async def _aw_response(resp):
+    return await resp
 
 
 def _encode_chunk(v) -> bytes:

@@ -670,3 +670,43 @@ def test_router_max_ongoing_backpressure(serve_mod):
     assert len(results) == 6
     assert max(results) <= 2, f"router admitted beyond the cap: {results}"
     serve.delete("gauged")
+
+
+def test_request_timeout(serve_mod):
+    """http_options request_timeout_s: slow requests get 408 at the proxy
+    instead of holding the connection open forever."""
+    import subprocess
+    import sys
+    import textwrap
+
+    # separate session: the module fixture's proxy has no timeout
+    script = textwrap.dedent("""
+        import urllib.error
+        import urllib.request
+
+        import ant_ray_amd as ray
+        from ant_ray_amd import serve
+
+        ray.init(num_cpus=4)
+        serve.start(http_options={"port": 8899, "request_timeout_s": 1.0})
+
+        @serve.deployment
+        class Sluggish:
+            async def __call__(self, request):
+                import asyncio
+
+                await asyncio.sleep(10)
+                return "late"
+
+        serve.run(Sluggish.bind(), name="slug", route_prefix="/slug")
+        try:
+            urllib.request.urlopen("http://127.0.0.1:8899/slug", timeout=30)
+            print("NO_TIMEOUT")
+        except urllib.error.HTTPError as e:
+            print("STATUS", e.code)
+        serve.shutdown()
+        ray.shutdown()
+    """)
+    out = subprocess.run([sys.executable, "-c", script], capture_output=True,
+                         text=True, timeout=180, cwd="/root/repo")
+    assert "STATUS 408" in out.stdout, out.stdout[-800:] + out.stderr[-800:]
