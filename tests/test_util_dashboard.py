@@ -88,7 +88,6 @@ def test_metrics_and_dashboard(ray_mod):
 
     port = start_dashboard(port=0 or 8277)
     base = f"http://127.0.0.1:{port}"
-    time.sleep(1.2)  # metrics publish is fire-and-forget
 
     with urllib.request.urlopen(f"{base}/healthz", timeout=10) as r:
         assert r.read() == b"ok"
@@ -109,8 +108,15 @@ def test_metrics_and_dashboard(ray_mod):
         with urllib.request.urlopen(f"{base}/api/logs/{logs[0]}",
                                     timeout=10) as r:
             r.read()
-    with urllib.request.urlopen(f"{base}/metrics", timeout=10) as r:
-        text = r.read().decode()
+    # metrics publish is fire-and-forget: retry the scrape
+    deadline = time.time() + 15
+    text = ""
+    while time.time() < deadline:
+        with urllib.request.urlopen(f"{base}/metrics", timeout=10) as r:
+            text = r.read().decode()
+        if 'test_requests_total{route="/a"} 3.0' in text:
+            break
+        time.sleep(0.5)
     assert 'test_requests_total{route="/a"} 3.0' in text
     assert "test_temp 42.5" in text
     assert 'test_lat_bucket{le="0.1"} 1' in text
