@@ -591,6 +591,17 @@ class GcsServer:
         key = p["key"]
         if not p.get("overwrite", True) and key in ns:
             return {"added": False}
+        seq = p.get("seq")
+        if seq is not None:
+            # last-writer-wins by CLIENT sequence: concurrent handler
+            # tasks (or chaos-delayed ones) must not let a stale write
+            # overwrite a newer value (metrics publishes rely on this)
+            if not hasattr(self, "_kv_seq"):
+                self._kv_seq = {}
+            k = (p.get("ns", ""), key)
+            if seq < self._kv_seq.get(k, -1):
+                return {"added": False, "stale": True}
+            self._kv_seq[k] = seq
         ns[key] = p["value"]
         self._persist_soon()
         return {"added": True}

@@ -26,13 +26,21 @@ def _publish(name: str, kind: str, value, tags: Dict[str, str],
                "ts": time.time()}
     if extra:
         payload.update(extra)
+    global _pub_seq
+    _pub_seq += 1
     try:
         cw.io.submit(cw.gcs.call("kv_put", {
             "ns": "metrics", "key": key,
             "value": json.dumps(payload).encode(), "overwrite": True,
+            # ordered last-writer-wins: fire-and-forget publishes may be
+            # handled out of order server-side
+            "seq": _pub_seq,
         }))
     except Exception:
         pass
+
+
+_pub_seq = 0
 
 
 class _Metric:
