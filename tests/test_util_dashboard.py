@@ -177,3 +177,26 @@ def test_inspect_serializability():
     assert not ok and any("lk" in f for f in fails)
     ok, fails = inspect_serializability({"x": 1})
     assert ok and not fails
+
+
+def test_kv_put_seq_ordering(ray_mod):
+    """Regression: sequenced kv_put must reject stale overwrites so
+    fire-and-forget metric publishes stay last-writer-wins even when the
+    GCS handles them out of order (chaos-jitter repro)."""
+    from ant_ray_amd._private.worker import global_worker
+
+    cw = global_worker.core_worker
+
+    def put(val, seq):
+        return cw.io.submit(cw.gcs.call("kv_put", {
+            "ns": "metrics", "key": b"seqtest",
+            "value": val, "overwrite": True, "seq": seq,
+        })).result(10)
+
+    assert put(b"v1", 1)["added"]
+    assert put(b"v3", 3)["added"]
+    r = put(b"v2", 2)  # stale: must not win
+    assert r.get("stale") or not r.get("added")
+    got = cw.io.submit(cw.gcs.call("kv_get", {
+        "ns": "metrics", "key": b"seqtest"})).result(10)
+    assert got["value"] == b"v3"
