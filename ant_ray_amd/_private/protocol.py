@@ -197,6 +197,9 @@ class Connection:
         for fut in self._pending.values():
             if not fut.done():
                 fut.set_exception(ConnectionLost(self.name))
+                # abandoned awaiters (cancelled mid-call) are expected on a
+                # lost conn; mark retrieved so GC doesn't log a warning
+                fut.exception()
         self._pending.clear()
         try:
             self.writer.close()
