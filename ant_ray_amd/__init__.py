@@ -33,6 +33,8 @@ __all__ = [
     "nodes", "cluster_resources", "available_resources", "method",
     "ObjectRef", "ObjectRefGenerator", "ActorHandle", "exceptions", "actor", "remote_function",
     "util", "train", "serve", "data", "tune",
+    "LoggingConfig", "ClientBuilder", "client", "Language", "SCRIPT_MODE",
+    "show_in_dashboard", "java_function", "java_actor_class", "cpp_function",
 ]
 
 
@@ -60,6 +62,10 @@ def init(
         if ignore_reinit_error:
             return RuntimeContext(global_worker)
         raise RuntimeError("ray.init() called twice; pass ignore_reinit_error=True")
+
+    lc = kwargs.pop("logging_config", None)
+    if lc is not None:
+        lc._apply()
 
     global_worker.namespace = namespace or f"ns-{os.getpid()}"
 
@@ -378,6 +384,72 @@ def timeline(filename: Optional[str] = None):
 def get_runtime_context() -> RuntimeContext:
     _check_connected()
     return RuntimeContext(global_worker)
+
+
+from ant_ray_amd._private.logging_config import LoggingConfig  # noqa: E402
+from ant_ray_amd.client_builder import ClientBuilder, client  # noqa: E402
+
+SCRIPT_MODE = DRIVER_MODE  # legacy alias (reference worker.py:111)
+
+
+class Language:
+    """Cross-language markers (reference Language proto enum). Only PYTHON
+    tasks execute in this build; JAVA/CPP exist so code that inspects the
+    enum imports cleanly."""
+
+    PYTHON = "PYTHON"
+    JAVA = "JAVA"
+    CPP = "CPP"
+
+
+def _xlang_unsupported(kind):
+    def fn(*_a, **_k):
+        raise NotImplementedError(
+            f"cross-language ({kind}) is not supported in this MI355X build; "
+            "see PARITY.md §2.2")
+
+    return fn
+
+
+java_function = _xlang_unsupported("java_function")
+java_actor_class = _xlang_unsupported("java_actor_class")
+cpp_function = _xlang_unsupported("cpp_function")
+
+
+class _Config:
+    """ray._config parity: read access to the env-overridable system-config
+    registry (`_private/ray_constants.py`)."""
+
+    def __getattr__(self, name):
+        from ant_ray_amd._private import ray_constants
+
+        def get(default=None):
+            sysc = getattr(ray_constants, "_system_config", {}) or {}
+            if name in sysc:
+                return sysc[name]
+            return os.environ.get(f"RAY_{name}", default)
+
+        return get
+
+
+_config = _Config()
+
+
+def show_in_dashboard(message: str, key: str = "", dtype: str = "text"):
+    """Display a message for the current task/actor in the dashboard
+    (parity: reference worker.py:2821). Stored in the GCS KV under the
+    "dashboard_display" namespace; served at GET /api/display."""
+    _check_connected()
+    assert dtype in ("text", "html"), f"dtype accepts only: text, html"
+    import json as _json
+
+    cw = global_worker.core_worker
+    owner = (cw.actor_id or cw.worker_id).hex()[:16]
+    cw.io.submit(cw.gcs.call("kv_put", {
+        "ns": "dashboard_display", "key": f"{owner}|{key}".encode(),
+        "value": _json.dumps({"message": message, "dtype": dtype}).encode(),
+        "overwrite": True,
+    }))
 
 
 # Submodules are imported lazily to keep `import ant_ray_amd` light.

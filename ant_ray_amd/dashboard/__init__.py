@@ -141,6 +141,17 @@ class DashboardHead:
                 return js({"error": str(e)}, 500)
         if path == "/api/virtual_clusters":
             return js(await self._gcs("list_virtual_clusters"))
+        if path == "/api/display":
+            # ray.show_in_dashboard messages (ns "dashboard_display")
+            keys = (await self._gcs("kv_keys", {"ns": "dashboard_display",
+                                                "prefix": b""}))["keys"]
+            out = {}
+            for k in keys:
+                r = await self._gcs("kv_get", {"ns": "dashboard_display",
+                                               "key": k})
+                if r.get("value"):
+                    out[k.decode()] = json.loads(r["value"].decode())
+            return js(out)
         if path == "/metrics":
             from ant_ray_amd.util.metrics import prometheus_text
 

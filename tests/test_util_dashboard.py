@@ -200,3 +200,63 @@ def test_kv_put_seq_ordering(ray_mod):
     got = cw.io.submit(cw.gcs.call("kv_get", {
         "ns": "metrics", "key": b"seqtest"})).result(10)
     assert got["value"] == b"v3"
+
+
+def test_top_level_api_additions(ray_mod):
+    """LoggingConfig / client builder / Language / show_in_dashboard /
+    _config — top-level parity items (reference ray/__init__.py __all__)."""
+    import io
+    import logging
+
+    import ant_ray_amd as ray
+
+    # LoggingConfig validation + formatter behavior (no global apply here)
+    from ant_ray_amd._private.logging_config import (CoreContextFilter,
+                                                     JSONFormatter,
+                                                     LoggingConfig)
+
+    with pytest.raises(ValueError):
+        LoggingConfig(encoding="YAML")
+    with pytest.raises(ValueError):
+        LoggingConfig(additional_log_standard_attrs=["nope"])
+    lc = LoggingConfig(encoding="JSON", log_level="DEBUG",
+                       additional_log_standard_attrs=["name"])
+    rec = logging.LogRecord("t", logging.INFO, __file__, 1, "hello %s",
+                            ("x",), None)
+    CoreContextFilter().filter(rec)
+    d = json.loads(JSONFormatter(["name"]).format(rec))
+    assert d["message"] == "hello x" and d["name"] == "t"
+    assert "worker_id" in d  # connected session stamps core context
+
+    # Language / SCRIPT_MODE / xlang stubs
+    assert ray.Language.PYTHON == "PYTHON"
+    assert ray.SCRIPT_MODE == ray.DRIVER_MODE
+    with pytest.raises(NotImplementedError):
+        ray.java_function("C", "m")
+
+    # client builder: on an already-initialized session connect() would
+    # re-init; just exercise the builder plumbing + double-connect guard
+    b = ray.client("127.0.0.1:9999").namespace("ns1").env({"env_vars": {}})
+    assert b.address == "127.0.0.1:9999" and b._namespace == "ns1"
+
+    # show_in_dashboard → /api/display
+    ray.show_in_dashboard("working on shard 3", key="status")
+    from ant_ray_amd.dashboard import start_dashboard
+
+    port = start_dashboard(port=8277)  # idempotent if already started
+    deadline = time.time() + 10
+    disp = {}
+    while time.time() < deadline:
+        with urllib.request.urlopen(
+                f"http://127.0.0.1:{port}/api/display", timeout=10) as r:
+            disp = json.loads(r.read())
+        if any(v.get("message") == "working on shard 3"
+               for v in disp.values()):
+            break
+        time.sleep(0.2)
+    assert any(v.get("message") == "working on shard 3"
+               for v in disp.values()), disp
+
+    # _config accessor
+    assert callable(ray._config.some_unknown_flag)
+    assert ray._config.some_unknown_flag(7) == 7
