@@ -24,11 +24,29 @@ PROXY_NAME = "SERVE_PROXY_ACTOR"
 
 __all__ = [
     "Application", "AutoscalingConfig", "Deployment", "DeploymentHandle",
-    "DeploymentResponse", "batch", "delete", "deployment",
-    "get_app_handle", "get_deployment_handle", "get_multiplexed_model_id",
-    "ingress", "multiplexed", "run", "shutdown",
+    "DeploymentResponse", "HTTPOptions", "RunTarget", "batch", "delete",
+    "deployment", "get_app_handle", "get_deployment_handle",
+    "get_multiplexed_model_id", "get_replica_context", "ingress",
+    "multiplexed", "run", "run_many", "shutdown", "shutdown_async",
     "start", "status",
 ]
+
+from ant_ray_amd.serve._private.replica import (  # noqa: E402
+    get_replica_context,
+)
+
+
+@dataclass
+class HTTPOptions:
+    """HTTP proxy options (parity: reference serve/config.py:620).
+    Accepted by serve.start(http_options=...) as an instance or dict."""
+
+    host: str = "127.0.0.1"
+    port: int = 8000
+    root_path: str = ""
+    request_timeout_s: Optional[float] = None
+    keep_alive_timeout_s: float = 5.0
+    location: str = "HeadOnly"
 
 
 @dataclass
@@ -165,11 +183,13 @@ def _get_or_create_controller(http_host="127.0.0.1", http_port=8000):
     return c
 
 
-def start(detached: bool = True, http_options: Optional[dict] = None,
+def start(detached: bool = True, http_options=None,
           grpc_options: Optional[dict] = None, **_):
     """Start Serve system actors (controller + HTTP proxy)."""
     import ant_ray_amd as ray
 
+    if isinstance(http_options, HTTPOptions):
+        http_options = http_options.__dict__
     http_options = http_options or {}
     host = http_options.get("host", "127.0.0.1")
     port = http_options.get("port", 8000)
@@ -259,6 +279,48 @@ def run(target: Application, *, name: str = DEFAULT_APP_NAME,
         except KeyboardInterrupt:
             pass
     return handle
+
+
+@dataclass(frozen=True)
+class RunTarget:
+    """One application for serve.run_many (parity serve/api.py:520)."""
+
+    target: Application
+    name: str = DEFAULT_APP_NAME
+    route_prefix: Optional[str] = "/"
+    logging_config: Optional[dict] = None
+
+
+def run_many(targets, blocking: bool = False, **_) -> List[DeploymentHandle]:
+    """Deploy several applications; returns their ingress handles
+    (parity serve/api.py:645)."""
+    if not targets:
+        raise ValueError("No applications provided.")
+    handles = []
+    for t in targets:
+        if not t.name:
+            raise ValueError("Application name must be a non-empty string.")
+        handles.append(run(t.target, name=t.name,
+                           route_prefix=t.route_prefix or f"/{t.name}"))
+    if blocking:
+        try:
+            while True:
+                time.sleep(1)
+        except KeyboardInterrupt:
+            pass
+    return handles
+
+
+_run = run
+_run_many = run_many
+
+
+async def shutdown_async():
+    """Async serve.shutdown (parity serve/api.py:130): awaitable from an
+    async context (e.g. inside a deployment), no blocking ray.get."""
+    import asyncio
+
+    await asyncio.get_event_loop().run_in_executor(None, shutdown)
 
 
 def delete(name: str, _blocking: bool = True):

@@ -186,9 +186,13 @@ class ServeController:
             opts.setdefault("num_cpus", 1)
             opts["max_concurrency"] = max(d.get("max_ongoing_requests", 100), 8)
             ReplicaCls = ray.remote(Replica)
+            rank = len(dr.replicas)
             h = ReplicaCls.options(**opts).remote(
                 d["callable_bytes"], d.get("init_args") or (),
                 d.get("init_kwargs") or {}, d.get("user_config"),
+                context={"app_name": app, "deployment": dep_name,
+                         "replica_tag": f"{app}#{dep_name}#{rank}",
+                         "rank": rank, "world_size": target},
             )
             dr.replicas.append(h)
             logger.info("started replica %d of %s/%s", len(dr.replicas), app,

@@ -48,13 +48,52 @@ class HTTPRequestData:
         return Request(scope, receive)
 
 
+class ReplicaContext:
+    """Runtime info for the current replica (parity: reference
+    serve/context.py:37 ReplicaContext — app_name/deployment/replica_tag/
+    servable_object/rank/world_size). Available inside a replica via
+    serve.get_replica_context()."""
+
+    def __init__(self, app_name, deployment, replica_tag, servable_object,
+                 rank, world_size):
+        self.app_name = app_name
+        self.deployment = deployment
+        self.replica_tag = replica_tag
+        self.servable_object = servable_object
+        self.rank = rank
+        self.world_size = world_size
+
+    def __repr__(self):
+        return (f"ReplicaContext(app={self.app_name!r}, "
+                f"deployment={self.deployment!r}, tag={self.replica_tag!r}, "
+                f"rank={self.rank}/{self.world_size})")
+
+
+_replica_context = None  # set once per replica actor process
+
+
+def get_replica_context() -> ReplicaContext:
+    if _replica_context is None:
+        raise RuntimeError(
+            "get_replica_context() may only be called from within a Ray "
+            "Serve replica (deployment constructor or request handler).")
+    return _replica_context
+
+
 class Replica:
     """The actor. Created by the controller with the deployment's pickled
     callable + init args (inner DeploymentHandles arrive ready to use)."""
 
     def __init__(self, callable_bytes: bytes, init_args, init_kwargs,
-                 user_config=None):
+                 user_config=None, context=None):
+        global _replica_context
         target = pickle.loads(callable_bytes)
+        if context:
+            _replica_context = ReplicaContext(
+                context.get("app_name", "default"),
+                context.get("deployment", ""),
+                context.get("replica_tag", ""),
+                None, context.get("rank", 0), context.get("world_size", 1))
         self._is_function = inspect.isfunction(target)
         if self._is_function:
             self._callable = target
@@ -63,6 +102,8 @@ class Replica:
             if user_config is not None and hasattr(self._callable,
                                                    "reconfigure"):
                 self._callable.reconfigure(user_config)
+        if _replica_context is not None:
+            _replica_context.servable_object = self._callable
         self._num_ongoing = 0
 
     def reconfigure(self, user_config):

@@ -11,13 +11,18 @@ from ant_ray_amd.train.base_trainer import (
     TrainingFailedError,
 )
 from ant_ray_amd.train.config import (
+    BackendConfig,
     CheckpointConfig,
+    DataConfig,
     FailureConfig,
     Result,
     RunConfig,
     ScalingConfig,
+    SyncConfig,
     TorchConfig,
 )
+
+TRAIN_DATASET_KEY = "train"
 from ant_ray_amd.train.session import (
     TrainContext,
     get_checkpoint,
@@ -27,6 +32,7 @@ from ant_ray_amd.train.session import (
 )
 
 __all__ = [
+    "BackendConfig", "DataConfig", "SyncConfig", "TRAIN_DATASET_KEY",
     "Checkpoint", "CheckpointConfig", "DataParallelTrainer", "FailureConfig",
     "Result", "RunConfig", "ScalingConfig", "TorchConfig", "TrainContext",
     "TrainingFailedError", "get_checkpoint", "get_context",

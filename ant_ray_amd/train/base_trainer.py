@@ -38,6 +38,7 @@ class DataParallelTrainer:
         self.scaling_config = scaling_config or ScalingConfig()
         self.run_config = run_config or RunConfig()
         self.datasets = datasets
+        self.dataset_config = dataset_config
         self.resume_from_checkpoint = resume_from_checkpoint
         self.metadata = metadata
         self.backend_config = backend_config
@@ -58,6 +59,7 @@ class DataParallelTrainer:
             torch_config=self._torch_config(),
             datasets=self.datasets,
             resume_from_checkpoint=self.resume_from_checkpoint,
+            dataset_config=self.dataset_config,
         )
         result = controller.run()
         if result.error is not None:

@@ -114,3 +114,58 @@ class Result:
             return None
         keyed.sort(key=lambda t: t[0], reverse=(mode == "max"))
         return keyed[0][1]
+
+
+@dataclass
+class SyncConfig:
+    """Artifact/driver sync knobs (parity: reference train/_internal/
+    syncer.py SyncConfig). Single shared-FS node class: syncing is a
+    no-op, the fields exist so user configs carry through unchanged."""
+
+    sync_period: int = 300
+    sync_timeout: int = 1800
+    sync_artifacts: bool = False
+    sync_artifacts_on_checkpoint: bool = True
+
+
+class BackendConfig:
+    """Parent class for training-backend configurations (parity:
+    reference train/backend.py:16). TorchConfig subclasses this; custom
+    backends override backend_cls/train_func_context."""
+
+    @property
+    def backend_cls(self):
+        return None
+
+    @property
+    def train_func_context(self):
+        import contextlib
+
+        return contextlib.nullcontext
+
+
+class DataConfig:
+    """Configures which datasets are split across train workers (parity:
+    reference train/_internal/data_config.py:13). datasets_to_split="all"
+    (default) shards every dataset via streaming split; a list limits
+    sharding to those names — the rest are replicated whole to each
+    worker."""
+
+    def __init__(self, datasets_to_split="all", execution_options=None,
+                 enable_shard_locality: bool = True):
+        if not (datasets_to_split == "all"
+                or isinstance(datasets_to_split, list)):
+            raise TypeError(
+                "`datasets_to_split` should be 'all' or a list of dataset "
+                f"names, got {datasets_to_split!r}")
+        self._datasets_to_split = datasets_to_split
+        self._execution_options = execution_options
+        self._enable_shard_locality = enable_shard_locality
+
+    def _should_split(self, name: str) -> bool:
+        return (self._datasets_to_split == "all"
+                or name in self._datasets_to_split)
+
+    @staticmethod
+    def default_ingest_options():
+        return None

@@ -86,6 +86,7 @@ class TrainController:
         torch_config: Optional[TorchConfig] = None,
         datasets: Optional[Dict[str, Any]] = None,
         resume_from_checkpoint: Optional[Checkpoint] = None,
+        dataset_config=None,
     ):
         self.train_fn = train_fn
         self.train_loop_config = train_loop_config
@@ -93,6 +94,7 @@ class TrainController:
         self.run_config = run_config
         self.torch_config = torch_config or TorchConfig()
         self.datasets = datasets or {}
+        self.dataset_config = dataset_config
         self.name = run_config.name or f"train_{int(time.time())}"
         self.experiment_path = os.path.join(run_config.storage_path, self.name)
         os.makedirs(self.experiment_path, exist_ok=True)
@@ -124,7 +126,9 @@ class TrainController:
         shards = [dict() for _ in range(n)]
         for name, ds in self.datasets.items():
             splits = None
-            if hasattr(ds, "streaming_split"):
+            dc = self.dataset_config
+            split_this = dc._should_split(name) if dc is not None else True
+            if split_this and hasattr(ds, "streaming_split"):
                 try:
                     splits = ds.streaming_split(n, equal=True)
                 except Exception:

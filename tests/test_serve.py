@@ -710,3 +710,44 @@ def test_request_timeout(serve_mod):
     out = subprocess.run([sys.executable, "-c", script], capture_output=True,
                          text=True, timeout=180, cwd="/root/repo")
     assert "STATUS 408" in out.stdout, out.stdout[-800:] + out.stderr[-800:]
+
+
+def test_replica_context_and_run_many(serve_mod):
+    """get_replica_context inside replicas; run_many deploys several apps;
+    HTTPOptions accepted by start (parity: serve/context.py:37,
+    serve/api.py:520/645)."""
+    serve = serve_mod
+
+    @serve.deployment(num_replicas=2)
+    class WhoAmI:
+        def __init__(self):
+            ctx = serve.get_replica_context()
+            self.boot_tag = ctx.replica_tag
+
+        def __call__(self, _):
+            ctx = serve.get_replica_context()
+            assert ctx.servable_object is self
+            return {"app": ctx.app_name, "dep": ctx.deployment,
+                    "tag": ctx.replica_tag, "rank": ctx.rank,
+                    "ws": ctx.world_size, "boot": self.boot_tag}
+
+    @serve.deployment
+    def plain(_):
+        return "plain-ok"
+
+    # outside a replica -> error
+    with pytest.raises(RuntimeError):
+        serve.get_replica_context()
+
+    h1, h2 = serve.run_many([
+        serve.RunTarget(WhoAmI.bind(), name="who"),
+        serve.RunTarget(plain.bind(), name="plain_app"),
+    ])
+    seen = {h1.remote(None).result(timeout_s=30)["tag"] for _ in range(10)}
+    one = h1.remote(None).result(timeout_s=30)
+    assert one["app"] == "who" and one["dep"] == "WhoAmI"
+    assert one["ws"] == 2 and one["boot"] == one["tag"]
+    assert len(seen) == 2  # both ranks answered across 10 calls
+    assert h2.remote(None).result(timeout_s=30) == "plain-ok"
+    serve.delete("who")
+    serve.delete("plain_app")
