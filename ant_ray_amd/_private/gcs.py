@@ -920,6 +920,17 @@ class GcsServer:
         info = self.actors.get(p["actor_id"])
         return info.view() if info else None
 
+    async def rpc_list_named_actors(self, conn, p):
+        # parity: GCS ListNamedActors (ray.util.list_named_actors)
+        out = []
+        for (namespace, name), aid in self.named_actors.items():
+            info = self.actors.get(aid)
+            if info is None or info.state == DEAD:
+                continue
+            if p.get("all_namespaces") or namespace == (p.get("namespace") or ""):
+                out.append({"namespace": namespace, "name": name})
+        return {"actors": out}
+
     async def rpc_get_actor_by_name(self, conn, p):
         aid = self.named_actors.get((p.get("namespace") or "", p["name"]))
         if aid is None:

@@ -115,12 +115,25 @@ def _rebuild_torch_tensor(buf, dtype_str, shape):
     return t.view(dtype).reshape(shape)
 
 
+# ray.util.register_serializer registry (exact-type match, like the
+# reference's per-class cloudpickle dispatch registration)
+_custom_serializers = {}
+
+
+def _apply_custom_deserializer(deser, data):
+    return deser(data)
+
+
 class _Pickler(cloudpickle.CloudPickler):
     def __init__(self, file, buffer_callback=None):
         super().__init__(file, protocol=5, buffer_callback=buffer_callback)
         self.contained_refs = []
 
     def reducer_override(self, obj):
+        cs = _custom_serializers.get(type(obj))
+        if cs is not None:
+            ser, deser = cs
+            return (_apply_custom_deserializer, (deser, ser(obj)))
         from ant_ray_amd._private.object_ref import ObjectRef
 
         if isinstance(obj, ObjectRef):

@@ -114,5 +114,20 @@ def placement_group_table(pg: Optional[PlacementGroup] = None):
     return out
 
 
+def get_placement_group(placement_group_name: str) -> PlacementGroup:
+    """Look up a placement group by the name it was created with
+    (parity: reference util/placement_group.py get_placement_group)."""
+    from ant_ray_amd._private.worker import global_worker
+
+    cw = global_worker.core_worker
+    views = cw.io.run(cw.gcs.call("list_placement_groups", {}, timeout=30),
+                      timeout=35)
+    for v in views:
+        if v.get("name") == placement_group_name:
+            return PlacementGroup(v["pg_id"], v.get("bundles"))
+    raise ValueError(
+        f"Failed to look up placement group with name {placement_group_name!r}")
+
+
 def get_current_placement_group() -> Optional[PlacementGroup]:
     return None

@@ -260,3 +260,72 @@ def test_top_level_api_additions(ray_mod):
     # _config accessor
     assert callable(ray._config.some_unknown_flag)
     assert ray._config.some_unknown_flag(7) == 7
+
+
+def test_util_parity_helpers(ray_mod):
+    """as_completed / map_unordered / log_once / custom serializers /
+    list_named_actors / get_placement_group (reference util/__init__.py
+    surface)."""
+    import ant_ray_amd as ray
+    from ant_ray_amd.util import (as_completed, deregister_serializer,
+                                  list_named_actors, map_unordered,
+                                  register_serializer)
+    from ant_ray_amd.util.debug import log_once, reset_log_once
+
+    @ray.remote
+    def sq(x):
+        return x * x
+
+    # as_completed yields everything, unordered
+    refs = [sq.remote(i) for i in range(20)]
+    assert sorted(as_completed(refs, chunk_size=4)) == [i * i for i in range(20)]
+    # map_unordered with backpressure
+    out = sorted(map_unordered(sq, range(25), backpressure_size=5,
+                               chunk_size=3))
+    assert out == [i * i for i in range(25)]
+
+    # log_once
+    reset_log_once("k1")
+    assert log_once("k1") is True
+    assert log_once("k1") is False
+
+    # custom serializer round-trips through a task
+    class Vec:
+        def __init__(self, x, y):
+            self.x, self.y = x, y
+
+    register_serializer(Vec, serializer=lambda v: (v.x, v.y),
+                        deserializer=lambda s: Vec(s[0] * 10, s[1] * 10))
+    try:
+        @ray.remote
+        def through(v):
+            return (v.x, v.y)
+
+        assert ray.get(through.remote(Vec(1, 2)), timeout=30) == (10, 20)
+    finally:
+        deregister_serializer(Vec)
+
+    # list_named_actors
+    @ray.remote
+    class Named:
+        def ping(self):
+            return "pong"
+
+    a = Named.options(name="util_listed_actor").remote()
+    ray.get(a.ping.remote(), timeout=30)
+    assert "util_listed_actor" in list_named_actors()
+    assert {"name": "util_listed_actor", "namespace": ""} in \
+        list_named_actors(all_namespaces=True)
+    ray.kill(a)
+
+    # get_placement_group by name
+    from ant_ray_amd.util import get_placement_group, placement_group, \
+        remove_placement_group
+
+    pg = placement_group([{"CPU": 1}], name="util_pg")
+    assert pg.wait(30)
+    found = get_placement_group("util_pg")
+    assert found.id == pg.id
+    remove_placement_group(pg)
+    with pytest.raises(ValueError):
+        get_placement_group("no_such_pg")
