@@ -64,6 +64,8 @@ class RunConfig:
     verbose: int = 1
     log_to_file: bool = False
     callbacks: Optional[list] = None
+    stop: Optional[Any] = None  # tune: dict {metric: threshold} / fn / Stopper
+    progress_reporter: Optional[Any] = None  # tune: ProgressReporter
 
     def __post_init__(self):
         if self.storage_path is None:

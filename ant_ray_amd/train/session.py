@@ -118,6 +118,11 @@ def report(metrics: Dict[str, Any], checkpoint: Optional[Checkpoint] = None,
             shutil.copytree(checkpoint.path, target, dirs_exist_ok=True)
         persisted = Checkpoint(target)
         ctx._latest_checkpoint = persisted
+    stop_ev = getattr(ctx, "stop_requested", None)
+    if stop_ev is not None and stop_ev.is_set():
+        # tune cooperative stop: the Tuner decided this trial is done
+        # (scheduler / stop criteria); unwind the trainable cleanly
+        raise SystemExit(0)
     ctx.report_queue.put(
         {
             "seq": seq,

@@ -30,17 +30,135 @@ from ant_ray_amd.tune.search import (  # noqa: F401
     BasicVariantGenerator,
     choice,
     grid_search,
+    lograndint,
     loguniform,
+    qlograndint,
+    qloguniform,
+    qrandint,
+    qrandn,
+    quniform,
     randint,
+    randn,
+    sample_from,
     uniform,
 )
+from ant_ray_amd.tune.analysis import ExperimentAnalysis  # noqa: F401
+from ant_ray_amd.tune.progress_reporter import (  # noqa: F401
+    CLIReporter,
+    JupyterNotebookReporter,
+    ProgressReporter,
+)
+from ant_ray_amd.tune.registry import (  # noqa: F401
+    create_scheduler,
+    create_searcher,
+    register_env,
+    register_trainable,
+)
+from ant_ray_amd.tune.stopper import Stopper  # noqa: F401
+from ant_ray_amd.tune.trainable import Trainable  # noqa: F401
+from ant_ray_amd.train.config import (  # noqa: F401 (shared AIR configs)
+    CheckpointConfig,
+    FailureConfig,
+    SyncConfig,
+)
+
+
+class TuneError(Exception):
+    """Parity: reference tune/error.py TuneError."""
+
+
+@dataclass
+class ResumeConfig:
+    """Parity: tune/tuner.py ResumeConfig — what to do with unfinished /
+    errored trials on Tuner.restore."""
+
+    resume_unfinished: bool = True
+    resume_errored: bool = False
+    restart_errored: bool = False
+
+
+class PlacementGroupFactory:
+    """Parity: tune/execution/placement_groups.py — bundle list a
+    trainable's trial reserves. with_resources() accepts one; the first
+    bundle is the trial actor's own resources."""
+
+    def __init__(self, bundles, strategy: str = "PACK"):
+        if not bundles:
+            raise ValueError("PlacementGroupFactory needs >=1 bundle")
+        self.bundles = [dict(b) for b in bundles]
+        self.strategy = strategy
+
+    @property
+    def head_bundle(self):
+        return self.bundles[0]
+
+
+class TuneContext:
+    """Parity: tune/context.py TuneContext (tune.get_context()) — trial
+    info inside a trainable function."""
+
+    def __init__(self, train_ctx):
+        self._ctx = train_ctx
+
+    def get_trial_id(self) -> str:
+        return self._ctx.experiment_name
+
+    def get_trial_name(self) -> str:
+        return self._ctx.experiment_name
+
+    def get_trial_dir(self) -> str:
+        return self._ctx.experiment_path
+
+    def get_experiment_name(self) -> str:
+        return os.path.basename(
+            os.path.dirname(self._ctx.experiment_path.rstrip("/")))
+
+
+def get_context() -> TuneContext:
+    from ant_ray_amd.train.session import get_context as _train_get
+
+    return TuneContext(_train_get())
+
 
 __all__ = [
-    "ASHAScheduler", "BasicVariantGenerator", "FIFOScheduler",
-    "MedianStoppingRule", "ResultGrid", "TuneConfig", "Tuner", "choice",
-    "grid_search", "loguniform", "randint", "report", "run", "uniform",
-    "with_parameters", "with_resources",
+    "ASHAScheduler", "BasicVariantGenerator", "CLIReporter", "Callback",
+    "CheckpointConfig", "Experiment", "ExperimentAnalysis", "FIFOScheduler",
+    "FailureConfig", "JupyterNotebookReporter", "MedianStoppingRule",
+    "PlacementGroupFactory", "ProgressReporter", "ResultGrid", "ResumeConfig",
+    "Stopper", "SyncConfig", "Trainable", "TuneConfig", "TuneContext",
+    "TuneError", "Tuner", "choice", "create_scheduler", "create_searcher",
+    "get_context", "grid_search", "lograndint", "loguniform", "qlograndint",
+    "qloguniform", "qrandint", "qrandn", "quniform", "randint", "randn",
+    "register_env", "register_trainable", "report", "run",
+    "run_experiments", "sample_from", "uniform", "with_parameters",
+    "with_resources",
 ]
+
+
+class Callback:
+    """Parity: tune/callback.py Callback — hooks fired by the Tuner loop.
+    Override any of the on_* methods."""
+
+    def setup(self, **info):
+        pass
+
+    def on_trial_start(self, iteration, trials, trial, **info):
+        pass
+
+    def on_trial_result(self, iteration, trials, trial, result, **info):
+        pass
+
+    def on_trial_complete(self, iteration, trials, trial, **info):
+        pass
+
+    def on_trial_error(self, iteration, trials, trial, **info):
+        pass
+
+    def on_checkpoint(self, iteration, trials, trial, checkpoint, **info):
+        pass
+
+    def on_experiment_end(self, trials, **info):
+        pass
 
 
 @dataclass
@@ -135,6 +253,8 @@ class _TrialRunner:
         ctx = TrainContext(experiment_name=trial_id,
                            experiment_path=experiment_path)
         ctx.report_queue = self._q
+        ctx.stop_requested = self._threading.Event()
+        self._stop_ev = ctx.stop_requested
         if restore_path:
             ctx.restore_checkpoint = Checkpoint(restore_path)
 
@@ -142,6 +262,8 @@ class _TrialRunner:
             set_train_context(ctx)
             try:
                 fn(config)
+            except SystemExit:
+                pass  # cooperative stop via request_stop()
             except BaseException:
                 self._error = traceback.format_exc()
             finally:
@@ -149,6 +271,15 @@ class _TrialRunner:
 
         self._thread = self._threading.Thread(target=run, daemon=True)
         self._thread.start()
+        return True
+
+    def request_stop(self):
+        """Cooperative stop: the next tune.report() in the trial raises
+        SystemExit (parity: the reference stops a trial at its next
+        result boundary rather than mid-step)."""
+        ev = getattr(self, "_stop_ev", None)
+        if ev is not None:
+            ev.set()
         return True
 
     def poll(self):
@@ -225,7 +356,46 @@ class Tuner:
 
         if not ray.is_initialized():
             ray.init()
+        if isinstance(self.trainable, str):
+            from ant_ray_amd.tune.registry import get_trainable_cls
+
+            self.trainable = get_trainable_cls(self.trainable)
+        if isinstance(self.trainable, type) and issubclass(self.trainable,
+                                                           Trainable):
+            from ant_ray_amd.tune.trainable import _wrap_trainable_cls
+
+            self.trainable = _wrap_trainable_cls(self.trainable,
+                                                 stop=self.run_config.stop)
         tc = self.tune_config
+        stop_cfg = self.run_config.stop
+        callbacks = list(self.run_config.callbacks or [])
+        reporter = self.run_config.progress_reporter
+
+        def _stop_hit(trial_id, metrics):
+            if stop_cfg is None:
+                return False
+            if isinstance(stop_cfg, dict):
+                return any(metrics.get(k) is not None and metrics[k] >= v
+                           for k, v in stop_cfg.items())
+            return bool(stop_cfg(trial_id, metrics))
+
+        def _fire_cb(hook, *a, **kw):
+            for cb in callbacks:
+                try:
+                    getattr(cb, hook, lambda *x, **y: None)(*a, **kw)
+                except Exception:
+                    pass
+
+        def _trial_rows():
+            rows = []
+            for i, t in running.items():
+                rows.append({"trial_id": t["trial_id"], "status": "running",
+                             "metrics": t["last_metrics"]})
+            for i, r in results.items():
+                rows.append({"trial_id": f"{name}_{i:05d}",
+                             "status": "errored" if r.error else "finished",
+                             "metrics": r.metrics})
+            return rows
         if self._restore_state is not None:
             variants = self._restore_state["variants"]
         else:
@@ -287,6 +457,7 @@ class Tuner:
                             "ckpt": prev["ckpt"] if prev else None}
             if hasattr(scheduler, "on_trial_start"):
                 scheduler.on_trial_start(trial_id, config)
+            _fire_cb("on_trial_start", 0, None, trial_id)
 
         try:
             self._save_state(exp_path, variants, results, running)
@@ -318,6 +489,10 @@ class Tuner:
                         stop = True
                     elif decision == "PERTURB":
                         perturb = True
+                    if _stop_hit(t["trial_id"], metrics):
+                        stop = True
+                    _fire_cb("on_trial_result", t["iter"], None,
+                             t["trial_id"], metrics)
                 if perturb and st["status"] == "running":
                     # PBT exploit/explore: clone a top trial's checkpoint,
                     # mutate hyperparams, relaunch this slot
@@ -330,6 +505,11 @@ class Tuner:
                             pass
                         launch(idx, new_cfg, restore=restore_path, prev=t)
                         continue
+                if stop and st["status"] == "running":
+                    try:
+                        t["actor"].request_stop.remote()
+                    except Exception:
+                        pass
                 if st["status"] in ("finished", "errored") or stop:
                     err = None
                     if st["status"] == "errored":
@@ -339,6 +519,8 @@ class Tuner:
                         checkpoint=Checkpoint(t["ckpt"]) if t["ckpt"] else None,
                         path=t["path"], error=err,
                     )
+                    _fire_cb("on_trial_error" if err else "on_trial_complete",
+                             t["iter"], None, t["trial_id"])
                     try:
                         ray.kill(t["actor"])
                     except Exception:
@@ -351,8 +533,15 @@ class Tuner:
                     self._save_state(exp_path, variants, results, running)
                 except Exception:
                     pass
+            if reporter is not None and reporter.should_report(
+                    _trial_rows(), done=not (pending or running)):
+                try:
+                    reporter.report(_trial_rows(), not (pending or running))
+                except Exception:
+                    pass
             if running:
                 time.sleep(0.1)
+        _fire_cb("on_experiment_end", _trial_rows())
         return ResultGrid([results[i] for i in sorted(results)])
 
 
@@ -364,5 +553,35 @@ def run(trainable, *, config: Optional[dict] = None, num_samples: int = 1,
         trainable, param_space=config or {},
         tune_config=TuneConfig(metric=metric, mode=mode,
                                num_samples=num_samples, scheduler=scheduler),
+        run_config=RunConfig(stop=kwargs.pop("stop", None),
+                             name=kwargs.pop("name", None)),
     )
     return tuner.fit()
+
+
+@dataclass
+class Experiment:
+    """Parity: tune/experiment/experiment.py — a named spec for
+    run_experiments."""
+
+    name: str
+    run: Any  # trainable (fn / Trainable subclass / registered name)
+    config: Optional[dict] = None
+    stop: Optional[Any] = None
+    num_samples: int = 1
+
+
+def run_experiments(experiments, **_) -> Dict[str, ResultGrid]:
+    """Run one or more Experiment specs; returns {name: ResultGrid}
+    (parity tune/tune.py run_experiments)."""
+    if isinstance(experiments, Experiment):
+        experiments = [experiments]
+    out = {}
+    for e in experiments:
+        tuner = Tuner(
+            e.run, param_space=e.config or {},
+            tune_config=TuneConfig(num_samples=e.num_samples),
+            run_config=RunConfig(name=e.name, stop=e.stop),
+        )
+        out[e.name] = tuner.fit()
+    return out

@@ -105,3 +105,83 @@ class BasicVariantGenerator:
                         cfg[k] = v
                 out.append(cfg)
         return out
+
+
+class Normal(Domain):
+    def __init__(self, mean, sd):
+        self.mean, self.sd = mean, sd
+
+    def sample(self, rng):
+        return rng.gauss(self.mean, self.sd)
+
+
+class LogRandint(Domain):
+    def __init__(self, low, high):
+        import math
+
+        self.lo, self.hi = math.log(low), math.log(high)
+
+    def sample(self, rng):
+        import math
+
+        return int(math.exp(rng.uniform(self.lo, self.hi)))
+
+
+class Quantized(Domain):
+    """Round another domain's samples to multiples of q (reference
+    sample.py quantized spaces)."""
+
+    def __init__(self, inner: Domain, q):
+        self.inner, self.q = inner, q
+
+    def sample(self, rng):
+        v = round(self.inner.sample(rng) / self.q) * self.q
+        if isinstance(self.inner, (Randint, LogRandint)):
+            return int(v)
+        return v
+
+
+class Function(Domain):
+    """tune.sample_from — draw from a user callable; the callable may take
+    an optional spec argument (ignored here, passed as None)."""
+
+    def __init__(self, fn):
+        self.fn = fn
+
+    def sample(self, rng):
+        try:
+            return self.fn(None)
+        except TypeError:
+            return self.fn()
+
+
+def randn(mean: float = 0.0, sd: float = 1.0) -> Normal:
+    return Normal(mean, sd)
+
+
+def qrandn(mean, sd, q) -> Quantized:
+    return Quantized(Normal(mean, sd), q)
+
+
+def quniform(low, high, q) -> Quantized:
+    return Quantized(Uniform(low, high), q)
+
+
+def qloguniform(low, high, q) -> Quantized:
+    return Quantized(LogUniform(low, high), q)
+
+
+def lograndint(low, high) -> LogRandint:
+    return LogRandint(low, high)
+
+
+def qrandint(low, high, q) -> Quantized:
+    return Quantized(Randint(low, high), q)
+
+
+def qlograndint(low, high, q) -> Quantized:
+    return Quantized(LogRandint(low, high), q)
+
+
+def sample_from(fn) -> Function:
+    return Function(fn)

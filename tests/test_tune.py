@@ -229,3 +229,133 @@ def test_tuner_restore(ray_mod, tmp_path_factory):
     by_x = {r.metrics["x"]: r.metrics for r in grid2}
     assert by_x[1]["it"] == 3          # finished trial carried over
     assert by_x[2]["it"] == 3          # interrupted trial completed
+
+
+def test_trainable_class_api(ray_mod, tmp_path_factory):
+    """Class-API Trainable: setup/step loop with stop criteria, checkpoint
+    at end, registry launch by name (parity: trainable/trainable.py,
+    tune/registry.py)."""
+    import os
+
+    from ant_ray_amd import tune
+    from ant_ray_amd.train import RunConfig
+
+    class MyTrainable(tune.Trainable):
+        def setup(self, config):
+            self.base = config["base"]
+
+        def step(self):
+            return {"score": self.base * self.training_iteration}
+
+        def save_checkpoint(self, d):
+            with open(os.path.join(d, "state.txt"), "w") as f:
+                f.write(str(self.training_iteration))
+            return d
+
+    storage = str(tmp_path_factory.mktemp("tune_cls"))
+    tuner = tune.Tuner(
+        MyTrainable,
+        param_space={"base": tune.grid_search([1, 2])},
+        run_config=RunConfig(storage_path=storage,
+                             stop={"training_iteration": 4}),
+    )
+    rg = tuner.fit()
+    assert len(rg) == 2 and not rg.errors
+    best = rg.get_best_result(metric="score", mode="max")
+    # step() sees the pre-increment iteration count (reference Trainable
+    # numbering): the stopping result (training_iteration=4) computed 2*3
+    assert best.metrics["score"] == 2 * 3
+    assert best.checkpoint is not None
+    with open(os.path.join(best.checkpoint.path, "state.txt")) as f:
+        assert f.read() == "4"
+
+    # by-name via registry
+    tune.register_trainable("my_trainable", MyTrainable)
+    rg2 = tune.Tuner(
+        "my_trainable", param_space={"base": [3]},
+        run_config=RunConfig(storage_path=storage,
+                             stop={"training_iteration": 2}),
+    ).fit()
+    assert not rg2.errors
+
+
+def test_stopper_callback_reporter(ray_mod, tmp_path_factory):
+    """RunConfig stop dict + Stopper + Callback hooks + CLIReporter
+    (parity: tune/stopper, tune/callback.py, progress_reporter.py)."""
+    from ant_ray_amd import tune
+    from ant_ray_amd.train import RunConfig
+    from ant_ray_amd.tune.stopper import MaximumIterationStopper
+
+    def loop(config):
+        import time as _t
+
+        for i in range(100):
+            tune.report({"loss": 1.0 / (i + 1)})
+            _t.sleep(0.05)  # slow enough for the Tuner to stop us in-band
+
+    events = []
+
+    class Rec(tune.Callback):
+        def on_trial_start(self, iteration, trials, trial, **info):
+            events.append(("start", trial))
+
+        def on_trial_result(self, iteration, trials, trial, result, **info):
+            events.append(("result", result["training_iteration"]))
+
+        def on_trial_complete(self, iteration, trials, trial, **info):
+            events.append(("complete", trial))
+
+    storage = str(tmp_path_factory.mktemp("tune_stop"))
+    rg = tune.Tuner(
+        loop, param_space={},
+        run_config=RunConfig(
+            storage_path=storage,
+            stop=MaximumIterationStopper(5),
+            callbacks=[Rec()],
+            progress_reporter=tune.CLIReporter(max_report_frequency=0.0),
+        ),
+    ).fit()
+    assert not rg.errors
+    # stopped well before 100 iterations
+    assert rg[0].metrics["training_iteration"] <= 60
+    kinds = [e[0] for e in events]
+    assert "start" in kinds and "result" in kinds and "complete" in kinds
+
+
+def test_sampling_domains_and_analysis(ray_mod, tmp_path_factory):
+    """Quantized/log domains + sample_from resolve in param spaces;
+    ExperimentAnalysis reloads a finished experiment (parity:
+    tune/search/sample.py, analysis/experiment_analysis.py)."""
+    import os
+
+    from ant_ray_amd import tune
+    from ant_ray_amd.train import RunConfig
+
+    def trial(config):
+        assert config["q"] in {0.0, 0.25, 0.5, 0.75, 1.0}
+        assert 1 <= config["li"] <= 1000 and isinstance(config["li"], int)
+        assert config["fixed"] == 42
+        tune.report({"obj": config["q"]})
+
+    storage = str(tmp_path_factory.mktemp("tune_dom"))
+    rg = tune.Tuner(
+        trial,
+        param_space={"q": tune.quniform(0, 1, 0.25),
+                     "li": tune.lograndint(1, 1000),
+                     "fixed": tune.sample_from(lambda spec: 42)},
+        tune_config=tune.TuneConfig(num_samples=6, seed=3),
+        run_config=RunConfig(storage_path=storage, name="dom_exp"),
+    ).fit()
+    assert len(rg) == 6 and not rg.errors
+
+    ea = tune.ExperimentAnalysis(os.path.join(storage, "dom_exp"))
+    assert len(ea.trials) == 6
+    cfg = ea.get_best_config(metric="obj", mode="max")
+    assert cfg["fixed"] == 42
+    df = ea.dataframe()
+    assert len(df) == 6
+
+    # create_scheduler / create_searcher factories
+    assert type(tune.create_scheduler("asha")).__name__ == "ASHAScheduler"
+    with pytest.raises(ValueError):
+        tune.create_scheduler("nope")
