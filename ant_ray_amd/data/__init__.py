@@ -10,6 +10,7 @@ from __future__ import annotations
 
 import glob as _glob
 import os
+from dataclasses import dataclass as _dataclass, field as _field
 from typing import Any, Dict, List, Optional
 
 import numpy as np
@@ -21,11 +22,78 @@ from ant_ray_amd.data.iterator import DataIterator
 from ant_ray_amd.data.expressions import col, lit
 from ant_ray_amd.data.plan import ActorPoolStrategy, ReadOp
 
+from ant_ray_amd.data.datasource import (  # noqa: E402
+    BlockBasedFileDatasink,
+    Datasink,
+    Datasource,
+    ReadTask,
+    RowBasedFileDatasink,
+    SaveMode,
+    SinkMode,
+)
+from ant_ray_amd.data.dataset import Schema  # noqa: E402
+from ant_ray_amd.data.preprocessor import Preprocessor  # noqa: E402
+from ant_ray_amd.data.stats import DatasetSummary  # noqa: E402
+
+# reference-name aliases
+DatasetContext = DataContext
+DatasetIterator = DataIterator
+NodeIdStr = str
+
+
+@_dataclass
+class ExecutionResources:
+    """Resource budget for a Data execution (parity: reference
+    execution_options.py ExecutionResources)."""
+
+    cpu: Optional[float] = None
+    gpu: Optional[float] = None
+    object_store_memory: Optional[float] = None
+
+
+@_dataclass
+class ExecutionOptions:
+    """Execution knobs (parity: execution_options.py ExecutionOptions);
+    carried on DataContext / Train DataConfig."""
+
+    resource_limits: ExecutionResources = _field(
+        default_factory=ExecutionResources)
+    exclude_resources: ExecutionResources = _field(
+        default_factory=ExecutionResources)
+    locality_with_output: bool = False
+    preserve_order: bool = False
+    actor_locality_enabled: bool = False
+    verbose_progress: bool = False
+
+
+@_dataclass
+class TaskPoolStrategy:
+    """Plain-task compute strategy marker (parity: reference
+    TaskPoolStrategy); Data ops treat it the same as compute=None."""
+
+    size: Optional[int] = None
+
+
+@_dataclass
+class FileShuffleConfig:
+    """Seeded shuffle of input-file order for read_* APIs (parity:
+    reference FileShuffleConfig)."""
+
+    seed: Optional[int] = None
+
+
 __all__ = [
-    "ActorPoolStrategy", "DataContext", "DataIterator", "Dataset",
-    "MaterializedDataset", "range", "range_tensor", "from_items",
-    "from_numpy", "from_pandas", "from_arrow", "from_torch",
-    "from_huggingface", "read_parquet", "read_csv", "read_json", "read_text",
+    "ActorPoolStrategy", "BlockBasedFileDatasink", "DataContext",
+    "DataIterator", "Dataset", "DatasetContext", "DatasetIterator",
+    "DatasetSummary", "Datasink", "Datasource", "ExecutionOptions",
+    "ExecutionResources", "FileShuffleConfig", "MaterializedDataset",
+    "NodeIdStr", "Preprocessor", "ReadTask", "RowBasedFileDatasink",
+    "SaveMode", "Schema", "SinkMode", "TaskPoolStrategy",
+    "range", "range_tensor", "from_items",
+    "from_numpy", "from_numpy_refs", "from_pandas", "from_pandas_refs",
+    "from_arrow", "from_arrow_refs", "from_torch",
+    "from_huggingface", "read_parquet", "read_csv", "read_datasource",
+    "read_json", "read_sql", "read_text",
     "read_binary_files", "read_numpy", "col", "lit",
 ]
 
@@ -208,3 +276,146 @@ def read_numpy(paths, *, parallelism: int = -1, **kw) -> Dataset:
 
     return Dataset([ReadOp(name="ReadNumpy",
                            read_tasks=_file_read_tasks(paths, reader, [".npy"]))])
+
+
+def read_datasource(datasource, *, parallelism: int = -1,
+                    override_num_blocks=None, **read_args) -> Dataset:
+    """Read via a custom Datasource (parity: reference read_datasource →
+    datasource.get_read_tasks, one Ray task per ReadTask)."""
+    p = override_num_blocks or (parallelism if parallelism > 0 else 8)
+    tasks = datasource.get_read_tasks(p, **read_args) \
+        if read_args else datasource.get_read_tasks(p)
+
+    def _one(t):
+        def run(t=t):
+            blocks = list(t() if callable(t) else t)
+            if len(blocks) == 1:
+                return blocks[0]
+            return blocks
+
+        return run
+
+    return Dataset([ReadOp(name=f"Read{datasource.get_name()}",
+                           read_tasks=[_one(t) for t in tasks])])
+
+
+def from_pandas_refs(dfs) -> Dataset:
+    """Dataset from ObjectRefs holding pandas DataFrames (parity:
+    reference from_pandas_refs); frames stay in the object store and are
+    converted to blocks inside read tasks."""
+    import ant_ray_amd as ray
+
+    refs = [dfs] if not isinstance(dfs, list) else dfs
+
+    def conv(ref):
+        def run(ref=ref):
+            import pyarrow as pa
+
+            return pa.Table.from_pandas(ray.get(ref), preserve_index=False)
+
+        return run
+
+    return Dataset([ReadOp(name="FromPandasRefs",
+                           read_tasks=[conv(r) for r in refs])])
+
+
+def from_numpy_refs(arrs, column: str = "data") -> Dataset:
+    import ant_ray_amd as ray
+
+    refs = [arrs] if not isinstance(arrs, list) else arrs
+
+    def conv(ref):
+        def run(ref=ref):
+            return {column: ray.get(ref)}
+
+        return run
+
+    return Dataset([ReadOp(name="FromNumpyRefs",
+                           read_tasks=[conv(r) for r in refs])])
+
+
+def from_arrow_refs(tables) -> Dataset:
+    import ant_ray_amd as ray
+
+    refs = [tables] if not isinstance(tables, list) else tables
+
+    def conv(ref):
+        def run(ref=ref):
+            return ray.get(ref)
+
+        return run
+
+    return Dataset([ReadOp(name="FromArrowRefs",
+                           read_tasks=[conv(r) for r in refs])])
+
+
+def read_sql(sql: str, connection_factory, *, parallelism: int = -1,
+             **_) -> Dataset:
+    """Read the results of a SQL query via a DB-API connection factory
+    (parity: reference read_sql; works offline with sqlite3)."""
+
+    def run():
+        conn = connection_factory()
+        try:
+            cur = conn.cursor()
+            cur.execute(sql)
+            cols = [d[0] for d in cur.description]
+            rows = cur.fetchall()
+        finally:
+            conn.close()
+        return {c: [r[i] for r in rows] for i, c in enumerate(cols)}
+
+    return Dataset([ReadOp(name="ReadSQL", read_tasks=[run])])
+
+
+def _unavailable_reader(name: str, needs: str):
+    def fn(*_a, **_k):
+        raise NotImplementedError(
+            f"ray.data.{name} requires {needs}, which is not available in "
+            "this air-gapped MI355X image (no network egress, package not "
+            "installed). See PARITY.md 'Known reductions'.")
+
+    fn.__name__ = name
+    return fn
+
+
+# cloud / third-party-backed readers and converters: explicit,
+# informative stubs (the packages/services they need don't exist here)
+read_images = _unavailable_reader("read_images", "an image decoder (PIL)")
+read_audio = _unavailable_reader("read_audio", "an audio decoder")
+read_videos = _unavailable_reader("read_videos", "a video decoder")
+read_avro = _unavailable_reader("read_avro", "fastavro")
+read_tfrecords = _unavailable_reader("read_tfrecords", "tensorflow protos")
+read_webdataset = _unavailable_reader("read_webdataset", "webdataset")
+read_clickhouse = _unavailable_reader("read_clickhouse", "clickhouse-connect")
+read_delta = _unavailable_reader("read_delta", "deltalake")
+read_delta_sharing_tables = _unavailable_reader(
+    "read_delta_sharing_tables", "delta-sharing")
+read_hudi = _unavailable_reader("read_hudi", "hudi")
+read_iceberg = _unavailable_reader("read_iceberg", "pyiceberg")
+read_kafka = _unavailable_reader("read_kafka", "kafka-python")
+read_lance = _unavailable_reader("read_lance", "lance")
+read_mcap = _unavailable_reader("read_mcap", "mcap")
+read_mongo = _unavailable_reader("read_mongo", "pymongo")
+read_snowflake = _unavailable_reader("read_snowflake", "snowflake-connector")
+read_unity_catalog = _unavailable_reader("read_unity_catalog",
+                                         "databricks APIs")
+from_daft = _unavailable_reader("from_daft", "daft")
+from_dask = _unavailable_reader("from_dask", "dask")
+from_mars = _unavailable_reader("from_mars", "mars")
+from_modin = _unavailable_reader("from_modin", "modin")
+from_spark = _unavailable_reader("from_spark", "pyspark")
+from_tf = _unavailable_reader("from_tf", "tensorflow")
+
+
+class KafkaAuthConfig:
+    """Stub config (parity name: reference read_kafka auth config);
+    read_kafka itself is unavailable offline."""
+
+
+class ClickHouseTableSettings:
+    """Stub config (parity name: reference read_clickhouse settings)."""
+
+
+class TFXReadOptions:
+    """Stub config (parity name: reference read_tfrecords TFX options)."""

@@ -296,7 +296,7 @@ class Dataset:
 
     def schema(self):
         for block in self.iter_blocks():
-            return BlockAccessor(block).schema()
+            return Schema(BlockAccessor(block).schema())
         return None
 
     def columns(self) -> List[str]:
@@ -414,6 +414,39 @@ class Dataset:
             refs.append(write_remote.remote(ref, path, fmt, i))
         ray.get(refs)
 
+    def write_datasink(self, datasink, *, concurrency=None, **_):
+        """Write through a custom Datasink (parity: reference
+        Dataset.write_datasink → datasink.py lifecycle): driver runs
+        on_write_start, one Ray task per block bundle runs write(),
+        driver finishes with on_write_complete / on_write_failed."""
+        datasink.on_write_start()
+
+        def _task(block, sink, idx):
+            return sink.write([block], {"task_index": idx})
+
+        try:
+            w = ray.remote(_task)
+            results = ray.get([w.remote(ref, datasink, i)
+                               for i, ref in
+                               enumerate(self.iter_internal_ref_bundles())])
+        except Exception as e:
+            datasink.on_write_failed(e)
+            raise
+        datasink.on_write_complete(results)
+        return results
+
+    def summary(self, *, columns=None):
+        """Per-column statistics (parity: reference Dataset.summary →
+        data/stats.py DatasetSummary): count/min/max/mean/std for numeric
+        columns, count/missing for the rest."""
+        from ant_ray_amd.data.stats import DatasetSummary, _summarize_blocks
+
+        schema = self.schema()
+        cols = columns or (schema.names if schema else [])
+        stats = _summarize_blocks(self, cols)
+        return DatasetSummary(dataset_schema=schema, columns=cols,
+                              stats=stats)
+
     def __repr__(self):
         return self.stats()
 
@@ -521,3 +554,28 @@ class GroupedData:
             return groupby_exchange(refs, key, finalize)
 
         return self.ds._with(AllToAllOp(name="MapGroups", fn=_do))
+
+
+class Schema:
+    """Dataset schema wrapper (parity: reference data/dataset.py Schema —
+    .names / .types over the underlying pyarrow schema)."""
+
+    def __init__(self, base_schema):
+        self.base_schema = base_schema
+
+    @property
+    def names(self):
+        return list(self.base_schema.names)
+
+    @property
+    def types(self):
+        return list(self.base_schema.types)
+
+    def __eq__(self, other):
+        o = other.base_schema if isinstance(other, Schema) else other
+        return self.base_schema == o
+
+    def __repr__(self):
+        cols = ", ".join(f"{n}: {t}" for n, t in
+                         zip(self.names, self.types))
+        return f"Schema({cols})"

@@ -218,3 +218,115 @@ def test_join_and_zip(data):
     assert len(z) == 6
 
     assert left.unique("id") == [0, 1, 2, 3, 4, 5]
+
+
+def test_custom_datasource_and_datasink(ray_mod, tmp_path):
+    """Datasource/ReadTask → read_datasource; RowBased/BlockBased file
+    datasinks with SaveMode semantics (parity: data/datasource/)."""
+    import os
+
+    import ant_ray_amd.data as rd
+    from ant_ray_amd.data import (Datasource, ReadTask,
+                                  RowBasedFileDatasink, SaveMode)
+
+    class SquaresSource(Datasource):
+        def __init__(self, n):
+            self.n = n
+
+        def get_read_tasks(self, parallelism):
+            per = max(1, self.n // parallelism)
+            tasks = []
+            for s in range(0, self.n, per):
+                e = min(s + per, self.n)
+                tasks.append(ReadTask(
+                    lambda s=s, e=e: [{"id": list(range(s, e)),
+                                       "sq": [i * i for i in range(s, e)]}],
+                    metadata={"num_rows": e - s}))
+            return tasks
+
+    ds = rd.read_datasource(SquaresSource(100), parallelism=5)
+    assert ds.count() == 100
+    assert sorted(r["sq"] for r in ds.take_all())[-1] == 99 * 99
+
+    class CsvRowSink(RowBasedFileDatasink):
+        def write_row_to_file(self, row, file):
+            file.write(f"{row['id']},{row['sq']}\n".encode())
+
+    out = str(tmp_path / "sink_out")
+    sink = CsvRowSink(out, file_format="csv")
+    ds.write_datasink(sink)
+    lines = []
+    for f in os.listdir(out):
+        with open(os.path.join(out, f)) as fh:
+            lines += [ln for ln in fh.read().splitlines() if ln]
+    assert len(lines) == 100
+
+    # SaveMode.ERROR refuses a non-empty dir
+    with pytest.raises(ValueError):
+        rd.range(3).write_datasink(CsvRowSink(out, file_format="csv",
+                                              mode=SaveMode.ERROR))
+
+
+def test_from_refs_read_sql_summary(ray_mod, tmp_path):
+    """from_pandas_refs/from_numpy_refs/from_arrow_refs, read_sql over
+    sqlite3, Dataset.summary, Preprocessor (parity: data/read_api.py,
+    stats.py, preprocessor.py)."""
+    import sqlite3
+
+    import numpy as np
+    import pandas as pd
+    import pyarrow as pa
+
+    import ant_ray_amd as ray
+    import ant_ray_amd.data as rd
+
+    dfs = [pd.DataFrame({"x": [1, 2]}), pd.DataFrame({"x": [3, 4]})]
+    ds = rd.from_pandas_refs([ray.put(d) for d in dfs])
+    assert sorted(r["x"] for r in ds.take_all()) == [1, 2, 3, 4]
+
+    ds = rd.from_numpy_refs([ray.put(np.arange(3))], column="v")
+    assert sorted(r["v"] for r in ds.take_all()) == [0, 1, 2]
+
+    ds = rd.from_arrow_refs([ray.put(pa.table({"y": [7, 8]}))])
+    assert sorted(r["y"] for r in ds.take_all()) == [7, 8]
+
+    # read_sql against sqlite
+    db = str(tmp_path / "t.db")
+    conn = sqlite3.connect(db)
+    conn.execute("create table t (a int, b real)")
+    conn.executemany("insert into t values (?, ?)",
+                     [(i, i * 0.5) for i in range(10)])
+    conn.commit()
+    conn.close()
+    ds = rd.read_sql("select * from t where a >= 2",
+                     lambda: sqlite3.connect(db))
+    assert ds.count() == 8 and ds.schema().names == ["a", "b"]
+
+    # summary stats
+    summ = rd.range(100).summary()
+    st = summ.column_stats("id")
+    assert st["count"] == 100 and st["min"] == 0 and st["max"] == 99
+    assert abs(st["mean"] - 49.5) < 1e-9
+    assert "id" in summ.to_pandas().columns
+
+    # Preprocessor
+    from ant_ray_amd.data import Preprocessor
+
+    class Center(Preprocessor):
+        def _fit(self, ds):
+            self.mean_ = ds.mean("id").take_all()[0]["mean(id)"]
+
+        def _transform_pandas(self, df):
+            df["id"] = df["id"] - self.mean_
+            return df
+
+    pre = Center()
+    with pytest.raises(Exception):
+        pre.transform(rd.range(10))  # not fitted
+    out = pre.fit_transform(rd.range(10))
+    vals = sorted(r["id"] for r in out.take_all())
+    assert abs(vals[0] + 4.5) < 1e-9 and abs(vals[-1] - 4.5) < 1e-9
+
+    # unavailable readers raise informative errors
+    with pytest.raises(NotImplementedError):
+        rd.read_images("/tmp/x")
