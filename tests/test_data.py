@@ -314,7 +314,7 @@ def test_from_refs_read_sql_summary(ray_mod, tmp_path):
 
     class Center(Preprocessor):
         def _fit(self, ds):
-            self.mean_ = ds.mean("id").take_all()[0]["mean(id)"]
+            self.mean_ = ds.summary(columns=["id"]).column_stats("id")["mean"]
 
         def _transform_pandas(self, df):
             df["id"] = df["id"] - self.mean_
