@@ -47,3 +47,44 @@ import cloudpickle as _cp  # noqa: E402
 
 _sys.modules["ray.cloudpickle"] = _cp
 cloudpickle = _cp
+
+# Any other `ray.X.Y` import resolves to `ant_ray_amd.X.Y` via a meta-path
+# finder, so arbitrary-depth reference imports (ray.runtime_context,
+# ray.workflow, ray.serve.handle, ray._private.worker, ...) work without
+# enumerating them.
+import importlib as _importlib  # noqa: E402
+import importlib.abc as _ilabc  # noqa: E402
+import importlib.util as _ilutil  # noqa: E402
+
+
+class _AliasLoader(_ilabc.Loader):
+    def __init__(self, impl):
+        self._impl = impl
+
+    def create_module(self, spec):
+        return self._impl
+
+    def exec_module(self, module):
+        pass
+
+
+class _RayAliasFinder(_ilabc.MetaPathFinder):
+    def find_spec(self, fullname, path=None, target=None):
+        if not fullname.startswith("ray."):
+            return None
+        try:
+            impl = _importlib.import_module("ant_ray_amd." + fullname[4:])
+        except ImportError:
+            return None
+        return _ilutil.spec_from_loader(fullname, _AliasLoader(impl))
+
+
+_sys.meta_path.append(_RayAliasFinder())
+
+
+def __getattr__(name):
+    # `from ray import workflow` etc without a prior submodule import
+    try:
+        return _importlib.import_module(f"ant_ray_amd.{name}")
+    except ImportError:
+        raise AttributeError(f"module 'ray' has no attribute {name!r}")

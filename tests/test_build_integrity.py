@@ -27,3 +27,24 @@ def test_shm_store_loads():
 def test_gpu_ipc_loads():
     pytest.importorskip("torch")
     import ant_ray_amd._gpu_ipc  # noqa: F401
+
+
+def test_ray_alias_deep_imports():
+    """Arbitrary-depth `ray.*` imports resolve to ant_ray_amd modules via
+    the alias meta-path finder (user code written for the reference works
+    unchanged)."""
+    import importlib
+
+    import ray
+
+    for name in ["ray.runtime_context", "ray.types", "ray.cross_language",
+                 "ray.job_config", "ray.workflow", "ray.air",
+                 "ray.serve.handle", "ray.data.datasource",
+                 "ray.tune.stopper", "ray.util.debug",
+                 "ray._private.serialization"]:
+        mod = importlib.import_module(name)
+        assert mod.__name__.startswith("ant_ray_amd"), (name, mod.__name__)
+    from ray.job_config import JobConfig
+
+    assert JobConfig(ray_namespace="n").serialize()["ray_namespace"] == "n"
+    assert ray.types.ObjectRef is not None
