@@ -60,12 +60,23 @@ import importlib.util as _ilutil  # noqa: E402
 class _AliasLoader(_ilabc.Loader):
     def __init__(self, impl):
         self._impl = impl
+        # module_from_spec clobbers __name__/__spec__/__loader__ on the
+        # (shared) impl module; save them to restore in exec_module so the
+        # canonical ant_ray_amd identity survives (pickle-by-reference of
+        # its functions depends on __name__)
+        self._saved = {k: getattr(impl, k, None)
+                       for k in ("__name__", "__spec__", "__loader__",
+                                 "__package__")}
 
     def create_module(self, spec):
         return self._impl
 
     def exec_module(self, module):
-        pass
+        for k, v in self._saved.items():
+            try:
+                setattr(module, k, v)
+            except Exception:
+                pass
 
 
 class _RayAliasFinder(_ilabc.MetaPathFinder):
@@ -79,7 +90,7 @@ class _RayAliasFinder(_ilabc.MetaPathFinder):
         return _ilutil.spec_from_loader(fullname, _AliasLoader(impl))
 
 
-_sys.meta_path.append(_RayAliasFinder())
+_sys.meta_path.insert(0, _RayAliasFinder())
 
 
 def __getattr__(name):
