@@ -48,3 +48,23 @@ def test_ray_alias_deep_imports():
 
     assert JobConfig(ray_namespace="n").serialize()["ray_namespace"] == "n"
     assert ray.types.ObjectRef is not None
+
+
+def test_experimental_namespace():
+    """ray.experimental parity members: tqdm_ray, compiled_dag_ref, queue,
+    object locations (reference python/ray/experimental/)."""
+    import io
+
+    import ray
+    from ray.experimental import tqdm_ray
+    from ray.experimental.compiled_dag_ref import CompiledDAGRef  # noqa: F401
+    from ray.experimental.queue import Empty, Queue  # noqa: F401
+
+    bar = tqdm_ray.tqdm(total=5, desc="x")
+    for _ in range(5):
+        bar.update(1)
+    bar.set_description("done")
+    bar.close()
+    tqdm_ray.safe_print("safe")
+    assert list(tqdm_ray.tqdm(iterable=[1, 2, 3])) == [1, 2, 3]
+    tqdm_ray.instance().unhide_bars()
