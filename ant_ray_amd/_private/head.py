@@ -93,6 +93,9 @@ async def run_head(args):
 
 
 def main():
+    from ant_ray_amd._private.stack_dump import install as _stack_install
+
+    _stack_install()
     ap = argparse.ArgumentParser()
     ap.add_argument("--host", default="127.0.0.1")
     ap.add_argument("--port", type=int, default=0)

@@ -12,6 +12,9 @@ import threading
 
 
 def main():
+    from ant_ray_amd._private.stack_dump import install as _stack_install
+
+    _stack_install()
     logging.basicConfig(level=logging.INFO)
     worker_id = bytes.fromhex(os.environ["ANTRAY_WORKER_ID"])
     gcs_host, gcs_port = os.environ["ANTRAY_GCS"].rsplit(":", 1)

@@ -219,6 +219,149 @@ def summary(kind):
 # ----------------------------------------------------------------------- job
 
 
+def _latest_session():
+    """session_dir of the most recent local cluster (head.json)."""
+    import json as _json
+
+    base = "/tmp/antray"
+    cands = []
+    if os.path.isdir(base):
+        for d in os.listdir(base):
+            hp = os.path.join(base, d, "head.json")
+            if os.path.exists(hp):
+                cands.append(hp)
+    if not cands:
+        return None
+    with open(sorted(cands)[-1]) as f:
+        return _json.load(f).get("session_dir")
+
+
+@cli.command()
+@click.argument("filename", required=False)
+@click.option("--tail", type=int, default=100, help="lines from the end")
+def logs(filename, tail):
+    """List session log files, or tail one (parity `ray logs`)."""
+    sd = _latest_session()
+    if not sd:
+        click.echo("no running cluster found")
+        raise SystemExit(1)
+    logs_dir = os.path.join(sd, "logs")
+    if not os.path.isdir(logs_dir):
+        click.echo("no logs directory")
+        return
+    if not filename:
+        for f in sorted(os.listdir(logs_dir)):
+            sz = os.path.getsize(os.path.join(logs_dir, f))
+            click.echo(f"{f}\t{sz}")
+        return
+    fp = os.path.join(logs_dir, os.path.basename(filename))
+    if not os.path.isfile(fp):
+        click.echo(f"no such log: {filename}")
+        raise SystemExit(1)
+    with open(fp, errors="replace") as f:
+        for line in f.read().splitlines()[-tail:]:
+            click.echo(line)
+
+
+@cli.command("health-check")
+def health_check():
+    """Exit 0 if the local cluster's GCS answers (parity
+    `ray health-check`)."""
+    import ant_ray_amd as ray
+
+    try:
+        ray.init(address="auto", ignore_reinit_error=True)
+        ray.cluster_resources()
+        click.echo("ok")
+    except Exception as e:
+        click.echo(f"unhealthy: {e}")
+        raise SystemExit(1)
+
+
+@cli.command()
+def stack():
+    """Dump python stacks of this node's runtime processes into the
+    session logs (parity `ray stack`; SIGUSR1 → faulthandler — py-spy is
+    not in this image)."""
+    import signal as _signal
+    import time as _time
+
+    import psutil
+
+    me = os.getpid()
+    hit = []
+    for proc in psutil.process_iter(["pid", "cmdline"]):
+        try:
+            cmd = " ".join(proc.info["cmdline"] or [])
+            if proc.info["pid"] != me and (
+                    "_private/head.py" in cmd
+                    or "default_worker.py" in cmd):
+                proc.send_signal(_signal.SIGUSR1)
+                hit.append((proc.info["pid"], cmd.split()[-1][-60:]))
+        except (psutil.NoSuchProcess, psutil.AccessDenied):
+            continue
+    if not hit:
+        click.echo("no runtime processes found")
+        return
+    _time.sleep(0.5)
+    for pid, what in hit:
+        click.echo(f"signaled {pid} ({what})")
+    click.echo("stacks dumped to each process's stderr "
+               "(see `ray logs`)")
+
+
+@cli.command("drain-node")
+@click.option("--node-id", default=None,
+              help="hex node id (default: every alive node)")
+@click.option("--timeout-s", type=float, default=30.0)
+def drain_node(node_id, timeout_s):
+    """Gracefully drain a raylet: running leases finish, queued work
+    spills elsewhere (parity `ray drain-node`)."""
+    import asyncio
+
+    import ant_ray_amd as ray
+    from ant_ray_amd._private import protocol
+    from ant_ray_amd._private.worker import global_worker
+
+    ray.init(address="auto", ignore_reinit_error=True)
+    cw = global_worker.core_worker
+    nodes = cw.io.run(cw.gcs.call("node_table", {}, timeout=10), timeout=15)
+    targets = [n for n in nodes if n.get("alive")
+               and (node_id is None or n["node_id"].hex() == node_id)]
+    if not targets:
+        click.echo("no matching alive node")
+        raise SystemExit(1)
+
+    async def _drain(addr):
+        conn = await protocol.connect(tuple(addr), None, name="drain-cli")
+        return await conn.call("drain", {"timeout_s": timeout_s}, timeout=10)
+
+    for n in targets:
+        try:
+            r = asyncio.run(_drain(n["addr"]))
+            click.echo(f"node {n['node_id'].hex()[:16]}: draining={r.get('draining')}")
+        except Exception as e:
+            click.echo(f"node {n['node_id'].hex()[:16]}: drain failed: {e}")
+
+
+@cli.command("disable-usage-stats")
+def disable_usage_stats():
+    """Persist usage-stats opt-out (parity; no data ever leaves this
+    air-gapped deployment either way)."""
+    os.makedirs(os.path.expanduser("~/.ray"), exist_ok=True)
+    with open(os.path.expanduser("~/.ray/usage_stats.json"), "w") as f:
+        f.write('{"usage_stats": false}')
+    click.echo("usage stats disabled")
+
+
+@cli.command("enable-usage-stats")
+def enable_usage_stats():
+    os.makedirs(os.path.expanduser("~/.ray"), exist_ok=True)
+    with open(os.path.expanduser("~/.ray/usage_stats.json"), "w") as f:
+        f.write('{"usage_stats": true}')
+    click.echo("usage stats enabled (local only: no egress)")
+
+
 @cli.group()
 def job():
     """Job submission (parity `ray job ...`)."""

@@ -1,4 +1,5 @@
 """State API, CLI, and job submission tests."""
+import os
 import time
 
 import pytest
@@ -144,3 +145,31 @@ def test_llm_alias_imports():
 
     assert hasattr(sllm, "build_openai_app")
     assert hasattr(dllm, "build_llm_processor")
+
+
+def test_cli_logs_health_stack_drain(ray_mod):
+    """New CLI commands: logs / health-check / stack / drain-node
+    (parity: reference scripts/scripts.py surface)."""
+    import subprocess
+    import sys
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = "/root/repo"
+
+    def run(*args):
+        return subprocess.run(
+            [sys.executable, "-m", "ant_ray_amd.scripts.cli", *args],
+            capture_output=True, text=True, env=env, timeout=120)
+
+    r = run("health-check")
+    assert r.returncode == 0 and "ok" in r.stdout, (r.stdout, r.stderr)
+
+    r = run("logs")
+    assert r.returncode == 0, r.stderr
+
+    r = run("stack")
+    assert r.returncode == 0, r.stderr
+    assert "signaled" in r.stdout or "no runtime processes" in r.stdout
+
+    r = run("disable-usage-stats")
+    assert r.returncode == 0 and "disabled" in r.stdout
