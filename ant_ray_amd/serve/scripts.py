@@ -99,6 +99,57 @@ def deploy(config_file):
 
 
 @cli.command()
+def config():
+    """Print the currently-deployed applications as a YAML config
+    (parity: `serve config` — the inverse of `serve deploy`)."""
+    import yaml
+
+    import ant_ray_amd as ray
+    from ant_ray_amd import serve
+
+    if not ray.is_initialized():
+        try:
+            ray.init(address="auto", ignore_reinit_error=True)
+        except Exception:
+            raise click.ClickException("no running cluster found")
+    st = serve.status().get("applications", {})
+    apps = []
+    for name, info in st.items():
+        apps.append({"name": name,
+                     "route_prefix": info.get("route_prefix", "/"),
+                     "deployments": sorted((info.get("deployments") or {}))})
+    click.echo(yaml.safe_dump({"applications": apps}, sort_keys=False))
+
+
+@cli.command()
+@click.argument("import_paths", nargs=-1, required=True)
+@click.option("-o", "--output-path", default=None,
+              help="write the config YAML here instead of stdout")
+def build(import_paths, output_path):
+    """Generate a deployable YAML config from application import paths
+    (parity: `serve build`)."""
+    import yaml
+
+    apps = []
+    for i, path in enumerate(import_paths):
+        _load_target(path)  # validate it imports and is an Application
+        name = f"app{i + 1}" if len(import_paths) > 1 else "default"
+        apps.append({"name": name,
+                     "route_prefix": "/" if len(import_paths) == 1
+                     else f"/{name}",
+                     "import_path": path})
+    doc = yaml.safe_dump(
+        {"http_options": {"host": "127.0.0.1", "port": 8000},
+         "applications": apps}, sort_keys=False)
+    if output_path:
+        with open(output_path, "w") as f:
+            f.write(doc)
+        click.echo(f"wrote {output_path}")
+    else:
+        click.echo(doc)
+
+
+@cli.command()
 def status():
     """Show application/deployment status."""
     import ant_ray_amd as ray

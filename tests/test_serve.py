@@ -751,3 +751,47 @@ def test_replica_context_and_run_many(serve_mod):
     assert h2.remote(None).result(timeout_s=30) == "plain-ok"
     serve.delete("who")
     serve.delete("plain_app")
+
+
+def test_serve_cli_build_and_config(serve_mod, tmp_path, monkeypatch):
+    """serve build generates a deployable YAML; serve config prints the
+    live state (parity: reference serve CLI build/config)."""
+    import sys
+
+    import yaml
+    from click.testing import CliRunner
+
+    from ant_ray_amd.serve.scripts import cli
+
+    serve = serve_mod
+    app_mod = tmp_path / "cli_bc_mod.py"
+    app_mod.write_text(
+        "from ant_ray_amd import serve\n"
+        "@serve.deployment\n"
+        "def ping(request=None):\n"
+        "    return 'pong'\n"
+        "app = ping.bind()\n"
+    )
+    monkeypatch.chdir(tmp_path)
+    sys.path.insert(0, str(tmp_path))
+    try:
+        out_yaml = tmp_path / "built.yaml"
+        r = CliRunner().invoke(cli, ["build", "cli_bc_mod:app",
+                                     "-o", str(out_yaml)])
+        assert r.exit_code == 0, r.output
+        doc = yaml.safe_load(out_yaml.read_text())
+        assert doc["applications"][0]["import_path"] == "cli_bc_mod:app"
+
+        for app_name in list(serve.status().get("applications", {})):
+            serve.delete(app_name)
+        r = CliRunner().invoke(cli, ["deploy", str(out_yaml)])
+        assert r.exit_code == 0, r.output
+
+        r = CliRunner().invoke(cli, ["config"])
+        assert r.exit_code == 0, r.output
+        cfg = yaml.safe_load(r.output)
+        names = [a["name"] for a in cfg["applications"]]
+        assert "default" in names
+        serve.delete("default")
+    finally:
+        sys.path.remove(str(tmp_path))
