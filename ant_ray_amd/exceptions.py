@@ -109,3 +109,133 @@ class PendingCallsLimitExceeded(RayError):
     """Raised when an actor handle with max_pending_calls set already has
     that many calls outstanding (parity: ray.exceptions
     .PendingCallsLimitExceeded)."""
+
+
+class WorkerCrashedError(RayError):
+    """The worker executing a task died unexpectedly (parity:
+    reference exceptions.py WorkerCrashedError)."""
+
+
+class LocalRayletDiedError(RayError):
+    """This node's raylet died while the worker depended on it."""
+
+
+class NodeDiedError(RayError):
+    """The node a task/object lived on died."""
+
+
+class TaskUnschedulableError(RayError):
+    """The task can never be scheduled (infeasible resources after
+    cluster-shape changes, or scheduling constraints)."""
+
+    def __init__(self, error_message: str = ""):
+        self.error_message = error_message
+        super().__init__(error_message)
+
+
+class ActorUnschedulableError(TaskUnschedulableError):
+    """The actor can never be scheduled."""
+
+
+class TaskPlacementGroupRemoved(RayError):
+    """The placement group a queued task targeted was removed."""
+
+
+class ActorPlacementGroupRemoved(RayError):
+    """The placement group a pending actor targeted was removed."""
+
+
+class ActorAlreadyExistsError(RayError):
+    """A named actor with this name already exists
+    (get_if_exists=False)."""
+
+
+class AsyncioActorExit(RayError):
+    """Internal marker used to unwind an async actor on exit_actor()."""
+
+
+class ObjectFreedError(ObjectLostError):
+    """The object was explicitly freed via ray.internal.free."""
+
+
+class ObjectFetchTimedOutError(ObjectLostError):
+    """Fetching the object from its holder timed out."""
+
+
+class ObjectReconstructionFailedError(ObjectLostError):
+    """Lineage reconstruction could not recompute the object."""
+
+
+class ObjectReconstructionFailedMaxAttemptsExceededError(
+        ObjectReconstructionFailedError):
+    """Reconstruction gave up after max_retries resubmissions."""
+
+
+class ObjectReconstructionFailedLineageEvictedError(
+        ObjectReconstructionFailedError):
+    """Reconstruction impossible: the lineage buffer evicted the
+    producing task (see the bounded lineage cache in
+    _private/worker.py)."""
+
+
+class ReferenceCountingAssertionError(ObjectLostError, AssertionError):
+    """The object was deleted while this process still held a
+    reference — a reference-protocol invariant was violated."""
+
+
+class ObjectRefStreamEndOfStreamError(RayError):
+    """Internal: a streaming generator's stream is exhausted (surfaced
+    as StopIteration/StopAsyncIteration to user code)."""
+
+
+class OutOfDiskError(RayError):
+    """The object store spill directory ran out of disk."""
+
+
+class OufOfBandObjectRefSerializationException(RayError):
+    """An ObjectRef was pickled outside ray serialization (name kept
+    with the reference's spelling, typo included)."""
+
+
+class PlasmaObjectNotAvailable(RayError):
+    """The requested object bytes are not available in the local store."""
+
+
+class UnserializableException(RayError):
+    """The exception raised by user code could not be pickled; this
+    carries its string form instead."""
+
+
+class UserCodeException(RayError):
+    """Wrapper marking that the failure originated in user code, not
+    the runtime."""
+
+
+class RpcError(RayError):
+    """A low-level RPC failed (parity: reference RpcError; the
+    transport's own error type lives in _private/protocol.py)."""
+
+    def __init__(self, message: str = "", rpc_code: int = None):
+        self.rpc_code = rpc_code
+        super().__init__(message)
+
+
+class RayChannelError(RaySystemError):
+    """Compiled-DAG channel failure (parity: experimental.channel
+    errors): the peer actor died or the channel was closed."""
+
+    def __init__(self, message: str = ""):
+        super().__init__(message)
+
+
+class RayChannelTimeoutError(RayChannelError, TimeoutError):
+    """Compiled-DAG channel read/write timed out."""
+
+
+class RayCgraphCapacityExceeded(RaySystemError):
+    """A compiled-graph channel's buffer capacity was exceeded."""
+
+
+class AuthenticationError(RayError):
+    """Cluster authentication failed (no auth layer in this air-gapped
+    single-tenant build; exists for API parity)."""
