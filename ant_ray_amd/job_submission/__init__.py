@@ -181,3 +181,72 @@ class JobStatus:
     SUCCEEDED = "SUCCEEDED"
     FAILED = "FAILED"
     STOPPED = "STOPPED"
+
+
+class JobType:
+    """How the job entered the cluster (parity: reference
+    job_submission JobType)."""
+
+    SUBMISSION = "SUBMISSION"  # via JobSubmissionClient / `ray job submit`
+    DRIVER = "DRIVER"          # a directly-connected driver
+
+
+class JobErrorType:
+    """Failure classification (parity name)."""
+
+    APPLICATION = "APPLICATION"   # user entrypoint exited non-zero
+    SYSTEM = "SYSTEM"             # runtime failure (worker/node death)
+
+
+from dataclasses import dataclass, field  # noqa: E402
+from typing import Any, Dict, Optional  # noqa: E402
+
+
+@dataclass
+class DriverInfo:
+    """Driver process of a job (parity: reference DriverInfo)."""
+
+    id: str
+    node_ip_address: str = "127.0.0.1"
+    pid: Optional[int] = None
+
+
+@dataclass
+class JobInfo:
+    """Submission-job metadata (parity: reference JobInfo)."""
+
+    status: str
+    entrypoint: str = ""
+    message: Optional[str] = None
+    error_type: Optional[str] = None
+    start_time: Optional[int] = None
+    end_time: Optional[int] = None
+    metadata: Dict[str, str] = field(default_factory=dict)
+    runtime_env: Dict[str, Any] = field(default_factory=dict)
+    entrypoint_num_cpus: Optional[float] = None
+    entrypoint_num_gpus: Optional[float] = None
+    submission_id: Optional[str] = None
+
+
+@dataclass
+class JobDetails(JobInfo):
+    """JobInfo + identity fields, the shape `ray job list`/the REST API
+    returns (parity: reference JobDetails)."""
+
+    job_id: Optional[str] = None
+    type: str = JobType.SUBMISSION
+    driver_info: Optional[DriverInfo] = None
+
+    @classmethod
+    def _from_info_dict(cls, d: Dict[str, Any]) -> "JobDetails":
+        return cls(
+            status=d.get("status", "PENDING"),
+            entrypoint=d.get("entrypoint", ""),
+            message=d.get("message"),
+            start_time=d.get("start_time"),
+            end_time=d.get("end_time"),
+            metadata=d.get("metadata") or {},
+            runtime_env=d.get("runtime_env") or {},
+            submission_id=d.get("submission_id") or d.get("job_id"),
+            job_id=d.get("job_id"),
+        )

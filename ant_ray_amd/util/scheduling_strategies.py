@@ -32,3 +32,38 @@ class NodeLabelSchedulingStrategy:
 
 DEFAULT_SCHEDULING_STRATEGY = "DEFAULT"
 SPREAD_SCHEDULING_STRATEGY = "SPREAD"
+
+
+class _LabelMatchExpression:
+    """One label predicate (parity: reference scheduling_strategies
+    label-selector helpers). The GCS label selector consumes its
+    dict form."""
+
+    def __init__(self, key: str, operator: str, values=None):
+        self.key = key
+        self.operator = operator
+        self.values = list(values or [])
+
+    def to_dict(self):
+        return {"key": self.key, "op": self.operator,
+                "values": self.values}
+
+
+class In(_LabelMatchExpression):
+    def __init__(self, *values):
+        super().__init__("", "in", values)
+
+
+class NotIn(_LabelMatchExpression):
+    def __init__(self, *values):
+        super().__init__("", "not_in", values)
+
+
+class Exists(_LabelMatchExpression):
+    def __init__(self):
+        super().__init__("", "exists")
+
+
+class DoesNotExist(_LabelMatchExpression):
+    def __init__(self):
+        super().__init__("", "does_not_exist")

@@ -211,3 +211,101 @@ def get_timeline(limit: int = 20000) -> List[dict]:
             "args": {"task_id": e.get("task_id"), "state": e.get("state")},
         })
     return trace
+
+
+def get_task(task_id: str) -> Optional[Dict[str, Any]]:
+    """Single task-event view by hex id (parity: state.get_task)."""
+    for t in list_tasks(limit=100000):
+        tid = t.get("task_id")
+        if isinstance(tid, bytes):
+            tid = tid.hex()
+        if tid == task_id:
+            return t
+    return None
+
+
+def get_worker(worker_id: str) -> Optional[Dict[str, Any]]:
+    for w in list_workers(limit=100000):
+        if w.get("worker_id") == worker_id:
+            return w
+    return None
+
+
+def get_job(job_id: str) -> Optional[Dict[str, Any]]:
+    for j in list_jobs(limit=100000):
+        if str(j.get("job_id")) == str(job_id) \
+                or j.get("submission_id") == job_id:
+            return j
+    return None
+
+
+def get_objects(object_id: str = None, **_) -> List[Dict[str, Any]]:
+    """Object views, optionally filtered to one hex id (parity:
+    state.get_objects)."""
+    rows = list_objects(limit=100000)
+    if object_id is None:
+        return rows
+    return [r for r in rows if r.get("object_id") == object_id]
+
+
+def list_runtime_envs(filters=None, limit: int = 1000, **_):
+    """Runtime envs currently attached to live workers (parity:
+    state.list_runtime_envs; this build materializes envs at worker
+    spawn, so the distinct env specs of live workers ARE the active
+    set)."""
+    envs = []
+    seen = set()
+    for w in list_workers(limit=100000):
+        env = w.get("runtime_env")
+        key = repr(env)
+        if env and key not in seen:
+            seen.add(key)
+            envs.append({"runtime_env": env, "ref_cnt": 1,
+                         "success": True})
+        elif env and key in seen:
+            for e in envs:
+                if repr(e["runtime_env"]) == key:
+                    e["ref_cnt"] += 1
+    out = [e for e in envs if _match(e, filters)]
+    return out[:limit]
+
+
+class StateApiClient:
+    """Programmatic façade over the state functions (parity: reference
+    util/state/api.py StateApiClient — list/get by resource name)."""
+
+    def __init__(self, address: str = None, **_):
+        self.address = address
+
+    _LIST = {
+        "actors": "list_actors", "tasks": "list_tasks",
+        "nodes": "list_nodes", "workers": "list_workers",
+        "jobs": "list_jobs", "placement_groups": "list_placement_groups",
+        "objects": "list_objects", "runtime_envs": "list_runtime_envs",
+        "cluster_events": "list_cluster_events",
+    }
+    _GET = {"actors": "get_actor", "nodes": "get_node",
+            "tasks": "get_task", "workers": "get_worker",
+            "jobs": "get_job",
+            "placement_groups": "get_placement_group"}
+
+    def list(self, resource: str, options=None, raise_on_missing_output=True,
+             **kwargs):
+        fn = self._LIST.get(resource)
+        if fn is None:
+            raise ValueError(f"unknown resource {resource!r}; "
+                             f"available: {sorted(self._LIST)}")
+        opts = dict(kwargs)
+        if options is not None:
+            for k in ("filters", "limit"):
+                v = getattr(options, k, None)
+                if v is not None:
+                    opts[k] = v
+        return globals()[fn](**opts)
+
+    def get(self, resource: str, id: str, **kwargs):
+        fn = self._GET.get(resource)
+        if fn is None:
+            raise ValueError(f"unknown resource {resource!r}; "
+                             f"available: {sorted(self._GET)}")
+        return globals()[fn](id)

@@ -266,3 +266,26 @@ def test_compiled_dag_actor_death_surfaces(ray_mod):
         ray.get(ref, timeout=90)
     assert time.time() - t0 < 60, "actor death took too long to surface"
     assert isinstance(ei.value, (ActorDiedError, RuntimeError)), ei.value
+
+
+def test_dag_context_and_plot(ray_mod):
+    """DAGContext env-config singleton; plot emits dot text (parity:
+    dag/context.py:38, vis utils)."""
+    import ant_ray_amd as ray
+    from ant_ray_amd.dag import DAGContext, DAGInputData, plot
+
+    ctx = DAGContext.get_current()
+    assert ctx.buffer_size_bytes >= 1
+    assert DAGContext.get_current() is ctx
+
+    d = DAGInputData(1, 2, k=3)
+    assert d[0] == 1 and d["k"] == 3
+
+    @ray.remote
+    def add(a, b):
+        return a + b
+
+    node = add.bind(add.bind(1, 2), 3)
+    dot = plot(node)
+    assert dot.startswith("digraph") and "FunctionNode" in dot
+    assert ray.get(node.execute(), timeout=30) == 6

@@ -173,3 +173,36 @@ def test_cli_logs_health_stack_drain(ray_mod):
 
     r = run("disable-usage-stats")
     assert r.returncode == 0 and "disabled" in r.stdout
+
+
+def test_state_getters_and_client(ray_mod):
+    """get_task/get_worker/get_job/get_objects/list_runtime_envs +
+    StateApiClient (parity: util/state/api.py)."""
+    import ant_ray_amd as ray
+    from ant_ray_amd.util import state as S
+    from ant_ray_amd.util.state import StateApiClient
+
+    @ray.remote
+    def probe():
+        return 1
+
+    ray.get([probe.remote() for _ in range(3)], timeout=60)
+    tasks = S.list_tasks(limit=100)
+    assert tasks
+    tid = tasks[0].get("task_id")
+    if isinstance(tid, bytes):
+        tid = tid.hex()
+    assert S.get_task(tid) is not None
+
+    workers = S.list_workers(limit=10)
+    if workers:
+        assert S.get_worker(workers[0]["worker_id"]) is not None
+
+    c = StateApiClient()
+    assert isinstance(c.list("actors"), list)
+    assert isinstance(c.list("nodes"), list)
+    assert isinstance(c.list("runtime_envs"), list)
+    with pytest.raises(ValueError):
+        c.list("nope")
+    nodes = c.list("nodes")
+    assert c.get("nodes", nodes[0]["node_id"]) is not None
