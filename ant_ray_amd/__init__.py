@@ -133,6 +133,17 @@ def init(
     global_worker.core_worker = cw
     global_worker.mode = DRIVER_MODE
     global_worker._head_proc = head
+    if lc is not None:
+        # propagate to workers: every spawned worker applies this at boot
+        import json as _json
+
+        try:
+            cw.io.run(cw.gcs.call("kv_put", {
+                "ns": "_cluster", "key": b"logging_config",
+                "value": _json.dumps(lc._to_dict()).encode(),
+                "overwrite": True}, timeout=10), timeout=15)
+        except Exception:
+            pass
     atexit.register(shutdown)
     if include_dashboard:
         from ant_ray_amd.dashboard import start_dashboard

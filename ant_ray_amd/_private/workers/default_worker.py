@@ -42,6 +42,19 @@ def main():
         store_path=store_path,
         node_id=node_id,
     )
+    try:
+        # driver-pushed structured logging config (ray.init(logging_config=))
+        r = cw.io.run(cw.gcs.call("kv_get", {"ns": "_cluster",
+                                             "key": b"logging_config"},
+                                  timeout=5), timeout=8)
+        if r and r.get("value"):
+            import json as _json
+
+            from ant_ray_amd._private.logging_config import LoggingConfig
+
+            LoggingConfig._from_dict(_json.loads(r["value"].decode()))._apply()
+    except Exception:
+        pass
     logging.getLogger("antray.worker").info(
         "worker %s ready on %s", worker_id.hex()[:8], cw.addr
     )

@@ -329,3 +329,39 @@ def test_util_parity_helpers(ray_mod):
     remove_placement_group(pg)
     with pytest.raises(ValueError):
         get_placement_group("no_such_pg")
+
+
+def test_logging_config_propagates_to_workers():
+    """ray.init(logging_config=JSON) applies in SPAWNED WORKERS too: a
+    task's log record comes out as JSON with core context (parity:
+    reference logging_config propagation via runtime env)."""
+    import subprocess
+    import sys
+
+    code = r"""
+import sys
+sys.path.insert(0, "/root/repo")
+import ray
+ray.init(num_cpus=1, logging_config=ray.LoggingConfig(encoding="JSON"))
+
+@ray.remote
+def noisy():
+    import logging as L
+    h = L.getLogger().handlers[-1]
+    r = L.LogRecord("app", L.INFO, "t.py", 1, "probe %s", ("x",), None)
+    for f in h.filters:
+        f.filter(r)
+    return h.format(r)
+
+out = ray.get(noisy.remote(), timeout=60)
+print("FORMATTED::" + out)
+ray.shutdown()
+"""
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=180)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [ln for ln in r.stdout.splitlines()
+            if ln.startswith("FORMATTED::")][0]
+    payload = json.loads(line[len("FORMATTED::"):])
+    assert payload["message"] == "probe x"
+    assert "worker_id" in payload and "task_id" in payload
