@@ -106,6 +106,8 @@ class LoggingConfig:
                     f"Unknown standard attr {a!r}; supported: {_STANDARD_ATTRS}")
 
     def _apply(self):
+        global _applied
+        _applied = True
         fmt_cls = JSONFormatter if self.encoding == "JSON" else TextFormatter
         handler = logging.StreamHandler()
         handler.setLevel(self.log_level)
@@ -129,3 +131,6 @@ class LoggingConfig:
     @classmethod
     def _from_dict(cls, d):
         return cls(**d)
+
+
+_applied = False  # set once any config applies in this process

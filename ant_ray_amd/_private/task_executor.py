@@ -271,6 +271,30 @@ class TaskExecutor:
             pass
 
     def _execute(self, payload) -> dict:
+        if not getattr(self, "_logging_cfg_checked", False):
+            # one-time re-check: a PRESTARTED worker boots before the
+            # driver's ray.init(logging_config=...) kv_put lands; the first
+            # task strictly follows init, so this read is race-free
+            self._logging_cfg_checked = True
+            try:
+                from ant_ray_amd._private import logging_config as _lc
+
+                if _lc._applied:
+                    raise StopIteration  # boot-time read already applied it
+                r = self.cw.io.run(self.cw.gcs.call(
+                    "kv_get", {"ns": "_cluster", "key": b"logging_config"},
+                    timeout=5), timeout=8)
+                if r and r.get("value"):
+                    import json as _json
+
+                    from ant_ray_amd._private.logging_config import (
+                        LoggingConfig,
+                    )
+
+                    LoggingConfig._from_dict(
+                        _json.loads(r["value"].decode()))._apply()
+            except Exception:
+                pass
         ttype = payload["type"]
         task_id = payload["task_id"]
         if task_id in self._cancelled:
