@@ -65,6 +65,26 @@ async def run_head(args):
     resources.setdefault(f"node:{args.host}", 1.0)
     capacity = args.object_store_memory or default_store_capacity()
     resources.setdefault("object_store_memory", float(capacity))
+    # GC arenas leaked by SIGKILLed sessions (name = antray_<headpid>_<ts>;
+    # if that pid is gone, nothing can still map the store legitimately)
+    try:
+        for f in os.listdir("/dev/shm"):
+            if not f.startswith("antray_"):
+                continue
+            try:
+                pid = int(f.split("_")[1])
+                os.kill(pid, 0)
+            except (ValueError, IndexError):
+                continue
+            except ProcessLookupError:
+                try:
+                    os.unlink(os.path.join("/dev/shm", f))
+                except OSError:
+                    pass
+            except PermissionError:
+                pass  # pid alive under another uid: leave it
+    except OSError:
+        pass
     store_path = os.path.join("/dev/shm", f"antray_{os.getpid()}_{int(time.time())}")
 
     raylet = Raylet(args.host, (args.host, gcs_port), resources, store_path, capacity, session_dir)
