@@ -190,6 +190,13 @@ class ActorHandleState:
         # must not reuse a pre-restart call's deterministic task id, or
         # ray.get would return the old call's cached result
         self.task_counter = 0
+        # all handles dropped while calls were still outstanding: defer the
+        # out-of-scope teardown until the last reply lands (reference
+        # semantics — __ray_terminate__ runs AFTER submitted tasks)
+        self.kill_on_drain = False
+        # oids of live hip_ipc objects pinned in this actor's HBM: teardown
+        # also waits for these (the tensors die with the actor process)
+        self.gpu_pinned_oids: set = set()
 
 
 class CoreWorker:
@@ -223,6 +230,13 @@ class CoreWorker:
         self._actor_retry_payloads: Dict[bytes, bytes] = {}  # task -> actor
         self._killed_workers: Dict[bytes, str] = {}  # worker_id -> kill reason
         self._object_locations: Dict[bytes, Tuple[str, int]] = {}
+        # hipIpc results: oid -> producer actor_id whose GPUObjectStore pins
+        # the device tensors; the actor outlives the owner's refs to them
+        self._gpu_object_holders: Dict[bytes, bytes] = {}
+        # refs serialized into a submitted task's args are borrowed until
+        # that task completes: task_id -> [oid], oid -> borrow count
+        self._task_contained: Dict[bytes, list] = {}
+        self._borrow_counts: Dict[bytes, int] = {}
         self._actors: Dict[bytes, ActorHandleState] = {}
         self._actor_results: Dict[bytes, dict] = {}
         # lease state per scheduling key
@@ -1037,6 +1051,7 @@ class CoreWorker:
                 self.store.memory.mark_pending(r.binary())
         sobj = serialization.serialize((args, kwargs))
         self._register_escapes(sobj)
+        self._begin_task_borrows(task_id, sobj)
         if fn_id not in self._pushed_fns:
             import cloudpickle
 
@@ -1415,11 +1430,13 @@ class CoreWorker:
 
     def _fail_task(self, payload, exc: BaseException):
         task_id = payload["task_id"]
+        self._end_task_borrows(task_id)
         if payload.get("type") == "actor_task":
             st = self._actors.get(payload.get("actor_id"))
             if st is not None:
                 with st.lock:
                     st.outstanding = max(0, st.outstanding - 1)
+                self._maybe_kill_on_drain(st)
         st = self._streams.get(task_id)
         if st is not None:
             with st.cv:
@@ -1431,6 +1448,7 @@ class CoreWorker:
 
     def _handle_task_reply(self, payload, reply):
         if payload.get("streaming"):
+            self._end_task_borrows(payload["task_id"])
             st = self._streams.get(payload["task_id"])
             if st is not None:
                 with st.cv:
@@ -1453,19 +1471,32 @@ class CoreWorker:
                     st.cv.notify_all()
             return
         if reply.get("status") == "ok":
+            gpu_actor = (payload.get("actor_id")
+                         if payload.get("tensor_transport") == "hip_ipc"
+                         else None)
             for r in reply.get("results", []):
                 oid = r["oid"]
                 if r.get("holder"):
                     self._object_locations[oid] = tuple(r["holder"])
+                if gpu_actor is not None:
+                    # device tensors live pinned in the producer actor: keep
+                    # it alive until our refs (and borrows) to them drain
+                    self._gpu_object_holders[oid] = gpu_actor
+                    ast = self._actors.get(gpu_actor)
+                    if ast is not None:
+                        with ast.lock:
+                            ast.gpu_pinned_oids.add(oid)
                 if r.get("inline") is not None:
                     self.store.memory.put(oid, _RawResult(r["inline"], r.get("meta", b"py")))
                 else:
                     self.store.memory.put(oid, IN_PLASMA)
+            self._end_task_borrows(payload["task_id"])
         else:
             data = reply.get("error_payload")
             meta = reply.get("error_meta", serialization.META_ERROR)
             if self._should_retry_app_error(payload, data, meta):
                 return
+            self._end_task_borrows(payload["task_id"])
             for i in range(payload.get("n_returns", 1)):
                 oid = ObjectID.for_return(TaskID(payload["task_id"]), i).binary()
                 if data is not None:
@@ -1572,6 +1603,7 @@ class CoreWorker:
                 self.store.memory.mark_pending(r.binary())
         sobj = serialization.serialize((args, kwargs))
         self._register_escapes(sobj)
+        self._begin_task_borrows(task_id, sobj)
         if not streaming:
             for r in refs:
                 self._task_of_oid[r.binary()] = ("actor", actor_id, task_id)
@@ -1669,6 +1701,7 @@ class CoreWorker:
             with st.lock:
                 st.inflight.pop(payload["seq"], None)
                 st.outstanding = max(0, st.outstanding - 1)
+            self._maybe_kill_on_drain(st)
         except Exception:
             # connection to actor lost: wait for GCS verdict (restart/dead)
             with st.lock:
@@ -1732,6 +1765,7 @@ class CoreWorker:
         st = self._get_actor_state(actor_id)
         with st.lock:
             st.handle_count += 1
+            st.kill_on_drain = False
 
     def actor_handle_removed(self, actor_id: bytes):
         st = self._actors.get(actor_id)
@@ -1742,11 +1776,84 @@ class CoreWorker:
             should_kill = (
                 st.handle_count <= 0 and st.is_owner and not st.detached
             )
-        if should_kill and self.connected and self.io is not None:
+            if should_kill and st.outstanding > 0:
+                # graceful: submitted calls run to completion first (the
+                # reference queues __ray_terminate__ behind them); the last
+                # reply/fail triggers _maybe_kill_on_drain
+                st.kill_on_drain = True
+                return
+        if should_kill:
+            self._send_actor_out_of_scope(actor_id)
+
+    def _send_actor_out_of_scope(self, actor_id: bytes):
+        if self.connected and self.io is not None:
             try:
                 self.io.submit(
                     self.gcs.call("actor_out_of_scope", {"actor_id": actor_id})
                 )
+            except Exception:
+                pass
+
+    def _maybe_kill_on_drain(self, st: ActorHandleState):
+        with st.lock:
+            fire = (st.kill_on_drain and st.outstanding <= 0
+                    and st.handle_count <= 0 and not st.gpu_pinned_oids)
+            if fire:
+                st.kill_on_drain = False
+        if fire:
+            self._send_actor_out_of_scope(st.actor_id)
+
+    def _begin_task_borrows(self, task_id: bytes, sobj):
+        """Record refs shipped in a task's args as borrowed until it replies
+        (keeps hip_ipc producer actors alive while a downstream task reads
+        their pinned tensors)."""
+        if not sobj.contained_refs:
+            return
+        with self._lock:
+            oids = [r.binary() for r in sobj.contained_refs]
+            self._task_contained[task_id] = oids
+            for oid in oids:
+                self._borrow_counts[oid] = self._borrow_counts.get(oid, 0) + 1
+
+    def _end_task_borrows(self, task_id: bytes):
+        with self._lock:
+            oids = self._task_contained.pop(task_id, None)
+            if not oids:
+                return
+            done = []
+            for oid in oids:
+                n = self._borrow_counts.get(oid, 1) - 1
+                if n <= 0:
+                    self._borrow_counts.pop(oid, None)
+                    done.append(oid)
+                else:
+                    self._borrow_counts[oid] = n
+        for oid in done:
+            if oid in self._gpu_object_holders and \
+                    self._local_refs.get(oid, 0) <= 0:
+                self._release_gpu_pin(oid)
+
+    def _release_gpu_pin(self, oid: bytes):
+        """Owner dropped its last ref to a hip_ipc object: unpin it from the
+        producer actor's lifetime bookkeeping (and free the HBM there)."""
+        actor_id = self._gpu_object_holders.pop(oid, None)
+        if actor_id is None:
+            return
+        st = self._actors.get(actor_id)
+        if st is not None:
+            with st.lock:
+                st.gpu_pinned_oids.discard(oid)
+            self._maybe_kill_on_drain(st)
+        holder = self._object_locations.get(oid)
+        if holder is not None and self.io is not None and self.connected:
+            async def _free():
+                try:
+                    conn = await self._get_worker_conn_async_cached(tuple(holder))
+                    await conn.call("free_objects", {"oids": [oid]}, timeout=10)
+                except Exception:
+                    pass
+            try:
+                self.io.submit(_free())
             except Exception:
                 pass
 
@@ -1779,6 +1886,13 @@ class CoreWorker:
                     return
                 del self._local_refs[oid]
                 info = self._owned.pop(oid, None)
+                gpu_held = (oid in self._gpu_object_holders
+                            and self._borrow_counts.get(oid, 0) <= 0)
+            if gpu_held:
+                # last local ref to a hip_ipc object with no in-flight
+                # borrows: unpin it from the producer actor (HBM freed there)
+                self._release_gpu_pin(oid)
+            with self._lock:
                 if info is None or info.get("escaped"):
                     return
             # owned, unreferenced, never escaped -> free storage

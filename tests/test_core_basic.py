@@ -374,3 +374,55 @@ def test_actor_task_ids_unique_across_callers(ray_start_regular):
     sa, sb = ray.get([ra, rb], timeout=120)
     assert sa == 200_000 * 1.0, sa
     assert sb == 200_000 * 2.0, sb
+
+
+def test_out_of_scope_actor_drains_submitted_tasks(ray_start_regular):
+    """Dropping the last handle to an actor with submitted tasks must NOT
+    kill it mid-call: the out-of-scope teardown waits for every submitted
+    task to complete (reference queues __ray_terminate__ behind them).
+    Regression for the r01 hipIpc mixed-payload failure — the same
+    chained-temporaries pattern, CPU-only."""
+    ray = ray_start_regular
+
+    @ray.remote
+    class P:
+        def make(self):
+            time.sleep(0.2)  # widen the window the old code lost the race in
+            return {"w": [1.0] * 256, "meta": "hello", "n": 5}
+
+    @ray.remote
+    class C:
+        def read(self, d):
+            time.sleep(0.2)
+            return (sum(d["w"]), d["meta"], d["n"])
+
+    # both handles are temporaries: refcount drops to zero right after
+    # submission, long before the tasks run
+    out = ray.get(C.remote().read.remote(P.remote().make.remote()), timeout=60)
+    assert out == (256.0, "hello", 5)
+
+
+def test_out_of_scope_actor_eventually_dies(ray_start_regular):
+    """After the drained teardown, the actor must actually terminate (no
+    leak): its state reaches DEAD within a bounded wait."""
+    ray = ray_start_regular
+    from ant_ray_amd.util import state as ray_state
+
+    @ray.remote
+    class Once:
+        def ping(self):
+            return "pong"
+
+    ref = Once.remote().ping.remote()  # handle is a temporary
+    assert ray.get(ref, timeout=60) == "pong"
+    deadline = time.time() + 30
+    dead = False
+    while time.time() < deadline:
+        actors = ray_state.list_actors()
+        states = {a.get("actor_id"): a.get("state") for a in actors}
+        alive = [s for s in states.values() if s not in ("DEAD",)]
+        if not alive:
+            dead = True
+            break
+        time.sleep(0.3)
+    assert dead, f"out-of-scope actor leaked: {states}"
