@@ -153,6 +153,7 @@ class GcsServer:
         # air-gapped image has no Redis, so the store client is a file)
         self.persist_path: Optional[str] = None
         self._persist_task: Optional[asyncio.Task] = None
+        self._persist_dirty = False
 
     # ------------------------------------------------------------------ serve
     async def start(self, host="127.0.0.1", port=0, persist_path=None):
@@ -201,9 +202,16 @@ class GcsServer:
 
     # ------------------------------------------------------- fault tolerance
     def _persist_soon(self):
-        """Debounced snapshot: coalesce bursts of mutations into one write."""
-        if not self.persist_path or (
-                self._persist_task and not self._persist_task.done()):
+        """Debounced snapshot: coalesce bursts of mutations into one write.
+
+        A mutation landing while a persist is already in flight sets the
+        dirty flag; _persist_later loops until it drains, so every
+        acknowledged mutation reaches disk even if it arrived during the
+        pack/write phase of the previous snapshot."""
+        if not self.persist_path:
+            return
+        self._persist_dirty = True
+        if self._persist_task and not self._persist_task.done():
             return
         try:
             loop = asyncio.get_running_loop()
@@ -213,19 +221,27 @@ class GcsServer:
 
     async def _persist_later(self):
         await asyncio.sleep(0.1)
-        try:
-            # build the snapshot ON the loop (cheap dict-of-refs walk, no
-            # concurrent-mutation hazard), then pack+write OFF the loop:
-            # with thousands of actors the bytes are tens of MB and would
-            # stall heartbeats
-            snap = self._snapshot()
-            await asyncio.get_running_loop().run_in_executor(
-                None, self._write_snapshot, snap)
-        except Exception:
-            logger.exception("GCS table persist failed")
+        while getattr(self, "_persist_dirty", False):
+            self._persist_dirty = False
+            try:
+                # pack the snapshot ON the loop (msgpack of plain dicts —
+                # no concurrent-mutation hazard, nested mutables included),
+                # then only the file write goes OFF the loop so tens-of-MB
+                # disk IO can't stall heartbeats
+                import msgpack
+
+                packed = msgpack.packb(self._snapshot(), use_bin_type=True)
+                await asyncio.get_running_loop().run_in_executor(
+                    None, self._write_snapshot_bytes, packed)
+            except Exception:
+                logger.exception("GCS table persist failed")
 
     def _persist_now(self):
-        self._write_snapshot(self._snapshot())
+        import msgpack
+
+        self._persist_dirty = False
+        self._write_snapshot_bytes(
+            msgpack.packb(self._snapshot(), use_bin_type=True))
 
     def _snapshot(self):
         snap = {
@@ -260,12 +276,10 @@ class GcsServer:
         }
         return snap
 
-    def _write_snapshot(self, snap):
-        import msgpack
-
+    def _write_snapshot_bytes(self, packed: bytes):
         tmp = self.persist_path + ".tmp"
         with open(tmp, "wb") as f:
-            f.write(msgpack.packb(snap, use_bin_type=True))
+            f.write(packed)
         os.replace(tmp, self.persist_path)
 
     def _restore_tables(self):
@@ -595,10 +609,13 @@ class GcsServer:
         if seq is not None:
             # last-writer-wins by CLIENT sequence: concurrent handler
             # tasks (or chaos-delayed ones) must not let a stale write
-            # overwrite a newer value (metrics publishes rely on this)
+            # overwrite a newer value (metrics publishes rely on this).
+            # Scoped per client (seq_id): each publisher's counter starts
+            # at 0, so comparing across clients would silently reject a
+            # restarted worker's writes until it caught up
             if not hasattr(self, "_kv_seq"):
                 self._kv_seq = {}
-            k = (p.get("ns", ""), key)
+            k = (p.get("seq_id", b""), p.get("ns", ""), key)
             if seq < self._kv_seq.get(k, -1):
                 return {"added": False, "stale": True}
             self._kv_seq[k] = seq
@@ -614,6 +631,13 @@ class GcsServer:
         ns = self.kv.get(p.get("ns", ""), {})
         existed = ns.pop(p["key"], None) is not None
         if existed:
+            if hasattr(self, "_kv_seq"):
+                # drop seq state for the deleted key (any client): a fresh
+                # publisher of a re-created key starts clean
+                nsk = p.get("ns", "")
+                for k in [k for k in self._kv_seq
+                          if k[1] == nsk and k[2] == p["key"]]:
+                    del self._kv_seq[k]
             self._persist_soon()
         return {"deleted": existed}
 

@@ -44,6 +44,7 @@ class TaskExecutor:
         self._serial_q: "queue.Queue" = queue.Queue()
         self._cancelled: set = set()        # task_ids cancelled pre-start
         self._running_serial: bytes = None  # task running on the serial thread
+        self._serial_cancel_lock = threading.Lock()
         self._serial_thread = threading.Thread(target=self._serial_loop, daemon=True)
         self._serial_thread.start()
         self._pool: Optional[concurrent.futures.ThreadPoolExecutor] = None
@@ -120,14 +121,24 @@ class TaskExecutor:
 
     def _serial_loop(self):
         while True:
-            payload, done = self._serial_q.get()
-            if payload is None:
-                return
-            self._running_serial = payload.get("task_id")
             try:
-                self._run_and_reply(payload, done)
-            finally:
-                self._running_serial = None
+                payload, done = self._serial_q.get()
+                if payload is None:
+                    return
+                with self._serial_cancel_lock:
+                    self._running_serial = payload.get("task_id")
+                try:
+                    self._run_and_reply(payload, done)
+                finally:
+                    with self._serial_cancel_lock:
+                        self._running_serial = None
+            except KeyboardInterrupt:
+                # a soft-cancel async-exc landed AFTER its target task
+                # finished (between cancel_task's _running_serial check and
+                # delivery at the next bytecode boundary): the interrupt is
+                # stale — swallow it and keep serving, never let it kill
+                # the serial thread (callers would hang forever)
+                continue
 
     def cancel_task(self, task_id: bytes, force: bool):
         """Cancel a queued or running task in this worker (parity:
@@ -135,7 +146,11 @@ class TaskExecutor:
         running on the serial thread -> KeyboardInterrupt raised into it
         via PyThreadState_SetAsyncExc (the reference sends SIGINT to the
         worker's main thread); force -> the whole worker exits (reference
-        force kill), the owner maps the dead push to TaskCancelledError."""
+        force kill), the owner maps the dead push to TaskCancelledError.
+        Tasks running on a concurrency-group/thread-pool thread are NOT
+        interrupted (no per-thread soft cancel, matching the reference's
+        main-thread-only SIGINT); they are dropped if still queued and
+        force-kill works regardless."""
         self._cancelled.add(task_id)
         t = getattr(self, "_async_tasks", {}).get(task_id)
         if t is not None and not force:
@@ -150,13 +165,18 @@ class TaskExecutor:
 
             _t.Timer(0.05, lambda: os._exit(1)).start()  # let the RPC reply
             return {"cancelled": "force-killing worker"}
-        if self._running_serial == task_id:
-            import ctypes
+        with self._serial_cancel_lock:
+            # the lock pins _running_serial across the check+inject so the
+            # interrupt can only be pended while the task is still the
+            # current one (the residual stale-delivery window is handled by
+            # _serial_loop's KeyboardInterrupt guard)
+            if self._running_serial == task_id:
+                import ctypes
 
-            ctypes.pythonapi.PyThreadState_SetAsyncExc(
-                ctypes.c_long(self._serial_thread.ident),
-                ctypes.py_object(KeyboardInterrupt))
-            return {"cancelled": "interrupted running task"}
+                ctypes.pythonapi.PyThreadState_SetAsyncExc(
+                    ctypes.c_long(self._serial_thread.ident),
+                    ctypes.py_object(KeyboardInterrupt))
+                return {"cancelled": "interrupted running task"}
         return {"cancelled": "queued"}
 
     def request_exit(self):

@@ -1096,15 +1096,17 @@ class CoreWorker:
             while len(self._retry_exceptions) > 20000:
                 self._retry_exceptions.pop(next(iter(self._retry_exceptions)))
         if not streaming:
+            # cost is charged PER REF and refunded per pop (every pop path
+            # goes through _lineage_pop), so the 256MB bound stays accurate
+            # for multi-return tasks and free-on-zero eviction alike
             cost = len(payload.get("args") or b"") + 512
             for r in refs:
+                if r.binary() not in self._lineage:
+                    self._lineage_bytes += cost
                 self._lineage[r.binary()] = (key, payload, resources, opts)
-            self._lineage_bytes += cost
             while ((self._lineage_bytes > 256 * 1024 * 1024
                     or len(self._lineage) > 50_000) and self._lineage):
-                old_oid = next(iter(self._lineage))
-                _, old_p, _, _ = self._lineage.pop(old_oid)
-                self._lineage_bytes -= len(old_p.get("args") or b"") + 512
+                self._lineage_pop(next(iter(self._lineage)))
         # locality hint: an arg ref held by a worker on ANOTHER node pulls
         # the lease request toward that node (reference locality-aware
         # LeasePolicy); first remote-held ref wins
@@ -1420,8 +1422,11 @@ class CoreWorker:
         logger.warning("task %s raised a retryable exception; retrying "
                        "(%d retries left)", task_id.hex()[:8],
                        lpayload["max_retries"])
+        rcost = len(lpayload.get("args") or b"") + 512
         for i in range(lpayload.get("n_returns", 1)):
             roid = ObjectID.for_return(TaskID(task_id), i).binary()
+            if roid not in self._lineage:
+                self._lineage_bytes += rcost
             self._lineage[roid] = (key, lpayload, res, opts)
             self.store.memory.delete(roid)
             self.store.memory.mark_pending(roid)
@@ -1803,6 +1808,17 @@ class CoreWorker:
         if fire:
             self._send_actor_out_of_scope(st.actor_id)
 
+    def _lineage_pop(self, oid: bytes):
+        """Remove a lineage record, refunding its byte charge (one charge
+        per oid entry — ADVICE r01: multi-return tasks used to drive the
+        counter negative and free-on-zero never refunded)."""
+        rec = self._lineage.pop(oid, None)
+        if rec is not None:
+            self._lineage_bytes -= len(rec[1].get("args") or b"") + 512
+            if self._lineage_bytes < 0:
+                self._lineage_bytes = 0
+        return rec
+
     def _begin_task_borrows(self, task_id: bytes, sobj):
         """Record refs shipped in a task's args as borrowed until it replies
         (keeps hip_ipc producer actors alive while a downstream task reads
@@ -1910,7 +1926,7 @@ class CoreWorker:
             self.store.free([oid])
             # bookkeeping tied to this oid goes with it
             self._object_locations.pop(oid, None)
-            self._lineage.pop(oid, None)
+            self._lineage_pop(oid)
             self._task_of_oid.pop(oid, None)
         except Exception:
             pass

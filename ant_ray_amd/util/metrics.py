@@ -33,8 +33,11 @@ def _publish(name: str, kind: str, value, tags: Dict[str, str],
             "ns": "metrics", "key": key,
             "value": json.dumps(payload).encode(), "overwrite": True,
             # ordered last-writer-wins: fire-and-forget publishes may be
-            # handled out of order server-side
+            # handled out of order server-side. The seq is scoped to THIS
+            # client (seq_id): another worker publishing the same metric
+            # starts its own sequence and is never judged against ours
             "seq": _pub_seq,
+            "seq_id": cw.worker_id,
         }))
     except Exception:
         pass
