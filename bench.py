@@ -1,14 +1,25 @@
-"""Flagship benchmark: tokens/sec, Llama-3-8B DDP training (BASELINE.json).
+"""Flagship benchmark: tokens/sec, Llama-3-8B Ray Train DDP (BASELINE.json).
 
-Single process per GPU. The driver launches N>1 as:
+Default path (--via-ray, on): the measured number is produced by an actual
+Ray Train run — ray.init + TorchTrainer + placement group pinning one
+worker actor per GPU — the same stack a user of the framework runs
+(parity: reference python/ray/train/v2/api/data_parallel_trainer.py:155).
+The training engine inside each worker is FlatDDP (bucketed RCCL
+all-reduce overlapped with backward) + fused flat AdamW on the CDNA4 HIP
+kernels. `--bare` keeps the round-1 no-runtime path for A/B overhead
+checks (tools/bench_parity.py measures the delta).
+
+Launch contract (driver): N=1 is `python bench.py --gpus 1 ...`; N>1 is
   python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
-      --master-addr 127.0.0.1 --master-port P bench.py --gpus N --steps K --warmup W
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N ...
+Under torchrun, only RANK 0 proceeds in via-ray mode (it drives the Ray
+cluster that spawns one worker actor per GPU); other ranks exit 0
+immediately without touching the GPU. In --bare mode every rank is a DDP
+rank as in round 1.
 
-Each rank: Llama-3-8B (random init, bf16) + FlatDDP (bucketed RCCL all-reduce
-overlapped with backward) + fused flat AdamW — the same training engine Ray
-Train's TorchTrainer workers use (ant_ray_amd/train). Synthetic token data.
-
-Rank 0 prints ONE JSON line with the whole-job aggregate tokens/sec.
+Timing: W untimed warmup steps, then exactly K steps bracketed by
+barrier+synchronize on both sides, elapsed MAX-reduced over ranks, ONE
+JSON line from rank 0 with the whole-job aggregate tokens/s.
 """
 from __future__ import annotations
 
@@ -16,8 +27,6 @@ import argparse
 import json
 import os
 import time
-
-import torch
 
 
 def _setup_tunableop(local_rank: int):
@@ -53,7 +62,7 @@ def _setup_tunableop(local_rank: int):
             pass
 
 
-def main():
+def _parse_args(argv=None):
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=8)
@@ -63,32 +72,29 @@ def main():
     ap.add_argument("--seq", type=int, default=4096)
     ap.add_argument("--bucket-mb", type=int, default=64)
     ap.add_argument("--lr", type=float, default=3e-4)
-    args = ap.parse_args()
+    ap.add_argument("--bare", action="store_true",
+                    help="skip the Ray runtime: raw torchrun DDP ranks")
+    return ap.parse_args(argv)
 
-    rank = int(os.environ.get("RANK", 0))
-    local_rank = int(os.environ.get("LOCAL_RANK", 0))
-    world = int(os.environ.get("WORLD_SIZE", 1))
-    if world > 1:
-        import torch.distributed as dist
 
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group(backend="nccl", rank=rank, world_size=world)
-    else:
-        torch.cuda.set_device(local_rank)
-
-    device = f"cuda:{local_rank}"
-    _setup_tunableop(local_rank)
+def _resolve_batch(args) -> int:
     if args.model.startswith("gpt2"):
         args.seq = min(args.seq, 1024)
-        batch = args.batch or 16
-    else:
-        # batch 6 x seq 4096 ~ 250 GB of 288 GB HBM (measured): biggest safe
-        # per-GPU batch with headroom for RCCL workspaces at 8 GPUs
-        batch = args.batch or 6
+        return args.batch or 16
+    # batch 6 x seq 4096 ~ 250 GB of 288 GB HBM (measured): biggest safe
+    # per-GPU batch with headroom for RCCL workspaces at 8 GPUs
+    return args.batch or 6
+
+
+def _train_core(args, device, rank: int, world: int):
+    """The measured loop, identical for both paths. Requires the process
+    group to already exist when world > 1. Returns (elapsed_s, loss)."""
+    import torch
 
     from ant_ray_amd.models import build_model
     from ant_ray_amd.parallel import FlatAdamW, FlatDDP, FlatParamManager
 
+    batch = _resolve_batch(args)
     torch.manual_seed(1234 + rank)
     t_build = time.time()
     model = build_model(args.model, device=device, seq_len=args.seq)
@@ -104,13 +110,9 @@ def main():
             flush=True,
         )
 
-    def make_batch():
+    def step():
         tokens = torch.randint(0, vocab, (batch, args.seq), device=device)
         targets = torch.randint(0, vocab, (batch, args.seq), device=device)
-        return tokens, targets
-
-    def step():
-        tokens, targets = make_batch()
         loss = ddp(tokens, targets)
         loss.backward()
         ddp.finish_grad_sync()
@@ -118,12 +120,11 @@ def main():
         opt.zero_grad()
         return loss
 
-    # warmup
+    import torch.distributed as dist
+
     for _ in range(args.warmup):
         loss = step()
     if world > 1:
-        import torch.distributed as dist
-
         dist.barrier()
     torch.cuda.synchronize()
     t0 = time.time()
@@ -131,48 +132,126 @@ def main():
         loss = step()
     torch.cuda.synchronize()
     if world > 1:
-        import torch.distributed as dist
-
         dist.barrier()
     elapsed = time.time() - t0
     if world > 1:
-        import torch.distributed as dist
-
         t = torch.tensor([elapsed], device=device)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
+    return elapsed, float(loss.item())
 
+
+def _emit(args, world: int, elapsed: float, final_loss: float):
+    batch = _resolve_batch(args)
     tokens_total = batch * args.seq * args.steps * world
-    value = tokens_total / elapsed
+    out = {
+        "metric": "tokens/sec Llama-3-8B Ray Train DDP"
+        if not args.model.startswith("gpt2")
+        else "tokens/sec GPT-2-small DDP",
+        "value": round(tokens_total / elapsed, 1),
+        "unit": "tokens/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed / args.steps * 1000, 2),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16",
+        "data": "synthetic",
+        "config": {
+            "model": args.model,
+            "global_batch": batch * world,
+            "seq_len": args.seq,
+            "parallelism": f"dp{world}",
+            "via": "bare-torchrun" if args.bare else "ray-train",
+            "final_loss": round(final_loss, 4),
+        },
+    }
+    print(json.dumps(out), flush=True)
+
+
+def _main_bare(args):
+    import torch
+
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    torch.cuda.set_device(local_rank)
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.init_process_group(backend="nccl", rank=rank, world_size=world)
+    device = f"cuda:{local_rank}"
+    _setup_tunableop(local_rank)
+    elapsed, final_loss = _train_core(args, device, rank, world)
     if rank == 0:
-        out = {
-            "metric": "tokens/sec Llama-3-8B Ray Train DDP"
-            if not args.model.startswith("gpt2")
-            else "tokens/sec GPT-2-small DDP",
-            "value": round(value, 1),
-            "unit": "tokens/s",
-            "n_gpus": world,
-            "steps": args.steps,
-            "warmup": args.warmup,
-            "ms_per_step": round(elapsed / args.steps * 1000, 2),
-            "higher_is_better": True,
-            "scaling": "weak",
-            "vs_baseline": None,
-            "dtype": "bf16",
-            "data": "synthetic",
-            "config": {
-                "model": args.model,
-                "global_batch": batch * world,
-                "seq_len": args.seq,
-                "parallelism": f"dp{world}",
-                "final_loss": round(float(loss.item()), 4),
-            },
-        }
-        print(json.dumps(out), flush=True)
+        _emit(args, world, elapsed, final_loss)
     if world > 1:
         import torch.distributed as dist
 
         dist.destroy_process_group()
+
+
+def _train_fn(config):
+    """Runs inside each Ray Train worker actor (1 GPU each)."""
+    import torch
+
+    from ant_ray_amd import train
+
+    args = argparse.Namespace(**config)
+    ctx = train.get_context()
+    rank = ctx.get_world_rank()
+    world = ctx.get_world_size()
+    device = train.torch.get_device()
+    torch.cuda.set_device(device)
+    _setup_tunableop(ctx.get_local_rank())
+    elapsed, final_loss = _train_core(args, str(device), rank, world)
+    train.report({"elapsed": elapsed, "final_loss": final_loss,
+                  "world": world})
+
+
+def _main_via_ray(args):
+    # Under torchrun only rank 0 drives the Ray cluster; sibling ranks
+    # exit clean without touching the GPU (the worker ACTORS are the
+    # ranks). torchrun waits for all workers, so the early exits are fine.
+    if int(os.environ.get("RANK", 0)) != 0:
+        return
+    # the worker group does its own rendezvous on a free port; drop the
+    # torchrun-provided one so nothing collides with it
+    for k in ("MASTER_ADDR", "MASTER_PORT", "RANK", "LOCAL_RANK",
+              "WORLD_SIZE", "LOCAL_WORLD_SIZE", "GROUP_RANK", "NODE_RANK"):
+        os.environ.pop(k, None)
+
+    import ant_ray_amd as ray
+    from ant_ray_amd.train import RunConfig, ScalingConfig
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    n = args.gpus
+    ray.init(num_cpus=max(8, 2 * n), num_gpus=n)
+    try:
+        trainer = TorchTrainer(
+            _train_fn,
+            train_loop_config=dict(vars(args)),
+            scaling_config=ScalingConfig(num_workers=n, use_gpu=True),
+            run_config=RunConfig(name="bench", storage_path="/tmp/antray_bench"),
+        )
+        result = trainer.fit()
+        if result.error is not None:
+            raise result.error
+        m = result.metrics
+        _emit(args, int(m["world"]), float(m["elapsed"]),
+              float(m["final_loss"]))
+    finally:
+        ray.shutdown()
+
+
+def main():
+    args = _parse_args()
+    if args.bare:
+        _main_bare(args)
+    else:
+        _main_via_ray(args)
 
 
 if __name__ == "__main__":
