@@ -37,6 +37,9 @@ void launch_attn_bwd_reduce_kv(const void*, const void*, void*, void*, int,
                                int, int, int, void*);
 void launch_bf16_to_f32(const void*, float*, int64_t, hipStream_t);
 void launch_f32_to_bf16(const float*, void*, int64_t, hipStream_t);
+void launch_attn_decode(const void*, const void*, const void*, void*, float*,
+                        const int*, int, int, int, int, int, long, long, long,
+                        float, void*);
 }
 
 namespace {
@@ -278,6 +281,49 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> attn_bwd(
   return {dq, dk, dv};
 }
 
+// Flash-decode: one new token per sequence over a bf16 KV cache.
+// q: [B, Hq, D] contiguous; k/v: [B, Hk, Tmax, D] (strided along b/h ok,
+// rows contiguous); T = valid cache length (uniform), or per-sequence
+// lengths in `lens` (int32 [B] on device; then T = max). Returns o
+// [B, Hq, D] bf16.
+torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                          int64_t T, double scale,
+                          c10::optional<torch::Tensor> lens) {
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16, "q bf16 gpu");
+  TORCH_CHECK(q.dim() == 3 && q.is_contiguous(), "q must be [B,Hq,D] contig");
+  TORCH_CHECK(k.dim() == 4 && v.dim() == 4, "k/v must be [B,Hk,T,D]");
+  int B = q.size(0), Hq = q.size(1), D = q.size(2);
+  int Hk = k.size(1);
+  TORCH_CHECK(D == 128, "attn_decode: D must be 128");
+  TORCH_CHECK(Hq % Hk == 0 && Hq / Hk <= 8, "attn_decode: Hq/Hk must be <=8");
+  TORCH_CHECK(k.stride(3) == 1 && v.stride(3) == 1, "rows must be contiguous");
+  TORCH_CHECK(k.strides() == v.strides(), "k/v must share layout");
+  TORCH_CHECK(T >= 1 && T <= k.size(2), "bad cache length");
+  const int* lens_ptr = nullptr;
+  if (lens.has_value()) {
+    TORCH_CHECK(lens->is_cuda() && lens->scalar_type() == torch::kInt32 &&
+                lens->numel() == B, "lens must be int32 [B] on device");
+    lens_ptr = lens->data_ptr<int>();
+  }
+  // chunk count: fill the chip (>=512 workgroups) without splitting below
+  // 256 keys per chunk
+  int C = (int)std::min<long>(std::max<long>(1, 512 / std::max(1, B * Hk)),
+                              std::max<long>(1, ((long)T + 255) / 256));
+  auto o = torch::empty({B, Hq, D}, q.options());
+  torch::Tensor part;
+  float* part_ptr = nullptr;
+  if (C > 1) {
+    part = torch::empty({(long)B * Hq * C * (D + 2)},
+                        q.options().dtype(torch::kFloat32));
+    part_ptr = part.data_ptr<float>();
+  }
+  launch_attn_decode(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                     part_ptr, lens_ptr, B, (int)T, Hq, Hk, C, k.stride(0),
+                     k.stride(1), k.stride(2), (float)scale,
+                     (void*)cur_stream());
+  return o;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused (add+)rmsnorm forward",
         py::arg("x"), py::arg("w"), py::arg("residual") = py::none(),
@@ -297,6 +343,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd, py::arg("dout"), py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("o"), py::arg("lse"), py::arg("scale"),
         py::arg("causal") = true);
+  m.def("attn_decode", &attn_decode, py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("T"), py::arg("scale"), py::arg("lens") = py::none());
   m.def("bf16_scale_", &bf16_scale_);
   m.def("bf16_to_f32", &bf16_to_f32);
   m.def("f32_to_bf16_", &f32_to_bf16_);

@@ -52,6 +52,26 @@ class LlamaConfig:
                    intermediate=512, vocab=1024, max_seq=max_seq)
 
 
+class KVCache:
+    """Per-model KV cache for generation: [n_layers, B, Hk, max_seq, D]
+    bf16, preallocated. `pos` = number of cached positions (uniform across
+    the batch; ragged serve batches pass per-sequence lens to the decode
+    kernel instead)."""
+
+    def __init__(self, cfg: LlamaConfig, batch: int, max_seq: int, device):
+        shape = (cfg.n_layers, batch, cfg.n_kv_heads, max_seq, cfg.head_dim)
+        self.k = torch.empty(shape, device=device, dtype=torch.bfloat16)
+        self.v = torch.empty(shape, device=device, dtype=torch.bfloat16)
+        self.max_seq = max_seq
+        self.pos = 0
+
+    def layer(self, i: int):
+        return self.k[i], self.v[i]
+
+    def nbytes(self) -> int:
+        return self.k.numel() * 2 * 2
+
+
 class LlamaAttention(nn.Module):
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
@@ -60,18 +80,29 @@ class LlamaAttention(nn.Module):
         self.wqkv = nn.Linear(cfg.hidden, (Hq + 2 * Hk) * D, bias=False)
         self.wo = nn.Linear(Hq * D, cfg.hidden, bias=False)
 
-    def forward(self, y, cos, sin):
+    def forward(self, y, cos, sin, cache=None, layer_idx=0, pos=0):
         cfg = self.cfg
         Hq, Hk, D = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim
         B, S, _ = y.shape
         qkv = self.wqkv(y)
-        qkv = ops.rope_qkv(qkv, cos, sin, Hq, Hk, D)
+        qkv = ops.rope_qkv(qkv, cos[pos : pos + S], sin[pos : pos + S],
+                           Hq, Hk, D)
         q = qkv[..., : Hq * D].view(B, S, Hq, D).transpose(1, 2)
         k = qkv[..., Hq * D : (Hq + Hk) * D].view(B, S, Hk, D).transpose(1, 2)
         v = qkv[..., (Hq + Hk) * D :].view(B, S, Hk, D).transpose(1, 2)
+        if cache is not None:
+            ck, cv = cache.layer(layer_idx)
+            ck[:, :, pos : pos + S] = k
+            cv[:, :, pos : pos + S] = v
+            if S == 1:
+                # single-token decode over the cache (flash-decode kernel)
+                o = ops.attention_decode(q.reshape(B, Hq, D), ck, cv,
+                                         seq_len=pos + 1)
+                return self.wo(o.view(B, 1, Hq * D))
+            assert pos == 0, "chunked prefill not supported"
         if qkv.is_cuda and D == 128 and os.environ.get("ANTRAY_FLASH", "0") == "1":
             # hand-written CDNA4 flash attention (fwd: attention.hip at
-            # AOTriton parity; bwd: attention_bwd.hip at 0.94x) — opt-in
+            # AOTriton parity; bwd: attention_bwd.hip — fused dKV) — opt-in
             # until the bwd beats AOTriton; numerics covered by
             # tests/test_attention_gpu.py either way
             o = ops.attention(q, k, v, causal=True)
@@ -101,14 +132,14 @@ class LlamaBlock(nn.Module):
         self.attn = LlamaAttention(cfg)
         self.mlp = LlamaMLP(cfg)
 
-    def forward(self, h, res, cos, sin):
+    def forward(self, h, res, cos, sin, cache=None, layer_idx=0, pos=0):
         # h = contribution from the previous sublayer; res = residual stream
         if res is None:
             res = h
             y = ops.rmsnorm(h, self.attn_norm, self.cfg.eps)
         else:
             y, res = ops.fused_add_rmsnorm(h, res, self.attn_norm, self.cfg.eps)
-        a = self.attn(y, cos, sin)
+        a = self.attn(y, cos, sin, cache=cache, layer_idx=layer_idx, pos=pos)
         y2, res = ops.fused_add_rmsnorm(a, res, self.mlp_norm, self.cfg.eps)
         return self.mlp(y2), res
 
@@ -147,21 +178,59 @@ class LlamaForCausalLM(nn.Module):
     def head_weight(self):
         return self.embed.weight if self.lm_head is None else self.lm_head.weight
 
-    def forward(self, tokens, targets=None):
+    def forward(self, tokens, targets=None, cache=None, pos=0):
         B, S = tokens.shape
         h = self.embed(tokens)
         res = None
         cos, sin = self.rope_cos, self.rope_sin
-        for blk in self.blocks:
-            h, res = blk(h, res, cos, sin)
+        for i, blk in enumerate(self.blocks):
+            h, res = blk(h, res, cos, sin, cache=cache, layer_idx=i, pos=pos)
         if res is None:
             y = ops.rmsnorm(h, self.final_norm, self.cfg.eps)
         else:
             y, _ = ops.fused_add_rmsnorm(h, res, self.final_norm, self.cfg.eps)
+        if cache is not None and targets is None:
+            # generation: logits only for the LAST position (saves the
+            # full-sequence vocab projection on prefill)
+            y_last = y[:, -1, :]
+            return y_last @ self.head_weight.t()
         flat = y.reshape(B * S, self.cfg.hidden)
         if targets is not None:
             return ops.linear_cross_entropy(flat, self.head_weight, targets.reshape(-1))
         return (flat @ self.head_weight.t()).view(B, S, -1)
+
+    @torch.no_grad()
+    def generate(self, tokens, max_new_tokens: int, cache: "KVCache" = None,
+                 temperature: float = 0.0):
+        """Greedy (temperature=0) or sampled generation with the KV cache +
+        flash-decode kernel. tokens: [B, S_prompt] int64. Returns
+        [B, S_prompt + max_new_tokens].
+
+        Role parity: the reference serves generation through vLLM
+        (reference python/ray/llm/_internal/serve/engines/vllm/
+        vllm_engine.py:1); this is the in-tree MI355X-native decode loop
+        used by the framework's own Serve LLM path."""
+        B, S = tokens.shape
+        dev = tokens.device
+        if cache is None:
+            cache = KVCache(self.cfg, B, min(self.cfg.max_seq,
+                                             S + max_new_tokens), dev)
+        logits = self.forward(tokens, cache=cache, pos=0)  # prefill
+        cache.pos = S
+        out = [tokens]
+        cur = None
+        for _ in range(max_new_tokens):
+            if temperature > 0:
+                probs = torch.softmax(logits.float() / temperature, dim=-1)
+                cur = torch.multinomial(probs, 1)
+            else:
+                cur = logits.argmax(dim=-1, keepdim=True)
+            out.append(cur)
+            if cache.pos >= cache.max_seq:
+                break
+            logits = self.forward(cur, cache=cache, pos=cache.pos)
+            cache.pos += 1
+        return torch.cat(out, dim=1)
 
     def num_params(self):
         return sum(p.numel() for p in self.parameters())

@@ -13,6 +13,7 @@ from ant_ray_amd.ops import reference
 from ant_ray_amd.ops.functional import (  # noqa: F401
     adamw_step,
     attention,
+    attention_decode,
     fused_add_rmsnorm,
     rmsnorm,
     rope_qkv,

@@ -192,6 +192,23 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return o
 
 
+def attention_decode(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                     seq_len: int, scale=None, lens=None) -> torch.Tensor:
+    """Flash-decode: one new token per sequence over a bf16 KV cache
+    (csrc/kernels/attention_decode.hip — flash-decode two-level combine,
+    GQA heads share one K/V read). q: [B,Hq,D]; k/v: [B,Hk,Tmax,D] with
+    seq_len valid rows (or per-sequence int32 `lens` on device). Inference
+    only (no autograd). CPU fallback: exact fp32 reference."""
+    if scale is None:
+        scale = q.shape[-1] ** -0.5
+    if _use_hip(q):
+        return _hip().attn_decode(q.contiguous(), k, v, int(seq_len),
+                                  float(scale), lens)
+    kc = k[:, :, :seq_len]
+    vc = v[:, :, :seq_len]
+    return ref.attention_decode_ref(q, kc, vc, lens=lens, scale=float(scale))
+
+
 # ------------------------------------------------------------------ adamw
 
 

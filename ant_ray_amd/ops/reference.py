@@ -75,6 +75,25 @@ def adamw_step(p32, p_bf16, g, m, v, lr, b1, b2, eps, wd, step, grad_scale):
     p_bf16.copy_(p32.to(torch.bfloat16))
 
 
+def attention_decode_ref(q, k, v, lens=None, scale=None):
+    """fp32 reference decode: q [B,Hq,D] (one new token), k/v [B,Hk,T,D]
+    caches; lens optional per-sequence valid lengths."""
+    B, Hq, D = q.shape
+    Hk, T = k.shape[1], k.shape[2]
+    if scale is None:
+        scale = D ** -0.5
+    rep = Hq // Hk
+    kf = k.float().repeat_interleave(rep, dim=1)   # [B,Hq,T,D]
+    vf = v.float().repeat_interleave(rep, dim=1)
+    s = torch.einsum("bhd,bhtd->bht", q.float(), kf) * scale
+    if lens is not None:
+        pos = torch.arange(T, device=q.device).view(1, 1, T)
+        mask = pos >= lens.view(B, 1, 1)
+        s = s.masked_fill(mask, float("-inf"))
+    p = torch.softmax(s, dim=-1)
+    return torch.einsum("bht,bhtd->bhd", p, vf).to(q.dtype)
+
+
 def attention_ref(q, k, v, causal=True, scale=None):
     """fp32 reference attention with GQA (q:[B,Hq,S,D], k/v:[B,Hk,S,D])."""
     import torch

@@ -181,3 +181,71 @@ class TestModelGPU:
         loss = m(tok, tok)
         loss.backward()
         assert torch.isfinite(loss)
+
+
+class TestAttentionDecode:
+    """Flash-decode kernel (attention_decode.hip) vs fp32 reference."""
+
+    def test_uniform_len(self):
+        B, Hq, Hk, T, D = 4, 32, 8, 777, 128
+        q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
+        k = torch.randn(B, Hk, 1024, D, dtype=torch.bfloat16, device=DEV)
+        v = torch.randn(B, Hk, 1024, D, dtype=torch.bfloat16, device=DEV)
+        o = ops.attention_decode(q, k, v, seq_len=T)
+        o_ref = ref.attention_decode_ref(q, k[:, :, :T], v[:, :, :T])
+        _assert_close(o, o_ref, what="attn_decode uniform")
+
+    def test_long_cache_multichunk(self):
+        # T large enough that the two-level (chunked) combine path runs
+        B, Hq, Hk, T, D = 1, 32, 8, 4096, 128
+        q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
+        k = torch.randn(B, Hk, T, D, dtype=torch.bfloat16, device=DEV)
+        v = torch.randn(B, Hk, T, D, dtype=torch.bfloat16, device=DEV)
+        o = ops.attention_decode(q, k, v, seq_len=T)
+        o_ref = ref.attention_decode_ref(q, k, v)
+        _assert_close(o, o_ref, what="attn_decode multichunk")
+
+    def test_ragged_lens(self):
+        B, Hq, Hk, T, D = 5, 8, 4, 512, 128
+        q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
+        k = torch.randn(B, Hk, T, D, dtype=torch.bfloat16, device=DEV)
+        v = torch.randn(B, Hk, T, D, dtype=torch.bfloat16, device=DEV)
+        lens = torch.tensor([3, 100, 512, 77, 256], dtype=torch.int32,
+                            device=DEV)
+        o = ops.attention_decode(q, k, v, seq_len=T, lens=lens)
+        o_ref = ref.attention_decode_ref(q, k, v, lens=lens)
+        _assert_close(o, o_ref, what="attn_decode ragged")
+
+    def test_spiked_key(self):
+        # forces the defer-max rescale branch (rule 26): one huge score
+        B, Hq, Hk, T, D = 2, 4, 2, 300, 128
+        q = torch.randn(B, Hq, D, dtype=torch.bfloat16, device=DEV)
+        k = torch.randn(B, Hk, T, D, dtype=torch.bfloat16, device=DEV) * 0.1
+        v = torch.randn(B, Hk, T, D, dtype=torch.bfloat16, device=DEV)
+        k[:, :, 237] = q[:, ::2, :] * 3.0  # aligned spike late in the walk
+        o = ops.attention_decode(q, k, v, seq_len=T)
+        o_ref = ref.attention_decode_ref(q, k, v)
+        _assert_close(o, o_ref, what="attn_decode spiked")
+
+
+class TestGenerate:
+    def test_kv_cache_generate_matches_full_forward(self):
+        """Greedy generation with the KV cache + decode kernel must match
+        argmax stepping with full no-cache forwards."""
+        from ant_ray_amd.models import build_model
+
+        torch.manual_seed(7)
+        m = build_model("llama-tiny-d128", device=DEV, seq_len=256)
+        m.eval()
+        tokens = torch.randint(0, 1024, (2, 17), device=DEV)
+        out = m.generate(tokens, max_new_tokens=8)
+        # reference: recompute the full sequence each step (no cache)
+        cur = tokens.clone()
+        with torch.no_grad():
+            for _ in range(8):
+                logits = m(cur)[:, -1, :]
+                cur = torch.cat([cur, logits.argmax(-1, keepdim=True)], 1)
+        # bf16 decode vs recompute can differ after an early divergence;
+        # require the first few tokens to match exactly
+        assert torch.equal(out[:, :17 + 4], cur[:, :17 + 4]), (
+            out.tolist(), cur.tolist())
