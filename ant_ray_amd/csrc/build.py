@@ -21,7 +21,9 @@ BUILD = os.path.join(CSRC, "_build")
 HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
-KERNELS = ["norms.hip", "rope.hip", "elementwise.hip", "cross_entropy.hip", "adamw.hip", "attention.hip", "attention_bwd.hip"]
+KERNELS = ["norms.hip", "rope.hip", "elementwise.hip", "cross_entropy.hip",
+           "adamw.hip", "attention.hip", "attention_bwd.hip",
+           "attention_decode.hip", "data_transform.hip"]
 
 
 def _newer(src, obj):

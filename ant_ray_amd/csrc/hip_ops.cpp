@@ -40,6 +40,12 @@ void launch_f32_to_bf16(const float*, void*, int64_t, hipStream_t);
 void launch_attn_decode(const void*, const void*, const void*, void*, float*,
                         const int*, int, int, int, int, int, long, long, long,
                         float, void*);
+void launch_cast_affine_u8(const void*, void*, const float*, const float*,
+                           long, int, int, int, void*);
+void launch_cast_affine_f32(const void*, void*, const float*, const float*,
+                            long, int, int, int, void*);
+void launch_nhwc_to_nchw_u8(const void*, void*, const float*, const float*,
+                            long, long, int, int, void*);
 }
 
 namespace {
@@ -324,6 +330,57 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   return o;
 }
 
+// Fused cast + per-channel affine: y = (x - shift) * scale. x u8 or f32;
+// out bf16 or f32. scale/shift: f32 tensors of numel 1 (scalar) or C
+// (= x's innermost dim). Data-plane collate hot path.
+torch::Tensor cast_affine(torch::Tensor x, torch::Tensor scale,
+                          torch::Tensor shift, torch::ScalarType out_dtype) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be contiguous GPU");
+  TORCH_CHECK(scale.is_cuda() && scale.scalar_type() == torch::kFloat32 &&
+              shift.is_cuda() && shift.scalar_type() == torch::kFloat32,
+              "scale/shift must be f32 on GPU");
+  TORCH_CHECK(out_dtype == torch::kBFloat16 || out_dtype == torch::kFloat32,
+              "out dtype must be bf16 or f32");
+  long n = x.numel();
+  int C = (int)x.size(-1);
+  int per_channel = scale.numel() > 1;
+  if (per_channel) {
+    TORCH_CHECK(scale.numel() == C && shift.numel() == C,
+                "per-channel scale/shift must match innermost dim");
+  }
+  auto y = torch::empty(x.sizes(), x.options().dtype(out_dtype));
+  int out_f32 = out_dtype == torch::kFloat32;
+  if (x.scalar_type() == torch::kUInt8) {
+    launch_cast_affine_u8(x.data_ptr(), y.data_ptr(),
+                          scale.data_ptr<float>(), shift.data_ptr<float>(), n,
+                          C, per_channel, out_f32, (void*)cur_stream());
+  } else if (x.scalar_type() == torch::kFloat32) {
+    launch_cast_affine_f32(x.data_ptr(), y.data_ptr(),
+                           scale.data_ptr<float>(), shift.data_ptr<float>(),
+                           n, C, per_channel, out_f32, (void*)cur_stream());
+  } else {
+    TORCH_CHECK(false, "cast_affine supports u8 or f32 input");
+  }
+  return y;
+}
+
+// Fused NHWC u8 -> NCHW normalize: x [N,H,W,C] u8 -> [N,C,H,W] bf16/f32.
+torch::Tensor nhwc_to_nchw(torch::Tensor x, torch::Tensor scale,
+                           torch::Tensor shift, torch::ScalarType out_dtype) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 4 &&
+              x.scalar_type() == torch::kUInt8, "x must be [N,H,W,C] u8 GPU");
+  long N = x.size(0), H = x.size(1), W = x.size(2);
+  int C = (int)x.size(3);
+  TORCH_CHECK(scale.numel() == C && shift.numel() == C &&
+              scale.is_cuda() && shift.is_cuda(), "scale/shift must be [C]");
+  auto y = torch::empty({N, C, H, W}, x.options().dtype(out_dtype));
+  launch_nhwc_to_nchw_u8(x.data_ptr(), y.data_ptr(), scale.data_ptr<float>(),
+                         shift.data_ptr<float>(), N * H * W, H * W, C,
+                         out_dtype == torch::kFloat32,
+                         (void*)cur_stream());
+  return y;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "fused (add+)rmsnorm forward",
         py::arg("x"), py::arg("w"), py::arg("residual") = py::none(),
@@ -345,6 +402,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("causal") = true);
   m.def("attn_decode", &attn_decode, py::arg("q"), py::arg("k"), py::arg("v"),
         py::arg("T"), py::arg("scale"), py::arg("lens") = py::none());
+  m.def("cast_affine", &cast_affine, py::arg("x"), py::arg("scale"),
+        py::arg("shift"), py::arg("out_dtype"));
+  m.def("nhwc_to_nchw", &nhwc_to_nchw, py::arg("x"), py::arg("scale"),
+        py::arg("shift"), py::arg("out_dtype"));
   m.def("bf16_scale_", &bf16_scale_);
   m.def("bf16_to_f32", &bf16_to_f32);
   m.def("f32_to_bf16_", &f32_to_bf16_);

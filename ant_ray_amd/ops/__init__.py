@@ -14,7 +14,9 @@ from ant_ray_amd.ops.functional import (  # noqa: F401
     adamw_step,
     attention,
     attention_decode,
+    cast_affine,
     fused_add_rmsnorm,
+    nhwc_to_nchw,
     rmsnorm,
     rope_qkv,
     swiglu,

@@ -209,6 +209,48 @@ def attention_decode(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return ref.attention_decode_ref(q, kc, vc, lens=lens, scale=float(scale))
 
 
+# ------------------------------------------------------- data transforms
+
+
+def cast_affine(x: torch.Tensor, scale=1.0, shift=0.0,
+                out_dtype=torch.bfloat16) -> torch.Tensor:
+    """Fused y = (cast(x) - shift) * scale for the Data collate path
+    (csrc/kernels/data_transform.hip). x: u8 or f32; scale/shift scalars
+    or per-channel (innermost dim) sequences. Memory-bound one-pass."""
+    def as_t(v):
+        if isinstance(v, torch.Tensor):
+            return v.to(device=x.device, dtype=torch.float32)
+        if isinstance(v, (list, tuple)):
+            return torch.tensor(v, device=x.device, dtype=torch.float32)
+        return torch.tensor([float(v)], device=x.device, dtype=torch.float32)
+
+    s, h = as_t(scale), as_t(shift)
+    if _use_hip(x):
+        return _hip().cast_affine(x.contiguous(), s, h, out_dtype)
+    v = x.float()
+    if s.numel() > 1:
+        v = (v - h) * s
+    else:
+        v = (v - h.item()) * s.item()
+    return v.to(out_dtype)
+
+
+def nhwc_to_nchw(x: torch.Tensor, mean, std,
+                 out_dtype=torch.bfloat16) -> torch.Tensor:
+    """Fused image collate: [N,H,W,C] u8 -> [N,C,H,W] normalized
+    ((x - mean) / std), one HBM pass. mean/std per-channel sequences."""
+    def as_t(v):
+        if isinstance(v, torch.Tensor):
+            return v.to(device=x.device, dtype=torch.float32)
+        return torch.tensor(v, device=x.device, dtype=torch.float32)
+
+    m, sd = as_t(mean), as_t(std)
+    if _use_hip(x):
+        return _hip().nhwc_to_nchw(x.contiguous(), 1.0 / sd, m, out_dtype)
+    v = (x.float() - m.view(1, 1, 1, -1)) / sd.view(1, 1, 1, -1)
+    return v.permute(0, 3, 1, 2).contiguous().to(out_dtype)
+
+
 # ------------------------------------------------------------------ adamw
 
 

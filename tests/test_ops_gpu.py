@@ -249,3 +249,41 @@ class TestGenerate:
         # require the first few tokens to match exactly
         assert torch.equal(out[:, :17 + 4], cur[:, :17 + 4]), (
             out.tolist(), cur.tolist())
+
+
+class TestDataTransforms:
+    """Fused Data-plane kernels (data_transform.hip) vs torch reference."""
+
+    def test_cast_affine_u8_scalar(self):
+        x = torch.randint(0, 256, (4096, 128), dtype=torch.uint8, device=DEV)
+        y = ops.cast_affine(x, scale=1 / 255.0, shift=127.5,
+                            out_dtype=torch.bfloat16)
+        y_ref = ((x.float() - 127.5) / 255.0).to(torch.bfloat16)
+        _assert_close(y, y_ref, atol=1e-3, rtol=1e-2, what="cast_affine u8")
+
+    def test_cast_affine_u8_per_channel_f32_out(self):
+        C = 16
+        x = torch.randint(0, 256, (1000, C), dtype=torch.uint8, device=DEV)
+        scale = torch.rand(C, device=DEV) + 0.5
+        shift = torch.rand(C, device=DEV) * 100
+        y = ops.cast_affine(x, scale, shift, out_dtype=torch.float32)
+        y_ref = (x.float() - shift) * scale
+        _assert_close(y, y_ref, atol=1e-4, rtol=1e-4, what="cast_affine chan")
+
+    def test_cast_affine_ragged_tail(self):
+        x = torch.randint(0, 256, (3, 7, 13), dtype=torch.uint8, device=DEV)
+        y = ops.cast_affine(x, 2.0, 1.0, out_dtype=torch.float32)
+        _assert_close(y, (x.float() - 1.0) * 2.0, atol=1e-5, rtol=1e-5,
+                      what="cast_affine tail")
+
+    def test_nhwc_to_nchw(self):
+        x = torch.randint(0, 256, (8, 32, 32, 3), dtype=torch.uint8,
+                          device=DEV)
+        mean = [123.7, 116.3, 103.5]
+        std = [58.4, 57.1, 57.4]
+        y = ops.nhwc_to_nchw(x, mean, std, out_dtype=torch.float32)
+        m = torch.tensor(mean, device=DEV).view(1, 1, 1, 3)
+        s = torch.tensor(std, device=DEV).view(1, 1, 1, 3)
+        y_ref = ((x.float() - m) / s).permute(0, 3, 1, 2).contiguous()
+        assert y.shape == (8, 3, 32, 32)
+        _assert_close(y, y_ref, atol=1e-3, rtol=1e-3, what="nhwc_to_nchw")
