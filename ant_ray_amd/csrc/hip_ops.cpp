@@ -115,14 +115,26 @@ void rope_(torch::Tensor q, torch::Tensor k, torch::Tensor cos_t,
   TORCH_CHECK(q.dim() == 4 && k.dim() == 4, "q/k must be [B,S,H,D]");
   TORCH_CHECK(q.stride(3) == 1 && q.stride(2) == q.size(3), "q [H,D] must be contiguous");
   TORCH_CHECK(k.stride(3) == 1 && k.stride(2) == k.size(3), "k [H,D] must be contiguous");
-  TORCH_CHECK(q.stride(0) == q.size(1) * q.stride(1), "q batch stride must fold");
-  TORCH_CHECK(k.stride(0) == k.size(1) * k.stride(1), "k batch stride must fold");
   TORCH_CHECK(cos_t.scalar_type() == torch::kFloat32 && cos_t.is_contiguous());
   int64_t B = q.size(0), S = q.size(1);
   int Hq = (int)q.size(2), Hk = (int)k.size(2), D = (int)q.size(3);
+  // the kernel walks a FLAT token index (B*S tokens at one uniform
+  // stride), so dim0 must fold into dim1. A size-1 S dim (single-token
+  // decode step) reports a synthetic stride — use the batch stride as the
+  // token stride there (tok == b).
+  int64_t q_tok = q.stride(1), k_tok = k.stride(1);
+  if (S == 1) {
+    q_tok = B > 1 ? q.stride(0) : q_tok;
+    k_tok = B > 1 ? k.stride(0) : k_tok;
+  } else {
+    TORCH_CHECK(B == 1 || q.stride(0) == S * q.stride(1),
+                "q batch stride must fold");
+    TORCH_CHECK(B == 1 || k.stride(0) == S * k.stride(1),
+                "k batch stride must fold");
+  }
   TORCH_CHECK(cos_t.size(0) >= S && cos_t.size(1) == D / 2, "cos table too small");
   launch_rope(q.data_ptr(), k.data_ptr(), cos_t.data_ptr<float>(),
-              sin_t.data_ptr<float>(), B, S, Hq, Hk, D, q.stride(1), k.stride(1),
+              sin_t.data_ptr<float>(), B, S, Hq, Hk, D, q_tok, k_tok,
               backward ? 1 : 0, cur_stream());
 }
 

@@ -135,15 +135,21 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   const int st_row = threadIdx.x >> 3;
   const int st_col = (threadIdx.x & 7) * 16;
 
+  // T14 double-buffered prefetch (same pattern as the dv/dk kernels): tile
+  // t+1's global loads are issued right after tile t is staged, so the
+  // inter-barrier critical path never waits on HBM latency.
+  ushortx8 k0, k1, v0, v1;
+  {
+    const int krow_g = min(st_row, S - 1);
+    k0 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
+    k1 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col + 8);
+    v0 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col);
+    v1 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col + 8);
+  }
   for (int t = 0; t < n_tiles; ++t) {
     const int n0 = t * BLOCK;
     __syncthreads();
     {
-      const int krow_g = min(n0 + st_row, S - 1);
-      const ushortx8 k0 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
-      const ushortx8 k1 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col + 8);
-      const ushortx8 v0 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col);
-      const ushortx8 v1 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col + 8);
       *(ushortx8*)(&k_rm[rm_off(st_row, st_col / 8)]) = k0;
       *(ushortx8*)(&k_rm[rm_off(st_row, st_col / 8 + 1)]) = k1;
       *(ushortx8*)(&v_rm[rm_off(st_row, st_col / 8)]) = v0;
@@ -155,6 +161,13 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       }
     }
     __syncthreads();
+    if (t + 1 < n_tiles) {
+      const int krow_g = min((t + 1) * BLOCK + st_row, S - 1);
+      k0 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
+      k1 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col + 8);
+      v0 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col);
+      v1 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col + 8);
+    }
     if (t > my_last_tile || q0 >= S) continue;
 
     // S^T and dP^T (C: n = q lane-local, m = key)
@@ -795,11 +808,14 @@ extern "C" void launch_attn_bwd_dkv(const void* q, const void* k,
                                     long oh, long os, float scale,
                                     int causal, void* stream) {
   dim3 grid((S + 127) / 128, Hk, B);
-  // fused dK+dV (one S^T/dP^T per tile, 7 matmul units total) vs the split
-  // pair (8 units, higher occupancy). ANTRAY_BWD_SPLIT=1 forces the split.
+  // fused dK+dV does one S^T/dP^T per tile (7 matmul units total) but its
+  // 374-reg footprint caps it at 1 wave/SIMD: measured 11.9 ms vs the
+  // split pair's 10.06 ms on the llama3-8b shape (r2c A/B) — the split
+  // pair (8 units, 2 waves/SIMD) stays the default; ANTRAY_BWD_FUSED=1
+  // keeps the fused kernel selectable for future register work.
   static const bool use_split = [] {
-    const char* e = getenv("ANTRAY_BWD_SPLIT");
-    return e && e[0] == '1';
+    const char* e = getenv("ANTRAY_BWD_FUSED");
+    return !(e && e[0] == '1');
   }();
   if (!use_split) {
     hipLaunchKernelGGL(attn_bwd_dkv_kernel, grid, dim3(256), 0,

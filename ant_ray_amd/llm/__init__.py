@@ -55,9 +55,26 @@ class LLMConfig:
         return self.tensor_parallel_size() * self.pipeline_parallel_size()
 
 
+_NATIVE_MODELS = ("llama3-8b", "llama-3-8b", "llama3_8b", "llama-tiny",
+                  "tiny", "llama-tiny-d128", "tiny-d128")
+
+
 def build_llm_deployment(config: LLMConfig):
-    """Serve deployment wrapping a vLLM AsyncLLMEngine (parity:
-    serve/llm build_llm_deployment)."""
+    """Serve deployment for an LLM. In-tree model ids (llama3-8b family)
+    are served by the NATIVE MI355X engine (ant_ray_amd.llm.native_engine:
+    hand-written prefill/decode kernels, KV cache, dynamic batching);
+    other model ids delegate to vLLM when importable (parity:
+    serve/llm build_llm_deployment -> vllm_engine.py)."""
+    if config.model_id.lower() in _NATIVE_MODELS:
+        from ant_ray_amd.llm.native_engine import build_native_llm_deployment
+
+        return build_native_llm_deployment(
+            model_name=config.model_id.lower(),
+            num_replicas=config.deployment_config.get("num_replicas", 1),
+            num_gpus=config.num_gpus_per_replica(),
+            max_seq=int(config.engine_kwargs.get("max_model_len", 4096)),
+            max_batch_size=int(config.engine_kwargs.get("max_num_seqs", 16)),
+        )
     vllm = _require_vllm()
     from ant_ray_amd import serve
 

@@ -98,3 +98,25 @@ def test_serve_gpu_deployment(ray_mod):
     h = serve.run(GpuNorm.bind(), name="gpunorm", route_prefix="/gpunorm")
     assert h.remote(8).result(timeout_s=120) == [8, 256]
     serve.shutdown()
+
+
+def test_serve_native_llm_gpu(ray_mod):
+    """Native LLM engine behind Serve on the GPU: prefill (flash fwd path
+    uses SDPA default) + flash-decode kernel + dynamic batching."""
+    import random
+
+    from ant_ray_amd import serve
+    from ant_ray_amd.llm import LLMConfig, build_llm_deployment
+
+    app = build_llm_deployment(LLMConfig(
+        model_loading_config={"model_id": "llama-tiny-d128"},
+        engine_kwargs={"max_model_len": 128, "max_num_seqs": 4},
+        deployment_config={"num_replicas": 1},
+    ))
+    h = serve.run(app, name="llm-g", route_prefix="/llm-g")
+    rng = random.Random(0)
+    reqs = [h.remote({"prompt_ids": [rng.randrange(1024) for _ in range(16)],
+                      "max_new_tokens": 6}) for _ in range(5)]
+    outs = [r.result(timeout_s=300) for r in reqs]
+    assert all(len(o["token_ids"]) == 6 for o in outs), outs
+    serve.shutdown()

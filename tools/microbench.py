@@ -64,19 +64,22 @@ def main():
            lambda: ray.get([do_put_small.remote() for _ in range(10)]),
            1000, dur)
 
-    def put_gb():
-        for _ in range(8):
-            ray.put(arr_1mb)
+    # single client put gigabytes: ONE 800 MB array per put, exactly the
+    # reference's shape (ray_perf.py:120 arr = np.zeros(100Mi, int64);
+    # put_large puts it whole — sustained large-copy bandwidth, not
+    # per-put overhead amortization)
+    arr_800mb = np.zeros(100 * 1024 * 1024, dtype=np.int64)
 
     n = [0]
     start = time.time()
-    put_gb()
-    while time.time() - start < dur:
-        put_gb()
+    ray.put(arr_800mb)
+    while time.time() - start < max(dur, 3.0):
+        ray.put(arr_800mb)
         n[0] += 1
-    gbps = (n[0] + 1) * 8 / 1024 / (time.time() - start)
+    gbps = (n[0] + 1) * 0.8 / (time.time() - start)
     print(json.dumps({"name": "single client put gigabytes",
                       "value": round(gbps, 2), "unit": "GB/s"}), flush=True)
+    del arr_800mb
 
     # multi client put gigabytes: 10 workers x 10 puts of 80 MB
     # (parity ray_perf.py:140 do_put / put_multi)
