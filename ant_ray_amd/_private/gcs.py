@@ -960,7 +960,12 @@ class GcsServer:
         if aid is None:
             return None
         info = self.actors.get(aid)
-        return info.view() if info else None
+        if info is None or info.state == DEAD:
+            # a dead actor's name is free (reference GCS removes the name
+            # on death); callers re-creating under the same name must not
+            # receive a handle to the corpse
+            return None
+        return info.view()
 
     async def rpc_wait_actor_ready(self, conn, p):
         info = self.actors.get(p["actor_id"])

@@ -133,6 +133,11 @@ class Raylet:
         asyncio.get_running_loop().create_task(self._heartbeat_loop())
         asyncio.get_running_loop().create_task(self._reap_loop())
         asyncio.get_running_loop().create_task(self._memory_monitor_loop())
+        # per-node physical/GPU stats reporter (parity: reference
+        # dashboard/modules/reporter/reporter_agent.py per-node agent)
+        from ant_ray_amd.dashboard.reporter import reporter_loop
+
+        asyncio.get_running_loop().create_task(reporter_loop(self))
         logger.info(
             "raylet %s on %s:%s (%s)", self.node_id.hex()[:8], self.node_ip, self.port,
             {k: v for k, v in self.resources_total.items()},

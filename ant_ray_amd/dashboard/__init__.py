@@ -80,6 +80,17 @@ class DashboardHead:
 
         return await asyncio.wrap_future(fut)
 
+    async def _node_stats(self):
+        """Latest per-node reporter samples from the GCS KV."""
+        keys = (await self._gcs("kv_keys", {"ns": "node_stats",
+                                            "prefix": b""}))["keys"]
+        out = {}
+        for k in keys:
+            r = await self._gcs("kv_get", {"ns": "node_stats", "key": k})
+            if r.get("value"):
+                out[k.decode()] = json.loads(r["value"].decode())
+        return out
+
     async def _route(self, scope, receive):
         path = scope["path"].rstrip("/")
         method = scope["method"]
@@ -99,7 +110,21 @@ class DashboardHead:
             total = await self._gcs("cluster_resources")
             return js({"nodes": nodes, "resources": total})
         if path == "/api/nodes":
-            return js(await self._gcs("node_table"))
+            nodes = await self._gcs("node_table")
+            stats = await self._node_stats()
+            for n in nodes if isinstance(nodes, list) else []:
+                nid = n.get("node_id")
+                nid = nid.hex() if isinstance(nid, bytes) else str(nid)
+                if nid in stats:
+                    n["physical_stats"] = stats[nid]
+            return js(nodes)
+        if path == "/api/node_stats" or path == "/api/gpus":
+            # per-node reporter samples (reference reporter_agent shape)
+            stats = await self._node_stats()
+            if path == "/api/gpus":
+                return js({nid: s.get("gpus", [])
+                           for nid, s in stats.items()})
+            return js(stats)
         if path == "/api/actors":
             return js(await self._gcs("list_actors"))
         if path == "/api/tasks":

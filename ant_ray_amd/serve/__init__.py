@@ -360,6 +360,19 @@ def shutdown():
             ray.kill(ray.get_actor(actor_name))
         except Exception:
             pass
+    # wait for the names to actually free: a serve.start right after
+    # shutdown must not resolve a dying controller (the kill is processed
+    # asynchronously by the GCS)
+    import time as _time
+
+    deadline = _time.time() + 15
+    for actor_name in (PROXY_NAME, CONTROLLER_NAME):
+        while _time.time() < deadline:
+            try:
+                ray.get_actor(actor_name)
+            except Exception:
+                break
+            _time.sleep(0.1)
 
 
 def get_app_handle(name: str = DEFAULT_APP_NAME) -> DeploymentHandle:

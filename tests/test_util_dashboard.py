@@ -104,6 +104,25 @@ def test_metrics_and_dashboard(ray_mod):
     with urllib.request.urlopen(f"{base}/api/logs", timeout=10) as r:
         logs = json.loads(r.read())
     assert isinstance(logs, list)
+    # per-node reporter (dashboard/reporter.py in each raylet, 5s period):
+    # wait for the first sample and check the psutil-backed fields
+    deadline = time.time() + 30
+    stats = {}
+    while time.time() < deadline and not stats:
+        with urllib.request.urlopen(f"{base}/api/node_stats",
+                                    timeout=10) as r:
+            stats = json.loads(r.read())
+        if not stats:
+            time.sleep(1)
+    assert stats, "no reporter samples arrived"
+    rec = next(iter(stats.values()))
+    assert rec["cpus"] >= 1 and 0 <= rec["cpu_percent"] <= 100 * rec["cpus"]
+    assert rec["mem"]["total"] > 0 and "gpus" in rec
+    with urllib.request.urlopen(f"{base}/api/gpus", timeout=10) as r:
+        assert isinstance(json.loads(r.read()), dict)
+    with urllib.request.urlopen(f"{base}/api/nodes", timeout=10) as r:
+        nodes = json.loads(r.read())
+    assert any("physical_stats" in n for n in nodes)
     if logs:
         with urllib.request.urlopen(f"{base}/api/logs/{logs[0]}",
                                     timeout=10) as r:
