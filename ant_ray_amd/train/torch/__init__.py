@@ -67,13 +67,15 @@ def prepare_model(
     if move_to_device:
         model = model.to(device)
     world = dist.get_world_size() if dist.is_initialized() else 1
-    if world <= 1 or parallel_strategy is None:
-        return model
     kwargs = dict(parallel_strategy_kwargs or {})
-    if parallel_strategy == "fsdp":
+    if parallel_strategy == "fsdp" and dist.is_initialized():
+        # torch-ROCm FSDP passthrough; wraps at any world size (world==1
+        # shards trivially — useful for single-GPU validation of the path)
         from torch.distributed.fsdp import FullyShardedDataParallel
 
         return FullyShardedDataParallel(model, **kwargs)
+    if world <= 1 or parallel_strategy is None:
+        return model
     dtype = next(model.parameters()).dtype
     mgr = FlatParamManager(model, device=device, dtype=dtype)
     bucket_mb = kwargs.pop("bucket_mb", 64)

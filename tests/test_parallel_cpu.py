@@ -35,11 +35,14 @@ def _worker(rank, world, port, q):
         dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("world", [2])
+@pytest.mark.parametrize("world", [2, 4, 8])
 def test_flatddp_grad_sync(world):
+    """2/4/8-rank gloo: the rank counts the driver's round-end multi-GPU
+    scaling bench uses (8-GPU readiness pre-staged on CPU — VERDICT r01
+    item 5). bucket_mb=1 gives several buckets incl. a partial tail."""
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29511
+    port = 29511 + world
     procs = [ctx.Process(target=_worker, args=(r, world, port, q)) for r in range(world)]
     for p in procs:
         p.start()
@@ -49,9 +52,10 @@ def test_flatddp_grad_sync(world):
         results[rank] = (gradsum, psum, loss)
     for p in procs:
         p.join(timeout=60)
-    # after all-reduce both ranks see identical summed grads and params
-    assert results[0][0] == pytest.approx(results[1][0], rel=1e-5)
-    assert results[0][1] == pytest.approx(results[1][1], rel=1e-6)
+    # after all-reduce every rank sees identical summed grads and params
+    for r in range(1, world):
+        assert results[0][0] == pytest.approx(results[r][0], rel=1e-5)
+        assert results[0][1] == pytest.approx(results[r][1], rel=1e-6)
     # losses differ (different shards) -> the all-reduce really combined them
     assert results[0][2] != results[1][2]
 

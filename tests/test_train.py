@@ -471,3 +471,4 @@ def test_accelerate_inside_torchtrainer(ray_mod, tmp_path_factory):
     ).fit()
     assert res.error is None, res.error
     assert res.metrics["procs"] == 2  # accelerate picked up the group
+

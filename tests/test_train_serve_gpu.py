@@ -120,3 +120,53 @@ def test_serve_native_llm_gpu(ray_mod):
     outs = [r.result(timeout_s=300) for r in reqs]
     assert all(len(o["token_ids"]) == 6 for o in outs), outs
     serve.shutdown()
+
+
+def test_fsdp_passthrough_gpu(ray_mod, tmp_path_factory):
+    """prepare_model(parallel_strategy="fsdp") wraps torch-ROCm FSDP and
+    trains on the MI355X (the passthrough claim was untested).
+    world=1 + gloo pg: FSDP requires an accelerator device, so the CPU CI
+    cannot run this — sharded multi-rank behavior is exercised by the
+    driver's round-end multi-GPU tier."""
+    from ant_ray_amd.train import RunConfig, ScalingConfig
+    from ant_ray_amd.train.config import TorchConfig
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    storage = str(tmp_path_factory.mktemp("fsdp"))
+
+    def train_fn(config):
+        import torch
+        import torch.distributed as dist
+
+        from ant_ray_amd import train
+
+        assert dist.is_initialized()
+        torch.manual_seed(0)
+        model = torch.nn.Sequential(
+            torch.nn.Linear(16, 64), torch.nn.ReLU(),
+            torch.nn.Linear(64, 1)).to("cuda")
+        model = train.torch.prepare_model(model, parallel_strategy="fsdp")
+        from torch.distributed.fsdp import FullyShardedDataParallel
+
+        assert isinstance(model, FullyShardedDataParallel)
+        opt = torch.optim.SGD(model.parameters(), lr=0.05)
+        g = torch.Generator().manual_seed(1)
+        losses = []
+        for _ in range(20):
+            x = torch.randn(32, 16, generator=g).to("cuda")
+            y = (x.sum(-1, keepdim=True) > 0).float()
+            loss = torch.nn.functional.mse_loss(model(x), y)
+            opt.zero_grad()
+            loss.backward()
+            opt.step()
+            losses.append(float(loss))
+        train.report({"first": losses[0], "last": losses[-1]})
+
+    res = TorchTrainer(
+        train_fn,
+        scaling_config=ScalingConfig(num_workers=1, use_gpu=True),
+        torch_config=TorchConfig(backend="gloo"),
+        run_config=RunConfig(name="fsdp", storage_path=storage),
+    ).fit()
+    assert res.error is None, res.error
+    assert res.metrics["last"] < res.metrics["first"]
