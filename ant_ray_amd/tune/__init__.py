@@ -396,8 +396,18 @@ class Tuner:
                              "status": "errored" if r.error else "finished",
                              "metrics": r.metrics})
             return rows
+        search_alg = tc.search_alg
+        if isinstance(search_alg, str):
+            search_alg = create_searcher(search_alg)
         if self._restore_state is not None:
             variants = self._restore_state["variants"]
+            search_alg = None  # restored runs replay the recorded variants
+        elif search_alg is not None:
+            # sequential suggest/observe searcher (BayesOpt/TPE/wrappers):
+            # trials are created on demand, up to num_samples
+            search_alg.set_search_properties(tc.metric, tc.mode or "max",
+                                             self.param_space)
+            variants = []
         else:
             variants = BasicVariantGenerator(
                 self.param_space, tc.num_samples, seed=tc.seed).variants()
@@ -463,10 +473,29 @@ class Tuner:
             self._save_state(exp_path, variants, results, running)
         except Exception:
             pass
-        while pending or running:
+        suggested = [len(variants)]
+
+        def _more_suggestions():
+            return (search_alg is not None
+                    and suggested[0] < tc.num_samples)
+
+        while pending or running or _more_suggestions():
             while pending and len(running) < max_conc:
                 idx, cfg = pending.pop(0)
                 launch(idx, cfg)
+            while _more_suggestions() and len(running) < max_conc:
+                idx = suggested[0]
+                cfg = search_alg.suggest(f"{name}_{idx:05d}")
+                if cfg is None:
+                    break  # searcher is concurrency-limited right now
+                variants.append(cfg)
+                suggested[0] += 1
+                launch(idx, cfg)
+            if not running:
+                if pending or _more_suggestions():
+                    time.sleep(0.05)
+                    continue
+                break
             polls = ray.get([t["actor"].poll.remote()
                              for t in running.values()])
             finished_idx = []
@@ -521,6 +550,13 @@ class Tuner:
                     )
                     _fire_cb("on_trial_error" if err else "on_trial_complete",
                              t["iter"], None, t["trial_id"])
+                    if search_alg is not None:
+                        try:
+                            search_alg.on_trial_complete(
+                                t["trial_id"], t["last_metrics"],
+                                error=err is not None)
+                        except Exception:
+                            pass
                     try:
                         ray.kill(t["actor"])
                     except Exception:

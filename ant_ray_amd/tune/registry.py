@@ -50,15 +50,27 @@ def create_scheduler(name: str, **kwargs):
 
 
 def create_searcher(name: str, **kwargs):
+    """Factory (parity tune/search create_searcher). Sequential searchers
+    return INSTANCES (the Tuner drives suggest/observe); the basic variant
+    generator returns its class (Tuner constructs it with the space)."""
     from ant_ray_amd.tune.search import BasicVariantGenerator
+    from ant_ray_amd.tune.search.bayesopt import BayesOptSearch
+    from ant_ray_amd.tune.search.bohb import TuneBOHB
+    from ant_ray_amd.tune.search.hyperopt import HyperOptSearch
+    from ant_ray_amd.tune.search.optuna import OptunaSearch
+    from ant_ray_amd.tune.search.searcher import TPESearch
 
     table = {"variant_generator": BasicVariantGenerator,
-             "random": BasicVariantGenerator}
+             "random": BasicVariantGenerator,
+             "bayesopt": BayesOptSearch,
+             "tpe": TPESearch,
+             "hyperopt": HyperOptSearch,
+             "optuna": OptunaSearch,
+             "bohb": TuneBOHB}
     if name not in table:
         raise ValueError(
-            f"unknown searcher {name!r}; available: {sorted(table)} "
-            "(wrapped third-party searchers are not in this image)")
+            f"unknown searcher {name!r}; available: {sorted(table)}")
     cls = table[name]
-    # BasicVariantGenerator is constructed by the Tuner with the param
-    # space; return the class so Tuner can instantiate it
-    return cls if not kwargs else cls
+    if cls is BasicVariantGenerator:
+        return cls
+    return cls(**kwargs)

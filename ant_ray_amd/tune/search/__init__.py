@@ -185,3 +185,13 @@ def qlograndint(low, high, q) -> Quantized:
 
 def sample_from(fn) -> Function:
     return Function(fn)
+
+
+# sequential searchers (suggest/observe) + meta-searchers
+from ant_ray_amd.tune.search.searcher import (  # noqa: F401,E402
+    BayesOptSearch,
+    ConcurrencyLimiter,
+    Repeater,
+    Searcher,
+    TPESearch,
+)

@@ -359,3 +359,103 @@ def test_sampling_domains_and_analysis(ray_mod, tmp_path_factory):
     assert type(tune.create_scheduler("asha")).__name__ == "ASHAScheduler"
     with pytest.raises(ValueError):
         tune.create_scheduler("nope")
+
+
+def test_bayesopt_searcher_finds_optimum(ray_mod, tmp_path_factory):
+    """Native GP-EI searcher drives the Tuner toward the quadratic's
+    optimum better than its own random warmup."""
+    from ant_ray_amd import tune
+    from ant_ray_amd.train import RunConfig
+    from ant_ray_amd.tune.search import BayesOptSearch
+
+    def objective(config):
+        x, y = config["x"], config["y"]
+        tune.report({"score": -((x - 0.7) ** 2) - (y + 0.3) ** 2})
+
+    search = BayesOptSearch(metric="score", mode="max",
+                            random_search_steps=6, seed=0)
+    tuner = tune.Tuner(
+        objective,
+        param_space={"x": tune.uniform(-2, 2), "y": tune.uniform(-2, 2)},
+        tune_config=tune.TuneConfig(metric="score", mode="max",
+                                    num_samples=24, search_alg=search,
+                                    max_concurrent_trials=2),
+        run_config=RunConfig(name="bo",
+                             storage_path=str(tmp_path_factory.mktemp("bo"))),
+    )
+    res = tuner.fit()
+    best = res.get_best_result(metric="score", mode="max")
+    assert len(list(res)) == 24
+    assert best.metrics["score"] > -0.25, best.metrics
+    # the model-based phase must beat the pure-random warmup's best
+    warm = sorted((r.metrics["score"] for r in list(res)[:6]), reverse=True)
+    assert best.metrics["score"] >= warm[0]
+
+
+def test_tpe_searcher_and_concurrency_limiter(ray_mod, tmp_path_factory):
+    from ant_ray_amd import tune
+    from ant_ray_amd.train import RunConfig
+    from ant_ray_amd.tune.search import ConcurrencyLimiter, TPESearch
+
+    seen = []
+
+    def objective(config):
+        seen.append(config["lr"])
+        tune.report({"loss": abs(config["lr"] - 1e-2)})
+
+    search = ConcurrencyLimiter(
+        TPESearch(metric="loss", mode="min", n_startup=6, seed=1),
+        max_concurrent=2)
+    res = tune.Tuner(
+        objective,
+        param_space={"lr": tune.loguniform(1e-5, 1.0)},
+        tune_config=tune.TuneConfig(metric="loss", mode="min",
+                                    num_samples=18, search_alg=search,
+                                    max_concurrent_trials=4),
+        run_config=RunConfig(name="tpe",
+                             storage_path=str(tmp_path_factory.mktemp("tpe"))),
+    ).fit()
+    best = res.get_best_result(metric="loss", mode="min")
+    assert best.metrics["loss"] < 0.05
+
+
+def test_repeater_averages(ray_mod, tmp_path_factory):
+    from ant_ray_amd import tune
+    from ant_ray_amd.train import RunConfig
+    from ant_ray_amd.tune.search import Repeater, TPESearch
+
+    inner = TPESearch(metric="v", mode="max", n_startup=2, seed=2)
+    rep = Repeater(inner, repeat=3)
+
+    def objective(config):
+        # noisy objective; __trial_index__ marks the repeat index
+        assert "__trial_index__" in config
+        tune.report({"v": config["x"] + 0.01 * config["__trial_index__"]})
+
+    tune.Tuner(
+        objective,
+        param_space={"x": tune.uniform(0, 1)},
+        tune_config=tune.TuneConfig(metric="v", mode="max", num_samples=9,
+                                    search_alg=rep),
+        run_config=RunConfig(name="rep",
+                             storage_path=str(tmp_path_factory.mktemp("rp"))),
+    ).fit()
+    # 9 trials = 3 groups of 3; inner searcher observed exactly 3 results
+    assert len(inner._y) == 3
+
+
+def test_searcher_wrapper_paths():
+    from ant_ray_amd.tune.search.bayesopt import BayesOptSearch  # noqa
+    from ant_ray_amd.tune.search.bohb import TuneBOHB  # noqa
+    from ant_ray_amd.tune.search.hyperopt import HyperOptSearch
+    from ant_ray_amd.tune.search.optuna import OptunaSearch
+    from ant_ray_amd import tune
+
+    s = OptunaSearch(metric="m", mode="min")
+    s.set_search_properties("m", "min", {"x": tune.uniform(0, 1)})
+    cfg = s.suggest("t1")
+    assert 0 <= cfg["x"] <= 1
+    s.on_trial_complete("t1", {"m": 0.5})
+    h = HyperOptSearch(space={"x": tune.uniform(0, 1)}, metric="m")
+    assert 0 <= h.suggest("t2")["x"] <= 1
+    assert type(tune.create_searcher("bayesopt")).__name__ == "BayesOptSearch"
