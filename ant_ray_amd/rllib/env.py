@@ -44,3 +44,40 @@ class CartPoleEnv:
         terminated = bool(abs(self.state[0]) > 2.4 or abs(self.state[2]) > 0.2095)
         truncated = self.t >= self.max_steps
         return self.state.copy(), 1.0, terminated, truncated, {}
+
+
+class Reacher1DEnv:
+    """Continuous-control smoke env: drive a 1-D point mass to a target.
+    obs = [pos, vel, target]; action in [-1, 1]; reward = -|pos - target|
+    - 0.01*a^2. Learnable by SAC in a few thousand steps."""
+
+    observation_dim = 3
+    action_dim = 1
+    continuous = True
+
+    def __init__(self, seed=None, max_steps: int = 60):
+        self.rng = np.random.RandomState(seed)
+        self.max_steps = max_steps
+        self.pos = self.vel = self.target = 0.0
+        self.t = 0
+
+    def _obs(self):
+        return np.array([self.pos, self.vel, self.target], dtype=np.float32)
+
+    def reset(self, *, seed=None, options=None):
+        if seed is not None:
+            self.rng = np.random.RandomState(seed)
+        self.pos = float(self.rng.uniform(-1, 1))
+        self.vel = 0.0
+        self.target = float(self.rng.uniform(-1, 1))
+        self.t = 0
+        return self._obs(), {}
+
+    def step(self, action):
+        a = float(np.clip(np.asarray(action).reshape(-1)[0], -1, 1))
+        self.vel = 0.8 * self.vel + 0.2 * a
+        self.pos = float(np.clip(self.pos + 0.2 * self.vel, -2, 2))
+        self.t += 1
+        reward = -abs(self.pos - self.target) - 0.01 * a * a
+        truncated = self.t >= self.max_steps
+        return self._obs(), reward, False, truncated, {}

@@ -74,6 +74,8 @@ class EnvRunner:
             "values": np.array(val_l, dtype=np.float32),
             "dones": np.array(done_l, dtype=bool),
             "last_value": last_val,
+            "last_obs": self.obs.copy(),  # V-trace re-values with the
+            # LEARNER's value fn (impala.py); the bootstrap state travels
             "episode_returns": rets,
         }
 
