@@ -323,10 +323,10 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                 lens->numel() == B, "lens must be int32 [B] on device");
     lens_ptr = lens->data_ptr<int>();
   }
-  // chunk count: fill the chip (>=512 workgroups) without splitting below
-  // 256 keys per chunk
-  int C = (int)std::min<long>(std::max<long>(1, 512 / std::max(1, B * Hk)),
-                              std::max<long>(1, ((long)T + 255) / 256));
+  // chunk count: fill the chip (target ~1024 workgroups = 2 per CU at
+  // 256 threads) without splitting below 128 keys per chunk
+  int C = (int)std::min<long>(std::max<long>(1, 1024 / std::max(1, B * Hk)),
+                              std::max<long>(1, ((long)T + 127) / 128));
   auto o = torch::empty({B, Hq, D}, q.options());
   torch::Tensor part;
   float* part_ptr = nullptr;

@@ -76,10 +76,22 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
 #pragma unroll
   for (int g = 0; g < DEC_MAX_GQ; ++g) { m_run[g] = -1e30f; l_run[g] = 0.f; }
 
-  // keys walk: group `group` handles keys c0+group, +16, +32 ...
-  for (int key = c0 + group; key < c1; key += 16) {
-    const ushortx8 kv = *(const ushortx8*)(Kp + (long)key * ks + d0);
-    const ushortx8 vv = *(const ushortx8*)(Vp + (long)key * ks + d0);
+  // keys walk: group `group` handles keys c0+group, +16, +32 ... K/V rows
+  // for the NEXT key are prefetched while the current key computes (a
+  // load->use chain per key would serialize ~900-cycle HBM misses behind
+  // ~40 VALU ops; the compiler keeps the prefetch loads in flight across
+  // the softmax body)
+  ushortx8 kv, vv, kv_n, vv_n;
+  const int key0 = c0 + group;
+  if (key0 < c1) {
+    kv = *(const ushortx8*)(Kp + (long)key0 * ks + d0);
+    vv = *(const ushortx8*)(Vp + (long)key0 * ks + d0);
+  }
+  for (int key = key0; key < c1; key += 16) {
+    if (key + 16 < c1) {
+      kv_n = *(const ushortx8*)(Kp + (long)(key + 16) * ks + d0);
+      vv_n = *(const ushortx8*)(Vp + (long)(key + 16) * ks + d0);
+    }
     const floatx8 kf = bf8_to_f32x8(kv);
     const floatx8 vf = bf8_to_f32x8(vv);
 #pragma unroll
@@ -107,6 +119,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
 #pragma unroll
       for (int j = 0; j < 8; ++j) o_acc[g][j] = fmaf(p, vf[j], o_acc[g][j]);
     }
+    kv = kv_n;
+    vv = vv_n;
   }
 
   // ---- combine the 16 groups through LDS.

@@ -59,6 +59,11 @@ class TrainWorker:
         import torch
         import torch.distributed as dist
 
+        # a leaked TORCHELASTIC_USE_AGENT_STORE (e.g. the cluster was
+        # started from inside a torchrun rank) would make this join a
+        # nonexistent agent store at MASTER_PORT and hang the rendezvous
+        for k in [k for k in os.environ if k.startswith("TORCHELASTIC")]:
+            os.environ.pop(k, None)
         os.environ["RANK"] = str(world_rank)
         os.environ["LOCAL_RANK"] = str(local_rank)
         os.environ["WORLD_SIZE"] = str(world_size)

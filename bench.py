@@ -91,6 +91,8 @@ def _train_core(args, device, rank: int, world: int):
     group to already exist when world > 1. Returns (elapsed_s, loss)."""
     import torch
 
+    on_gpu = str(device).startswith("cuda")
+
     from ant_ray_amd.models import build_model
     from ant_ray_amd.parallel import FlatAdamW, FlatDDP, FlatParamManager
 
@@ -126,11 +128,13 @@ def _train_core(args, device, rank: int, world: int):
         loss = step()
     if world > 1:
         dist.barrier()
-    torch.cuda.synchronize()
+    if on_gpu:
+        torch.cuda.synchronize()
     t0 = time.time()
     for _ in range(args.steps):
         loss = step()
-    torch.cuda.synchronize()
+    if on_gpu:
+        torch.cuda.synchronize()
     if world > 1:
         dist.barrier()
     elapsed = time.time() - t0
@@ -204,8 +208,9 @@ def _train_fn(config):
     rank = ctx.get_world_rank()
     world = ctx.get_world_size()
     device = train.torch.get_device()
-    torch.cuda.set_device(device)
-    _setup_tunableop(ctx.get_local_rank())
+    if str(device).startswith("cuda"):
+        torch.cuda.set_device(device)
+        _setup_tunableop(ctx.get_local_rank())
     elapsed, final_loss = _train_core(args, str(device), rank, world)
     train.report({"elapsed": elapsed, "final_loss": final_loss,
                   "world": world})
@@ -219,21 +224,33 @@ def _main_via_ray(args):
         return
     # the worker group does its own rendezvous on a free port; drop the
     # torchrun-provided one so nothing collides with it
-    for k in ("MASTER_ADDR", "MASTER_PORT", "RANK", "LOCAL_RANK",
-              "WORLD_SIZE", "LOCAL_WORLD_SIZE", "GROUP_RANK", "NODE_RANK"):
-        os.environ.pop(k, None)
+    for k in list(os.environ):
+        if k in ("MASTER_ADDR", "MASTER_PORT", "RANK", "LOCAL_RANK",
+                 "WORLD_SIZE", "LOCAL_WORLD_SIZE", "GROUP_RANK",
+                 "NODE_RANK", "ROLE_RANK", "ROLE_WORLD_SIZE",
+                 "ROLE_NAME") or k.startswith("TORCHELASTIC"):
+            # TORCHELASTIC_USE_AGENT_STORE=True makes every descendant's
+            # init_process_group try to JOIN torchrun's agent store at
+            # MASTER_PORT instead of creating its own -> the worker
+            # actors' gloo/nccl rendezvous hangs forever
+            os.environ.pop(k, None)
 
     import ant_ray_amd as ray
     from ant_ray_amd.train import RunConfig, ScalingConfig
     from ant_ray_amd.train.torch import TorchTrainer
 
     n = args.gpus
-    ray.init(num_cpus=max(8, 2 * n), num_gpus=n)
+    # ANTRAY_BENCH_CPU=1: plumbing-test mode — same torchrun->rank0->ray->
+    # TorchTrainer shape on CPU/gloo with a tiny model (CI covers the
+    # exact launch path the driver uses for the N>1 scaling bench)
+    cpu_mode = os.environ.get("ANTRAY_BENCH_CPU") == "1"
+    ray.init(num_cpus=max(8, 2 * n), num_gpus=0 if cpu_mode else n)
     try:
         trainer = TorchTrainer(
             _train_fn,
             train_loop_config=dict(vars(args)),
-            scaling_config=ScalingConfig(num_workers=n, use_gpu=True),
+            scaling_config=ScalingConfig(num_workers=n,
+                                         use_gpu=not cpu_mode),
             run_config=RunConfig(name="bench", storage_path="/tmp/antray_bench"),
         )
         result = trainer.fit()
