@@ -216,14 +216,34 @@ def from_huggingface(hf_dataset) -> Dataset:
         return from_items(list(hf_dataset))
 
 
-def read_parquet(paths, *, columns=None, parallelism: int = -1, **kw) -> Dataset:
-    def reader(f):
-        import pyarrow.parquet as pq
+def read_parquet(paths, *, columns=None, parallelism: int = -1,
+                 filter_expr=None, **kw) -> Dataset:
+    """Parquet source with native projection + filter pushdown: the
+    optimizer folds downstream select_columns/filter_expr into the read
+    tasks (pyarrow.parquet reads only the pruned columns/row groups —
+    parity: reference logical/rules + ParquetDatasource pushdown)."""
 
-        return pq.read_table(f, columns=columns)
+    def make_read(cols, fexpr):
+        def reader(f):
+            import pyarrow.parquet as pq
 
-    return Dataset([ReadOp(name="ReadParquet",
-                           read_tasks=_file_read_tasks(paths, reader, [".parquet"]))])
+            t = pq.read_table(f, columns=cols)
+            if fexpr is not None:
+                from ant_ray_amd.data.expressions import eval_expr_to_column
+
+                t = t.filter(eval_expr_to_column(t, fexpr))
+            return t
+
+        return ReadOp(
+            name="ReadParquet",
+            read_tasks=_file_read_tasks(paths, reader, [".parquet"]),
+            pushdown=lambda columns=None, filter_expr=None: make_read(
+                sorted(set(cols or []) | set(columns)) if (cols and columns)
+                else (columns or cols), fexpr if filter_expr is None
+                else (filter_expr if fexpr is None else fexpr & filter_expr)),
+        )
+
+    return Dataset([make_read(columns, filter_expr)])
 
 
 def read_csv(paths, *, parallelism: int = -1, **kw) -> Dataset:

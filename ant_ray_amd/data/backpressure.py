@@ -1,0 +1,77 @@
+"""Backpressure policies for the streaming executor.
+
+Role parity: reference python/ray/data/_internal/execution/
+backpressure_policy/ — ConcurrencyCapBackpressurePolicy (per-stage task
+cap) and StreamingOutputBackpressurePolicy /
+ObjectStoreMemoryBackpressurePolicy (throttle task admission when the
+shm object store fills). Policies are consulted by executor.py before
+launching another in-flight task; configure via
+DataContext.backpressure_policies.
+"""
+from __future__ import annotations
+
+import time
+
+
+class BackpressurePolicy:
+    def can_add_input(self, stage_name: str, in_flight: int) -> bool:
+        return True
+
+
+class ConcurrencyCapBackpressurePolicy(BackpressurePolicy):
+    """Hard per-stage in-flight cap (reference
+    concurrency_cap_backpressure_policy.py)."""
+
+    def __init__(self, cap: int = 16):
+        self.cap = cap
+
+    def can_add_input(self, stage_name: str, in_flight: int) -> bool:
+        return in_flight < self.cap
+
+
+class ObjectStoreMemoryBackpressurePolicy(BackpressurePolicy):
+    """Stop admitting tasks while the node's shm object store is above
+    `high_watermark` full (reference streaming-output/resource-budget
+    backpressure). The store fill level is polled at most every
+    `poll_interval_s` (a GCS round-trip)."""
+
+    def __init__(self, high_watermark: float = 0.8,
+                 poll_interval_s: float = 1.0):
+        self.high_watermark = high_watermark
+        self.poll_interval_s = poll_interval_s
+        self._last_poll = 0.0
+        self._last_frac = 0.0
+
+    def _store_fraction(self) -> float:
+        now = time.monotonic()
+        if now - self._last_poll < self.poll_interval_s:
+            return self._last_frac
+        self._last_poll = now
+        try:
+            from ant_ray_amd._private.worker import global_worker
+
+            cw = global_worker.core_worker
+            stats = cw.io.run(cw.gcs.call("store_stats", {}, timeout=5),
+                              timeout=10)
+            used = sum(s.get("used_bytes", s.get("bytes_in_use", 0))
+                       for s in stats)
+            cap = sum(s.get("arena_size", 0) for s in stats) or 1
+            self._last_frac = used / cap
+        except Exception:
+            self._last_frac = 0.0
+        return self._last_frac
+
+    def can_add_input(self, stage_name: str, in_flight: int) -> bool:
+        # always allow ONE in-flight task so the pipeline can drain the
+        # store rather than deadlock
+        if in_flight == 0:
+            return True
+        return self._store_fraction() < self.high_watermark
+
+
+def default_policies(ctx) -> list:
+    pols = getattr(ctx, "backpressure_policies", None)
+    if pols is None:
+        pols = [ConcurrencyCapBackpressurePolicy(ctx.max_concurrent_tasks),
+                ObjectStoreMemoryBackpressurePolicy()]
+    return pols

@@ -95,8 +95,10 @@ class Dataset:
         return self.map_batches(_drop, batch_format="pyarrow", batch_size=None)
 
     def select_columns(self, cols: List[str], **kw):
-        return self.map_batches(lambda t: t.select(cols),
-                                batch_format="pyarrow", batch_size=None)
+        return self._with(MapOp(
+            name=f"Select{cols}", kind="map_batches",
+            fn=lambda t: t.select(cols), batch_format="pyarrow",
+            batch_size=None, meta={"type": "select", "columns": list(cols)}))
 
     def rename_columns(self, mapping: Dict[str, str], **kw):
         def _ren(t: pa.Table):
@@ -118,7 +120,10 @@ class Dataset:
                     t = t.append_column(name, colv)
             return t
 
-        return self.map_batches(_apply, batch_format="pyarrow", batch_size=None)
+        return self._with(MapOp(
+            name="WithColumns", kind="map_batches", fn=_apply,
+            batch_format="pyarrow", batch_size=None,
+            meta={"type": "with_columns", "exprs": dict(exprs)}))
 
     def filter_expr(self, expr) -> "Dataset":
         """Vectorized filter by expression (the reference overloads
@@ -129,7 +134,10 @@ class Dataset:
             mask = eval_expr_to_column(t, expr)
             return t.filter(mask)
 
-        return self.map_batches(_apply, batch_format="pyarrow", batch_size=None)
+        return self._with(MapOp(
+            name="FilterExpr", kind="map_batches", fn=_apply,
+            batch_format="pyarrow", batch_size=None,
+            meta={"type": "filter_expr", "expr": expr}))
 
     def limit(self, n: int) -> "Dataset":
         return self._with(LimitOp(name=f"Limit[{n}]", limit=n))

@@ -104,8 +104,10 @@ class _MapWorker:
 
 def execute_plan(ops: List[Any]) -> Iterator[Any]:
     """Yields ObjectRefs of output blocks, streaming."""
+    from ant_ray_amd.data.optimizer import optimize
+
     ctx = DataContext.get_current()
-    stages = fuse_stages(ops)
+    stages = fuse_stages(optimize(ops))
     stream: Iterator[Any] = iter(())
     for stage in stages:
         if isinstance(stage, ReadOp):
@@ -123,21 +125,27 @@ def execute_plan(ops: List[Any]) -> Iterator[Any]:
 
 
 def _read_stage(op: ReadOp, ctx) -> Iterator[Any]:
+    from ant_ray_amd.data.backpressure import default_policies
+
+    policies = default_policies(ctx)
     read_remote = ray.remote(num_cpus=1)(_run_read_task)
     window = collections.deque()
     tasks = iter(op.read_tasks)
-    try:
-        for _ in range(ctx.max_concurrent_tasks):
-            window.append(read_remote.remote(next(tasks)))
-    except StopIteration:
-        tasks = None
-    while window:
-        ref = window.popleft()
-        if tasks is not None:
+
+    def admit():
+        nonlocal tasks
+        while (tasks is not None
+               and all(p.can_add_input(op.name, len(window))
+                       for p in policies)):
             try:
                 window.append(read_remote.remote(next(tasks)))
             except StopIteration:
                 tasks = None
+
+    admit()
+    while window:
+        ref = window.popleft()
+        admit()
         yield ref
 
 
@@ -158,12 +166,18 @@ def _map_stage(stream: Iterator[Any], chain: List[MapOp], ctx) -> Iterator[Any]:
     res = {"num_cpus": chain[0].num_cpus or 1}
     if any(op.num_gpus for op in chain):
         res["num_gpus"] = max(op.num_gpus or 0 for op in chain)
+    from ant_ray_amd.data.backpressure import default_policies
+
+    policies = default_policies(ctx)
+    name = chain[0].name
     map_remote = ray.remote(**res)(_apply_chain)
     window = collections.deque()
     exhausted = False
     it = iter(stream)
     while True:
-        while not exhausted and len(window) < ctx.max_concurrent_tasks:
+        while (not exhausted
+               and all(p.can_add_input(name, len(window))
+                       for p in policies)):
             try:
                 block_ref = next(it)
             except StopIteration:

@@ -20,6 +20,10 @@ class ReadOp:
     read_tasks: List[Callable]
     # estimated row count if cheaply known (range/from_items)
     num_rows: Optional[int] = None
+    # optimizer hook: pushdown(columns=..., filter_expr=...) -> new ReadOp
+    # with pruning/filtering folded into the read tasks (parquet supports
+    # both natively; None = the source cannot push down)
+    pushdown: Optional[Callable] = None
 
 
 @dataclass
@@ -39,6 +43,9 @@ class MapOp:
     num_cpus: Optional[float] = None
     num_gpus: Optional[float] = None
     concurrency: Optional[Any] = None
+    # optimizer metadata for declarative ops: {"type": "select"|"filter_expr"
+    # |"with_columns"|"drop"|"rename", ...} — None for opaque user UDFs
+    meta: Optional[Dict[str, Any]] = None
 
 
 @dataclass

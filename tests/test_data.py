@@ -330,3 +330,91 @@ def test_from_refs_read_sql_summary(ray_mod, tmp_path):
     # unavailable readers raise informative errors
     with pytest.raises(NotImplementedError):
         rd.read_images("/tmp/x")
+
+
+def test_expression_namespaces(ray_mod):
+    import pyarrow as pa
+
+    from ant_ray_amd import data
+    from ant_ray_amd.data.expressions import col, lit
+
+    ds = data.from_items([
+        {"s": " Hello ", "n": -3.7, "tags": [1, 2, 3]},
+        {"s": "world", "n": 2.2, "tags": [4]},
+    ])
+    out = ds.with_columns({
+        "lower": col("s").str.strip().str.lower(),
+        "has_l": col("s").str.contains("l"),
+        "absn": col("n").abs().round(0),
+        "ntags": col("tags").list.len(),
+        "inset": col("n").round(0).cast(pa.int64()).is_in([-4, 2]),
+    }).take_all()
+    assert [r["lower"] for r in out] == ["hello", "world"]
+    assert [r["absn"] for r in out] == [4.0, 2.0]
+    assert [r["ntags"] for r in out] == [3, 1]
+    assert [r["inset"] for r in out] == [True, True]
+    # between + fill_null
+    ds2 = data.from_items([{"x": 1}, {"x": None}, {"x": 9}])
+    vals = ds2.with_columns({"ok": col("x").fill_null(0).between(0, 5)}
+                            ).take_all()
+    assert [r["ok"] for r in vals] == [True, True, False]
+
+
+def test_parquet_projection_and_filter_pushdown(ray_mod, tmp_path):
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from ant_ray_amd import data
+    from ant_ray_amd.data.expressions import col
+    from ant_ray_amd.data.optimizer import optimize
+    from ant_ray_amd.data.plan import ReadOp
+
+    t = pa.table({"a": list(range(100)), "b": [f"s{i}" for i in range(100)],
+                  "c": [i * 1.5 for i in range(100)]})
+    pq.write_table(t, str(tmp_path / "f.parquet"))
+
+    ds = (data.read_parquet(str(tmp_path / "f.parquet"))
+          .filter_expr(col("a") < 10)
+          .select_columns(["a", "c"]))
+    # the optimizer folds both into the ReadOp
+    opt = optimize(ds._ops)
+    assert len(opt) == 1 and isinstance(opt[0], ReadOp)
+    rows = ds.take_all()
+    assert len(rows) == 10
+    assert set(rows[0].keys()) == {"a", "c"}
+
+
+def test_filter_reorder_before_with_columns(ray_mod):
+    from ant_ray_amd import data
+    from ant_ray_amd.data.expressions import col
+    from ant_ray_amd.data.optimizer import filter_reorder
+
+    ds = (data.from_items([{"x": i} for i in range(20)])
+          .with_columns({"y": col("x") * 2})
+          .filter_expr(col("x") > 15))
+    ops = filter_reorder(ds._ops)
+    metas = [(o.meta or {}).get("type") for o in ops[1:]]
+    assert metas == ["filter_expr", "with_columns"]  # swapped
+    # and NOT swapped when the filter uses the created column
+    ds2 = (data.from_items([{"x": i} for i in range(20)])
+           .with_columns({"y": col("x") * 2})
+           .filter_expr(col("y") > 15))
+    metas2 = [(o.meta or {}).get("type") for o in filter_reorder(ds2._ops)[1:]]
+    assert metas2 == ["with_columns", "filter_expr"]
+    assert len(ds.take_all()) == 4
+
+
+def test_backpressure_policies():
+    from ant_ray_amd.data.backpressure import (
+        ConcurrencyCapBackpressurePolicy,
+        ObjectStoreMemoryBackpressurePolicy,
+    )
+
+    cap = ConcurrencyCapBackpressurePolicy(cap=2)
+    assert cap.can_add_input("s", 0) and cap.can_add_input("s", 1)
+    assert not cap.can_add_input("s", 2)
+    mem = ObjectStoreMemoryBackpressurePolicy(high_watermark=0.8)
+    mem._last_frac = 0.9
+    mem._last_poll = __import__("time").monotonic() + 100  # pin the poll
+    assert mem.can_add_input("s", 0)  # always drainable
+    assert not mem.can_add_input("s", 3)
