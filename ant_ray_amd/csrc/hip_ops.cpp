@@ -323,9 +323,10 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                 lens->numel() == B, "lens must be int32 [B] on device");
     lens_ptr = lens->data_ptr<int>();
   }
-  // chunk count: fill the chip (target ~1024 workgroups = 2 per CU at
-  // 256 threads) without splitting below 128 keys per chunk
-  int C = (int)std::min<long>(std::max<long>(1, 1024 / std::max(1, B * Hk)),
+  // chunk count: fill the chip (~512 workgroups) without splitting below
+  // 128 keys per chunk (r2e A/B: a 1024-WG target halved per-chunk work
+  // and lost 15% at B=8 T=4096; the 128-key floor keeps the small-B win)
+  int C = (int)std::min<long>(std::max<long>(1, 512 / std::max(1, B * Hk)),
                               std::max<long>(1, ((long)T + 127) / 128));
   auto o = torch::empty({B, Hq, D}, q.options());
   torch::Tensor part;

@@ -280,9 +280,10 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
 
   __shared__ ushort_t q_rm[32 * KROW];
   __shared__ ushort_t do_tr[ATTN_D * TROW];
-  __shared__ ushort_t scr_p[4][32 * SCR_ROW];
 
-  // own K rows as A-fragments in registers
+  // own K rows in registers — used as the B operand of S = Q.K^T
+  // (B[k=d][n=key]: n = lane-local key, k-slices = d — the same per-lane
+  // bytes an A-fragment holds, so the load is unchanged)
   bf16x8 kfr[8];
   {
     const long kg = (long)min(key_row, S - 1) * ks;
@@ -332,10 +333,11 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
       da2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
     }
 
-    const int q_here = tq0 + (lane & 31);
-    const bool qv = q_here < S;
-    const float L_q = qv ? Lp[q_here] : 0.f;
-
+    // S = Q.K^T with q REG-SPREAD, key LANE-LOCAL (operand roles swapped
+    // vs the old scratch version): C[m=q][n=key] — P^T's A-fragment
+    // (A[m=key][k=q]) then comes from the SAME in-register
+    // cvt_pk+permlane transpose the forward kernel uses for P (T12),
+    // removing the wave-scratch LDS round trip entirely.
     floatx16 st_acc = {};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -343,36 +345,54 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
       bf16x8 qfr = __builtin_bit_cast(
           bf16x8,
           *(const ushortx8*)(&q_rm[rm_off(lane & 31, 2 * stp + (lane >> 5))]));
-      st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr[stp], qfr, st_acc,
+      st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfr, kfr[stp], st_acc,
                                                        0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
 
+    // per-q-row lse fetched once per lane, broadcast per reg row by shfl
+    const float L_lane = Lp[min(tq0 + (lane & 31), S - 1)];
     const int mrow_base = 4 * (lane >> 5);
-    ushort_t* my_scr = (ushort_t*)scr_p[wave];
+    const int key_here = key0 + (lane & 31);
+    float p_val[16];
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
-      const int key = key0 + (j & 3) + 8 * (j >> 2) + mrow_base;
-      const bool dead = (causal && key > q_here) || key >= S || !qv;
-      const float p = dead ? 0.f : __builtin_exp2f(st_acc[j] * c_log2 - L_q);
-      const int krow = (j & 3) + 8 * (j >> 2) + mrow_base;
-      my_scr[krow * SCR_ROW + (lane & 31)] = f2bf(p);
+      const int qrow_j = (j & 3) + 8 * (j >> 2) + mrow_base;
+      const int q_abs = tq0 + qrow_j;
+      const float L_q = __shfl(L_lane, qrow_j);
+      const bool dead = (causal && key_here > q_abs) || q_abs >= S ||
+                        key_here >= S;
+      p_val[j] = dead ? 0.f
+                      : __builtin_exp2f(st_acc[j] * c_log2 - L_q);
     }
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    // C[q][key] -> A[m=key][k=q] fragments in-register (fwd T12 pattern)
+    unsigned own_pk[8];
+#pragma unroll
+    for (int b2 = 0; b2 < 4; ++b2) {
+      own_pk[2 * b2] = cvt_pk_bf16b(p_val[4 * b2], p_val[4 * b2 + 1]);
+      own_pk[2 * b2 + 1] = cvt_pk_bf16b(p_val[4 * b2 + 2], p_val[4 * b2 + 3]);
+    }
+    bf16x8 pf[2];
+    {
+      auto r0 = __builtin_amdgcn_permlane32_swap(own_pk[0], own_pk[2], false, false);
+      auto r1 = __builtin_amdgcn_permlane32_swap(own_pk[1], own_pk[3], false, false);
+      auto r2 = __builtin_amdgcn_permlane32_swap(own_pk[4], own_pk[6], false, false);
+      auto r3 = __builtin_amdgcn_permlane32_swap(own_pk[5], own_pk[7], false, false);
+      unsigned w0[4] = {r0[0], r1[0], r0[1], r1[1]};
+      unsigned w1[4] = {r2[0], r3[0], r2[1], r3[1]};
+      pf[0] = __builtin_bit_cast(bf16x8, *(uint32x4_t*)w0);
+      pf[1] = __builtin_bit_cast(bf16x8, *(uint32x4_t*)w1);
+    }
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt) {
       const int d0 = dt * 32 + (lane & 31);
 #pragma unroll
       for (int kstep = 0; kstep < 2; ++kstep) {
-        bf16x8 pa = __builtin_bit_cast(
-            bf16x8, *(const ushortx8*)(
-                        &my_scr[(lane & 31) * SCR_ROW + kstep * 16 +
-                                (lane >> 5) * 8]));
         bf16x8 dob = __builtin_bit_cast(
             bf16x8, *(const ushortx8*)(
                         &do_tr[tr_off(d0, kstep * 16 + (lane >> 5) * 8)]));
-        dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob,
+        dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf[kstep], dob,
                                                              dv_acc[dt], 0, 0, 0);
       }
     }
@@ -418,8 +438,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
   __shared__ ushort_t q_rm[32 * KROW];
   __shared__ ushort_t do_rm[32 * KROW];
   __shared__ ushort_t q_tr[ATTN_D * TROW];
-  __shared__ ushort_t scr_ds[4][32 * SCR_ROW];
 
+  // own K/V rows in registers — B operands of S = Q.K^T and dP = dO.V^T
+  // (key lane-local; the same per-lane bytes an A-fragment holds)
   bf16x8 kfr[8], vfr[8];
   {
     const long kg = (long)min(key_row, S - 1) * ks;
@@ -477,11 +498,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
       da2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
     }
 
-    const int q_here = tq0 + (lane & 31);
-    const bool qv = q_here < S;
-    const float L_q = qv ? Lp[q_here] : 0.f;
-    const float D_q = qv ? Dp[q_here] : 0.f;
-
+    // S and dP with q REG-SPREAD, key LANE-LOCAL (roles swapped, as in
+    // attn_bwd_dv): dS^T's A-fragment comes from the in-register T12
+    // transpose — no wave-scratch LDS round trip.
     floatx16 st_acc = {}, dp_acc = {};
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -489,41 +508,61 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
       const int blk = 2 * stp + (lane >> 5);
       bf16x8 qfr = __builtin_bit_cast(
           bf16x8, *(const ushortx8*)(&q_rm[rm_off(lane & 31, blk)]));
-      st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr[stp], qfr, st_acc,
+      st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfr, kfr[stp], st_acc,
                                                        0, 0, 0);
       bf16x8 dofr = __builtin_bit_cast(
           bf16x8, *(const ushortx8*)(&do_rm[rm_off(lane & 31, blk)]));
-      dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr[stp], dofr, dp_acc,
+      dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dofr, vfr[stp], dp_acc,
                                                        0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
 
+    // per-q-row lse/delta fetched once per lane, broadcast by shfl
+    const int q_lane = min(tq0 + (lane & 31), S - 1);
+    const float L_lane = Lp[q_lane];
+    const float D_lane = Dp[q_lane];
     const int mrow_base = 4 * (lane >> 5);
-    ushort_t* my_scr = (ushort_t*)scr_ds[wave];
+    const int key_here = key0 + (lane & 31);
+    float ds_val[16];
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
-      const int key = key0 + (j & 3) + 8 * (j >> 2) + mrow_base;
-      const bool dead = (causal && key > q_here) || key >= S || !qv;
-      const float p = dead ? 0.f : __builtin_exp2f(st_acc[j] * c_log2 - L_q);
-      const int krow = (j & 3) + 8 * (j >> 2) + mrow_base;
-      my_scr[krow * SCR_ROW + (lane & 31)] =
-          f2bf(scale * p * (dp_acc[j] - D_q));
+      const int qrow_j = (j & 3) + 8 * (j >> 2) + mrow_base;
+      const int q_abs = tq0 + qrow_j;
+      const float L_q = __shfl(L_lane, qrow_j);
+      const float D_q = __shfl(D_lane, qrow_j);
+      const bool dead = (causal && key_here > q_abs) || q_abs >= S ||
+                        key_here >= S;
+      const float p = dead ? 0.f
+                           : __builtin_exp2f(st_acc[j] * c_log2 - L_q);
+      ds_val[j] = scale * p * (dp_acc[j] - D_q);
     }
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    unsigned own_pk[8];
+#pragma unroll
+    for (int b2 = 0; b2 < 4; ++b2) {
+      own_pk[2 * b2] = cvt_pk_bf16b(ds_val[4 * b2], ds_val[4 * b2 + 1]);
+      own_pk[2 * b2 + 1] = cvt_pk_bf16b(ds_val[4 * b2 + 2], ds_val[4 * b2 + 3]);
+    }
+    bf16x8 dsf[2];
+    {
+      auto r0 = __builtin_amdgcn_permlane32_swap(own_pk[0], own_pk[2], false, false);
+      auto r1 = __builtin_amdgcn_permlane32_swap(own_pk[1], own_pk[3], false, false);
+      auto r2 = __builtin_amdgcn_permlane32_swap(own_pk[4], own_pk[6], false, false);
+      auto r3 = __builtin_amdgcn_permlane32_swap(own_pk[5], own_pk[7], false, false);
+      unsigned w0[4] = {r0[0], r1[0], r0[1], r1[1]};
+      unsigned w1[4] = {r2[0], r3[0], r2[1], r3[1]};
+      dsf[0] = __builtin_bit_cast(bf16x8, *(uint32x4_t*)w0);
+      dsf[1] = __builtin_bit_cast(bf16x8, *(uint32x4_t*)w1);
+    }
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt) {
       const int d0 = dt * 32 + (lane & 31);
 #pragma unroll
       for (int kstep = 0; kstep < 2; ++kstep) {
-        bf16x8 dsa = __builtin_bit_cast(
-            bf16x8, *(const ushortx8*)(
-                        &my_scr[(lane & 31) * SCR_ROW + kstep * 16 +
-                                (lane >> 5) * 8]));
         bf16x8 qtb = __builtin_bit_cast(
             bf16x8, *(const ushortx8*)(
                         &q_tr[tr_off(d0, kstep * 16 + (lane >> 5) * 8)]));
-        dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsa, qtb,
+        dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsf[kstep], qtb,
                                                              dk_acc[dt], 0, 0, 0);
       }
     }
