@@ -11,10 +11,14 @@
 //     4 permlanes + 8 cvt_pk replace 16 ds_bpermute + scalar bf16 packing,
 //   * -inf masking (exp2(-inf - m) == 0) so the exp path has NO branches,
 //   * K staged row-major with the T2 XOR swizzle (block ^= row&7) and V
-//     staged TRANSPOSED (VROW=40, k ^= 8*((d>>4)&3)): conflict-reduced
+//     staged TRANSPOSED (VROW=40+, k ^= 8*((d>>4)&3)): conflict-reduced
 //     ds_read_b128 fragment reads on both,
 //   * T14 issue-early staging: tile t+1's global loads are in flight while
-//     tile t computes.
+//     tile t computes,
+//   * tile width templated (NT sub-tiles of 32 keys): NT=2 = 64-key tiles,
+//     32 MFMAs per barrier interval instead of 16 — the 2-phase-stall
+//     analysis in the playbook shows barrier/staging overhead amortizes
+//     with MFMA-per-phase.
 //
 // Layout contract (one workgroup = 4 waves = one 128-row Q block):
 //   wave w handles q rows [m0 + 32w, m0 + 32w + 32)
@@ -34,7 +38,6 @@ typedef unsigned int uint32x4 __attribute__((ext_vector_type(4)));
 
 #define ATTN_D 128
 #define BLOCK_M 128   // q rows per workgroup (32 per wave)
-#define BLOCK_N 32    // keys per tile
 #define KROW 128      // K LDS row stride (T2 swizzle instead of padding)
 #define VROW 40       // V^T LDS row stride (odd multiple of 8)
 #define RESCALE_THR 8.0f  // T13: defer O-rescale until max grows > 2^8
@@ -57,10 +60,12 @@ __device__ __forceinline__ int k_lds_off(int row, int blk) {
 
 // V^T LDS: element offset of (d row, k element), k-block swizzled by d
 __device__ __forceinline__ int v_lds_off(int d, int k) {
-  return d * VROW + (k ^ (8 * ((d >> 4) & 3)));
+  return d * VROW + ((k & 31) ^ (8 * ((d >> 4) & 3))) + (k >> 5) * (ATTN_D * VROW);
 }
 
-extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
+// NT = 32-key sub-tiles per staged tile (1 -> 32-key tiles, 2 -> 64).
+template <int NT>
+__global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     const ushort_t* __restrict__ Q,  // [B, Hq, S, D] via strides
     const ushort_t* __restrict__ K,  // [B, Hk, S, D]
     const ushort_t* __restrict__ V,  // [B, Hk, S, D]
@@ -72,6 +77,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     long ob, long oh, long os,
     float scale_log2,                // softmax_scale * log2(e)
     int causal) {
+  constexpr int BN = 32 * NT;       // keys per staged tile
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int m_block = blockIdx.x;
@@ -88,8 +94,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   const ushort_t* Kp = K + (long)b * kb + (long)hk * kh;
   const ushort_t* Vp = V + (long)b * kb + (long)hk * kh;
 
-  __shared__ ushort_t k_lds[2][32 * KROW];
-  __shared__ ushort_t v_lds[2][ATTN_D * VROW];
+  __shared__ ushort_t k_lds[2][BN * KROW];
+  __shared__ ushort_t v_lds[2][NT * ATTN_D * VROW];
 
   // ---- Q fragments: B[k=d][n=q]
   bf16x8 qf[8];
@@ -107,79 +113,97 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
 
   const int wg_last_row = m_block * BLOCK_M + (BLOCK_M - 1);
   const int n_end_row = causal ? min(wg_last_row, S - 1) : (S - 1);
-  const int n_tiles = (n_end_row / BLOCK_N) + 1;
-  const int my_last_tile = causal ? ((q0 + 31) / BLOCK_N) : (n_tiles - 1);
+  const int n_tiles = (n_end_row / BN) + 1;
+  const int my_last_tile = causal ? ((q0 + 31) / BN) : (n_tiles - 1);
 
-  // staging geometry: thread covers (row = tid>>3, 16 elems at (tid&7)*16)
+  // staging geometry: thread covers NT rows 32 apart, 16 elems each at
+  // (row = tid>>3 (+32), col = (tid&7)*16)
   const int st_row = threadIdx.x >> 3;
   const int st_col = (threadIdx.x & 7) * 16;
 
   // T14: tile 0 loads issued before the loop
-  ushortx8 kv0, kv1, vv0, vv1;
-  {
-    const int krow_g = min(st_row, S - 1);
-    kv0 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
-    kv1 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col + 8);
-    vv0 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col);
-    vv1 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col + 8);
+  ushortx8 kv0[NT], kv1[NT], vv0[NT], vv1[NT];
+#pragma unroll
+  for (int h = 0; h < NT; ++h) {
+    const int krow_g = min(st_row + 32 * h, S - 1);
+    kv0[h] = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
+    kv1[h] = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col + 8);
+    vv0[h] = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col);
+    vv1[h] = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col + 8);
   }
 
   for (int t = 0; t < n_tiles; ++t) {
-    const int n0 = t * BLOCK_N;
+    const int n0 = t * BN;
     const int buf = t & 1;
     // double buffer: write tile t into buf while tile t-1 (other buf) may
     // still be read; ONE barrier per tile orders write(t) vs read(t)
-    *(ushortx8*)(&k_lds[buf][k_lds_off(st_row, st_col / 8)]) = kv0;
-    *(ushortx8*)(&k_lds[buf][k_lds_off(st_row, st_col / 8 + 1)]) = kv1;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      v_lds[buf][v_lds_off(st_col + j, st_row)] = vv0[j];
-      v_lds[buf][v_lds_off(st_col + 8 + j, st_row)] = vv1[j];
+    for (int h = 0; h < NT; ++h) {
+      *(ushortx8*)(&k_lds[buf][k_lds_off(st_row + 32 * h, st_col / 8)]) = kv0[h];
+      *(ushortx8*)(&k_lds[buf][k_lds_off(st_row + 32 * h, st_col / 8 + 1)]) = kv1[h];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        v_lds[buf][v_lds_off(st_col + j, st_row + 32 * h)] = vv0[h][j];
+        v_lds[buf][v_lds_off(st_col + 8 + j, st_row + 32 * h)] = vv1[h][j];
+      }
     }
     __syncthreads();  // tile t staged in buf; tile t-1 reads are also done
     if (t + 1 < n_tiles) {
-      const int krow_g = min((t + 1) * BLOCK_N + st_row, S - 1);
-      kv0 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
-      kv1 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col + 8);
-      vv0 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col);
-      vv1 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col + 8);
+#pragma unroll
+      for (int h = 0; h < NT; ++h) {
+        const int krow_g = min((t + 1) * BN + st_row + 32 * h, S - 1);
+        kv0[h] = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
+        kv1[h] = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col + 8);
+        vv0[h] = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col);
+        vv1[h] = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col + 8);
+      }
     }
     if (t > my_last_tile || q0 >= S) continue;  // masked wave: stage only
 
     // ---- S^T = K . Q^T  (C: n = q = lane&31, m = key offset)
-    floatx16 st_acc = {};
-    __builtin_amdgcn_s_setprio(1);  // T5: keep the matrix pipe fed
+    floatx16 st_acc[NT] = {};
+    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int stp = 0; stp < 8; ++stp) {
-      bf16x8 kf = __builtin_bit_cast(
-          bf16x8, *(const ushortx8*)(
-                      &k_lds[buf][k_lds_off(lane & 31, 2 * stp + (lane >> 5))]));
-      st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[stp], st_acc,
-                                                       0, 0, 0);
-    }
+    for (int h = 0; h < NT; ++h)
+#pragma unroll
+      for (int stp = 0; stp < 8; ++stp) {
+        bf16x8 kf = __builtin_bit_cast(
+            bf16x8, *(const ushortx8*)(
+                        &k_lds[buf][k_lds_off((lane & 31) + 32 * h,
+                                              2 * stp + (lane >> 5))]));
+        st_acc[h] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[stp],
+                                                            st_acc[h], 0, 0, 0);
+      }
     __builtin_amdgcn_s_setprio(0);
 
     // ---- mask + scale; dead scores are -inf so exp2 underflows to 0
     const int mrow_base = 4 * (lane >> 5);
-    float s_val[16];
+    float s_val[NT][16];
     const bool needs_mask =
-        (causal && n0 + BLOCK_N - 1 > q0) || (n0 + BLOCK_N > S) || !q_valid;
+        (causal && n0 + BN - 1 > q0) || (n0 + BN > S) || !q_valid;
     if (needs_mask) {
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const int key = n0 + (j & 3) + 8 * (j >> 2) + mrow_base;
-        const bool dead = (causal && key > q_row) || key >= S || !q_valid;
-        s_val[j] = dead ? NEG_INF : st_acc[j] * scale_log2;
-      }
+      for (int h = 0; h < NT; ++h)
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const int key = n0 + 32 * h + (j & 3) + 8 * (j >> 2) + mrow_base;
+          const bool dead = (causal && key > q_row) || key >= S || !q_valid;
+          s_val[h][j] = dead ? NEG_INF : st_acc[h][j] * scale_log2;
+        }
     } else {
 #pragma unroll
-      for (int j = 0; j < 16; ++j) s_val[j] = st_acc[j] * scale_log2;
+      for (int h = 0; h < NT; ++h)
+#pragma unroll
+        for (int j = 0; j < 16; ++j) s_val[h][j] = st_acc[h][j] * scale_log2;
     }
 
     // ---- online softmax with defer-max (T13)
-    float tmax = fmaxf(s_val[0], s_val[1]);
+    float tmax = fmaxf(s_val[0][0], s_val[0][1]);
 #pragma unroll
-    for (int j = 2; j < 16; ++j) tmax = fmaxf(tmax, s_val[j]);
+    for (int h = 0; h < NT; ++h)
+#pragma unroll
+      for (int j = h == 0 ? 2 : 0; j < 16; ++j)
+        tmax = fmaxf(tmax, s_val[h][j]);
     tmax = fmaxf(tmax, xor32(tmax));
     if (__any(tmax > m_run + RESCALE_THR)) {
       const float m_new = fmaxf(m_run, tmax);
@@ -196,42 +220,46 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
         for (int j = 0; j < 16; ++j) o_acc[dt][j] *= alpha_j[j];
     }
     float psum = 0.f;
-    float p_val[16];
+    float p_val[NT][16];
 #pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      p_val[j] = __builtin_exp2f(s_val[j] - m_run);  // bounded by 2^THR
-      psum += p_val[j];
-    }
+    for (int h = 0; h < NT; ++h)
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        p_val[h][j] = __builtin_exp2f(s_val[h][j] - m_run);  // <= 2^THR
+        psum += p_val[h][j];
+      }
     psum += xor32(psum);
     l_run += psum;
 
     // ---- P A-fragments via cvt_pk + permlane32_swap (T12).
-    // own_pk[2b+c] packs this lane-half's keys (8b + 4h + 2c, +1);
-    // permlane32_swap(D, S) exchanges D's hi lanes with S's lo lanes, so
-    // swap(own_pk[x], own_pk[x+2]) returns the pf words for BOTH halves.
-    unsigned own_pk[8];
+    bf16x8 pf[2 * NT];
 #pragma unroll
-    for (int b2 = 0; b2 < 4; ++b2) {
-      own_pk[2 * b2] = cvt_pk_bf16(p_val[4 * b2], p_val[4 * b2 + 1]);
-      own_pk[2 * b2 + 1] = cvt_pk_bf16(p_val[4 * b2 + 2], p_val[4 * b2 + 3]);
+    for (int h = 0; h < NT; ++h) {
+      unsigned own_pk[8];
+#pragma unroll
+      for (int b2 = 0; b2 < 4; ++b2) {
+        own_pk[2 * b2] = cvt_pk_bf16(p_val[h][4 * b2], p_val[h][4 * b2 + 1]);
+        own_pk[2 * b2 + 1] =
+            cvt_pk_bf16(p_val[h][4 * b2 + 2], p_val[h][4 * b2 + 3]);
+      }
+      uint32x4 pw0, pw1;
+      {
+        auto r0 = __builtin_amdgcn_permlane32_swap(own_pk[0], own_pk[2],
+                                                   false, false);
+        auto r1 = __builtin_amdgcn_permlane32_swap(own_pk[1], own_pk[3],
+                                                   false, false);
+        pw0[0] = r0[0]; pw0[2] = r0[1];
+        pw0[1] = r1[0]; pw0[3] = r1[1];
+        auto r2 = __builtin_amdgcn_permlane32_swap(own_pk[4], own_pk[6],
+                                                   false, false);
+        auto r3 = __builtin_amdgcn_permlane32_swap(own_pk[5], own_pk[7],
+                                                   false, false);
+        pw1[0] = r2[0]; pw1[2] = r2[1];
+        pw1[1] = r3[0]; pw1[3] = r3[1];
+      }
+      pf[2 * h] = __builtin_bit_cast(bf16x8, pw0);
+      pf[2 * h + 1] = __builtin_bit_cast(bf16x8, pw1);
     }
-    uint32x4 pw0, pw1;
-    {
-      auto r0 = __builtin_amdgcn_permlane32_swap(own_pk[0], own_pk[2], false,
-                                                 false);
-      auto r1 = __builtin_amdgcn_permlane32_swap(own_pk[1], own_pk[3], false,
-                                                 false);
-      pw0[0] = r0[0]; pw0[2] = r0[1];
-      pw0[1] = r1[0]; pw0[3] = r1[1];
-      auto r2 = __builtin_amdgcn_permlane32_swap(own_pk[4], own_pk[6], false,
-                                                 false);
-      auto r3 = __builtin_amdgcn_permlane32_swap(own_pk[5], own_pk[7], false,
-                                                 false);
-      pw1[0] = r2[0]; pw1[2] = r2[1];
-      pw1[1] = r3[0]; pw1[3] = r3[1];
-    }
-    bf16x8 pf[2] = {__builtin_bit_cast(bf16x8, pw0),
-                    __builtin_bit_cast(bf16x8, pw1)};
 
     // ---- O += P . V   (B[k][n=d] = one b128 from transposed V)
     __builtin_amdgcn_s_setprio(1);
@@ -239,7 +267,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     for (int dt = 0; dt < 4; ++dt) {
       const int d0 = dt * 32 + (lane & 31);
 #pragma unroll
-      for (int kstep = 0; kstep < 2; ++kstep) {
+      for (int kstep = 0; kstep < 2 * NT; ++kstep) {
         bf16x8 vf = __builtin_bit_cast(
             bf16x8, *(const ushortx8*)(
                         &v_lds[buf][v_lds_off(d0, kstep * 16 + (lane >> 5) * 8)]));
@@ -280,9 +308,23 @@ extern "C" void launch_attn_fwd(const void* q, const void* k, const void* v,
                                 float scale, int causal, void* stream) {
   dim3 grid((S + BLOCK_M - 1) / BLOCK_M, Hq, B);
   const float scale_log2 = scale * 1.4426950408889634f;
-  hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(256), 0,
-                     (hipStream_t)stream, (const ushort_t*)q,
-                     (const ushort_t*)k, (const ushort_t*)v, (ushort_t*)o,
-                     lse, S, Hq, Hk, qb, qh, qs, kb, kh, ks, ob, oh, os,
-                     scale_log2, causal);
+  // tile width: 64-key tiles by default (32 MFMAs per barrier interval);
+  // ANTRAY_FWD_NT=1 keeps the 32-key variant for A/B
+  static const int nt = [] {
+    const char* e = getenv("ANTRAY_FWD_NT");
+    return (e && e[0] == '1') ? 1 : 2;
+  }();
+  if (nt == 2) {
+    hipLaunchKernelGGL(attn_fwd_kernel<2>, grid, dim3(256), 0,
+                       (hipStream_t)stream, (const ushort_t*)q,
+                       (const ushort_t*)k, (const ushort_t*)v, (ushort_t*)o,
+                       lse, S, Hq, Hk, qb, qh, qs, kb, kh, ks, ob, oh, os,
+                       scale_log2, causal);
+  } else {
+    hipLaunchKernelGGL(attn_fwd_kernel<1>, grid, dim3(256), 0,
+                       (hipStream_t)stream, (const ushort_t*)q,
+                       (const ushort_t*)k, (const ushort_t*)v, (ushort_t*)o,
+                       lse, S, Hq, Hk, qb, qh, qs, kb, kh, ks, ob, oh, os,
+                       scale_log2, causal);
+  }
 }
