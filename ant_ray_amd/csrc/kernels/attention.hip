@@ -39,12 +39,32 @@ typedef unsigned int uint32x4 __attribute__((ext_vector_type(4)));
 #define ATTN_D 128
 #define BLOCK_M 128   // q rows per workgroup (32 per wave)
 #define KROW 128      // K LDS row stride (T2 swizzle instead of padding)
-#define VROW 40       // V^T LDS row stride (odd multiple of 8)
 #define RESCALE_THR 8.0f  // T13: defer O-rescale until max grows > 2^8
+
+// V image for the hardware transpose read (T10, ds_read_b64_tr_b16):
+// subtiled [key/4][d/16] of row-major [4 key][16 d] 64-element tiles.
+// Strides are bank-staggered (empirically derived, see tools/probes/
+// tr16_probe.hip for the instruction's lane->element mapping):
+//  * d-subtile stride 72 elems (144 B): the 8 d-subtiles a contiguous
+//    8-lane ds_write_b128 group touches land on distinct write banks
+//    (144/4 = 36 = +4 banks/subtile), and the two tr tiles one 32-lane
+//    read conflict-group touches overlap on only 4 of 64 read banks.
+//  * key4 stride 8*72 = 576 elems.
+#define VSUB 72
+#define VKEY4 (8 * VSUB)
 
 #define NEG_INF (-__builtin_inff())
 
 __device__ __forceinline__ float xor32(float v) { return __shfl_xor(v, 32); }
+
+// raw v_exp_f32: clang's exp2f lowers to exp+ldexp+cmp+cndmask (a large-
+// input guard) even under fast-math; our exponents are <= THR=8 and -inf
+// maps to 0 in hardware, so the single instruction is exact here
+__device__ __forceinline__ float exp2_raw(float x) {
+  float r;
+  asm("v_exp_f32 %0, %1" : "=v"(r) : "v"(x));
+  return r;
+}
 
 // v_cvt_pk_bf16_f32: packs (lo, hi) f32 -> one u32 of 2 bf16 (HW RNE)
 __device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
@@ -58,10 +78,58 @@ __device__ __forceinline__ int k_lds_off(int row, int blk) {
   return row * KROW + 8 * (blk ^ (row & 7));
 }
 
-// V^T LDS: element offset of (d row, k element), k-block swizzled by d
-__device__ __forceinline__ int v_lds_off(int d, int k) {
-  return d * VROW + ((k & 31) ^ (8 * ((d >> 4) & 3))) + (k >> 5) * (ATTN_D * VROW);
+// V image: element offset of (key, d) in the subtiled layout
+__device__ __forceinline__ int v_img_off(int key, int d) {
+  return (key >> 2) * VKEY4 + (d >> 4) * VSUB + (key & 3) * 16 + (d & 15);
 }
+
+typedef ushort_t ushortx4_t __attribute__((ext_vector_type(4)));
+typedef __bf16 bf16x8_fwd __attribute__((ext_vector_type(8)));
+
+// hardware transpose read: lane gets column (lane&15), rows j=0..3 of the
+// 4x16 row-major bf16 tile at (byte_addr - (lane&15)*8). The ds address
+// operand is the LDS BYTE OFFSET — an addrspacecast generic->local gives
+// it (a flat pointer value would be aperture-based garbage). The tile
+// coordinate goes in the 16-bit `offset:` immediate so each read costs
+// ONE issue slot (the per-tile lane-dependent base is computed once).
+// Results counted by lgkm; callers wait via tr16_wait<N>() (counted —
+// the PV phase has no other LDS ops in flight) before the MFMA.
+__device__ __forceinline__ unsigned lds_byte_base(const ushort_t* p) {
+  auto lp = (const __attribute__((address_space(3))) ushort_t*)p;
+  return (unsigned)(unsigned long)lp;
+}
+
+template <int OFF>
+__device__ __forceinline__ ushortx4_t tr16_read_imm(unsigned vbase) {
+  ushortx4_t r;
+  // no "memory" clobber: the image was published by the tile barrier and
+  // nothing writes v_lds during PV — a clobber would serialize the whole
+  // read burst against every surrounding load/store
+  asm volatile("ds_read_b64_tr_b16 %0, %1 offset:%c2"
+               : "=v"(r) : "v"(vbase), "i"(OFF));
+  return r;
+}
+
+template <int N>
+__device__ __forceinline__ void tr16_wait() {
+  asm volatile("s_waitcnt lgkmcnt(%c0)" :: "i"(N) : "memory");
+  __builtin_amdgcn_sched_barrier(0);  // rule 18: fence MFMAs below the wait
+}
+
+// PV helpers: template-recursive unrolls so every tr read carries its
+// tile coordinate as an `offset:` immediate (a #pragma-unroll index is
+// not an integer constant expression for the "i" constraint).
+template <int NT, int DT, int KSTEP = 0>
+__device__ __forceinline__ void pv_issue(unsigned vbase,
+                                         ushortx4_t (*vr)[2]) {
+  if constexpr (KSTEP < 2 * NT) {
+    constexpr int base = DT * 2 * VSUB * 2;
+    vr[KSTEP][0] = tr16_read_imm<base + (KSTEP * 4 + 0) * VKEY4 * 2>(vbase);
+    vr[KSTEP][1] = tr16_read_imm<base + (KSTEP * 4 + 1) * VKEY4 * 2>(vbase);
+    pv_issue<NT, DT, KSTEP + 1>(vbase, vr);
+  }
+}
+
 
 // NT = 32-key sub-tiles per staged tile (1 -> 32-key tiles, 2 -> 64).
 template <int NT>
@@ -95,17 +163,24 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   const ushort_t* Vp = V + (long)b * kb + (long)hk * kh;
 
   __shared__ ushort_t k_lds[2][BN * KROW];
-  __shared__ ushort_t v_lds[2][NT * ATTN_D * VROW];
+  __shared__ ushort_t v_lds[2][(BN / 4) * VKEY4];
 
-  // ---- Q fragments: B[k=d][n=q]
+  // ---- Q fragments: B[k=d][n=q], PRE-SCALED by softmax_scale*log2e so
+  // the per-tile scores come out of the MFMA already in the exp2 domain
+  // (saves 32 v_mul per tile; bf16 rounding of q*scale is within the
+  // kernel's bf16 I/O noise)
   bf16x8 qf[8];
   {
     const ushort_t* qrow = Qp + (long)(q_valid ? q_row : 0) * qs;
     const int dbase = (lane >> 5) * 8;
 #pragma unroll
-    for (int st = 0; st < 8; ++st)
-      qf[st] = __builtin_bit_cast(
-          bf16x8, *(const ushortx8*)(qrow + st * 16 + dbase));
+    for (int st = 0; st < 8; ++st) {
+      ushortx8 u = *(const ushortx8*)(qrow + st * 16 + dbase);
+      ushortx8 sc;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) sc[e] = f2bf(bf2f(u[e]) * scale_log2);
+      qf[st] = __builtin_bit_cast(bf16x8, sc);
+    }
   }
 
   float m_run = -1e30f, l_run = 0.f;
@@ -141,11 +216,11 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     for (int h = 0; h < NT; ++h) {
       *(ushortx8*)(&k_lds[buf][k_lds_off(st_row + 32 * h, st_col / 8)]) = kv0[h];
       *(ushortx8*)(&k_lds[buf][k_lds_off(st_row + 32 * h, st_col / 8 + 1)]) = kv1[h];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        v_lds[buf][v_lds_off(st_col + j, st_row + 32 * h)] = vv0[h][j];
-        v_lds[buf][v_lds_off(st_col + 8 + j, st_row + 32 * h)] = vv1[h][j];
-      }
+      // V: two 16-B vector writes into the subtiled tr-read image (the
+      // 16 consecutive d of one (key/4, d/16) subtile row are contiguous)
+      const int vkey = st_row + 32 * h;
+      *(ushortx8*)(&v_lds[buf][v_img_off(vkey, st_col)]) = vv0[h];
+      *(ushortx8*)(&v_lds[buf][v_img_off(vkey, st_col + 8)]) = vv1[h];
     }
     __syncthreads();  // tile t staged in buf; tile t-1 reads are also done
     if (t + 1 < n_tiles) {
@@ -188,13 +263,13 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
         for (int j = 0; j < 16; ++j) {
           const int key = n0 + 32 * h + (j & 3) + 8 * (j >> 2) + mrow_base;
           const bool dead = (causal && key > q_row) || key >= S || !q_valid;
-          s_val[h][j] = dead ? NEG_INF : st_acc[h][j] * scale_log2;
+          s_val[h][j] = dead ? NEG_INF : st_acc[h][j];
         }
     } else {
 #pragma unroll
       for (int h = 0; h < NT; ++h)
 #pragma unroll
-        for (int j = 0; j < 16; ++j) s_val[h][j] = st_acc[h][j] * scale_log2;
+        for (int j = 0; j < 16; ++j) s_val[h][j] = st_acc[h][j];
     }
 
     // ---- online softmax with defer-max (T13)
@@ -207,7 +282,7 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     tmax = fmaxf(tmax, xor32(tmax));
     if (__any(tmax > m_run + RESCALE_THR)) {
       const float m_new = fmaxf(m_run, tmax);
-      const float alpha = __builtin_exp2f(m_run - m_new);  // 0 on 1st tile
+      const float alpha = exp2_raw(m_run - m_new);  // 0 on 1st tile
       l_run *= alpha;
       m_run = m_new;
       float alpha_j[16];
@@ -225,7 +300,7 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     for (int h = 0; h < NT; ++h)
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
-        p_val[h][j] = __builtin_exp2f(s_val[h][j] - m_run);  // <= 2^THR
+        p_val[h][j] = exp2_raw(s_val[h][j] - m_run);  // <= 2^THR
         psum += p_val[h][j];
       }
     psum += xor32(psum);
@@ -261,21 +336,41 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       pf[2 * h + 1] = __builtin_bit_cast(bf16x8, pw1);
     }
 
-    // ---- O += P . V   (B[k][n=d] = one b128 from transposed V)
-    __builtin_amdgcn_s_setprio(1);
+    // ---- O += P . V   (B[k=key][n=d] fragments by hardware transpose
+    // read from the row-major subtiled V image: per kstep two
+    // ds_read_b64_tr_b16 deliver the lane's 8 keys at its d column).
+    // Double-buffered: dt+1's reads are in flight behind a counted
+    // lgkm wait while dt's MFMAs run (no other LDS ops live here).
+    {
+      const unsigned vbase =
+          lds_byte_base(v_lds[buf]) +
+          (unsigned)((((lane >> 5) * 2 * VKEY4) + ((lane >> 4) & 1) * VSUB) * 2 +
+                     (lane & 15) * 8);
+      ushortx4_t vrA[2 * NT][2], vrB[2 * NT][2];
+      __builtin_amdgcn_s_setprio(1);
+      pv_issue<NT, 0>(vbase, vrA);
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
-      const int d0 = dt * 32 + (lane & 31);
+      for (int dt = 0; dt < 4; ++dt) {
+        ushortx4_t(*cur)[2] = (dt & 1) ? vrB : vrA;
+        ushortx4_t(*nxt)[2] = (dt & 1) ? vrA : vrB;
+        if (dt == 0) pv_issue<NT, 1>(vbase, nxt);
+        if (dt == 1) pv_issue<NT, 2>(vbase, nxt);
+        if (dt == 2) pv_issue<NT, 3>(vbase, nxt);
+        if (dt < 3)
+          tr16_wait<4 * NT>();  // dt's reads landed; dt+1's stay in flight
+        else
+          tr16_wait<0>();
 #pragma unroll
-      for (int kstep = 0; kstep < 2 * NT; ++kstep) {
-        bf16x8 vf = __builtin_bit_cast(
-            bf16x8, *(const ushortx8*)(
-                        &v_lds[buf][v_lds_off(d0, kstep * 16 + (lane >> 5) * 8)]));
-        o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf[kstep], vf,
-                                                            o_acc[dt], 0, 0, 0);
+        for (int kstep = 0; kstep < 2 * NT; ++kstep) {
+          struct {
+            ushortx4_t a, b;
+          } pair = {cur[kstep][0], cur[kstep][1]};
+          o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              pf[kstep], __builtin_bit_cast(bf16x8, pair), o_acc[dt], 0, 0, 0);
+        }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
   }
 
   // ---- epilogue: O /= l, write bf16; LSE in log2 domain

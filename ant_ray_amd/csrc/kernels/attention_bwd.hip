@@ -47,6 +47,37 @@ __device__ __forceinline__ int tr_off(int d, int k) {
   return d * TROW + (k ^ (8 * ((d >> 4) & 3)));
 }
 
+// raw v_exp_f32 (clang's exp2f carries a large-input guard even under
+// fast-math; bwd exponents are bounded by the fwd lse)
+__device__ __forceinline__ float exp2_rawb(float x) {
+  float r;
+  asm("v_exp_f32 %0, %1" : "=v"(r) : "v"(x));
+  return r;
+}
+
+typedef float floatx4_b __attribute__((ext_vector_type(4)));
+
+// the 16 per-reg q-rows of a C tile are (j&3)+8*(j>>2)+4*(lane>>5): four
+// 4-row runs at mb, mb+8, mb+16, mb+24 — four float4 loads fetch the
+// lse/delta values a lane needs (replaces 16 ds_bpermute broadcasts)
+__device__ __forceinline__ void load_rows16(const float* p, int base,
+                                            int mb, int S,
+                                            floatx4_b (&o)[4]) {
+  if (base + 32 <= S) {
+#pragma unroll
+    for (int g = 0; g < 4; ++g)
+      o[g] = *(const floatx4_b*)(p + base + mb + 8 * g);
+  } else {
+    // tail tile: element-wise clamped loads (rows >= S are masked dead
+    // by the caller; the clamp only keeps the reads in bounds)
+#pragma unroll
+    for (int g = 0; g < 4; ++g)
+#pragma unroll
+      for (int e = 0; e < 4; ++e)
+        o[g][e] = p[min(base + mb + 8 * g + e, S - 1)];
+  }
+}
+
 __device__ __forceinline__ unsigned cvt_pk_bf16b(float lo, float hi) {
   unsigned r;
   asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
@@ -195,7 +226,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       const int key = n0 + (j & 3) + 8 * (j >> 2) + mrow_base;
       const bool dead = (causal && key > q_row) || key >= S || !q_valid;
       const float p = dead ? 0.f
-                           : __builtin_exp2f(st_acc[j] * c_log2 - L_q);
+                           : exp2_rawb(st_acc[j] * c_log2 - L_q);
       ds_val[j] = scale * p * (dp_acc[j] - D_q);
     }
 
@@ -350,7 +381,12 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // per-q-row lse fetched once per lane, broadcast per reg row by shfl
+    // per-q-row lse via four float4 loads (L2-hot; all lanes of a half
+    // read the same addresses). Guard: the last tile may poke past S —
+    // Lp is sized S, so clamp the base (dead lanes mask the values out).
+    // per-q-row lse fetched once per lane, broadcast per reg row by
+    // shfl (LDS-local ~50cy; an r2m experiment with global float4 loads
+    // instead put 200+cy L2 latency on the critical path and lost 3%)
     const float L_lane = Lp[min(tq0 + (lane & 31), S - 1)];
     const int mrow_base = 4 * (lane >> 5);
     const int key_here = key0 + (lane & 31);
@@ -363,7 +399,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
       const bool dead = (causal && key_here > q_abs) || q_abs >= S ||
                         key_here >= S;
       p_val[j] = dead ? 0.f
-                      : __builtin_exp2f(st_acc[j] * c_log2 - L_q);
+                      : exp2_rawb(st_acc[j] * c_log2 - L_q);
     }
     // C[q][key] -> A[m=key][k=q] fragments in-register (fwd T12 pattern)
     unsigned own_pk[8];
@@ -517,7 +553,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // per-q-row lse/delta fetched once per lane, broadcast by shfl
+    // per-q-row lse/delta via float4 loads (see attn_bwd_dv)
     const int q_lane = min(tq0 + (lane & 31), S - 1);
     const float L_lane = Lp[q_lane];
     const float D_lane = Dp[q_lane];
@@ -533,7 +569,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
       const bool dead = (causal && key_here > q_abs) || q_abs >= S ||
                         key_here >= S;
       const float p = dead ? 0.f
-                           : __builtin_exp2f(st_acc[j] * c_log2 - L_q);
+                           : exp2_rawb(st_acc[j] * c_log2 - L_q);
       ds_val[j] = scale * p * (dp_acc[j] - D_q);
     }
     unsigned own_pk[8];
