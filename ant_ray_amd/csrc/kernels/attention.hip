@@ -41,18 +41,6 @@ typedef unsigned int uint32x4 __attribute__((ext_vector_type(4)));
 #define KROW 128      // K LDS row stride (T2 swizzle instead of padding)
 #define RESCALE_THR 8.0f  // T13: defer O-rescale until max grows > 2^8
 
-// V image for the hardware transpose read (T10, ds_read_b64_tr_b16):
-// subtiled [key/4][d/16] of row-major [4 key][16 d] 64-element tiles.
-// Strides are bank-staggered (empirically derived, see tools/probes/
-// tr16_probe.hip for the instruction's lane->element mapping):
-//  * d-subtile stride 72 elems (144 B): the 8 d-subtiles a contiguous
-//    8-lane ds_write_b128 group touches land on distinct write banks
-//    (144/4 = 36 = +4 banks/subtile), and the two tr tiles one 32-lane
-//    read conflict-group touches overlap on only 4 of 64 read banks.
-//  * key4 stride 8*72 = 576 elems.
-#define VSUB 72
-#define VKEY4 (8 * VSUB)
-
 #define NEG_INF (-__builtin_inff())
 
 __device__ __forceinline__ float xor32(float v) { return __shfl_xor(v, 32); }
@@ -78,58 +66,10 @@ __device__ __forceinline__ int k_lds_off(int row, int blk) {
   return row * KROW + 8 * (blk ^ (row & 7));
 }
 
-// V image: element offset of (key, d) in the subtiled layout
-__device__ __forceinline__ int v_img_off(int key, int d) {
-  return (key >> 2) * VKEY4 + (d >> 4) * VSUB + (key & 3) * 16 + (d & 15);
-}
-
-typedef ushort_t ushortx4_t __attribute__((ext_vector_type(4)));
-typedef __bf16 bf16x8_fwd __attribute__((ext_vector_type(8)));
-
-// hardware transpose read: lane gets column (lane&15), rows j=0..3 of the
-// 4x16 row-major bf16 tile at (byte_addr - (lane&15)*8). The ds address
-// operand is the LDS BYTE OFFSET — an addrspacecast generic->local gives
-// it (a flat pointer value would be aperture-based garbage). The tile
-// coordinate goes in the 16-bit `offset:` immediate so each read costs
-// ONE issue slot (the per-tile lane-dependent base is computed once).
-// Results counted by lgkm; callers wait via tr16_wait<N>() (counted —
-// the PV phase has no other LDS ops in flight) before the MFMA.
-__device__ __forceinline__ unsigned lds_byte_base(const ushort_t* p) {
-  auto lp = (const __attribute__((address_space(3))) ushort_t*)p;
-  return (unsigned)(unsigned long)lp;
-}
-
-template <int OFF>
-__device__ __forceinline__ ushortx4_t tr16_read_imm(unsigned vbase) {
-  ushortx4_t r;
-  // no "memory" clobber: the image was published by the tile barrier and
-  // nothing writes v_lds during PV — a clobber would serialize the whole
-  // read burst against every surrounding load/store
-  asm volatile("ds_read_b64_tr_b16 %0, %1 offset:%c2"
-               : "=v"(r) : "v"(vbase), "i"(OFF));
-  return r;
-}
-
-template <int N>
-__device__ __forceinline__ void tr16_wait() {
-  asm volatile("s_waitcnt lgkmcnt(%c0)" :: "i"(N) : "memory");
-  __builtin_amdgcn_sched_barrier(0);  // rule 18: fence MFMAs below the wait
-}
-
-// PV helpers: template-recursive unrolls so every tr read carries its
-// tile coordinate as an `offset:` immediate (a #pragma-unroll index is
-// not an integer constant expression for the "i" constraint).
-template <int NT, int DT, int KSTEP = 0>
-__device__ __forceinline__ void pv_issue(unsigned vbase,
-                                         ushortx4_t (*vr)[2]) {
-  if constexpr (KSTEP < 2 * NT) {
-    constexpr int base = DT * 2 * VSUB * 2;
-    vr[KSTEP][0] = tr16_read_imm<base + (KSTEP * 4 + 0) * VKEY4 * 2>(vbase);
-    vr[KSTEP][1] = tr16_read_imm<base + (KSTEP * 4 + 1) * VKEY4 * 2>(vbase);
-    pv_issue<NT, DT, KSTEP + 1>(vbase, vr);
-  }
-}
-
+#define ATTN_D 128
+#define BLOCK_M 128   // q rows per workgroup (32 per wave)
+#define KROW 128      // K LDS row stride (T2 swizzle instead of padding)
+#define RESCALE_THR 8.0f  // T13: defer O-rescale until max grows > 2^8
 
 // NT = 32-key sub-tiles per staged tile (1 -> 32-key tiles, 2 -> 64).
 template <int NT>
@@ -163,7 +103,7 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
   const ushort_t* Vp = V + (long)b * kb + (long)hk * kh;
 
   __shared__ ushort_t k_lds[2][BN * KROW];
-  __shared__ ushort_t v_lds[2][(BN / 4) * VKEY4];
+  __shared__ ushort_t v_lds[2][(BN / 4) * TRKEY4];
 
   // ---- Q fragments: B[k=d][n=q], PRE-SCALED by softmax_scale*log2e so
   // the per-tile scores come out of the MFMA already in the exp2 domain
@@ -219,8 +159,8 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
       // V: two 16-B vector writes into the subtiled tr-read image (the
       // 16 consecutive d of one (key/4, d/16) subtile row are contiguous)
       const int vkey = st_row + 32 * h;
-      *(ushortx8*)(&v_lds[buf][v_img_off(vkey, st_col)]) = vv0[h];
-      *(ushortx8*)(&v_lds[buf][v_img_off(vkey, st_col + 8)]) = vv1[h];
+      *(ushortx8*)(&v_lds[buf][tr_img_off(vkey, st_col)]) = vv0[h];
+      *(ushortx8*)(&v_lds[buf][tr_img_off(vkey, st_col + 8)]) = vv1[h];
     }
     __syncthreads();  // tile t staged in buf; tile t-1 reads are also done
     if (t + 1 < n_tiles) {
@@ -342,28 +282,25 @@ __global__ __launch_bounds__(256, 2) void attn_fwd_kernel(
     // Double-buffered: dt+1's reads are in flight behind a counted
     // lgkm wait while dt's MFMAs run (no other LDS ops live here).
     {
-      const unsigned vbase =
-          lds_byte_base(v_lds[buf]) +
-          (unsigned)((((lane >> 5) * 2 * VKEY4) + ((lane >> 4) & 1) * VSUB) * 2 +
-                     (lane & 15) * 8);
-      ushortx4_t vrA[2 * NT][2], vrB[2 * NT][2];
+      const unsigned vbase = tr16_lane_base(v_lds[buf], lane);
+      ushortx4_tr vrA[2 * NT][2], vrB[2 * NT][2];
       __builtin_amdgcn_s_setprio(1);
-      pv_issue<NT, 0>(vbase, vrA);
+      tr16_issue_dt<NT, 0>(vbase, vrA);
 #pragma unroll
       for (int dt = 0; dt < 4; ++dt) {
-        ushortx4_t(*cur)[2] = (dt & 1) ? vrB : vrA;
-        ushortx4_t(*nxt)[2] = (dt & 1) ? vrA : vrB;
-        if (dt == 0) pv_issue<NT, 1>(vbase, nxt);
-        if (dt == 1) pv_issue<NT, 2>(vbase, nxt);
-        if (dt == 2) pv_issue<NT, 3>(vbase, nxt);
+        ushortx4_tr(*cur)[2] = (dt & 1) ? vrB : vrA;
+        ushortx4_tr(*nxt)[2] = (dt & 1) ? vrA : vrB;
+        if (dt == 0) tr16_issue_dt<NT, 1>(vbase, nxt);
+        if (dt == 1) tr16_issue_dt<NT, 2>(vbase, nxt);
+        if (dt == 2) tr16_issue_dt<NT, 3>(vbase, nxt);
         if (dt < 3)
-          tr16_wait<4 * NT>();  // dt's reads landed; dt+1's stay in flight
+          tr16_wait_n<4 * NT>();  // dt's reads landed; dt+1's stay in flight
         else
-          tr16_wait<0>();
+          tr16_wait_n<0>();
 #pragma unroll
         for (int kstep = 0; kstep < 2 * NT; ++kstep) {
           struct {
-            ushortx4_t a, b;
+            ushortx4_tr a, b;
           } pair = {cur[kstep][0], cur[kstep][1]};
           o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               pf[kstep], __builtin_bit_cast(bf16x8, pair), o_acc[dt], 0, 0, 0);

@@ -137,7 +137,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 
   __shared__ ushort_t k_rm[32 * KROW];
   __shared__ ushort_t v_rm[32 * KROW];
-  __shared__ ushort_t k_tr[ATTN_D * TROW];
+  __shared__ ushort_t k_img[8 * TRKEY4];  // tr16 image: K^T B-fragments
 
   // Q and dO fragments (B: lane n = q, kdim = d slices)
   bf16x8 qf[8], dof[8];
@@ -185,11 +185,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       *(ushortx8*)(&k_rm[rm_off(st_row, st_col / 8 + 1)]) = k1;
       *(ushortx8*)(&v_rm[rm_off(st_row, st_col / 8)]) = v0;
       *(ushortx8*)(&v_rm[rm_off(st_row, st_col / 8 + 1)]) = v1;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        k_tr[tr_off(st_col + j, st_row)] = k0[j];
-        k_tr[tr_off(st_col + 8 + j, st_row)] = k1[j];
-      }
+      *(ushortx8*)(&k_img[tr_img_off(st_row, st_col)]) = k0;
+      *(ushortx8*)(&k_img[tr_img_off(st_row, st_col + 8)]) = k1;
     }
     __syncthreads();
     if (t + 1 < n_tiles) {
@@ -249,21 +246,37 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       dsf[1] = __builtin_bit_cast(bf16x8, *(uint32x4_t*)w1);
     }
 
-    // dQ += dS . K  (B = K^T from transposed KT)
-    __builtin_amdgcn_s_setprio(1);
+    // dQ += dS . K  (B[k=key][n=d] fragments via ds_read_b64_tr_b16
+    // from the row-major subtiled K image; dt+1's reads pipeline behind
+    // dt's MFMAs — only our tr ops are on lgkm in this phase)
+    {
+      const unsigned kbase = tr16_lane_base(k_img, lane);
+      ushortx4_tr vrA[2][2], vrB[2][2];
+      __builtin_amdgcn_s_setprio(1);
+      tr16_issue_dt<1, 0>(kbase, vrA);
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
-      const int d0 = dt * 32 + (lane & 31);
+      for (int dt = 0; dt < 4; ++dt) {
+        ushortx4_tr(*cur)[2] = (dt & 1) ? vrB : vrA;
+        ushortx4_tr(*nxt)[2] = (dt & 1) ? vrA : vrB;
+        if (dt == 0) tr16_issue_dt<1, 1>(kbase, nxt);
+        if (dt == 1) tr16_issue_dt<1, 2>(kbase, nxt);
+        if (dt == 2) tr16_issue_dt<1, 3>(kbase, nxt);
+        if (dt < 3)
+          tr16_wait_n<4>();
+        else
+          tr16_wait_n<0>();
 #pragma unroll
-      for (int kstep = 0; kstep < 2; ++kstep) {
-        bf16x8 ktf = __builtin_bit_cast(
-            bf16x8, *(const ushortx8*)(
-                        &k_tr[tr_off(d0, kstep * 16 + (lane >> 5) * 8)]));
-        dq_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsf[kstep], ktf,
-                                                             dq_acc[dt], 0, 0, 0);
+        for (int kstep = 0; kstep < 2; ++kstep) {
+          struct {
+            ushortx4_tr a, b;
+          } pair = {cur[kstep][0], cur[kstep][1]};
+          dq_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              dsf[kstep], __builtin_bit_cast(bf16x8, pair), dq_acc[dt],
+              0, 0, 0);
+        }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
   }
 
   // epilogue: write dQ (C: m = q reg-spread, n = d lane)
@@ -310,7 +323,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
   const ushort_t* Kp = K + (long)b * kb + (long)hk * kh;
 
   __shared__ ushort_t q_rm[32 * KROW];
-  __shared__ ushort_t do_tr[ATTN_D * TROW];
+  __shared__ ushort_t do_img[8 * TRKEY4];  // tr16 image: dO^T B-fragments
 
   // own K rows in registers — used as the B operand of S = Q.K^T
   // (B[k=d][n=key]: n = lane-local key, k-slices = d — the same per-lane
@@ -350,11 +363,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
     __syncthreads();
     *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8)]) = qa;
     *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8 + 1)]) = qa2;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      do_tr[tr_off(st_col + j, st_row)] = da[j];
-      do_tr[tr_off(st_col + 8 + j, st_row)] = da2[j];
-    }
+    *(ushortx8*)(&do_img[tr_img_off(st_row, st_col)]) = da;
+    *(ushortx8*)(&do_img[tr_img_off(st_row, st_col + 8)]) = da2;
     __syncthreads();
     if (t + 1 < n_q_tiles) {
       const int qg = min((t + 1) * BLOCK + st_row, S - 1);
@@ -419,20 +429,34 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
       pf[0] = __builtin_bit_cast(bf16x8, *(uint32x4_t*)w0);
       pf[1] = __builtin_bit_cast(bf16x8, *(uint32x4_t*)w1);
     }
-    __builtin_amdgcn_s_setprio(1);
+    {
+      const unsigned dbase = tr16_lane_base(do_img, lane);
+      ushortx4_tr vrA[2][2], vrB[2][2];
+      __builtin_amdgcn_s_setprio(1);
+      tr16_issue_dt<1, 0>(dbase, vrA);
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
-      const int d0 = dt * 32 + (lane & 31);
+      for (int dt = 0; dt < 4; ++dt) {
+        ushortx4_tr(*cur)[2] = (dt & 1) ? vrB : vrA;
+        ushortx4_tr(*nxt)[2] = (dt & 1) ? vrA : vrB;
+        if (dt == 0) tr16_issue_dt<1, 1>(dbase, nxt);
+        if (dt == 1) tr16_issue_dt<1, 2>(dbase, nxt);
+        if (dt == 2) tr16_issue_dt<1, 3>(dbase, nxt);
+        if (dt < 3)
+          tr16_wait_n<4>();
+        else
+          tr16_wait_n<0>();
 #pragma unroll
-      for (int kstep = 0; kstep < 2; ++kstep) {
-        bf16x8 dob = __builtin_bit_cast(
-            bf16x8, *(const ushortx8*)(
-                        &do_tr[tr_off(d0, kstep * 16 + (lane >> 5) * 8)]));
-        dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pf[kstep], dob,
-                                                             dv_acc[dt], 0, 0, 0);
+        for (int kstep = 0; kstep < 2; ++kstep) {
+          struct {
+            ushortx4_tr a, b;
+          } pair = {cur[kstep][0], cur[kstep][1]};
+          dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              pf[kstep], __builtin_bit_cast(bf16x8, pair), dv_acc[dt],
+              0, 0, 0);
+        }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
   }
   }  // q-head group loop
 
@@ -473,7 +497,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
 
   __shared__ ushort_t q_rm[32 * KROW];
   __shared__ ushort_t do_rm[32 * KROW];
-  __shared__ ushort_t q_tr[ATTN_D * TROW];
+  __shared__ ushort_t q_img[8 * TRKEY4];  // tr16 image: Q^T B-fragments
 
   // own K/V rows in registers — B operands of S = Q.K^T and dP = dO.V^T
   // (key lane-local; the same per-lane bytes an A-fragment holds)
@@ -520,11 +544,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
     *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8 + 1)]) = qa2;
     *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8)]) = da;
     *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8 + 1)]) = da2;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      q_tr[tr_off(st_col + j, st_row)] = qa[j];
-      q_tr[tr_off(st_col + 8 + j, st_row)] = qa2[j];
-    }
+    *(ushortx8*)(&q_img[tr_img_off(st_row, st_col)]) = qa;
+    *(ushortx8*)(&q_img[tr_img_off(st_row, st_col + 8)]) = qa2;
     __syncthreads();
     if (t + 1 < n_q_tiles) {
       const int qg = min((t + 1) * BLOCK + st_row, S - 1);
@@ -589,20 +610,34 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
       dsf[0] = __builtin_bit_cast(bf16x8, *(uint32x4_t*)w0);
       dsf[1] = __builtin_bit_cast(bf16x8, *(uint32x4_t*)w1);
     }
-    __builtin_amdgcn_s_setprio(1);
+    {
+      const unsigned qb2 = tr16_lane_base(q_img, lane);
+      ushortx4_tr vrA[2][2], vrB[2][2];
+      __builtin_amdgcn_s_setprio(1);
+      tr16_issue_dt<1, 0>(qb2, vrA);
 #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
-      const int d0 = dt * 32 + (lane & 31);
+      for (int dt = 0; dt < 4; ++dt) {
+        ushortx4_tr(*cur)[2] = (dt & 1) ? vrB : vrA;
+        ushortx4_tr(*nxt)[2] = (dt & 1) ? vrA : vrB;
+        if (dt == 0) tr16_issue_dt<1, 1>(qb2, nxt);
+        if (dt == 1) tr16_issue_dt<1, 2>(qb2, nxt);
+        if (dt == 2) tr16_issue_dt<1, 3>(qb2, nxt);
+        if (dt < 3)
+          tr16_wait_n<4>();
+        else
+          tr16_wait_n<0>();
 #pragma unroll
-      for (int kstep = 0; kstep < 2; ++kstep) {
-        bf16x8 qtb = __builtin_bit_cast(
-            bf16x8, *(const ushortx8*)(
-                        &q_tr[tr_off(d0, kstep * 16 + (lane >> 5) * 8)]));
-        dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dsf[kstep], qtb,
-                                                             dk_acc[dt], 0, 0, 0);
+        for (int kstep = 0; kstep < 2; ++kstep) {
+          struct {
+            ushortx4_tr a, b;
+          } pair = {cur[kstep][0], cur[kstep][1]};
+          dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              dsf[kstep], __builtin_bit_cast(bf16x8, pair), dk_acc[dt],
+              0, 0, 0);
+        }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
   }
   }  // q-head group loop
 
@@ -614,207 +649,6 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt)
       dKh[(long)kr * ATTN_D + dt * 32 + (lane & 31)] = f2bf(dk_acc[dt][j]);
-  }
-}
-
-// ------------------------------------------------- fused dK+dV kernel
-//
-// One workgroup = 128 keys (4 waves x 32), loops q-tiles from the causal
-// diagonal, computing BOTH dV and dK per tile: S^T and dP^T are computed
-// once (the split kernels recompute S^T in each and dP^T in dK — fusing
-// removes one full S^T matmul pass AND one Q/dO staging pass: 8 -> 7
-// matmul units for the whole backward). Register budget is the reason the
-// split existed: dk_acc+dv_acc = 128 VGPRs + two C tiles + fragments.
-// This version fits by (a) staging the NEXT tile's Q/dO prefetch through
-// the two scratch LDS buffers' write path directly (no 16-reg double
-// buffer), (b) sharing one LDS scratch for P then dS sequentially (two
-// lgkm waits), (c) q_tr/do_tr staged from the just-written row-major LDS
-// image instead of held in registers.
-extern "C" __global__ __launch_bounds__(256, 1) void attn_bwd_dkv_kernel(
-    const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
-    const ushort_t* __restrict__ V, const ushort_t* __restrict__ dO,
-    const float* __restrict__ LSE, const float* __restrict__ Delta,
-    ushort_t* __restrict__ dK_out,   // [B,Hk,S,D] direct
-    ushort_t* __restrict__ dV_out,   // [B,Hk,S,D] direct
-    int S, int Hq, int Hk,
-    long qb, long qh, long qs,
-    long kb, long kh, long ks,
-    long ob, long oh, long os,
-    float scale, int causal) {
-  const int lane = threadIdx.x & 63;
-  const int wave = threadIdx.x >> 6;
-  const int k_block = blockIdx.x;
-  const int hk = blockIdx.y;
-  const int b = blockIdx.z;
-  const int group = Hq / Hk;
-  const float c_log2 = scale * 1.4426950408889634f;
-
-  const int key0 = k_block * 128 + wave * 32;
-  const int key_row = key0 + (lane & 31);
-
-  const ushort_t* Kp = K + (long)b * kb + (long)hk * kh;
-  const ushort_t* Vp = V + (long)b * kb + (long)hk * kh;
-
-  __shared__ ushort_t q_rm[32 * KROW];
-  __shared__ ushort_t do_rm[32 * KROW];
-  __shared__ ushort_t q_tr[ATTN_D * TROW];
-  __shared__ ushort_t do_tr[ATTN_D * TROW];
-  __shared__ ushort_t scr[4][32 * SCR_ROW];
-
-  bf16x8 kfr[8], vfr[8];
-  {
-    const long kg = (long)min(key_row, S - 1) * ks;
-    const int dbase = (lane >> 5) * 8;
-#pragma unroll
-    for (int st = 0; st < 8; ++st) {
-      kfr[st] = __builtin_bit_cast(
-          bf16x8, *(const ushortx8*)(Kp + kg + st * 16 + dbase));
-      vfr[st] = __builtin_bit_cast(
-          bf16x8, *(const ushortx8*)(Vp + kg + st * 16 + dbase));
-    }
-  }
-
-  floatx16 dk_acc[4] = {}, dv_acc[4] = {};
-  const int q_start_tile = causal ? (k_block * 128) / BLOCK : 0;
-  const int n_q_tiles = (S + BLOCK - 1) / BLOCK;
-  const int st_row = threadIdx.x >> 3;
-  const int st_col = (threadIdx.x & 7) * 16;
-
-  for (int g = 0; g < group; ++g) {
-    const int hq = hk * group + g;
-    const ushort_t* Qp = Q + (long)b * qb + (long)hq * qh;
-    const ushort_t* dOp = dO + (long)b * ob + (long)hq * oh;
-    const float* Lp = LSE + ((long)b * Hq + hq) * S;
-    const float* Dp = Delta + ((long)b * Hq + hq) * S;
-    // double-buffered q/do tile prefetch (issue-early, write-late: T14)
-    ushortx8 qa, qa2, da, da2;
-    {
-      const int qg = min(q_start_tile * BLOCK + st_row, S - 1);
-      qa = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
-      qa2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
-      da = *(const ushortx8*)(dOp + (long)qg * os + st_col);
-      da2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
-    }
-    for (int t = q_start_tile; t < n_q_tiles; ++t) {
-      const int tq0 = t * BLOCK;
-      __syncthreads();
-      *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8)]) = qa;
-      *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8 + 1)]) = qa2;
-      *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8)]) = da;
-      *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8 + 1)]) = da2;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        q_tr[tr_off(st_col + j, st_row)] = qa[j];
-        q_tr[tr_off(st_col + 8 + j, st_row)] = qa2[j];
-        do_tr[tr_off(st_col + j, st_row)] = da[j];
-        do_tr[tr_off(st_col + 8 + j, st_row)] = da2[j];
-      }
-      __syncthreads();
-      if (t + 1 < n_q_tiles) {
-        const int qg = min((t + 1) * BLOCK + st_row, S - 1);
-        qa = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
-        qa2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
-        da = *(const ushortx8*)(dOp + (long)qg * os + st_col);
-        da2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
-      }
-
-      const int q_here = tq0 + (lane & 31);
-      const bool qv = q_here < S;
-      const float L_q = qv ? Lp[q_here] : 0.f;
-      const float D_q = qv ? Dp[q_here] : 0.f;
-
-      // S^T and dP^T once per tile (the fusion win)
-      floatx16 st_acc = {}, dp_acc = {};
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int stp = 0; stp < 8; ++stp) {
-        const int blk = 2 * stp + (lane >> 5);
-        bf16x8 qfr = __builtin_bit_cast(
-            bf16x8, *(const ushortx8*)(&q_rm[rm_off(lane & 31, blk)]));
-        st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfr[stp], qfr,
-                                                         st_acc, 0, 0, 0);
-        bf16x8 dofr = __builtin_bit_cast(
-            bf16x8, *(const ushortx8*)(&do_rm[rm_off(lane & 31, blk)]));
-        dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfr[stp], dofr,
-                                                         dp_acc, 0, 0, 0);
-      }
-      __builtin_amdgcn_s_setprio(0);
-
-      const int mrow_base = 4 * (lane >> 5);
-      ushort_t* my_scr = (ushort_t*)scr[wave];
-      // write P and dS into one interleaved scratch image: column c holds
-      // P at even 8-blocks, dS at odd (SCR_ROW covers 32 + pad; we use two
-      // row bands: P rows [0,32), dS via a second pass after the dV MFMAs)
-#pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const int key = key0 + (j & 3) + 8 * (j >> 2) + mrow_base;
-        const bool dead = (causal && key > q_here) || key >= S || !qv;
-        const float p = dead ? 0.f : __builtin_exp2f(st_acc[j] * c_log2 - L_q);
-        st_acc[j] = p;  // reuse the C tile: st_acc now holds P
-        const int krow = (j & 3) + 8 * (j >> 2) + mrow_base;
-        my_scr[krow * SCR_ROW + (lane & 31)] = f2bf(p);
-      }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      // dV += P^T . dO
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-        const int d0 = dt * 32 + (lane & 31);
-#pragma unroll
-        for (int kstep = 0; kstep < 2; ++kstep) {
-          bf16x8 pa = __builtin_bit_cast(
-              bf16x8, *(const ushortx8*)(
-                          &my_scr[(lane & 31) * SCR_ROW + kstep * 16 +
-                                  (lane >> 5) * 8]));
-          bf16x8 dob = __builtin_bit_cast(
-              bf16x8, *(const ushortx8*)(
-                          &do_tr[tr_off(d0, kstep * 16 + (lane >> 5) * 8)]));
-          dv_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              pa, dob, dv_acc[dt], 0, 0, 0);
-        }
-      }
-      __builtin_amdgcn_s_setprio(0);
-      // dS = scale * P * (dP - delta) into the same scratch (overwrite)
-#pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const int krow = (j & 3) + 8 * (j >> 2) + mrow_base;
-        my_scr[krow * SCR_ROW + (lane & 31)] =
-            f2bf(scale * st_acc[j] * (dp_acc[j] - D_q));
-      }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      // dK += dS^T . Q
-      __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-        const int d0 = dt * 32 + (lane & 31);
-#pragma unroll
-        for (int kstep = 0; kstep < 2; ++kstep) {
-          bf16x8 dsa = __builtin_bit_cast(
-              bf16x8, *(const ushortx8*)(
-                          &my_scr[(lane & 31) * SCR_ROW + kstep * 16 +
-                                  (lane >> 5) * 8]));
-          bf16x8 qtb = __builtin_bit_cast(
-              bf16x8, *(const ushortx8*)(
-                          &q_tr[tr_off(d0, kstep * 16 + (lane >> 5) * 8)]));
-          dk_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              dsa, qtb, dk_acc[dt], 0, 0, 0);
-        }
-      }
-      __builtin_amdgcn_s_setprio(0);
-    }
-  }  // q-head group loop
-
-  ushort_t* dKh = dK_out + ((long)b * Hk + hk) * (long)S * ATTN_D;
-  ushort_t* dVh = dV_out + ((long)b * Hk + hk) * (long)S * ATTN_D;
-#pragma unroll
-  for (int j = 0; j < 16; ++j) {
-    const int kr = key0 + (j & 3) + 8 * (j >> 2) + 4 * (lane >> 5);
-    if (kr >= S) continue;
-#pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
-      dKh[(long)kr * ATTN_D + dt * 32 + (lane & 31)] = f2bf(dk_acc[dt][j]);
-      dVh[(long)kr * ATTN_D + dt * 32 + (lane & 31)] = f2bf(dv_acc[dt][j]);
-    }
   }
 }
 
@@ -883,24 +717,6 @@ extern "C" void launch_attn_bwd_dkv(const void* q, const void* k,
                                     long oh, long os, float scale,
                                     int causal, void* stream) {
   dim3 grid((S + 127) / 128, Hk, B);
-  // fused dK+dV does one S^T/dP^T per tile (7 matmul units total) but its
-  // 374-reg footprint caps it at 1 wave/SIMD: measured 11.9 ms vs the
-  // split pair's 10.06 ms on the llama3-8b shape (r2c A/B) — the split
-  // pair (8 units, 2 waves/SIMD) stays the default; ANTRAY_BWD_FUSED=1
-  // keeps the fused kernel selectable for future register work.
-  static const bool use_split = [] {
-    const char* e = getenv("ANTRAY_BWD_FUSED");
-    return !(e && e[0] == '1');
-  }();
-  if (!use_split) {
-    hipLaunchKernelGGL(attn_bwd_dkv_kernel, grid, dim3(256), 0,
-                       (hipStream_t)stream, (const ushort_t*)q,
-                       (const ushort_t*)k, (const ushort_t*)v,
-                       (const ushort_t*)dO, lse, delta, (ushort_t*)dkp,
-                       (ushort_t*)dvp, S, Hq, Hk, qb, qh, qs, kb, kh, ks,
-                       ob, oh, os, scale, causal);
-    return;
-  }
   hipLaunchKernelGGL(attn_bwd_dv_kernel, grid, dim3(256), 0,
                      (hipStream_t)stream, (const ushort_t*)q,
                      (const ushort_t*)k, (const ushort_t*)dO, lse,
