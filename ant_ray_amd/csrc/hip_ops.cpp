@@ -28,11 +28,11 @@ void launch_attn_bwd_preprocess(const void*, const void*, float*, long, void*);
 void launch_attn_bwd_dq(const void*, const void*, const void*, const void*,
                         const float*, const float*, void*, int, int, int, int,
                         long, long, long, long, long, long, long, long, long,
-                        float, int, void*);
+                        long, long, long, float, int, void*);
 void launch_attn_bwd_dkv(const void*, const void*, const void*, const void*,
                          const float*, const float*, void*, void*, int, int,
                          int, int, long, long, long, long, long, long, long,
-                         long, long, float, int, void*);
+                         long, long, long, long, long, float, int, void*);
 void launch_attn_bwd_reduce_kv(const void*, const void*, void*, void*, int,
                                int, int, int, void*);
 void launch_bf16_to_f32(const void*, float*, int64_t, hipStream_t);
@@ -133,6 +133,7 @@ void rope_(torch::Tensor q, torch::Tensor k, torch::Tensor cos_t,
                 "k batch stride must fold");
   }
   TORCH_CHECK(cos_t.size(0) >= S && cos_t.size(1) == D / 2, "cos table too small");
+  TORCH_CHECK(D % 16 == 0, "rope: head_dim must be a multiple of 16");
   launch_rope(q.data_ptr(), k.data_ptr(), cos_t.data_ptr<float>(),
               sin_t.data_ptr<float>(), B, S, Hq, Hk, D, q_tok, k_tok,
               backward ? 1 : 0, cur_stream());
@@ -264,7 +265,9 @@ std::tuple<torch::Tensor, torch::Tensor> attn_fwd(torch::Tensor q,
 // (dq, dk, dv) in [B,Hq,S,D]/[B,Hk,S,D] contiguous bf16.
 std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> attn_bwd(
     torch::Tensor dout, torch::Tensor q, torch::Tensor k, torch::Tensor v,
-    torch::Tensor o, torch::Tensor lse, double scale, bool causal) {
+    torch::Tensor o, torch::Tensor lse, double scale, bool causal,
+    c10::optional<torch::Tensor> dq_out, c10::optional<torch::Tensor> dk_out,
+    c10::optional<torch::Tensor> dv_out) {
   TORCH_CHECK(q.dim() == 4 && q.size(3) == 128, "attn_bwd: q [B,Hq,S,128]");
   int B = q.size(0), Hq = q.size(1), S = q.size(2);
   int Hk = k.size(1);
@@ -278,23 +281,36 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> attn_bwd(
   auto delta = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat32));
   launch_attn_bwd_preprocess(dc.data_ptr(), oc.data_ptr(),
                              delta.data_ptr<float>(), rows, (void*)stream);
-  auto dq = torch::empty({B, Hq, S, 128}, q.options());
+  // gradients may land in caller-provided STRIDED tensors (e.g. views of
+  // one fused dqkv buffer — the rope+attention fused backward skips the
+  // autograd slice-scatter entirely); rows must be contiguous
+  auto pick = [&](c10::optional<torch::Tensor>& t, int H) {
+    if (t.has_value()) {
+      TORCH_CHECK(t->dim() == 4 && t->size(1) == H && t->stride(3) == 1,
+                  "grad out must be [B,H,S,128] with contiguous rows");
+      return *t;
+    }
+    return torch::empty({B, H, S, 128}, q.options());
+  };
+  auto dq = pick(dq_out, Hq);
+  auto dk = pick(dk_out, Hk);
+  auto dv = pick(dv_out, Hk);
+  TORCH_CHECK(dk.strides() == dv.strides(), "dk/dv must share layout");
   launch_attn_bwd_dq(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                      dc.data_ptr(), lse.data_ptr<float>(),
                      delta.data_ptr<float>(), dq.data_ptr(), B, S, Hq, Hk,
                      q.stride(0), q.stride(1), q.stride(2), k.stride(0),
                      k.stride(1), k.stride(2), dc.stride(0), dc.stride(1),
-                     dc.stride(2), (float)scale, causal ? 1 : 0,
+                     dc.stride(2), dq.stride(0), dq.stride(1), dq.stride(2),
+                     (float)scale, causal ? 1 : 0,
                      (void*)stream);
-  // dK/dV written directly per KV head (the kernels loop the q-head group)
-  auto dk = torch::empty({B, Hk, S, 128}, q.options());
-  auto dv = torch::empty({B, Hk, S, 128}, q.options());
   launch_attn_bwd_dkv(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                       dc.data_ptr(), lse.data_ptr<float>(),
                       delta.data_ptr<float>(), dk.data_ptr(), dv.data_ptr(),
                       B, S, Hq, Hk, q.stride(0), q.stride(1), q.stride(2),
                       k.stride(0), k.stride(1), k.stride(2), dc.stride(0),
-                      dc.stride(1), dc.stride(2), (float)scale,
+                      dc.stride(1), dc.stride(2), dk.stride(0), dk.stride(1),
+                      dk.stride(2), (float)scale,
                       causal ? 1 : 0, (void*)stream);
   return {dq, dk, dv};
 }
@@ -412,7 +428,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("need_lse") = false);
   m.def("attn_bwd", &attn_bwd, py::arg("dout"), py::arg("q"), py::arg("k"),
         py::arg("v"), py::arg("o"), py::arg("lse"), py::arg("scale"),
-        py::arg("causal") = true);
+        py::arg("causal") = true, py::arg("dq_out") = py::none(),
+        py::arg("dk_out") = py::none(), py::arg("dv_out") = py::none());
   m.def("attn_decode", &attn_decode, py::arg("q"), py::arg("k"), py::arg("v"),
         py::arg("T"), py::arg("scale"), py::arg("lens") = py::none());
   m.def("cast_affine", &cast_affine, py::arg("x"), py::arg("scale"),

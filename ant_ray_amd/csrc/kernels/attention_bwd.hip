@@ -110,11 +110,12 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     const ushort_t* __restrict__ V, const ushort_t* __restrict__ dO,
     const float* __restrict__ LSE,    // [B,Hq,S] log2 domain
     const float* __restrict__ Delta,  // [B,Hq,S]
-    ushort_t* __restrict__ dQ,        // [B,Hq,S,D] contiguous
+    ushort_t* __restrict__ dQ,        // [B,Hq,S,D] via its own strides
     int S, int Hq, int Hk,
     long qb, long qh, long qs,        // Q strides (elements)
     long kb, long kh, long ks,        // K/V strides
-    long ob, long oh, long os,        // dO/dQ (contiguous) strides
+    long ob, long oh, long os,        // dO strides
+    long gb, long gh, long gs,        // dQ strides (rows contiguous)
     float scale, int causal) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -280,14 +281,14 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   }
 
   // epilogue: write dQ (C: m = q reg-spread, n = d lane)
-  ushort_t* dQp = dQ + (long)b * ob + (long)hq * oh;
+  ushort_t* dQp = dQ + (long)b * gb + (long)hq * gh;
 #pragma unroll
   for (int j = 0; j < 16; ++j) {
     const int qr = q0 + (j & 3) + 8 * (j >> 2) + 4 * (lane >> 5);
     if (qr >= S) continue;
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt)
-      dQp[(long)qr * os + dt * 32 + (lane & 31)] = f2bf(dq_acc[dt][j]);
+      dQp[(long)qr * gs + dt * 32 + (lane & 31)] = f2bf(dq_acc[dt][j]);
   }
 }
 
@@ -303,11 +304,12 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
     const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
     const ushort_t* __restrict__ dO,
     const float* __restrict__ LSE,
-    ushort_t* __restrict__ dV_out,   // [B,Hk,S,D] direct (q-heads looped)
+    ushort_t* __restrict__ dV_out,   // [B,Hk,S,D] via its own strides
     int S, int Hq, int Hk,
     long qb, long qh, long qs,
     long kb, long kh, long ks,
-    long ob, long oh, long os,       // dO (contiguous) strides
+    long ob, long oh, long os,       // dO strides
+    long gb, long gh, long gs,       // dV strides (rows contiguous)
     float scale, int causal) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -460,14 +462,14 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
   }
   }  // q-head group loop
 
-  ushort_t* dVh = dV_out + ((long)b * Hk + hk) * (long)S * ATTN_D;
+  ushort_t* dVh = dV_out + (long)b * gb + (long)hk * gh;
 #pragma unroll
   for (int j = 0; j < 16; ++j) {
     const int kr = key0 + (j & 3) + 8 * (j >> 2) + 4 * (lane >> 5);
     if (kr >= S) continue;
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt)
-      dVh[(long)kr * ATTN_D + dt * 32 + (lane & 31)] = f2bf(dv_acc[dt][j]);
+      dVh[(long)kr * gs + dt * 32 + (lane & 31)] = f2bf(dv_acc[dt][j]);
   }
 }
 
@@ -475,11 +477,12 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
     const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
     const ushort_t* __restrict__ V, const ushort_t* __restrict__ dO,
     const float* __restrict__ LSE, const float* __restrict__ Delta,
-    ushort_t* __restrict__ dK_out,   // [B,Hk,S,D] direct (q-heads looped)
+    ushort_t* __restrict__ dK_out,   // [B,Hk,S,D] via its own strides
     int S, int Hq, int Hk,
     long qb, long qh, long qs,
     long kb, long kh, long ks,
-    long ob, long oh, long os,       // dO (contiguous) strides
+    long ob, long oh, long os,       // dO strides
+    long gb, long gh, long gs,       // dK strides (rows contiguous)
     float scale, int causal) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -641,14 +644,14 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
   }
   }  // q-head group loop
 
-  ushort_t* dKh = dK_out + ((long)b * Hk + hk) * (long)S * ATTN_D;
+  ushort_t* dKh = dK_out + (long)b * gb + (long)hk * gh;
 #pragma unroll
   for (int j = 0; j < 16; ++j) {
     const int kr = key0 + (j & 3) + 8 * (j >> 2) + 4 * (lane >> 5);
     if (kr >= S) continue;
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt)
-      dKh[(long)kr * ATTN_D + dt * 32 + (lane & 31)] = f2bf(dk_acc[dt][j]);
+      dKh[(long)kr * gs + dt * 32 + (lane & 31)] = f2bf(dk_acc[dt][j]);
   }
 }
 
@@ -698,14 +701,16 @@ extern "C" void launch_attn_bwd_dq(const void* q, const void* k, const void* v,
                                    const float* delta, void* dq, int B, int S,
                                    int Hq, int Hk, long qb, long qh, long qs,
                                    long kb, long kh, long ks, long ob,
-                                   long oh, long os, float scale,
+                                   long oh, long os, long gb, long gh,
+                                   long gs, float scale,
                                    int causal, void* stream) {
   dim3 grid((S + 127) / 128, Hq, B);
   hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, dim3(256), 0,
                      (hipStream_t)stream, (const ushort_t*)q,
                      (const ushort_t*)k, (const ushort_t*)v,
                      (const ushort_t*)dO, lse, delta, (ushort_t*)dq, S, Hq,
-                     Hk, qb, qh, qs, kb, kh, ks, ob, oh, os, scale, causal);
+                     Hk, qb, qh, qs, kb, kh, ks, ob, oh, os, gb, gh, gs,
+                     scale, causal);
 }
 
 extern "C" void launch_attn_bwd_dkv(const void* q, const void* k,
@@ -714,19 +719,21 @@ extern "C" void launch_attn_bwd_dkv(const void* q, const void* k,
                                     void* dkp, void* dvp, int B, int S,
                                     int Hq, int Hk, long qb, long qh, long qs,
                                     long kb, long kh, long ks, long ob,
-                                    long oh, long os, float scale,
+                                    long oh, long os, long gb, long gh,
+                                    long gs, float scale,
                                     int causal, void* stream) {
   dim3 grid((S + 127) / 128, Hk, B);
   hipLaunchKernelGGL(attn_bwd_dv_kernel, grid, dim3(256), 0,
                      (hipStream_t)stream, (const ushort_t*)q,
                      (const ushort_t*)k, (const ushort_t*)dO, lse,
                      (ushort_t*)dvp, S, Hq, Hk, qb, qh, qs, kb, kh, ks,
-                     ob, oh, os, scale, causal);
+                     ob, oh, os, gb, gh, gs, scale, causal);
   hipLaunchKernelGGL(attn_bwd_dk_kernel, grid, dim3(256), 0,
                      (hipStream_t)stream, (const ushort_t*)q,
                      (const ushort_t*)k, (const ushort_t*)v,
                      (const ushort_t*)dO, lse, delta, (ushort_t*)dkp, S, Hq,
-                     Hk, qb, qh, qs, kb, kh, ks, ob, oh, os, scale, causal);
+                     Hk, qb, qh, qs, kb, kh, ks, ob, oh, os, gb, gh, gs,
+                     scale, causal);
 }
 
 extern "C" void launch_attn_bwd_reduce_kv(const void* dkp, const void* dvp,

@@ -85,6 +85,17 @@ class LlamaAttention(nn.Module):
         Hq, Hk, D = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim
         B, S, _ = y.shape
         qkv = self.wqkv(y)
+        if (cache is None and qkv.is_cuda and D == 128
+                and torch.is_grad_enabled() and qkv.requires_grad
+                and os.environ.get("ANTRAY_FLASH", "1") != "0"):
+            # training hot path: fused RoPE + flash attention with the
+            # backward assembling dqkv in one buffer (no autograd
+            # slice-scatter glue)
+            o = ops.rope_attention(qkv, cos[pos : pos + S],
+                                   sin[pos : pos + S], Hq, Hk, D,
+                                   causal=True)
+            o = o.transpose(1, 2).reshape(B, S, Hq * D)
+            return self.wo(o)
         qkv = ops.rope_qkv(qkv, cos[pos : pos + S], sin[pos : pos + S],
                            Hq, Hk, D)
         q = qkv[..., : Hq * D].view(B, S, Hq, D).transpose(1, 2)

@@ -18,6 +18,7 @@ from ant_ray_amd.ops.functional import (  # noqa: F401
     fused_add_rmsnorm,
     nhwc_to_nchw,
     rmsnorm,
+    rope_attention,
     rope_qkv,
     swiglu,
 )
