@@ -32,7 +32,8 @@ void launch_attn_bwd_dq(const void*, const void*, const void*, const void*,
 void launch_attn_bwd_dkv(const void*, const void*, const void*, const void*,
                          const float*, const float*, void*, void*, int, int,
                          int, int, long, long, long, long, long, long, long,
-                         long, long, long, long, long, float, int, void*);
+                         long, long, long, long, long, float, int, void*,
+                         void*);
 void launch_attn_bwd_reduce_kv(const void*, const void*, void*, void*, int,
                                int, int, int, void*);
 void launch_bf16_to_f32(const void*, float*, int64_t, hipStream_t);
@@ -296,6 +297,21 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> attn_bwd(
   auto dk = pick(dk_out, Hk);
   auto dv = pick(dv_out, Hk);
   TORCH_CHECK(dk.strides() == dv.strides(), "dk/dv must share layout");
+  // the three gradient kernels are independent and each leaves the MFMA
+  // pipe ~65% idle at 2 waves/SIMD: run them CONCURRENTLY on two side
+  // streams so their workgroups interleave on the CUs
+  static hipStream_t side1 = nullptr, side2 = nullptr;
+  static hipEvent_t ev_pre = nullptr, ev_s1 = nullptr, ev_s2 = nullptr;
+  if (side1 == nullptr) {
+    hipStreamCreateWithFlags(&side1, hipStreamNonBlocking);
+    hipStreamCreateWithFlags(&side2, hipStreamNonBlocking);
+    hipEventCreateWithFlags(&ev_pre, hipEventDisableTiming);
+    hipEventCreateWithFlags(&ev_s1, hipEventDisableTiming);
+    hipEventCreateWithFlags(&ev_s2, hipEventDisableTiming);
+  }
+  hipEventRecord(ev_pre, stream);       // delta/lse ready
+  hipStreamWaitEvent(side1, ev_pre, 0);
+  hipStreamWaitEvent(side2, ev_pre, 0);
   launch_attn_bwd_dq(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                      dc.data_ptr(), lse.data_ptr<float>(),
                      delta.data_ptr<float>(), dq.data_ptr(), B, S, Hq, Hk,
@@ -311,8 +327,12 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> attn_bwd(
                       k.stride(0), k.stride(1), k.stride(2), dc.stride(0),
                       dc.stride(1), dc.stride(2), dk.stride(0), dk.stride(1),
                       dk.stride(2), (float)scale,
-                      causal ? 1 : 0, (void*)stream);
-  return {dq, dk, dv};
+                      causal ? 1 : 0, (void*)side1, (void*)side2);
+  hipEventRecord(ev_s1, side1);
+  hipEventRecord(ev_s2, side2);
+  hipStreamWaitEvent(stream, ev_s1, 0);
+  hipStreamWaitEvent(stream, ev_s2, 0);
+    return {dq, dk, dv};
 }
 
 // Flash-decode: one new token per sequence over a bf16 KV cache.

@@ -721,7 +721,8 @@ extern "C" void launch_attn_bwd_dkv(const void* q, const void* k,
                                     long kb, long kh, long ks, long ob,
                                     long oh, long os, long gb, long gh,
                                     long gs, float scale,
-                                    int causal, void* stream) {
+                                    int causal, void* stream,
+                                    void* stream2) {
   dim3 grid((S + 127) / 128, Hk, B);
   hipLaunchKernelGGL(attn_bwd_dv_kernel, grid, dim3(256), 0,
                      (hipStream_t)stream, (const ushort_t*)q,
@@ -729,7 +730,8 @@ extern "C" void launch_attn_bwd_dkv(const void* q, const void* k,
                      (ushort_t*)dvp, S, Hq, Hk, qb, qh, qs, kb, kh, ks,
                      ob, oh, os, gb, gh, gs, scale, causal);
   hipLaunchKernelGGL(attn_bwd_dk_kernel, grid, dim3(256), 0,
-                     (hipStream_t)stream, (const ushort_t*)q,
+                     (hipStream_t)(stream2 ? stream2 : stream),
+                     (const ushort_t*)q,
                      (const ushort_t*)k, (const ushort_t*)v,
                      (const ushort_t*)dO, lse, delta, (ushort_t*)dkp, S, Hq,
                      Hk, qb, qh, qs, kb, kh, ks, ob, oh, os, gb, gh, gs,
