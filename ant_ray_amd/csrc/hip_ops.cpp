@@ -24,7 +24,9 @@ void launch_bf16_scale(void*, float, int64_t, hipStream_t);
 void launch_attn_fwd(const void*, const void*, const void*, void*, float*,
                      int, int, int, int, long, long, long, long, long, long,
                      long, long, long, float, int, void*);
-void launch_attn_bwd_preprocess(const void*, const void*, float*, long, void*);
+void launch_attn_bwd_preprocess(const void*, const void*, float*, long, int,
+                                int, long, long, long, long, long, long,
+                                void*);
 void launch_attn_bwd_dq(const void*, const void*, const void*, const void*,
                         const float*, const float*, void*, int, int, int, int,
                         long, long, long, long, long, long, long, long, long,
@@ -242,8 +244,10 @@ std::tuple<torch::Tensor, torch::Tensor> attn_fwd(torch::Tensor q,
               "innermost dim must be contiguous");
   TORCH_CHECK(k.strides() == v.strides() && k.sizes() == v.sizes(),
               "k and v must share layout");
-  auto o = torch::empty({B, Hq, S, D},
-                        q.options().memory_format(torch::MemoryFormat::Contiguous));
+  // o in [B, S, Hq, D] memory layout (presented as a [B,Hq,S,D] strided
+  // view): the model consumes o as [B, S, Hq*D] without a transpose copy
+  auto o = torch::empty({B, S, Hq, D}, q.options())
+               .permute({0, 2, 1, 3});
   torch::Tensor lse;
   float* lse_ptr = nullptr;
   if (need_lse) {
@@ -275,13 +279,18 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> attn_bwd(
   TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1,
               "innermost dim must be contiguous");
   TORCH_CHECK(k.strides() == v.strides(), "k/v must share layout");
-  auto dc = dout.contiguous();
-  auto oc = o.contiguous();
+  // dout/o may be [B,S,H,D]-layout strided views (independently); the
+  // kernels read rows through strides — no contiguous copies
+  auto dc = dout.stride(3) == 1 ? dout : dout.contiguous();
+  TORCH_CHECK(o.stride(3) == 1, "o rows must be contiguous");
   auto stream = cur_stream();
   long rows = (long)B * Hq * S;
   auto delta = torch::empty({B, Hq, S}, q.options().dtype(torch::kFloat32));
-  launch_attn_bwd_preprocess(dc.data_ptr(), oc.data_ptr(),
-                             delta.data_ptr<float>(), rows, (void*)stream);
+  launch_attn_bwd_preprocess(dc.data_ptr(), o.data_ptr(),
+                             delta.data_ptr<float>(), rows, Hq, S,
+                             dc.stride(0), dc.stride(1), dc.stride(2),
+                             o.stride(0), o.stride(1), o.stride(2),
+                             (void*)stream);
   // gradients may land in caller-provided STRIDED tensors (e.g. views of
   // one fused dqkv buffer — the rope+attention fused backward skips the
   // autograd slice-scatter entirely); rows must be contiguous

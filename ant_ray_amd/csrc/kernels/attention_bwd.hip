@@ -88,13 +88,21 @@ __device__ __forceinline__ unsigned cvt_pk_bf16b(float lo, float hi) {
 
 extern "C" __global__ void attn_bwd_preprocess_kernel(
     const ushort_t* __restrict__ dO, const ushort_t* __restrict__ O,
-    float* __restrict__ delta, long rows) {
-  // one wave per 4 rows: 16 lanes per row, each lane 8 elems
+    float* __restrict__ delta, long rows, int Hq, int S,
+    long dob, long doh, long dos, long ob, long oh, long os) {
+  // one wave per 4 rows: 16 lanes per row, each lane 8 elems. dO and O
+  // may each be STRIDED ([B,S,H,D]-layout outputs); delta stays [B,Hq,S]
+  // row-major.
   const long row = (long)blockIdx.x * (blockDim.x >> 4) + (threadIdx.x >> 4);
   if (row >= rows) return;
   const int sub = threadIdx.x & 15;
-  const ushortx8 a = *(const ushortx8*)(dO + row * ATTN_D + sub * 8);
-  const ushortx8 b = *(const ushortx8*)(O + row * ATTN_D + sub * 8);
+  const long s = row % S;
+  const long h = (row / S) % Hq;
+  const long b2 = row / ((long)S * Hq);
+  const ushortx8 a =
+      *(const ushortx8*)(dO + b2 * dob + h * doh + s * dos + sub * 8);
+  const ushortx8 b =
+      *(const ushortx8*)(O + b2 * ob + h * oh + s * os + sub * 8);
   float acc = 0.f;
 #pragma unroll
   for (int i = 0; i < 8; ++i) acc += bf2f(a[i]) * bf2f(b[i]);
@@ -686,14 +694,17 @@ extern "C" __global__ void attn_bwd_reduce_kv_kernel(
 // ---------------------------------------------------------------- hosts
 
 extern "C" void launch_attn_bwd_preprocess(const void* dO, const void* o,
-                                           float* delta, long rows,
-                                           void* stream) {
+                                           float* delta, long rows, int Hq,
+                                           int S, long dob, long doh,
+                                           long dos, long ob, long oh,
+                                           long os, void* stream) {
   const int waves_per_block = 4;
   const long rows_per_block = waves_per_block * 4;
   dim3 grid((rows + rows_per_block - 1) / rows_per_block);
   hipLaunchKernelGGL(attn_bwd_preprocess_kernel, grid, dim3(256), 0,
                      (hipStream_t)stream, (const ushort_t*)dO,
-                     (const ushort_t*)o, delta, rows);
+                     (const ushort_t*)o, delta, rows, Hq, S, dob, doh, dos,
+                     ob, oh, os);
 }
 
 extern "C" void launch_attn_bwd_dq(const void* q, const void* k, const void* v,

@@ -136,19 +136,20 @@ rmsnorm_bwd_kernel(const ushort_t* __restrict__ dy, const ushort_t* __restrict__
   }
 }
 
-// Second stage: dw[H] = sum over G partial rows (column-parallel).
+// Second stage: dw[H] = sum over G partial rows. Parallel over BOTH h
+// and G-slices (H=4096 columns alone gave only 4 workgroups — measured
+// 63 GB/s; slicing G across blockIdx.y with one atomicAdd per column
+// per slice fills the chip). dw must be zeroed before launch.
 extern "C" __global__ void __launch_bounds__(256)
 dw_reduce_kernel(const float* __restrict__ partials, float* __restrict__ dw,
-                 int G, int H) {
-  int h = (blockIdx.x * blockDim.x + threadIdx.x) * 4;
+                 int G, int H, int g_per_slice) {
+  const int h = blockIdx.x * blockDim.x + threadIdx.x;
   if (h >= H) return;
-  floatx4 acc = {0.f, 0.f, 0.f, 0.f};
-  for (int g = 0; g < G; ++g) {
-    floatx4 v = *reinterpret_cast<const floatx4*>(partials + (int64_t)g * H + h);
-#pragma unroll
-    for (int i = 0; i < 4; ++i) acc[i] += v[i];
-  }
-  *reinterpret_cast<floatx4*>(dw + h) = acc;
+  const int g0 = blockIdx.y * g_per_slice;
+  const int g1 = min(g0 + g_per_slice, G);
+  float acc = 0.f;
+  for (int g = g0; g < g1; ++g) acc += partials[(int64_t)g * H + h];
+  atomicAdd(&dw[h], acc);
 }
 
 extern "C" void launch_rmsnorm_fwd(const void* x, const void* w, void* y,
@@ -182,7 +183,10 @@ extern "C" void launch_rmsnorm_bwd(const void* dy, const void* r, const void* w,
     default: LAUNCH_BWD(4); break;
   }
 #undef LAUNCH_BWD
-  int rgrid = (H / 4 + 255) / 256;
-  hipLaunchKernelGGL(dw_reduce_kernel, dim3(rgrid), dim3(256), 0, s,
-                     dw_partials, dw, grid, H);
+  int rgrid = (H + 255) / 256;
+  int slices = 16;
+  int g_per_slice = (grid + slices - 1) / slices;
+  hipMemsetAsync(dw, 0, (size_t)H * sizeof(float), s);
+  hipLaunchKernelGGL(dw_reduce_kernel, dim3(rgrid, slices), dim3(256), 0, s,
+                     dw_partials, dw, grid, H, g_per_slice);
 }
