@@ -267,3 +267,75 @@ def test_autoscaler_idle_downscale():
         assert provider.non_terminated_nodes().get("worker", 0) == 0
     finally:
         c.shutdown()
+
+
+def test_autoscaler_bin_packing_scores():
+    """Utilization scoring + GPU avoidance + strict-spread (parity:
+    reference resource_demand_scheduler.py _utilization_score)."""
+    from ant_ray_amd.autoscaler import NodeTypeConfig, ResourceDemandScheduler
+
+    types = {
+        "cpu4": NodeTypeConfig("cpu4", {"CPU": 4}, max_workers=10),
+        "cpu16": NodeTypeConfig("cpu16", {"CPU": 16}, max_workers=10),
+        "gpu": NodeTypeConfig("gpu", {"CPU": 8, "GPU": 1}, max_workers=10),
+    }
+    sched = ResourceDemandScheduler(types)
+    # CPU-only demand must NOT land on the GPU node type
+    out = sched.get_nodes_to_launch([{"CPU": 4}], {})
+    assert out == {"cpu4": 1}, out
+    # 16 1-CPU demands bin-pack onto ONE cpu16 (utilization beats many cpu4s)
+    out = sched.get_nodes_to_launch([{"CPU": 1}] * 16, {})
+    assert sum(out.values()) <= 4 and "gpu" not in out, out
+    # GPU demand picks the gpu type
+    out = sched.get_nodes_to_launch([{"GPU": 1}], {})
+    assert out == {"gpu": 1}, out
+    # existing capacity absorbs demand first
+    out = sched.get_nodes_to_launch([{"CPU": 2}], {"cpu4": 1})
+    assert out == {}, out
+    # strict-spread: 3 bundles need 3 DISTINCT nodes even though one
+    # cpu16 could fit all
+    out = sched.get_nodes_to_launch(
+        [], {}, strict_spread=[[{"CPU": 2}, {"CPU": 2}, {"CPU": 2}]])
+    assert sum(out.values()) == 3, out
+    # global max_workers cap
+    capped = ResourceDemandScheduler(types, max_workers=2)
+    out = capped.get_nodes_to_launch([{"CPU": 4}] * 5, {})
+    assert sum(out.values()) == 2, out
+
+
+def test_instance_manager_lifecycle():
+    from ant_ray_amd.autoscaler import (
+        IM_RAY_RUNNING,
+        IM_TERMINATED,
+        InstanceManager,
+        NodeTypeConfig,
+    )
+
+    class FakeProvider:
+        def __init__(self):
+            self.created = []
+            self.fail_next = False
+
+        def create_node(self, cfg):
+            if self.fail_next:
+                self.fail_next = False
+                raise RuntimeError("cloud error")
+            self.created.append(cfg.name)
+            return f"node-{len(self.created)}"
+
+        def terminate_node(self, t):
+            self.created.remove(t)
+
+    types = {"w": NodeTypeConfig("w", {"CPU": 4})}
+    prov = FakeProvider()
+    im = InstanceManager(prov, types)
+    a = im.queue("w")
+    im.reconcile()
+    assert a.status == IM_RAY_RUNNING
+    prov.fail_next = True
+    b = im.queue("w")
+    im.reconcile()
+    assert b.status == IM_TERMINATED  # failed launch is GC'd
+    assert im.running() == {"w": 1}
+    assert im.terminate_one("w")
+    assert im.running() == {}
