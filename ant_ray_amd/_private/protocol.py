@@ -27,8 +27,21 @@ import msgpack
 
 logger = logging.getLogger(__name__)
 
-REQUEST, REPLY, ERROR, NOTIFY = 0, 1, 2, 3
+REQUEST, REPLY, ERROR, NOTIFY, HELLO, HELLO_NAK = 0, 1, 2, 3, 4, 5
 _LEN = struct.Struct("<I")
+
+# Wire-protocol version (parity: the reference's proto-schema'd messages
+# carry versioned service definitions; here every connection opens with a
+# HELLO frame [4, 0, {"v": version, "token": token}] and the server
+# rejects mismatched versions or bad auth tokens with HELLO_NAK + close —
+# token auth parity: reference src/ray/rpc/authentication/).
+PROTOCOL_VERSION = 1
+
+
+def auth_token():
+    """Cluster auth token (None = auth disabled). Set RAY_AUTH_TOKEN on
+    the head; spawned components inherit it through the environment."""
+    return os.environ.get("RAY_AUTH_TOKEN") or None
 
 # Handlers receive (connection, method, payload) and return the reply payload.
 Handler = Callable[["Connection", str, Any], Awaitable[Any]]
@@ -139,6 +152,26 @@ class Connection:
                     asyncio.get_running_loop().create_task(
                         self._handle_notify(msg[2], msg[3])
                     )
+                elif kind == HELLO:
+                    info = msg[2] or {}
+                    if info.get("v") != PROTOCOL_VERSION:
+                        await self.send([HELLO_NAK, 0,
+                                         f"protocol version mismatch: peer "
+                                         f"{info.get('v')} != "
+                                         f"{PROTOCOL_VERSION}"])
+                        break
+                    tok = auth_token()
+                    if tok is not None and info.get("token") != tok:
+                        await self.send([HELLO_NAK, 0,
+                                         "authentication failed: bad or "
+                                         "missing RAY_AUTH_TOKEN"])
+                        break
+                    self.session["hello"] = info
+                elif kind == HELLO_NAK:
+                    self._nak_reason = msg[2]
+                    logger.error("connection %s rejected by peer: %s",
+                                 self.name, msg[2])
+                    break
         except (
             asyncio.IncompleteReadError,
             ConnectionResetError,
@@ -194,9 +227,11 @@ class Connection:
         if self._closed:
             return
         self._closed = True
+        reason = getattr(self, "_nak_reason", None)
         for fut in self._pending.values():
             if not fut.done():
-                fut.set_exception(ConnectionLost(self.name))
+                fut.set_exception(RpcError(reason) if reason
+                                  else ConnectionLost(self.name))
                 # abandoned awaiters (cancelled mid-call) are expected on a
                 # lost conn; mark retrieved so GC doesn't log a warning
                 fut.exception()
@@ -230,6 +265,8 @@ async def connect(
         )
     conn = Connection(reader, writer, handler, name=name)
     conn.start()
+    await conn.send([HELLO, 0, {"v": PROTOCOL_VERSION,
+                                "token": auth_token()}])
     return conn
 
 

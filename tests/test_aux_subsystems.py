@@ -339,3 +339,54 @@ def test_instance_manager_lifecycle():
     assert im.running() == {"w": 1}
     assert im.terminate_one("w")
     assert im.running() == {}
+
+
+def test_protocol_version_and_auth(monkeypatch):
+    """HELLO handshake: version mismatch and bad tokens are rejected with
+    a reasoned NAK; matching token accepted (parity: reference
+    rpc/authentication token auth + schema'd protocol)."""
+    import asyncio
+
+    from ant_ray_amd._private import protocol
+
+    async def run():
+        async def handler(conn, method, payload):
+            return {"echo": payload}
+
+        server, port = await protocol.serve(handler, port=0)
+        # 1) happy path, auth disabled
+        c = await protocol.connect(("127.0.0.1", port), handler)
+        assert (await c.call("ping", {"x": 1}, timeout=5)) == {"echo": {"x": 1}}
+        await c.close()
+        import msgpack
+        import struct as _struct
+
+        async def raw_hello(info):
+            """Hand-roll a HELLO and return the server's first frame."""
+            reader, writer = await asyncio.open_connection("127.0.0.1", port)
+            raw = msgpack.packb([protocol.HELLO, 0, info], use_bin_type=True)
+            writer.write(_struct.pack("<I", len(raw)) + raw)
+            await writer.drain()
+            hdr = await asyncio.wait_for(reader.readexactly(4), timeout=5)
+            (n,) = _struct.unpack("<I", hdr)
+            msg = msgpack.unpackb(await reader.readexactly(n), raw=False)
+            writer.close()
+            return msg
+
+        # 2) auth required: wrong token rejected with a reasoned NAK
+        monkeypatch.setenv("RAY_AUTH_TOKEN", "s3cret")
+        msg = await raw_hello({"v": protocol.PROTOCOL_VERSION,
+                               "token": "WRONG"})
+        assert msg[0] == protocol.HELLO_NAK and "authentication" in msg[2]
+        # 3) correct token accepted (both ends read the env)
+        c3 = await protocol.connect(("127.0.0.1", port), handler)
+        assert (await c3.call("ping", {}, timeout=5)) == {"echo": {}}
+        await c3.close()
+        monkeypatch.delenv("RAY_AUTH_TOKEN")
+        # 4) version mismatch rejected
+        msg = await raw_hello({"v": protocol.PROTOCOL_VERSION + 7,
+                               "token": None})
+        assert msg[0] == protocol.HELLO_NAK and "version" in msg[2]
+        server.close()
+
+    asyncio.new_event_loop().run_until_complete(run())
