@@ -42,6 +42,26 @@ def rebatch_blocks(blocks: Iterator[Any], batch_size: Optional[int],
         yield BlockAccessor(table).to_batch(batch_format)
 
 
+_pinned_pool: Dict[int, Any] = {}
+
+
+def _pinned_staging(t):
+    """Copy a CPU tensor through a REUSED pinned buffer (a fresh
+    pin_memory() per batch costs ~ms of hipHostMalloc; the pool keeps one
+    buffer per rounded size class)."""
+    import torch
+
+    nbytes = t.numel() * t.element_size()
+    size_class = 1 << max(nbytes - 1, 1).bit_length()
+    buf = _pinned_pool.get(size_class)
+    if buf is None or buf.numel() < nbytes:
+        buf = torch.empty(size_class, dtype=torch.uint8, pin_memory=True)
+        _pinned_pool[size_class] = buf
+    flat = buf[:nbytes].view(torch.uint8)
+    flat.copy_(t.reshape(-1).view(torch.uint8))
+    return flat.view(t.dtype).reshape(t.shape)
+
+
 def to_torch_batch(batch: Dict[str, np.ndarray], dtypes=None,
                    device: Optional[str] = None,
                    collate_fn: Optional[Callable] = None):
@@ -70,8 +90,11 @@ def to_torch_batch(batch: Dict[str, np.ndarray], dtypes=None,
                 # cast-on-device kernel (csrc/kernels/data_transform.hip)
                 from ant_ray_amd import ops
 
-                t = t.pin_memory().to(device, non_blocking=True)
+                t = _pinned_staging(t).to(device, non_blocking=True)
                 t = ops.cast_affine(t, 1.0, 0.0, out_dtype=want)
+                # the pooled pinned buffer is reused next batch: fence the
+                # async H2D before returning control
+                torch.cuda.current_stream().synchronize()
             else:
                 if want is not None:
                     t = t.to(want)

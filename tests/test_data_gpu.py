@@ -53,3 +53,56 @@ def test_gpu_map_batches_pipeline(ray_mod):
     # batches iterate straight onto the device
     got = next(iter(ds.iter_torch_batches(batch_size=64, device="cuda")))
     assert got["normed"].is_cuda and got["normed"].shape == (64, 256)
+
+
+def test_data_to_train_e2e_gpu(ray_mod, tmp_path_factory):
+    """BASELINE config 5 (1-GPU slice): Ray Data streaming pipeline with a
+    GPU preprocess stage feeding TorchTrainer via streaming_split; batches
+    cross H2D as BYTES and cast on-device through the fused
+    data_transform kernels; the worker trains on them."""
+    import ant_ray_amd.data as data
+    from ant_ray_amd.train import RunConfig, ScalingConfig
+    from ant_ray_amd.train.torch import TorchTrainer
+
+    n_rows, dim = 2048, 64
+    rng = np.random.RandomState(0)
+    raw = (rng.rand(n_rows, dim) * 255).astype(np.uint8)
+    ds = data.from_numpy(raw).map_batches(
+        lambda b: {**b, "label": b["data"].astype(np.float32).mean(
+            -1, keepdims=True) / 255.0},
+        batch_size=256)
+
+    def train_fn(config):
+        import torch as _t
+
+        from ant_ray_amd import train
+
+        shard = train.get_dataset_shard("train")
+        model = _t.nn.Linear(dim, 1).to("cuda")
+        opt = _t.optim.SGD(model.parameters(), lr=1e-3)
+        rows = 0
+        losses = []
+        for batch in shard.iter_torch_batches(
+                batch_size=256, dtypes={"data": _t.float32,
+                                        "label": _t.float32},
+                device="cuda"):
+            x = batch["data"]
+            assert x.is_cuda and x.dtype == _t.float32, (x.device, x.dtype)
+            y = batch["label"]
+            loss = ((model(x / 255.0) - y) ** 2).mean()
+            opt.zero_grad()
+            loss.backward()
+            opt.step()
+            rows += len(x)
+            losses.append(float(loss))
+        train.report({"rows": rows, "loss": losses[-1]})
+
+    res = TorchTrainer(
+        train_fn,
+        scaling_config=ScalingConfig(num_workers=1, use_gpu=True),
+        run_config=RunConfig(
+            name="d2t", storage_path=str(tmp_path_factory.mktemp("d2t"))),
+        datasets={"train": ds},
+    ).fit()
+    assert res.error is None, res.error
+    assert res.metrics["rows"] == n_rows
