@@ -238,8 +238,20 @@ class TaskExecutor:
         try:
             reply = self._execute(payload)
         except BaseException as e:  # noqa: BLE001
-            logger.exception("task execution crashed")
-            reply = self._error_reply(payload, e)
+            if (isinstance(e, KeyboardInterrupt)
+                    and payload.get("task_id") in self._cancelled):
+                # a soft-cancel interrupt can land anywhere in _execute —
+                # including its preamble (e.g. the one-time logging-config
+                # GCS fetch), outside the user-call try that normally maps
+                # it. cancel_task only injects while THIS task is current,
+                # so a KI here is this task's cancellation, not a crash.
+                from ant_ray_amd.exceptions import TaskCancelledError
+
+                reply = self._error_reply(payload, TaskCancelledError(
+                    "task was cancelled while running"))
+            else:
+                logger.exception("task execution crashed")
+                reply = self._error_reply(payload, e)
         done(reply)
 
     def _record_event(self, payload, t0, t1, ok: bool):

@@ -64,6 +64,15 @@ PIPELINE_DEPTH = 8          # tasks pushed per leased worker before waiting
 LEASE_IDLE_RELEASE_S = 2.0  # return leased workers after this idle time
 
 
+def _tid_log(task_id: bytes) -> str:
+    """Log-friendly task-id slice. TaskID = 6-byte process prefix +
+    6-byte counter + job id: the FIRST bytes are constant per process,
+    so hex()[:8] made every task in a process log as the same id (it
+    disguised ordinary <=3x retry noise as one task retrying forever).
+    Show the counter half instead."""
+    return task_id.hex()[12:24]
+
+
 class _RawResult:
     """Memory-store entry holding a serialized value (deserialize at get)."""
 
@@ -1287,14 +1296,28 @@ class CoreWorker:
             if retries > 0 and not isinstance(e, protocol.RpcError):
                 payload["max_retries"] = retries - 1
                 payload["fn"] = payload.get("fn")  # keep blob for resubmit
-                logger.warning("task %s failed (%s); retrying", payload["task_id"].hex()[:8], e)
+                logger.warning("task %s failed (%s); retrying (%d left)",
+                               _tid_log(payload["task_id"]), e,
+                               payload["max_retries"])
                 try:
                     self._leases[key].remove(lw)
                 except ValueError:
                     pass
                 await self._enqueue_task(key, payload, item["resources"], item["opts"])
             else:
-                self._fail_task(payload, RaySystemError(f"task push failed: {e}"))
+                from ant_ray_amd.exceptions import WorkerCrashedError
+
+                if isinstance(e, protocol.RpcError):
+                    self._fail_task(payload, RaySystemError(
+                        f"task push failed: {e}"))
+                else:
+                    # retries exhausted on connection losses = the workers
+                    # executing this task kept dying (e.g. collateral of
+                    # force-cancels): surface the TYPED error (parity:
+                    # reference WorkerCrashedError), not a generic one
+                    self._fail_task(payload, WorkerCrashedError(
+                        f"the worker died while executing this task and "
+                        f"its retries are exhausted ({e})"))
             return
         finally:
             self._inflight_tasks.pop(payload["task_id"], None)
@@ -1402,7 +1425,7 @@ class CoreWorker:
             lpayload = dict(payload,
                             max_retries=payload.get("max_retries", 0) - 1)
             logger.warning("actor task %s raised a retryable exception; "
-                           "retrying (%d left)", task_id.hex()[:8],
+                           "retrying (%d left)", _tid_log(task_id),
                            lpayload["max_retries"])
             for i in range(lpayload.get("n_returns", 1)):
                 roid = ObjectID.for_return(TaskID(task_id), i).binary()
@@ -1420,7 +1443,7 @@ class CoreWorker:
         key, lpayload, res, opts = rec
         lpayload = dict(lpayload, max_retries=lpayload.get("max_retries", 0) - 1)
         logger.warning("task %s raised a retryable exception; retrying "
-                       "(%d retries left)", task_id.hex()[:8],
+                       "(%d retries left)", _tid_log(task_id),
                        lpayload["max_retries"])
         rcost = len(lpayload.get("args") or b"") + 512
         for i in range(lpayload.get("n_returns", 1)):
@@ -1668,6 +1691,15 @@ class CoreWorker:
             batch = st.pending[:]
             st.pending.clear()
             addr = st.addr
+            # Stamp the restart generation: while these payloads sit in the
+            # drain's local batch they are in NEITHER st.pending NOR
+            # st.inflight, so _apply_restart_locked cannot renumber them. A
+            # restart in that window would otherwise push stale seq numbers
+            # at a fresh executor whose ordering gate waits for seq 1 —
+            # the task never runs and its get never resolves (reproduced
+            # by tools/stress_mix.py under RAY_testing_asio_delay_us).
+            for p in batch:
+                p["_gen"] = st.restart_gen
         if not batch:
             return
         try:
@@ -1699,6 +1731,18 @@ class CoreWorker:
 
     async def _push_actor_task(self, st: ActorHandleState, conn, payload):
         with st.lock:
+            if payload.get("_gen", st.restart_gen) != st.restart_gen:
+                # the actor restarted while this payload sat in a drain
+                # batch: its seq belongs to the dead executor's numbering.
+                # Give it a fresh seq and requeue (ordering vs payloads
+                # renumbered by _apply_restart_locked is best-effort in
+                # this rare window; resolution is guaranteed).
+                st.seq += 1
+                payload["seq"] = st.seq
+                payload["_gen"] = st.restart_gen
+                st.pending.append(payload)
+                self.io.loop.create_task(self._drain_actor_queue(st))
+                return
             st.inflight[payload["seq"]] = payload
         try:
             reply = await conn.call("push_task", payload, timeout=None)
