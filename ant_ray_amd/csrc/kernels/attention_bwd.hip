@@ -308,7 +308,10 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 // and dP^T. +25% MFMA vs fused, but 2 waves/SIMD instead of 1 hides the
 // LDS/HBM latency that dominated the fused kernel (PMC: 37% WAIT_ANY).
 
-extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
+// NQ = 32-row q sub-tiles staged per barrier interval (NQ=2 halves the
+// __syncthreads count per q row, same win as the forward kernel's NT).
+template <int NQ>
+__global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
     const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
     const ushort_t* __restrict__ dO,
     const float* __restrict__ LSE,
@@ -332,8 +335,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
 
   const ushort_t* Kp = K + (long)b * kb + (long)hk * kh;
 
-  __shared__ ushort_t q_rm[32 * KROW];
-  __shared__ ushort_t do_img[8 * TRKEY4];  // tr16 image: dO^T B-fragments
+  __shared__ ushort_t q_rm[NQ * 32 * KROW];
+  __shared__ ushort_t do_img[NQ * 8 * TRKEY4];  // tr16 image: dO^T B-frags
 
   // own K rows in registers — used as the B operand of S = Q.K^T
   // (B[k=d][n=key]: n = lane-local key, k-slices = d — the same per-lane
@@ -349,8 +352,11 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
   }
 
   floatx16 dv_acc[4] = {};
-  const int q_start_tile = causal ? (k_block * 128) / BLOCK : 0;
-  const int n_q_tiles = (S + BLOCK - 1) / BLOCK;
+  constexpr int QB = 32 * NQ;       // q rows per staged tile
+  // 128 is a multiple of QB for NQ in {1,2}, so the causal start tile is
+  // exact (no dead half-tiles at the diagonal)
+  const int q_start_tile = causal ? (k_block * 128) / QB : 0;
+  const int n_q_tiles = (S + QB - 1) / QB;
   const int st_row = threadIdx.x >> 3;
   const int st_col = (threadIdx.x & 7) * 16;
 
@@ -359,31 +365,42 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
   const ushort_t* Qp = Q + (long)b * qb + (long)hq * qh;
   const ushort_t* dOp = dO + (long)b * ob + (long)hq * oh;
   const float* Lp = LSE + ((long)b * Hq + hq) * S;
-  ushortx8 qa, qa2, da, da2;
-  {
-    const int qg = min(q_start_tile * BLOCK + st_row, S - 1);
-    qa = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
-    qa2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
-    da = *(const ushortx8*)(dOp + (long)qg * os + st_col);
-    da2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
+  ushortx8 qa[NQ], qa2[NQ], da[NQ], da2[NQ];
+#pragma unroll
+  for (int h = 0; h < NQ; ++h) {
+    const int qg = min(q_start_tile * QB + 32 * h + st_row, S - 1);
+    qa[h] = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
+    qa2[h] = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
+    da[h] = *(const ushortx8*)(dOp + (long)qg * os + st_col);
+    da2[h] = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
   }
 
   for (int t = q_start_tile; t < n_q_tiles; ++t) {
-    const int tq0 = t * BLOCK;
     __syncthreads();
-    *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8)]) = qa;
-    *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8 + 1)]) = qa2;
-    *(ushortx8*)(&do_img[tr_img_off(st_row, st_col)]) = da;
-    *(ushortx8*)(&do_img[tr_img_off(st_row, st_col + 8)]) = da2;
+#pragma unroll
+    for (int h = 0; h < NQ; ++h) {
+      *(ushortx8*)(&q_rm[rm_off(32 * h + st_row, st_col / 8)]) = qa[h];
+      *(ushortx8*)(&q_rm[rm_off(32 * h + st_row, st_col / 8 + 1)]) = qa2[h];
+      *(ushortx8*)(&do_img[h * 8 * TRKEY4 + tr_img_off(st_row, st_col)]) =
+          da[h];
+      *(ushortx8*)(&do_img[h * 8 * TRKEY4 +
+                           tr_img_off(st_row, st_col + 8)]) = da2[h];
+    }
     __syncthreads();
     if (t + 1 < n_q_tiles) {
-      const int qg = min((t + 1) * BLOCK + st_row, S - 1);
-      qa = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
-      qa2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
-      da = *(const ushortx8*)(dOp + (long)qg * os + st_col);
-      da2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
+#pragma unroll
+      for (int h = 0; h < NQ; ++h) {
+        const int qg = min((t + 1) * QB + 32 * h + st_row, S - 1);
+        qa[h] = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
+        qa2[h] = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
+        da[h] = *(const ushortx8*)(dOp + (long)qg * os + st_col);
+        da2[h] = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
+      }
     }
 
+#pragma unroll
+    for (int h = 0; h < NQ; ++h) {
+    const int tq0 = t * QB + 32 * h;
     // S = Q.K^T with q REG-SPREAD, key LANE-LOCAL (operand roles swapped
     // vs the old scratch version): C[m=q][n=key] — P^T's A-fragment
     // (A[m=key][k=q]) then comes from the SAME in-register
@@ -395,15 +412,13 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
     for (int stp = 0; stp < 8; ++stp) {
       bf16x8 qfr = __builtin_bit_cast(
           bf16x8,
-          *(const ushortx8*)(&q_rm[rm_off(lane & 31, 2 * stp + (lane >> 5))]));
+          *(const ushortx8*)(&q_rm[rm_off(32 * h + (lane & 31),
+                                          2 * stp + (lane >> 5))]));
       st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfr, kfr[stp], st_acc,
                                                        0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // per-q-row lse via four float4 loads (L2-hot; all lanes of a half
-    // read the same addresses). Guard: the last tile may poke past S —
-    // Lp is sized S, so clamp the base (dead lanes mask the values out).
     // per-q-row lse fetched once per lane, broadcast per reg row by
     // shfl (LDS-local ~50cy; an r2m experiment with global float4 loads
     // instead put 200+cy L2 latency on the critical path and lost 3%)
@@ -440,7 +455,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
       pf[1] = __builtin_bit_cast(bf16x8, *(uint32x4_t*)w1);
     }
     {
-      const unsigned dbase = tr16_lane_base(do_img, lane);
+      const unsigned dbase =
+          tr16_lane_base(do_img + h * 8 * TRKEY4, lane);
       ushortx4_tr vrA[2][2], vrB[2][2];
       __builtin_amdgcn_s_setprio(1);
       tr16_issue_dt<1, 0>(dbase, vrA);
@@ -467,6 +483,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
       }
       __builtin_amdgcn_s_setprio(0);
     }
+    }  // NQ sub-tile loop
   }
   }  // q-head group loop
 
@@ -481,7 +498,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
   }
 }
 
-extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
+template <int NQ>
+__global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
     const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
     const ushort_t* __restrict__ V, const ushort_t* __restrict__ dO,
     const float* __restrict__ LSE, const float* __restrict__ Delta,
@@ -506,9 +524,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
   const ushort_t* Kp = K + (long)b * kb + (long)hk * kh;
   const ushort_t* Vp = V + (long)b * kb + (long)hk * kh;
 
-  __shared__ ushort_t q_rm[32 * KROW];
-  __shared__ ushort_t do_rm[32 * KROW];
-  __shared__ ushort_t q_img[8 * TRKEY4];  // tr16 image: Q^T B-fragments
+  __shared__ ushort_t q_rm[NQ * 32 * KROW];
+  __shared__ ushort_t do_rm[NQ * 32 * KROW];
+  __shared__ ushort_t q_img[NQ * 8 * TRKEY4];  // tr16 image: Q^T B-frags
 
   // own K/V rows in registers — B operands of S = Q.K^T and dP = dO.V^T
   // (key lane-local; the same per-lane bytes an A-fragment holds)
@@ -526,8 +544,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
   }
 
   floatx16 dk_acc[4] = {};
-  const int q_start_tile = causal ? (k_block * 128) / BLOCK : 0;
-  const int n_q_tiles = (S + BLOCK - 1) / BLOCK;
+  constexpr int QB = 32 * NQ;       // q rows per staged tile
+  const int q_start_tile = causal ? (k_block * 128) / QB : 0;
+  const int n_q_tiles = (S + QB - 1) / QB;
   const int st_row = threadIdx.x >> 3;
   const int st_col = (threadIdx.x & 7) * 16;
 
@@ -540,32 +559,43 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
   // double-buffered q/do tile prefetch (same pattern as attn_bwd_dv):
   // next tile's global loads issue OUTSIDE the LDS critical section so
   // the barrier never waits on HBM latency
-  ushortx8 qa, qa2, da, da2;
-  {
-    const int qg = min(q_start_tile * BLOCK + st_row, S - 1);
-    qa = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
-    qa2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
-    da = *(const ushortx8*)(dOp + (long)qg * os + st_col);
-    da2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
+  ushortx8 qa[NQ], qa2[NQ], da[NQ], da2[NQ];
+#pragma unroll
+  for (int h = 0; h < NQ; ++h) {
+    const int qg = min(q_start_tile * QB + 32 * h + st_row, S - 1);
+    qa[h] = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
+    qa2[h] = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
+    da[h] = *(const ushortx8*)(dOp + (long)qg * os + st_col);
+    da2[h] = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
   }
   for (int t = q_start_tile; t < n_q_tiles; ++t) {
-    const int tq0 = t * BLOCK;
     __syncthreads();
-    *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8)]) = qa;
-    *(ushortx8*)(&q_rm[rm_off(st_row, st_col / 8 + 1)]) = qa2;
-    *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8)]) = da;
-    *(ushortx8*)(&do_rm[rm_off(st_row, st_col / 8 + 1)]) = da2;
-    *(ushortx8*)(&q_img[tr_img_off(st_row, st_col)]) = qa;
-    *(ushortx8*)(&q_img[tr_img_off(st_row, st_col + 8)]) = qa2;
+#pragma unroll
+    for (int h = 0; h < NQ; ++h) {
+      *(ushortx8*)(&q_rm[rm_off(32 * h + st_row, st_col / 8)]) = qa[h];
+      *(ushortx8*)(&q_rm[rm_off(32 * h + st_row, st_col / 8 + 1)]) = qa2[h];
+      *(ushortx8*)(&do_rm[rm_off(32 * h + st_row, st_col / 8)]) = da[h];
+      *(ushortx8*)(&do_rm[rm_off(32 * h + st_row, st_col / 8 + 1)]) = da2[h];
+      *(ushortx8*)(&q_img[h * 8 * TRKEY4 + tr_img_off(st_row, st_col)]) =
+          qa[h];
+      *(ushortx8*)(&q_img[h * 8 * TRKEY4 +
+                          tr_img_off(st_row, st_col + 8)]) = qa2[h];
+    }
     __syncthreads();
     if (t + 1 < n_q_tiles) {
-      const int qg = min((t + 1) * BLOCK + st_row, S - 1);
-      qa = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
-      qa2 = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
-      da = *(const ushortx8*)(dOp + (long)qg * os + st_col);
-      da2 = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
+#pragma unroll
+      for (int h = 0; h < NQ; ++h) {
+        const int qg = min((t + 1) * QB + 32 * h + st_row, S - 1);
+        qa[h] = *(const ushortx8*)(Qp + (long)qg * qs + st_col);
+        qa2[h] = *(const ushortx8*)(Qp + (long)qg * qs + st_col + 8);
+        da[h] = *(const ushortx8*)(dOp + (long)qg * os + st_col);
+        da2[h] = *(const ushortx8*)(dOp + (long)qg * os + st_col + 8);
+      }
     }
 
+#pragma unroll
+    for (int h = 0; h < NQ; ++h) {
+    const int tq0 = t * QB + 32 * h;
     // S and dP with q REG-SPREAD, key LANE-LOCAL (roles swapped, as in
     // attn_bwd_dv): dS^T's A-fragment comes from the in-register T12
     // transpose — no wave-scratch LDS round trip.
@@ -575,11 +605,13 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
     for (int stp = 0; stp < 8; ++stp) {
       const int blk = 2 * stp + (lane >> 5);
       bf16x8 qfr = __builtin_bit_cast(
-          bf16x8, *(const ushortx8*)(&q_rm[rm_off(lane & 31, blk)]));
+          bf16x8,
+          *(const ushortx8*)(&q_rm[rm_off(32 * h + (lane & 31), blk)]));
       st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfr, kfr[stp], st_acc,
                                                        0, 0, 0);
       bf16x8 dofr = __builtin_bit_cast(
-          bf16x8, *(const ushortx8*)(&do_rm[rm_off(lane & 31, blk)]));
+          bf16x8,
+          *(const ushortx8*)(&do_rm[rm_off(32 * h + (lane & 31), blk)]));
       dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dofr, vfr[stp], dp_acc,
                                                        0, 0, 0);
     }
@@ -622,7 +654,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
       dsf[1] = __builtin_bit_cast(bf16x8, *(uint32x4_t*)w1);
     }
     {
-      const unsigned qb2 = tr16_lane_base(q_img, lane);
+      const unsigned qb2 =
+          tr16_lane_base(q_img + h * 8 * TRKEY4, lane);
       ushortx4_tr vrA[2][2], vrB[2][2];
       __builtin_amdgcn_s_setprio(1);
       tr16_issue_dt<1, 0>(qb2, vrA);
@@ -649,6 +682,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dk_kernel(
       }
       __builtin_amdgcn_s_setprio(0);
     }
+    }  // NQ sub-tile loop
   }
   }  // q-head group loop
 
@@ -735,18 +769,46 @@ extern "C" void launch_attn_bwd_dkv(const void* q, const void* k,
                                     int causal, void* stream,
                                     void* stream2) {
   dim3 grid((S + 127) / 128, Hk, B);
-  hipLaunchKernelGGL(attn_bwd_dv_kernel, grid, dim3(256), 0,
-                     (hipStream_t)stream, (const ushort_t*)q,
-                     (const ushort_t*)k, (const ushort_t*)dO, lse,
-                     (ushort_t*)dvp, S, Hq, Hk, qb, qh, qs, kb, kh, ks,
-                     ob, oh, os, gb, gh, gs, scale, causal);
-  hipLaunchKernelGGL(attn_bwd_dk_kernel, grid, dim3(256), 0,
-                     (hipStream_t)(stream2 ? stream2 : stream),
-                     (const ushort_t*)q,
-                     (const ushort_t*)k, (const ushort_t*)v,
-                     (const ushort_t*)dO, lse, delta, (ushort_t*)dkp, S, Hq,
-                     Hk, qb, qh, qs, kb, kh, ks, ob, oh, os, gb, gh, gs,
-                     scale, causal);
+  // staged-q-tile width. Measured A/B (gpurun_out/r2g/ab.log, B6 S4096):
+  // dv<2> (64-row stages, 214 VGPR no spill) is NEUTRAL vs dv<1> (8.96
+  // vs 8.91 ms f+b — barrier count is not the dv bottleneck) and dk<2>
+  // needs ~280 VGPR -> spills 104 B/lane and LOSES (9.72 ms). Default
+  // stays 1; ANTRAY_BWD_NQ=2 -> dv 64-row, =3 -> both 64-row, for
+  // re-measuring on future silicon/compilers.
+  static const int nq = [] {
+    const char* e = getenv("ANTRAY_BWD_NQ");
+    return e ? atoi(e) : 1;
+  }();
+  if (nq >= 2) {
+    hipLaunchKernelGGL(attn_bwd_dv_kernel<2>, grid, dim3(256), 0,
+                       (hipStream_t)stream, (const ushort_t*)q,
+                       (const ushort_t*)k, (const ushort_t*)dO, lse,
+                       (ushort_t*)dvp, S, Hq, Hk, qb, qh, qs, kb, kh, ks,
+                       ob, oh, os, gb, gh, gs, scale, causal);
+  } else {
+    hipLaunchKernelGGL(attn_bwd_dv_kernel<1>, grid, dim3(256), 0,
+                       (hipStream_t)stream, (const ushort_t*)q,
+                       (const ushort_t*)k, (const ushort_t*)dO, lse,
+                       (ushort_t*)dvp, S, Hq, Hk, qb, qh, qs, kb, kh, ks,
+                       ob, oh, os, gb, gh, gs, scale, causal);
+  }
+  if (nq >= 3) {
+    hipLaunchKernelGGL(attn_bwd_dk_kernel<2>, grid, dim3(256), 0,
+                       (hipStream_t)(stream2 ? stream2 : stream),
+                       (const ushort_t*)q,
+                       (const ushort_t*)k, (const ushort_t*)v,
+                       (const ushort_t*)dO, lse, delta, (ushort_t*)dkp, S, Hq,
+                       Hk, qb, qh, qs, kb, kh, ks, ob, oh, os, gb, gh, gs,
+                       scale, causal);
+  } else {
+    hipLaunchKernelGGL(attn_bwd_dk_kernel<1>, grid, dim3(256), 0,
+                       (hipStream_t)(stream2 ? stream2 : stream),
+                       (const ushort_t*)q,
+                       (const ushort_t*)k, (const ushort_t*)v,
+                       (const ushort_t*)dO, lse, delta, (ushort_t*)dkp, S, Hq,
+                       Hk, qb, qh, qs, kb, kh, ks, ob, oh, os, gb, gh, gs,
+                       scale, causal);
+  }
 }
 
 extern "C" void launch_attn_bwd_reduce_kv(const void* dkp, const void* dvp,
