@@ -125,6 +125,14 @@ def execute_plan(ops: List[Any]) -> Iterator[Any]:
 
 
 def _read_stage(op: ReadOp, ctx) -> Iterator[Any]:
+    if op.block_refs:
+        # blocks were put driver-side (from_numpy/from_arrow): hand the
+        # refs straight to the next stage — a read TASK here would only
+        # get+return the same bytes through a worker (one extra copy of
+        # the whole dataset + a task round trip per block)
+        yield from op.block_refs
+        if not op.read_tasks:
+            return
     from ant_ray_amd.data.backpressure import default_policies
 
     policies = default_policies(ctx)

@@ -24,6 +24,10 @@ class ReadOp:
     # with pruning/filtering folded into the read tasks (parquet supports
     # both natively; None = the source cannot push down)
     pushdown: Optional[Callable] = None
+    # pre-materialized arrow-block refs (from_numpy/from_arrow put their
+    # blocks driver-side): the executor yields these directly — no read
+    # task, no worker round trip, no second serialization of the data
+    block_refs: Optional[List] = None
 
 
 @dataclass
