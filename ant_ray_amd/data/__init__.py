@@ -175,43 +175,73 @@ def from_items(items: List[Any], *, parallelism: int = -1,
                            num_rows=len(items))])
 
 
+def _connected() -> bool:
+    from ant_ray_amd._private.worker import global_worker
+
+    return bool(getattr(global_worker, "connected", False))
+
+
 def from_numpy(arr: np.ndarray, column: str = "data") -> Dataset:
     """Blocks are converted to arrow and put into the object store HERE
     (reference from_numpy eagerly stores blocks too): downstream stages
     receive shm refs — no closure-captured 200 MB array pickled per read
     task, no worker round trip re-storing the same bytes. Large inputs
-    split into ~64 MB blocks for pipeline parallelism."""
-    import ant_ray_amd as ray
+    split into ~64 MB blocks for pipeline parallelism. Before ray.init
+    the source stays lazy (read tasks) for compatibility."""
+    import builtins
+
     from ant_ray_amd.data.block import BlockAccessor
 
     n = len(arr)
-    if n == 0:
-        return Dataset([ReadOp(name="FromNumpy", read_tasks=[], num_rows=0)])
-    import builtins
+    if n and _connected():
+        import ant_ray_amd as ray
 
-    per = max(1, (64 << 20) // max(arr.nbytes // n, 1))
-    refs = [
-        ray.put(BlockAccessor.for_block({column: arr[s : s + per]}).to_arrow())
-        for s in builtins.range(0, n, per)  # data.range shadows builtins
-    ]
-    return Dataset([ReadOp(name="FromNumpy", read_tasks=[], num_rows=n,
-                           block_refs=refs)])
+        per = max(1, (64 << 20) // max(arr.nbytes // n, 1))
+        refs = [
+            ray.put(
+                BlockAccessor.for_block({column: arr[s : s + per]}).to_arrow())
+            for s in builtins.range(0, n, per)  # data.range shadows builtins
+        ]
+        return Dataset([ReadOp(name="FromNumpy", read_tasks=[], num_rows=n,
+                               block_refs=refs)])
+
+    def read():
+        return {column: arr}
+
+    return Dataset([ReadOp(name="FromNumpy", read_tasks=[read] if n else [],
+                           num_rows=n)])
 
 
 def from_pandas(dfs) -> Dataset:
     import pyarrow as pa
 
     dfs = [dfs] if not isinstance(dfs, list) else dfs
-    tasks = [lambda d=d: pa.Table.from_pandas(d, preserve_index=False) for d in dfs]
-    return Dataset([ReadOp(name="FromPandas", read_tasks=tasks,
-                           num_rows=sum(len(d) for d in dfs))])
+    n = sum(len(d) for d in dfs)
+    if _connected():
+        import ant_ray_amd as ray
+
+        refs = [ray.put(pa.Table.from_pandas(d, preserve_index=False))
+                for d in dfs]
+        return Dataset([ReadOp(name="FromPandas", read_tasks=[],
+                               num_rows=n, block_refs=refs)])
+    tasks = [lambda d=d: pa.Table.from_pandas(d, preserve_index=False)
+             for d in dfs]
+    return Dataset([ReadOp(name="FromPandas", read_tasks=tasks, num_rows=n)])
 
 
 def from_arrow(tables) -> Dataset:
     tables = [tables] if not isinstance(tables, list) else tables
+    n = sum(t.num_rows for t in tables)
+    if _connected():
+        import ant_ray_amd as ray
+
+        # driver-resident tables: put them (zero-copy for arrow buffers)
+        # and skip the read-task round trip, as in from_numpy
+        refs = [ray.put(t) for t in tables]
+        return Dataset([ReadOp(name="FromArrow", read_tasks=[],
+                               num_rows=n, block_refs=refs)])
     tasks = [lambda t=t: t for t in tables]
-    return Dataset([ReadOp(name="FromArrow", read_tasks=tasks,
-                           num_rows=sum(t.num_rows for t in tables))])
+    return Dataset([ReadOp(name="FromArrow", read_tasks=tasks, num_rows=n)])
 
 
 def from_torch(torch_dataset) -> Dataset:
