@@ -370,9 +370,15 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   }
   // chunk count: fill the chip (~512 workgroups) without splitting below
   // 128 keys per chunk (r2e A/B: a 1024-WG target halved per-chunk work
-  // and lost 15% at B=8 T=4096; the 128-key floor keeps the small-B win)
-  int C = (int)std::min<long>(std::max<long>(1, 512 / std::max(1, B * Hk)),
-                              std::max<long>(1, ((long)T + 127) / 128));
+  // and lost 15% at B=8 T=4096; the 128-key floor keeps the small-B win).
+  // ANTRAY_DEC_WGS overrides the target for occupancy experiments.
+  static const long wg_target = [] {
+    const char* e = getenv("ANTRAY_DEC_WGS");
+    return e ? atol(e) : 512L;
+  }();
+  int C = (int)std::min<long>(
+      std::max<long>(1, wg_target / std::max(1, B * Hk)),
+      std::max<long>(1, ((long)T + 127) / 128));
   auto o = torch::empty({B, Hq, D}, q.options());
   torch::Tensor part;
   float* part_ptr = nullptr;

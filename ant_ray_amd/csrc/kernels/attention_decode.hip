@@ -32,7 +32,14 @@ typedef float floatx2 __attribute__((ext_vector_type(2)));
 
 // One partial record per (b, hq, chunk): unnormalized o (f32[D]) + m + l.
 // Stored [B, Hq, C, D+2] f32.
-extern "C" __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
+//
+// GQT is a TEMPLATE parameter (0 = runtime GQ): the per-head register
+// arrays (qf/o_acc/m/l) sized for the runtime maximum (8) cost 191
+// VGPRs and cap the kernel at 2 waves/SIMD; llama3-8b runs GQ=4 (~120
+// VGPRs -> 4 waves/SIMD, twice the latency hiding for the HBM-bound KV
+// walk). MAXOCC passes the matching __launch_bounds__ occupancy.
+template <int GQT, int MAXOCC>
+__global__ __launch_bounds__(256, MAXOCC) void attn_decode_kernel(
     const ushort_t* __restrict__ Q,   // [B, Hq, D] contiguous
     const ushort_t* __restrict__ K,   // [B, Hk, Tmax, D] via strides
     const ushort_t* __restrict__ V,   // same layout as K
@@ -45,7 +52,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
   const int chunk = blockIdx.x;
   const int hk = blockIdx.y;
   const int b = blockIdx.z;
-  const int GQ = Hq / Hk;
+  constexpr int GA = GQT > 0 ? GQT : DEC_MAX_GQ;  // register-array bound
+  const int GQ = GQT > 0 ? GQT : (Hq / Hk);
 
   const int seq_len = lens ? lens[b] : T;
   // chunk covers keys [c0, c1)
@@ -62,19 +70,18 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
   const ushort_t* Vp = V + (long)b * kb + (long)hk * kh;
 
   // Q fragments for the GQ query heads of this kv group (8 f32 per head)
-  floatx8 qf[DEC_MAX_GQ];
+  floatx8 qf[GA];
 #pragma unroll
-  for (int g = 0; g < DEC_MAX_GQ; ++g) {
-    if (g < GQ) {
-      const ushort_t* qrow = Q + (((long)b * Hq) + hk * GQ + g) * DEC_D + d0;
-      qf[g] = bf8_to_f32x8(*(const ushortx8*)qrow);
-    }
+  for (int g = 0; g < GA; ++g) {
+    if (GQT == 0 && g >= GQ) break;
+    const ushort_t* qrow = Q + (((long)b * Hq) + hk * GQ + g) * DEC_D + d0;
+    qf[g] = bf8_to_f32x8(*(const ushortx8*)qrow);
   }
 
-  float m_run[DEC_MAX_GQ], l_run[DEC_MAX_GQ];
-  floatx8 o_acc[DEC_MAX_GQ] = {};
+  float m_run[GA], l_run[GA];
+  floatx8 o_acc[GA] = {};
 #pragma unroll
-  for (int g = 0; g < DEC_MAX_GQ; ++g) { m_run[g] = -1e30f; l_run[g] = 0.f; }
+  for (int g = 0; g < GA; ++g) { m_run[g] = -1e30f; l_run[g] = 0.f; }
 
   // keys walk: group `group` handles keys c0+group, +16, +32 ... K/V rows
   // for the NEXT key are prefetched while the current key computes (a
@@ -95,8 +102,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
     const floatx8 kf = bf8_to_f32x8(kv);
     const floatx8 vf = bf8_to_f32x8(vv);
 #pragma unroll
-    for (int g = 0; g < DEC_MAX_GQ; ++g) {
-      if (g >= GQ) break;
+    for (int g = 0; g < GA; ++g) {
+      if (GQT == 0 && g >= GQ) break;
       // dot(q, k) over this lane's 8 dims, then 16-lane tree reduce
       float s = qf[g][0] * kf[0];
 #pragma unroll
@@ -129,8 +136,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
   extern __shared__ __attribute__((aligned(16))) float red[];
   const int rec = DEC_D + 2;
 #pragma unroll
-  for (int g = 0; g < DEC_MAX_GQ; ++g) {
-    if (g >= GQ) break;
+  for (int g = 0; g < GA; ++g) {
+    if (GQT == 0 && g >= GQ) break;
     float* dst = &red[(group * GQ + g) * rec];
 #pragma unroll
     for (int j = 0; j < 8; ++j) dst[d0 + j] = o_acc[g][j];
@@ -198,10 +205,19 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
   dim3 grid(C, Hk, B);
   const int GQ = Hq / Hk;
   const size_t lds = (size_t)16 * GQ * (DEC_D + 2) * sizeof(float);
-  hipLaunchKernelGGL(attn_decode_kernel, grid, dim3(256), lds,
-                     (hipStream_t)stream, (const ushort_t*)q,
-                     (const ushort_t*)k, (const ushort_t*)v, (ushort_t*)o,
-                     part, lens, T, Hq, Hk, C, kb, kh, ks, scale_log2);
+#define ANTRAY_DEC_LAUNCH(GQV, OCC)                                          \
+  hipLaunchKernelGGL((attn_decode_kernel<GQV, OCC>), grid, dim3(256), lds,   \
+                     (hipStream_t)stream, (const ushort_t*)q,                \
+                     (const ushort_t*)k, (const ushort_t*)v, (ushort_t*)o,   \
+                     part, lens, T, Hq, Hk, C, kb, kh, ks, scale_log2)
+  switch (GQ) {
+    case 1: ANTRAY_DEC_LAUNCH(1, 4); break;
+    case 2: ANTRAY_DEC_LAUNCH(2, 4); break;
+    case 4: ANTRAY_DEC_LAUNCH(4, 4); break;   // llama3-8b (Hq32/Hk8)
+    case 8: ANTRAY_DEC_LAUNCH(8, 2); break;   // 66 KiB LDS caps blocks/CU
+    default: ANTRAY_DEC_LAUNCH(0, 2); break;  // runtime-GQ guard path
+  }
+#undef ANTRAY_DEC_LAUNCH
   if (C > 1) {
     hipLaunchKernelGGL(attn_decode_combine_kernel, dim3(B * Hq), dim3(128), 0,
                        (hipStream_t)stream, part, (ushort_t*)o, C);

@@ -46,20 +46,41 @@ _pinned_pool: Dict[int, Any] = {}
 
 
 def _pinned_staging(t):
-    """Copy a CPU tensor through a REUSED pinned buffer (a fresh
-    pin_memory() per batch costs ~ms of hipHostMalloc; the pool keeps one
-    buffer per rounded size class)."""
+    """Copy a CPU tensor through a ROTATING pair of pinned buffers (a
+    fresh pin_memory() per batch costs ~ms of hipHostMalloc). Each
+    buffer carries a CUDA event recorded after its async H2D is
+    enqueued; before the CPU overwrites a buffer again it waits on that
+    event — so consecutive batches overlap host copy with device work
+    instead of a full stream sync per batch."""
     import torch
 
     nbytes = t.numel() * t.element_size()
     size_class = 1 << max(nbytes - 1, 1).bit_length()
-    buf = _pinned_pool.get(size_class)
+    pair = _pinned_pool.get(size_class)
+    if pair is None:
+        pair = {"bufs": [None, None], "events": [None, None], "next": 0}
+        _pinned_pool[size_class] = pair
+    i = pair["next"]
+    pair["next"] = 1 - i
+    buf = pair["bufs"][i]
     if buf is None or buf.numel() < nbytes:
         buf = torch.empty(size_class, dtype=torch.uint8, pin_memory=True)
-        _pinned_pool[size_class] = buf
+        pair["bufs"][i] = buf
+    ev = pair["events"][i]
+    if ev is not None:
+        ev.synchronize()  # prior H2D from THIS buffer has completed
     flat = buf[:nbytes].view(torch.uint8)
     flat.copy_(t.reshape(-1).view(torch.uint8))
-    return flat.view(t.dtype).reshape(t.shape)
+    return flat.view(t.dtype).reshape(t.shape), pair, i
+
+
+def _record_staging_event(pair, i):
+    import torch
+
+    ev = pair["events"][i]
+    if ev is None:
+        ev = pair["events"][i] = torch.cuda.Event()
+    ev.record()
 
 
 def to_torch_batch(batch: Dict[str, np.ndarray], dtypes=None,
@@ -90,11 +111,13 @@ def to_torch_batch(batch: Dict[str, np.ndarray], dtypes=None,
                 # cast-on-device kernel (csrc/kernels/data_transform.hip)
                 from ant_ray_amd import ops
 
-                t = _pinned_staging(t).to(device, non_blocking=True)
+                staged, pair, slot = _pinned_staging(t)
+                t = staged.to(device, non_blocking=True)
                 t = ops.cast_affine(t, 1.0, 0.0, out_dtype=want)
-                # the pooled pinned buffer is reused next batch: fence the
-                # async H2D before returning control
-                torch.cuda.current_stream().synchronize()
+                # record completion of this buffer's H2D; the rotating
+                # pool waits on it only when the SAME buffer comes around
+                # again (two batches later) instead of syncing every batch
+                _record_staging_event(pair, slot)
             else:
                 if want is not None:
                     t = t.to(want)
