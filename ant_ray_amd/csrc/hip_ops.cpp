@@ -368,13 +368,15 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                 lens->numel() == B, "lens must be int32 [B] on device");
     lens_ptr = lens->data_ptr<int>();
   }
-  // chunk count: fill the chip (~512 workgroups) without splitting below
-  // 128 keys per chunk (r2e A/B: a 1024-WG target halved per-chunk work
-  // and lost 15% at B=8 T=4096; the 128-key floor keeps the small-B win).
-  // ANTRAY_DEC_WGS overrides the target for occupancy experiments.
+  // chunk count: fill the chip (~1024 workgroups) without splitting
+  // below 128 keys per chunk. At the kernel's original 2 waves/SIMD a
+  // 1024-WG target lost 15% (r2e A/B); after the GQ-templated register
+  // diet raised occupancy to 4 waves/SIMD it WINS (r2k sweep: B8 T4096
+  // 2.86 -> 3.41 TB/s, B32 T2048 3.05 -> 4.43; 2048+ regress again).
+  // ANTRAY_DEC_WGS overrides the target for experiments.
   static const long wg_target = [] {
     const char* e = getenv("ANTRAY_DEC_WGS");
-    return e ? atol(e) : 512L;
+    return e ? atol(e) : 1024L;
   }();
   int C = (int)std::min<long>(
       std::max<long>(1, wg_target / std::max(1, B * Hk)),
