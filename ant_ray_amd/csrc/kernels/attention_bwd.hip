@@ -310,8 +310,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 
 // NQ = 32-row q sub-tiles staged per barrier interval (NQ=2 halves the
 // __syncthreads count per q row, same win as the forward kernel's NT).
-template <int NQ>
-__global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
+template <int NQ, int OCC = 2, bool KLDS = false>
+__global__ __launch_bounds__(256, OCC) void attn_bwd_dv_kernel(
     const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
     const ushort_t* __restrict__ dO,
     const float* __restrict__ LSE,
@@ -337,18 +337,38 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
 
   __shared__ ushort_t q_rm[NQ * 32 * KROW];
   __shared__ ushort_t do_img[NQ * 8 * TRKEY4];  // tr16 image: dO^T B-frags
+  // KLDS: the block's 128 K rows live in LDS instead of 64 VGPRs of
+  // per-lane registers — the register diet (190 -> ~130) buys a third
+  // wave per SIMD, which matters because PMC shows dv 46% wait-bound
+  // (gpurun_out/r2l). Reads come back as one ds_read_b128 per MFMA.
+  __shared__ ushort_t k_lds[KLDS ? 128 * KROW : 8];
 
-  // own K rows in registers — used as the B operand of S = Q.K^T
-  // (B[k=d][n=key]: n = lane-local key, k-slices = d — the same per-lane
-  // bytes an A-fragment holds, so the load is unchanged)
-  bf16x8 kfr[8];
-  {
+  // own K rows — the B operand of S = Q.K^T (B[k=d][n=key]: n =
+  // lane-local key, k-slices = d — the same per-lane bytes an
+  // A-fragment holds, so the load is unchanged)
+  bf16x8 kfr[KLDS ? 1 : 8];
+  if (!KLDS) {
     const long kg = (long)min(key_row, S - 1) * ks;
     const int dbase = (lane >> 5) * 8;
 #pragma unroll
     for (int st = 0; st < 8; ++st)
-      kfr[st] = __builtin_bit_cast(
+      kfr[KLDS ? 0 : st] = __builtin_bit_cast(
           bf16x8, *(const ushortx8*)(Kp + kg + st * 16 + dbase));
+  } else {
+    // stage all 128 rows once: thread covers rows (tid>>3) + 32*r, 16
+    // dims at (tid&7)*16 — same T2-swizzled row-major image as q_rm
+    const int sr = threadIdx.x >> 3;
+    const int sc = (threadIdx.x & 7) * 16;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int krow = k_block * 128 + 32 * r + sr;
+      const long kg = (long)min(krow, S - 1) * ks;
+      *(ushortx8*)(&k_lds[rm_off(32 * r + sr, sc / 8)]) =
+          *(const ushortx8*)(Kp + kg + sc);
+      *(ushortx8*)(&k_lds[rm_off(32 * r + sr, sc / 8 + 1)]) =
+          *(const ushortx8*)(Kp + kg + sc + 8);
+    }
+    // the first q-tile barrier below makes these writes visible
   }
 
   floatx16 dv_acc[4] = {};
@@ -414,7 +434,13 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dv_kernel(
           bf16x8,
           *(const ushortx8*)(&q_rm[rm_off(32 * h + (lane & 31),
                                           2 * stp + (lane >> 5))]));
-      st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfr, kfr[stp], st_acc,
+      const bf16x8 kop =
+          KLDS ? __builtin_bit_cast(
+                     bf16x8,
+                     *(const ushortx8*)(&k_lds[rm_off(
+                         wave * 32 + (lane & 31), 2 * stp + (lane >> 5))]))
+               : kfr[KLDS ? 0 : stp];
+      st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfr, kop, st_acc,
                                                        0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
@@ -779,8 +805,29 @@ extern "C" void launch_attn_bwd_dkv(const void* q, const void* k,
     const char* e = getenv("ANTRAY_BWD_NQ");
     return e ? atoi(e) : 1;
   }();
+  // dv variant: default 4 = K rows in LDS, 160 VGPR, 3 waves/SIMD —
+  // measured 8.70 vs 8.92 ms f+b against the 2-wave register-resident
+  // variant (PMC r2l showed dv 46% wait-bound; the third wave hides
+  // it). 3 = force-capped 3-wave variant (spills 80 B/lane, LOSES at
+  // 9.32 ms — kept for re-measurement); 2 = original.
+  static const int dv_occ = [] {
+    const char* e = getenv("ANTRAY_BWD_DV_OCC");
+    return e ? atoi(e) : 4;
+  }();
   if (nq >= 2) {
     hipLaunchKernelGGL(attn_bwd_dv_kernel<2>, grid, dim3(256), 0,
+                       (hipStream_t)stream, (const ushort_t*)q,
+                       (const ushort_t*)k, (const ushort_t*)dO, lse,
+                       (ushort_t*)dvp, S, Hq, Hk, qb, qh, qs, kb, kh, ks,
+                       ob, oh, os, gb, gh, gs, scale, causal);
+  } else if (dv_occ >= 4) {
+    hipLaunchKernelGGL((attn_bwd_dv_kernel<1, 3, true>), grid, dim3(256), 0,
+                       (hipStream_t)stream, (const ushort_t*)q,
+                       (const ushort_t*)k, (const ushort_t*)dO, lse,
+                       (ushort_t*)dvp, S, Hq, Hk, qb, qh, qs, kb, kh, ks,
+                       ob, oh, os, gb, gh, gs, scale, causal);
+  } else if (dv_occ >= 3) {
+    hipLaunchKernelGGL((attn_bwd_dv_kernel<1, 3>), grid, dim3(256), 0,
                        (hipStream_t)stream, (const ushort_t*)q,
                        (const ushort_t*)k, (const ushort_t*)dO, lse,
                        (ushort_t*)dvp, S, Hq, Hk, qb, qh, qs, kb, kh, ks,
