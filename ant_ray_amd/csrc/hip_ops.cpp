@@ -42,7 +42,10 @@ void launch_bf16_to_f32(const void*, float*, int64_t, hipStream_t);
 void launch_f32_to_bf16(const float*, void*, int64_t, hipStream_t);
 void launch_attn_decode(const void*, const void*, const void*, void*, float*,
                         const int*, int, int, int, int, int, long, long, long,
-                        float, void*);
+                        long, float, void*);
+void launch_rope_cache_write(void*, void*, void*, const int*, const float*,
+                             const float*, int, int, int, int, long, long,
+                             long, hipStream_t);
 void launch_cast_affine_u8(const void*, void*, const float*, const float*,
                            long, int, int, int, void*);
 void launch_cast_affine_f32(const void*, void*, const float*, const float*,
@@ -391,7 +394,62 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   }
   launch_attn_decode(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                      part_ptr, lens_ptr, B, (int)T, Hq, Hk, C, k.stride(0),
-                     k.stride(1), k.stride(2), (float)scale,
+                     k.stride(1), k.stride(2), (long)Hq * D, (float)scale,
+                     (void*)cur_stream());
+  return o;
+}
+
+// Device-pos single-token decode step (graph-capturable): rope q/k +
+// cache writes + flash-decode in one call, with the position read from
+// the DEVICE lens buffer (pos = lens[b]-1) — no host scalar depends on
+// the step index, so generate() captures the whole token step in a
+// hipGraph and replays it (the host-pos path pays ~260 Python launches
+// per token). qkv: [B, 1, (Hq+2Hk)*D] or [B, (Hq+2Hk)*D] contiguous.
+torch::Tensor decode_step_attn(torch::Tensor qkv, torch::Tensor ck,
+                               torch::Tensor cv, torch::Tensor lens,
+                               torch::Tensor cos_t, torch::Tensor sin_t,
+                               int64_t Hq, int64_t Hk, double scale) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.scalar_type() == torch::kBFloat16 &&
+              qkv.is_contiguous(), "qkv must be contiguous bf16 gpu");
+  TORCH_CHECK(ck.dim() == 4 && cv.dim() == 4 && ck.strides() == cv.strides(),
+              "caches must be [B,Hk,T,D] with a shared layout");
+  TORCH_CHECK(ck.stride(3) == 1, "cache rows must be contiguous");
+  TORCH_CHECK(lens.is_cuda() && lens.scalar_type() == torch::kInt32,
+              "lens must be int32 on device");
+  int B = ck.size(0), D = ck.size(3);
+  int Tmax = ck.size(2);
+  TORCH_CHECK(D == 128, "decode_step: D must be 128");
+  TORCH_CHECK(qkv.numel() == (long)B * (Hq + 2 * Hk) * D, "qkv shape");
+  TORCH_CHECK(lens.numel() == B, "lens must be [B]");
+  TORCH_CHECK(cos_t.scalar_type() == torch::kFloat32 &&
+              cos_t.is_contiguous() && cos_t.size(0) >= Tmax,
+              "cos table must cover the cache");
+  launch_rope_cache_write(qkv.data_ptr(), ck.data_ptr(), cv.data_ptr(),
+                          lens.data_ptr<int>(), cos_t.data_ptr<float>(),
+                          sin_t.data_ptr<float>(), B, (int)Hq, (int)Hk, D,
+                          ck.stride(0), ck.stride(1), ck.stride(2),
+                          cur_stream());
+  static const long wg_target2 = [] {
+    const char* e = getenv("ANTRAY_DEC_WGS");
+    return e ? atol(e) : 1024L;
+  }();
+  // C is sized from the STATIC Tmax (graph shapes cannot depend on the
+  // step); chunks past lens[b] compute empty ranges and combine to 0
+  int C = (int)std::min<long>(
+      std::max<long>(1, wg_target2 / std::max(1, B * (int)Hk)),
+      std::max<long>(1, ((long)Tmax + 127) / 128));
+  auto o = torch::empty({B, Hq, D}, qkv.options());
+  torch::Tensor part;
+  float* part_ptr = nullptr;
+  if (C > 1) {
+    part = torch::empty({(long)B * Hq * C * (D + 2)},
+                        qkv.options().dtype(torch::kFloat32));
+    part_ptr = part.data_ptr<float>();
+  }
+  launch_attn_decode(qkv.data_ptr(), ck.data_ptr(), cv.data_ptr(),
+                     o.data_ptr(), part_ptr, lens.data_ptr<int>(), B, Tmax,
+                     (int)Hq, (int)Hk, C, ck.stride(0), ck.stride(1),
+                     ck.stride(2), (long)(Hq + 2 * Hk) * D, (float)scale,
                      (void*)cur_stream());
   return o;
 }
@@ -467,6 +525,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("v"), py::arg("o"), py::arg("lse"), py::arg("scale"),
         py::arg("causal") = true, py::arg("dq_out") = py::none(),
         py::arg("dk_out") = py::none(), py::arg("dv_out") = py::none());
+  m.def("decode_step_attn", &decode_step_attn, py::arg("qkv"), py::arg("ck"),
+        py::arg("cv"), py::arg("lens"), py::arg("cos"), py::arg("sin"),
+        py::arg("hq"), py::arg("hk"), py::arg("scale"));
   m.def("attn_decode", &attn_decode, py::arg("q"), py::arg("k"), py::arg("v"),
         py::arg("T"), py::arg("scale"), py::arg("lens") = py::none());
   m.def("cast_affine", &cast_affine, py::arg("x"), py::arg("scale"),

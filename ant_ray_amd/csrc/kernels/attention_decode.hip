@@ -48,6 +48,8 @@ __global__ __launch_bounds__(256, MAXOCC) void attn_decode_kernel(
     const int* __restrict__ lens,     // [B] valid lengths (nullptr -> T)
     int T, int Hq, int Hk, int C,
     long kb, long kh, long ks,        // K/V strides (elements)
+    long qbs,                         // Q batch stride (Hq*D, or the fused
+                                      // qkv row stride when q is a region)
     float scale_log2) {
   const int chunk = blockIdx.x;
   const int hk = blockIdx.y;
@@ -74,7 +76,8 @@ __global__ __launch_bounds__(256, MAXOCC) void attn_decode_kernel(
 #pragma unroll
   for (int g = 0; g < GA; ++g) {
     if (GQT == 0 && g >= GQ) break;
-    const ushort_t* qrow = Q + (((long)b * Hq) + hk * GQ + g) * DEC_D + d0;
+    const ushort_t* qrow =
+        Q + (long)b * qbs + (long)(hk * GQ + g) * DEC_D + d0;
     qf[g] = bf8_to_f32x8(*(const ushortx8*)qrow);
   }
 
@@ -199,8 +202,8 @@ extern "C" __global__ __launch_bounds__(128, 8) void attn_decode_combine_kernel(
 extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
                                    void* o, float* part, const int* lens,
                                    int B, int T, int Hq, int Hk, int C,
-                                   long kb, long kh, long ks, float scale,
-                                   void* stream) {
+                                   long kb, long kh, long ks, long qbs,
+                                   float scale, void* stream) {
   const float scale_log2 = scale * 1.4426950408889634f;
   dim3 grid(C, Hk, B);
   const int GQ = Hq / Hk;
@@ -209,7 +212,7 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
   hipLaunchKernelGGL((attn_decode_kernel<GQV, OCC>), grid, dim3(256), lds,   \
                      (hipStream_t)stream, (const ushort_t*)q,                \
                      (const ushort_t*)k, (const ushort_t*)v, (ushort_t*)o,   \
-                     part, lens, T, Hq, Hk, C, kb, kh, ks, scale_log2)
+                     part, lens, T, Hq, Hk, C, kb, kh, ks, qbs, scale_log2)
   switch (GQ) {
     case 1: ANTRAY_DEC_LAUNCH(1, 4); break;
     case 2: ANTRAY_DEC_LAUNCH(2, 4); break;

@@ -80,11 +80,20 @@ class LlamaAttention(nn.Module):
         self.wqkv = nn.Linear(cfg.hidden, (Hq + 2 * Hk) * D, bias=False)
         self.wo = nn.Linear(Hq * D, cfg.hidden, bias=False)
 
-    def forward(self, y, cos, sin, cache=None, layer_idx=0, pos=0):
+    def forward(self, y, cos, sin, cache=None, layer_idx=0, pos=0,
+                lens=None):
         cfg = self.cfg
         Hq, Hk, D = cfg.n_heads, cfg.n_kv_heads, cfg.head_dim
         B, S, _ = y.shape
         qkv = self.wqkv(y)
+        if lens is not None:
+            # graph-capturable decode step: position comes from the DEVICE
+            # lens buffer (rope + cache write + flash-decode in one call —
+            # no host scalar depends on the step index)
+            ck, cv = cache.layer(layer_idx)
+            o = ops.decode_step_attn(qkv.view(B, (Hq + 2 * Hk) * D), ck, cv,
+                                     lens, cos, sin, Hq, Hk)
+            return self.wo(o.view(B, 1, Hq * D))
         if (cache is None and qkv.is_cuda and D == 128
                 and torch.is_grad_enabled() and qkv.requires_grad
                 and os.environ.get("ANTRAY_FLASH", "1") != "0"):
@@ -143,14 +152,16 @@ class LlamaBlock(nn.Module):
         self.attn = LlamaAttention(cfg)
         self.mlp = LlamaMLP(cfg)
 
-    def forward(self, h, res, cos, sin, cache=None, layer_idx=0, pos=0):
+    def forward(self, h, res, cos, sin, cache=None, layer_idx=0, pos=0,
+                lens=None):
         # h = contribution from the previous sublayer; res = residual stream
         if res is None:
             res = h
             y = ops.rmsnorm(h, self.attn_norm, self.cfg.eps)
         else:
             y, res = ops.fused_add_rmsnorm(h, res, self.attn_norm, self.cfg.eps)
-        a = self.attn(y, cos, sin, cache=cache, layer_idx=layer_idx, pos=pos)
+        a = self.attn(y, cos, sin, cache=cache, layer_idx=layer_idx, pos=pos,
+                      lens=lens)
         y2, res = ops.fused_add_rmsnorm(a, res, self.mlp_norm, self.cfg.eps)
         return self.mlp(y2), res
 
@@ -189,13 +200,14 @@ class LlamaForCausalLM(nn.Module):
     def head_weight(self):
         return self.embed.weight if self.lm_head is None else self.lm_head.weight
 
-    def forward(self, tokens, targets=None, cache=None, pos=0):
+    def forward(self, tokens, targets=None, cache=None, pos=0, lens=None):
         B, S = tokens.shape
         h = self.embed(tokens)
         res = None
         cos, sin = self.rope_cos, self.rope_sin
         for i, blk in enumerate(self.blocks):
-            h, res = blk(h, res, cos, sin, cache=cache, layer_idx=i, pos=pos)
+            h, res = blk(h, res, cos, sin, cache=cache, layer_idx=i, pos=pos,
+                         lens=lens)
         if res is None:
             y = ops.rmsnorm(h, self.final_norm, self.cfg.eps)
         else:
@@ -228,6 +240,12 @@ class LlamaForCausalLM(nn.Module):
                                              S + max_new_tokens), dev)
         logits = self.forward(tokens, cache=cache, pos=0)  # prefill
         cache.pos = S
+        use_graph = (tokens.is_cuda and self.cfg.head_dim == 128
+                     and temperature == 0
+                     and os.environ.get("ANTRAY_DECODE_GRAPH", "1") != "0")
+        if use_graph:
+            return self._generate_graphed(tokens, logits, cache,
+                                          max_new_tokens)
         out = [tokens]
         cur = None
         for _ in range(max_new_tokens):
@@ -241,6 +259,66 @@ class LlamaForCausalLM(nn.Module):
                 break
             logits = self.forward(cur, cache=cache, pos=cache.pos)
             cache.pos += 1
+        return torch.cat(out, dim=1)
+
+    @torch.no_grad()
+    def _generate_graphed(self, tokens, logits, cache: "KVCache",
+                          max_new_tokens: int):
+        """Greedy decode with the whole token step captured in a hipGraph.
+
+        The device-pos decode path (ops.decode_step_attn: rope + cache
+        write + flash-decode reading pos from the device lens buffer)
+        makes every kernel argument independent of the step index, so
+        ONE capture serves all steps: each replay advances `cur` (next
+        token) and `lens` in place on the GPU. The eager host-pos loop
+        pays ~260 Python-side launches per token; the replay is one
+        hipGraphLaunch. Capture happens once per (B, cache) generate
+        call (first token runs eagerly as the capture warmup)."""
+        B, S = tokens.shape
+        dev = tokens.device
+        lens = torch.full((B,), S, dtype=torch.int32, device=dev)
+        cur = logits.argmax(dim=-1, keepdim=True)  # token 1 (from prefill)
+        out = [tokens]
+        n_left = min(max_new_tokens, cache.max_seq - S + 1)
+
+        def step():
+            lg = self.forward(cur, cache=cache, lens=lens)
+            cur.copy_(lg.argmax(dim=-1, keepdim=True))
+
+        emitted = 0
+        # warmup steps run eagerly (they also produce real tokens)
+        for _ in range(min(2, n_left)):
+            out.append(cur.clone())
+            emitted += 1
+            if emitted >= n_left:
+                break
+            lens += 1
+            cache.pos += 1
+            step()
+        if emitted < n_left and n_left - emitted >= 8:
+            # capture costs ~10s of ms once; only worth it for a real run
+            g = torch.cuda.CUDAGraph()
+            torch.cuda.synchronize()
+            with torch.cuda.graph(g):
+                lens.add_(1)
+                step()
+            # capture records without executing: lens/cur are unchanged
+            while emitted < n_left:
+                out.append(cur.clone())
+                emitted += 1
+                if emitted >= n_left:
+                    break
+                g.replay()
+                cache.pos += 1
+        else:
+            while emitted < n_left:
+                out.append(cur.clone())
+                emitted += 1
+                if emitted >= n_left:
+                    break
+                lens += 1
+                cache.pos += 1
+                step()
         return torch.cat(out, dim=1)
 
     def num_params(self):

@@ -273,6 +273,23 @@ def attention_decode(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return ref.attention_decode_ref(q, kc, vc, lens=lens, scale=float(scale))
 
 
+def decode_step_attn(qkv: torch.Tensor, ck: torch.Tensor, cv: torch.Tensor,
+                     lens: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+                     hq: int, hk: int, scale=None) -> torch.Tensor:
+    """Device-pos single-token decode step: rope q/k + cache writes +
+    flash-decode in ONE call, with the token position read from the
+    DEVICE `lens` buffer (pos = lens[b]-1). No host scalar depends on
+    the step index, so models/llama.py captures the whole token step in
+    a hipGraph and replays it. qkv: [B,1,(Hq+2Hk)*D] from the fused
+    projection; caches [B,Hk,Tmax,D]; lens int32 [B] (valid length
+    INCLUDING the new token). GPU-only (generation path)."""
+    D = ck.shape[-1]
+    if scale is None:
+        scale = D ** -0.5
+    return _hip().decode_step_attn(qkv, ck, cv, lens, cos, sin, hq, hk,
+                                   float(scale))
+
+
 # ------------------------------------------------------- data transforms
 
 
