@@ -34,6 +34,23 @@ class NativeLLMEngine:
         self.max_seq = max_seq
         self.vocab = self.model.cfg.vocab
         self._lock = threading.Lock()
+        # persistent per-batch-bucket graphed decoders (greedy + CUDA +
+        # D=128): KV cache and hipGraph capture are paid once per bucket,
+        # then every token of every request is one graph replay
+        self._decoders = {}
+
+    def _graphed_decoder(self, batch: int):
+        from ant_ray_amd.models.llama import GraphedDecoder
+
+        bucket = 1
+        while bucket < batch:
+            bucket *= 2
+        dec = self._decoders.get(bucket)
+        if dec is None:
+            dec = GraphedDecoder(self.model, bucket, self.max_seq,
+                                 self.device)
+            self._decoders[bucket] = dec
+        return dec, bucket
 
     @torch.no_grad()
     def generate_tokens(self, prompts: List[List[int]], max_new_tokens: int,
@@ -45,12 +62,22 @@ class NativeLLMEngine:
             for i, p in enumerate(prompts):
                 by_len.setdefault(len(p), []).append(i)
             out: List[Optional[List[int]]] = [None] * len(prompts)
+            use_graph = (temperature == 0
+                         and str(self.device).startswith("cuda")
+                         and self.model.cfg.head_dim == 128)
             for plen, idxs in by_len.items():
                 toks = torch.tensor([prompts[i] for i in idxs],
                                     dtype=torch.long, device=self.device)
-                gen = self.model.generate(toks, max_new_tokens,
-                                          temperature=temperature)
-                new = gen[:, plen:].tolist()
+                if use_graph:
+                    dec, bucket = self._graphed_decoder(len(idxs))
+                    if len(idxs) < bucket:  # pad rows replay row 0
+                        toks = torch.cat(
+                            [toks, toks[:1].expand(bucket - len(idxs), -1)])
+                    gen = dec.generate(toks, max_new_tokens)
+                else:
+                    gen = self.model.generate(toks, max_new_tokens,
+                                              temperature=temperature)
+                new = gen[: len(idxs), plen:].tolist()
                 for j, i in enumerate(idxs):
                     out[i] = new[j]
             return out

@@ -72,6 +72,67 @@ class KVCache:
         return self.k.numel() * 2 * 2
 
 
+class GraphedDecoder:
+    """Persistent hipGraph-captured greedy decode loop for a FIXED
+    (model, batch, cache): capture happens ONCE ever, then every token
+    of every subsequent generate() on this decoder is one
+    hipGraphLaunch. Serve replicas keep one per batch bucket
+    (llm/native_engine.py) so repeated requests never pay capture or
+    KV-cache allocation again. Requires head_dim 128 + CUDA + greedy."""
+
+    def __init__(self, model: "LlamaForCausalLM", batch: int, max_seq: int,
+                 device):
+        self.model = model
+        self.B = batch
+        self.cache = KVCache(model.cfg, batch, max_seq, device)
+        self.lens = torch.zeros(batch, dtype=torch.int32, device=device)
+        self.cur = torch.zeros(batch, 1, dtype=torch.long, device=device)
+        self.graph = None
+
+    def _step(self):
+        lg = self.model.forward(self.cur, cache=self.cache, lens=self.lens)
+        self.cur.copy_(lg.argmax(dim=-1, keepdim=True))
+
+    @torch.no_grad()
+    def generate(self, tokens, max_new_tokens: int):
+        B, S = tokens.shape
+        assert B == self.B, "decoder is fixed-batch; pad or re-bucket"
+        logits = self.model.forward(tokens, cache=self.cache, pos=0)
+        self.cache.pos = S
+        self.lens.fill_(S)
+        self.cur.copy_(logits.argmax(dim=-1, keepdim=True))
+        out = [tokens]
+        n_left = min(max_new_tokens, self.cache.max_seq - S + 1)
+        emitted = 0
+        while (emitted < n_left and self.graph is None and emitted < 2):
+            out.append(self.cur.clone())
+            emitted += 1
+            if emitted >= n_left:
+                break
+            self.lens += 1
+            self.cache.pos += 1
+            self._step()
+        if emitted < n_left and self.graph is None and n_left - emitted >= 8:
+            g = torch.cuda.CUDAGraph()
+            torch.cuda.synchronize()
+            with torch.cuda.graph(g):
+                self.lens.add_(1)
+                self._step()
+            self.graph = g
+        while emitted < n_left:
+            out.append(self.cur.clone())
+            emitted += 1
+            if emitted >= n_left:
+                break
+            if self.graph is not None:
+                self.graph.replay()
+            else:
+                self.lens += 1
+                self._step()
+            self.cache.pos += 1
+        return torch.cat(out, dim=1)
+
+
 class LlamaAttention(nn.Module):
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
