@@ -25,8 +25,11 @@ class NativeLLMEngine:
 
     def __init__(self, model_name: str, max_seq: int = 4096,
                  seed: int = 0, device: str = "cuda"):
-        from ant_ray_amd.models import build_model
+        from ant_ray_amd.models import build_model, setup_tunableop
 
+        if str(device).startswith("cuda"):
+            setup_tunableop()  # tuned hipBLASLt algos for the skinny
+            # decode GEMM shapes (profiles/tunableop_gfx950.csv)
         torch.manual_seed(seed)
         self.model = build_model(model_name, device=device, seq_len=max_seq)
         self.model.eval()

@@ -21,3 +21,38 @@ def build_model(name: str, device="cuda", seq_len: int = 4096):
         cfg = GPT2Config.small(max_seq=min(seq_len, 1024))
         return GPT2LMHeadModel(cfg, device=device)
     raise ValueError(f"unknown model {name}")
+
+
+def setup_tunableop(local_rank: int = 0):
+    """Load the committed hipBLASLt TunableOp table (profiles/
+    tunableop_gfx950.csv) for this process, or (re)tune with
+    ANTRAY_TUNE=1 (rows appended for shapes not yet in the table —
+    decode/serve GEMMs are M=1..16 skinny shapes the training sweep
+    never sees). Mirrors bench.py's setup; safe no-op on CPU."""
+    import os
+
+    if os.environ.get("ANTRAY_TUNEOP") == "0":
+        return
+    try:
+        import torch.cuda.tunable as tun
+    except ImportError:
+        return
+    here = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    table = os.path.join(here, "..", "profiles", "tunableop_gfx950.csv")
+    table = os.path.abspath(table)
+    tuning = os.environ.get("ANTRAY_TUNE") == "1"
+    if not tuning and not os.path.exists(table):
+        return
+    tun.enable(True)
+    out = os.environ.get("ANTRAY_TUNE_OUT",
+                         f"/tmp/tunableop_rank{local_rank}.csv")
+    tun.set_filename(out if tuning else table)
+    if tuning:
+        tun.tuning_enable(True)
+        tun.set_max_tuning_duration(100)
+    else:
+        tun.tuning_enable(False)
+        try:
+            tun.read_file(table)
+        except Exception:
+            pass

@@ -49,7 +49,9 @@ def main():
         print(f"B={B:3d} T={T:5d}: {t*1e6:8.1f} us  {bytes_rd/t/1e12:6.2f} TB/s")
 
     print(f"== generate() {args.model} ==")
-    from ant_ray_amd.models import build_model
+    from ant_ray_amd.models import build_model, setup_tunableop
+
+    setup_tunableop()
 
     m = build_model(args.model, device="cuda",
                     seq_len=args.prompt + args.new_tokens + 8)
