@@ -41,6 +41,17 @@ class NativeLLMEngine:
         # D=128): KV cache and hipGraph capture are paid once per bucket,
         # then every token of every request is one graph replay
         self._decoders = {}
+        if (str(device).startswith("cuda")
+                and self.model.cfg.head_dim == 128):
+            # pre-capture the B=1 bucket: the process's FIRST hipGraph
+            # instantiation costs ~0.7 s — pay it at replica startup,
+            # not on the first request
+            try:
+                dec, _ = self._graphed_decoder(1)
+                dummy = torch.zeros(1, 8, dtype=torch.long, device=device)
+                dec.generate(dummy, 12)
+            except Exception:
+                pass
 
     def _graphed_decoder(self, batch: int):
         from ant_ray_amd.models.llama import GraphedDecoder
