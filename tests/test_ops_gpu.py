@@ -250,6 +250,32 @@ class TestGenerate:
         assert torch.equal(out[:, :17 + 4], cur[:, :17 + 4]), (
             out.tolist(), cur.tolist())
 
+    def test_graphed_decoder_matches_eager_and_reuses_capture(self):
+        """The persistent GraphedDecoder (hipGraph token step, device-pos
+        rope/cache-write/flash-decode) must produce the SAME tokens as
+        the eager host-pos loop, and keep producing them on a SECOND
+        generate call that reuses the captured graph + KV cache."""
+        import os
+
+        from ant_ray_amd.models import build_model
+        from ant_ray_amd.models.llama import GraphedDecoder
+
+        torch.manual_seed(11)
+        m = build_model("llama-tiny-d128", device=DEV, seq_len=256)
+        m.eval()
+        tokens = torch.randint(0, 1024, (2, 13), device=DEV)
+        os.environ["ANTRAY_DECODE_GRAPH"] = "0"
+        try:
+            eager = m.generate(tokens, max_new_tokens=16)
+        finally:
+            os.environ.pop("ANTRAY_DECODE_GRAPH", None)
+        dec = GraphedDecoder(m, batch=2, max_seq=256, device=DEV)
+        first = dec.generate(tokens, 16)
+        assert dec.graph is not None, "capture did not happen"
+        second = dec.generate(tokens, 16)  # replays the SAME graph
+        assert torch.equal(first, eager), (first.tolist(), eager.tolist())
+        assert torch.equal(second, eager)
+
 
 class TestDataTransforms:
     """Fused Data-plane kernels (data_transform.hip) vs torch reference."""
