@@ -113,7 +113,11 @@ extern "C" __global__ void attn_bwd_preprocess_kernel(
 
 // ------------------------------------------------------------ dQ kernel
 
-extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
+// NT = 32-key sub-tiles staged per barrier interval (NT=2 = 64-key
+// stages, halves the __syncthreads count — the same lever as the
+// forward kernel's NT; LDS 2x25.2 KB still fits 2 blocks/CU).
+template <int NT>
+__global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
     const ushort_t* __restrict__ V, const ushort_t* __restrict__ dO,
     const float* __restrict__ LSE,    // [B,Hq,S] log2 domain
@@ -144,9 +148,9 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   const float* Lp = LSE + ((long)b * Hq + hq) * S;
   const float* Dp = Delta + ((long)b * Hq + hq) * S;
 
-  __shared__ ushort_t k_rm[32 * KROW];
-  __shared__ ushort_t v_rm[32 * KROW];
-  __shared__ ushort_t k_img[8 * TRKEY4];  // tr16 image: K^T B-fragments
+  __shared__ ushort_t k_rm[NT * 32 * KROW];
+  __shared__ ushort_t v_rm[NT * 32 * KROW];
+  __shared__ ushort_t k_img[NT * 8 * TRKEY4];  // tr16 image: K^T B-frags
 
   // Q and dO fragments (B: lane n = q, kdim = d slices)
   bf16x8 qf[8], dof[8];
@@ -178,34 +182,47 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   // T14 double-buffered prefetch (same pattern as the dv/dk kernels): tile
   // t+1's global loads are issued right after tile t is staged, so the
   // inter-barrier critical path never waits on HBM latency.
-  ushortx8 k0, k1, v0, v1;
-  {
-    const int krow_g = min(st_row, S - 1);
-    k0 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
-    k1 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col + 8);
-    v0 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col);
-    v1 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col + 8);
+  ushortx8 k0[NT], k1[NT], v0[NT], v1[NT];
+#pragma unroll
+  for (int hh = 0; hh < NT; ++hh) {
+    const int krow_g = min(32 * hh + st_row, S - 1);
+    k0[hh] = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
+    k1[hh] = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col + 8);
+    v0[hh] = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col);
+    v1[hh] = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col + 8);
   }
-  for (int t = 0; t < n_tiles; ++t) {
+  const int n_tiles_st = (n_tiles + NT - 1) / NT;   // staged (NT-wide) tiles
+  for (int ts = 0; ts < n_tiles_st; ++ts) {
+    __syncthreads();
+#pragma unroll
+    for (int hh = 0; hh < NT; ++hh) {
+      *(ushortx8*)(&k_rm[rm_off(32 * hh + st_row, st_col / 8)]) = k0[hh];
+      *(ushortx8*)(&k_rm[rm_off(32 * hh + st_row, st_col / 8 + 1)]) = k1[hh];
+      *(ushortx8*)(&v_rm[rm_off(32 * hh + st_row, st_col / 8)]) = v0[hh];
+      *(ushortx8*)(&v_rm[rm_off(32 * hh + st_row, st_col / 8 + 1)]) = v1[hh];
+      *(ushortx8*)(&k_img[hh * 8 * TRKEY4 + tr_img_off(st_row, st_col)]) =
+          k0[hh];
+      *(ushortx8*)(&k_img[hh * 8 * TRKEY4 +
+                          tr_img_off(st_row, st_col + 8)]) = k1[hh];
+    }
+    __syncthreads();
+    if (ts + 1 < n_tiles_st) {
+#pragma unroll
+      for (int hh = 0; hh < NT; ++hh) {
+        const int krow_g = min((ts + 1) * NT * BLOCK + 32 * hh + st_row,
+                               S - 1);
+        k0[hh] = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
+        k1[hh] = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col + 8);
+        v0[hh] = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col);
+        v1[hh] = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col + 8);
+      }
+    }
+
+#pragma unroll
+    for (int h = 0; h < NT; ++h) {
+    const int t = ts * NT + h;
+    if (t >= n_tiles || t > my_last_tile || q0 >= S) continue;
     const int n0 = t * BLOCK;
-    __syncthreads();
-    {
-      *(ushortx8*)(&k_rm[rm_off(st_row, st_col / 8)]) = k0;
-      *(ushortx8*)(&k_rm[rm_off(st_row, st_col / 8 + 1)]) = k1;
-      *(ushortx8*)(&v_rm[rm_off(st_row, st_col / 8)]) = v0;
-      *(ushortx8*)(&v_rm[rm_off(st_row, st_col / 8 + 1)]) = v1;
-      *(ushortx8*)(&k_img[tr_img_off(st_row, st_col)]) = k0;
-      *(ushortx8*)(&k_img[tr_img_off(st_row, st_col + 8)]) = k1;
-    }
-    __syncthreads();
-    if (t + 1 < n_tiles) {
-      const int krow_g = min((t + 1) * BLOCK + st_row, S - 1);
-      k0 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col);
-      k1 = *(const ushortx8*)(Kp + (long)krow_g * ks + st_col + 8);
-      v0 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col);
-      v1 = *(const ushortx8*)(Vp + (long)krow_g * ks + st_col + 8);
-    }
-    if (t > my_last_tile || q0 >= S) continue;
 
     // S^T and dP^T (C: n = q lane-local, m = key)
     floatx16 st_acc = {}, dp_acc = {};
@@ -214,9 +231,11 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     for (int stp = 0; stp < 8; ++stp) {
       const int blk = 2 * stp + (lane >> 5);
       bf16x8 kf = __builtin_bit_cast(
-          bf16x8, *(const ushortx8*)(&k_rm[rm_off(lane & 31, blk)]));
+          bf16x8,
+          *(const ushortx8*)(&k_rm[rm_off(32 * h + (lane & 31), blk)]));
       bf16x8 vf = __builtin_bit_cast(
-          bf16x8, *(const ushortx8*)(&v_rm[rm_off(lane & 31, blk)]));
+          bf16x8,
+          *(const ushortx8*)(&v_rm[rm_off(32 * h + (lane & 31), blk)]));
       st_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[stp], st_acc,
                                                        0, 0, 0);
       dp_acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dof[stp], dp_acc,
@@ -259,7 +278,8 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     // from the row-major subtiled K image; dt+1's reads pipeline behind
     // dt's MFMAs — only our tr ops are on lgkm in this phase)
     {
-      const unsigned kbase = tr16_lane_base(k_img, lane);
+      const unsigned kbase =
+          tr16_lane_base(k_img + h * 8 * TRKEY4, lane);
       ushortx4_tr vrA[2][2], vrB[2][2];
       __builtin_amdgcn_s_setprio(1);
       tr16_issue_dt<1, 0>(kbase, vrA);
@@ -286,6 +306,7 @@ extern "C" __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
       }
       __builtin_amdgcn_s_setprio(0);
     }
+    }  // NT sub-tile loop
   }
 
   // epilogue: write dQ (C: m = q reg-spread, n = d lane)
@@ -776,12 +797,25 @@ extern "C" void launch_attn_bwd_dq(const void* q, const void* k, const void* v,
                                    long gs, float scale,
                                    int causal, void* stream) {
   dim3 grid((S + 127) / 128, Hq, B);
-  hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, dim3(256), 0,
-                     (hipStream_t)stream, (const ushort_t*)q,
-                     (const ushort_t*)k, (const ushort_t*)v,
-                     (const ushort_t*)dO, lse, delta, (ushort_t*)dq, S, Hq,
-                     Hk, qb, qh, qs, kb, kh, ks, ob, oh, os, gb, gh, gs,
-                     scale, causal);
+  static const int dq_nt = [] {
+    const char* e = getenv("ANTRAY_BWD_DQ_NT");
+    return e ? atoi(e) : 1;
+  }();
+  if (dq_nt >= 2) {
+    hipLaunchKernelGGL(attn_bwd_dq_kernel<2>, grid, dim3(256), 0,
+                       (hipStream_t)stream, (const ushort_t*)q,
+                       (const ushort_t*)k, (const ushort_t*)v,
+                       (const ushort_t*)dO, lse, delta, (ushort_t*)dq, S, Hq,
+                       Hk, qb, qh, qs, kb, kh, ks, ob, oh, os, gb, gh, gs,
+                       scale, causal);
+  } else {
+    hipLaunchKernelGGL(attn_bwd_dq_kernel<1>, grid, dim3(256), 0,
+                       (hipStream_t)stream, (const ushort_t*)q,
+                       (const ushort_t*)k, (const ushort_t*)v,
+                       (const ushort_t*)dO, lse, delta, (ushort_t*)dq, S, Hq,
+                       Hk, qb, qh, qs, kb, kh, ks, ob, oh, os, gb, gh, gs,
+                       scale, causal);
+  }
 }
 
 extern "C" void launch_attn_bwd_dkv(const void* q, const void* k,
