@@ -42,7 +42,18 @@ def rebatch_blocks(blocks: Iterator[Any], batch_size: Optional[int],
         yield BlockAccessor(table).to_batch(batch_format)
 
 
-_pinned_pool: Dict[int, Any] = {}
+import threading as _threading
+
+_pinned_pool_tls = _threading.local()
+
+
+def _pinned_pool() -> Dict[int, Any]:
+    # per-THREAD pools: two iterator streams in one process must not
+    # rotate through the same buffers (the event fencing is per-pair)
+    pool = getattr(_pinned_pool_tls, "pool", None)
+    if pool is None:
+        pool = _pinned_pool_tls.pool = {}
+    return pool
 
 
 def _pinned_staging(t):
@@ -56,10 +67,11 @@ def _pinned_staging(t):
 
     nbytes = t.numel() * t.element_size()
     size_class = 1 << max(nbytes - 1, 1).bit_length()
-    pair = _pinned_pool.get(size_class)
+    pool = _pinned_pool()
+    pair = pool.get(size_class)
     if pair is None:
         pair = {"bufs": [None, None], "events": [None, None], "next": 0}
-        _pinned_pool[size_class] = pair
+        pool[size_class] = pair
     i = pair["next"]
     pair["next"] = 1 - i
     buf = pair["bufs"][i]
