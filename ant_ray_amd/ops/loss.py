@@ -58,11 +58,15 @@ def linear_cross_entropy(
     weight: torch.Tensor,
     targets: torch.Tensor,
     ignore_index: int = -100,
-    chunk_rows: int = 4096,
+    chunk_rows: int = 0,
 ) -> torch.Tensor:
     """mean CE over valid targets of (x @ weight.T) without materializing the
     full logits. x: [N,H] bf16; weight: [V,H] bf16; targets: [N] int."""
     if x.is_cuda:
+        if chunk_rows <= 0:
+            import os
+
+            chunk_rows = int(os.environ.get("ANTRAY_CE_CHUNK", "4096"))
         return _LinearCrossEntropyFn.apply(x, weight, targets, ignore_index, chunk_rows)
     logits = x.float() @ weight.float().t()
     return torch.nn.functional.cross_entropy(
