@@ -48,6 +48,11 @@ def setup_tunableop(local_rank: int = 0):
                          f"/tmp/tunableop_rank{local_rank}.csv")
     tun.set_filename(out if tuning else table)
     if tuning:
+        if os.path.exists(table):
+            try:
+                tun.read_file(table)  # only UNSEEN shapes get tuned
+            except Exception:
+                pass
         tun.tuning_enable(True)
         tun.set_max_tuning_duration(100)
     else:
