@@ -35,6 +35,10 @@ from ant_ray_amd.rllib.impala import (  # noqa: F401
 from ant_ray_amd.rllib.offline import (  # noqa: F401
     BC,
     BCConfig,
+    CQL,
+    CQLConfig,
+    IQL,
+    IQLConfig,
     MARWIL,
     MARWILConfig,
     load_episodes,

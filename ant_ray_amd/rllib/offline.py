@@ -204,3 +204,273 @@ class MARWIL(BC):
     """BC weighted by exp(beta * advantage) with a learned baseline."""
 
     beta = None  # use cfg.beta
+
+
+def _flatten_transitions(episodes: EpisodeData):
+    """(obs, act, rew, next_obs, done) tensors from episode dicts —
+    next_obs within an episode; the final step is terminal."""
+    O, A, R, NO, D = [], [], [], [], []
+    for e in episodes:
+        o = e["obs"]
+        T = len(o)
+        O.append(o)
+        A.append(e["actions"])
+        R.append(e["rewards"])
+        no = np.concatenate([o[1:], o[-1:]], axis=0)
+        NO.append(no)
+        d = np.zeros(T, dtype=np.float32)
+        d[-1] = 1.0
+        D.append(d)
+    return (torch.from_numpy(np.concatenate(O)),
+            torch.from_numpy(np.concatenate(A)),
+            torch.from_numpy(np.concatenate(R)),
+            torch.from_numpy(np.concatenate(NO)),
+            torch.from_numpy(np.concatenate(D)))
+
+
+class _DiscreteQ(torch.nn.Module):
+    def __init__(self, obs_dim: int, act_dim: int, hidden: int = 128):
+        super().__init__()
+        self.net = torch.nn.Sequential(
+            torch.nn.Linear(obs_dim, hidden), torch.nn.Tanh(),
+            torch.nn.Linear(hidden, hidden), torch.nn.Tanh(),
+            torch.nn.Linear(hidden, act_dim))
+
+    def forward(self, obs):
+        return self.net(obs)
+
+
+class CQLConfig(_OfflineConfig):
+    """cql_alpha weighs the conservative regularizer
+    (logsumexp_a Q(s,a) - Q(s, a_data))."""
+
+    def __init__(self):
+        super().__init__()
+        self.cql_alpha = 1.0
+        self.tau = 0.01
+
+    def training(self, lr=1e-3, gamma=0.99, train_batch_size=256,
+                 updates_per_iteration=64, beta=1.0, cql_alpha=1.0,
+                 tau=0.01):
+        super().training(lr, gamma, train_batch_size,
+                         updates_per_iteration, beta)
+        self.cql_alpha = cql_alpha
+        self.tau = tau
+        return self
+
+    def build(self) -> "CQL":
+        return CQL(self)
+
+
+class CQL:
+    """Conservative Q-Learning, discrete-action variant.
+
+    Role parity: reference rllib/algorithms/cql/ (continuous SAC-CQL;
+    this build's testbed envs are discrete, so the Q-learning body is
+    DQN-style). TD loss on the offline transitions plus the CQL
+    penalty that pushes down out-of-distribution action values:
+    alpha * E[logsumexp_a Q(s,a) - Q(s, a_data)]. Acting = argmax Q.
+    """
+
+    def __init__(self, cfg: CQLConfig):
+        assert cfg.data is not None, "call .offline_data() first"
+        episodes = (load_episodes(cfg.data) if isinstance(cfg.data, str)
+                    else cfg.data)
+        self.cfg = cfg
+        obs_dim = cfg.obs_dim or episodes[0]["obs"].shape[1]
+        act_dim = cfg.act_dim or int(
+            max(e["actions"].max() for e in episodes) + 1)
+        self.q = _DiscreteQ(obs_dim, act_dim)
+        self.q_t = _DiscreteQ(obs_dim, act_dim)
+        self.q_t.load_state_dict(self.q.state_dict())
+        self.opt = torch.optim.Adam(self.q.parameters(), lr=cfg.lr)
+        (self.obs, self.acts, self.rews, self.next_obs,
+         self.dones) = _flatten_transitions(episodes)
+        self.iteration = 0
+
+    def train(self) -> Dict:
+        n = len(self.obs)
+        stats = {}
+        for _ in range(self.cfg.updates_per_iter):
+            idx = torch.randint(0, n, (self.cfg.batch,))
+            qs = self.q(self.obs[idx])
+            q_data = qs.gather(1, self.acts[idx].unsqueeze(1)).squeeze(1)
+            with torch.no_grad():
+                q_next = self.q_t(self.next_obs[idx]).max(dim=1).values
+                target = (self.rews[idx]
+                          + self.cfg.gamma * (1 - self.dones[idx]) * q_next)
+            td = (q_data - target).pow(2).mean()
+            cql = (torch.logsumexp(qs, dim=1) - q_data).mean()
+            loss = td + self.cfg.cql_alpha * cql
+            self.opt.zero_grad()
+            loss.backward()
+            self.opt.step()
+            with torch.no_grad():
+                for p, pt in zip(self.q.parameters(),
+                                 self.q_t.parameters()):
+                    pt.mul_(1 - self.cfg.tau).add_(self.cfg.tau * p)
+            stats = {"td_loss": float(td), "cql_loss": float(cql)}
+        self.iteration += 1
+        return {"training_iteration": self.iteration, **stats}
+
+    def compute_single_action(self, obs, explore: bool = False) -> int:
+        with torch.no_grad():
+            return int(self.q(torch.from_numpy(
+                np.asarray(obs, dtype=np.float32)).unsqueeze(0)).argmax())
+
+    evaluate = BC.evaluate
+
+    def save(self, path: str) -> str:
+        import os
+
+        os.makedirs(path, exist_ok=True)
+        torch.save({"q": self.q.state_dict(),
+                    "iteration": self.iteration},
+                   os.path.join(path, "algorithm_state.pt"))
+        return path
+
+    def restore(self, path: str):
+        import os
+
+        st = torch.load(os.path.join(path, "algorithm_state.pt"),
+                        weights_only=False)
+        self.q.load_state_dict(st["q"])
+        self.q_t.load_state_dict(st["q"])
+        self.iteration = st["iteration"]
+
+    def stop(self):
+        pass
+
+
+class IQLConfig(_OfflineConfig):
+    """expectile: the V-regression expectile (0.5 = mean, ->1 approaches
+    max_a Q — the in-distribution optimal); awr_beta: advantage-weighted
+    regression temperature for policy extraction."""
+
+    def __init__(self):
+        super().__init__()
+        self.expectile = 0.8
+        self.awr_beta = 3.0
+        self.tau = 0.01
+
+    def training(self, lr=1e-3, gamma=0.99, train_batch_size=256,
+                 updates_per_iteration=64, beta=1.0, expectile=0.8,
+                 awr_beta=3.0, tau=0.01):
+        super().training(lr, gamma, train_batch_size,
+                         updates_per_iteration, beta)
+        self.expectile = expectile
+        self.awr_beta = awr_beta
+        self.tau = tau
+        return self
+
+    def build(self) -> "IQL":
+        return IQL(self)
+
+
+class IQL:
+    """Implicit Q-Learning, discrete-action variant.
+
+    Role parity: reference rllib/algorithms (IQL). Three learners, all
+    strictly in-distribution (no action sampled outside the dataset):
+      V  <- expectile regression towards Q_target(s, a_data)
+      Q  <- TD towards r + gamma * V(s')
+      pi <- advantage-weighted BC with w = exp(awr_beta * (Q - V))
+    Acting = argmax pi.
+    """
+
+    def __init__(self, cfg: IQLConfig):
+        assert cfg.data is not None, "call .offline_data() first"
+        episodes = (load_episodes(cfg.data) if isinstance(cfg.data, str)
+                    else cfg.data)
+        self.cfg = cfg
+        obs_dim = cfg.obs_dim or episodes[0]["obs"].shape[1]
+        act_dim = cfg.act_dim or int(
+            max(e["actions"].max() for e in episodes) + 1)
+        self.q = _DiscreteQ(obs_dim, act_dim)
+        self.q_t = _DiscreteQ(obs_dim, act_dim)
+        self.q_t.load_state_dict(self.q.state_dict())
+        self.v = torch.nn.Sequential(
+            torch.nn.Linear(obs_dim, 128), torch.nn.Tanh(),
+            torch.nn.Linear(128, 128), torch.nn.Tanh(),
+            torch.nn.Linear(128, 1))
+        self.policy = _MLPPolicy(obs_dim, act_dim)
+        self.opt_q = torch.optim.Adam(self.q.parameters(), lr=cfg.lr)
+        self.opt_v = torch.optim.Adam(self.v.parameters(), lr=cfg.lr)
+        self.opt_pi = torch.optim.Adam(self.policy.parameters(), lr=cfg.lr)
+        (self.obs, self.acts, self.rews, self.next_obs,
+         self.dones) = _flatten_transitions(episodes)
+        self.iteration = 0
+
+    def train(self) -> Dict:
+        n = len(self.obs)
+        stats = {}
+        for _ in range(self.cfg.updates_per_iter):
+            idx = torch.randint(0, n, (self.cfg.batch,))
+            obs, acts = self.obs[idx], self.acts[idx]
+            with torch.no_grad():
+                q_t = self.q_t(obs).gather(
+                    1, acts.unsqueeze(1)).squeeze(1)
+            v = self.v(obs).squeeze(1)
+            diff = q_t - v
+            ex = self.cfg.expectile
+            v_loss = (torch.where(diff > 0, ex, 1 - ex)
+                      * diff.pow(2)).mean()
+            self.opt_v.zero_grad()
+            v_loss.backward()
+            self.opt_v.step()
+
+            with torch.no_grad():
+                v_next = self.v(self.next_obs[idx]).squeeze(1)
+                target = (self.rews[idx]
+                          + self.cfg.gamma * (1 - self.dones[idx]) * v_next)
+            q = self.q(obs).gather(1, acts.unsqueeze(1)).squeeze(1)
+            q_loss = (q - target).pow(2).mean()
+            self.opt_q.zero_grad()
+            q_loss.backward()
+            self.opt_q.step()
+
+            with torch.no_grad():
+                adv = (self.q_t(obs).gather(1, acts.unsqueeze(1)).squeeze(1)
+                       - self.v(obs).squeeze(1))
+                w = torch.exp(self.cfg.awr_beta * adv).clamp(max=100.0)
+            logits, _ = self.policy(obs)
+            logp = torch.distributions.Categorical(
+                logits=logits).log_prob(acts)
+            pi_loss = -(w * logp).mean()
+            self.opt_pi.zero_grad()
+            pi_loss.backward()
+            self.opt_pi.step()
+
+            with torch.no_grad():
+                for p, pt in zip(self.q.parameters(),
+                                 self.q_t.parameters()):
+                    pt.mul_(1 - self.cfg.tau).add_(self.cfg.tau * p)
+            stats = {"v_loss": float(v_loss), "q_loss": float(q_loss),
+                     "pi_loss": float(pi_loss)}
+        self.iteration += 1
+        return {"training_iteration": self.iteration, **stats}
+
+    compute_single_action = BC.compute_single_action
+    evaluate = BC.evaluate
+
+    def save(self, path: str) -> str:
+        import os
+
+        os.makedirs(path, exist_ok=True)
+        torch.save({"policy": self.policy.state_dict(),
+                    "q": self.q.state_dict(),
+                    "iteration": self.iteration},
+                   os.path.join(path, "algorithm_state.pt"))
+        return path
+
+    def restore(self, path: str):
+        import os
+
+        st = torch.load(os.path.join(path, "algorithm_state.pt"),
+                        weights_only=False)
+        self.policy.load_state_dict(st["policy"])
+        self.q.load_state_dict(st["q"])
+        self.iteration = st["iteration"]
+
+    def stop(self):
+        pass
