@@ -96,10 +96,92 @@ def build_llm_deployment(config: LLMConfig):
 
 
 def build_openai_app(configs: List[LLMConfig]):
-    _require_vllm()
-    raise NotImplementedError(
-        "OpenAI-compatible ingress requires vllm; see reference "
-        "llm/_internal/serve/core/ingress/")
+    """OpenAI-compatible ingress (parity: reference llm/_internal/serve/
+    core/ingress/ — /v1/models + /v1/completions). In-tree model ids are
+    served by the NATIVE engine; `prompt` may be a string (byte-level
+    fallback tokenization — the in-tree models run random-init weights,
+    there is no pretrained tokenizer in this air-gapped image) or a list
+    of token ids; the response carries both `text` (byte-rendered) and
+    `token_ids`. Other model ids require vllm (as the reference)."""
+    native = [c for c in configs if c.model_id.lower() in _NATIVE_MODELS]
+    if not native:
+        _require_vllm()
+        raise NotImplementedError(
+            "OpenAI ingress for non-native models requires vllm; see "
+            "reference llm/_internal/serve/core/ingress/")
+    from ant_ray_amd import serve
+
+    cfg = native[0]
+
+    def _build_api():
+        # built INSIDE the replica (a function-local FastAPI app cannot
+        # survive cloudpickle — starlette State recursion); route
+        # functions reach the replica through app.state.serve_self
+        from fastapi import FastAPI
+
+        api = FastAPI()
+
+        @api.get("/v1/models")
+        async def models():  # noqa: ANN202
+            replica = api.state.serve_self
+            return {"object": "list",
+                    "data": [{"id": replica.model_id, "object": "model",
+                              "owned_by": "ant-ray-amd"}]}
+
+        @api.post("/v1/completions")
+        async def completions(body: dict):  # noqa: ANN202
+            import asyncio
+            import time as _time
+
+            replica = api.state.serve_self
+            prompt = body.get("prompt", "")
+            if isinstance(prompt, str):
+                ids = [b % replica.engine.vocab
+                       for b in prompt.encode("utf-8")] or [0]
+            else:
+                ids = [int(t) % replica.engine.vocab for t in prompt] or [0]
+            max_tokens = int(body.get("max_tokens", 16))
+            temperature = float(body.get("temperature", 0.0))
+            loop = asyncio.get_running_loop()
+            out = await loop.run_in_executor(
+                None, lambda: replica.engine.generate_tokens(
+                    [ids], max_tokens, temperature)[0])
+            text = bytes(t % 256 for t in out).decode("utf-8",
+                                                      errors="replace")
+            return {
+                "id": f"cmpl-{int(_time.time() * 1e6):x}",
+                "object": "text_completion",
+                "created": int(_time.time()),
+                "model": replica.model_id,
+                "choices": [{"index": 0, "text": text, "token_ids": out,
+                             "finish_reason": "length"}],
+                "usage": {"prompt_tokens": len(ids),
+                          "completion_tokens": len(out),
+                          "total_tokens": len(ids) + len(out)},
+            }
+
+        return api
+
+    @serve.deployment(
+        num_replicas=cfg.deployment_config.get("num_replicas", 1),
+        ray_actor_options={"num_gpus": cfg.num_gpus_per_replica()},
+    )
+    @serve.ingress(_build_api)
+    class OpenAIIngress:
+        def __init__(self):
+            import torch
+
+            from ant_ray_amd.llm.native_engine import NativeLLMEngine
+
+            device = ("cuda" if (cfg.num_gpus_per_replica() > 0
+                                 and torch.cuda.is_available()) else "cpu")
+            self.engine = NativeLLMEngine(
+                cfg.model_id.lower(),
+                max_seq=int(cfg.engine_kwargs.get("max_model_len", 4096)),
+                device=device)
+            self.model_id = cfg.model_id
+
+    return OpenAIIngress.bind()
 
 
 def build_llm_processor(config: LLMConfig, preprocess=None, postprocess=None):

@@ -114,14 +114,25 @@ def deployment(_target=None, **kwargs):
 
 
 def ingress(asgi_app):
-    """@serve.ingress(fastapi_app): class decorator mounting the deployment's
-    methods behind a FastAPI/starlette app (parity serve/api.py ingress)."""
+    """@serve.ingress(fastapi_app_or_factory): class decorator mounting the
+    deployment's methods behind a FastAPI/starlette app (parity
+    serve/api.py ingress). A zero-arg FACTORY may be passed instead of an
+    app: it is called lazily inside the replica — required when the app
+    is built in a function scope (a function-local FastAPI app pickles by
+    VALUE, and starlette State's __getattr__ recurses infinitely during
+    unpickling)."""
 
     def wrap(cls):
         class ASGIWrapped(cls):
             __name__ = cls.__name__
 
             async def __call__(self, request):
+                app = getattr(self, "_resolved_asgi_app", None)
+                if app is None:
+                    app = asgi_app
+                    if callable(app) and not hasattr(app, "router"):
+                        app = app()  # factory
+                    self._resolved_asgi_app = app
                 # run one request through the ASGI app
                 scope = dict(request.scope)
                 body = await request.body()
@@ -138,8 +149,8 @@ def ingress(asgi_app):
                     sent.append(msg)
 
                 # make `self` reachable from route functions via app state
-                asgi_app.state.serve_self = self
-                await asgi_app(scope, receive, send)
+                app.state.serve_self = self
+                await app(scope, receive, send)
                 status = 200
                 headers: List = []
                 chunks = []

@@ -48,3 +48,46 @@ def test_serve_native_llm_deployment(ray_cpu):
     outs = [r.result(timeout_s=300) for r in reqs]
     assert all(len(o["token_ids"]) == 5 for o in outs)
     serve.shutdown()
+
+
+def test_openai_ingress_native_cpu(ray_cpu):
+    """OpenAI-compatible /v1/completions over the native engine (CPU,
+    tiny model): request/response shapes mirror the OpenAI API."""
+    import urllib.request
+    import json as _json
+
+    from ant_ray_amd import serve
+    from ant_ray_amd.llm import LLMConfig, build_openai_app
+
+    app = build_openai_app([LLMConfig(
+        model_loading_config={"model_id": "llama-tiny-d128"},
+        engine_kwargs={"max_model_len": 128,
+                       "tensor_parallel_size": 0},  # 0 GPUs -> CPU engine
+        deployment_config={"num_replicas": 1},
+    )])
+    serve.run(app, name="oai", route_prefix="/")
+    try:
+        base = "http://127.0.0.1:8000"
+        with urllib.request.urlopen(base + "/v1/models", timeout=30) as r:
+            models = _json.load(r)
+        assert models["data"][0]["id"] == "llama-tiny-d128"
+        body = _json.dumps({"prompt": "hello world", "max_tokens": 4,
+                            "temperature": 0.0}).encode()
+        req = urllib.request.Request(
+            base + "/v1/completions", data=body,
+            headers={"Content-Type": "application/json"})
+        with urllib.request.urlopen(req, timeout=120) as r:
+            out = _json.load(r)
+        assert out["object"] == "text_completion"
+        assert len(out["choices"][0]["token_ids"]) == 4
+        assert out["usage"]["completion_tokens"] == 4
+        # token-ids prompt form
+        body = _json.dumps({"prompt": [1, 2, 3, 4], "max_tokens": 3}).encode()
+        req = urllib.request.Request(
+            base + "/v1/completions", data=body,
+            headers={"Content-Type": "application/json"})
+        with urllib.request.urlopen(req, timeout=120) as r:
+            out = _json.load(r)
+        assert len(out["choices"][0]["token_ids"]) == 3
+    finally:
+        serve.shutdown()
