@@ -332,6 +332,222 @@ class Dataset:
         for batch in self.iter_batches(batch_size=batch_size, drop_last=drop_last):
             yield to_torch_batch(batch, dtypes, device, collate_fn)
 
+    # ------------------------------------------------- aggregates / stats
+    def _column_array(self, on: str):
+        import pyarrow as pa
+
+        blocks = list(self.iter_blocks())
+        if not blocks:
+            return None
+        t = pa.concat_tables(blocks, promote_options="default")
+        return t.column(on).combine_chunks()
+
+    def aggregate(self, *aggs):
+        """Run AggregateFn-style aggregations (parity Dataset.aggregate);
+        also accepts ("name", col) pairs for sum/min/max/mean/std."""
+        out = {}
+        for a in aggs:
+            if isinstance(a, tuple) and len(a) == 2:
+                name, col = a
+                out[f"{name}({col})"] = getattr(self, name)(col)
+            else:  # AggregateFn-like: init/accumulate_block/merge/name
+                acc = a.init(None)
+                for b in self.iter_blocks():
+                    acc = a.merge(acc, a.accumulate_block(a.init(None), b))
+                out[getattr(a, "name", repr(a))] = (
+                    a.finalize(acc) if hasattr(a, "finalize") else acc)
+        return out
+
+    def sum(self, on: str):
+        import pyarrow.compute as pc
+
+        arr = self._column_array(on)
+        return None if arr is None else pc.sum(arr).as_py()
+
+    def min(self, on: str):
+        import pyarrow.compute as pc
+
+        arr = self._column_array(on)
+        return None if arr is None else pc.min(arr).as_py()
+
+    def max(self, on: str):
+        import pyarrow.compute as pc
+
+        arr = self._column_array(on)
+        return None if arr is None else pc.max(arr).as_py()
+
+    def mean(self, on: str):
+        import pyarrow.compute as pc
+
+        arr = self._column_array(on)
+        return None if arr is None else pc.mean(arr).as_py()
+
+    def std(self, on: str, ddof: int = 1):
+        import pyarrow.compute as pc
+
+        arr = self._column_array(on)
+        return None if arr is None else pc.stddev(arr, ddof=ddof).as_py()
+
+    # ---------------------------------------------------- splits / order
+    def split_at_indices(self, indices: List[int]) -> List["Dataset"]:
+        """Row-exact split (parity Dataset.split_at_indices)."""
+        import pyarrow as pa
+
+        blocks = list(self.iter_blocks())
+        table = (pa.concat_tables(blocks, promote_options="default")
+                 if blocks else pa.table({}))
+        bounds = [0] + list(indices) + [table.num_rows]
+        out = []
+        import ant_ray_amd as ray
+
+        for s0, e0 in zip(bounds[:-1], bounds[1:]):
+            piece = table.slice(s0, max(0, e0 - s0))
+            out.append(MaterializedDataset([ray.put(piece)]))
+        return out
+
+    def split_proportionately(self, proportions: List[float]) -> List["Dataset"]:
+        n = self.count()
+        idx, acc = [], 0.0
+        for p in proportions:
+            acc += p
+            idx.append(int(n * acc))
+        return self.split_at_indices(idx)
+
+    def train_test_split(self, test_size: float, *, shuffle: bool = False,
+                         seed=None) -> List["Dataset"]:
+        ds = self.random_shuffle(seed=seed) if shuffle else self
+        return ds.split_proportionately([1.0 - test_size])
+
+    streaming_train_test_split = train_test_split
+
+    def randomize_block_order(self, *, seed: Optional[int] = None) -> "Dataset":
+        import random as _random
+
+        def _shuf(refs: List[Any]) -> List[Any]:
+            refs = list(refs)
+            _random.Random(seed).shuffle(refs)
+            return refs
+
+        return self._with(AllToAllOp(name="RandomizeBlockOrder", fn=_shuf))
+
+    def with_column(self, name: str, expr) -> "Dataset":
+        return self.with_columns({name: expr})
+
+    # -------------------------------------------------- metadata / plumbing
+    def name(self) -> Optional[str]:
+        return getattr(self, "_name", None)
+
+    def set_name(self, name: Optional[str]):
+        self._name = name
+
+    def get_dataset_id(self) -> str:
+        if not hasattr(self, "_dataset_id"):
+            import uuid
+
+            self._dataset_id = uuid.uuid4().hex
+        return self._dataset_id
+
+    def context(self):
+        from ant_ray_amd.data.context import DataContext
+
+        return DataContext.get_current()
+
+    def copy(self) -> "Dataset":
+        return Dataset(list(self._ops))
+
+    def explain(self) -> str:
+        """Logical plan rendering (parity Dataset.explain)."""
+        lines = ["Execution plan:"]
+        for op in self._ops:
+            lines.append(f"  {type(op).__name__}: {getattr(op, 'name', '')}")
+        return "\n".join(lines)
+
+    def names(self) -> List[str]:
+        return self.columns()
+
+    def types(self) -> List[Any]:
+        sch = self.schema()
+        return list(getattr(sch, "types", []))
+
+    def input_files(self) -> List[str]:
+        op = self._ops[0] if self._ops else None
+        return list(getattr(op, "input_files", []) or [])
+
+    def get_internal_block_refs(self) -> List[Any]:
+        return list(self.iter_internal_ref_bundles())
+
+    def iterator(self):
+        """DataIterator over this dataset (parity Dataset.iterator)."""
+        return self.split(1)[0]
+
+    def to_numpy_refs(self) -> List[Any]:
+        import ant_ray_amd as ray
+
+        out = []
+        for b in self.iter_blocks():
+            out.append(ray.put(
+                {c: b.column(c).combine_chunks().to_numpy(zero_copy_only=False)
+                 for c in b.column_names}))
+        return out
+
+    def to_pandas_refs(self) -> List[Any]:
+        import ant_ray_amd as ray
+
+        return [ray.put(b.to_pandas()) for b in self.iter_blocks()]
+
+    def write_numpy(self, path: str, *, column: str = "data", **_):
+        """Write each block's `column` as an .npy file under path."""
+        import os
+
+        import numpy as np
+
+        os.makedirs(path, exist_ok=True)
+        for i, b in enumerate(self.iter_blocks()):
+            arr = b.column(column).combine_chunks().to_numpy(
+                zero_copy_only=False)
+            np.save(os.path.join(path, f"block_{i:05d}.npy"), arr)
+
+    def write_sql(self, sql: str, connection_factory, **_):
+        """INSERT each row via DB-API (parity Dataset.write_sql; pairs
+        with read_sql). `sql` must be an INSERT with placeholders."""
+        conn = connection_factory()
+        cur = conn.cursor()
+        for b in self.iter_blocks():
+            cols = b.column_names
+            for row in b.to_pylist():
+                cur.execute(sql, tuple(row[c] for c in cols))
+        conn.commit()
+
+    def _no_lib(self, lib: str):
+        raise ImportError(
+            f"Dataset conversion requires {lib}, which is not installed in "
+            f"this air-gapped image (reference parity: the method exists "
+            f"and delegates to {lib}).")
+
+    def to_tf(self, *a, **k):
+        self._no_lib("tensorflow")
+
+    def iter_tf_batches(self, *a, **k):
+        self._no_lib("tensorflow")
+
+    def to_dask(self, *a, **k):
+        self._no_lib("dask")
+
+    def to_modin(self, *a, **k):
+        self._no_lib("modin")
+
+    def to_mars(self, *a, **k):
+        self._no_lib("mars")
+
+    def to_spark(self, *a, **k):
+        self._no_lib("pyspark")
+
+    def to_daft(self, *a, **k):
+        self._no_lib("daft")
+
+    def to_random_access_dataset(self, *a, **k):
+        self._no_lib("random-access datasets (reference experimental)")
+
     def to_pandas(self, limit: Optional[int] = None):
         import pandas as pd
 

@@ -418,3 +418,23 @@ def test_backpressure_policies():
     mem._last_poll = __import__("time").monotonic() + 100  # pin the poll
     assert mem.can_add_input("s", 0)  # always drainable
     assert not mem.can_add_input("s", 3)
+
+
+def test_dataset_aggregates_and_splits(ray_mod):
+    import ant_ray_amd.data as data
+
+    ds = data.from_items([{"x": i, "y": float(i) * 2} for i in range(100)])
+    assert ds.sum("x") == 4950
+    assert abs(ds.mean("y") - 99.0) < 1e-6
+    assert ds.max("x") == 99 and ds.min("x") == 0
+    assert ds.std("x") > 0
+    parts = ds.split_at_indices([30, 60])
+    assert [p.count() for p in parts] == [30, 30, 40]
+    tr, te = ds.train_test_split(0.2)
+    assert (tr.count(), te.count()) == (80, 20)
+    assert ds.randomize_block_order(seed=1).count() == 100
+    assert ds.aggregate(("sum", "x"))["sum(x)"] == 4950
+    assert "Execution plan" in ds.explain()
+    assert ds.names() == ds.columns()
+    ds.set_name("bench")
+    assert ds.name() == "bench"
