@@ -23,8 +23,12 @@ from ant_ray_amd.tune.schedulers import (  # noqa: F401
     STOP,
     ASHAScheduler,
     FIFOScheduler,
+    HyperBandForBOHB,
+    HyperBandScheduler,
     MedianStoppingRule,
+    PB2,
     PopulationBasedTraining,
+    ResourceChangingScheduler,
 )
 from ant_ray_amd.tune.search import (  # noqa: F401
     BasicVariantGenerator,
@@ -123,8 +127,10 @@ def get_context() -> TuneContext:
 __all__ = [
     "ASHAScheduler", "BasicVariantGenerator", "CLIReporter", "Callback",
     "CheckpointConfig", "Experiment", "ExperimentAnalysis", "FIFOScheduler",
-    "FailureConfig", "JupyterNotebookReporter", "MedianStoppingRule",
-    "PlacementGroupFactory", "ProgressReporter", "ResultGrid", "ResumeConfig",
+    "FailureConfig", "HyperBandForBOHB", "HyperBandScheduler",
+    "JupyterNotebookReporter", "MedianStoppingRule", "PB2",
+    "PlacementGroupFactory", "ProgressReporter",
+    "ResourceChangingScheduler", "ResultGrid", "ResumeConfig",
     "Stopper", "SyncConfig", "Trainable", "TuneConfig", "TuneContext",
     "TuneError", "Tuner", "choice", "create_scheduler", "create_searcher",
     "get_context", "grid_search", "lograndint", "loguniform", "qlograndint",

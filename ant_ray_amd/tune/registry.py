@@ -36,13 +36,20 @@ def get_env_creator(name: str):
 
 def create_scheduler(name: str, **kwargs):
     from ant_ray_amd.tune.schedulers import (ASHAScheduler, FIFOScheduler,
-                                             MedianStoppingRule,
-                                             PopulationBasedTraining)
+                                             HyperBandForBOHB,
+                                             HyperBandScheduler,
+                                             MedianStoppingRule, PB2,
+                                             PopulationBasedTraining,
+                                             ResourceChangingScheduler)
 
     table = {"fifo": FIFOScheduler, "asha": ASHAScheduler,
              "async_hyperband": ASHAScheduler,
+             "hyperband": HyperBandScheduler,
+             "hb_bohb": HyperBandForBOHB,
              "median_stopping_rule": MedianStoppingRule,
-             "pbt": PopulationBasedTraining}
+             "pbt": PopulationBasedTraining,
+             "pb2": PB2,
+             "resource_changing": ResourceChangingScheduler}
     if name not in table:
         raise ValueError(f"unknown scheduler {name!r}; "
                          f"available: {sorted(table)}")

@@ -204,3 +204,144 @@ class PopulationBasedTraining:
         if hasattr(space, "sample"):
             return space.sample(self._rng)
         return space
+
+
+class HyperBandScheduler:
+    """Classic (synchronous-bracket) HyperBand (reference tune/schedulers/
+    hyperband.py), adapted to the streaming on_trial_result protocol:
+    trials are assigned round-robin to brackets with different
+    (grace, reduction) trade-offs; inside a bracket, successive-halving
+    rungs cut the worst performers at each milestone."""
+
+    def __init__(self, metric: Optional[str] = None,
+                 mode: Optional[str] = None, max_t: int = 81,
+                 reduction_factor: int = 3,
+                 time_attr: str = "training_iteration"):
+        import math
+
+        self.metric, self.mode = metric, mode
+        self.max_t = max_t
+        self.rf = reduction_factor
+        self.time_attr = time_attr
+        s_max = int(math.log(max_t, reduction_factor))
+        # bracket k starts at grace rf^k
+        self._brackets = [ASHAScheduler(metric=metric, mode=mode,
+                                        max_t=max_t,
+                                        grace_period=reduction_factor ** k,
+                                        reduction_factor=reduction_factor,
+                                        time_attr=time_attr)
+                          for k in range(s_max + 1)]
+        self._assign: Dict[str, int] = {}
+        self._next = 0
+
+    def set_objective(self, metric, mode):
+        self.metric = self.metric or metric
+        self.mode = self.mode or mode
+        for b in self._brackets:
+            b.set_objective(self.metric, self.mode)
+
+    def on_trial_result(self, trial_id: str, result: dict) -> str:
+        if trial_id not in self._assign:
+            self._assign[trial_id] = self._next % len(self._brackets)
+            self._next += 1
+        return self._brackets[self._assign[trial_id]].on_trial_result(
+            trial_id, result)
+
+
+class HyperBandForBOHB(HyperBandScheduler):
+    """BOHB's bracket scheduler (reference tune/schedulers/hb_bohb.py) —
+    pair with TPESearch (the model-based half of BOHB) via
+    Tuner(search_alg=TPESearch(), scheduler=HyperBandForBOHB())."""
+
+
+class PB2(PopulationBasedTraining):
+    """PBT with GP-guided exploration (reference tune/schedulers/pb2.py):
+    instead of random multiply/resample mutations, continuous
+    hyperparameters of the exploited config are chosen by a GP-UCB fit
+    on (config -> latest score) observations. hyperparam_bounds maps
+    name -> (low, high)."""
+
+    def __init__(self, metric: Optional[str] = None,
+                 mode: Optional[str] = None, perturbation_interval: int = 4,
+                 hyperparam_bounds: Optional[Dict] = None,
+                 quantile_fraction: float = 0.25,
+                 time_attr: str = "training_iteration",
+                 seed: Optional[int] = None):
+        super().__init__(metric=metric, mode=mode,
+                         perturbation_interval=perturbation_interval,
+                         hyperparam_mutations={},
+                         quantile_fraction=quantile_fraction,
+                         time_attr=time_attr, seed=seed)
+        self.bounds = hyperparam_bounds or {}
+        self._obs: List = []  # (param-vector, score)
+
+    def on_trial_result(self, trial_id: str, result: dict) -> str:
+        v = result.get(self.metric)
+        if v is not None and trial_id in self._config:
+            cfg = self._config[trial_id]
+            vec = [float(cfg.get(k, (lo + hi) / 2))
+                   for k, (lo, hi) in self.bounds.items()]
+            if vec:
+                self._obs.append((vec, float(v)
+                                  * (1.0 if self.mode != "min" else -1.0)))
+        return super().on_trial_result(trial_id, result)
+
+    def exploit(self, trial_id: str):
+        base = super().exploit(trial_id)
+        if base is None or not self.bounds:
+            return base
+        path, cfg = base
+        cfg = dict(cfg)
+        new_vals = self._gp_ucb_suggest()
+        for k, val in new_vals.items():
+            cfg[k] = val
+        self._config[trial_id] = dict(cfg)
+        return path, cfg
+
+    def _gp_ucb_suggest(self) -> Dict[str, float]:
+        import numpy as np
+
+        keys = list(self.bounds)
+        lo = np.array([self.bounds[k][0] for k in keys])
+        hi = np.array([self.bounds[k][1] for k in keys])
+        rng = np.random.RandomState(self._rng.randrange(2 ** 31))
+        cand = rng.uniform(lo, hi, size=(64, len(keys)))
+        if len(self._obs) >= 3:
+            try:
+                from sklearn.gaussian_process import GaussianProcessRegressor
+                from sklearn.gaussian_process.kernels import Matern
+
+                X = np.array([o[0] for o in self._obs])
+                y = np.array([o[1] for o in self._obs])
+                y = (y - y.mean()) / (y.std() + 1e-9)
+                gp = GaussianProcessRegressor(
+                    kernel=Matern(nu=2.5), normalize_y=False,
+                    random_state=0).fit(X, y)
+                mu, sd = gp.predict(cand, return_std=True)
+                best = cand[int(np.argmax(mu + 1.0 * sd))]
+            except Exception:
+                best = cand[0]
+        else:
+            best = cand[0]
+        return {k: float(b) for k, b in zip(keys, best)}
+
+
+class ResourceChangingScheduler:
+    """Wrapper delegating trial decisions to a base scheduler while
+    exposing a resources_allocation_function hook (reference
+    tune/schedulers/resource_changing_scheduler.py; this build's trials
+    are single-process so reallocation is advisory)."""
+
+    def __init__(self, base_scheduler=None,
+                 resources_allocation_function=None):
+        self.base = base_scheduler or FIFOScheduler()
+        self.alloc_fn = resources_allocation_function
+
+    def set_objective(self, metric, mode):
+        self.base.set_objective(metric, mode)
+
+    def on_trial_result(self, trial_id: str, result: dict) -> str:
+        return self.base.on_trial_result(trial_id, result)
+
+    def __getattr__(self, item):
+        return getattr(self.base, item)

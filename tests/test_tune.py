@@ -459,3 +459,56 @@ def test_searcher_wrapper_paths():
     h = HyperOptSearch(space={"x": tune.uniform(0, 1)}, metric="m")
     assert 0 <= h.suggest("t2")["x"] <= 1
     assert type(tune.create_searcher("bayesopt")).__name__ == "BayesOptSearch"
+
+
+def test_hyperband_scheduler_prunes(ray_mod):
+    """Protocol-level: brackets assign round-robin; a bad trial arriving
+    at a rung AFTER its peers is cut, a good one survives to max_t."""
+    from ant_ray_amd.tune.schedulers import (CONTINUE, STOP,
+                                             HyperBandScheduler)
+
+    s = HyperBandScheduler(metric="score", mode="max", max_t=27,
+                           reduction_factor=3)
+    s.set_objective("score", "max")
+    # 4 trials -> all land in DIFFERENT brackets; add 4 more so bracket 0
+    # (grace 1) holds t0, t4, t8, t12 as peers
+    tids = [f"t{i}" for i in range(16)]
+    for tid in tids:
+        assert s.on_trial_result(tid, {"score": 0.0,
+                                       "training_iteration": 0}) == CONTINUE
+    # good peers of bracket 0 hit rung 1 first with high scores
+    for tid, v in [("t4", 10.0), ("t8", 11.0), ("t12", 12.0)]:
+        assert s.on_trial_result(tid, {"score": v,
+                                       "training_iteration": 1}) == CONTINUE
+    # the straggler with a terrible score is cut at the rung
+    assert s.on_trial_result("t0", {"score": 0.1,
+                                    "training_iteration": 1}) == STOP
+    # a late GOOD trial in the same bracket survives... (new peer set)
+    assert s.on_trial_result("t4", {"score": 99.0,
+                                    "training_iteration": 3}) == CONTINUE
+    # and everything stops at max_t
+    assert s.on_trial_result("t12", {"score": 50.0,
+                                     "training_iteration": 27}) == STOP
+
+
+def test_pb2_gp_exploit_moves_params(ray_mod):
+    from ant_ray_amd.tune.schedulers import PB2, PERTURB
+
+    sched = PB2(metric="score", mode="max", perturbation_interval=1,
+                hyperparam_bounds={"lr": (0.001, 0.1)}, seed=0)
+    sched.set_objective("score", "max")
+    for i, tid in enumerate(["a", "b", "c", "d"]):
+        sched.on_trial_start(tid, {"lr": 0.01 * (i + 1)})
+        sched.on_checkpoint(tid, f"/tmp/ck_{tid}")
+    decisions = {}
+    for rep in range(3):
+        for i, tid in enumerate(["a", "b", "c", "d"]):
+            decisions[tid] = sched.on_trial_result(
+                tid, {"score": float(i) + rep * 0.1,
+                      "training_iteration": rep + 1})
+    assert decisions["a"] == PERTURB  # worst trial exploits
+    out = sched.exploit("a")
+    assert out is not None
+    path, cfg = out
+    assert 0.001 <= cfg["lr"] <= 0.1
+    assert path.startswith("/tmp/ck_")
