@@ -37,6 +37,11 @@ class NativeLLMEngine:
         self.max_seq = max_seq
         self.vocab = self.model.cfg.vocab
         self._lock = threading.Lock()
+        # hash-based block prefix cache (vLLM-parity; see prefix_cache.py):
+        # shared prompt prefixes skip their prefill FLOPs
+        from ant_ray_amd.llm.prefix_cache import prefix_cache_from_env
+
+        self.prefix_cache = prefix_cache_from_env()
         # persistent per-batch-bucket graphed decoders (greedy + CUDA +
         # D=128): KV cache and hipGraph capture are paid once per bucket,
         # then every token of every request is one graph replay
@@ -80,17 +85,51 @@ class NativeLLMEngine:
                          and str(self.device).startswith("cuda")
                          and self.model.cfg.head_dim == 128)
             for plen, idxs in by_len.items():
-                toks = torch.tensor([prompts[i] for i in idxs],
-                                    dtype=torch.long, device=self.device)
+                group = [prompts[i] for i in idxs]
+                toks = torch.tensor(group, dtype=torch.long,
+                                    device=self.device)
+                # prefix-cache lookup on the group's COMMON prefix (serve
+                # batches typically share a system prompt); keep >=1 token
+                # to prefill
+                hit, blocks, common = 0, [], plen
+                if self.prefix_cache is not None:
+                    for c, col in enumerate(zip(*group)):
+                        if any(t != col[0] for t in col):
+                            common = c
+                            break
+                    hit, blocks = self.prefix_cache.lookup(
+                        group[0][: min(common, plen - 1)])
+                cache = None
                 if use_graph:
                     dec, bucket = self._graphed_decoder(len(idxs))
                     if len(idxs) < bucket:  # pad rows replay row 0
                         toks = torch.cat(
                             [toks, toks[:1].expand(bucket - len(idxs), -1)])
-                    gen = dec.generate(toks, max_new_tokens)
+                    if hit:
+                        self.prefix_cache.seed(dec.cache, blocks)
+                    gen = dec.generate(toks, max_new_tokens, prefix_len=hit)
+                    cache = dec.cache
                 else:
-                    gen = self.model.generate(toks, max_new_tokens,
-                                              temperature=temperature)
+                    from ant_ray_amd.models.llama import (KVCache,
+                                                          LlamaForCausalLM)
+
+                    if isinstance(self.model, LlamaForCausalLM):
+                        cache = KVCache(
+                            self.model.cfg, toks.shape[0],
+                            min(self.max_seq, plen + max_new_tokens),
+                            self.device)
+                        if hit:
+                            self.prefix_cache.seed(cache, blocks)
+                        gen = self.model.generate(
+                            toks, max_new_tokens, cache=cache,
+                            temperature=temperature, start_pos=hit)
+                    else:  # model family without a llama KVCache
+                        gen = self.model.generate(toks, max_new_tokens,
+                                                  temperature=temperature)
+                if self.prefix_cache is not None and cache is not None:
+                    # store the group's common prompt prefix for reuse
+                    self.prefix_cache.insert(group[0], cache, row=0,
+                                             upto=common)
                 new = gen[: len(idxs), plen:].tolist()
                 for j, i in enumerate(idxs):
                     out[i] = new[j]

@@ -94,10 +94,14 @@ class GraphedDecoder:
         self.cur.copy_(lg.argmax(dim=-1, keepdim=True))
 
     @torch.no_grad()
-    def generate(self, tokens, max_new_tokens: int):
+    def generate(self, tokens, max_new_tokens: int, prefix_len: int = 0):
         B, S = tokens.shape
         assert B == self.B, "decoder is fixed-batch; pad or re-bucket"
-        logits = self.model.forward(tokens, cache=self.cache, pos=0)
+        # prefix_len > 0: cache[0:prefix_len) pre-seeded by the caller
+        # (prefix-cache hit) — prefill only the suffix at pos=prefix_len
+        logits = self.model.forward(
+            tokens[:, prefix_len:] if prefix_len else tokens,
+            cache=self.cache, pos=prefix_len)
         self.cache.pos = S
         self.lens.fill_(S)
         self.cur.copy_(logits.argmax(dim=-1, keepdim=True))
@@ -180,7 +184,21 @@ class LlamaAttention(nn.Module):
                 o = ops.attention_decode(q.reshape(B, Hq, D), ck, cv,
                                          seq_len=pos + 1)
                 return self.wo(o.view(B, 1, Hq * D))
-            assert pos == 0, "chunked prefill not supported"
+            if pos > 0:
+                # chunked prefill (prefix-cache hit): queries at absolute
+                # positions pos..pos+S-1 attend over cache[0:pos+S] with a
+                # bottom-right-aligned causal mask (query i sees keys
+                # j <= pos+i). One-shot per request, so masked SDPA is
+                # fine here; the per-token decode loop stays on the
+                # flash-decode kernel.
+                kf = ck[:, :, : pos + S]
+                vf = cv[:, :, : pos + S]
+                mask = torch.ones(S, pos + S, dtype=torch.bool,
+                                  device=q.device).tril_(diagonal=pos)
+                o = F.scaled_dot_product_attention(q, kf, vf, attn_mask=mask,
+                                                   enable_gqa=True)
+                o = o.transpose(1, 2).reshape(B, S, Hq * D)
+                return self.wo(o)
         if qkv.is_cuda and D == 128 and os.environ.get("ANTRAY_FLASH", "1") != "0":
             # hand-written CDNA4 flash attention — THE DEFAULT (fwd 1.30x
             # AOTriton at 514 TF/s, f+b 1.01x after the r2 instruction-diet
@@ -285,10 +303,14 @@ class LlamaForCausalLM(nn.Module):
 
     @torch.no_grad()
     def generate(self, tokens, max_new_tokens: int, cache: "KVCache" = None,
-                 temperature: float = 0.0):
+                 temperature: float = 0.0, start_pos: int = 0):
         """Greedy (temperature=0) or sampled generation with the KV cache +
         flash-decode kernel. tokens: [B, S_prompt] int64. Returns
         [B, S_prompt + max_new_tokens].
+
+        start_pos > 0 = prefix-cache hit: positions [0, start_pos) of
+        `cache` were seeded by the caller (llm/prefix_cache.py) and only
+        tokens[:, start_pos:] are prefilled (chunked prefill).
 
         Role parity: the reference serves generation through vLLM
         (reference python/ray/llm/_internal/serve/engines/vllm/
@@ -299,7 +321,8 @@ class LlamaForCausalLM(nn.Module):
         if cache is None:
             cache = KVCache(self.cfg, B, min(self.cfg.max_seq,
                                              S + max_new_tokens), dev)
-        logits = self.forward(tokens, cache=cache, pos=0)  # prefill
+        logits = self.forward(tokens[:, start_pos:] if start_pos else tokens,
+                              cache=cache, pos=start_pos)  # prefill
         cache.pos = S
         use_graph = (tokens.is_cuda and self.cfg.head_dim == 128
                      and temperature == 0
