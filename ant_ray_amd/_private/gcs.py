@@ -667,7 +667,9 @@ class GcsServer:
 
         if not hasattr(self, "cluster_events"):
             self.cluster_events = collections.deque(maxlen=10000)
+        self._event_seq = getattr(self, "_event_seq", 0) + 1
         self.cluster_events.append({
+            "seq": self._event_seq,    # export-pipeline high-water mark
             "timestamp": time.time(),
             "source": source,          # NODE | ACTOR | JOB | WORKER
             "event": event,            # REGISTERED | ALIVE | DEAD | ...
@@ -1032,7 +1034,10 @@ class GcsServer:
 
         if not hasattr(self, "task_events"):
             self.task_events = collections.deque(maxlen=20000)
-        self.task_events.extend(p["events"])
+        for ev in p["events"]:
+            self._event_seq = getattr(self, "_event_seq", 0) + 1
+            ev["seq"] = self._event_seq
+            self.task_events.append(ev)
         return {"ok": True}
 
     async def rpc_list_task_events(self, conn, p):

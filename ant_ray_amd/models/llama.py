@@ -186,17 +186,19 @@ class LlamaAttention(nn.Module):
                 return self.wo(o.view(B, 1, Hq * D))
             if pos > 0:
                 # chunked prefill (prefix-cache hit): queries at absolute
-                # positions pos..pos+S-1 attend over cache[0:pos+S] with a
-                # bottom-right-aligned causal mask (query i sees keys
-                # j <= pos+i). One-shot per request, so masked SDPA is
-                # fine here; the per-token decode loop stays on the
-                # flash-decode kernel.
-                kf = ck[:, :, : pos + S]
-                vf = cv[:, :, : pos + S]
-                mask = torch.ones(S, pos + S, dtype=torch.bool,
-                                  device=q.device).tril_(diagonal=pos)
-                o = F.scaled_dot_product_attention(q, kf, vf, attn_mask=mask,
-                                                   enable_gqa=True)
+                # positions pos..pos+S-1 attend over cache[0:pos+S]
+                # (prefix unmasked + suffix causal). GPU: suffix flash +
+                # prefix GEMM/logsumexp, LSE-combined; CPU: masked SDPA.
+                if qkv.is_cuda and D == 128 and os.environ.get(
+                        "ANTRAY_FLASH", "1") != "0":
+                    o = ops.chunked_prefill_attention(q, k, v, ck, cv, pos)
+                else:
+                    kf = ck[:, :, : pos + S]
+                    vf = cv[:, :, : pos + S]
+                    mask = torch.ones(S, pos + S, dtype=torch.bool,
+                                      device=q.device).tril_(diagonal=pos)
+                    o = F.scaled_dot_product_attention(
+                        q, kf, vf, attn_mask=mask, enable_gqa=True)
                 o = o.transpose(1, 2).reshape(B, S, Hq * D)
                 return self.wo(o)
         if qkv.is_cuda and D == 128 and os.environ.get("ANTRAY_FLASH", "1") != "0":

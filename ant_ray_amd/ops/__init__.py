@@ -15,6 +15,7 @@ from ant_ray_amd.ops.functional import (  # noqa: F401
     attention,
     attention_decode,
     cast_affine,
+    chunked_prefill_attention,
     decode_step_attn,
     fused_add_rmsnorm,
     nhwc_to_nchw,

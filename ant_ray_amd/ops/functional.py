@@ -273,6 +273,58 @@ def attention_decode(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return ref.attention_decode_ref(q, kc, vc, lens=lens, scale=float(scale))
 
 
+def chunked_prefill_attention(q: torch.Tensor, k_new: torch.Tensor,
+                              v_new: torch.Tensor, ck: torch.Tensor,
+                              cv: torch.Tensor, pos: int,
+                              scale=None) -> torch.Tensor:
+    """Attention for CHUNKED PREFILL (prefix-cache hit): suffix queries
+    attend over `pos` cached prefix positions (unmasked) plus the suffix
+    itself (causal). GPU: square causal flash on the suffix (need_lse) +
+    a bf16-GEMM / fp32-logsumexp pass over the prefix, combined by LSE
+    weights — the flash-decode two-part combine, at prefill granularity.
+    (A masked SDPA here falls back to the math backend and forfeits the
+    prefix-cache win; this path keeps the GEMMs on hipBLASLt.)
+
+    q: [B,Hq,S,D]; k_new/v_new: [B,Hk,S,D] rope'd suffix keys/values
+    (also already written to the cache at [pos, pos+S)); ck/cv:
+    [B,Hk,Tmax,D]. Returns o [B,Hq,S,D]. Inference only.
+    """
+    B, Hq, S, D = q.shape
+    Hk = ck.shape[1]
+    G = Hq // Hk
+    if scale is None:
+        scale = D ** -0.5
+    if not _use_hip(q):
+        # exact fallback: masked SDPA over prefix+suffix
+        import torch.nn.functional as F
+
+        kf = ck[:, :, : pos + S]
+        vf = cv[:, :, : pos + S]
+        mask = torch.ones(S, pos + S, dtype=torch.bool,
+                          device=q.device).tril_(diagonal=pos)
+        return F.scaled_dot_product_attention(q, kf, vf, attn_mask=mask,
+                                              enable_gqa=True)
+    oB, lseB = _hip().attn_fwd(q, k_new, v_new, float(scale), True, True)
+    kp = ck[:, :, :pos]
+    vp = cv[:, :, :pos]
+    # prefix part: grouped bf16 GEMMs + fp32 running-softmax stats
+    qg = q.reshape(B, Hk, G * S, D)  # q heads are kv-head-major
+    sp = torch.matmul(qg, kp.transpose(-1, -2)).float() * float(scale)
+    mA = sp.amax(-1, keepdim=True)                    # [B,Hk,G*S,1]
+    eA = torch.exp(sp - mA)
+    sA = eA.sum(-1)                                   # [B,Hk,G*S]
+    oA = torch.matmul(eA.to(q.dtype), vp).float() / sA.unsqueeze(-1)
+    lA = mA.squeeze(-1) + torch.log(sA)               # natural-log LSE
+    # suffix flash LSE is log2-domain
+    lB = lseB.view(B, Hk, G * S) * 0.6931471805599453
+    m = torch.maximum(lA, lB)
+    wA = torch.exp(lA - m).unsqueeze(-1)
+    wB = torch.exp(lB - m).unsqueeze(-1)
+    ob = oB.reshape(B, Hk, G * S, D).float()
+    o = (oA * wA + ob * wB) / (wA + wB)
+    return o.view(B, Hq, S, D).to(q.dtype)
+
+
 def decode_step_attn(qkv: torch.Tensor, ck: torch.Tensor, cv: torch.Tensor,
                      lens: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
                      hq: int, hk: int, scale=None) -> torch.Tensor:
