@@ -71,7 +71,11 @@ class ExportEventAggregator:
         self.out_dir = out_dir
         self.http_target = http_target
         self.period = period
-        self._last_seq = 0
+        # per-ring high-water marks: the rings share one GCS seq counter,
+        # so a single mark would skip a task event whose seq is lower
+        # than a cluster event drained in the same pass
+        self._last_cluster_seq = 0
+        self._last_task_seq = 0
         self._files: Dict[str, object] = {}
         self._http_buf: List[dict] = []
         self._max_buffer = max_buffer
@@ -85,13 +89,13 @@ class ExportEventAggregator:
     def _collect_new(self) -> List[dict]:
         out = []
         for ev in getattr(self.gcs, "cluster_events", []):
-            if ev.get("seq", 0) > self._last_seq:
+            if ev.get("seq", 0) > self._last_cluster_seq:
                 out.append(_normalize_cluster_event(ev))
-                self._last_seq = max(self._last_seq, ev["seq"])
+                self._last_cluster_seq = ev["seq"]
         for ev in getattr(self.gcs, "task_events", []):
-            if ev.get("seq", 0) > self._last_seq:
+            if ev.get("seq", 0) > self._last_task_seq:
                 out.append(_normalize_task_event(ev))
-                self._last_seq = max(self._last_seq, ev["seq"])
+                self._last_task_seq = ev["seq"]
         out.sort(key=lambda r: r["timestamp"] or 0)
         return out
 
@@ -170,7 +174,8 @@ class ExportEventAggregator:
     def stats(self) -> dict:
         return {"written": self.written, "published": self.published,
                 "dropped": self.dropped, "buffered": len(self._http_buf),
-                "last_seq": self._last_seq}
+                "last_cluster_seq": self._last_cluster_seq,
+                "last_task_seq": self._last_task_seq}
 
 
 def maybe_start(gcs) -> Optional[ExportEventAggregator]:

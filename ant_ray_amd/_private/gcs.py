@@ -168,6 +168,10 @@ class GcsServer:
                 logger.exception("GCS table restore failed; starting fresh")
         self._server, self.port = await protocol.serve(self._handle, host, port)
         asyncio.get_running_loop().create_task(self._health_loop())
+        # export-event pipeline (file sink / HTTP aggregator), env-gated
+        from ant_ray_amd._private import event_export
+
+        self._event_exporter = event_export.maybe_start(self)
         if self.actors:
             asyncio.get_running_loop().create_task(
                 self._verify_restored_actors())
