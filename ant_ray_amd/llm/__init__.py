@@ -74,6 +74,10 @@ def build_llm_deployment(config: LLMConfig):
             num_gpus=config.num_gpus_per_replica(),
             max_seq=int(config.engine_kwargs.get("max_model_len", 4096)),
             max_batch_size=int(config.engine_kwargs.get("max_num_seqs", 16)),
+            # vLLM serves with continuous batching unconditionally; the
+            # native engine keeps it opt-in (greedy-only token scheduler)
+            continuous=bool(config.engine_kwargs.get(
+                "continuous_batching", False)),
         )
     vllm = _require_vllm()
     from ant_ray_amd import serve

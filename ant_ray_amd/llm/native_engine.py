@@ -141,13 +141,46 @@ def build_native_llm_deployment(model_name: str = "llama3-8b",
                                 num_gpus: float = 1,
                                 max_seq: int = 4096,
                                 max_batch_size: int = 16,
-                                batch_wait_timeout_s: float = 0.02):
-    """Serve deployment serving the native engine with dynamic batching.
+                                batch_wait_timeout_s: float = 0.02,
+                                continuous: bool = False):
+    """Serve deployment serving the native engine.
+
+    continuous=False: dynamic request batching (@serve.batch groups by
+    prompt length). continuous=True: token-level continuous batching —
+    each request joins the running decode batch immediately
+    (llm/continuous.py ContinuousLLMEngine; greedy only).
 
     Request payload: {"prompt_ids": [int], "max_new_tokens": int,
     "temperature": float} -> {"token_ids": [int]}.
     """
     from ant_ray_amd import serve
+
+    if continuous:
+        @serve.deployment(
+            num_replicas=num_replicas,
+            ray_actor_options={"num_gpus": num_gpus},
+            max_ongoing_requests=max(64, 4 * max_batch_size),
+        )
+        class NativeLLMServerCB:
+            def __init__(self):
+                import asyncio  # noqa: F401
+
+                from ant_ray_amd.llm.continuous import ContinuousLLMEngine
+
+                self.engine = ContinuousLLMEngine(
+                    model_name, slots=max_batch_size, max_seq=max_seq,
+                    device="cuda" if num_gpus > 0 else "cpu",
+                    start_thread=True)
+
+            async def __call__(self, request: dict) -> dict:
+                import asyncio
+
+                fut = self.engine.submit(
+                    request["prompt_ids"],
+                    int(request.get("max_new_tokens", 32)))
+                return {"token_ids": await asyncio.wrap_future(fut)}
+
+        return NativeLLMServerCB.bind()
 
     @serve.deployment(
         num_replicas=num_replicas,

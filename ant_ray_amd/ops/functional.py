@@ -334,12 +334,35 @@ def decode_step_attn(qkv: torch.Tensor, ck: torch.Tensor, cv: torch.Tensor,
     the step index, so models/llama.py captures the whole token step in
     a hipGraph and replays it. qkv: [B,1,(Hq+2Hk)*D] from the fused
     projection; caches [B,Hk,Tmax,D]; lens int32 [B] (valid length
-    INCLUDING the new token). GPU-only (generation path)."""
+    INCLUDING the new token). CPU fallback: exact fp32 reference (so the
+    ragged/continuous decode path is testable without a GPU)."""
     D = ck.shape[-1]
     if scale is None:
         scale = D ** -0.5
-    return _hip().decode_step_attn(qkv, ck, cv, lens, cos, sin, hq, hk,
-                                   float(scale))
+    if _use_hip(qkv):
+        return _hip().decode_step_attn(qkv, ck, cv, lens, cos, sin, hq, hk,
+                                       float(scale))
+    B = qkv.shape[0]
+    flat = qkv.reshape(B, -1)
+    q = flat[:, : hq * D].view(B, hq, D)
+    k = flat[:, hq * D : (hq + hk) * D].view(B, hk, D)
+    v = flat[:, (hq + hk) * D :].view(B, hk, D)
+    pos = (lens.long() - 1).clamp(min=0)              # per-row position
+    half = D // 2
+    c = cos[pos].view(B, 1, half).float()
+    s = sin[pos].view(B, 1, half).float()
+
+    def rot(x):
+        xf = x.float()
+        x1, x2 = xf[..., :half], xf[..., half:]
+        return torch.cat([x1 * c - x2 * s, x1 * s + x2 * c], -1).to(x.dtype)
+
+    q, k = rot(q), rot(k)
+    rows = torch.arange(B, device=qkv.device)
+    ck[rows, :, pos] = k
+    cv[rows, :, pos] = v
+    return ref.attention_decode_ref(q, ck, cv, lens=lens,
+                                    scale=float(scale))
 
 
 # ------------------------------------------------------- data transforms

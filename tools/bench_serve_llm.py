@@ -27,6 +27,11 @@ def main():
     ap.add_argument("--prompt", type=int, default=128)
     ap.add_argument("--new-tokens", type=int, default=32)
     ap.add_argument("--out", default="")
+    ap.add_argument("--continuous", action="store_true",
+                    help="token-level continuous batching engine")
+    ap.add_argument("--ragged", action="store_true",
+                    help="sample prompt lengths in [16, --prompt] instead "
+                         "of fixed (exposes batching-policy differences)")
     args = ap.parse_args()
 
     import ant_ray_amd as ray
@@ -37,7 +42,8 @@ def main():
     app = build_llm_deployment(LLMConfig(
         model_loading_config={"model_id": args.model},
         engine_kwargs={"max_model_len": args.prompt + args.new_tokens + 8,
-                       "max_num_seqs": args.concurrency},
+                       "max_num_seqs": args.concurrency,
+                       "continuous_batching": args.continuous},
         deployment_config={"num_replicas": args.replicas},
     ))
     h = serve.run(app, name="llm", route_prefix="/llm")
@@ -46,8 +52,9 @@ def main():
     rng = random.Random(0)
 
     def req_payload():
-        return {"prompt_ids": [rng.randrange(vocab)
-                               for _ in range(args.prompt)],
+        plen = rng.randrange(16, args.prompt + 1) if args.ragged \
+            else args.prompt
+        return {"prompt_ids": [rng.randrange(vocab) for _ in range(plen)],
                 "max_new_tokens": args.new_tokens}
 
     # warm (model build + first kernels)
@@ -101,6 +108,8 @@ def main():
         "p95_s": round(lat[int(n * 0.95)] if n > 1 else lat[0], 3) if n else None,
         "wall_s": round(wall, 2),
         "completed": n,
+        "continuous_batching": args.continuous,
+        "ragged_prompts": args.ragged,
     }
     print(json.dumps(result), flush=True)
     if args.out:
