@@ -75,13 +75,22 @@ class SlotDecoder:
         self.graph = None
         self._eager_steps = 0
 
-    def prefill(self, row: int, tokens: List[int]) -> None:
+    def prefill(self, row: int, tokens: List[int],
+                prefix_cache=None) -> None:
         toks = torch.tensor([tokens], dtype=torch.long, device=self.device)
         rv = _RowCache(self.cache, row)
+        hit = 0
+        if prefix_cache is not None:
+            hit, blocks = prefix_cache.lookup(tokens[: len(tokens) - 1])
+            if hit:
+                prefix_cache.seed_row(self.cache, blocks, row)
         with torch.no_grad():
-            logits = self.model.forward(toks, cache=rv, pos=0)
+            logits = self.model.forward(
+                toks[:, hit:] if hit else toks, cache=rv, pos=hit)
         self.cur[row] = logits.argmax(-1)
         self.lens[row] = len(tokens)
+        if prefix_cache is not None:
+            prefix_cache.insert(tokens, self.cache, row=row)
 
     def _step(self):
         lg = self.model.forward(self.cur, cache=self.cache, lens=self.lens)
@@ -133,6 +142,9 @@ class ContinuousLLMEngine:
         self.device = device
         self.max_seq = max_seq
         self.dec = SlotDecoder(self.model, slots, max_seq, device)
+        from ant_ray_amd.llm.prefix_cache import prefix_cache_from_env
+
+        self.prefix_cache = prefix_cache_from_env()
         self._ids = itertools.count()
         self._lock = threading.Lock()
         self._queue: List[_Request] = []
@@ -183,7 +195,7 @@ class ContinuousLLMEngine:
                     return
                 req = self._queue.pop(0)
             row = free[0]
-            self.dec.prefill(row, req.prompt)
+            self.dec.prefill(row, req.prompt, self.prefix_cache)
             req.row = row
             req.start = self._trace_base + len(self._trace)
             req.emitted = 0

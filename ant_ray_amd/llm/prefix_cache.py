@@ -124,6 +124,13 @@ class PrefixKVCache:
             cache.k[:, :, :, sl] = ent.k.unsqueeze(1)
             cache.v[:, :, :, sl] = ent.v.unsqueeze(1)
 
+    def seed_row(self, cache, blocks: List[_Block], row: int) -> None:
+        """Copy hit blocks into ONE row (continuous-batching slots)."""
+        for i, ent in enumerate(blocks):
+            sl = slice(i * self.bs, (i + 1) * self.bs)
+            cache.k[:, row, :, sl] = ent.k
+            cache.v[:, row, :, sl] = ent.v
+
     def stats(self) -> dict:
         return {"blocks": len(self._blocks), "bytes": self._bytes,
                 "hits": self.hits, "misses": self.misses,

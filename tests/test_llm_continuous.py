@@ -78,6 +78,27 @@ def test_continuous_background_thread():
         eng.shutdown()
 
 
+def test_continuous_with_prefix_cache(monkeypatch):
+    """Requests sharing a system prompt reuse its KV blocks across slot
+    admissions; outputs stay exact."""
+    monkeypatch.setenv("ANTRAY_PREFIX_CACHE", "1")
+    monkeypatch.setenv("ANTRAY_PREFIX_BLOCK", "8")
+    eng = ContinuousLLMEngine("llama-tiny", slots=2, max_seq=128,
+                              device="cpu")
+    assert eng.prefix_cache is not None
+    import random
+
+    rng = random.Random(3)
+    sysp = [rng.randrange(256) for _ in range(24)]
+    prompts = [sysp + [rng.randrange(256) for _ in range(4)]
+               for _ in range(4)]
+    futs = [eng.submit(p, 5) for p in prompts]
+    eng.run_until_idle()
+    assert eng.prefix_cache.stats()["tokens_reused"] >= 24 * 2
+    for p, f in zip(prompts, futs):
+        assert f.result(timeout=0) == _ref_tokens(eng.model, p, 5)
+
+
 def test_continuous_rejects_oversized():
     eng = ContinuousLLMEngine("llama-tiny", slots=1, max_seq=32,
                               device="cpu")
