@@ -49,6 +49,15 @@ class _Request:
     row: int = -1
     start: int = -1        # trace index of the first emitted token
     emitted: int = 0
+    stop_ids: Optional[frozenset] = None  # terminate early on these
+    on_token: Optional[object] = None     # streaming callback(token)
+    # host-side tokens, filled only when stop_ids/on_token force a
+    # per-tick readback for this request
+    tokens: List[int] = field(default_factory=list)
+
+    @property
+    def eager_host(self) -> bool:
+        return self.stop_ids is not None or self.on_token is not None
 
 
 class SlotDecoder:
@@ -173,11 +182,20 @@ class ContinuousLLMEngine:
             self._thread.start()
 
     # ------------------------------------------------------------ submit
-    def submit(self, prompt_ids: List[int], max_new_tokens: int) -> Future:
+    def submit(self, prompt_ids: List[int], max_new_tokens: int,
+               stop_token_ids=None, on_token=None) -> Future:
+        """stop_token_ids: terminate the request early when one is
+        produced (the stop token is included in the output). on_token:
+        per-token streaming callback, called from the pump thread.
+        Either option switches the request to per-tick host readback
+        (one batched D2H per step while any such request is active)."""
         if len(prompt_ids) + max_new_tokens > self.max_seq:
             raise ValueError("prompt + max_new_tokens exceeds max_seq")
         req = _Request(next(self._ids), list(prompt_ids),
-                       int(max_new_tokens))
+                       int(max_new_tokens),
+                       stop_ids=(frozenset(stop_token_ids)
+                                 if stop_token_ids else None),
+                       on_token=on_token)
         with self._lock:
             self._queue.append(req)
         self._wake.set()
@@ -203,10 +221,12 @@ class ContinuousLLMEngine:
 
     def _retire(self, req: _Request) -> None:
         del self._active[req.row]
-        lo = req.start - self._trace_base
-        cols = torch.cat(self._trace[lo : lo + req.max_new], dim=1)
-        toks = cols[req.row].tolist()
-        req.future.set_result(toks)
+        if len(req.tokens) == req.emitted:  # host copy already complete
+            req.future.set_result(list(req.tokens))
+        else:
+            lo = req.start - self._trace_base
+            cols = torch.cat(self._trace[lo : lo + req.emitted], dim=1)
+            req.future.set_result(cols[req.row].tolist())
         # prune trace entries no active request still needs
         if self._active:
             keep = min(r.start for r in self._active.values())
@@ -224,9 +244,24 @@ class ContinuousLLMEngine:
         if not self._active:
             return False
         self._trace.append(self.dec.cur.clone())
+        host_cur = None
+        if any(r.eager_host for r in self._active.values()):
+            # one batched D2H for every streaming/stop-scanning request
+            host_cur = self._trace[-1].squeeze(1).tolist()
         for req in list(self._active.values()):
             req.emitted += 1
-            if req.emitted >= req.max_new:
+            stop_hit = False
+            if req.eager_host:
+                tok = host_cur[req.row]
+                req.tokens.append(tok)
+                if req.on_token is not None:
+                    try:
+                        req.on_token(tok)
+                    except Exception:
+                        pass
+                stop_hit = (req.stop_ids is not None
+                            and tok in req.stop_ids)
+            if stop_hit or req.emitted >= req.max_new:
                 self._retire(req)
         # freed slots admit at the NEXT tick's _admit: a row admitted
         # here would lose its first token (step() advances cur before

@@ -177,7 +177,8 @@ def build_native_llm_deployment(model_name: str = "llama3-8b",
 
                 fut = self.engine.submit(
                     request["prompt_ids"],
-                    int(request.get("max_new_tokens", 32)))
+                    int(request.get("max_new_tokens", 32)),
+                    stop_token_ids=request.get("stop_token_ids"))
                 return {"token_ids": await asyncio.wrap_future(fut)}
 
         return NativeLLMServerCB.bind()

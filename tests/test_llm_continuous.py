@@ -99,6 +99,28 @@ def test_continuous_with_prefix_cache(monkeypatch):
         assert f.result(timeout=0) == _ref_tokens(eng.model, p, 5)
 
 
+def test_continuous_stop_tokens_and_streaming():
+    """stop_token_ids ends a request early (stop token included);
+    on_token streams every emitted token; a plain request sharing the
+    batch is unaffected."""
+    eng = ContinuousLLMEngine("llama-tiny", slots=2, max_seq=128,
+                              device="cpu")
+    p = [21, 22, 23, 24]
+    ref = _ref_tokens(eng.model, p, 10)
+    stop = ref[3]
+    streamed = []
+    f = eng.submit(p, 10, stop_token_ids=[stop],
+                   on_token=streamed.append)
+    p2 = [9, 9, 9]
+    f2 = eng.submit(p2, 6)
+    eng.run_until_idle()
+    out = f.result(timeout=0)
+    k = ref.index(stop) + 1
+    assert out == ref[:k], (out, ref)
+    assert streamed == out
+    assert f2.result(timeout=0) == _ref_tokens(eng.model, p2, 6)
+
+
 def test_continuous_rejects_oversized():
     eng = ContinuousLLMEngine("llama-tiny", slots=1, max_seq=32,
                               device="cpu")
